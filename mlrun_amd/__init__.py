@@ -1,0 +1,108 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""mlrun_amd — an MI355X-native MLOps orchestration & serving framework.
+
+A from-scratch rebuild of the mlrun/mlrun capability surface for a
+single 8xMI355X node: projects/functions/runs/artifacts + a serving
+graph engine whose hot path runs hand-written CDNA4 HIP kernels, a
+GPU-resident feature store, and RCCL-over-xGMI distributed runtimes.
+
+Public API parity: reference mlrun/__init__.py (__all__ :17,
+set_environment :107).
+"""
+
+__version__ = "0.1.0"
+
+__all__ = [
+    "get_version",
+    "set_environment",
+    "code_to_function",
+    "import_function",
+    "new_function",
+    "run_local",
+    "get_or_create_ctx",
+    "new_task",
+    "new_project",
+    "load_project",
+    "get_or_create_project",
+    "run_function",
+    "build_function",
+    "deploy_function",
+    "get_run_db",
+    "mlconf",
+    "handler",
+]
+
+from .config import config as mlconf  # noqa: E402
+from .errors import (  # noqa: F401,E402
+    MLRunBaseError,
+    MLRunInvalidArgumentError,
+    MLRunNotFoundError,
+    MLRunRuntimeError,
+)
+from .model import RunObject, RunTemplate, new_task  # noqa: F401,E402
+from .execution import MLClientCtx  # noqa: F401,E402
+from .db import get_run_db  # noqa: F401,E402
+from .datastore import DataItem, store_manager, get_dataitem, get_object  # noqa: F401,E402
+from .run import (  # noqa: F401,E402
+    code_to_function,
+    function_to_module,
+    get_or_create_ctx,
+    import_function,
+    new_function,
+    run_local,
+)
+from .projects import (  # noqa: F401,E402
+    MlrunProject,
+    ProjectMetadata,
+    build_function,
+    deploy_function,
+    get_or_create_project,
+    load_project,
+    new_project,
+    run_function,
+)
+from .package import handler, ArtifactType  # noqa: F401,E402
+
+
+def get_version() -> str:
+    return __version__
+
+
+def set_environment(api_path: str = None, artifact_path: str = "",
+                    env_file: str = None, mock_functions: str = None):
+    """Set global configuration: run-DB target + default artifact path.
+
+    Returns (default_project_name, artifact_path) — parity with the
+    reference set_environment.
+    """
+    if env_file:
+        _load_env_file(env_file)
+    if api_path:
+        mlconf.dbpath = api_path
+        from .db import get_run_db as _get
+
+        _get(api_path, force_reconnect=True)
+    if artifact_path:
+        import os
+
+        if not artifact_path.startswith("/") and "://" not in artifact_path:
+            artifact_path = os.path.abspath(artifact_path)
+        mlconf.artifact_path = artifact_path
+    return mlconf.default_project, mlconf.artifact_path
+
+
+def _load_env_file(path: str):
+    import os
+
+    from .utils import list_to_dict
+
+    with open(os.path.expanduser(path)) as fp:
+        for key, value in list_to_dict(fp.readlines()).items():
+            if key and not key.startswith("#"):
+                os.environ[key] = value
+    from .config import _populate
+
+    _populate()

@@ -1,0 +1,54 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Run-DB factory.  ``get_run_db()`` dispatches on config.dbpath:
+
+- ``""`` / ``"local"`` / filesystem path -> node-local SQLite DB
+- ``http(s)://...``                      -> HTTP client to the API service
+- ``"nop"``                              -> NopDB (offline)
+
+Parity: reference mlrun/db/__init__.py get_run_db + run-db singleton.
+"""
+
+import threading
+
+from ..config import config
+from .base import RunDBInterface
+from .httpdb import HTTPRunDB
+from .nopdb import NopDB
+from .sqldb import SQLRunDB
+
+_lock = threading.Lock()
+_run_db = None
+_run_db_url = None
+
+
+def create_run_db(url: str = "", secrets=None) -> RunDBInterface:
+    url = url if url is not None else ""
+    if url == "nop":
+        return NopDB()
+    if url.startswith("http://") or url.startswith("https://"):
+        return HTTPRunDB(url).connect(secrets)
+    return SQLRunDB(url if url not in ("", "local") else "")
+
+
+def get_run_db(url: str = None, secrets=None, force_reconnect=False) -> RunDBInterface:
+    """Return the process-wide run DB (created on first use)."""
+    global _run_db, _run_db_url
+
+    url = url if url is not None else str(config.dbpath or "")
+    with _lock:
+        if _run_db is None or force_reconnect or url != _run_db_url:
+            _run_db = create_run_db(url, secrets)
+            _run_db_url = url
+        return _run_db
+
+
+def set_run_db(db: RunDBInterface):
+    """Install a DB instance (used by tests to inject mocks)."""
+    global _run_db, _run_db_url
+
+    with _lock:
+        _run_db = db
+        _run_db_url = getattr(db, "dsn", getattr(db, "base_url", "injected"))

@@ -1,0 +1,20 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Model monitoring: endpoint records, event stream stats, drift apps.
+
+Parity target: reference mlrun/model_monitoring (stream_processing.py
+EventStreamProcessor, controller, histogram_data_drift).  Implemented
+in mlrun_amd as: per-endpoint event stream -> window stats (count/avg/
+latency percentiles) -> parquet + in-memory TSDB; drift application
+computing TVD/Hellinger/KL vs. the reference histogram.
+"""
+
+from .stream import (  # noqa: F401
+    EventStreamProcessor,
+    ModelMonitoringEvent,
+    get_stream_processor,
+)
+from .drift import histogram_drift_metrics  # noqa: F401
+from .controller import enable_model_monitoring, MonitoringController  # noqa: F401

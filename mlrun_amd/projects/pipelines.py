@@ -1,0 +1,141 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Workflow execution: the local pipeline runner.
+
+The reference compiles workflows to Kubeflow Pipelines
+(projects/pipelines.py:542 _KFPRunner) or runs them locally
+(:673 _LocalRunner).  The MI355X-native build keeps the local runner
+as the only engine: a workflow is a python function using
+mlrun_amd.run_function / project.run_function, executed in-process
+with pipeline_context set.
+"""
+
+import os
+import typing
+
+from ..errors import MLRunInvalidArgumentError, MLRunRuntimeError
+from ..model import RunObject, generate_uid
+from ..utils import logger
+from .project import pipeline_context
+
+
+class _PipelineRunStatus:
+    def __init__(self, run_id, project, workflow_name, state="completed",
+                 runs=None, error=None):
+        self.run_id = run_id
+        self._project = project
+        self.workflow_name = workflow_name
+        self.state = state
+        self.runs = runs or []
+        self.error = error
+
+    def wait_for_completion(self, timeout=None, expected_statuses=None):
+        return self.state
+
+    def __str__(self):
+        return str(self.run_id)
+
+
+class FunctionStep:
+    """A lazily-executed function invocation inside a workflow
+    (returned by function.as_step — parity: reference runtimes/base.py:666)."""
+
+    def __init__(self, function, runspec=None, handler=None, name="",
+                 params=None, inputs=None, outputs=None, artifact_path=""):
+        self.function = function
+        self.runspec = runspec
+        self.handler = handler
+        self.name = name
+        self.params = params
+        self.inputs = inputs
+        self.outputs = outputs or []
+        self.artifact_path = artifact_path
+        self._run: typing.Optional[RunObject] = None
+
+    def run(self) -> RunObject:
+        self._run = self.function.run(
+            self.runspec, handler=self.handler, name=self.name,
+            params=self.params, inputs=self.inputs,
+            artifact_path=self.artifact_path, watch=False)
+        return self._run
+
+    @property
+    def outputs_map(self) -> dict:
+        if self._run is None:
+            self.run()
+        return self._run.outputs
+
+    def output(self, key):
+        if self._run is None:
+            self.run()
+        return self._run.output(key)
+
+    # KFP-compat sugar: step.after(other) -> self (execution is already
+    # sequential in the local runner)
+    def after(self, *steps):
+        return self
+
+
+def load_workflow_module(path: str):
+    from ..runtimes.local import load_module
+
+    return load_module(path)
+
+
+def run_workflow(project, path=None, handler=None, arguments=None,
+                 artifact_path=None, watch=True) -> _PipelineRunStatus:
+    """Execute a workflow python file's handler with pipeline_context
+    bound to the project."""
+    arguments = arguments or {}
+    run_id = generate_uid()[:8]
+    pipeline_context.set(project)
+    error = None
+    try:
+        if callable(handler):
+            workflow_fn = handler
+        else:
+            if not path:
+                raise MLRunInvalidArgumentError(
+                    "workflow needs a path or a callable handler")
+            if not os.path.isabs(path) and project.context:
+                candidate = os.path.join(project.context, path)
+                if os.path.isfile(candidate):
+                    path = candidate
+            module = load_workflow_module(path)
+            fn_name = handler or "pipeline"
+            if not hasattr(module, fn_name):
+                # fall back: first public function
+                candidates = [n for n in dir(module)
+                              if callable(getattr(module, n))
+                              and not n.startswith("_")]
+                if not candidates:
+                    raise MLRunInvalidArgumentError(
+                        f"no workflow handler found in {path}")
+                fn_name = candidates[0]
+            workflow_fn = getattr(module, fn_name)
+        import inspect
+
+        sig = inspect.signature(workflow_fn)
+        kwargs = {}
+        for pname, param in sig.parameters.items():
+            if pname in arguments:
+                kwargs[pname] = arguments[pname]
+            elif pname in ("project",):
+                kwargs[pname] = project
+        workflow_fn(**kwargs)
+        state = "completed"
+    except Exception as exc:
+        logger.error("workflow failed", error=str(exc))
+        state = "error"
+        error = str(exc)
+    runs = list(pipeline_context.runs)
+    pipeline_context.clear()
+    status = _PipelineRunStatus(run_id, project,
+                                getattr(workflow_fn, "__name__", "workflow")
+                                if "workflow_fn" in dir() else "workflow",
+                                state=state, runs=runs, error=error)
+    if state == "error":
+        raise MLRunRuntimeError(f"workflow failed: {error}")
+    return status

@@ -1,0 +1,86 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Runtime kinds registry.
+
+The reference exposes ~11 runtime kinds (job/mpijob/dask/spark/nuclio/
+serving/...).  The MI355X-native framework maps them to node-local
+engines:
+
+- local / handler  -> in-process execution
+- job              -> managed local process w/ GPU lease
+- mpijob           -> N local ranks (1/GPU) over RCCL/xGMI
+- serving          -> serving-graph engine (HIP kernels + hipGraph)
+- application      -> long-lived local HTTP app
+- remote (nuclio)  -> local HTTP function host
+- dask / spark / databricks kinds from the reference have no MI355X
+  analog (cluster-external engines) — requesting them raises with a
+  pointer to the node-local equivalents (job with parallel_runs /
+  mpijob).
+"""
+
+from ..errors import MLRunInvalidArgumentError
+from .base import BaseRuntime, FunctionSpec, FunctionMetadata  # noqa: F401
+from .local import HandlerRuntime, LocalRuntime  # noqa: F401
+from .job import KubejobRuntime  # noqa: F401
+
+
+class RuntimeKinds:
+    local = "local"
+    handler = "handler"
+    job = "job"
+    mpijob = "mpijob"
+    serving = "serving"
+    remote = "remote"
+    nuclio = "nuclio"
+    application = "application"
+    dask = "dask"
+    spark = "spark"
+    databricks = "databricks"
+
+    @staticmethod
+    def all():
+        return [RuntimeKinds.local, RuntimeKinds.handler, RuntimeKinds.job,
+                RuntimeKinds.mpijob, RuntimeKinds.serving,
+                RuntimeKinds.remote, RuntimeKinds.nuclio,
+                RuntimeKinds.application]
+
+    @staticmethod
+    def runtime_with_handlers():
+        return [RuntimeKinds.local, RuntimeKinds.handler, RuntimeKinds.job,
+                RuntimeKinds.mpijob]
+
+    @staticmethod
+    def local_runtimes():
+        return [RuntimeKinds.local, RuntimeKinds.handler]
+
+
+def get_runtime_class(kind: str):
+    if kind in (None, "", "local"):
+        return LocalRuntime
+    if kind == RuntimeKinds.handler:
+        return HandlerRuntime
+    if kind == RuntimeKinds.job:
+        return KubejobRuntime
+    if kind == RuntimeKinds.mpijob:
+        from .mpijob import MpiRuntime
+
+        return MpiRuntime
+    if kind in (RuntimeKinds.serving,):
+        from .serving import ServingRuntime
+
+        return ServingRuntime
+    if kind in (RuntimeKinds.remote, RuntimeKinds.nuclio,
+                RuntimeKinds.application):
+        from .remote import RemoteRuntime, ApplicationRuntime
+
+        return ApplicationRuntime if kind == RuntimeKinds.application \
+            else RemoteRuntime
+    if kind in (RuntimeKinds.dask, RuntimeKinds.spark,
+                RuntimeKinds.databricks):
+        raise MLRunInvalidArgumentError(
+            f"runtime kind {kind!r} has no MI355X node-local engine; use "
+            f"kind='job' with hyper_param_options.parallel_runs for task "
+            f"fan-out or kind='mpijob' for distributed (RCCL/xGMI) runs")
+    raise MLRunInvalidArgumentError(f"unsupported runtime kind {kind!r}")

@@ -1,0 +1,244 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Core spine tests: config, model objects, DB, datastore, artifacts."""
+
+import json
+import os
+
+import pytest
+
+from mlrun_amd.config import Config, read_env
+from mlrun_amd.errors import MLRunNotFoundError
+from mlrun_amd.model import (
+    RunObject,
+    RunSpec,
+    RunStates,
+    RunTemplate,
+    new_task,
+)
+
+
+class TestConfig:
+    def test_defaults(self, config_test_base):
+        cfg = config_test_base
+        assert cfg.default_project == "default"
+        assert cfg.gpu.arch == "gfx950"
+        assert int(cfg.gpu.devices_per_node) == 8
+
+    def test_env_override(self):
+        overrides = read_env({"MLRUN_HTTPDB__PORT": "9999",
+                              "MLRUN_LOG_LEVEL": '"DEBUG"',
+                              "MLRUN_DBPATH": "http://localhost:8080"})
+        assert overrides["httpdb"]["port"] == 9999
+        assert overrides["log_level"] == "DEBUG"
+        assert overrides["dbpath"] == "http://localhost:8080"
+
+    def test_nested_set(self):
+        cfg = Config({"a": {"b": 1}})
+        assert cfg.a.b == 1
+        cfg.update({"a": {"c": 2}})
+        assert cfg.a.b == 1 and cfg.a.c == 2
+
+
+class TestModelObjects:
+    def test_task_roundtrip(self):
+        task = new_task(name="t1", project="p1", params={"x": 5},
+                        inputs={"data": "/tmp/x.csv"})
+        struct = task.to_dict()
+        back = RunTemplate.from_dict(struct)
+        assert back.metadata.name == "t1"
+        assert back.spec.parameters == {"x": 5}
+        assert back.spec.inputs == {"data": "/tmp/x.csv"}
+
+    def test_run_object_outputs(self):
+        run = RunObject.from_template(new_task(name="r"))
+        run.status.results = {"acc": 0.9}
+        run.status.artifacts = [
+            {"kind": "model", "metadata": {"key": "m", "project": "p",
+                                           "tree": "abc"},
+             "spec": {"target_path": "/tmp/m"}}]
+        assert run.output("acc") == 0.9
+        assert run.output("m").startswith("store://")
+        outputs = run.outputs
+        assert set(outputs) == {"acc", "m"}
+
+    def test_hyper_params(self):
+        task = new_task(name="h").with_hyper_params(
+            {"p1": [1, 2]}, selector="max.acc", strategy="grid")
+        assert task.spec.hyperparams == {"p1": [1, 2]}
+        assert task.spec.hyper_param_options.strategy == "grid"
+
+    def test_states(self):
+        assert RunStates.is_terminal(RunStates.completed)
+        assert not RunStates.is_terminal(RunStates.running)
+
+
+class TestSQLRunDB:
+    def test_runs_crud(self, rundb):
+        struct = {"metadata": {"name": "r1", "uid": "u1"},
+                  "status": {"state": "running"}}
+        rundb.store_run(struct, "u1", "proj")
+        run = rundb.read_run("u1", "proj")
+        assert run["metadata"]["name"] == "r1"
+        rundb.update_run({"status.state": "completed"}, "u1", "proj")
+        assert rundb.read_run("u1", "proj")["status"]["state"] == "completed"
+        runs = rundb.list_runs(project="proj")
+        assert len(runs) == 1
+        runs = rundb.list_runs(project="proj", state="completed")
+        assert len(runs) == 1
+        rundb.del_run("u1", "proj")
+        with pytest.raises(MLRunNotFoundError):
+            rundb.read_run("u1", "proj")
+
+    def test_list_runs_filters(self, rundb):
+        for i in range(5):
+            rundb.store_run(
+                {"metadata": {"name": f"run{i}",
+                              "labels": {"kind": "job" if i % 2 else "local"}},
+                 "status": {"state": "completed"}}, f"uid{i}", "p")
+        assert len(rundb.list_runs(project="p")) == 5
+        assert len(rundb.list_runs(project="p", labels={"kind": "job"})) == 2
+        assert len(rundb.list_runs(project="p", last=3)) == 3
+        assert len(rundb.list_runs(project="p", name="run1")) == 1
+
+    def test_artifacts_crud(self, rundb):
+        artifact = {"kind": "artifact", "metadata": {"key": "a1"},
+                    "spec": {"target_path": "/tmp/a1"}}
+        rundb.store_artifact("a1", artifact, project="p", tree="t1",
+                             tag="v1")
+        read = rundb.read_artifact("a1", project="p", tag="v1")
+        assert read["spec"]["target_path"] == "/tmp/a1"
+        # latest tag also resolves
+        read = rundb.read_artifact("a1", project="p")
+        assert read["metadata"]["key"] == "a1"
+        assert len(rundb.list_artifacts(project="p")) == 1
+        rundb.del_artifact("a1", project="p")
+        with pytest.raises(MLRunNotFoundError):
+            rundb.read_artifact("a1", project="p")
+
+    def test_functions_crud(self, rundb):
+        hash_key = rundb.store_function(
+            {"kind": "job", "metadata": {"name": "f1"}}, "f1", "p",
+            tag="latest", versioned=True)
+        assert hash_key
+        func = rundb.get_function("f1", "p")
+        assert func["kind"] == "job"
+        func = rundb.get_function("f1", "p", hash_key=hash_key)
+        assert func["kind"] == "job"
+        assert len(rundb.list_functions(project="p")) == 1
+        rundb.delete_function("f1", "p")
+        with pytest.raises(MLRunNotFoundError):
+            rundb.get_function("f1", "p")
+
+    def test_projects_crud(self, rundb):
+        rundb.create_project({"metadata": {"name": "p1"}})
+        assert rundb.get_project("p1")["metadata"]["name"] == "p1"
+        from mlrun_amd.errors import MLRunConflictError
+
+        with pytest.raises(MLRunConflictError):
+            rundb.create_project({"metadata": {"name": "p1"}})
+        rundb.delete_project("p1")
+        with pytest.raises(MLRunNotFoundError):
+            rundb.get_project("p1")
+
+    def test_logs(self, rundb):
+        rundb.store_log("u1", "p", b"hello ")
+        rundb.store_log("u1", "p", b"world", append=True)
+        _, log = rundb.get_log("u1", "p")
+        assert log == b"hello world"
+
+    def test_schedules(self, rundb):
+        rundb.create_schedule("p", {"name": "s1", "kind": "job",
+                                    "cron_trigger": "*/5 * * * *"})
+        sched = rundb.get_schedule("p", "s1")
+        assert sched["cron_trigger"] == "*/5 * * * *"
+        assert len(rundb.list_schedules("p")) == 1
+        rundb.delete_schedule("p", "s1")
+        assert rundb.list_schedules("p") == []
+
+    def test_feature_sets(self, rundb):
+        rundb.store_feature_set({"metadata": {"name": "fs1"},
+                                 "spec": {"entities": ["id"]}}, project="p")
+        feature_set = rundb.get_feature_set("fs1", "p")
+        assert feature_set["spec"]["entities"] == ["id"]
+        assert len(rundb.list_feature_sets("p")) == 1
+
+
+class TestDatastore:
+    def test_file_store(self, tmp_path):
+        from mlrun_amd.datastore import store_manager
+
+        path = tmp_path / "x.txt"
+        item = store_manager.object(str(path))
+        item.put("hello")
+        assert item.get(encoding="utf-8") == "hello"
+        assert item.stat()["size"] == 5
+        assert item.local() == str(path)
+
+    def test_memory_store(self):
+        from mlrun_amd.datastore import store_manager
+
+        item = store_manager.object("memory://buf1")
+        item.put(b"data")
+        assert item.get() == b"data"
+        item.delete()
+
+    def test_dataframe(self, tmp_path):
+        import pandas as pd
+
+        from mlrun_amd.datastore import store_manager
+
+        df = pd.DataFrame({"a": [1, 2], "b": [3.0, 4.0]})
+        path = tmp_path / "d.csv"
+        df.to_csv(path, index=False)
+        loaded = store_manager.object(str(path)).as_df()
+        assert list(loaded.columns) == ["a", "b"]
+        assert len(loaded) == 2
+
+    def test_store_uri_parse(self):
+        from mlrun_amd.datastore import parse_store_uri
+
+        kind, project, key, tag, tree, it = parse_store_uri(
+            "store://artifacts/proj/mykey:v2")
+        assert (kind, project, key, tag) == ("artifacts", "proj", "mykey",
+                                             "v2")
+        kind, project, key, tag, tree, it = parse_store_uri(
+            "store://models/proj/m@tree123")
+        assert tree == "tree123"
+
+
+class TestArtifacts:
+    def test_store_and_resolve(self, rundb, tmp_path):
+        from mlrun_amd.artifacts import ArtifactManager, ArtifactProducer
+        from mlrun_amd.datastore import store_manager
+
+        manager = ArtifactManager(db=rundb)
+        producer = ArtifactProducer("run", "p", "myrun", uid="tree1")
+        item = manager.log_artifact(producer, "results", body="abc",
+                                    artifact_path=str(tmp_path))
+        assert os.path.isfile(item.spec.target_path)
+        # resolve through store://
+        data_item = store_manager.object("store://artifacts/p/results")
+        assert data_item.get() == b"abc"
+
+    def test_log_model_layout(self, rundb, tmp_path):
+        from mlrun_amd.artifacts import (
+            ArtifactManager, ArtifactProducer, get_model)
+
+        manager = ArtifactManager(db=rundb)
+        producer = ArtifactProducer("run", "p", "myrun", uid="tree2")
+        model = manager.log_model(
+            producer, "mymodel", body=b"\x00weights", framework="torch",
+            parameters={"layers": 2}, artifact_path=str(tmp_path),
+            extra_data={"config.json": json.dumps({"a": 1})})
+        target = model.spec.target_path.rstrip("/")
+        assert os.path.isfile(os.path.join(target, "model_spec.yaml"))
+        assert os.path.isfile(os.path.join(target, "mymodel.bin"))
+        assert os.path.isfile(os.path.join(target, "config.json"))
+        # resolve via store uri
+        model_file, spec, extra = get_model(model.uri)
+        assert spec.framework == "torch"
+        assert os.path.basename(model_file) == "mymodel.bin"
+        assert "config.json" in extra

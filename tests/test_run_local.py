@@ -1,0 +1,174 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Local execution-path tests (baseline config 1: run_local plumbing)."""
+
+import os
+
+import pytest
+
+import mlrun_amd
+from mlrun_amd.model import RunStates
+
+
+def my_handler(context, x: int = 1, y: int = 2):
+    context.log_result("total", x + y)
+    context.log_artifact("note", body=f"x={x}")
+    return {"product": x * y}
+
+
+def failing_handler(context):
+    raise ValueError("boom")
+
+
+class TestRunLocal:
+    def test_handler_run(self):
+        run = mlrun_amd.run_local(handler=my_handler, name="t1",
+                                  params={"x": 3, "y": 4})
+        assert run.status.state == RunStates.completed
+        assert run.status.results["total"] == 7
+        assert run.status.results["product"] == 12
+        assert run.output("note").startswith("store://")
+
+    def test_run_is_in_db(self, rundb):
+        run = mlrun_amd.run_local(handler=my_handler, name="t2",
+                                  params={"x": 1})
+        stored = rundb.read_run(run.metadata.uid, run.metadata.project)
+        assert stored["status"]["state"] == RunStates.completed
+        assert stored["status"]["results"]["total"] == 3
+
+    def test_failed_run(self):
+        run = mlrun_amd.run_local(handler=failing_handler, name="bad")
+        assert run.status.state == RunStates.error
+        assert "boom" in run.status.error
+
+    def test_file_handler(self, tmp_path):
+        code = (
+            "def entry(context, a=0):\n"
+            "    context.log_result('a2', a * 2)\n"
+        )
+        path = tmp_path / "job.py"
+        path.write_text(code)
+        fn = mlrun_amd.new_function(name="filefn", command=str(path))
+        run = fn.run(handler="entry", params={"a": 21})
+        assert run.status.state == RunStates.completed
+        assert run.status.results["a2"] == 42
+
+    def test_command_mode_subprocess(self, tmp_path):
+        code = (
+            "import mlrun_amd\n"
+            "ctx = mlrun_amd.get_or_create_ctx('sub')\n"
+            "ctx.log_result('from_child', 99)\n"
+            "ctx.commit(completed=True)\n"
+        )
+        path = tmp_path / "script.py"
+        path.write_text(code)
+        fn = mlrun_amd.new_function(name="subfn", command=str(path))
+        run = fn.run(name="subrun")
+        assert run.status.state == RunStates.completed
+        assert run.status.results.get("from_child") == 99
+
+    def test_hyperparam_grid(self):
+        run = mlrun_amd.run_local(
+            handler=my_handler, name="sweep",
+            task=None) if False else None
+        fn = mlrun_amd.new_function(name="sweepfn", kind="handler")
+        fn.handler = my_handler
+        run = fn.run(name="sweep", hyperparams={"x": [1, 2], "y": [10, 20]},
+                     hyper_param_options={"selector": "max.total",
+                                          "strategy": "grid"})
+        assert run.status.state == RunStates.completed
+        assert run.status.results["best_iteration"] == 4
+        assert run.status.results["total"] == 22
+        assert len(run.status.iterations) == 4
+
+    def test_hyperparam_list_parallel(self):
+        fn = mlrun_amd.new_function(name="parfn", kind="handler")
+        fn.handler = my_handler
+        run = fn.run(name="par", hyperparams={"x": [1, 2, 3]},
+                     hyper_param_options={"strategy": "list",
+                                          "parallel_runs": 3,
+                                          "selector": "max.total"})
+        assert run.status.state == RunStates.completed
+        assert run.status.results["total"] == 5  # x=3, y=2
+
+    def test_code_to_function(self, tmp_path):
+        code = (
+            "def train(context, lr=0.1):\n"
+            "    '''train a model'''\n"
+            "    context.log_result('lr_used', lr)\n"
+        )
+        path = tmp_path / "train.py"
+        path.write_text(code)
+        fn = mlrun_amd.code_to_function(name="trainer", filename=str(path),
+                                        handler="train", kind="job")
+        assert "train" in fn.spec.entry_points
+        run = fn.run(params={"lr": 0.5}, local=True)
+        assert run.status.results["lr_used"] == 0.5
+
+    def test_get_or_create_ctx_standalone(self):
+        ctx = mlrun_amd.get_or_create_ctx("manual")
+        ctx.log_result("k", 1)
+        ctx.commit(completed=True)
+        assert ctx.state == RunStates.completed
+
+    def test_notifications_console(self, capsys):
+        run = mlrun_amd.run_local(
+            handler=my_handler, name="notif",
+            notifications=[{"kind": "console", "when": ["completed"]}])
+        assert run.status.state == RunStates.completed
+        assert "console" in run.status.notifications
+
+
+class TestHandlerDecorator:
+    def test_decorated_outputs(self):
+        @mlrun_amd.handler(outputs=["doubled"])
+        def fn(context, v: int = 0):
+            return v * 2
+
+        run = mlrun_amd.run_local(handler=fn, name="deco", params={"v": 5})
+        assert run.status.results["doubled"] == 10
+
+
+class TestProjects:
+    def test_project_lifecycle(self, tmp_path):
+        project = mlrun_amd.new_project("proj-a", context=str(tmp_path))
+        assert project.name == "proj-a"
+        fn = project.set_function(my_handler, name="h1", kind="handler")
+        run = project.run_function("h1", params={"x": 10})
+        assert run.status.results["total"] == 12
+        # reload from DB
+        loaded = mlrun_amd.get_or_create_project("proj-a",
+                                                 context=str(tmp_path))
+        assert loaded.name == "proj-a"
+
+    def test_project_artifacts(self, tmp_path):
+        project = mlrun_amd.new_project("proj-b", context=str(tmp_path))
+        project.log_artifact("doc", body="text")
+        artifacts = project.list_artifacts()
+        assert len(artifacts) == 1
+        model = project.log_model("m1", body=b"bin", framework="torch")
+        assert model.uri.startswith("store://models/proj-b/")
+
+    def test_workflow(self, tmp_path):
+        project = mlrun_amd.new_project("proj-c", context=str(tmp_path))
+        project.set_function(my_handler, name="step1", kind="handler")
+
+        def pipeline(project):
+            mlrun_amd.run_function("step1", params={"x": 2})
+            mlrun_amd.run_function("step1", params={"x": 3})
+
+        status = project.run(workflow_handler=pipeline)
+        assert status.state == "completed"
+        assert len(status.runs) == 2
+
+
+class TestScheduleCreation:
+    def test_schedule_run(self, rundb):
+        fn = mlrun_amd.new_function(name="schedfn", kind="handler")
+        fn.handler = my_handler
+        run = fn.run(name="sched", schedule="*/5 * * * *")
+        scheds = rundb.list_schedules("default")
+        assert len(scheds) == 1
+        assert scheds[0]["cron_trigger"] == "*/5 * * * *"
