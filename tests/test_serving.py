@@ -1,0 +1,298 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Serving-graph engine tests (mirror the reference's
+tests/serving/test_serving.py / test_flow.py / test_parallel.py tiers,
+all in-process via mock servers)."""
+
+import json
+
+import pytest
+
+import mlrun_amd
+from mlrun_amd.serving import (
+    Event,
+    GraphServer,
+    RouterStep,
+    TaskStep,
+    V2ModelServer,
+    create_graph_server,
+)
+
+
+class EchoModel(V2ModelServer):
+    def load(self):
+        self.model = "loaded"
+
+    def predict(self, request):
+        return [x * 2 for x in request["inputs"]]
+
+
+class ConstModel(V2ModelServer):
+    def load(self):
+        self.value = self.get_param("value", 1)
+
+    def predict(self, request):
+        return [self.value] * len(request["inputs"])
+
+
+class FailModel(V2ModelServer):
+    def load(self):
+        pass
+
+    def predict(self, request):
+        raise RuntimeError("predict exploded")
+
+
+def double_handler(body):
+    body["x"] = body.get("x", 0) * 2
+    return body
+
+
+def add_one(body):
+    body["x"] = body.get("x", 0) + 1
+    return body
+
+
+def _serving_fn():
+    return mlrun_amd.new_function(name="srv", kind="serving")
+
+
+class TestRouterTopology:
+    def test_infer(self):
+        fn = _serving_fn()
+        fn.add_model("m1", class_name=EchoModel)
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/m1/infer", body={"inputs": [1, 2, 3]})
+        assert resp["outputs"] == [2, 4, 6]
+        assert resp["model_name"] == "m1"
+
+    def test_multiple_models_and_list(self):
+        fn = _serving_fn()
+        fn.add_model("a", class_name=EchoModel)
+        fn.add_model("b", class_name=ConstModel, value=7)
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/b/infer", body={"inputs": [0, 0]})
+        assert resp["outputs"] == [7, 7]
+        listing = server.test("/v2/models/", method="GET")
+        assert set(listing["models"]) == {"a", "b"}
+
+    def test_body_routing(self):
+        fn = _serving_fn()
+        fn.add_model("a", class_name=EchoModel)
+        fn.add_model("b", class_name=ConstModel, value=5)
+        server = fn.to_mock_server()
+        resp = server.test("/", body={"model": "b", "inputs": [1]})
+        assert resp["outputs"] == [5]
+
+    def test_json_string_body(self):
+        fn = _serving_fn()
+        fn.add_model("m", class_name=EchoModel)
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/m/infer",
+                           body=json.dumps({"inputs": [4]}))
+        assert resp["outputs"] == [8]
+
+    def test_model_error(self):
+        fn = _serving_fn()
+        fn.add_model("bad", class_name=FailModel)
+        server = fn.to_mock_server()
+        with pytest.raises(RuntimeError):
+            server.test("/v2/models/bad/infer", body={"inputs": [1]})
+        resp = server.test("/v2/models/bad/infer", body={"inputs": [1]},
+                           silent=True)
+        assert resp.error
+
+    def test_ready_and_metadata(self):
+        fn = _serving_fn()
+        fn.add_model("m", class_name=EchoModel)
+        server = fn.to_mock_server()
+        assert server.test("/v2/models/m/ready")["ready"] is True
+        meta = server.test("/v2/models/m", method="GET")
+        assert meta["name"] == "m"
+
+    def test_unknown_model(self):
+        fn = _serving_fn()
+        fn.add_model("m", class_name=EchoModel)
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/nope/infer", body={"inputs": [1]},
+                           silent=True)
+        assert resp.error
+
+
+class TestFlowTopology:
+    def test_handler_chain(self):
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(handler=double_handler, name="d1") \
+            .to(handler=add_one, name="a1")
+        server = fn.to_mock_server()
+        resp = server.test("/", body={"x": 5})
+        assert resp["x"] == 11
+
+    def test_flow_with_router(self):
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(handler=double_handler, name="pre") \
+            .to(RouterStep(name="router")) \
+            .to(handler=add_one, name="post")
+        fn.add_model("m", class_name=EchoModel)
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/m/infer", body={"x": 1, "inputs": [3]})
+        # pre doubles x (ignored by model), model doubles inputs,
+        # post runs on the response body
+        assert resp["outputs"] == [6]
+        assert resp["x"] == 1
+
+    def test_error_handler(self):
+        def boom(body):
+            raise ValueError("nope")
+
+        def catcher(event):
+            event.body = {"caught": event.error}
+            event.error = None
+            event.terminated = True
+            return event
+
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        step = graph.to(handler=boom, name="boom")
+        graph.add_step(handler=catcher, name="catcher", after=[],
+                       full_event=True)
+        # detach catcher from the implicit chain
+        graph["catcher"].after = []
+        graph._build_links()
+        step.error_handler(name="catcher")
+        server = fn.to_mock_server()
+        resp = server.test("/", body={"x": 1})
+        assert "nope" in resp["caught"]
+
+    def test_respond_semantics(self):
+        """A responder step's output is the response even when the chain
+        continues."""
+        seen = []
+
+        def tail(body):
+            seen.append(dict(body))
+            body["tail"] = True
+            return body
+
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(handler=double_handler, name="first").respond() \
+            .to(handler=tail, name="tail")
+        server = fn.to_mock_server()
+        resp = server.test("/", body={"x": 2}, get_body=False)
+        assert resp.responded
+
+    def test_queue_step(self):
+        import time
+
+        results = []
+
+        def sink(body):
+            results.append(body)
+            return body
+
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(handler=double_handler, name="head") \
+            .to("$queue", name="q1") \
+            .to(handler=sink, name="sink")
+        server = fn.to_mock_server()
+        server.test("/", body={"x": 2})
+        deadline = time.monotonic() + 5
+        while not results and time.monotonic() < deadline:
+            time.sleep(0.01)
+        assert results and results[0]["x"] == 4
+
+
+class TestEnsembles:
+    def test_voting_classification(self):
+        fn = _serving_fn()
+        fn.set_topology("router", class_name="VotingEnsemble")
+        fn.add_model("m1", class_name=ConstModel, value=1)
+        fn.add_model("m2", class_name=ConstModel, value=1)
+        fn.add_model("m3", class_name=ConstModel, value=0)
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/infer", body={"inputs": [0, 0]})
+        assert resp["outputs"] == [1, 1]
+
+    def test_voting_regression_mean(self):
+        class Half(V2ModelServer):
+            def load(self):
+                pass
+
+            def predict(self, request):
+                return [0.5 for _ in request["inputs"]]
+
+        class OneAndHalf(V2ModelServer):
+            def load(self):
+                pass
+
+            def predict(self, request):
+                return [1.5 for _ in request["inputs"]]
+
+        fn = _serving_fn()
+        fn.set_topology("router", class_name="VotingEnsemble",
+                        vote_type="regression")
+        fn.add_model("a", class_name=Half)
+        fn.add_model("b", class_name=OneAndHalf)
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/infer", body={"inputs": [0]})
+        assert resp["outputs"] == [1.0]
+
+    def test_parallel_run_merge(self):
+        fn = _serving_fn()
+        fn.set_topology("router", class_name="ParallelRun",
+                        executor_type="thread")
+        fn.add_model("a", class_name=ConstModel, value=1)
+        fn.add_model("b", class_name=ConstModel, value=2)
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/infer", body={"inputs": [9]})
+        assert resp["results"]["a"]["outputs"] == [1]
+        assert resp["results"]["b"]["outputs"] == [2]
+
+
+class TestGraphServerDict:
+    def test_roundtrip(self):
+        fn = _serving_fn()
+        fn.add_model("m1", class_name="EchoModel", model_path="")
+        struct = fn.to_dict()
+        fn2 = mlrun_amd.new_function(runtime=struct)
+        assert fn2.kind == "serving"
+        server = fn2.to_mock_server(namespace={"EchoModel": EchoModel})
+        resp = server.test("/v2/models/m1/infer", body={"inputs": [1]})
+        assert resp["outputs"] == [2]
+
+
+class TestHTTPDeploy:
+    def test_deploy_and_invoke(self):
+        fn = _serving_fn()
+        fn.add_model("m1", class_name=EchoModel)
+        address = fn.deploy()
+        assert address.startswith("http://")
+        resp = fn.invoke("/v2/models/m1/infer", body={"inputs": [10]})
+        assert resp["outputs"] == [20]
+        import requests
+
+        health = requests.get(address + "/healthz", timeout=5).json()
+        assert health["status"] == "ok"
+        fn.stop()
+
+
+class TestMonitoringIntegration:
+    def test_tracked_model_pushes_events(self):
+        from mlrun_amd.model_monitoring import get_stream_processor
+
+        fn = _serving_fn()
+        fn.set_tracking()
+        fn.add_model("tracked", class_name=EchoModel)
+        server = fn.to_mock_server(track_models=True)
+        for _ in range(5):
+            server.test("/v2/models/tracked/infer", body={"inputs": [1]})
+        processor = get_stream_processor("default")
+        stats = processor.endpoint_stats("tracked")
+        assert stats["300"]["count"] == 5
+        assert stats["300"]["avg_latency_ms"] >= 0
