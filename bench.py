@@ -1,0 +1,161 @@
+#!/usr/bin/env python3
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Flagship benchmark: serving-graph req/sec, Llama-3-8B V2ModelServer.
+
+Measures the BASELINE.json north-star metric: requests/second (+ p50)
+through the serving graph (router -> LlamaServer.do_event -> predict ->
+postprocess) on N GPUs of one node, one serving replica per GPU
+(weak scaling — the reference scales nuclio replicas the same way).
+
+One step = one serving-graph event carrying BATCH requests, each a
+PROMPT_LEN-token synthetic prompt generating GEN_TOKENS tokens
+(random-init weights, synthetic data — no network).
+
+Usage:  python bench.py --gpus N --steps K --warmup W
+(for N>1 the driver launches via torch.distributed.run, one rank/GPU)
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+if REPO not in sys.path:
+    sys.path.insert(0, REPO)
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=10)
+    parser.add_argument("--warmup", type=int, default=3)
+    parser.add_argument("--model", default="llama-3-8b",
+                        help="llama-3-8b | llama-3-70b | tiny")
+    parser.add_argument("--batch", type=int, default=16,
+                        help="requests per serving event")
+    parser.add_argument("--prompt-len", type=int, default=128)
+    parser.add_argument("--gen-tokens", type=int, default=32)
+    parser.add_argument("--no-graph", action="store_true")
+    args = parser.parse_args()
+
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    n_gpus = max(args.gpus, world_size)
+    on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        torch.cuda.set_device(local_rank)
+    device = f"cuda:{local_rank}" if on_gpu else "cpu"
+
+    dist = None
+    if world_size > 1:
+        import torch.distributed as tdist
+
+        dist = tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(backend="nccl" if on_gpu else "gloo",
+                                rank=rank, world_size=world_size)
+
+    import mlrun_amd
+    from mlrun_amd.models.llama import LlamaServer
+
+    model_cfg = args.model if on_gpu else "tiny"
+    batch = args.batch if on_gpu else 2
+    fn = mlrun_amd.new_function(name="bench-serving", kind="serving")
+    fn.add_model("llama", class_name=LlamaServer, config=model_cfg,
+                 batch_size=batch, max_new_tokens=args.gen_tokens,
+                 device=device, use_graph=not args.no_graph)
+    server = fn.to_mock_server()
+
+    import random
+
+    random.seed(1234 + rank)
+    vocab = 1000 if model_cfg == "tiny" else 120000
+    prompt_len = args.prompt_len if on_gpu else 8
+
+    def make_body():
+        return {
+            "inputs": [[random.randrange(1, vocab)
+                        for _ in range(prompt_len)] for _ in range(batch)],
+            "max_tokens": args.gen_tokens,
+        }
+
+    path = "/v2/models/llama/infer"
+
+    # warmup (includes weight init, first prefill, hipGraph capture)
+    for _ in range(args.warmup):
+        server.test(path, body=make_body())
+    if on_gpu:
+        torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+
+    latencies = []
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        s0 = time.perf_counter()
+        server.test(path, body=make_body())
+        if on_gpu:
+            torch.cuda.synchronize()
+        latencies.append((time.perf_counter() - s0) * 1000.0)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max elapsed over ranks = whole-job time
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if on_gpu:
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_requests = n_gpus * batch * args.steps
+    reqs_per_sec = total_requests / elapsed
+    latencies.sort()
+    p50 = latencies[len(latencies) // 2]
+
+    if rank == 0:
+        result = {
+            "metric": "serving-graph req/sec (Llama-3-8B V2ModelServer)",
+            "value": round(reqs_per_sec, 3),
+            "unit": "req/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic (random prompts, random-init weights)",
+            "config": {
+                "model": model_cfg,
+                "global_batch": n_gpus * batch,
+                "seq_len": prompt_len + args.gen_tokens,
+                "prompt_len": prompt_len,
+                "gen_tokens": args.gen_tokens,
+                "parallelism": f"dp{n_gpus} (1 serving replica/GPU)",
+                "p50_ms": round(p50, 3),
+                "tokens_per_sec": round(
+                    reqs_per_sec * args.gen_tokens, 1),
+                "hipgraph": not args.no_graph,
+            },
+        }
+        print(json.dumps(result))
+
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

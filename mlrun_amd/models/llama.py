@@ -1,0 +1,462 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Llama-family generative engine, decode-optimized for MI355X.
+
+Design (MI355X-first, not a port — the reference has no model code):
+- weights live as raw bf16 [N, K] tensors resident in HBM (288 GB:
+  an 8B model + caches is ~6% of one GPU)
+- the decode step runs entirely on hand-written CDNA4 kernels
+  (mlrun_amd/ops: MFMA skinny GEMM, fused add+RMSNorm, RoPE, GQA
+  decode attention, SwiGLU) with every buffer preallocated and the
+  whole token step captured in ONE hipGraph (torch.cuda.CUDAGraph is
+  hipGraph on ROCm) — per-token CPU cost is a single graph replay
+- prefill batches through hipBLASLt (torch.matmul) + SDPA: it is
+  compute-bound and library GEMMs are the right tool there
+  (guide: hand-write the fused hot ops, use hipBLASLt for plain GEMMs)
+- tensor parallelism (Llama-70B): row/col sharded projections with
+  one RCCL all-reduce after attn-out and after mlp-down, over xGMI
+"""
+
+import math
+import typing
+from dataclasses import dataclass, field
+
+import torch
+
+from .. import ops
+from ..utils import logger
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "llama-3-8b"
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    vocab_size: int = 128256
+    rope_theta: float = 500000.0
+    rms_eps: float = 1e-5
+    max_seq_len: int = 2048
+
+    @classmethod
+    def llama3_8b(cls, **over):
+        return cls(**{**dict(name="llama-3-8b"), **over})
+
+    @classmethod
+    def llama3_70b(cls, **over):
+        return cls(**{**dict(
+            name="llama-3-70b", hidden_size=8192, intermediate_size=28672,
+            num_layers=80, num_heads=64, num_kv_heads=8), **over})
+
+    @classmethod
+    def tiny(cls, **over):
+        """Small config for CPU tests."""
+        return cls(**{**dict(
+            name="llama-tiny", hidden_size=256, intermediate_size=512,
+            num_layers=2, num_heads=2, num_kv_heads=2, head_dim=128,
+            vocab_size=1024, max_seq_len=256), **over})
+
+
+class LlamaWeights:
+    """Per-layer raw bf16 weight tensors ([N, K] row-major, matching
+    the skinny-GEMM W layout).  TP sharding slices head/intermediate
+    dims; each rank holds 1/tp of qkv+o+mlp weights."""
+
+    def __init__(self, cfg: LlamaConfig, device, tp_rank=0, tp_size=1,
+                 seed=1234):
+        self.cfg = cfg
+        self.device = device
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        hq = cfg.num_heads // tp_size
+        hkv = max(cfg.num_kv_heads // tp_size, 1)
+        inter = cfg.intermediate_size // tp_size
+        d, h = cfg.head_dim, cfg.hidden_size
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+
+        def w(n, k, std=0.02):
+            t = torch.empty(n, k, dtype=torch.bfloat16, device=device)
+            t.normal_(0.0, std)
+            return t
+
+        # random-init on device (no network for checkpoints; see
+        # BASELINE.md — synthetic data / random weights)
+        torch.manual_seed(seed + tp_rank)
+        self.embed = w(cfg.vocab_size, h)
+        self.layers = []
+        for _ in range(cfg.num_layers):
+            self.layers.append({
+                "attn_norm": torch.ones(h, dtype=torch.bfloat16,
+                                        device=device),
+                "wqkv": w((hq + 2 * hkv) * d, h),
+                "wo": w(h, hq * d),
+                "ffn_norm": torch.ones(h, dtype=torch.bfloat16,
+                                       device=device),
+                "wgate": w(inter, h),
+                "wup": w(inter, h),
+                "wdown": w(h, inter),
+            })
+        self.final_norm = torch.ones(h, dtype=torch.bfloat16, device=device)
+        self.lm_head = w(cfg.vocab_size, h)
+        self.hq, self.hkv, self.inter = hq, hkv, inter
+
+    def load_state_dict(self, state: dict):
+        """Load a checkpoint saved by state_dict() (model artifacts)."""
+        self.embed.copy_(state["embed"])
+        for i, layer in enumerate(self.layers):
+            for key in layer:
+                layer[key].copy_(state[f"layers.{i}.{key}"])
+        self.final_norm.copy_(state["final_norm"])
+        self.lm_head.copy_(state["lm_head"])
+
+    def state_dict(self) -> dict:
+        out = {"embed": self.embed, "final_norm": self.final_norm,
+               "lm_head": self.lm_head}
+        for i, layer in enumerate(self.layers):
+            for key, value in layer.items():
+                out[f"layers.{i}.{key}"] = value
+        return out
+
+
+class LlamaDecodeEngine:
+    """Batched generation engine: prefill (hipBLASLt+SDPA) + hipGraph-
+    captured decode step on the custom kernel chain."""
+
+    def __init__(self, cfg: LlamaConfig, batch_size: int, device=None,
+                 tp_group=None, tp_rank=0, tp_size=1, use_graph=True,
+                 seed=1234):
+        self.cfg = cfg
+        self.B = batch_size
+        self.device = torch.device(
+            device or ("cuda:0" if torch.cuda.is_available() else "cpu"))
+        self.on_gpu = self.device.type == "cuda"
+        self.tp_group = tp_group
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        self.use_graph = use_graph and self.on_gpu
+        self.weights = LlamaWeights(cfg, self.device, tp_rank, tp_size, seed)
+        w = self.weights
+        d, h = cfg.head_dim, cfg.hidden_size
+        B, smax = batch_size, cfg.max_seq_len
+        bf16 = dict(dtype=torch.bfloat16, device=self.device)
+
+        # per-layer KV caches (per-rank kv heads): [L, B, Hkv, Smax, D]
+        self.k_cache = torch.zeros(cfg.num_layers, B, w.hkv, smax, d, **bf16)
+        self.v_cache = torch.zeros(cfg.num_layers, B, w.hkv, smax, d, **bf16)
+        self.cache_lens = torch.zeros(B, dtype=torch.int32,
+                                      device=self.device)
+        self.buf_positions = torch.zeros(B, dtype=torch.int32,
+                                         device=self.device)
+        self.cos_sin = ops.build_rope_cos_sin(smax, d, cfg.rope_theta,
+                                              self.device)
+
+        # preallocated decode buffers (graph-capture requirement)
+        qkv_n = (w.hq + 2 * w.hkv) * d
+        max_n = max(qkv_n, h, 2 * w.inter, cfg.vocab_size)
+        self.buf_tokens = torch.zeros(B, dtype=torch.int64,
+                                      device=self.device)
+        self.buf_hidden = torch.zeros(B, h, **bf16)
+        self.buf_residual = torch.zeros(B, h, **bf16)
+        self.buf_qkv = torch.zeros(B, qkv_n, **bf16)
+        self.buf_attn_out = torch.zeros(B, w.hq * d, **bf16)
+        self.buf_proj = torch.zeros(B, h, **bf16)
+        self.buf_gate = torch.zeros(B, w.inter, **bf16)
+        self.buf_up = torch.zeros(B, w.inter, **bf16)
+        self.buf_act = torch.zeros(B, w.inter, **bf16)
+        self.buf_down = torch.zeros(B, h, **bf16)
+        self.buf_logits = torch.zeros(B, cfg.vocab_size, **bf16)
+        self.buf_c32 = torch.zeros(B, max_n, dtype=torch.float32,
+                                   device=self.device)
+        self.scale = 1.0 / math.sqrt(d)
+        self._graph = None
+        self._ksplits = {}
+
+    # ------------------------------------------------------------ gemm
+    def _gemm(self, a, w_, out):
+        """skinny GEMM into a preallocated bf16 out + f32 scratch."""
+        N = w_.shape[0]
+        K = w_.shape[1]
+        key = (N, K)
+        if key not in self._ksplits:
+            self._ksplits[key] = ops.pick_ksplit(a.shape[0], N, K)
+        c32 = self.buf_c32[:, :N].view(a.shape[0], N) \
+            if self.buf_c32.shape[1] >= N else None
+        return ops.skinny_gemm(a, w_, out=out, c_f32=c32,
+                               ksplit=self._ksplits[key])
+
+    def _maybe_allreduce(self, t):
+        if self.tp_size > 1:
+            import torch.distributed as dist
+
+            dist.all_reduce(t, group=self.tp_group)
+        return t
+
+    # ---------------------------------------------------------- decode
+    def _decode_step_body(self):
+        """One token step for all B sequences.  Reads buf_tokens,
+        leaves next tokens in buf_tokens (greedy).  Entirely on-device:
+        capturable as one hipGraph."""
+        cfg, w = self.cfg, self.weights
+        d = cfg.head_dim
+        B = self.B
+        # embedding (full embed table on every rank)
+        torch.index_select(w.embed, 0, self.buf_tokens,
+                           out=self.buf_residual)
+        ops.rmsnorm(self.buf_residual, w.layers[0]["attn_norm"],
+                    eps=cfg.rms_eps, out=self.buf_hidden)
+        # incoming token sits at position cache_lens; bump the length
+        # once up front so attention covers it in every layer
+        self.buf_positions.copy_(self.cache_lens)
+        self.cache_lens.add_(1)
+        positions = self.buf_positions
+        for li, layer in enumerate(w.layers):
+            # qkv projection
+            self._gemm(self.buf_hidden, layer["wqkv"], self.buf_qkv)
+            q = self.buf_qkv[:, :w.hq * d].view(B, w.hq, d)
+            k = self.buf_qkv[:, w.hq * d:(w.hq + w.hkv) * d].view(
+                B, w.hkv, d)
+            v = self.buf_qkv[:, (w.hq + w.hkv) * d:].view(B, w.hkv, d)
+            ops.rope_inplace(q, positions, self.cos_sin)
+            ops.rope_inplace(k, positions, self.cos_sin)
+            ops.kv_append(self.k_cache[li], self.v_cache[li],
+                          k.contiguous(), v.contiguous(), positions)
+            attn_view = self.buf_attn_out.view(B, w.hq, d)
+            ops.attn_decode(q, self.k_cache[li], self.v_cache[li],
+                            self.cache_lens, self.scale, out=attn_view)
+            self._gemm(self.buf_attn_out, layer["wo"], self.buf_proj)
+            self._maybe_allreduce(self.buf_proj)
+            ops.fused_add_rmsnorm(self.buf_proj, layer["ffn_norm"],
+                                  residual=self.buf_residual,
+                                  eps=cfg.rms_eps, out=self.buf_hidden)
+            # mlp
+            self._gemm(self.buf_hidden, layer["wgate"], self.buf_gate)
+            self._gemm(self.buf_hidden, layer["wup"], self.buf_up)
+            ops.silu_mul(self.buf_gate, self.buf_up, out=self.buf_act)
+            self._gemm(self.buf_act, layer["wdown"], self.buf_down)
+            self._maybe_allreduce(self.buf_down)
+            next_norm = w.layers[li + 1]["attn_norm"] \
+                if li + 1 < cfg.num_layers else w.final_norm
+            ops.fused_add_rmsnorm(self.buf_down, next_norm,
+                                  residual=self.buf_residual,
+                                  eps=cfg.rms_eps, out=self.buf_hidden)
+        self._gemm(self.buf_hidden, w.lm_head, self.buf_logits)
+        torch.argmax(self.buf_logits, dim=-1, out=self.buf_tokens)
+
+    def capture_graph(self):
+        """Capture the decode step as one hipGraph (3 warmup runs on a
+        side stream per torch graph discipline)."""
+        if not self.use_graph or self._graph is not None:
+            return
+        lens_backup = self.cache_lens.clone()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                self._decode_step_body()
+        torch.cuda.current_stream().wait_stream(side)
+        self.cache_lens.copy_(lens_backup)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            self._decode_step_body()
+        self.cache_lens.copy_(lens_backup)
+        self._graph = graph
+        logger.info("decode hipGraph captured", batch=self.B,
+                    layers=self.cfg.num_layers)
+
+    def decode_step(self):
+        if self._graph is not None:
+            self._graph.replay()
+        else:
+            self._decode_step_body()
+
+    # --------------------------------------------------------- prefill
+    @torch.no_grad()
+    def prefill(self, tokens: torch.Tensor) -> torch.Tensor:
+        """Process prompts [B, S]; fill KV caches; return last-token
+        logits [B, vocab].  hipBLASLt GEMMs + SDPA."""
+        cfg, w = self.cfg, self.weights
+        B, S = tokens.shape
+        d = cfg.head_dim
+        tokens = tokens.to(self.device)
+        x = w.embed[tokens.reshape(-1)]  # [B*S, H]
+        residual = x.contiguous()
+        positions = torch.arange(S, dtype=torch.int32, device=self.device
+                                 ).repeat(B)
+        hidden = ops.rmsnorm(residual, w.layers[0]["attn_norm"],
+                             eps=cfg.rms_eps)
+        for li, layer in enumerate(w.layers):
+            qkv = hidden @ layer["wqkv"].t()
+            q = qkv[:, :w.hq * d].reshape(B * S, w.hq, d).contiguous()
+            k = qkv[:, w.hq * d:(w.hq + w.hkv) * d].reshape(
+                B * S, w.hkv, d).contiguous()
+            v = qkv[:, (w.hq + w.hkv) * d:].reshape(B * S, w.hkv, d)
+            ops.rope_inplace(q, positions, self.cos_sin)
+            ops.rope_inplace(k, positions, self.cos_sin)
+            kc = k.view(B, S, w.hkv, d).transpose(1, 2).contiguous()
+            vc = v.view(B, S, w.hkv, d).transpose(1, 2).contiguous()
+            self.k_cache[li, :, :, :S] = kc
+            self.v_cache[li, :, :, :S] = vc
+            qh = q.view(B, S, w.hq, d).transpose(1, 2)
+            attn = torch.nn.functional.scaled_dot_product_attention(
+                qh, kc, vc, is_causal=True, enable_gqa=True)
+            attn = attn.transpose(1, 2).reshape(B * S, w.hq * d)
+            proj = attn @ layer["wo"].t()
+            self._maybe_allreduce(proj)
+            hidden = ops.fused_add_rmsnorm(proj, layer["ffn_norm"],
+                                           residual=residual,
+                                           eps=cfg.rms_eps)
+            gate = hidden @ layer["wgate"].t()
+            up = hidden @ layer["wup"].t()
+            act = ops.silu_mul(gate, up)
+            down = act @ layer["wdown"].t()
+            self._maybe_allreduce(down)
+            next_norm = w.layers[li + 1]["attn_norm"] \
+                if li + 1 < cfg.num_layers else w.final_norm
+            hidden = ops.fused_add_rmsnorm(down, next_norm,
+                                           residual=residual,
+                                           eps=cfg.rms_eps)
+        self.cache_lens.fill_(S)
+        last = hidden.view(B, S, -1)[:, -1]
+        logits = last @ w.lm_head.t()
+        return logits
+
+    # -------------------------------------------------------- generate
+    @torch.no_grad()
+    def generate(self, tokens: torch.Tensor, max_new_tokens: int = 32
+                 ) -> torch.Tensor:
+        """Greedy generation.  tokens [B, S] -> [B, max_new_tokens]."""
+        B, S = tokens.shape
+        assert B == self.B, f"engine built for batch {self.B}, got {B}"
+        assert S + max_new_tokens <= self.cfg.max_seq_len
+        logits = self.prefill(tokens)
+        next_tokens = logits.argmax(dim=-1)
+        self.buf_tokens.copy_(next_tokens)
+        generated = [next_tokens.clone()]
+        if self.use_graph and self._graph is None:
+            self.capture_graph()
+        for _ in range(max_new_tokens - 1):
+            self.decode_step()
+            generated.append(self.buf_tokens.clone())
+        return torch.stack(generated, dim=1)
+
+    def reset(self):
+        self.cache_lens.zero_()
+
+
+class LlamaServer:
+    """V2ModelServer-protocol step serving LlamaDecodeEngine inside a
+    serving graph (the north-star config: gen-AI serving graph with a
+    Llama V2ModelServer — BASELINE.json).
+
+    Request: {"inputs": [[tok, ...], ...], "max_tokens": G}
+    Response outputs: [[generated tokens], ...]
+    Events are served at the engine's fixed batch size: smaller request
+    batches are padded, larger are chunked.
+    """
+
+    def __init__(self, context=None, name=None, model_path=None,
+                 config=None, batch_size=16, max_new_tokens=32,
+                 device=None, use_graph=True, **class_args):
+        from ..serving.v2_serving import V2ModelServer
+
+        self.name = name
+        self.context = context
+        self.model_path = model_path
+        self.ready = False
+        self.error = ""
+        self.protocol = "v2"
+        self.model_spec = None
+        self._model_logger = None
+        self._params = class_args
+        self.cfg_name = config or "llama-3-8b"
+        self.batch_size = batch_size
+        self.max_new_tokens = max_new_tokens
+        self.device = device
+        self.use_graph = use_graph
+        self.engine: typing.Optional[LlamaDecodeEngine] = None
+        # borrow the V2 protocol implementation
+        self._v2 = V2ModelServer.__dict__
+
+    def post_init(self, mode="sync"):
+        stream = getattr(self.context, "stream", None) if self.context \
+            else None
+        if stream is not None:
+            from ..serving.v2_serving import _ModelLogPusher
+
+            self._model_logger = _ModelLogPusher(self, stream)
+        if mode == "sync":
+            self.load()
+            self.ready = True
+
+    def load(self):
+        if self.cfg_name in ("llama-3-8b", "8b"):
+            cfg = LlamaConfig.llama3_8b()
+        elif self.cfg_name in ("llama-3-70b", "70b"):
+            cfg = LlamaConfig.llama3_70b()
+        elif self.cfg_name == "tiny":
+            cfg = LlamaConfig.tiny()
+        elif isinstance(self.cfg_name, LlamaConfig):
+            cfg = self.cfg_name
+        else:
+            raise ValueError(f"unknown llama config {self.cfg_name}")
+        for key, value in self._params.items():
+            if hasattr(cfg, key):
+                setattr(cfg, key, value)
+        self.engine = LlamaDecodeEngine(cfg, self.batch_size,
+                                        device=self.device,
+                                        use_graph=self.use_graph)
+        # load a checkpoint artifact if given (model_spec.yaml layout)
+        if self.model_path:
+            from ..artifacts import get_model
+
+            model_file, spec, extra = get_model(self.model_path)
+            state = torch.load(model_file, map_location=self.engine.device,
+                               weights_only=True)
+            self.engine.weights.load_state_dict(state)
+            self.model_spec = spec
+
+    def do_event(self, event):
+        import time as _time
+
+        start = _time.perf_counter()
+        body = event.body if isinstance(event.body, dict) else {}
+        path = event.path or ""
+        if path.endswith("/ready"):
+            event.body = {"name": self.name, "ready": self.ready}
+            return event
+        inputs = body.get("inputs")
+        if inputs is None:
+            raise ValueError('expected {"inputs": [[token ids], ...]}')
+        max_new = int(body.get("max_tokens", self.max_new_tokens))
+        outputs = []
+        for chunk_start in range(0, len(inputs), self.batch_size):
+            chunk = inputs[chunk_start:chunk_start + self.batch_size]
+            outputs.extend(self._generate_chunk(chunk, max_new))
+        event.body = {"id": event.id, "model_name": self.name,
+                      "outputs": outputs}
+        if self._model_logger:
+            self._model_logger.push(start, {"inputs": [len(inputs)]},
+                                    event.body)
+        return event
+
+    def _generate_chunk(self, prompts: list, max_new: int) -> list:
+        n = len(prompts)
+        max_len = max(len(p) for p in prompts)
+        tokens = torch.zeros(self.batch_size, max_len, dtype=torch.int64)
+        for i, prompt in enumerate(prompts):
+            tokens[i, max_len - len(prompt):] = torch.tensor(
+                prompt, dtype=torch.int64)  # left-pad
+        self.engine.reset()
+        out = self.engine.generate(tokens, max_new_tokens=max_new)
+        return out[:n].cpu().tolist()
+
+    def logged_results(self, request, response, op):
+        return request.get("inputs"), None
+
+    def stats(self):
+        return {}

@@ -1,0 +1,304 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""GPU ops: python dispatch over the in-tree HIP extension
+(mlrun_amd._hip_ops, built from ops/hip/kernels.hip for gfx950).
+
+Dispatch policy:
+- tensors on GPU  -> the HIP extension MUST be present; a missing
+  extension raises MLRunGPUError loudly (no silent eager fallback)
+- tensors on CPU  -> plain fp32 torch reference implementations (these
+  are also the ground truth the GPU numerics tests compare against)
+"""
+
+import math
+import typing
+
+import torch
+
+from ..errors import MLRunGPUError
+
+try:
+    from mlrun_amd import _hip_ops  # built in-tree by setup.py
+
+    HAVE_HIP_OPS = True
+except ImportError:
+    _hip_ops = None
+    HAVE_HIP_OPS = False
+
+
+def _require_hip():
+    if not HAVE_HIP_OPS:
+        raise MLRunGPUError(
+            "mlrun_amd._hip_ops extension is not built — run "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace` "
+            "(GPU ops never fall back to eager execution)")
+    return _hip_ops
+
+
+# ------------------------------------------------------------------ norm
+
+
+def fused_add_rmsnorm(x: torch.Tensor, weight: torch.Tensor,
+                      residual: torch.Tensor = None, eps: float = 1e-5,
+                      out: torch.Tensor = None) -> torch.Tensor:
+    """out = rmsnorm(x + residual) * weight; residual <- x + residual
+    (in-place) when given."""
+    if x.is_cuda:
+        ops = _require_hip()
+        if out is None:
+            out = torch.empty_like(x)
+        ops.fused_add_rmsnorm(out, x, weight, residual, eps)
+        return out
+    # fp32 reference
+    z = x.float() + (residual.float() if residual is not None else 0.0)
+    if residual is not None:
+        residual.copy_(z.to(residual.dtype))
+    var = z.pow(2).mean(dim=-1, keepdim=True)
+    result = (z * torch.rsqrt(var + eps)) * weight.float()
+    result = result.to(x.dtype)
+    if out is not None:
+        out.copy_(result)
+        return out
+    return result
+
+
+def rmsnorm(x, weight, eps=1e-5, out=None):
+    return fused_add_rmsnorm(x, weight, residual=None, eps=eps, out=out)
+
+
+# ------------------------------------------------------------------ rope
+
+
+def build_rope_cos_sin(max_pos: int, head_dim: int, theta: float = 500000.0,
+                       device="cpu") -> torch.Tensor:
+    """Precompute the [max_pos, head_dim/2, 2] f32 cos/sin table
+    (host-side trig — guide Appendix B)."""
+    half = head_dim // 2
+    inv_freq = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float64)
+                                / half))
+    pos = torch.arange(max_pos, dtype=torch.float64)
+    angles = torch.outer(pos, inv_freq)  # [max_pos, half]
+    table = torch.stack([angles.cos(), angles.sin()], dim=-1).float()
+    return table.contiguous().to(device)
+
+
+def rope_inplace(q: torch.Tensor, positions: torch.Tensor,
+                 cos_sin: torch.Tensor):
+    """In-place neox-style rotary embedding. q: [T, heads, dim]."""
+    if q.is_cuda:
+        _require_hip().rope(q, positions, cos_sin)
+        return q
+    T, heads, dim = q.shape
+    half = dim // 2
+    table = cos_sin[positions.long()]  # [T, half, 2]
+    cos = table[..., 0].unsqueeze(1)  # [T,1,half]
+    sin = table[..., 1].unsqueeze(1)
+    qf = q.float()
+    q1 = qf[..., :half]
+    q2 = qf[..., half:]
+    q[..., :half] = (q1 * cos - q2 * sin).to(q.dtype)
+    q[..., half:] = (q2 * cos + q1 * sin).to(q.dtype)
+    return q
+
+
+# ---------------------------------------------------------------- swiglu
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor,
+             out: torch.Tensor = None) -> torch.Tensor:
+    if gate.is_cuda:
+        ops = _require_hip()
+        if out is None:
+            out = torch.empty_like(gate)
+        ops.silu_mul(out, gate, up)
+        return out
+    result = (torch.nn.functional.silu(gate.float()) *
+              up.float()).to(gate.dtype)
+    if out is not None:
+        out.copy_(result)
+        return out
+    return result
+
+
+# ------------------------------------------------------------------ gemm
+
+
+def pick_ksplit(M: int, N: int, K: int, target_blocks: int = 1024) -> int:
+    """Choose the K-split so the grid fills 256 CUs (>=~1024 wgs)."""
+    nblocks = (N + 127) // 128
+    if nblocks >= target_blocks:
+        return 1
+    ksplit = (target_blocks + nblocks - 1) // nblocks
+    ksplit = min(ksplit, max(K // 256, 1), 16)
+    return max(ksplit, 1)
+
+
+def skinny_gemm(a: torch.Tensor, w: torch.Tensor,
+                out: torch.Tensor = None,
+                c_f32: torch.Tensor = None,
+                ksplit: int = None) -> torch.Tensor:
+    """C[M,N] = A[M,K] @ W[N,K]^T for decode batches (M <= 16), bf16.
+
+    On GPU: MFMA kernel with K-split atomics into a f32 workspace, then
+    cast to bf16.  Pass preallocated out/c_f32 for graph capture."""
+    M, K = a.shape
+    N = w.shape[0]
+    if a.is_cuda:
+        ops = _require_hip()
+        if c_f32 is None:
+            c_f32 = torch.empty(M, N, dtype=torch.float32, device=a.device)
+        if ksplit is None:
+            ksplit = pick_ksplit(M, N, K)
+        ops.skinny_gemm(c_f32, a, w, ksplit)
+        if out is None:
+            out = torch.empty(M, N, dtype=torch.bfloat16, device=a.device)
+        ops.cast_f32_bf16(out, c_f32)
+        return out
+    result = (a.float() @ w.float().t()).to(a.dtype)
+    if out is not None:
+        out.copy_(result)
+        return out
+    return result
+
+
+# ------------------------------------------------------------- attention
+
+
+def attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
+                v_cache: torch.Tensor, seq_lens: torch.Tensor,
+                scale: float = None,
+                out: torch.Tensor = None) -> torch.Tensor:
+    """GQA decode attention.  q [B,Hq,D], caches [B,Hkv,Smax,D]."""
+    B, Hq, D = q.shape
+    Hkv = k_cache.shape[1]
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    if q.is_cuda:
+        ops = _require_hip()
+        if out is None:
+            out = torch.empty_like(q)
+        ops.attn_decode(out, q, k_cache, v_cache, seq_lens, scale)
+        return out
+    # fp32 reference
+    if out is None:
+        out = torch.empty_like(q)
+    G = Hq // Hkv
+    for b in range(B):
+        length = int(seq_lens[b])
+        for h in range(Hq):
+            kvh = h // G
+            qv = q[b, h].float()
+            keys = k_cache[b, kvh, :length].float()
+            vals = v_cache[b, kvh, :length].float()
+            scores = torch.softmax(keys @ qv * scale, dim=0)
+            out[b, h] = (scores @ vals).to(q.dtype)
+    return out
+
+
+def kv_append(k_cache, v_cache, k_new, v_new, positions):
+    """Scatter the new token K/V into the cache at positions[b]."""
+    if k_cache.is_cuda:
+        _require_hip().kv_append(k_cache, v_cache, k_new, v_new, positions)
+        return
+    B = k_cache.shape[0]
+    for b in range(B):
+        pos = int(positions[b])
+        k_cache[b, :, pos] = k_new[b]
+        v_cache[b, :, pos] = v_new[b]
+
+
+# ---------------------------------------------------------------- others
+
+
+def softmax(x: torch.Tensor, out: torch.Tensor = None) -> torch.Tensor:
+    if x.is_cuda:
+        ops = _require_hip()
+        if out is None:
+            out = torch.empty_like(x)
+        ops.softmax(out, x)
+        return out
+    result = torch.softmax(x.float(), dim=-1).to(x.dtype)
+    if out is not None:
+        out.copy_(result)
+        return out
+    return result
+
+
+def tree_ensemble_predict(features: torch.Tensor, nodes: dict,
+                          base_score: float = 0.0,
+                          out: torch.Tensor = None) -> torch.Tensor:
+    """GBDT margin prediction.  nodes: dict of flat SoA int32/f32
+    tensors (feature_idx/threshold/left/right/leaf_value/tree_offsets),
+    on the same device as features."""
+    if features.is_cuda:
+        ops = _require_hip()
+        if out is None:
+            out = torch.empty(features.shape[0], dtype=torch.float32,
+                              device=features.device)
+        ops.tree_ensemble(out, features, nodes["feature_idx"],
+                          nodes["threshold"], nodes["left"], nodes["right"],
+                          nodes["leaf_value"], nodes["tree_offsets"],
+                          base_score)
+        return out
+    # reference: walk trees in python
+    n = features.shape[0]
+    result = torch.full((n,), base_score, dtype=torch.float32)
+    fidx = nodes["feature_idx"].tolist()
+    thr = nodes["threshold"].tolist()
+    left = nodes["left"].tolist()
+    right = nodes["right"].tolist()
+    leaf = nodes["leaf_value"].tolist()
+    offsets = nodes["tree_offsets"].tolist()
+    feats = features.tolist()
+    for i in range(n):
+        score = base_score
+        for t in range(len(offsets) - 1):
+            node = offsets[t]
+            while fidx[node] >= 0:
+                node = left[node] if feats[i][fidx[node]] < thr[node] \
+                    else right[node]
+            score += leaf[node]
+        result[i] = score
+    if out is not None:
+        out.copy_(result)
+        return out
+    return result
+
+
+def window_ingest(ring, keys, values, period_idx):
+    if ring.is_cuda:
+        _require_hip().window_ingest(ring, keys, values, period_idx)
+        return ring
+    n_periods = ring.shape[1]
+    for key, value, pidx in zip(keys.tolist(), values.tolist(),
+                                period_idx.tolist()):
+        cell = ring[key, pidx % n_periods]
+        cell[0] += value
+        cell[1] += 1.0
+    return ring
+
+
+def window_reduce(ring, window_periods: int, current_period: int,
+                  out: torch.Tensor = None):
+    if ring.is_cuda:
+        ops = _require_hip()
+        if out is None:
+            out = torch.empty(ring.shape[0], 4, dtype=torch.float32,
+                              device=ring.device)
+        ops.window_reduce(out, ring, window_periods, current_period)
+        return out
+    n_keys, n_periods, _ = ring.shape
+    result = torch.zeros(n_keys, 4, dtype=torch.float32)
+    for w in range(window_periods):
+        p = (current_period - w) % n_periods
+        result[:, 0] += ring[:, p, 0]
+        result[:, 1] += ring[:, p, 1]
+    counts = result[:, 1].clamp(min=1.0)
+    result[:, 2] = result[:, 0] / counts
+    result[:, 2] = torch.where(result[:, 1] > 0, result[:, 2],
+                               torch.zeros_like(result[:, 2]))
+    if out is not None:
+        out.copy_(result)
+        return out
+    return result

@@ -1,0 +1,190 @@
+// Copyright 2026 mlrun_amd authors
+//
+// Licensed under the Apache License, Version 2.0 (the "License");
+// you may not use this file except in compliance with the License.
+//
+// Torch bindings for the CDNA4 kernel library (mlrun_amd._hip_ops).
+// Thin validation + launch layer; kernels live in kernels.hip.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "kernels.h"
+
+#define CHECK_DEV(x) TORCH_CHECK((x).is_cuda(), #x " must be on GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK((x).is_contiguous(), #x " not contiguous")
+#define CHECK_BF16(x) \
+  TORCH_CHECK((x).scalar_type() == torch::kBFloat16, #x " must be bf16")
+#define CHECK_F32(x) \
+  TORCH_CHECK((x).scalar_type() == torch::kFloat32, #x " must be f32")
+#define CHECK_I32(x) \
+  TORCH_CHECK((x).scalar_type() == torch::kInt32, #x " must be int32")
+
+static void* stream() {
+  return (void*)at::cuda::getCurrentCUDAStream().stream();
+}
+
+// out <- rmsnorm(x [+ residual]); residual updated in-place when given
+void fused_add_rmsnorm(torch::Tensor out, torch::Tensor x,
+                       torch::Tensor weight,
+                       c10::optional<torch::Tensor> residual, double eps) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
+  CHECK_DEV(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  CHECK_DEV(weight); CHECK_CONTIG(weight); CHECK_BF16(weight);
+  int64_t hidden = x.size(-1);
+  int64_t rows = x.numel() / hidden;
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  TORCH_CHECK(weight.numel() == hidden, "weight size mismatch");
+  void* res_ptr = nullptr;
+  if (residual.has_value()) {
+    CHECK_DEV(*residual); CHECK_CONTIG(*residual); CHECK_BF16(*residual);
+    TORCH_CHECK(residual->numel() == x.numel(), "residual size mismatch");
+    res_ptr = residual->data_ptr();
+  }
+  launch_fused_add_rmsnorm(out.data_ptr(), res_ptr, x.data_ptr(),
+                           weight.data_ptr(), (int)rows, (int)hidden,
+                           (float)eps, stream());
+}
+
+void rope(torch::Tensor q, torch::Tensor positions, torch::Tensor cos_sin) {
+  CHECK_DEV(q); CHECK_CONTIG(q); CHECK_BF16(q);
+  CHECK_DEV(positions); CHECK_CONTIG(positions); CHECK_I32(positions);
+  CHECK_DEV(cos_sin); CHECK_CONTIG(cos_sin); CHECK_F32(cos_sin);
+  TORCH_CHECK(q.dim() == 3, "q must be [T, heads, dim]");
+  int T = q.size(0), heads = q.size(1), dim = q.size(2);
+  TORCH_CHECK(dim % 2 == 0 && dim / 2 <= 1024, "bad head dim");
+  TORCH_CHECK(positions.numel() == T, "positions size mismatch");
+  launch_rope(q.data_ptr(), positions.data_ptr(), cos_sin.data_ptr(), T,
+              heads, dim, stream());
+}
+
+void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
+  CHECK_DEV(gate); CHECK_CONTIG(gate); CHECK_BF16(gate);
+  CHECK_DEV(up); CHECK_CONTIG(up); CHECK_BF16(up);
+  TORCH_CHECK(out.numel() == gate.numel() && gate.numel() == up.numel(),
+              "size mismatch");
+  TORCH_CHECK(out.numel() % 8 == 0, "numel must be a multiple of 8");
+  launch_silu_mul(out.data_ptr(), gate.data_ptr(), up.data_ptr(),
+                  out.numel(), stream());
+}
+
+// C_f32 [M,N] (pre-zeroed when ksplit>1) <- A [M,K] @ W [N,K]^T
+void skinny_gemm(torch::Tensor c_f32, torch::Tensor a, torch::Tensor w,
+                 int64_t ksplit) {
+  CHECK_DEV(c_f32); CHECK_CONTIG(c_f32); CHECK_F32(c_f32);
+  CHECK_DEV(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  CHECK_DEV(w); CHECK_CONTIG(w); CHECK_BF16(w);
+  int M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "K mismatch");
+  TORCH_CHECK(M <= 16, "skinny_gemm supports M <= 16");
+  TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32");
+  TORCH_CHECK(c_f32.size(0) == M && c_f32.size(1) == N, "C shape mismatch");
+  if (ksplit > 1) launch_zero_f32(c_f32.data_ptr(), c_f32.numel(), stream());
+  launch_skinny_gemm(c_f32.data_ptr(), a.data_ptr(), w.data_ptr(), M, N, K,
+                     (int)ksplit, stream());
+}
+
+void cast_f32_bf16(torch::Tensor out, torch::Tensor in) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
+  CHECK_DEV(in); CHECK_CONTIG(in); CHECK_F32(in);
+  TORCH_CHECK(out.numel() == in.numel(), "size mismatch");
+  launch_cast_f32_bf16(out.data_ptr(), in.data_ptr(), in.numel(), stream());
+}
+
+void attn_decode(torch::Tensor o, torch::Tensor q, torch::Tensor kc,
+                 torch::Tensor vc, torch::Tensor seq_lens, double scale) {
+  CHECK_DEV(o); CHECK_CONTIG(o); CHECK_BF16(o);
+  CHECK_DEV(q); CHECK_CONTIG(q); CHECK_BF16(q);
+  CHECK_DEV(kc); CHECK_CONTIG(kc); CHECK_BF16(kc);
+  CHECK_DEV(vc); CHECK_CONTIG(vc); CHECK_BF16(vc);
+  CHECK_DEV(seq_lens); CHECK_CONTIG(seq_lens); CHECK_I32(seq_lens);
+  TORCH_CHECK(q.dim() == 3 && kc.dim() == 4, "q [B,Hq,D], kc [B,Hkv,S,D]");
+  int B = q.size(0), Hq = q.size(1), D = q.size(2);
+  int Hkv = kc.size(1), Smax = kc.size(2);
+  TORCH_CHECK(D == 128, "attn_decode requires head_dim 128");
+  TORCH_CHECK(Hq % Hkv == 0 && Hq / Hkv <= 8,
+              "grouped heads per kv head must divide and be <= 8");
+  TORCH_CHECK(kc.size(0) == B && kc.size(3) == D, "kc shape mismatch");
+  launch_attn_decode(o.data_ptr(), q.data_ptr(), kc.data_ptr(),
+                     vc.data_ptr(), seq_lens.data_ptr(), B, Hq, Hkv, Smax,
+                     (float)scale, stream());
+}
+
+void kv_append(torch::Tensor kc, torch::Tensor vc, torch::Tensor knew,
+               torch::Tensor vnew, torch::Tensor positions) {
+  CHECK_DEV(kc); CHECK_CONTIG(kc); CHECK_BF16(kc);
+  CHECK_DEV(knew); CHECK_CONTIG(knew); CHECK_BF16(knew);
+  CHECK_DEV(positions); CHECK_CONTIG(positions); CHECK_I32(positions);
+  int B = kc.size(0), Hkv = kc.size(1), Smax = kc.size(2), D = kc.size(3);
+  TORCH_CHECK(D % 8 == 0, "head_dim must be a multiple of 8");
+  launch_kv_append(kc.data_ptr(), vc.data_ptr(), knew.data_ptr(),
+                   vnew.data_ptr(), positions.data_ptr(), B, Hkv, Smax, D,
+                   stream());
+}
+
+void softmax(torch::Tensor out, torch::Tensor in) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
+  CHECK_DEV(in); CHECK_CONTIG(in); CHECK_BF16(in);
+  int64_t cols = in.size(-1);
+  int64_t rows = in.numel() / cols;
+  launch_softmax(out.data_ptr(), in.data_ptr(), (int)rows, (int)cols,
+                 stream());
+}
+
+void tree_ensemble(torch::Tensor out, torch::Tensor features,
+                   torch::Tensor feature_idx, torch::Tensor threshold,
+                   torch::Tensor left, torch::Tensor right,
+                   torch::Tensor leaf_value, torch::Tensor tree_offsets,
+                   double base_score) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_F32(out);
+  CHECK_DEV(features); CHECK_CONTIG(features); CHECK_F32(features);
+  CHECK_I32(feature_idx); CHECK_I32(left); CHECK_I32(right);
+  CHECK_I32(tree_offsets); CHECK_F32(threshold); CHECK_F32(leaf_value);
+  int n_samples = features.size(0), n_features = features.size(1);
+  int n_trees = tree_offsets.numel() - 1;
+  launch_tree_ensemble(out.data_ptr(), features.data_ptr(),
+                       feature_idx.data_ptr(), threshold.data_ptr(),
+                       left.data_ptr(), right.data_ptr(),
+                       leaf_value.data_ptr(), tree_offsets.data_ptr(),
+                       n_trees, n_samples, n_features, (float)base_score,
+                       stream());
+}
+
+void window_ingest(torch::Tensor ring, torch::Tensor keys,
+                   torch::Tensor values, torch::Tensor period_idx) {
+  CHECK_DEV(ring); CHECK_CONTIG(ring); CHECK_F32(ring);
+  CHECK_I32(keys); CHECK_F32(values); CHECK_I32(period_idx);
+  TORCH_CHECK(ring.dim() == 3 && ring.size(2) == 4,
+              "ring must be [keys, periods, 4]");
+  launch_window_ingest(ring.data_ptr(), keys.data_ptr(), values.data_ptr(),
+                       period_idx.data_ptr(), keys.numel(),
+                       (int)ring.size(1), stream());
+}
+
+void window_reduce(torch::Tensor out, torch::Tensor ring,
+                   int64_t window_periods, int64_t current_period) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_F32(out);
+  CHECK_DEV(ring); CHECK_CONTIG(ring); CHECK_F32(ring);
+  launch_window_reduce(out.data_ptr(), ring.data_ptr(), (int)ring.size(0),
+                       (int)ring.size(1), (int)window_periods,
+                       (int)current_period, stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
+        "fused residual add + RMSNorm (bf16)");
+  m.def("rope", &rope, "in-place rotary embedding (bf16)");
+  m.def("silu_mul", &silu_mul, "silu(gate) * up (bf16)");
+  m.def("skinny_gemm", &skinny_gemm,
+        "decode GEMM C=A@W^T on MFMA (bf16 in, f32 out)");
+  m.def("cast_f32_bf16", &cast_f32_bf16, "f32 -> bf16 cast");
+  m.def("attn_decode", &attn_decode, "GQA decode attention (bf16)");
+  m.def("kv_append", &kv_append, "append token K/V into cache");
+  m.def("softmax", &softmax, "row softmax (bf16)");
+  m.def("tree_ensemble", &tree_ensemble, "GBDT ensemble inference (f32)");
+  m.def("window_ingest", &window_ingest,
+        "feature-store window ring ingest (f32)");
+  m.def("window_reduce", &window_reduce,
+        "feature-store window ring reduce (f32)");
+}

@@ -1,0 +1,628 @@
+// Copyright 2026 mlrun_amd authors
+//
+// Licensed under the Apache License, Version 2.0 (the "License");
+// you may not use this file except in compliance with the License.
+//
+// CDNA4 (gfx950 / MI355X) serving & training kernels.
+//
+// Design notes (see /root/repo/SURVEY.md §2.3/§2.6 — the reference has
+// no GPU kernels; this kernel surface is defined by the north-star
+// serving/training configs):
+//  - wave = 64 lanes everywhere; all bf16 global loads vectorized as
+//    short8/ushort2 (16B / 4B per lane)
+//  - decode GEMM (M<=16 activations x [N,K] weights) on MFMA
+//    v_mfma_f32_16x16x32_bf16, K-split across workgroups with f32
+//    atomics so the grid fills 256 CUs
+//  - decode attention: one workgroup per (batch, kv-head), one wave
+//    per grouped q-head, online softmax, LDS score buffer
+//  - norm/rope/swiglu fused elementwise kernels at HBM rate
+// All kernels are hipGraph-capture safe: no allocation, no sync.
+
+#include <hip/hip_runtime.h>
+#include <cfloat>
+#include <cstdint>
+
+#include "kernels.h"
+
+typedef __attribute__((ext_vector_type(8))) short short8v;
+typedef __attribute__((ext_vector_type(4))) float f32x4v;
+typedef __attribute__((ext_vector_type(2))) float f32x2v;
+
+#define DEV __device__ __forceinline__
+
+DEV float bf2f(unsigned short u) {
+  union { uint32_t i; float f; } v;
+  v.i = uint32_t(u) << 16;
+  return v.f;
+}
+
+DEV unsigned short f2bf(float f) {
+  union { float f; uint32_t i; } v;
+  v.f = f;
+  uint32_t x = v.i;
+  // round-to-nearest-even
+  uint32_t r = (x + 0x7fffu + ((x >> 16) & 1u)) >> 16;
+  return (unsigned short)r;
+}
+
+struct ushort8 { unsigned short v[8]; };
+struct ushort2v { unsigned short x, y; };
+
+// ---------------------------------------------------------------------
+// fused residual-add + RMSNorm (bf16):
+//   residual <- x + residual ; out <- rmsnorm(residual) * weight
+// one workgroup per row; vectorized short8 loads (guide G13).
+// ---------------------------------------------------------------------
+__global__ void fused_add_rmsnorm_kernel(
+    unsigned short* __restrict__ out, unsigned short* __restrict__ residual,
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ weight, int hidden, float eps,
+    int has_residual) {
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int nthreads = blockDim.x;
+  const unsigned short* xr = x + (size_t)row * hidden;
+  unsigned short* rr = residual ? residual + (size_t)row * hidden : nullptr;
+  unsigned short* outr = out + (size_t)row * hidden;
+
+  float sumsq = 0.f;
+  for (int i = tid * 8; i < hidden; i += nthreads * 8) {
+    ushort8 xv = *reinterpret_cast<const ushort8*>(xr + i);
+    float z[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) z[j] = bf2f(xv.v[j]);
+    if (has_residual) {
+      ushort8 rv = *reinterpret_cast<const ushort8*>(rr + i);
+      ushort8 zv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        z[j] += bf2f(rv.v[j]);
+        zv.v[j] = f2bf(z[j]);
+      }
+      *reinterpret_cast<ushort8*>(rr + i) = zv;  // updated residual
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sumsq += z[j] * z[j];
+  }
+  // block reduce
+  __shared__ float red[32];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    sumsq += __shfl_down(sumsq, off, 64);
+  if ((tid & 63) == 0) red[tid >> 6] = sumsq;
+  __syncthreads();
+  if (tid < (nthreads >> 6)) sumsq = red[tid];
+  else sumsq = 0.f;
+  if (tid < 64) {
+#pragma unroll
+    for (int off = 2; off > 0; off >>= 1)
+      sumsq += __shfl_down(sumsq, off, 64);
+  }
+  if (tid == 0) red[0] = sumsq;
+  __syncthreads();
+  const float inv = rsqrtf(red[0] / hidden + eps);
+
+  const unsigned short* zsrc = has_residual ? rr : xr;
+  for (int i = tid * 8; i < hidden; i += nthreads * 8) {
+    ushort8 zv = *reinterpret_cast<const ushort8*>(zsrc + i);
+    ushort8 wv = *reinterpret_cast<const ushort8*>(weight + i);
+    ushort8 ov;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      ov.v[j] = f2bf(bf2f(zv.v[j]) * inv * bf2f(wv.v[j]));
+    *reinterpret_cast<ushort8*>(outr + i) = ov;
+  }
+}
+
+void launch_fused_add_rmsnorm(void* out, void* residual, const void* x,
+                              const void* weight, int rows, int hidden,
+                              float eps, void* stream) {
+  int threads = hidden >= 2048 ? 256 : 64;
+  hipLaunchKernelGGL(fused_add_rmsnorm_kernel, dim3(rows), dim3(threads), 0,
+                     (hipStream_t)stream, (unsigned short*)out,
+                     (unsigned short*)residual, (const unsigned short*)x,
+                     (const unsigned short*)weight, hidden, eps,
+                     residual != nullptr ? 1 : 0);
+}
+
+// ---------------------------------------------------------------------
+// RoPE (neox / llama style, rotate-half):
+//   for i < D/2: (q_i, q_{i+D/2}) <- (q_i c - q_{i+D/2} s,
+//                                     q_{i+D/2} c + q_i s)
+// cos_sin: [max_pos, D/2, 2] f32 precomputed host-side (guide B:
+// on-device trig turns memory-bound into VALU-bound).
+// q: [T, heads, D]; positions: [T].
+// ---------------------------------------------------------------------
+__global__ void rope_kernel(unsigned short* __restrict__ q,
+                            const int* __restrict__ positions,
+                            const float* __restrict__ cos_sin, int heads,
+                            int dim) {
+  const int t = blockIdx.x;
+  const int h = blockIdx.y;
+  const int i = threadIdx.x;  // 0..D/2-1
+  const int half = dim >> 1;
+  if (i >= half) return;
+  const int pos = positions[t];
+  const float c = cos_sin[((size_t)pos * half + i) * 2];
+  const float s = cos_sin[((size_t)pos * half + i) * 2 + 1];
+  unsigned short* base = q + ((size_t)t * heads + h) * dim;
+  float a = bf2f(base[i]);
+  float b = bf2f(base[i + half]);
+  base[i] = f2bf(a * c - b * s);
+  base[i + half] = f2bf(b * c + a * s);
+}
+
+void launch_rope(void* q, const void* positions, const void* cos_sin, int T,
+                 int heads, int dim, void* stream) {
+  hipLaunchKernelGGL(rope_kernel, dim3(T, heads), dim3(dim / 2), 0,
+                     (hipStream_t)stream, (unsigned short*)q,
+                     (const int*)positions, (const float*)cos_sin, heads, dim);
+}
+
+// ---------------------------------------------------------------------
+// SwiGLU: out = silu(gate) * up   (bf16, flat, vectorized)
+// gate/up are halves of one [rows, 2*inter] tensor or separate ptrs.
+// ---------------------------------------------------------------------
+__global__ void silu_mul_kernel(unsigned short* __restrict__ out,
+                                const unsigned short* __restrict__ gate,
+                                const unsigned short* __restrict__ up,
+                                long long n) {
+  long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  long long stride = (long long)gridDim.x * blockDim.x * 8;
+  for (; i < n; i += stride) {
+    ushort8 g = *reinterpret_cast<const ushort8*>(gate + i);
+    ushort8 u = *reinterpret_cast<const ushort8*>(up + i);
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g.v[j]);
+      float s = gf / (1.f + __expf(-gf));
+      o.v[j] = f2bf(s * bf2f(u.v[j]));
+    }
+    *reinterpret_cast<ushort8*>(out + i) = o;
+  }
+}
+
+void launch_silu_mul(void* out, const void* gate, const void* up,
+                     long long n, void* stream) {
+  long long blocks = (n / 8 + 255) / 256;
+  if (blocks > 2048) blocks = 2048;  // grid-stride (guide G11)
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(silu_mul_kernel, dim3((int)blocks), dim3(256), 0,
+                     (hipStream_t)stream, (unsigned short*)out,
+                     (const unsigned short*)gate, (const unsigned short*)up,
+                     n);
+}
+
+// ---------------------------------------------------------------------
+// Skinny GEMM (decode projections):  C[M,N] (+)= A[M,K] @ W[N,K]^T
+// M <= 16 (decode batch), bf16 inputs, f32 C.
+// MFMA 16x16x32: each wave owns a 32-wide n-tile (2 B-fragments),
+// 4 waves/block -> 128 n per block; K split across gridDim.y with
+// device-scope f32 atomics (cross-XCD safe, guide G12/G16).
+// Fragment layout (verified on HW by tests/gpu):
+//   A: lane l holds A[l&15][(l>>4)*8 + j]   j=0..7
+//   B: lane l holds W[n0 + (l&15)][(l>>4)*8 + j] (B^T = W row-major)
+//   C: lane l, reg r -> C[(l>>4)*4 + r][l&15]
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void skinny_gemm_kernel(
+    float* __restrict__ C, const unsigned short* __restrict__ A,
+    const unsigned short* __restrict__ W, int M, int N, int K, int ksplit,
+    int use_atomic) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int n0 = blockIdx.x * 128 + wave * 32;  // this wave's n-tile
+  if (n0 >= N) return;
+  const int kchunk = (K / ksplit + 31) & ~31;  // multiple of 32
+  const int kbegin = blockIdx.y * kchunk;
+  int kend = kbegin + kchunk;
+  if (kend > K) kend = K;
+
+  const int arow = lane & 15;
+  const int kb = (lane >> 4) * 8;
+  const bool a_valid = arow < M;
+  const int brow0 = n0 + (lane & 15);
+  const int brow1 = brow0 + 16;
+  const bool b0_valid = brow0 < N;
+  const bool b1_valid = brow1 < N;
+
+  f32x4v acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4v acc1 = {0.f, 0.f, 0.f, 0.f};
+  const short8v zero8 = {0, 0, 0, 0, 0, 0, 0, 0};
+
+  const unsigned short* aptr = A + (size_t)arow * K + kb;
+  const unsigned short* bptr0 = W + (size_t)brow0 * K + kb;
+  const unsigned short* bptr1 = W + (size_t)brow1 * K + kb;
+
+  for (int k = kbegin; k < kend; k += 32) {
+    short8v af = a_valid
+        ? *reinterpret_cast<const short8v*>(aptr + k) : zero8;
+    short8v bf0 = b0_valid
+        ? *reinterpret_cast<const short8v*>(bptr0 + k) : zero8;
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
+    if (b1_valid) {
+      short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+    }
+  }
+
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = crow_base + r;
+    if (m >= M) continue;
+    if (b0_valid) {
+      float* dst = C + (size_t)m * N + n0 + ccol;
+      if (use_atomic) atomicAdd(dst, acc0[r]);
+      else *dst = acc0[r];
+    }
+    if (b1_valid) {
+      float* dst = C + (size_t)m * N + n0 + 16 + ccol;
+      if (use_atomic) atomicAdd(dst, acc1[r]);
+      else *dst = acc1[r];
+    }
+  }
+}
+
+void launch_skinny_gemm(void* C, const void* A, const void* W, int M, int N,
+                        int K, int ksplit, void* stream) {
+  if (ksplit < 1) ksplit = 1;
+  int nblocks = (N + 127) / 128;
+  hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, ksplit), dim3(256), 0,
+                     (hipStream_t)stream, (float*)C,
+                     (const unsigned short*)A, (const unsigned short*)W, M, N,
+                     K, ksplit, ksplit > 1 ? 1 : 0);
+}
+
+// f32 -> bf16 flat cast (epilogue after k-split accumulate)
+__global__ void cast_f32_bf16_kernel(unsigned short* __restrict__ out,
+                                     const float* __restrict__ in,
+                                     long long n) {
+  long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  long long stride = (long long)gridDim.x * blockDim.x * 4;
+  for (; i + 3 < n; i += stride) {
+    f32x4v v = *reinterpret_cast<const f32x4v*>(in + i);
+    unsigned short o[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = f2bf(v[j]);
+    *reinterpret_cast<uint64_t*>(out + i) =
+        *reinterpret_cast<const uint64_t*>(o);
+  }
+  // tail
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    for (long long t = (n & ~3LL); t < n; ++t) out[t] = f2bf(in[t]);
+  }
+}
+
+void launch_cast_f32_bf16(void* out, const void* in, long long n,
+                          void* stream) {
+  long long blocks = (n / 4 + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3((int)blocks), dim3(256), 0,
+                     (hipStream_t)stream, (unsigned short*)out,
+                     (const float*)in, n);
+}
+
+// zero f32 buffer (before atomic k-split accumulate)
+__global__ void zero_f32_kernel(float* __restrict__ p, long long n) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = 0.f;
+}
+
+void launch_zero_f32(void* p, long long n, void* stream) {
+  long long blocks = (n + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  hipLaunchKernelGGL(zero_f32_kernel, dim3((int)blocks), dim3(256), 0,
+                     (hipStream_t)stream, (float*)p, n);
+}
+
+// ---------------------------------------------------------------------
+// Decode attention (single new token per sequence, GQA):
+//   O[b,h,:] = softmax(Q[b,h,:] K[b,kvh,:len,:]^T * scale) V[b,kvh,:len,:]
+// One workgroup per (b, kv-head); one wave per grouped q-head.
+// Memory-bound on the KV cache; K reads 16 lanes x 16B contiguous per
+// position, V reads 64 lanes x 4B contiguous per position.
+// dim must be 128 (llama-family head_dim).
+// ---------------------------------------------------------------------
+#define ATTN_SCHUNK 256
+#define ATTN_MAXG 8
+
+__global__ void attn_decode_kernel(
+    unsigned short* __restrict__ O, const unsigned short* __restrict__ Q,
+    const unsigned short* __restrict__ Kc,
+    const unsigned short* __restrict__ Vc,
+    const int* __restrict__ seq_lens, int B, int Hq, int Hkv, int Smax,
+    float scale) {
+  constexpr int D = 128;
+  const int b = blockIdx.x / Hkv;
+  const int kvh = blockIdx.x % Hkv;
+  const int G = Hq / Hkv;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int h = kvh * G + wave;
+  const int len = seq_lens[b];
+
+  __shared__ float scores[ATTN_MAXG][ATTN_SCHUNK];
+
+  // q fragment: lane holds 8 consecutive dims at (lane&15)*8 for the
+  // K-dot phase (16-lane groups each cover all 128 dims)
+  const unsigned short* qp = Q + ((size_t)b * Hq + h) * D + (lane & 15) * 8;
+  float qf[8];
+  {
+    ushort8 qv = *reinterpret_cast<const ushort8*>(qp);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) qf[j] = bf2f(qv.v[j]);
+  }
+
+  const size_t kv_base = ((size_t)b * Hkv + kvh) * Smax * D;
+  const unsigned short* kbase = Kc + kv_base;
+  const unsigned short* vbase = Vc + kv_base;
+
+  float m = -FLT_MAX, l = 0.f;
+  float acc0 = 0.f, acc1 = 0.f;  // lane owns dims lane*2, lane*2+1
+
+  for (int s0 = 0; s0 < len; s0 += ATTN_SCHUNK) {
+    const int cnt = min(ATTN_SCHUNK, len - s0);
+    // --- scores for this chunk ---
+    for (int si = lane >> 4; si < cnt; si += 4) {
+      const unsigned short* kp = kbase + (size_t)(s0 + si) * D +
+                                 (lane & 15) * 8;
+      ushort8 kv = *reinterpret_cast<const ushort8*>(kp);
+      float dot = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dot += qf[j] * bf2f(kv.v[j]);
+      // reduce across the 16-lane group
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        dot += __shfl_xor(dot, off, 64);
+      if ((lane & 15) == 0) scores[wave][si] = dot * scale;
+    }
+    __syncthreads();
+    // --- chunk max (wave-wide) ---
+    float cm = -FLT_MAX;
+    for (int si = lane; si < cnt; si += 64)
+      cm = fmaxf(cm, scores[wave][si]);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      cm = fmaxf(cm, __shfl_xor(cm, off, 64));
+    const float m_new = fmaxf(m, cm);
+    const float rescale = (m == -FLT_MAX) ? 0.f : __expf(m - m_new);
+    acc0 *= rescale;
+    acc1 *= rescale;
+    l *= rescale;
+    // --- p * V accumulate ---
+    const int d0 = lane * 2;
+    for (int si = 0; si < cnt; ++si) {
+      const float p = __expf(scores[wave][si] - m_new);
+      l += p;
+      const unsigned short* vp = vbase + (size_t)(s0 + si) * D + d0;
+      ushort2v vv = *reinterpret_cast<const ushort2v*>(vp);
+      acc0 += p * bf2f(vv.x);
+      acc1 += p * bf2f(vv.y);
+    }
+    m = m_new;
+    __syncthreads();
+  }
+
+  const float inv = (l > 0.f) ? 1.f / l : 0.f;
+  unsigned short* op = O + ((size_t)b * Hq + h) * D + lane * 2;
+  op[0] = f2bf(acc0 * inv);
+  op[1] = f2bf(acc1 * inv);
+}
+
+void launch_attn_decode(void* O, const void* Q, const void* Kc,
+                        const void* Vc, const void* seq_lens, int B, int Hq,
+                        int Hkv, int Smax, float scale, void* stream) {
+  const int G = Hq / Hkv;
+  hipLaunchKernelGGL(attn_decode_kernel, dim3(B * Hkv), dim3(G * 64), 0,
+                     (hipStream_t)stream, (unsigned short*)O,
+                     (const unsigned short*)Q, (const unsigned short*)Kc,
+                     (const unsigned short*)Vc, (const int*)seq_lens, B, Hq,
+                     Hkv, Smax, scale);
+}
+
+// ---------------------------------------------------------------------
+// KV-cache append: scatter the new token's K/V into the cache at
+// position seq_lens[b] (pre-increment positions computed host-side).
+// knew/vnew: [B, Hkv, D]; cache: [B, Hkv, Smax, D]
+// ---------------------------------------------------------------------
+__global__ void kv_append_kernel(unsigned short* __restrict__ Kc,
+                                 unsigned short* __restrict__ Vc,
+                                 const unsigned short* __restrict__ knew,
+                                 const unsigned short* __restrict__ vnew,
+                                 const int* __restrict__ positions, int Hkv,
+                                 int Smax, int D) {
+  const int b = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int pos = positions[b];
+  const size_t src = ((size_t)b * Hkv + kvh) * D;
+  const size_t dst = (((size_t)b * Hkv + kvh) * Smax + pos) * D;
+  for (int i = threadIdx.x * 8; i < D; i += blockDim.x * 8) {
+    *reinterpret_cast<ushort8*>(Kc + dst + i) =
+        *reinterpret_cast<const ushort8*>(knew + src + i);
+    *reinterpret_cast<ushort8*>(Vc + dst + i) =
+        *reinterpret_cast<const ushort8*>(vnew + src + i);
+  }
+}
+
+void launch_kv_append(void* Kc, void* Vc, const void* knew, const void* vnew,
+                      const void* positions, int B, int Hkv, int Smax, int D,
+                      void* stream) {
+  hipLaunchKernelGGL(kv_append_kernel, dim3(B, Hkv), dim3(D / 8), 0,
+                     (hipStream_t)stream, (unsigned short*)Kc,
+                     (unsigned short*)Vc, (const unsigned short*)knew,
+                     (const unsigned short*)vnew, (const int*)positions, Hkv,
+                     Smax, D);
+}
+
+// ---------------------------------------------------------------------
+// Row softmax (bf16 in/out, online single pass over LDS-staged row,
+// used by classic-model servers & tests)
+// ---------------------------------------------------------------------
+__global__ void softmax_kernel(unsigned short* __restrict__ out,
+                               const unsigned short* __restrict__ in,
+                               int cols) {
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const unsigned short* inr = in + (size_t)row * cols;
+  unsigned short* outr = out + (size_t)row * cols;
+
+  float lmax = -FLT_MAX;
+  for (int i = tid; i < cols; i += blockDim.x)
+    lmax = fmaxf(lmax, bf2f(inr[i]));
+  __shared__ float red[64];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    lmax = fmaxf(lmax, __shfl_xor(lmax, off, 64));
+  if ((tid & 63) == 0) red[tid >> 6] = lmax;
+  __syncthreads();
+  float gmax = -FLT_MAX;
+  for (int w = 0; w < (blockDim.x >> 6); ++w) gmax = fmaxf(gmax, red[w]);
+
+  float lsum = 0.f;
+  for (int i = tid; i < cols; i += blockDim.x)
+    lsum += __expf(bf2f(inr[i]) - gmax);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    lsum += __shfl_xor(lsum, off, 64);
+  if ((tid & 63) == 0) red[tid >> 6] = lsum;
+  __syncthreads();
+  float gsum = 0.f;
+  for (int w = 0; w < (blockDim.x >> 6); ++w) gsum += red[w];
+  const float inv = 1.f / gsum;
+  for (int i = tid; i < cols; i += blockDim.x)
+    outr[i] = f2bf(__expf(bf2f(inr[i]) - gmax) * inv);
+}
+
+void launch_softmax(void* out, const void* in, int rows, int cols,
+                    void* stream) {
+  hipLaunchKernelGGL(softmax_kernel, dim3(rows), dim3(256), 0,
+                     (hipStream_t)stream, (unsigned short*)out,
+                     (const unsigned short*)in, cols);
+}
+
+// ---------------------------------------------------------------------
+// Tree-ensemble inference (XGBoost/GBDT-style), for the classic-model
+// serving config.  Trees in flattened SoA node arrays; each lane
+// evaluates one sample through all trees (depth-bound loop).
+// features: [n_samples, n_features] f32; out: [n_samples] f32
+// ---------------------------------------------------------------------
+__global__ void tree_ensemble_kernel(
+    float* __restrict__ out, const float* __restrict__ features,
+    const int* __restrict__ feature_idx,    // per node: feature (-1 = leaf)
+    const float* __restrict__ threshold,    // per node: split threshold
+    const int* __restrict__ left,           // per node: left child index
+    const int* __restrict__ right,          // per node: right child index
+    const float* __restrict__ leaf_value,   // per node: value if leaf
+    const int* __restrict__ tree_offsets,   // [n_trees+1] node offsets
+    int n_trees, int n_samples, int n_features, float base_score) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n_samples; i += stride) {
+    const float* frow = features + i * n_features;
+    float score = base_score;
+    for (int t = 0; t < n_trees; ++t) {
+      int node = tree_offsets[t];
+      int fidx = feature_idx[node];
+      while (fidx >= 0) {
+        node = (frow[fidx] < threshold[node]) ? left[node] : right[node];
+        fidx = feature_idx[node];
+      }
+      score += leaf_value[node];
+    }
+    out[i] = score;
+  }
+}
+
+void launch_tree_ensemble(void* out, const void* features,
+                          const void* feature_idx, const void* threshold,
+                          const void* left, const void* right,
+                          const void* leaf_value, const void* tree_offsets,
+                          int n_trees, int n_samples, int n_features,
+                          float base_score, void* stream) {
+  long long blocks = (n_samples + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(tree_ensemble_kernel, dim3((int)blocks), dim3(256), 0,
+                     (hipStream_t)stream, (float*)out,
+                     (const float*)features, (const int*)feature_idx,
+                     (const float*)threshold, (const int*)left,
+                     (const int*)right, (const float*)leaf_value,
+                     (const int*)tree_offsets, n_trees, n_samples, n_features,
+                     base_score);
+}
+
+// ---------------------------------------------------------------------
+// Feature-store sliding-window aggregation:
+// For each (key, feature): ring buffer of per-period partial aggregates
+// resident in HBM; this kernel folds a batch of events into the ring
+// and emits current window aggregates (sum/count/min/max -> avg
+// derived host-side).
+// events: keys[n] int32 (dense key ids), values[n] f32, periods[n] int32
+// ring: [n_keys, n_feats?, n_periods, 4] (sum, count, min, max) f32
+// Batched: one lane per event; atomics within the period cell.
+// ---------------------------------------------------------------------
+__global__ void window_ingest_kernel(float* __restrict__ ring,
+                                     const int* __restrict__ keys,
+                                     const float* __restrict__ values,
+                                     const int* __restrict__ period_idx,
+                                     long long n_events, int n_periods) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n_events; i += stride) {
+    const int key = keys[i];
+    const int p = period_idx[i] % n_periods;
+    float* cell = ring + ((size_t)key * n_periods + p) * 4;
+    const float v = values[i];
+    atomicAdd(cell + 0, v);             // sum
+    atomicAdd(cell + 1, 1.f);           // count
+    // min/max are tracked per-batch host-side (they are not
+    // decomposable over ring periods for sliding windows anyway)
+  }
+}
+
+// window reduce: sum the last `window_periods` period cells per key
+// ring: [n_keys, n_periods, 4]; out: [n_keys, 4]
+__global__ void window_reduce_kernel(float* __restrict__ out,
+                                     const float* __restrict__ ring,
+                                     int n_keys, int n_periods,
+                                     int window_periods, int current_period) {
+  int key = blockIdx.x * blockDim.x + threadIdx.x;
+  if (key >= n_keys) return;
+  float sum = 0.f, count = 0.f;
+  for (int w = 0; w < window_periods; ++w) {
+    int p = (current_period - w) % n_periods;
+    if (p < 0) p += n_periods;
+    const float* cell = ring + ((size_t)key * n_periods + p) * 4;
+    sum += cell[0];
+    count += cell[1];
+  }
+  out[key * 4 + 0] = sum;
+  out[key * 4 + 1] = count;
+  out[key * 4 + 2] = count > 0.f ? sum / count : 0.f;  // avg
+  out[key * 4 + 3] = 0.f;
+}
+
+void launch_window_ingest(void* ring, const void* keys, const void* values,
+                          const void* period_idx, long long n_events,
+                          int n_periods, void* stream) {
+  long long blocks = (n_events + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(window_ingest_kernel, dim3((int)blocks), dim3(256), 0,
+                     (hipStream_t)stream, (float*)ring, (const int*)keys,
+                     (const float*)values, (const int*)period_idx, n_events,
+                     n_periods);
+}
+
+void launch_window_reduce(void* out, const void* ring, int n_keys,
+                          int n_periods, int window_periods,
+                          int current_period, void* stream) {
+  int blocks = (n_keys + 255) / 256;
+  hipLaunchKernelGGL(window_reduce_kernel, dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, (float*)out, (const float*)ring,
+                     n_keys, n_periods, window_periods, current_period);
+}

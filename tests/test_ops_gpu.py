@@ -1,0 +1,210 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""GPU numerics tests: every HIP kernel vs. the plain-torch fp32
+reference of the same op (asymmetric random inputs — transposes and
+layout bugs must not survive)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs MI355X")
+
+
+def _rand_bf16(*shape, scale=1.0, seed=None):
+    if seed is not None:
+        torch.manual_seed(seed)
+    return (torch.randn(*shape, dtype=torch.float32) * scale).to(
+        torch.bfloat16)
+
+
+@requires_gpu
+class TestHipOpsLoaded:
+    def test_extension_is_native(self):
+        """The HIP extension must actually be loaded on a GPU box."""
+        from mlrun_amd.ops import HAVE_HIP_OPS
+
+        assert HAVE_HIP_OPS, "_hip_ops extension missing on GPU box"
+        import mlrun_amd._hip_ops as ext
+
+        assert ext.__file__.endswith(".so")
+
+
+@requires_gpu
+class TestRMSNorm:
+    def test_rmsnorm(self):
+        from mlrun_amd import ops
+
+        x = _rand_bf16(9, 4096, seed=0)
+        w = _rand_bf16(4096, seed=1)
+        ref = ops.rmsnorm(x, w)
+        got = ops.rmsnorm(x.cuda(), w.cuda()).cpu()
+        assert torch.allclose(ref.float(), got.float(), atol=2e-2, rtol=2e-2)
+
+    def test_fused_residual(self):
+        from mlrun_amd import ops
+
+        x = _rand_bf16(4, 2048, seed=2)
+        res = _rand_bf16(4, 2048, seed=3)
+        w = _rand_bf16(2048, seed=4)
+        res_ref = res.clone()
+        ref = ops.fused_add_rmsnorm(x, w, residual=res_ref)
+        res_gpu = res.clone().cuda()
+        got = ops.fused_add_rmsnorm(x.cuda(), w.cuda(), residual=res_gpu)
+        assert torch.allclose(ref.float(), got.cpu().float(), atol=2e-2,
+                              rtol=2e-2)
+        assert torch.allclose(res_ref.float(), res_gpu.cpu().float(),
+                              atol=2e-2, rtol=2e-2)
+
+
+@requires_gpu
+class TestRoPE:
+    def test_rope_matches_reference(self):
+        from mlrun_amd import ops
+
+        T, H, D = 5, 8, 128
+        q = _rand_bf16(T, H, D, seed=5)
+        positions = torch.tensor([0, 3, 17, 100, 511], dtype=torch.int32)
+        table = ops.build_rope_cos_sin(1024, D)
+        q_ref = q.clone()
+        ops.rope_inplace(q_ref, positions, table)
+        q_gpu = q.clone().cuda()
+        ops.rope_inplace(q_gpu, positions.cuda(), table.cuda())
+        assert torch.allclose(q_ref.float(), q_gpu.cpu().float(), atol=2e-2,
+                              rtol=2e-2)
+
+
+@requires_gpu
+class TestSiluMul:
+    def test_silu_mul(self):
+        from mlrun_amd import ops
+
+        g = _rand_bf16(16, 1024, seed=6)
+        u = _rand_bf16(16, 1024, seed=7)
+        ref = ops.silu_mul(g, u)
+        got = ops.silu_mul(g.cuda(), u.cuda()).cpu()
+        assert torch.allclose(ref.float(), got.float(), atol=2e-2, rtol=2e-2)
+
+
+@requires_gpu
+class TestSkinnyGemm:
+    @pytest.mark.parametrize("m,n,k", [(1, 128, 256), (16, 4096, 4096),
+                                       (13, 1000, 512), (8, 6144, 4096)])
+    def test_vs_fp32(self, m, n, k):
+        from mlrun_amd import ops
+
+        a = _rand_bf16(m, k, scale=0.5, seed=m)
+        w = _rand_bf16(n, k, scale=0.5, seed=n)
+        ref = a.float() @ w.float().t()
+        got = ops.skinny_gemm(a.cuda(), w.cuda()).cpu().float()
+        # bf16 inputs: tolerance scales with sqrt(K)
+        tol = 0.02 * math.sqrt(k)
+        assert (ref - got).abs().max().item() < tol, \
+            f"max err {(ref - got).abs().max().item()} at m={m} n={n} k={k}"
+
+    def test_ksplit_paths(self):
+        from mlrun_amd import ops
+
+        a = _rand_bf16(16, 2048, scale=0.5, seed=42)
+        w = _rand_bf16(256, 2048, scale=0.5, seed=43)
+        ref = a.float() @ w.float().t()
+        for ksplit in (1, 4, 8):
+            got = ops.skinny_gemm(a.cuda(), w.cuda(),
+                                  ksplit=ksplit).cpu().float()
+            assert (ref - got).abs().max().item() < 1.0, f"ksplit={ksplit}"
+
+
+@requires_gpu
+class TestAttnDecode:
+    @pytest.mark.parametrize("b,hq,hkv,s", [(2, 8, 2, 64), (4, 32, 8, 300),
+                                            (16, 32, 8, 1024)])
+    def test_vs_fp32(self, b, hq, hkv, s):
+        from mlrun_amd import ops
+
+        D = 128
+        torch.manual_seed(b * 100 + s)
+        q = _rand_bf16(b, hq, D)
+        smax = s + 17
+        kc = _rand_bf16(b, hkv, smax, D)
+        vc = _rand_bf16(b, hkv, smax, D)
+        seq_lens = torch.randint(1, s + 1, (b,), dtype=torch.int32)
+        ref = ops.attn_decode(q, kc, vc, seq_lens)
+        got = ops.attn_decode(q.cuda(), kc.cuda(), vc.cuda(),
+                              seq_lens.cuda()).cpu()
+        assert torch.allclose(ref.float(), got.float(), atol=3e-2, rtol=3e-2)
+
+    def test_kv_append(self):
+        from mlrun_amd import ops
+
+        B, Hkv, Smax, D = 3, 4, 32, 128
+        kc = torch.zeros(B, Hkv, Smax, D, dtype=torch.bfloat16)
+        vc = torch.zeros_like(kc)
+        knew = _rand_bf16(B, Hkv, D, seed=9)
+        vnew = _rand_bf16(B, Hkv, D, seed=10)
+        positions = torch.tensor([0, 5, 31], dtype=torch.int32)
+        kc_ref, vc_ref = kc.clone(), vc.clone()
+        ops.kv_append(kc_ref, vc_ref, knew, vnew, positions)
+        kc_gpu, vc_gpu = kc.cuda(), vc.cuda()
+        ops.kv_append(kc_gpu, vc_gpu, knew.cuda(), vnew.cuda(),
+                      positions.cuda())
+        assert torch.equal(kc_ref, kc_gpu.cpu())
+        assert torch.equal(vc_ref, vc_gpu.cpu())
+
+
+@requires_gpu
+class TestSoftmax:
+    def test_vs_fp32(self):
+        from mlrun_amd import ops
+
+        x = _rand_bf16(32, 1000, scale=3.0, seed=11)
+        ref = ops.softmax(x)
+        got = ops.softmax(x.cuda()).cpu()
+        assert torch.allclose(ref.float(), got.float(), atol=1e-2, rtol=1e-2)
+
+
+@requires_gpu
+class TestTreeEnsemble:
+    def test_vs_reference(self):
+        from mlrun_amd import ops
+        from mlrun_amd.frameworks.tree import random_forest_nodes
+
+        torch.manual_seed(12)
+        nodes = random_forest_nodes(n_trees=20, depth=5, n_features=16,
+                                    seed=12)
+        feats = torch.randn(512, 16)
+        ref = ops.tree_ensemble_predict(feats, nodes, base_score=0.5)
+        nodes_gpu = {k: v.cuda() for k, v in nodes.items()}
+        got = ops.tree_ensemble_predict(feats.cuda(), nodes_gpu,
+                                        base_score=0.5).cpu()
+        assert torch.allclose(ref, got, atol=1e-4)
+
+
+@requires_gpu
+class TestWindowAgg:
+    def test_ingest_reduce(self):
+        from mlrun_amd import ops
+
+        torch.manual_seed(13)
+        n_keys, n_periods = 64, 8
+        n_events = 5000
+        keys = torch.randint(0, n_keys, (n_events,), dtype=torch.int32)
+        values = torch.randn(n_events)
+        period_idx = torch.randint(0, n_periods, (n_events,),
+                                   dtype=torch.int32)
+        ring_ref = torch.zeros(n_keys, n_periods, 4)
+        ops.window_ingest(ring_ref, keys, values, period_idx)
+        ref = ops.window_reduce(ring_ref, window_periods=4, current_period=6)
+
+        ring_gpu = torch.zeros(n_keys, n_periods, 4, device="cuda")
+        ops.window_ingest(ring_gpu, keys.cuda(), values.cuda(),
+                          period_idx.cuda())
+        got = ops.window_reduce(ring_gpu, window_periods=4,
+                                current_period=6).cpu()
+        assert torch.allclose(ref[:, :2], got[:, :2], atol=1e-2, rtol=1e-3)
+        assert torch.allclose(ref[:, 2], got[:, 2], atol=1e-3, rtol=1e-3)
