@@ -170,7 +170,8 @@ class LlamaDecodeEngine:
         self.buf_act = torch.zeros(B, w.inter, **bf16)
         self.buf_down = torch.zeros(B, h, **bf16)
         self.buf_logits = torch.zeros(B, cfg.vocab_size, **bf16)
-        self.buf_c32 = torch.zeros(B, max_n, dtype=torch.float32,
+        # flat f32 GEMM scratch: narrow(0, 0, B*N) stays contiguous
+        self.buf_c32 = torch.zeros(B * max_n, dtype=torch.float32,
                                    device=self.device)
         self.scale = 1.0 / math.sqrt(d)
         self._graph = None
@@ -184,8 +185,9 @@ class LlamaDecodeEngine:
         key = (N, K)
         if key not in self._ksplits:
             self._ksplits[key] = ops.pick_ksplit(a.shape[0], N, K)
-        c32 = self.buf_c32[:, :N].view(a.shape[0], N) \
-            if self.buf_c32.shape[1] >= N else None
+        M = a.shape[0]
+        c32 = self.buf_c32.narrow(0, 0, M * N).view(M, N) \
+            if self.buf_c32.numel() >= M * N else None
         return ops.skinny_gemm(a, w_, out=out, c_f32=c32,
                                ksplit=self._ksplits[key])
 
