@@ -225,8 +225,8 @@ class LlamaDecodeEngine:
             v = self.buf_qkv[:, (w.hq + w.hkv) * d:].view(B, w.hkv, d)
             ops.rope_inplace(q, positions, self.cos_sin)
             ops.rope_inplace(k, positions, self.cos_sin)
-            ops.kv_append(self.k_cache[li], self.v_cache[li],
-                          k.contiguous(), v.contiguous(), positions)
+            ops.kv_append(self.k_cache[li], self.v_cache[li], k, v,
+                          positions)
             attn_view = self.buf_attn_out.view(B, w.hq, d)
             ops.attn_decode(q, self.k_cache[li], self.v_cache[li],
                             self.cache_lens, self.scale, out=attn_view)

@@ -46,16 +46,25 @@ void fused_add_rmsnorm(torch::Tensor out, torch::Tensor x,
                            (float)eps, stream());
 }
 
+// q may be a strided row-view into a fused qkv buffer:
+// required layout [T, heads, dim] with strides (row_stride, dim, 1)
+static long long check_head_view(const torch::Tensor& q) {
+  TORCH_CHECK(q.dim() == 3, "expected [T, heads, dim]");
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2),
+              "inner dims must be dense (strides (*, dim, 1))");
+  return (long long)q.stride(0);
+}
+
 void rope(torch::Tensor q, torch::Tensor positions, torch::Tensor cos_sin) {
-  CHECK_DEV(q); CHECK_CONTIG(q); CHECK_BF16(q);
+  CHECK_DEV(q); CHECK_BF16(q);
   CHECK_DEV(positions); CHECK_CONTIG(positions); CHECK_I32(positions);
   CHECK_DEV(cos_sin); CHECK_CONTIG(cos_sin); CHECK_F32(cos_sin);
-  TORCH_CHECK(q.dim() == 3, "q must be [T, heads, dim]");
+  long long row_stride = check_head_view(q);
   int T = q.size(0), heads = q.size(1), dim = q.size(2);
   TORCH_CHECK(dim % 2 == 0 && dim / 2 <= 1024, "bad head dim");
   TORCH_CHECK(positions.numel() == T, "positions size mismatch");
   launch_rope(q.data_ptr(), positions.data_ptr(), cos_sin.data_ptr(), T,
-              heads, dim, stream());
+              heads, dim, row_stride, stream());
 }
 
 void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up) {
@@ -95,11 +104,12 @@ void cast_f32_bf16(torch::Tensor out, torch::Tensor in) {
 void attn_decode(torch::Tensor o, torch::Tensor q, torch::Tensor kc,
                  torch::Tensor vc, torch::Tensor seq_lens, double scale) {
   CHECK_DEV(o); CHECK_CONTIG(o); CHECK_BF16(o);
-  CHECK_DEV(q); CHECK_CONTIG(q); CHECK_BF16(q);
+  CHECK_DEV(q); CHECK_BF16(q);
   CHECK_DEV(kc); CHECK_CONTIG(kc); CHECK_BF16(kc);
   CHECK_DEV(vc); CHECK_CONTIG(vc); CHECK_BF16(vc);
   CHECK_DEV(seq_lens); CHECK_CONTIG(seq_lens); CHECK_I32(seq_lens);
   TORCH_CHECK(q.dim() == 3 && kc.dim() == 4, "q [B,Hq,D], kc [B,Hkv,S,D]");
+  long long q_row_stride = check_head_view(q);
   int B = q.size(0), Hq = q.size(1), D = q.size(2);
   int Hkv = kc.size(1), Smax = kc.size(2);
   TORCH_CHECK(D == 128, "attn_decode requires head_dim 128");
@@ -108,19 +118,22 @@ void attn_decode(torch::Tensor o, torch::Tensor q, torch::Tensor kc,
   TORCH_CHECK(kc.size(0) == B && kc.size(3) == D, "kc shape mismatch");
   launch_attn_decode(o.data_ptr(), q.data_ptr(), kc.data_ptr(),
                      vc.data_ptr(), seq_lens.data_ptr(), B, Hq, Hkv, Smax,
-                     (float)scale, stream());
+                     (float)scale, q_row_stride, stream());
 }
 
 void kv_append(torch::Tensor kc, torch::Tensor vc, torch::Tensor knew,
                torch::Tensor vnew, torch::Tensor positions) {
   CHECK_DEV(kc); CHECK_CONTIG(kc); CHECK_BF16(kc);
-  CHECK_DEV(knew); CHECK_CONTIG(knew); CHECK_BF16(knew);
+  CHECK_DEV(knew); CHECK_BF16(knew);
   CHECK_DEV(positions); CHECK_CONTIG(positions); CHECK_I32(positions);
+  long long src_stride = check_head_view(knew);
+  TORCH_CHECK(check_head_view(vnew) == src_stride,
+              "knew/vnew strides must match");
   int B = kc.size(0), Hkv = kc.size(1), Smax = kc.size(2), D = kc.size(3);
   TORCH_CHECK(D % 8 == 0, "head_dim must be a multiple of 8");
   launch_kv_append(kc.data_ptr(), vc.data_ptr(), knew.data_ptr(),
                    vnew.data_ptr(), positions.data_ptr(), B, Hkv, Smax, D,
-                   stream());
+                   src_stride, stream());
 }
 
 void softmax(torch::Tensor out, torch::Tensor in) {

@@ -14,7 +14,7 @@ void launch_fused_add_rmsnorm(void* out, void* residual, const void* x,
                               float eps, void* stream);
 
 void launch_rope(void* q, const void* positions, const void* cos_sin, int T,
-                 int heads, int dim, void* stream);
+                 int heads, int dim, long long row_stride, void* stream);
 
 void launch_silu_mul(void* out, const void* gate, const void* up,
                      long long n, void* stream);
@@ -29,11 +29,12 @@ void launch_zero_f32(void* p, long long n, void* stream);
 
 void launch_attn_decode(void* O, const void* Q, const void* Kc,
                         const void* Vc, const void* seq_lens, int B, int Hq,
-                        int Hkv, int Smax, float scale, void* stream);
+                        int Hkv, int Smax, float scale,
+                        long long q_row_stride, void* stream);
 
 void launch_kv_append(void* Kc, void* Vc, const void* knew, const void* vnew,
                       const void* positions, int B, int Hkv, int Smax, int D,
-                      void* stream);
+                      long long src_row_stride, void* stream);
 
 void launch_softmax(void* out, const void* in, int rows, int cols,
                     void* stream);

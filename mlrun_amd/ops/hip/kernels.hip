@@ -136,7 +136,7 @@ void launch_fused_add_rmsnorm(void* out, void* residual, const void* x,
 __global__ void rope_kernel(unsigned short* __restrict__ q,
                             const int* __restrict__ positions,
                             const float* __restrict__ cos_sin, int heads,
-                            int dim) {
+                            int dim, long long row_stride) {
   const int t = blockIdx.x;
   const int h = blockIdx.y;
   const int i = threadIdx.x;  // 0..D/2-1
@@ -145,7 +145,7 @@ __global__ void rope_kernel(unsigned short* __restrict__ q,
   const int pos = positions[t];
   const float c = cos_sin[((size_t)pos * half + i) * 2];
   const float s = cos_sin[((size_t)pos * half + i) * 2 + 1];
-  unsigned short* base = q + ((size_t)t * heads + h) * dim;
+  unsigned short* base = q + (size_t)t * row_stride + (size_t)h * dim;
   float a = bf2f(base[i]);
   float b = bf2f(base[i + half]);
   base[i] = f2bf(a * c - b * s);
@@ -153,10 +153,11 @@ __global__ void rope_kernel(unsigned short* __restrict__ q,
 }
 
 void launch_rope(void* q, const void* positions, const void* cos_sin, int T,
-                 int heads, int dim, void* stream) {
+                 int heads, int dim, long long row_stride, void* stream) {
   hipLaunchKernelGGL(rope_kernel, dim3(T, heads), dim3(dim / 2), 0,
                      (hipStream_t)stream, (unsigned short*)q,
-                     (const int*)positions, (const float*)cos_sin, heads, dim);
+                     (const int*)positions, (const float*)cos_sin, heads, dim,
+                     row_stride);
 }
 
 // ---------------------------------------------------------------------
@@ -335,7 +336,7 @@ __global__ void attn_decode_kernel(
     const unsigned short* __restrict__ Kc,
     const unsigned short* __restrict__ Vc,
     const int* __restrict__ seq_lens, int B, int Hq, int Hkv, int Smax,
-    float scale) {
+    float scale, long long q_row_stride) {
   constexpr int D = 128;
   const int b = blockIdx.x / Hkv;
   const int kvh = blockIdx.x % Hkv;
@@ -349,7 +350,8 @@ __global__ void attn_decode_kernel(
 
   // q fragment: lane holds 8 consecutive dims at (lane&15)*8 for the
   // K-dot phase (16-lane groups each cover all 128 dims)
-  const unsigned short* qp = Q + ((size_t)b * Hq + h) * D + (lane & 15) * 8;
+  const unsigned short* qp = Q + (size_t)b * q_row_stride +
+                             (size_t)h * D + (lane & 15) * 8;
   float qf[8];
   {
     ushort8 qv = *reinterpret_cast<const ushort8*>(qp);
@@ -415,13 +417,14 @@ __global__ void attn_decode_kernel(
 
 void launch_attn_decode(void* O, const void* Q, const void* Kc,
                         const void* Vc, const void* seq_lens, int B, int Hq,
-                        int Hkv, int Smax, float scale, void* stream) {
+                        int Hkv, int Smax, float scale,
+                        long long q_row_stride, void* stream) {
   const int G = Hq / Hkv;
   hipLaunchKernelGGL(attn_decode_kernel, dim3(B * Hkv), dim3(G * 64), 0,
                      (hipStream_t)stream, (unsigned short*)O,
                      (const unsigned short*)Q, (const unsigned short*)Kc,
                      (const unsigned short*)Vc, (const int*)seq_lens, B, Hq,
-                     Hkv, Smax, scale);
+                     Hkv, Smax, scale, q_row_stride);
 }
 
 // ---------------------------------------------------------------------
@@ -434,11 +437,11 @@ __global__ void kv_append_kernel(unsigned short* __restrict__ Kc,
                                  const unsigned short* __restrict__ knew,
                                  const unsigned short* __restrict__ vnew,
                                  const int* __restrict__ positions, int Hkv,
-                                 int Smax, int D) {
+                                 int Smax, int D, long long src_row_stride) {
   const int b = blockIdx.x;
   const int kvh = blockIdx.y;
   const int pos = positions[b];
-  const size_t src = ((size_t)b * Hkv + kvh) * D;
+  const size_t src = (size_t)b * src_row_stride + (size_t)kvh * D;
   const size_t dst = (((size_t)b * Hkv + kvh) * Smax + pos) * D;
   for (int i = threadIdx.x * 8; i < D; i += blockDim.x * 8) {
     *reinterpret_cast<ushort8*>(Kc + dst + i) =
@@ -450,12 +453,12 @@ __global__ void kv_append_kernel(unsigned short* __restrict__ Kc,
 
 void launch_kv_append(void* Kc, void* Vc, const void* knew, const void* vnew,
                       const void* positions, int B, int Hkv, int Smax, int D,
-                      void* stream) {
+                      long long src_row_stride, void* stream) {
   hipLaunchKernelGGL(kv_append_kernel, dim3(B, Hkv), dim3(D / 8), 0,
                      (hipStream_t)stream, (unsigned short*)Kc,
                      (unsigned short*)Vc, (const unsigned short*)knew,
                      (const unsigned short*)vnew, (const int*)positions, Hkv,
-                     Smax, D);
+                     Smax, D, src_row_stride);
 }
 
 // ---------------------------------------------------------------------
