@@ -255,6 +255,7 @@ class LlamaDecodeEngine:
         if not self.use_graph or self._graph is not None:
             return
         lens_backup = self.cache_lens.clone()
+        tokens_backup = self.buf_tokens.clone()
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
@@ -262,10 +263,14 @@ class LlamaDecodeEngine:
                 self._decode_step_body()
         torch.cuda.current_stream().wait_stream(side)
         self.cache_lens.copy_(lens_backup)
+        self.buf_tokens.copy_(tokens_backup)
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
             self._decode_step_body()
+        # stream capture records without executing, but restore anyway
+        # in case a backend executed eagerly during capture
         self.cache_lens.copy_(lens_backup)
+        self.buf_tokens.copy_(tokens_backup)
         self._graph = graph
         logger.info("decode hipGraph captured", batch=self.B,
                     layers=self.cfg.num_layers)
