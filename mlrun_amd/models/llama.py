@@ -97,8 +97,8 @@ class LlamaWeights:
                 "wo": w(h, hq * d),
                 "ffn_norm": torch.ones(h, dtype=torch.bfloat16,
                                        device=device),
-                "wgate": w(inter, h),
-                "wup": w(inter, h),
+                # fused gate|up: one [2*inter, h] GEMM per mlp
+                "wgu": w(2 * inter, h),
                 "wdown": w(h, inter),
             })
         self.final_norm = torch.ones(h, dtype=torch.bfloat16, device=device)
@@ -157,7 +157,10 @@ class LlamaDecodeEngine:
 
         # preallocated decode buffers (graph-capture requirement)
         qkv_n = (w.hq + 2 * w.hkv) * d
-        max_n = max(qkv_n, h, 2 * w.inter, cfg.vocab_size)
+        # split-K scratch: ksplit * B * N f32 for the largest case
+        gemm_shapes = [(qkv_n, h), (h, w.hq * d), (2 * w.inter, h),
+                       (h, w.inter), (cfg.vocab_size, h)]
+        scratch = max(ops.pick_ksplit(B, n, k) * n for n, k in gemm_shapes)
         self.buf_tokens = torch.zeros(B, dtype=torch.int64,
                                       device=self.device)
         self.buf_hidden = torch.zeros(B, h, **bf16)
@@ -165,13 +168,12 @@ class LlamaDecodeEngine:
         self.buf_qkv = torch.zeros(B, qkv_n, **bf16)
         self.buf_attn_out = torch.zeros(B, w.hq * d, **bf16)
         self.buf_proj = torch.zeros(B, h, **bf16)
-        self.buf_gate = torch.zeros(B, w.inter, **bf16)
-        self.buf_up = torch.zeros(B, w.inter, **bf16)
+        self.buf_gu = torch.zeros(B, 2 * w.inter, **bf16)
         self.buf_act = torch.zeros(B, w.inter, **bf16)
         self.buf_down = torch.zeros(B, h, **bf16)
         self.buf_logits = torch.zeros(B, cfg.vocab_size, **bf16)
-        # flat f32 GEMM scratch: narrow(0, 0, B*N) stays contiguous
-        self.buf_c32 = torch.zeros(B * max_n, dtype=torch.float32,
+        # flat f32 split-K scratch (shared across all decode GEMMs)
+        self.buf_c32 = torch.empty(B * scratch, dtype=torch.float32,
                                    device=self.device)
         self.scale = 1.0 / math.sqrt(d)
         self._graph = None
@@ -180,15 +182,10 @@ class LlamaDecodeEngine:
     # ------------------------------------------------------------ gemm
     def _gemm(self, a, w_, out):
         """skinny GEMM into a preallocated bf16 out + f32 scratch."""
-        N = w_.shape[0]
-        K = w_.shape[1]
-        key = (N, K)
+        key = (w_.shape[0], w_.shape[1])
         if key not in self._ksplits:
-            self._ksplits[key] = ops.pick_ksplit(a.shape[0], N, K)
-        M = a.shape[0]
-        c32 = self.buf_c32.narrow(0, 0, M * N).view(M, N) \
-            if self.buf_c32.numel() >= M * N else None
-        return ops.skinny_gemm(a, w_, out=out, c_f32=c32,
+            self._ksplits[key] = ops.pick_ksplit(a.shape[0], *key)
+        return ops.skinny_gemm(a, w_, out=out, c_f32=self.buf_c32,
                                ksplit=self._ksplits[key])
 
     def _maybe_allreduce(self, t):
@@ -219,14 +216,10 @@ class LlamaDecodeEngine:
         for li, layer in enumerate(w.layers):
             # qkv projection
             self._gemm(self.buf_hidden, layer["wqkv"], self.buf_qkv)
+            ops.rope_kv_fused(self.buf_qkv, self.k_cache[li],
+                              self.v_cache[li], positions, self.cos_sin,
+                              w.hq)
             q = self.buf_qkv[:, :w.hq * d].view(B, w.hq, d)
-            k = self.buf_qkv[:, w.hq * d:(w.hq + w.hkv) * d].view(
-                B, w.hkv, d)
-            v = self.buf_qkv[:, (w.hq + w.hkv) * d:].view(B, w.hkv, d)
-            ops.rope_inplace(q, positions, self.cos_sin)
-            ops.rope_inplace(k, positions, self.cos_sin)
-            ops.kv_append(self.k_cache[li], self.v_cache[li], k, v,
-                          positions)
             attn_view = self.buf_attn_out.view(B, w.hq, d)
             ops.attn_decode(q, self.k_cache[li], self.v_cache[li],
                             self.cache_lens, self.scale, out=attn_view)
@@ -235,10 +228,9 @@ class LlamaDecodeEngine:
             ops.fused_add_rmsnorm(self.buf_proj, layer["ffn_norm"],
                                   residual=self.buf_residual,
                                   eps=cfg.rms_eps, out=self.buf_hidden)
-            # mlp
-            self._gemm(self.buf_hidden, layer["wgate"], self.buf_gate)
-            self._gemm(self.buf_hidden, layer["wup"], self.buf_up)
-            ops.silu_mul(self.buf_gate, self.buf_up, out=self.buf_act)
+            # mlp (fused gate|up projection)
+            self._gemm(self.buf_hidden, layer["wgu"], self.buf_gu)
+            ops.swiglu_fused(self.buf_gu, out=self.buf_act)
             self._gemm(self.buf_act, layer["wdown"], self.buf_down)
             self._maybe_allreduce(self.buf_down)
             next_norm = w.layers[li + 1]["attn_norm"] \
@@ -317,9 +309,8 @@ class LlamaDecodeEngine:
             hidden = ops.fused_add_rmsnorm(proj, layer["ffn_norm"],
                                            residual=residual,
                                            eps=cfg.rms_eps)
-            gate = hidden @ layer["wgate"].t()
-            up = hidden @ layer["wup"].t()
-            act = ops.silu_mul(gate, up)
+            gu = hidden @ layer["wgu"].t()
+            act = ops.swiglu_fused(gu.contiguous())
             down = act @ layer["wdown"].t()
             self._maybe_allreduce(down)
             next_norm = w.layers[li + 1]["attn_norm"] \

@@ -106,6 +106,24 @@ def rope_inplace(q: torch.Tensor, positions: torch.Tensor,
 # ---------------------------------------------------------------- swiglu
 
 
+def swiglu_fused(gu: torch.Tensor, out: torch.Tensor = None) -> torch.Tensor:
+    """silu(gu[:, :I]) * gu[:, I:] over the fused gate|up projection."""
+    rows, two_i = gu.shape
+    inter = two_i // 2
+    if gu.is_cuda:
+        ops = _require_hip()
+        if out is None:
+            out = torch.empty(rows, inter, dtype=gu.dtype, device=gu.device)
+        ops.swiglu_fused(out, gu)
+        return out
+    result = (torch.nn.functional.silu(gu[:, :inter].float()) *
+              gu[:, inter:].float()).to(gu.dtype)
+    if out is not None:
+        out.copy_(result)
+        return out
+    return result
+
+
 def silu_mul(gate: torch.Tensor, up: torch.Tensor,
              out: torch.Tensor = None) -> torch.Tensor:
     if gate.is_cuda:
@@ -125,8 +143,8 @@ def silu_mul(gate: torch.Tensor, up: torch.Tensor,
 # ------------------------------------------------------------------ gemm
 
 
-def pick_ksplit(M: int, N: int, K: int, target_blocks: int = 1024) -> int:
-    """Choose the K-split so the grid fills 256 CUs (>=~1024 wgs)."""
+def pick_ksplit(M: int, N: int, K: int, target_blocks: int = 2048) -> int:
+    """Choose the K-split so the grid fills 256 CUs (~2048 wgs)."""
     nblocks = (N + 127) // 128
     if nblocks >= target_blocks:
         return 1
@@ -135,32 +153,62 @@ def pick_ksplit(M: int, N: int, K: int, target_blocks: int = 1024) -> int:
     return max(ksplit, 1)
 
 
+_EMPTY_F32 = None
+
+
 def skinny_gemm(a: torch.Tensor, w: torch.Tensor,
                 out: torch.Tensor = None,
                 c_f32: torch.Tensor = None,
                 ksplit: int = None) -> torch.Tensor:
     """C[M,N] = A[M,K] @ W[N,K]^T for decode batches (M <= 16), bf16.
 
-    On GPU: MFMA kernel with K-split atomics into a f32 workspace, then
-    cast to bf16.  Pass preallocated out/c_f32 for graph capture."""
+    On GPU: MFMA kernel; split-K writes per-chunk f32 slabs into c_f32
+    and a reduce kernel folds them to bf16 (deterministic, no atomics).
+    Pass preallocated out/c_f32 for graph capture."""
+    global _EMPTY_F32
+
     M, K = a.shape
     N = w.shape[0]
     if a.is_cuda:
         ops = _require_hip()
-        if c_f32 is None:
-            c_f32 = torch.empty(M, N, dtype=torch.float32, device=a.device)
         if ksplit is None:
             ksplit = pick_ksplit(M, N, K)
-        ops.skinny_gemm(c_f32, a, w, ksplit)
         if out is None:
             out = torch.empty(M, N, dtype=torch.bfloat16, device=a.device)
-        ops.cast_f32_bf16(out, c_f32)
+        if ksplit > 1 and c_f32 is None:
+            c_f32 = torch.empty(ksplit * M * N, dtype=torch.float32,
+                                device=a.device)
+        if c_f32 is None:
+            if _EMPTY_F32 is None or _EMPTY_F32.device != a.device:
+                _EMPTY_F32 = torch.empty(1, dtype=torch.float32,
+                                         device=a.device)
+            c_f32 = _EMPTY_F32
+        ops.skinny_gemm(out, c_f32, a, w, ksplit)
         return out
     result = (a.float() @ w.float().t()).to(a.dtype)
     if out is not None:
         out.copy_(result)
         return out
     return result
+
+
+def rope_kv_fused(qkv: torch.Tensor, k_cache: torch.Tensor,
+                  v_cache: torch.Tensor, positions: torch.Tensor,
+                  cos_sin: torch.Tensor, hq: int):
+    """Fused decode rope(q,k) + KV append from the fused qkv buffer
+    [B, (hq+2*hkv)*d]."""
+    B = qkv.shape[0]
+    hkv, d = k_cache.shape[1], k_cache.shape[3]
+    if qkv.is_cuda:
+        _require_hip().rope_kv_fused(qkv, k_cache, v_cache, positions,
+                                     cos_sin, hq)
+        return
+    q = qkv[:, :hq * d].view(B, hq, d)
+    k = qkv[:, hq * d:(hq + hkv) * d].view(B, hkv, d)
+    v = qkv[:, (hq + hkv) * d:(hq + 2 * hkv) * d].view(B, hkv, d)
+    rope_inplace(q, positions, cos_sin)
+    rope_inplace(k, positions, cos_sin)
+    kv_append(k_cache, v_cache, k, v, positions)
 
 
 # ------------------------------------------------------------- attention

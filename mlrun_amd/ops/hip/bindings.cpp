@@ -67,6 +67,16 @@ void rope(torch::Tensor q, torch::Tensor positions, torch::Tensor cos_sin) {
               heads, dim, row_stride, stream());
 }
 
+void swiglu_fused(torch::Tensor out, torch::Tensor gu) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
+  CHECK_DEV(gu); CHECK_CONTIG(gu); CHECK_BF16(gu);
+  TORCH_CHECK(gu.dim() == 2 && out.dim() == 2, "need 2-D tensors");
+  int rows = gu.size(0), inter = gu.size(1) / 2;
+  TORCH_CHECK(out.size(0) == rows && out.size(1) == inter, "shape mismatch");
+  TORCH_CHECK(inter % 8 == 0, "inter must be a multiple of 8");
+  launch_swiglu_fused(out.data_ptr(), gu.data_ptr(), rows, inter, stream());
+}
+
 void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up) {
   CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
   CHECK_DEV(gate); CHECK_CONTIG(gate); CHECK_BF16(gate);
@@ -78,20 +88,43 @@ void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up) {
                   out.numel(), stream());
 }
 
-// C_f32 [M,N] (pre-zeroed when ksplit>1) <- A [M,K] @ W [N,K]^T
-void skinny_gemm(torch::Tensor c_f32, torch::Tensor a, torch::Tensor w,
-                 int64_t ksplit) {
-  CHECK_DEV(c_f32); CHECK_CONTIG(c_f32); CHECK_F32(c_f32);
+// out_bf16 [M,N] <- A [M,K] @ W [N,K]^T.  part_f32 is the split-K
+// scratch ([ksplit*M*N] f32, unused when ksplit==1 — pass any tensor).
+void skinny_gemm(torch::Tensor out_bf16, torch::Tensor part_f32,
+                 torch::Tensor a, torch::Tensor w, int64_t ksplit) {
+  CHECK_DEV(out_bf16); CHECK_CONTIG(out_bf16); CHECK_BF16(out_bf16);
   CHECK_DEV(a); CHECK_CONTIG(a); CHECK_BF16(a);
   CHECK_DEV(w); CHECK_CONTIG(w); CHECK_BF16(w);
   int M = a.size(0), K = a.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "K mismatch");
   TORCH_CHECK(M <= 16, "skinny_gemm supports M <= 16");
   TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32");
-  TORCH_CHECK(c_f32.size(0) == M && c_f32.size(1) == N, "C shape mismatch");
-  if (ksplit > 1) launch_zero_f32(c_f32.data_ptr(), c_f32.numel(), stream());
-  launch_skinny_gemm(c_f32.data_ptr(), a.data_ptr(), w.data_ptr(), M, N, K,
-                     (int)ksplit, stream());
+  TORCH_CHECK(out_bf16.numel() == (int64_t)M * N, "out shape mismatch");
+  if (ksplit > 1) {
+    CHECK_DEV(part_f32); CHECK_CONTIG(part_f32); CHECK_F32(part_f32);
+    TORCH_CHECK(part_f32.numel() >= ksplit * (int64_t)M * N,
+                "part_f32 scratch too small");
+  }
+  launch_skinny_gemm(out_bf16.data_ptr(),
+                     ksplit > 1 ? part_f32.data_ptr() : nullptr,
+                     a.data_ptr(), w.data_ptr(), M, N, K, (int)ksplit,
+                     stream());
+}
+
+void rope_kv_fused(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
+                   torch::Tensor positions, torch::Tensor cos_sin,
+                   int64_t hq) {
+  CHECK_DEV(qkv); CHECK_CONTIG(qkv); CHECK_BF16(qkv);
+  CHECK_DEV(kc); CHECK_CONTIG(kc); CHECK_BF16(kc);
+  CHECK_DEV(vc); CHECK_CONTIG(vc); CHECK_BF16(vc);
+  CHECK_DEV(positions); CHECK_CONTIG(positions); CHECK_I32(positions);
+  CHECK_DEV(cos_sin); CHECK_CONTIG(cos_sin); CHECK_F32(cos_sin);
+  int B = kc.size(0), Hkv = kc.size(1), Smax = kc.size(2), D = kc.size(3);
+  TORCH_CHECK(qkv.dim() == 2 && qkv.size(0) == B, "qkv must be [B, rows]");
+  TORCH_CHECK(qkv.size(1) >= (hq + 2 * Hkv) * D, "qkv row too small");
+  launch_rope_kv_fused(qkv.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                       positions.data_ptr(), cos_sin.data_ptr(), B, (int)hq,
+                       Hkv, Smax, D, qkv.size(1), stream());
 }
 
 void cast_f32_bf16(torch::Tensor out, torch::Tensor in) {
@@ -189,8 +222,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused residual add + RMSNorm (bf16)");
   m.def("rope", &rope, "in-place rotary embedding (bf16)");
   m.def("silu_mul", &silu_mul, "silu(gate) * up (bf16)");
+  m.def("swiglu_fused", &swiglu_fused,
+        "silu(gu[:, :I]) * gu[:, I:] over a fused gate|up buffer");
   m.def("skinny_gemm", &skinny_gemm,
-        "decode GEMM C=A@W^T on MFMA (bf16 in, f32 out)");
+        "decode GEMM C=A@W^T on MFMA (bf16 in/out, split-K f32 slabs)");
+  m.def("rope_kv_fused", &rope_kv_fused,
+        "fused decode rope(q,k) + KV-cache append");
   m.def("cast_f32_bf16", &cast_f32_bf16, "f32 -> bf16 cast");
   m.def("attn_decode", &attn_decode, "GQA decode attention (bf16)");
   m.def("kv_append", &kv_append, "append token K/V into cache");

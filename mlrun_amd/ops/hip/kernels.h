@@ -16,11 +16,20 @@ void launch_fused_add_rmsnorm(void* out, void* residual, const void* x,
 void launch_rope(void* q, const void* positions, const void* cos_sin, int T,
                  int heads, int dim, long long row_stride, void* stream);
 
+void launch_swiglu_fused(void* out, const void* gu, int rows, int inter,
+                         void* stream);
+
 void launch_silu_mul(void* out, const void* gate, const void* up,
                      long long n, void* stream);
 
-void launch_skinny_gemm(void* C, const void* A, const void* W, int M, int N,
-                        int K, int ksplit, void* stream);
+void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
+                        const void* W, int M, int N, int K, int ksplit,
+                        void* stream);
+
+void launch_rope_kv_fused(void* qkv, void* Kc, void* Vc,
+                          const void* positions, const void* cos_sin, int B,
+                          int Hq, int Hkv, int Smax, int D,
+                          long long row_stride, void* stream);
 
 void launch_cast_f32_bf16(void* out, const void* in, long long n,
                           void* stream);

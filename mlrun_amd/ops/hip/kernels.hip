@@ -184,6 +184,42 @@ __global__ void silu_mul_kernel(unsigned short* __restrict__ out,
   }
 }
 
+// SwiGLU over a fused gate|up buffer: gu [rows, 2*inter] ->
+// out [rows, inter] = silu(gu[:, :inter]) * gu[:, inter:]
+__global__ void swiglu_fused_kernel(unsigned short* __restrict__ out,
+                                    const unsigned short* __restrict__ gu,
+                                    int rows, int inter) {
+  const long long total = (long long)rows * inter / 8;
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    const long long flat = i * 8;
+    const int r = (int)(flat / inter);
+    const int c = (int)(flat % inter);
+    const unsigned short* row = gu + (size_t)r * 2 * inter;
+    ushort8 g = *reinterpret_cast<const ushort8*>(row + c);
+    ushort8 u = *reinterpret_cast<const ushort8*>(row + inter + c);
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g.v[j]);
+      float s = gf / (1.f + __expf(-gf));
+      o.v[j] = f2bf(s * bf2f(u.v[j]));
+    }
+    *reinterpret_cast<ushort8*>(out + (size_t)r * inter + c) = o;
+  }
+}
+
+void launch_swiglu_fused(void* out, const void* gu, int rows, int inter,
+                         void* stream) {
+  long long blocks = ((long long)rows * inter / 8 + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(swiglu_fused_kernel, dim3((int)blocks), dim3(256), 0,
+                     (hipStream_t)stream, (unsigned short*)out,
+                     (const unsigned short*)gu, rows, inter);
+}
+
 void launch_silu_mul(void* out, const void* gate, const void* up,
                      long long n, void* stream) {
   long long blocks = (n / 8 + 255) / 256;
@@ -206,18 +242,25 @@ void launch_silu_mul(void* out, const void* gate, const void* up,
 //   B: lane l holds W[n0 + (l&15)][(l>>4)*8 + j] (B^T = W row-major)
 //   C: lane l, reg r -> C[(l>>4)*4 + r][l&15]
 // ---------------------------------------------------------------------
+// SPLIT=false: ksplit==1, epilogue writes bf16 straight to `out`.
+// SPLIT=true:  blockIdx.y writes its f32 partial slab part[y][M][N]
+//              (plain stores — no atomics, no pre-zero, deterministic);
+//              reduce_cast_kernel folds the slabs to bf16.
+template <bool SPLIT>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
-    float* __restrict__ C, const unsigned short* __restrict__ A,
-    const unsigned short* __restrict__ W, int M, int N, int K, int ksplit,
-    int use_atomic) {
+    void* __restrict__ out, const unsigned short* __restrict__ A,
+    const unsigned short* __restrict__ W, int M, int N, int K, int ksplit) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int n0 = blockIdx.x * 128 + wave * 32;  // this wave's n-tile
   if (n0 >= N) return;
-  const int kchunk = (K / ksplit + 31) & ~31;  // multiple of 32
-  const int kbegin = blockIdx.y * kchunk;
-  int kend = kbegin + kchunk;
-  if (kend > K) kend = K;
+  int kbegin = 0, kend = K;
+  if (SPLIT) {
+    const int kchunk = (K / ksplit + 31) & ~31;  // multiple of 32
+    kbegin = blockIdx.y * kchunk;
+    kend = kbegin + kchunk;
+    if (kend > K) kend = K;
+  }
 
   const int arow = lane & 15;
   const int kb = (lane >> 4) * 8;
@@ -235,6 +278,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   const unsigned short* bptr0 = W + (size_t)brow0 * K + kb;
   const unsigned short* bptr1 = W + (size_t)brow1 * K + kb;
 
+#pragma unroll 4
   for (int k = kbegin; k < kend; k += 32) {
     short8v af = a_valid
         ? *reinterpret_cast<const short8v*>(aptr + k) : zero8;
@@ -249,31 +293,78 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
+  if (SPLIT) {
+    float* part = (float*)out + (size_t)blockIdx.y * M * N;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int m = crow_base + r;
-    if (m >= M) continue;
-    if (b0_valid) {
-      float* dst = C + (size_t)m * N + n0 + ccol;
-      if (use_atomic) atomicAdd(dst, acc0[r]);
-      else *dst = acc0[r];
+    for (int r = 0; r < 4; ++r) {
+      const int m = crow_base + r;
+      if (m >= M) continue;
+      if (b0_valid) part[(size_t)m * N + n0 + ccol] = acc0[r];
+      if (b1_valid) part[(size_t)m * N + n0 + 16 + ccol] = acc1[r];
     }
-    if (b1_valid) {
-      float* dst = C + (size_t)m * N + n0 + 16 + ccol;
-      if (use_atomic) atomicAdd(dst, acc1[r]);
-      else *dst = acc1[r];
+  } else {
+    unsigned short* dst = (unsigned short*)out;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = crow_base + r;
+      if (m >= M) continue;
+      if (b0_valid) dst[(size_t)m * N + n0 + ccol] = f2bf(acc0[r]);
+      if (b1_valid) dst[(size_t)m * N + n0 + 16 + ccol] = f2bf(acc1[r]);
     }
   }
 }
 
-void launch_skinny_gemm(void* C, const void* A, const void* W, int M, int N,
-                        int K, int ksplit, void* stream) {
+// fold ksplit partial slabs [ksplit, M, N] f32 -> bf16 [M, N]
+__global__ void reduce_cast_kernel(unsigned short* __restrict__ out,
+                                   const float* __restrict__ part,
+                                   long long mn, int ksplit) {
+  long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  long long stride = (long long)gridDim.x * blockDim.x * 4;
+  for (; i + 3 < mn; i += stride) {
+    f32x4v sum = *reinterpret_cast<const f32x4v*>(part + i);
+    for (int s = 1; s < ksplit; ++s) {
+      f32x4v v = *reinterpret_cast<const f32x4v*>(part + (size_t)s * mn + i);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) sum[j] += v[j];
+    }
+    unsigned short o[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = f2bf(sum[j]);
+    *reinterpret_cast<uint64_t*>(out + i) =
+        *reinterpret_cast<const uint64_t*>(o);
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    for (long long t = (mn & ~3LL); t < mn; ++t) {
+      float sum = part[t];
+      for (int s = 1; s < ksplit; ++s) sum += part[(size_t)s * mn + t];
+      out[t] = f2bf(sum);
+    }
+  }
+}
+
+void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
+                        const void* W, int M, int N, int K, int ksplit,
+                        void* stream) {
   if (ksplit < 1) ksplit = 1;
   int nblocks = (N + 127) / 128;
-  hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, ksplit), dim3(256), 0,
-                     (hipStream_t)stream, (float*)C,
-                     (const unsigned short*)A, (const unsigned short*)W, M, N,
-                     K, ksplit, ksplit > 1 ? 1 : 0);
+  if (ksplit == 1) {
+    hipLaunchKernelGGL(skinny_gemm_kernel<false>, dim3(nblocks), dim3(256),
+                       0, (hipStream_t)stream, out_bf16,
+                       (const unsigned short*)A, (const unsigned short*)W,
+                       M, N, K, 1);
+    return;
+  }
+  hipLaunchKernelGGL(skinny_gemm_kernel<true>, dim3(nblocks, ksplit),
+                     dim3(256), 0, (hipStream_t)stream, part_f32,
+                     (const unsigned short*)A, (const unsigned short*)W, M,
+                     N, K, ksplit);
+  long long mn = (long long)M * N;
+  long long blocks = (mn / 4 + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(reduce_cast_kernel, dim3((int)blocks), dim3(256), 0,
+                     (hipStream_t)stream, (unsigned short*)out_bf16,
+                     (const float*)part_f32, mn, ksplit);
 }
 
 // f32 -> bf16 flat cast (epilogue after k-split accumulate)
@@ -318,6 +409,64 @@ void launch_zero_f32(void* p, long long n, void* stream) {
   if (blocks > 2048) blocks = 2048;
   hipLaunchKernelGGL(zero_f32_kernel, dim3((int)blocks), dim3(256), 0,
                      (hipStream_t)stream, (float*)p, n);
+}
+
+// ---------------------------------------------------------------------
+// Fused decode RoPE + KV-cache append: one launch per layer replaces
+// {rope(q), rope(k), kv_append}.  qkv is the fused projection buffer
+// [B, row_stride] with q at offset 0, k at Hq*D, v at (Hq+Hkv)*D.
+// grid: (B, Hq + 2*Hkv); block: D/2 threads.
+//  y < Hq:          rope q head in place
+//  Hq <= y < +Hkv:  rope k head in place, then copy into Kc[pos]
+//  else:            copy v head into Vc[pos]
+// ---------------------------------------------------------------------
+__global__ void rope_kv_fused_kernel(
+    unsigned short* __restrict__ qkv, unsigned short* __restrict__ Kc,
+    unsigned short* __restrict__ Vc, const int* __restrict__ positions,
+    const float* __restrict__ cos_sin, int Hq, int Hkv, int Smax, int D,
+    long long row_stride) {
+  const int b = blockIdx.x;
+  const int y = blockIdx.y;
+  const int i = threadIdx.x;  // 0..D/2-1
+  const int half = D >> 1;
+  const int pos = positions[b];
+  unsigned short* row = qkv + (size_t)b * row_stride;
+  if (y < Hq + Hkv) {
+    // rope a q or k head
+    unsigned short* head = row + (size_t)y * D;
+    const float c = cos_sin[((size_t)pos * half + i) * 2];
+    const float s = cos_sin[((size_t)pos * half + i) * 2 + 1];
+    const float a = bf2f(head[i]);
+    const float bvf = bf2f(head[i + half]);
+    const unsigned short lo = f2bf(a * c - bvf * s);
+    const unsigned short hi = f2bf(bvf * c + a * s);
+    head[i] = lo;
+    head[i + half] = hi;
+    if (y >= Hq) {
+      const int kvh = y - Hq;
+      unsigned short* dst = Kc +
+          (((size_t)b * Hkv + kvh) * Smax + pos) * D;
+      dst[i] = lo;
+      dst[i + half] = hi;
+    }
+  } else {
+    const int kvh = y - Hq - Hkv;
+    const unsigned short* src = row + (size_t)(Hq + Hkv + kvh) * D;
+    unsigned short* dst = Vc + (((size_t)b * Hkv + kvh) * Smax + pos) * D;
+    dst[i] = src[i];
+    dst[i + half] = src[i + half];
+  }
+}
+
+void launch_rope_kv_fused(void* qkv, void* Kc, void* Vc,
+                          const void* positions, const void* cos_sin, int B,
+                          int Hq, int Hkv, int Smax, int D,
+                          long long row_stride, void* stream) {
+  hipLaunchKernelGGL(rope_kv_fused_kernel, dim3(B, Hq + 2 * Hkv),
+                     dim3(D / 2), 0, (hipStream_t)stream,
+                     (unsigned short*)qkv, (unsigned short*)Kc,
+                     (unsigned short*)Vc, (const int*)positions,
+                     (const float*)cos_sin, Hq, Hkv, Smax, D, row_stride);
 }
 
 // ---------------------------------------------------------------------

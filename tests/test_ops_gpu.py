@@ -208,3 +208,35 @@ class TestWindowAgg:
                                 current_period=6).cpu()
         assert torch.allclose(ref[:, :2], got[:, :2], atol=1e-2, rtol=1e-3)
         assert torch.allclose(ref[:, 2], got[:, 2], atol=1e-3, rtol=1e-3)
+
+
+@requires_gpu
+class TestFusedDecodeOps:
+    def test_swiglu_fused(self):
+        from mlrun_amd import ops
+
+        gu = _rand_bf16(16, 2048, seed=20)
+        ref = ops.swiglu_fused(gu)
+        got = ops.swiglu_fused(gu.cuda()).cpu()
+        assert torch.allclose(ref.float(), got.float(), atol=2e-2, rtol=2e-2)
+
+    def test_rope_kv_fused(self):
+        from mlrun_amd import ops
+
+        B, Hq, Hkv, D, Smax = 4, 8, 2, 128, 64
+        qkv = _rand_bf16(B, (Hq + 2 * Hkv) * D, seed=21)
+        kc = torch.zeros(B, Hkv, Smax, D, dtype=torch.bfloat16)
+        vc = torch.zeros_like(kc)
+        positions = torch.tensor([0, 3, 10, 63], dtype=torch.int32)
+        table = ops.build_rope_cos_sin(Smax, D)
+
+        qkv_ref, kc_ref, vc_ref = qkv.clone(), kc.clone(), vc.clone()
+        ops.rope_kv_fused(qkv_ref, kc_ref, vc_ref, positions, table, Hq)
+        qkv_gpu, kc_gpu, vc_gpu = qkv.cuda(), kc.cuda(), vc.cuda()
+        ops.rope_kv_fused(qkv_gpu, kc_gpu, vc_gpu, positions.cuda(),
+                          table.cuda(), Hq)
+        assert torch.allclose(qkv_ref.float(), qkv_gpu.cpu().float(),
+                              atol=2e-2, rtol=2e-2)
+        assert torch.allclose(kc_ref.float(), kc_gpu.cpu().float(),
+                              atol=2e-2, rtol=2e-2)
+        assert torch.equal(vc_ref, vc_gpu.cpu())
