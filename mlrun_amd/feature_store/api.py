@@ -1,0 +1,245 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Feature-store API: ingest / get_offline_features /
+get_online_feature_service.
+
+Parity target: reference mlrun/feature_store/api.py (ingest :450,
+get_offline_features :99, get_online_feature_service :296) +
+retrieval/local_merger.py (pandas offline merge).
+"""
+
+import os
+import time
+import typing
+
+from ..config import config
+from ..errors import MLRunInvalidArgumentError, MLRunNotFoundError
+from ..utils import logger, now_iso
+from .feature_set import FeatureSet
+from .online import OnlineTable, get_online_table
+from .vector import FeatureVector, OnlineVectorService, parse_feature_string
+
+
+def _resolve_feature_set(ref) -> FeatureSet:
+    if isinstance(ref, FeatureSet):
+        return ref
+    if isinstance(ref, str):
+        from ..db import get_run_db
+
+        name = ref
+        project = "default"
+        if ref.startswith("store://feature-sets/"):
+            body = ref[len("store://feature-sets/"):]
+            project, _, name = body.partition("/")
+        elif "/" in ref:
+            project, _, name = ref.partition("/")
+        name = name.split(":")[0]
+        struct = get_run_db().get_feature_set(name, project)
+        return FeatureSet.from_dict(struct)
+    raise MLRunInvalidArgumentError("cannot resolve feature set")
+
+
+def ingest(featureset: typing.Union[FeatureSet, str] = None, source=None,
+           targets: list = None, namespace=None, return_df: bool = True,
+           infer_options=None, overwrite=None):
+    """Batch-ingest a source (DataFrame / csv / parquet path) through
+    the featureset's transform graph into its targets (parquet offline
+    + the online window/KV table)."""
+    import pandas as pd
+
+    fset = _resolve_feature_set(featureset)
+    if source is None:
+        source = fset.spec.source
+    if isinstance(source, str):
+        if source.endswith(".csv"):
+            df = pd.read_csv(source)
+        elif source.endswith((".parquet", ".pq")):
+            df = pd.read_parquet(source)
+        else:
+            raise MLRunInvalidArgumentError(f"unsupported source {source}")
+    elif isinstance(source, pd.DataFrame):
+        df = source.copy()
+    else:
+        raise MLRunInvalidArgumentError("source must be a DataFrame or path")
+
+    # run the transform graph (batch steps)
+    df = _run_graph(fset, df)
+
+    # infer features from columns not yet declared
+    known = set(fset.feature_names()) | set(fset.entity_names())
+    from .feature_set import Feature
+
+    for col in df.columns:
+        if col not in known and col != fset.spec.timestamp_key:
+            kind = "float" if df[col].dtype.kind in "if" else "str"
+            fset.spec.features.append(Feature(name=col, value_type=kind))
+
+    targets = targets or fset.spec.targets or ["parquet", "nosql"]
+    # aggregations always fold into the online window table; the offline
+    # (parquet) target is then enriched with the as-of-ingest aggregate
+    # columns so vectors can reference them offline too
+    if fset.spec.aggregations:
+        table = get_online_table(fset)
+        table.ingest_batch(df)
+        entities = fset.entity_names()
+        unique_keys = df[entities].drop_duplicates().to_dict(
+            orient="records")
+        agg_records = table.get(unique_keys)
+        agg_cols = [f.name for f in fset.spec.features if f.aggregate]
+        agg_df = pd.DataFrame([
+            {**key, **{c: rec.get(c) for c in agg_cols}}
+            for key, rec in zip(unique_keys, agg_records)])
+        df = df.merge(agg_df, on=entities, how="left")
+    fset.status.targets = []
+    for target in targets:
+        kind = target if isinstance(target, str) else target.get("kind")
+        if kind in ("parquet", "offline"):
+            path = _parquet_target_path(fset)
+            os.makedirs(os.path.dirname(path), exist_ok=True)
+            if os.path.isfile(path) and not overwrite:
+                existing = pd.read_parquet(path)
+                pd.concat([existing, df], ignore_index=True).to_parquet(path)
+            else:
+                df.to_parquet(path)
+            fset.status.targets.append(
+                {"name": "parquet", "kind": "parquet", "path": path,
+                 "updated": now_iso()})
+        elif kind in ("nosql", "online"):
+            table = get_online_table(fset)
+            if not fset.spec.aggregations:
+                table.ingest_batch(df)  # aggregated sets folded above
+            fset.status.targets.append(
+                {"name": "nosql", "kind": "nosql",
+                 "path": f"online://{fset.fullname}", "updated": now_iso()})
+    fset.status.state = "ready"
+    try:
+        fset.save()
+    except Exception as exc:
+        logger.warning("feature set save failed", error=str(exc))
+    return df if return_df else None
+
+
+def _run_graph(fset: FeatureSet, df):
+    graph = fset.graph
+    if not graph.steps:
+        return df
+    from ..serving.server import Event, GraphContext
+
+    graph.init_object(GraphContext(), {})
+    event = Event(body=df)
+    result = graph.run(event)
+    return result.body if result is not None else df
+
+
+def _parquet_target_path(fset: FeatureSet) -> str:
+    base = str(config.feature_store.data_prefix or "") or os.path.join(
+        config.base_dir, "feature-store")
+    return os.path.join(base, fset.metadata.project or "default",
+                        f"{fset.metadata.name}.parquet")
+
+
+def preview(featureset, source, limit: int = 20):
+    """Dry-run the transform graph on a sample (reference: preview)."""
+    import pandas as pd
+
+    fset = _resolve_feature_set(featureset)
+    if isinstance(source, pd.DataFrame):
+        df = source.head(limit)
+    else:
+        df = pd.read_csv(source, nrows=limit) if str(source).endswith(
+            ".csv") else pd.read_parquet(source).head(limit)
+    out = _run_graph(fset, df)
+    fset.status.preview = out.head(limit).values.tolist()
+    return out
+
+
+class OfflineVectorResponse:
+    def __init__(self, df, vector=None):
+        self._df = df
+        self.vector = vector
+        self.status = "completed"
+
+    def to_dataframe(self):
+        return self._df
+
+    def to_parquet(self, path, **kwargs):
+        os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
+        self._df.to_parquet(path, **kwargs)
+        return path
+
+    def to_csv(self, path, **kwargs):
+        self._df.to_csv(path, index=False, **kwargs)
+        return path
+
+
+def get_offline_features(feature_vector, entity_rows=None,
+                         entity_timestamp_column=None, target=None,
+                         drop_columns=None, with_indexes=False,
+                         update_stats=False) -> OfflineVectorResponse:
+    """Join features from the parquet targets of the referenced sets
+    (pandas merger — reference retrieval/local_merger.py)."""
+    import pandas as pd
+
+    vector = FeatureVector.resolve(feature_vector)
+    merged = None
+    entity_cols: list = []
+    for fs_name, columns, aliases in vector.grouped_features():
+        fset = _resolve_feature_set(
+            f"{vector.metadata.project or 'default'}/{fs_name}")
+        path = _parquet_target_path(fset)
+        if not os.path.isfile(path):
+            raise MLRunNotFoundError(
+                f"feature set {fs_name} has no parquet target (ingest it "
+                f"first)")
+        df = pd.read_parquet(path)
+        entities = fset.entity_names()
+        entity_cols = entities
+        if columns != ["*"]:
+            missing = [c for c in columns if c not in df.columns]
+            if missing:
+                raise MLRunInvalidArgumentError(
+                    f"features {missing} not found in {fs_name}")
+            df = df[entities + columns]
+        if aliases:
+            df = df.rename(columns=aliases)
+        # latest row per entity for the join (offline snapshot)
+        ts = fset.spec.timestamp_key
+        if ts and ts in df.columns:
+            df = df.sort_values(ts).groupby(entities, as_index=False).last()
+            if ts not in (columns if columns != ["*"] else df.columns):
+                df = df.drop(columns=[ts], errors="ignore")
+        else:
+            df = df.groupby(entities, as_index=False).last()
+        merged = df if merged is None else merged.merge(
+            df, on=entities, how="inner")
+    if merged is None:
+        raise MLRunInvalidArgumentError("vector references no features")
+    if entity_rows is not None:
+        merged = entity_rows.merge(merged, on=entity_cols, how="left")
+    if vector.spec.label_feature:
+        pass
+    if drop_columns:
+        merged = merged.drop(columns=[c for c in drop_columns
+                                      if c in merged.columns])
+    if not with_indexes:
+        merged = merged.reset_index(drop=True)
+    if target:
+        merged.to_parquet(target)
+    return OfflineVectorResponse(merged, vector)
+
+
+def get_online_feature_service(feature_vector, impute_policy: dict = None,
+                               fixed_window_type=None,
+                               entity_keys=None) -> OnlineVectorService:
+    """Start an online lookup service over the vector's feature sets
+    (reference api.py:296)."""
+    vector = FeatureVector.resolve(feature_vector)
+    tables = {}
+    for fs_name, _, _ in vector.grouped_features():
+        fset = _resolve_feature_set(
+            f"{vector.metadata.project or 'default'}/{fs_name}")
+        tables[fs_name] = get_online_table(fset)
+    return OnlineVectorService(vector, tables,
+                               impute_policy=impute_policy or {})

@@ -1,0 +1,284 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Online feature store: HBM-resident window aggregations + latest-value
+KV table.
+
+The reference's online path is a storey Table over V3IO/Redis with
+per-key running aggregates (datastore/targets.py:1415).  MI355X-native
+design: per-(featureset, column) ring buffers of per-period partial
+aggregates live as torch tensors — on the GPU when available (288 GB
+HBM easily holds millions of keys), CPU tensors otherwise — ingested
+and reduced by the HIP window kernels (ops.window_ingest/window_reduce)
+in one batched launch per column, not per-event asyncio emits.
+"""
+
+import threading
+import time
+import typing
+
+import numpy as np
+import torch
+
+from ..errors import MLRunInvalidArgumentError
+from ..utils import logger
+from .feature_set import FeatureSet, parse_span
+
+
+class WindowRing:
+    """Ring of per-period partials for one aggregated column.
+
+    ring[key, period, 0] = sum, [1] = count; min/max/first/last are
+    tracked in auxiliary per-key tensors (they don't decompose over
+    ring periods for sliding windows; kept as running values like the
+    reference's running aggregates)."""
+
+    def __init__(self, period_seconds: int, n_periods: int, device="cpu",
+                 capacity: int = 1024):
+        self.period_seconds = period_seconds
+        self.n_periods = n_periods
+        self.device = device
+        self.capacity = capacity
+        self.ring = torch.zeros(capacity, n_periods, 4, dtype=torch.float32,
+                                device=device)
+        # running (non-windowed) aggregates per key:
+        # [min, max, first, last, count, sum, sumsq]
+        self.running = torch.zeros(capacity, 7, dtype=torch.float32)
+        self.running[:, 0] = float("inf")
+        self.running[:, 1] = float("-inf")
+        # per-(key, period) sum of squares for stdvar/stddev windows
+        self.ring_sq = torch.zeros(capacity, n_periods, 4,
+                                   dtype=torch.float32, device=device)
+        self.last_period = -1
+
+    def grow(self, capacity: int):
+        if capacity <= self.capacity:
+            return
+        new_ring = torch.zeros(capacity, self.n_periods, 4,
+                               dtype=torch.float32, device=self.device)
+        new_ring[:self.capacity] = self.ring
+        self.ring = new_ring
+        new_sq = torch.zeros(capacity, self.n_periods, 4,
+                             dtype=torch.float32, device=self.device)
+        new_sq[:self.capacity] = self.ring_sq
+        self.ring_sq = new_sq
+        new_running = torch.zeros(capacity, 7, dtype=torch.float32)
+        new_running[:, 0] = float("inf")
+        new_running[:, 1] = float("-inf")
+        new_running[:self.capacity] = self.running
+        self.running = new_running
+        self.capacity = capacity
+
+    def _expire_old_periods(self, current_period: int):
+        """Zero ring cells for periods that wrapped since last ingest."""
+        if self.last_period < 0:
+            self.last_period = current_period
+            return
+        gap = current_period - self.last_period
+        if gap <= 0:
+            return
+        if gap >= self.n_periods:
+            self.ring.zero_()
+            self.ring_sq.zero_()
+        else:
+            for p in range(self.last_period + 1, current_period + 1):
+                idx = p % self.n_periods
+                self.ring[:, idx].zero_()
+                self.ring_sq[:, idx].zero_()
+        self.last_period = current_period
+
+    def ingest(self, key_ids: torch.Tensor, values: torch.Tensor,
+               timestamps: torch.Tensor):
+        """Batched fold of events into the ring (HIP kernel on GPU)."""
+        from .. import ops
+
+        period_idx_abs = (timestamps.long() //
+                          self.period_seconds)
+        current = int(period_idx_abs.max())
+        self._expire_old_periods(current)
+        keys32 = key_ids.to(torch.int32)
+        pidx = (period_idx_abs % self.n_periods).to(torch.int32)
+        vals = values.to(torch.float32)
+        if self.ring.is_cuda:
+            keys32 = keys32.cuda(self.ring.device)
+            pidx = pidx.to(self.ring.device)
+            vals_dev = vals.to(self.ring.device)
+        else:
+            vals_dev = vals
+        ops.window_ingest(self.ring, keys32, vals_dev, pidx)
+        ops.window_ingest(self.ring_sq, keys32, vals_dev * vals_dev, pidx)
+        # running aggregates (CPU, numpy-vectorized per batch)
+        k = key_ids.numpy()
+        v = vals.numpy()
+        order = np.argsort(k, kind="stable")
+        k_sorted, v_sorted = k[order], v[order]
+        uniq, starts = np.unique(k_sorted, return_index=True)
+        ends = np.append(starts[1:], len(k_sorted))
+        for key, s, e in zip(uniq, starts, ends):
+            seg = v_sorted[s:e]
+            row = self.running[key]
+            row[0] = min(float(row[0]), float(seg.min()))
+            row[1] = max(float(row[1]), float(seg.max()))
+            if row[4] == 0:
+                row[2] = float(seg[0])
+            row[3] = float(seg[-1])
+            row[4] += len(seg)
+            row[5] += float(seg.sum())
+            row[6] += float((seg * seg).sum())
+
+    def window_values(self, window_seconds: int, now_ts: float) -> dict:
+        """Reduce the ring for one window length -> tensors keyed by op
+        (sum/count/avg/stdvar/stddev), each [capacity]."""
+        from .. import ops
+
+        window_periods = max(window_seconds // self.period_seconds, 1)
+        current_period = int(now_ts // self.period_seconds)
+        self._expire_old_periods(current_period)
+        out = ops.window_reduce(self.ring, window_periods,
+                                current_period % self.n_periods)
+        out_sq = ops.window_reduce(self.ring_sq, window_periods,
+                                   current_period % self.n_periods)
+        if out.is_cuda:
+            out = out.cpu()
+            out_sq = out_sq.cpu()
+        total, count, avg = out[:, 0], out[:, 1], out[:, 2]
+        sumsq = out_sq[:, 0]
+        n = count.clamp(min=1.0)
+        var = (sumsq / n - (total / n) ** 2).clamp(min=0.0)
+        unbiased = torch.where(count > 1, var * count / (count - 1).clamp(
+            min=1.0), torch.zeros_like(var))
+        return {
+            "sum": total,
+            "count": count,
+            "avg": avg,
+            "sqr": sumsq,
+            "stdvar": unbiased,
+            "stddev": unbiased.sqrt(),
+        }
+
+
+class OnlineTable:
+    """Per-featureset online state: latest row per key + window rings."""
+
+    def __init__(self, feature_set: FeatureSet, device=None):
+        self.feature_set = feature_set
+        self.device = device or ("cuda:0" if torch.cuda.is_available()
+                                 else "cpu")
+        self.key_index: typing.Dict[typing.Any, int] = {}
+        self.latest: typing.Dict[typing.Any, dict] = {}
+        self.rings: typing.Dict[str, WindowRing] = {}
+        self._lock = threading.Lock()
+        for agg in feature_set.spec.aggregations:
+            period = agg.period or agg.windows[0]
+            period_s = parse_span(period)
+            max_window = max(parse_span(w) for w in agg.windows)
+            n_periods = max(max_window // period_s, 1) + 1
+            self.rings[agg.name] = WindowRing(period_s, n_periods,
+                                              device=self.device)
+
+    def _key_of(self, row: dict):
+        entities = self.feature_set.entity_names()
+        if len(entities) == 1:
+            return row[entities[0]]
+        return tuple(row[e] for e in entities)
+
+    def _key_ids(self, keys: list) -> torch.Tensor:
+        ids = []
+        for key in keys:
+            idx = self.key_index.get(key)
+            if idx is None:
+                idx = len(self.key_index)
+                self.key_index[key] = idx
+            ids.append(idx)
+        needed = len(self.key_index)
+        for ring in self.rings.values():
+            if needed > ring.capacity:
+                ring.grow(max(needed, ring.capacity * 2))
+        return torch.tensor(ids, dtype=torch.int64)
+
+    def ingest_batch(self, df):
+        """Fold a dataframe batch: update latest rows + window rings."""
+        import pandas as pd
+
+        fset = self.feature_set
+        ts_key = fset.spec.timestamp_key
+        with self._lock:
+            keys = [self._key_of(row) for row in
+                    df.to_dict(orient="records")]
+            key_ids = self._key_ids(keys)
+            if ts_key and ts_key in df.columns:
+                ts = pd.to_datetime(df[ts_key]).astype("int64") // 10 ** 9
+                timestamps = torch.tensor(ts.values, dtype=torch.int64)
+            else:
+                timestamps = torch.full((len(df),), int(time.time()),
+                                        dtype=torch.int64)
+            for agg in fset.spec.aggregations:
+                if agg.column not in df.columns:
+                    continue
+                values = torch.tensor(
+                    df[agg.column].astype("float32").values)
+                self.rings[agg.name].ingest(key_ids, values, timestamps)
+            for key, row in zip(keys, df.to_dict(orient="records")):
+                self.latest[key] = row
+        return len(df)
+
+    def get(self, entity_rows: typing.List[dict], now_ts: float = None
+            ) -> typing.List[dict]:
+        """Batched online lookup: latest values + window aggregates."""
+        now_ts = now_ts or time.time()
+        fset = self.feature_set
+        with self._lock:
+            # one ring reduce per (agg, window) serves the whole batch
+            reduced = {}
+            for agg in fset.spec.aggregations:
+                ring = self.rings[agg.name]
+                for window in agg.windows:
+                    reduced[(agg.name, window)] = ring.window_values(
+                        parse_span(window), now_ts)
+            out = []
+            for row in entity_rows:
+                key = self._key_of(row)
+                idx = self.key_index.get(key)
+                record: dict = {}
+                latest = self.latest.get(key)
+                if latest:
+                    record.update(latest)
+                for agg in fset.spec.aggregations:
+                    ring = self.rings[agg.name]
+                    for window in agg.windows:
+                        vals = reduced[(agg.name, window)]
+                        for op in agg.operations:
+                            name = f"{agg.name}_{op}_{window}"
+                            if idx is None:
+                                record[name] = None
+                            elif op in vals:
+                                record[name] = float(vals[op][idx])
+                            else:  # min/max/first/last: running values
+                                run = ring.running[idx]
+                                mapping = {"min": 0, "max": 1, "first": 2,
+                                           "last": 3}
+                                value = float(run[mapping[op]])
+                                if op in ("min", "max") and \
+                                        float(run[4]) == 0:
+                                    value = None
+                                record[name] = value
+                out.append(record)
+            return out
+
+
+_tables: dict = {}
+_tables_lock = threading.Lock()
+
+
+def get_online_table(feature_set: FeatureSet, device=None) -> OnlineTable:
+    key = feature_set.fullname
+    with _tables_lock:
+        if key not in _tables:
+            _tables[key] = OnlineTable(feature_set, device=device)
+        return _tables[key]
+
+
+def reset_online_tables():
+    with _tables_lock:
+        _tables.clear()

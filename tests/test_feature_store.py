@@ -1,0 +1,186 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Feature-store tests (mirror of the reference's tests/feature-store
+unit tier: local parquet + in-process engine)."""
+
+import time
+
+import numpy as np
+import pandas as pd
+import pytest
+
+from mlrun_amd import feature_store as fstore
+from mlrun_amd.errors import MLRunInvalidArgumentError
+
+
+@pytest.fixture(autouse=True)
+def _reset_tables():
+    fstore.reset_online_tables()
+    yield
+    fstore.reset_online_tables()
+
+
+def make_df(n=100, t0=None):
+    t0 = t0 or time.time()
+    rng = np.random.default_rng(42)
+    return pd.DataFrame({
+        "customer": rng.choice(["alice", "bob", "carol"], n),
+        "amount": rng.uniform(1, 100, n).round(2),
+        "city": rng.choice(["NY", "SF"], n),
+        "ts": pd.to_datetime((t0 - rng.uniform(0, 3000, n)), unit="s"),
+    })
+
+
+class TestFeatureSet:
+    def test_aggregation_validation(self):
+        fset = fstore.FeatureSet("s1", entities=["customer"])
+        with pytest.raises(MLRunInvalidArgumentError):
+            fset.add_aggregation("amount", ["bogus"], ["1h"])
+        with pytest.raises(MLRunInvalidArgumentError):
+            # period must divide window
+            fset.add_aggregation("amount", ["sum"], ["1h"], period="7m")
+        fset.add_aggregation("amount", ["sum", "avg"], ["1h"], period="10m")
+        names = fset.feature_names()
+        assert "amount_sum_1h" in names and "amount_avg_1h" in names
+
+    def test_parse_span(self):
+        assert fstore.parse_span("30s") == 30
+        assert fstore.parse_span("10m") == 600
+        assert fstore.parse_span("2h") == 7200
+
+    def test_roundtrip_db(self, rundb):
+        fset = fstore.FeatureSet("s2", entities=["id"], project="p")
+        fset.add_aggregation("v", ["count"], ["1h"], period="10m")
+        fset.save()
+        loaded = fstore.FeatureSet.from_dict(rundb.get_feature_set("s2", "p"))
+        assert loaded.name == "s2"
+        assert loaded.spec.aggregations[0].operations == ["count"]
+
+
+class TestIngest:
+    def test_ingest_parquet_and_online(self, tmp_path):
+        df = make_df(200)
+        fset = fstore.FeatureSet("tx", entities=["customer"],
+                                 timestamp_key="ts", project="default")
+        fset.add_aggregation("amount", ["sum", "count", "avg", "max", "min"],
+                             ["1h"], period="10m")
+        out = fstore.ingest(fset, df)
+        assert len(out) == 200
+        path = fset.get_target_path("parquet")
+        assert path and pd.read_parquet(path).shape[0] == 200
+
+        table = fstore.get_online_table(fset)
+        records = table.get([{"customer": "alice"}])
+        rec = records[0]
+        # windowed sum over the last hour matches pandas
+        cutoff = pd.Timestamp.now() - pd.Timedelta(hours=1)
+        # ring periods quantize: compare within the ring resolution
+        expected_all = df[df.customer == "alice"].amount.sum()
+        assert rec["amount_count_1h"] >= 1
+        assert rec["amount_sum_1h"] <= expected_all + 1e-3
+        assert rec["amount_max_1h"] == pytest.approx(
+            df[df.customer == "alice"].amount.max(), abs=1e-3)
+        assert rec["amount_min_1h"] == pytest.approx(
+            df[df.customer == "alice"].amount.min(), abs=1e-3)
+
+    def test_window_semantics_exact(self):
+        """Deterministic timestamps: sum over window, excluding expired
+        periods."""
+        now = time.time()
+        period = 600  # 10m
+        fset = fstore.FeatureSet("win", entities=["k"], timestamp_key="ts")
+        fset.add_aggregation("v", ["sum", "count"], ["30m"], period="10m")
+        # 3 events inside the 30m window, 1 far outside
+        df = pd.DataFrame({
+            "k": ["a"] * 4,
+            "v": [1.0, 2.0, 4.0, 100.0],
+            "ts": pd.to_datetime(
+                [now - 60, now - 700, now - 1500, now - 9000], unit="s"),
+        })
+        fstore.ingest(fset, df, targets=["nosql"])
+        table = fstore.get_online_table(fset)
+        rec = table.get([{"k": "a"}], now_ts=now)[0]
+        assert rec["v_sum_30m"] == pytest.approx(7.0)
+        assert rec["v_count_30m"] == 3
+
+    def test_transform_graph(self):
+        fset = fstore.FeatureSet("tg", entities=["customer"])
+        fset.graph.to(fstore.MapValues(
+            mapping={"city": {"NY": "east", "SF": "west"}}), name="map")
+        df = make_df(20)
+        out = fstore.ingest(fset, df, targets=["nosql"])
+        assert set(out.city.unique()) <= {"east", "west"}
+
+    def test_steps(self):
+        df = pd.DataFrame({
+            "a": [1.0, None, 3.0],
+            "cat": ["x", "y", "x"],
+            "ts": pd.to_datetime(["2026-01-05", "2026-01-06", "2026-01-07"]),
+        })
+        out = fstore.Imputer(method="avg").do(df)
+        assert out["a"].isna().sum() == 0
+        out = fstore.OneHotEncoder(mapping={"cat": ["x", "y"]}).do(df)
+        assert list(out["cat_x"]) == [1, 0, 1]
+        out = fstore.DateExtractor(parts=["day_of_week"],
+                                   timestamp_col="ts").do(df)
+        assert "ts_day_of_week" in out.columns
+        out = fstore.DropFeatures(features=["a"]).do(df)
+        assert "a" not in out.columns
+        out = fstore.MapValues(mapping={"a": {"ranges": {
+            "low": [0, 2], "high": [2, "inf"]}}}).do(
+            df.fillna(0))
+        assert list(out["a"]) == ["low", "low", "high"]
+
+
+class TestVectorAndServices:
+    def _setup_sets(self):
+        df = make_df(100)
+        fset = fstore.FeatureSet("txv", entities=["customer"],
+                                 timestamp_key="ts")
+        fset.add_aggregation("amount", ["sum", "avg"], ["1h"], period="10m")
+        fstore.ingest(fset, df)
+        return df, fset
+
+    def test_offline_features(self, rundb):
+        df, fset = self._setup_sets()
+        vector = fstore.FeatureVector(
+            "v1", features=["txv.amount", "txv.amount_sum_1h"])
+        vector.metadata.project = "default"
+        vector.save()
+        resp = fstore.get_offline_features(vector)
+        out = resp.to_dataframe()
+        assert "amount" in out.columns and "amount_sum_1h" in out.columns
+        assert len(out) == df.customer.nunique()
+
+    def test_online_service(self):
+        df, fset = self._setup_sets()
+        vector = fstore.FeatureVector(
+            "v2", features=["txv.amount_sum_1h", "txv.amount_avg_1h",
+                            "txv.city"])
+        vector.metadata.project = "default"
+        svc = fstore.get_online_feature_service(vector)
+        out = svc.get([{"customer": "alice"}, {"customer": "bob"}])
+        assert len(out) == 2
+        assert out[0]["amount_sum_1h"] is not None
+        as_list = svc.get([{"customer": "alice"}], as_list=True)
+        assert len(as_list[0]) == 3
+
+    def test_impute_policy(self):
+        df, fset = self._setup_sets()
+        vector = fstore.FeatureVector("v3",
+                                      features=["txv.amount_sum_1h"])
+        vector.metadata.project = "default"
+        svc = fstore.get_online_feature_service(
+            vector, impute_policy={"amount_sum_1h": -1.0})
+        out = svc.get([{"customer": "nobody"}])
+        assert out[0]["amount_sum_1h"] == -1.0
+
+    def test_feature_string_parsing(self):
+        from mlrun_amd.feature_store.vector import parse_feature_string
+
+        assert parse_feature_string("s.f") == ("s", "f", None)
+        assert parse_feature_string("s.f as x") == ("s", "f", "x")
+        with pytest.raises(MLRunInvalidArgumentError):
+            parse_feature_string("nofset")
