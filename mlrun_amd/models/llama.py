@@ -160,7 +160,8 @@ class LlamaDecodeEngine:
         # split-K scratch: ksplit * B * N f32 for the largest case
         gemm_shapes = [(qkv_n, h), (h, w.hq * d), (2 * w.inter, h),
                        (h, w.inter), (cfg.vocab_size, h)]
-        scratch = max(ops.pick_ksplit(B, n, k) * n for n, k in gemm_shapes)
+        scratch = max(ops.pick_gemm_plan(B, n, k)[0] * n
+                      for n, k in gemm_shapes)
         self.buf_tokens = torch.zeros(B, dtype=torch.int64,
                                       device=self.device)
         self.buf_hidden = torch.zeros(B, h, **bf16)
@@ -184,9 +185,10 @@ class LlamaDecodeEngine:
         """skinny GEMM into a preallocated bf16 out + f32 scratch."""
         key = (w_.shape[0], w_.shape[1])
         if key not in self._ksplits:
-            self._ksplits[key] = ops.pick_ksplit(a.shape[0], *key)
+            self._ksplits[key] = ops.pick_gemm_plan(a.shape[0], *key)
+        ksplit, variant = self._ksplits[key]
         return ops.skinny_gemm(a, w_, out=out, c_f32=self.buf_c32,
-                               ksplit=self._ksplits[key])
+                               ksplit=ksplit, variant=variant)
 
     def _maybe_allreduce(self, t):
         if self.tp_size > 1:

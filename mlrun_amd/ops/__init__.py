@@ -156,10 +156,21 @@ def pick_ksplit(M: int, N: int, K: int, target_blocks: int = 2048) -> int:
 _EMPTY_F32 = None
 
 
+def pick_gemm_plan(M: int, N: int, K: int) -> tuple:
+    """(ksplit, variant): variant 1 = wave-split-K (4 waves share one
+    32-wide n-tile) for small-N projections, else the 128-wide tiler."""
+    nblocks_ws = (N + 31) // 32
+    if nblocks_ws * 4 <= 2048 and K >= 2048:
+        # small-N: wave-split for 4x waves/SIMD at equal slab traffic
+        ksplit = max(1, min(1024 // nblocks_ws, K // 512, 16))
+        return ksplit, 1
+    return pick_ksplit(M, N, K), 0
+
+
 def skinny_gemm(a: torch.Tensor, w: torch.Tensor,
                 out: torch.Tensor = None,
                 c_f32: torch.Tensor = None,
-                ksplit: int = None) -> torch.Tensor:
+                ksplit: int = None, variant: int = None) -> torch.Tensor:
     """C[M,N] = A[M,K] @ W[N,K]^T for decode batches (M <= 16), bf16.
 
     On GPU: MFMA kernel; split-K writes per-chunk f32 slabs into c_f32
@@ -171,8 +182,12 @@ def skinny_gemm(a: torch.Tensor, w: torch.Tensor,
     N = w.shape[0]
     if a.is_cuda:
         ops = _require_hip()
-        if ksplit is None:
+        if ksplit is None and variant is None:
+            ksplit, variant = pick_gemm_plan(M, N, K)
+        elif ksplit is None:
             ksplit = pick_ksplit(M, N, K)
+        elif variant is None:
+            variant = 0
         if out is None:
             out = torch.empty(M, N, dtype=torch.bfloat16, device=a.device)
         if ksplit > 1 and c_f32 is None:
@@ -183,7 +198,7 @@ def skinny_gemm(a: torch.Tensor, w: torch.Tensor,
                 _EMPTY_F32 = torch.empty(1, dtype=torch.float32,
                                          device=a.device)
             c_f32 = _EMPTY_F32
-        ops.skinny_gemm(out, c_f32, a, w, ksplit)
+        ops.skinny_gemm(out, c_f32, a, w, ksplit, variant)
         return out
     result = (a.float() @ w.float().t()).to(a.dtype)
     if out is not None:

@@ -91,7 +91,8 @@ void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up) {
 // out_bf16 [M,N] <- A [M,K] @ W [N,K]^T.  part_f32 is the split-K
 // scratch ([ksplit*M*N] f32, unused when ksplit==1 — pass any tensor).
 void skinny_gemm(torch::Tensor out_bf16, torch::Tensor part_f32,
-                 torch::Tensor a, torch::Tensor w, int64_t ksplit) {
+                 torch::Tensor a, torch::Tensor w, int64_t ksplit,
+                 int64_t variant) {
   CHECK_DEV(out_bf16); CHECK_CONTIG(out_bf16); CHECK_BF16(out_bf16);
   CHECK_DEV(a); CHECK_CONTIG(a); CHECK_BF16(a);
   CHECK_DEV(w); CHECK_CONTIG(w); CHECK_BF16(w);
@@ -108,7 +109,7 @@ void skinny_gemm(torch::Tensor out_bf16, torch::Tensor part_f32,
   launch_skinny_gemm(out_bf16.data_ptr(),
                      ksplit > 1 ? part_f32.data_ptr() : nullptr,
                      a.data_ptr(), w.data_ptr(), M, N, K, (int)ksplit,
-                     stream());
+                     (int)variant, stream());
 }
 
 void rope_kv_fused(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,

@@ -24,7 +24,7 @@ void launch_silu_mul(void* out, const void* gate, const void* up,
 
 void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                         const void* W, int M, int N, int K, int ksplit,
-                        void* stream);
+                        int variant, void* stream);
 
 void launch_rope_kv_fused(void* qkv, void* Kc, void* Vc,
                           const void* positions, const void* cos_sin, int B,

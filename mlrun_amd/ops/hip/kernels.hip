@@ -314,6 +314,91 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   }
 }
 
+// Wave-split-K variant: the 4 waves of a workgroup cooperate on ONE
+// 32-wide n-tile, each covering K/4 of the k-chunk, combining through
+// LDS.  4x the workgroup count of skinny_gemm_kernel at the same
+// split-K slab traffic -> 4x the waves/SIMD for latency hiding
+// (profile: N=4096 projections were latency-bound at 2 waves/SIMD).
+template <bool SPLIT>
+__global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
+    void* __restrict__ out, const unsigned short* __restrict__ A,
+    const unsigned short* __restrict__ W, int M, int N, int K, int ksplit) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int n0 = blockIdx.x * 32;  // one 32-wide tile per block
+  if (n0 >= N) return;
+  int cbegin = 0, cend = K;
+  if (SPLIT) {
+    const int kchunk = (K / ksplit + 31) & ~31;
+    cbegin = blockIdx.y * kchunk;
+    cend = cbegin + kchunk;
+    if (cend > K) cend = K;
+  }
+  // each wave covers a quarter of [cbegin, cend)
+  const int clen = cend - cbegin;
+  const int per_wave = ((clen / 4) + 31) & ~31;
+  int kbegin = cbegin + wave * per_wave;
+  int kend = kbegin + per_wave;
+  if (kend > cend) kend = cend;
+
+  const int arow = lane & 15;
+  const int kb = (lane >> 4) * 8;
+  const bool a_valid = arow < M;
+  const int brow0 = n0 + (lane & 15);
+  const int brow1 = brow0 + 16;
+  const bool b0_valid = brow0 < N;
+  const bool b1_valid = brow1 < N;
+
+  f32x4v acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4v acc1 = {0.f, 0.f, 0.f, 0.f};
+  const short8v zero8 = {0, 0, 0, 0, 0, 0, 0, 0};
+  const unsigned short* aptr = A + (size_t)arow * K + kb;
+  const unsigned short* bptr0 = W + (size_t)brow0 * K + kb;
+  const unsigned short* bptr1 = W + (size_t)brow1 * K + kb;
+
+#pragma unroll 8
+  for (int k = kbegin; k < kend; k += 32) {
+    short8v af = a_valid
+        ? *reinterpret_cast<const short8v*>(aptr + k) : zero8;
+    short8v bf0 = b0_valid
+        ? *reinterpret_cast<const short8v*>(bptr0 + k) : zero8;
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
+    if (b1_valid) {
+      short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+    }
+  }
+
+  // combine the 4 waves' partials through LDS
+  __shared__ float comb[4][16][32];  // [wave][m][n] 8 KiB
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    comb[wave][crow_base + r][ccol] = acc0[r];
+    comb[wave][crow_base + r][ccol + 16] = acc1[r];
+  }
+  __syncthreads();
+  if (wave == 0) {
+    // 64 lanes fold 16x32 cells: lane covers 8 cells
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      const int cell = lane * 8 + c;
+      const int m = cell >> 5;
+      const int n = cell & 31;
+      if (m >= M || n0 + n >= N) continue;
+      float sum = comb[0][m][n] + comb[1][m][n] + comb[2][m][n] +
+                  comb[3][m][n];
+      if (SPLIT) {
+        float* part = (float*)out + (size_t)blockIdx.y * M * N;
+        part[(size_t)m * N + n0 + n] = sum;
+      } else {
+        ((unsigned short*)out)[(size_t)m * N + n0 + n] = f2bf(sum);
+      }
+    }
+  }
+}
+
 // fold ksplit partial slabs [ksplit, M, N] f32 -> bf16 [M, N]
 __global__ void reduce_cast_kernel(unsigned short* __restrict__ out,
                                    const float* __restrict__ part,
@@ -344,27 +429,40 @@ __global__ void reduce_cast_kernel(unsigned short* __restrict__ out,
 
 void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                         const void* W, int M, int N, int K, int ksplit,
-                        void* stream) {
+                        int variant, void* stream) {
   if (ksplit < 1) ksplit = 1;
-  int nblocks = (N + 127) / 128;
+  const int nblocks = variant == 1 ? (N + 31) / 32 : (N + 127) / 128;
   if (ksplit == 1) {
-    hipLaunchKernelGGL(skinny_gemm_kernel<false>, dim3(nblocks), dim3(256),
-                       0, (hipStream_t)stream, out_bf16,
-                       (const unsigned short*)A, (const unsigned short*)W,
-                       M, N, K, 1);
-    return;
+    if (variant == 1)
+      hipLaunchKernelGGL(skinny_gemm_ws_kernel<false>, dim3(nblocks),
+                         dim3(256), 0, (hipStream_t)stream, out_bf16,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, 1);
+    else
+      hipLaunchKernelGGL(skinny_gemm_kernel<false>, dim3(nblocks),
+                         dim3(256), 0, (hipStream_t)stream, out_bf16,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, 1);
+  } else {
+    if (variant == 1)
+      hipLaunchKernelGGL(skinny_gemm_ws_kernel<true>,
+                         dim3(nblocks, ksplit), dim3(256), 0,
+                         (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+    else
+      hipLaunchKernelGGL(skinny_gemm_kernel<true>, dim3(nblocks, ksplit),
+                         dim3(256), 0, (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+    long long mn = (long long)M * N;
+    long long blocks = (mn / 4 + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(reduce_cast_kernel, dim3((int)blocks), dim3(256), 0,
+                       (hipStream_t)stream, (unsigned short*)out_bf16,
+                       (const float*)part_f32, mn, ksplit);
   }
-  hipLaunchKernelGGL(skinny_gemm_kernel<true>, dim3(nblocks, ksplit),
-                     dim3(256), 0, (hipStream_t)stream, part_f32,
-                     (const unsigned short*)A, (const unsigned short*)W, M,
-                     N, K, ksplit);
-  long long mn = (long long)M * N;
-  long long blocks = (mn / 4 + 255) / 256;
-  if (blocks > 2048) blocks = 2048;
-  if (blocks < 1) blocks = 1;
-  hipLaunchKernelGGL(reduce_cast_kernel, dim3((int)blocks), dim3(256), 0,
-                     (hipStream_t)stream, (unsigned short*)out_bf16,
-                     (const float*)part_f32, mn, ksplit);
 }
 
 // f32 -> bf16 flat cast (epilogue after k-split accumulate)
