@@ -156,13 +156,27 @@ def pick_ksplit(M: int, N: int, K: int, target_blocks: int = 2048) -> int:
 _EMPTY_F32 = None
 
 
+# measured on MI355X (scripts/bench_gemm.py sweep, profiles/): best
+# (ksplit, variant) per llama projection shape; variant 1/2 = wave-split
+_GEMM_PLAN_TABLE = {
+    (6144, 4096): (4, 1),     # qkv      3.06 TB/s
+    (4096, 4096): (4, 1),     # wo       2.81 TB/s
+    (28672, 4096): (1, 1),    # gate|up  4.51 TB/s
+    (4096, 14336): (8, 1),    # down     3.90 TB/s
+    (128256, 4096): (1, 0),   # lm_head  4.16 TB/s
+}
+
+
 def pick_gemm_plan(M: int, N: int, K: int) -> tuple:
     """(ksplit, variant): variant 1 = wave-split-K (4 waves share one
-    32-wide n-tile) for small-N projections, else the 128-wide tiler."""
+    32-wide n-tile) for small-N projections, else the 128-wide tiler.
+    Known llama shapes use the measured sweep table."""
+    if (N, K) in _GEMM_PLAN_TABLE:
+        return _GEMM_PLAN_TABLE[(N, K)]
     nblocks_ws = (N + 31) // 32
-    if nblocks_ws * 4 <= 2048 and K >= 2048:
+    if nblocks_ws <= 512 and K >= 2048:
         # small-N: wave-split for 4x waves/SIMD at equal slab traffic
-        ksplit = max(1, min(1024 // nblocks_ws, K // 512, 16))
+        ksplit = max(1, min(1024 // nblocks_ws, K // 1024, 8))
         return ksplit, 1
     return pick_ksplit(M, N, K), 0
 

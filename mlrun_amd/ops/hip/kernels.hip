@@ -319,7 +319,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 // LDS.  4x the workgroup count of skinny_gemm_kernel at the same
 // split-K slab traffic -> 4x the waves/SIMD for latency hiding
 // (profile: N=4096 projections were latency-bound at 2 waves/SIMD).
-template <bool SPLIT>
+template <bool SPLIT, int UNROLL>
 __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
     void* __restrict__ out, const unsigned short* __restrict__ A,
     const unsigned short* __restrict__ W, int M, int N, int K, int ksplit) {
@@ -356,7 +356,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
   const unsigned short* bptr0 = W + (size_t)brow0 * K + kb;
   const unsigned short* bptr1 = W + (size_t)brow1 * K + kb;
 
-#pragma unroll 8
+#pragma unroll UNROLL
   for (int k = kbegin; k < kend; k += 32) {
     short8v af = a_valid
         ? *reinterpret_cast<const short8v*>(aptr + k) : zero8;
@@ -431,10 +431,15 @@ void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                         const void* W, int M, int N, int K, int ksplit,
                         int variant, void* stream) {
   if (ksplit < 1) ksplit = 1;
-  const int nblocks = variant == 1 ? (N + 31) / 32 : (N + 127) / 128;
+  const int nblocks = variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
   if (ksplit == 1) {
-    if (variant == 1)
-      hipLaunchKernelGGL(skinny_gemm_ws_kernel<false>, dim3(nblocks),
+    if (variant == 2)
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 16>), dim3(nblocks),
+                         dim3(256), 0, (hipStream_t)stream, out_bf16,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, 1);
+    else if (variant == 1)
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 8>), dim3(nblocks),
                          dim3(256), 0, (hipStream_t)stream, out_bf16,
                          (const unsigned short*)A, (const unsigned short*)W,
                          M, N, K, 1);
@@ -444,8 +449,14 @@ void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                          (const unsigned short*)A, (const unsigned short*)W,
                          M, N, K, 1);
   } else {
-    if (variant == 1)
-      hipLaunchKernelGGL(skinny_gemm_ws_kernel<true>,
+    if (variant == 2)
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 16>),
+                         dim3(nblocks, ksplit), dim3(256), 0,
+                         (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+    else if (variant == 1)
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 8>),
                          dim3(nblocks, ksplit), dim3(256), 0,
                          (hipStream_t)stream, part_f32,
                          (const unsigned short*)A, (const unsigned short*)W,

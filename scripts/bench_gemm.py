@@ -53,7 +53,7 @@ def main():
         ref = (a.float() @ w.float().t())
         bytes_w = n * k * 2
         best = None
-        for variant in (0, 1):
+        for variant in (0, 1, 2):
             for ksplit in (1, 2, 4, 8, 16):
                 scratch = torch.empty(max(ksplit * m * n, 1),
                                       dtype=torch.float32, device="cuda")
