@@ -177,6 +177,10 @@ class LlamaDecodeEngine:
         self.buf_c32 = torch.empty(B * scratch, dtype=torch.float32,
                                    device=self.device)
         self.scale = 1.0 / math.sqrt(d)
+        self.attn_nsplit = ops.pick_attn_nsplit(B, w.hkv)
+        self.buf_attn_ws = torch.empty(
+            B * w.hq * self.attn_nsplit * (d + 2), dtype=torch.float32,
+            device=self.device) if self.on_gpu else None
         self._graph = None
         self._ksplits = {}
 
@@ -224,7 +228,9 @@ class LlamaDecodeEngine:
             q = self.buf_qkv[:, :w.hq * d].view(B, w.hq, d)
             attn_view = self.buf_attn_out.view(B, w.hq, d)
             ops.attn_decode(q, self.k_cache[li], self.v_cache[li],
-                            self.cache_lens, self.scale, out=attn_view)
+                            self.cache_lens, self.scale, out=attn_view,
+                            partial_ws=self.buf_attn_ws,
+                            nsplit=self.attn_nsplit)
             self._gemm(self.buf_attn_out, layer["wo"], self.buf_proj)
             self._maybe_allreduce(self.buf_proj)
             ops.fused_add_rmsnorm(self.buf_proj, layer["ffn_norm"],

@@ -136,7 +136,8 @@ void cast_f32_bf16(torch::Tensor out, torch::Tensor in) {
 }
 
 void attn_decode(torch::Tensor o, torch::Tensor q, torch::Tensor kc,
-                 torch::Tensor vc, torch::Tensor seq_lens, double scale) {
+                 torch::Tensor vc, torch::Tensor seq_lens, double scale,
+                 c10::optional<torch::Tensor> partial_ws, int64_t nsplit) {
   CHECK_DEV(o); CHECK_CONTIG(o); CHECK_BF16(o);
   CHECK_DEV(q); CHECK_BF16(q);
   CHECK_DEV(kc); CHECK_CONTIG(kc); CHECK_BF16(kc);
@@ -150,9 +151,19 @@ void attn_decode(torch::Tensor o, torch::Tensor q, torch::Tensor kc,
   TORCH_CHECK(Hq % Hkv == 0 && Hq / Hkv <= 8,
               "grouped heads per kv head must divide and be <= 8");
   TORCH_CHECK(kc.size(0) == B && kc.size(3) == D, "kc shape mismatch");
+  float* ws = nullptr;
+  if (nsplit > 1) {
+    TORCH_CHECK(partial_ws.has_value(), "nsplit>1 needs partial_ws");
+    CHECK_DEV(*partial_ws); CHECK_CONTIG(*partial_ws);
+    CHECK_F32(*partial_ws);
+    TORCH_CHECK(partial_ws->numel() >=
+                (int64_t)B * Hq * nsplit * (D + 2),
+                "partial_ws too small");
+    ws = (float*)partial_ws->data_ptr();
+  }
   launch_attn_decode(o.data_ptr(), q.data_ptr(), kc.data_ptr(),
                      vc.data_ptr(), seq_lens.data_ptr(), B, Hq, Hkv, Smax,
-                     (float)scale, q_row_stride, stream());
+                     (float)scale, q_row_stride, ws, (int)nsplit, stream());
 }
 
 void kv_append(torch::Tensor kc, torch::Tensor vc, torch::Tensor knew,

@@ -39,7 +39,8 @@ void launch_zero_f32(void* p, long long n, void* stream);
 void launch_attn_decode(void* O, const void* Q, const void* Kc,
                         const void* Vc, const void* seq_lens, int B, int Hq,
                         int Hkv, int Smax, float scale,
-                        long long q_row_stride, void* stream);
+                        long long q_row_stride, float* partial_ws,
+                        int nsplit, void* stream);
 
 void launch_kv_append(void* Kc, void* Vc, const void* knew, const void* vnew,
                       const void* positions, int B, int Hkv, int Smax, int D,

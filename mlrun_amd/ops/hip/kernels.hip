@@ -673,16 +673,145 @@ __global__ void attn_decode_kernel(
   op[1] = f2bf(acc1 * inv);
 }
 
+// ---------------------------------------------------------------------
+// Flash-decode split-S: nsplit workgroups per (b, kv-head) each cover
+// a slice of the sequence and write an unnormalized partial
+// (acc[D], m, l) to the workspace; a combine kernel folds the slices.
+// Fills the chip even at small B*Hkv (the single-workgroup variant
+// peaked at 128 workgroups on 256 CUs).
+// partial layout: [B, Hq, nsplit, D+2] f32
+// ---------------------------------------------------------------------
+__global__ void attn_decode_split_kernel(
+    float* __restrict__ partial, const unsigned short* __restrict__ Q,
+    const unsigned short* __restrict__ Kc,
+    const unsigned short* __restrict__ Vc,
+    const int* __restrict__ seq_lens, int B, int Hq, int Hkv, int Smax,
+    float scale, long long q_row_stride, int nsplit) {
+  constexpr int D = 128;
+  const int b = blockIdx.x / Hkv;
+  const int kvh = blockIdx.x % Hkv;
+  const int split = blockIdx.y;
+  const int G = Hq / Hkv;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int h = kvh * G + wave;
+  const int len = seq_lens[b];
+  const int per = (len + nsplit - 1) / nsplit;
+  const int s_begin = split * per;
+  const int s_final = min(len, s_begin + per);
+  float* prow = partial + (((size_t)b * Hq + h) * nsplit + split) * (D + 2);
+
+  __shared__ float scores[ATTN_MAXG][ATTN_SCHUNK];
+
+  const unsigned short* qp = Q + (size_t)b * q_row_stride +
+                             (size_t)h * D + (lane & 15) * 8;
+  float qf[8];
+  {
+    ushort8 qv = *reinterpret_cast<const ushort8*>(qp);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) qf[j] = bf2f(qv.v[j]);
+  }
+  const size_t kv_base = ((size_t)b * Hkv + kvh) * Smax * D;
+  const unsigned short* kbase = Kc + kv_base;
+  const unsigned short* vbase = Vc + kv_base;
+
+  float m = -FLT_MAX, l = 0.f;
+  float acc0 = 0.f, acc1 = 0.f;
+  for (int s0 = s_begin; s0 < s_final; s0 += ATTN_SCHUNK) {
+    const int cnt = min(ATTN_SCHUNK, s_final - s0);
+    for (int si = lane >> 4; si < cnt; si += 4) {
+      const unsigned short* kp = kbase + (size_t)(s0 + si) * D +
+                                 (lane & 15) * 8;
+      ushort8 kv = *reinterpret_cast<const ushort8*>(kp);
+      float dot = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dot += qf[j] * bf2f(kv.v[j]);
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        dot += __shfl_xor(dot, off, 64);
+      if ((lane & 15) == 0) scores[wave][si] = dot * scale;
+    }
+    __syncthreads();
+    float cm = -FLT_MAX;
+    for (int si = lane; si < cnt; si += 64)
+      cm = fmaxf(cm, scores[wave][si]);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      cm = fmaxf(cm, __shfl_xor(cm, off, 64));
+    const float m_new = fmaxf(m, cm);
+    const float rescale = (m == -FLT_MAX) ? 0.f : __expf(m - m_new);
+    acc0 *= rescale;
+    acc1 *= rescale;
+    l *= rescale;
+    const int d0 = lane * 2;
+    for (int si = 0; si < cnt; ++si) {
+      const float p = __expf(scores[wave][si] - m_new);
+      l += p;
+      const unsigned short* vp = vbase + (size_t)(s0 + si) * D + d0;
+      ushort2v vv = *reinterpret_cast<const ushort2v*>(vp);
+      acc0 += p * bf2f(vv.x);
+      acc1 += p * bf2f(vv.y);
+    }
+    m = m_new;
+    __syncthreads();
+  }
+  prow[lane * 2] = acc0;
+  prow[lane * 2 + 1] = acc1;
+  if (lane == 0) {
+    prow[D] = m;
+    prow[D + 1] = l;
+  }
+}
+
+// fold the nsplit partials: one wave per (b, h)
+__global__ void attn_decode_combine_kernel(
+    unsigned short* __restrict__ O, const float* __restrict__ partial,
+    int Hq, int nsplit) {
+  constexpr int D = 128;
+  const int bh = blockIdx.x;
+  const int lane = threadIdx.x;
+  const float* base = partial + (size_t)bh * nsplit * (D + 2);
+  float m_star = -FLT_MAX;
+  for (int s = 0; s < nsplit; ++s)
+    m_star = fmaxf(m_star, base[s * (D + 2) + D]);
+  float l_total = 0.f, a0 = 0.f, a1 = 0.f;
+  for (int s = 0; s < nsplit; ++s) {
+    const float* prow = base + s * (D + 2);
+    const float ms = prow[D];
+    if (ms == -FLT_MAX) continue;
+    const float wgt = __expf(ms - m_star);
+    l_total += wgt * prow[D + 1];
+    a0 += wgt * prow[lane * 2];
+    a1 += wgt * prow[lane * 2 + 1];
+  }
+  const float inv = (l_total > 0.f) ? 1.f / l_total : 0.f;
+  unsigned short* op = O + (size_t)bh * D + lane * 2;
+  op[0] = f2bf(a0 * inv);
+  op[1] = f2bf(a1 * inv);
+}
+
 void launch_attn_decode(void* O, const void* Q, const void* Kc,
                         const void* Vc, const void* seq_lens, int B, int Hq,
                         int Hkv, int Smax, float scale,
-                        long long q_row_stride, void* stream) {
+                        long long q_row_stride, float* partial_ws,
+                        int nsplit, void* stream) {
   const int G = Hq / Hkv;
-  hipLaunchKernelGGL(attn_decode_kernel, dim3(B * Hkv), dim3(G * 64), 0,
-                     (hipStream_t)stream, (unsigned short*)O,
+  if (nsplit <= 1 || partial_ws == nullptr) {
+    hipLaunchKernelGGL(attn_decode_kernel, dim3(B * Hkv), dim3(G * 64), 0,
+                       (hipStream_t)stream, (unsigned short*)O,
+                       (const unsigned short*)Q, (const unsigned short*)Kc,
+                       (const unsigned short*)Vc, (const int*)seq_lens, B,
+                       Hq, Hkv, Smax, scale, q_row_stride);
+    return;
+  }
+  hipLaunchKernelGGL(attn_decode_split_kernel, dim3(B * Hkv, nsplit),
+                     dim3(G * 64), 0, (hipStream_t)stream, partial_ws,
                      (const unsigned short*)Q, (const unsigned short*)Kc,
                      (const unsigned short*)Vc, (const int*)seq_lens, B, Hq,
-                     Hkv, Smax, scale, q_row_stride);
+                     Hkv, Smax, scale, q_row_stride, nsplit);
+  hipLaunchKernelGGL(attn_decode_combine_kernel, dim3(B * Hq), dim3(64), 0,
+                     (hipStream_t)stream, (unsigned short*)O, partial_ws,
+                     Hq, nsplit);
 }
 
 // ---------------------------------------------------------------------
