@@ -1,0 +1,427 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""The mlrun_amd API service: FastAPI app over the node-local SQLite
+run DB.
+
+Parity target: reference server/api/main.py:93 (app assembly) + the
+endpoint modules under server/api/api/endpoints/ (runs, artifacts,
+functions, projects, logs, schedules, feature-store, model-endpoints,
+background tasks, submit).  Periodic loops (runs monitoring :608,
+scheduler ticks) run as in-process asyncio tasks instead of k8s
+informers.
+"""
+
+import asyncio
+import json
+import traceback
+import typing
+
+from fastapi import FastAPI, Query, Request, Response
+from fastapi.responses import JSONResponse
+
+from ..config import config
+from ..db.sqldb import SQLRunDB
+from ..errors import MLRunBaseError, MLRunNotFoundError, err_to_status
+from ..model import RunStates
+from ..utils import logger, now_iso
+from .scheduler import Scheduler
+
+
+def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
+    db = db or SQLRunDB(config.httpdb.dsn or "")
+    app = FastAPI(title="mlrun-amd-api", version="v1")
+    app.state.db = db
+    app.state.scheduler = Scheduler(db) if with_scheduler else None
+
+    @app.exception_handler(MLRunBaseError)
+    async def mlrun_error_handler(request: Request, exc: MLRunBaseError):
+        return JSONResponse(status_code=err_to_status(exc),
+                            content={"detail": str(exc)})
+
+    @app.on_event("startup")
+    async def startup():
+        if app.state.scheduler:
+            app.state.scheduler.start()
+        app.state.monitor_task = asyncio.create_task(_runs_monitor(db))
+
+    @app.on_event("shutdown")
+    async def shutdown():
+        if app.state.scheduler:
+            app.state.scheduler.stop()
+        task = getattr(app.state, "monitor_task", None)
+        if task:
+            task.cancel()
+
+    # ------------------------------------------------------------ misc
+    @app.get("/api/v1/healthz")
+    async def healthz():
+        from .. import __version__
+
+        return {"status": "ok", "version": __version__}
+
+    @app.get("/api/v1/client-spec")
+    async def client_spec():
+        return {"version": config.version,
+                "default_project": config.default_project}
+
+    # ------------------------------------------------------------ runs
+    @app.post("/api/v1/run/{project}/{uid}")
+    async def store_run(project: str, uid: str, request: Request,
+                        iter: int = 0):
+        body = await request.json()
+        db.store_run(body, uid, project, iter=iter)
+        return {}
+
+    @app.patch("/api/v1/run/{project}/{uid}")
+    async def update_run(project: str, uid: str, request: Request,
+                         iter: int = 0):
+        body = await request.json()
+        return db.update_run(body, uid, project, iter=iter)
+
+    @app.get("/api/v1/run/{project}/{uid}")
+    async def read_run(project: str, uid: str, iter: int = 0):
+        return {"data": db.read_run(uid, project, iter=iter)}
+
+    @app.delete("/api/v1/run/{project}/{uid}")
+    async def del_run(project: str, uid: str, iter: int = 0):
+        db.del_run(uid, project, iter=iter)
+        return {}
+
+    @app.post("/api/v1/run/{project}/{uid}/abort")
+    async def abort_run(project: str, uid: str, request: Request):
+        try:
+            body = await request.json()
+        except Exception:
+            body = {}
+        db.abort_run(uid, project,
+                     status_text=body.get("status_text", ""))
+        return {}
+
+    @app.get("/api/v1/runs")
+    async def list_runs(project: str = "default", name: str = "",
+                        state: str = "", sort: int = 1, last: int = 0,
+                        iter: int = 0, uid: typing.List[str] = Query(None),
+                        label: typing.List[str] = Query(None)):
+        runs = db.list_runs(name=name, uid=uid, project=project,
+                            labels=label, state=state or None,
+                            sort=bool(sort), last=last, iter=bool(iter))
+        return {"runs": runs}
+
+    # ------------------------------------------------------------ logs
+    @app.post("/api/v1/log/{project}/{uid}")
+    async def store_log(project: str, uid: str, request: Request,
+                        append: int = 0):
+        body = await request.body()
+        db.store_log(uid, project, body, append=bool(append))
+        return {}
+
+    @app.get("/api/v1/log/{project}/{uid}")
+    async def get_log(project: str, uid: str, offset: int = 0,
+                      size: int = 0):
+        state, data = db.get_log(uid, project, offset=offset, size=size)
+        return Response(content=data,
+                        headers={"x-mlrun-run-state": state or ""},
+                        media_type="application/octet-stream")
+
+    # -------------------------------------------------------- artifacts
+    @app.post("/api/v1/artifact/{project}/{key:path}")
+    async def store_artifact(project: str, key: str, request: Request,
+                             tag: str = "", iter: int = 0, tree: str = "",
+                             uid: str = ""):
+        body = await request.json()
+        return db.store_artifact(key, body, uid=uid or None, iter=iter,
+                                 tag=tag, project=project,
+                                 tree=tree or None)
+
+    @app.get("/api/v1/artifact/{project}/{key:path}")
+    async def read_artifact(project: str, key: str, tag: str = "",
+                            iter: int = 0, tree: str = "", uid: str = ""):
+        return {"data": db.read_artifact(key, tag=tag, iter=iter,
+                                         project=project,
+                                         tree=tree or None,
+                                         uid=uid or None)}
+
+    @app.delete("/api/v1/artifact/{project}/{key:path}")
+    async def del_artifact(project: str, key: str, tag: str = ""):
+        db.del_artifact(key, tag=tag, project=project)
+        return {}
+
+    @app.get("/api/v1/artifacts")
+    async def list_artifacts(project: str = "default", name: str = "",
+                             tag: str = "", kind: str = "", tree: str = "",
+                             label: typing.List[str] = Query(None)):
+        artifacts = db.list_artifacts(name=name, project=project, tag=tag,
+                                      labels=label, kind=kind or None,
+                                      tree=tree or None)
+        return {"artifacts": artifacts}
+
+    # -------------------------------------------------------- functions
+    @app.post("/api/v1/func/{project}/{name}")
+    async def store_function(project: str, name: str, request: Request,
+                             tag: str = "", versioned: int = 0):
+        body = await request.json()
+        hash_key = db.store_function(body, name, project, tag=tag,
+                                     versioned=bool(versioned))
+        return {"hash_key": hash_key}
+
+    @app.get("/api/v1/func/{project}/{name}")
+    async def get_function(project: str, name: str, tag: str = "",
+                           hash_key: str = ""):
+        return {"func": db.get_function(name, project, tag=tag,
+                                        hash_key=hash_key)}
+
+    @app.delete("/api/v1/func/{project}/{name}")
+    async def delete_function(project: str, name: str):
+        db.delete_function(name, project)
+        return {}
+
+    @app.get("/api/v1/funcs")
+    async def list_functions(project: str = "default", name: str = "",
+                             tag: str = "",
+                             label: typing.List[str] = Query(None)):
+        return {"funcs": db.list_functions(name=name or None,
+                                           project=project, tag=tag,
+                                           labels=label)}
+
+    # --------------------------------------------------------- projects
+    @app.post("/api/v1/projects")
+    async def create_project(request: Request):
+        body = await request.json()
+        return db.create_project(body)
+
+    @app.get("/api/v1/projects/{name}")
+    async def get_project(name: str):
+        return db.get_project(name)
+
+    @app.put("/api/v1/projects/{name}")
+    async def store_project(name: str, request: Request):
+        body = await request.json()
+        return db.store_project(name, body)
+
+    @app.get("/api/v1/projects")
+    async def list_projects():
+        return {"projects": db.list_projects()}
+
+    @app.delete("/api/v1/projects/{name}")
+    async def delete_project(name: str, deletion_strategy: str = ""):
+        db.delete_project(name, deletion_strategy or None)
+        return {}
+
+    # -------------------------------------------------------- schedules
+    @app.post("/api/v1/projects/{project}/schedules")
+    async def create_schedule(project: str, request: Request):
+        body = await request.json()
+        db.create_schedule(project, body)
+        if app.state.scheduler:
+            app.state.scheduler.reload()
+        return {}
+
+    @app.put("/api/v1/projects/{project}/schedules/{name}")
+    async def update_schedule(project: str, name: str, request: Request):
+        body = await request.json()
+        db.update_schedule(project, name, body)
+        if app.state.scheduler:
+            app.state.scheduler.reload()
+        return {}
+
+    @app.get("/api/v1/projects/{project}/schedules/{name}")
+    async def get_schedule(project: str, name: str):
+        return db.get_schedule(project, name)
+
+    @app.get("/api/v1/projects/{project}/schedules")
+    async def list_schedules(project: str, name: str = ""):
+        return {"schedules": db.list_schedules(project, name=name)}
+
+    @app.delete("/api/v1/projects/{project}/schedules/{name}")
+    async def delete_schedule(project: str, name: str):
+        db.delete_schedule(project, name)
+        if app.state.scheduler:
+            app.state.scheduler.reload()
+        return {}
+
+    @app.post("/api/v1/projects/{project}/schedules/{name}/invoke")
+    async def invoke_schedule(project: str, name: str):
+        scheduler = app.state.scheduler or Scheduler(db)
+        await asyncio.to_thread(scheduler.invoke, project, name)
+        return {}
+
+    # ----------------------------------------------------- feature store
+    @app.put("/api/v1/projects/{project}/feature-sets/{name}")
+    async def store_feature_set(project: str, name: str, request: Request,
+                                tag: str = ""):
+        body = await request.json()
+        return db.store_feature_set(body, name=name, project=project,
+                                    tag=tag or None)
+
+    @app.get("/api/v1/projects/{project}/feature-sets/{name}")
+    async def get_feature_set(project: str, name: str, tag: str = ""):
+        return db.get_feature_set(name, project, tag=tag or None)
+
+    @app.get("/api/v1/projects/{project}/feature-sets")
+    async def list_feature_sets(project: str, name: str = ""):
+        return {"feature_sets": db.list_feature_sets(project,
+                                                     name=name or None)}
+
+    @app.delete("/api/v1/projects/{project}/feature-sets/{name}")
+    async def delete_feature_set(project: str, name: str):
+        db.delete_feature_set(name, project)
+        return {}
+
+    @app.put("/api/v1/projects/{project}/feature-vectors/{name}")
+    async def store_feature_vector(project: str, name: str, request: Request,
+                                   tag: str = ""):
+        body = await request.json()
+        return db.store_feature_vector(body, name=name, project=project,
+                                       tag=tag or None)
+
+    @app.get("/api/v1/projects/{project}/feature-vectors/{name}")
+    async def get_feature_vector(project: str, name: str, tag: str = ""):
+        return db.get_feature_vector(name, project, tag=tag or None)
+
+    @app.get("/api/v1/projects/{project}/feature-vectors")
+    async def list_feature_vectors(project: str, name: str = ""):
+        return {"feature_vectors": db.list_feature_vectors(
+            project, name=name or None)}
+
+    @app.delete("/api/v1/projects/{project}/feature-vectors/{name}")
+    async def delete_feature_vector(project: str, name: str):
+        db.delete_feature_vector(name, project)
+        return {}
+
+    # --------------------------------------------------- model endpoints
+    @app.put("/api/v1/projects/{project}/model-endpoints/{endpoint_id}")
+    async def store_model_endpoint(project: str, endpoint_id: str,
+                                   request: Request):
+        body = await request.json()
+        db.store_model_endpoint(project, endpoint_id, body)
+        return {}
+
+    @app.get("/api/v1/projects/{project}/model-endpoints/{endpoint_id}")
+    async def get_model_endpoint(project: str, endpoint_id: str):
+        return db.get_model_endpoint(project, endpoint_id)
+
+    @app.get("/api/v1/projects/{project}/model-endpoints")
+    async def list_model_endpoints(project: str, model: str = "",
+                                   function: str = ""):
+        return {"endpoints": db.list_model_endpoints(
+            project, model=model or None, function=function or None)}
+
+    @app.delete("/api/v1/projects/{project}/model-endpoints/{endpoint_id}")
+    async def delete_model_endpoint(project: str, endpoint_id: str):
+        db.delete_model_endpoint(project, endpoint_id)
+        return {}
+
+    # ------------------------------------------------------------ alerts
+    @app.put("/api/v1/projects/{project}/alerts/{name}")
+    async def store_alert(project: str, name: str, request: Request):
+        body = await request.json()
+        db.store_alert_config(project, name, body)
+        return {}
+
+    @app.get("/api/v1/projects/{project}/alerts/{name}")
+    async def get_alert(project: str, name: str):
+        return db.get_alert_config(project, name)
+
+    @app.get("/api/v1/projects/{project}/alerts")
+    async def list_alerts(project: str):
+        return {"alerts": db.list_alert_configs(project)}
+
+    @app.delete("/api/v1/projects/{project}/alerts/{name}")
+    async def delete_alert(project: str, name: str):
+        db.delete_alert_config(project, name)
+        return {}
+
+    @app.post("/api/v1/projects/{project}/events/{name}")
+    async def post_event(project: str, name: str, request: Request):
+        body = await request.json()
+        from .events import process_event
+
+        results = await asyncio.to_thread(process_event, project, name,
+                                          body, db)
+        return {"alerts_fired": results}
+
+    # -------------------------------------------------- background tasks
+    @app.get("/api/v1/projects/{project}/background-tasks/{name}")
+    async def get_background_task(project: str, name: str):
+        return db.get_background_task(project, name)
+
+    @app.get("/api/v1/projects/{project}/background-tasks")
+    async def list_background_tasks(project: str):
+        return {"background_tasks": db.list_background_tasks(project)}
+
+    # ------------------------------------------------------------ submit
+    @app.post("/api/v1/submit_job")
+    async def submit_job(request: Request):
+        body = await request.json()
+        task = body.get("task", body)
+        schedule = body.get("schedule")
+        from ..model import RunObject
+
+        run = RunObject.from_dict(task)
+        result = await asyncio.to_thread(db.submit_job, run, schedule)
+        return result if "data" in result else {"data": result}
+
+    return app
+
+
+async def _runs_monitor(db: SQLRunDB):
+    """Periodic run monitoring: abort runs stuck past state thresholds
+    (parity: reference main.py:608 + runtime_handlers/base.py:1387)."""
+    import datetime
+
+    interval = int(config.runs.monitoring_interval)
+    while True:
+        try:
+            await asyncio.sleep(interval)
+            thresholds = config.runs.state_thresholds.to_dict()
+            now = datetime.datetime.now(datetime.timezone.utc)
+            for state, limit in (("running", thresholds.get("running")),
+                                 ("pending", thresholds.get("pending"))):
+                for run in db.list_runs(project="", state=state):
+                    pass  # project-scoped listing below
+            for project in [p.get("metadata", {}).get("name", "default")
+                            for p in db.list_projects()] or ["default"]:
+                for state in ("running", "pending"):
+                    limit = thresholds.get(state)
+                    if not limit:
+                        continue
+                    for run in db.list_runs(project=project, state=state):
+                        started = run.get("status", {}).get("start_time")
+                        if not started:
+                            continue
+                        try:
+                            from ..utils import parse_time
+
+                            start_dt = parse_time(started)
+                            if start_dt is None:
+                                continue
+                            if start_dt.tzinfo is None:
+                                start_dt = start_dt.replace(
+                                    tzinfo=datetime.timezone.utc)
+                            age = (now - start_dt).total_seconds()
+                        except Exception:
+                            continue
+                        if age > float(limit):
+                            uid = run.get("metadata", {}).get("uid")
+                            logger.warning(
+                                "aborting run past state threshold",
+                                uid=uid, state=state, age=age)
+                            db.abort_run(
+                                uid, project,
+                                status_text=f"aborted by monitor: {state} "
+                                            f"for {int(age)}s > {limit}s")
+        except asyncio.CancelledError:
+            return
+        except Exception as exc:
+            logger.warning("runs monitor iteration failed", error=str(exc))
+
+
+def serve(port: int = None, db=None):
+    """Run the API service (blocking)."""
+    import uvicorn
+
+    app = create_app(db)
+    uvicorn.run(app, host="0.0.0.0",
+                port=port or int(config.httpdb.port), log_level="warning")

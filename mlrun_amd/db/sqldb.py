@@ -263,6 +263,9 @@ class SQLRunDB(RunDBInterface):
             self.del_run(meta.get("uid"), project, meta.get("iteration") or 0)
 
     def abort_run(self, uid, project="", iter=0, status_text=""):
+        current = self.read_run(uid, project, iter)
+        if RunStates.is_terminal(current.get("status", {}).get("state", "")):
+            return  # terminal runs are not abortable (reference crud)
         self.update_run(
             {"status.state": RunStates.aborted,
              "status.status_text": status_text or "aborted by request",

@@ -1,0 +1,186 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""API-service tests: FastAPI TestClient + fresh SQLite per test
+(mirrors reference tests/api/conftest.py seam) — including the
+client->server round trip through HTTPRunDB."""
+
+import datetime
+
+import pytest
+from fastapi.testclient import TestClient
+
+from mlrun_amd.api import create_app, CronTrigger
+from mlrun_amd.db.sqldb import SQLRunDB
+
+
+@pytest.fixture
+def client(tmp_path):
+    db = SQLRunDB(str(tmp_path / "api.db"))
+    app = create_app(db, with_scheduler=False)
+    with TestClient(app) as test_client:
+        test_client.db = db
+        yield test_client
+
+
+class TestEndpoints:
+    def test_healthz(self, client):
+        resp = client.get("/api/v1/healthz")
+        assert resp.status_code == 200
+        assert resp.json()["status"] == "ok"
+
+    def test_runs_crud(self, client):
+        run = {"metadata": {"name": "r1", "uid": "u1"},
+               "status": {"state": "running"}}
+        assert client.post("/api/v1/run/p/u1", json=run).status_code == 200
+        resp = client.get("/api/v1/run/p/u1")
+        assert resp.json()["data"]["metadata"]["name"] == "r1"
+        client.patch("/api/v1/run/p/u1",
+                     json={"status.state": "completed"})
+        runs = client.get("/api/v1/runs",
+                          params={"project": "p"}).json()["runs"]
+        assert len(runs) == 1 and runs[0]["status"]["state"] == "completed"
+        client.post("/api/v1/run/p/u1/abort", json={})
+        assert client.get("/api/v1/run/p/u1").json()[
+            "data"]["status"]["state"] == "completed"  # terminal wins
+        client.delete("/api/v1/run/p/u1")
+        assert client.get("/api/v1/run/p/u1").status_code == 404
+
+    def test_artifact_and_function_routes(self, client):
+        artifact = {"kind": "model", "metadata": {"key": "m"},
+                    "spec": {"target_path": "/tmp/m"}}
+        client.post("/api/v1/artifact/p/m", json=artifact,
+                    params={"tree": "t1"})
+        assert client.get("/api/v1/artifact/p/m").json()[
+            "data"]["spec"]["target_path"] == "/tmp/m"
+        assert len(client.get("/api/v1/artifacts",
+                              params={"project": "p"}).json()[
+            "artifacts"]) == 1
+        func = {"kind": "job", "metadata": {"name": "f"}}
+        hash_key = client.post("/api/v1/func/p/f", json=func).json()[
+            "hash_key"]
+        assert hash_key
+        assert client.get("/api/v1/func/p/f").json()["func"]["kind"] == \
+            "job"
+
+    def test_logs(self, client):
+        client.post("/api/v1/log/p/u1", content=b"line one\n")
+        client.post("/api/v1/log/p/u1", content=b"line two",
+                    params={"append": 1})
+        resp = client.get("/api/v1/log/p/u1")
+        assert resp.content == b"line one\nline two"
+
+    def test_projects_and_schedules(self, client):
+        client.post("/api/v1/projects",
+                    json={"metadata": {"name": "proj"}})
+        assert client.get("/api/v1/projects/proj").status_code == 200
+        sched = {"name": "s1", "kind": "job",
+                 "cron_trigger": "0 * * * *", "task": {}}
+        client.post("/api/v1/projects/proj/schedules", json=sched)
+        listed = client.get("/api/v1/projects/proj/schedules").json()
+        assert len(listed["schedules"]) == 1
+        client.delete("/api/v1/projects/proj/schedules/s1")
+        assert client.get("/api/v1/projects/proj/schedules").json()[
+            "schedules"] == []
+
+    def test_submit_job(self, client, tmp_path):
+        task = {
+            "metadata": {"name": "apirun", "project": "default"},
+            "spec": {"handler": None, "function": ""},
+        }
+        resp = client.post("/api/v1/submit_job", json={"task": task})
+        assert resp.status_code == 200
+        # the run executed server-side and is in the DB
+        data = resp.json()["data"]
+        assert data["status"]["state"] in ("completed", "error")
+
+    def test_error_mapping(self, client):
+        assert client.get("/api/v1/run/p/missing").status_code == 404
+        assert client.get("/api/v1/projects/missing").status_code == 404
+
+    def test_alerts_and_events(self, client):
+        alert = {"name": "a1", "summary": "drift!", "severity": "high",
+                 "trigger": {"events": ["model-drift"]},
+                 "criteria": {"count": 2},
+                 "notifications": [{"kind": "console"}]}
+        client.put("/api/v1/projects/p/alerts/a1", json=alert)
+        # first event: below criteria count
+        resp = client.post("/api/v1/projects/p/events/model-drift",
+                           json={"endpoint": "e1"})
+        assert resp.json()["alerts_fired"] == []
+        resp = client.post("/api/v1/projects/p/events/model-drift",
+                           json={"endpoint": "e1"})
+        assert resp.json()["alerts_fired"] == ["a1"]
+        # unrelated event kind does not fire
+        resp = client.post("/api/v1/projects/p/events/other",
+                           json={})
+        assert resp.json()["alerts_fired"] == []
+
+
+class TestHTTPRunDBClient:
+    """Client round trip against the in-process app (reference seam:
+    config.dbpath -> TestClient)."""
+
+    def test_client_roundtrip(self, client, monkeypatch):
+        from mlrun_amd.db.httpdb import HTTPRunDB
+
+        httpdb = HTTPRunDB("http://testserver")
+        # route the requests session through the TestClient
+        httpdb.session = client
+        orig_request = client.request
+
+        def request(method, url, params=None, data=None, json=None,
+                    headers=None, timeout=None):
+            return orig_request(method, url, params=params, content=data,
+                                json=json, headers=headers)
+
+        httpdb.session.request = request
+        httpdb.store_run({"metadata": {"name": "rr", "uid": "uu"},
+                          "status": {"state": "running"}}, "uu", "p")
+        run = httpdb.read_run("uu", "p")
+        assert run["metadata"]["name"] == "rr"
+        assert len(httpdb.list_runs(project="p")) == 1
+        httpdb.store_log("uu", "p", b"log text")
+        state, log = httpdb.get_log("uu", "p")
+        assert log == b"log text"
+        hash_key = httpdb.store_function({"kind": "job"}, "fn", "p")
+        assert httpdb.get_function("fn", "p")["kind"] == "job"
+
+
+class TestCron:
+    def test_parse_and_match(self):
+        trigger = CronTrigger("*/5 * * * *")
+        dt = datetime.datetime(2026, 9, 12, 10, 5)
+        assert trigger.matches(dt)
+        assert not trigger.matches(dt.replace(minute=7))
+        nxt = trigger.next_fire_time(dt)
+        assert nxt.minute == 10
+
+    def test_invalid(self):
+        from mlrun_amd.errors import MLRunInvalidArgumentError
+
+        with pytest.raises(MLRunInvalidArgumentError):
+            CronTrigger("* * *")
+
+    def test_min_interval(self):
+        assert CronTrigger("*/5 * * * *").min_interval_seconds() == 300
+        assert CronTrigger("0 * * * *").min_interval_seconds() == 3600
+
+    def test_schedule_execution(self, tmp_path):
+        """Invoke a stored schedule and verify the run executed."""
+        from mlrun_amd.api import Scheduler
+        from mlrun_amd.db.sqldb import SQLRunDB
+
+        db = SQLRunDB(str(tmp_path / "sched.db"))
+        task = {"metadata": {"name": "sched-run", "project": "default"},
+                "spec": {}}
+        db.create_schedule("default", {"name": "s1", "kind": "job",
+                                       "cron_trigger": "0 0 1 1 *",
+                                       "task": task})
+        scheduler = Scheduler(db)
+        scheduler.invoke("default", "s1")
+        runs = db.list_runs(project="default")
+        assert len(runs) == 1
+        sched = db.get_schedule("default", "s1")
+        assert sched["last_run_uri"]
