@@ -97,9 +97,12 @@ class WindowRing:
                           self.period_seconds)
         current = int(period_idx_abs.max())
         self._expire_old_periods(current)
-        keys32 = key_ids.to(torch.int32)
-        pidx = (period_idx_abs % self.n_periods).to(torch.int32)
-        vals = values.to(torch.float32)
+        # drop late events older than the ring span (their slot would
+        # alias a live period)
+        fresh = period_idx_abs > (current - self.n_periods)
+        keys32 = key_ids[fresh].to(torch.int32)
+        pidx = (period_idx_abs[fresh] % self.n_periods).to(torch.int32)
+        vals = values[fresh].to(torch.float32)
         if self.ring.is_cuda:
             keys32 = keys32.cuda(self.ring.device)
             pidx = pidx.to(self.ring.device)
@@ -108,9 +111,10 @@ class WindowRing:
             vals_dev = vals
         ops.window_ingest(self.ring, keys32, vals_dev, pidx)
         ops.window_ingest(self.ring_sq, keys32, vals_dev * vals_dev, pidx)
-        # running aggregates (CPU, numpy-vectorized per batch)
+        # running aggregates over the FULL batch (storey keeps running
+        # first/last/min/max even for late events)
         k = key_ids.numpy()
-        v = vals.numpy()
+        v = values.to(torch.float32).numpy()
         order = np.argsort(k, kind="stable")
         k_sorted, v_sorted = k[order], v[order]
         uniq, starts = np.unique(k_sorted, return_index=True)
@@ -132,7 +136,11 @@ class WindowRing:
         (sum/count/avg/stdvar/stddev), each [capacity]."""
         from .. import ops
 
-        window_periods = max(window_seconds // self.period_seconds, 1)
+        # cover every bucket intersecting [now - window, now]: the
+        # current partial bucket + window/period full buckets (events in
+        # the oldest partial bucket are included — bucket-quantized
+        # sliding window, storey-compatible)
+        window_periods = max(window_seconds // self.period_seconds, 1) + 1
         current_period = int(now_ts // self.period_seconds)
         self._expire_old_periods(current_period)
         out = ops.window_reduce(self.ring, window_periods,
