@@ -196,9 +196,9 @@ class LlamaDecodeEngine:
 
     def _maybe_allreduce(self, t):
         if self.tp_size > 1:
-            import torch.distributed as dist
+            from ..parallel.tp import all_reduce_tensor
 
-            dist.all_reduce(t, group=self.tp_group)
+            all_reduce_tensor(t, group=self.tp_group)
         return t
 
     # ---------------------------------------------------------- decode

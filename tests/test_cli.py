@@ -1,0 +1,54 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""CLI tests (python -m mlrun_amd ...)."""
+
+import pytest
+
+from mlrun_amd.__main__ import main
+
+
+class TestCLI:
+    def test_version(self, capsys):
+        main(["version"])
+        assert "version" in capsys.readouterr().out
+
+    def test_config(self, capsys):
+        main(["config"])
+        assert "dbpath" in capsys.readouterr().out
+
+    def test_run_and_get_and_logs(self, tmp_path, capsys, rundb):
+        script = tmp_path / "job.py"
+        script.write_text(
+            "import mlrun_amd\n"
+            "ctx = mlrun_amd.get_or_create_ctx('cli')\n"
+            "ctx.log_result('out', ctx.get_param('x', 0) * 2)\n"
+            "print('hello from job')\n"
+            "ctx.commit(completed=True)\n")
+        main(["run", str(script), "--name", "clirun", "-p", "x=21",
+              "--kind", "local"])
+        out = capsys.readouterr().out
+        assert "completed" in out and "out: 42" in out
+
+        main(["get", "runs"])
+        out = capsys.readouterr().out
+        assert "clirun" in out
+
+        runs = rundb.list_runs(project="default", name="clirun")
+        uid = runs[0]["metadata"]["uid"]
+        main(["logs", uid])
+        out = capsys.readouterr().out
+        assert "hello from job" in out
+
+    def test_run_failure_exit_code(self, tmp_path):
+        script = tmp_path / "bad.py"
+        script.write_text("raise SystemExit(2)\n")
+        with pytest.raises(SystemExit):
+            main(["run", str(script), "--name", "bad", "--kind", "local"])
+
+    def test_clean(self, tmp_path, capsys, rundb):
+        rundb.store_run({"metadata": {"name": "x", "uid": "u9"},
+                         "status": {"state": "completed"}}, "u9", "default")
+        main(["clean"])
+        assert rundb.list_runs(project="default") == []
