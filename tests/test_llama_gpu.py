@@ -72,3 +72,21 @@ class TestLlamaGPU:
                                  "max_tokens": 4})
         assert len(resp["outputs"]) == 2
         assert len(resp["outputs"][0]) == 4
+
+
+@requires_gpu
+class TestTrainGPU:
+    def test_train_step_8b_single_gpu(self):
+        """Config-4 smoke on 1 GPU: a few 8B train steps fit in 288 GB
+        HBM and the loss is finite."""
+        import torch
+        from mlrun_amd.models.llama import LlamaConfig
+        from mlrun_amd.models.llama_train import LlamaTrainer
+
+        cfg = LlamaConfig.llama3_8b(max_seq_len=512)
+        trainer = LlamaTrainer(cfg, device="cuda:0")
+        batch = torch.randint(0, cfg.vocab_size, (2, 512))
+        losses = [trainer.train_step(batch) for _ in range(2)]
+        assert all(torch.isfinite(torch.tensor(losses))), losses
+        del trainer
+        torch.cuda.empty_cache()
