@@ -1,0 +1,140 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Batch sources for ingestion graphs.
+
+Parity target: reference mlrun/datastore/sources.py (CSVSource :162,
+ParquetSource, DataFrameSource, ... KafkaSource :1052).  Node-local
+set: file-based batch sources + an in-process stream source; the
+Kafka/V3IO stream sources of the reference map to StreamSource backed
+by the in-process queue engine.
+"""
+
+import typing
+
+from ..errors import MLRunInvalidArgumentError
+from ..model import ModelObj
+
+
+class BaseSource(ModelObj):
+    kind = "source"
+
+    def __init__(self, name: str = "", path: str = None, attributes=None,
+                 key_field: str = None, time_field: str = None,
+                 schedule: str = None):
+        self.name = name
+        self.path = path
+        self.attributes = attributes or {}
+        self.key_field = key_field
+        self.time_field = time_field
+        self.schedule = schedule
+
+    def to_dataframe(self, columns=None, df_module=None, start_time=None,
+                     end_time=None, time_field=None):
+        raise NotImplementedError
+
+    def filter_time(self, df, start_time=None, end_time=None,
+                    time_field=None):
+        time_field = time_field or self.time_field
+        if time_field and time_field in df.columns:
+            import pandas as pd
+
+            ts = pd.to_datetime(df[time_field])
+            if start_time is not None:
+                df = df[ts >= pd.Timestamp(start_time)]
+            if end_time is not None:
+                df = df[ts < pd.Timestamp(end_time)]
+        return df
+
+
+class CSVSource(BaseSource):
+    kind = "csv"
+
+    def __init__(self, name="", path=None, attributes=None, key_field=None,
+                 time_field=None, schedule=None, parse_dates=None):
+        super().__init__(name, path, attributes, key_field, time_field,
+                         schedule)
+        self.parse_dates = parse_dates
+
+    def to_dataframe(self, columns=None, df_module=None, start_time=None,
+                     end_time=None, time_field=None):
+        import pandas as pd
+
+        df = pd.read_csv(self.path, usecols=columns,
+                         parse_dates=self.parse_dates,
+                         **self.attributes)
+        return self.filter_time(df, start_time, end_time, time_field)
+
+
+class ParquetSource(BaseSource):
+    kind = "parquet"
+
+    def to_dataframe(self, columns=None, df_module=None, start_time=None,
+                     end_time=None, time_field=None):
+        import pandas as pd
+
+        df = pd.read_parquet(self.path, columns=columns, **self.attributes)
+        return self.filter_time(df, start_time, end_time, time_field)
+
+
+class DataFrameSource(BaseSource):
+    kind = "dataframe"
+
+    def __init__(self, df=None, name="", key_field=None, time_field=None):
+        super().__init__(name, None, None, key_field, time_field)
+        self._df = df
+
+    def to_dataframe(self, columns=None, df_module=None, start_time=None,
+                     end_time=None, time_field=None):
+        df = self._df
+        if columns:
+            df = df[columns]
+        return self.filter_time(df, start_time, end_time, time_field)
+
+
+class StreamSource(BaseSource):
+    """In-process stream source: push() events, drained in batches by
+    the ingestion service (the KafkaSource/V3IO-stream analog)."""
+
+    kind = "stream"
+
+    def __init__(self, name="", key_field=None, time_field=None,
+                 max_events: int = 65536):
+        super().__init__(name, None, None, key_field, time_field)
+        import queue
+
+        self._queue = queue.Queue(maxsize=max_events)
+
+    def push(self, event: typing.Union[dict, list]):
+        events = event if isinstance(event, list) else [event]
+        for item in events:
+            self._queue.put(item)
+
+    def drain(self, max_batch: int = 4096) -> list:
+        out = []
+        while len(out) < max_batch:
+            try:
+                out.append(self._queue.get_nowait())
+            except Exception:
+                break
+        return out
+
+    def to_dataframe(self, columns=None, df_module=None, start_time=None,
+                     end_time=None, time_field=None):
+        import pandas as pd
+
+        df = pd.DataFrame(self.drain())
+        if columns:
+            df = df[columns]
+        return df
+
+
+def get_source_from_dict(struct: dict) -> BaseSource:
+    kinds = {"csv": CSVSource, "parquet": ParquetSource,
+             "dataframe": DataFrameSource, "stream": StreamSource}
+    kind = struct.get("kind", "csv")
+    cls = kinds.get(kind)
+    if cls is None:
+        raise MLRunInvalidArgumentError(f"unknown source kind {kind}")
+    return cls.from_dict(struct)

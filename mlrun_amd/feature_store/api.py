@@ -49,10 +49,14 @@ def ingest(featureset: typing.Union[FeatureSet, str] = None, source=None,
     + the online window/KV table)."""
     import pandas as pd
 
+    from ..datastore.sources import BaseSource
+
     fset = _resolve_feature_set(featureset)
     if source is None:
         source = fset.spec.source
-    if isinstance(source, str):
+    if isinstance(source, BaseSource):
+        df = source.to_dataframe()
+    elif isinstance(source, str):
         if source.endswith(".csv"):
             df = pd.read_csv(source)
         elif source.endswith((".parquet", ".pq")):
@@ -62,7 +66,8 @@ def ingest(featureset: typing.Union[FeatureSet, str] = None, source=None,
     elif isinstance(source, pd.DataFrame):
         df = source.copy()
     else:
-        raise MLRunInvalidArgumentError("source must be a DataFrame or path")
+        raise MLRunInvalidArgumentError(
+            "source must be a DataFrame, path, or Source object")
 
     # run the transform graph (batch steps)
     df = _run_graph(fset, df)
@@ -92,8 +97,20 @@ def ingest(featureset: typing.Union[FeatureSet, str] = None, source=None,
             {**key, **{c: rec.get(c) for c in agg_cols}}
             for key, rec in zip(unique_keys, agg_records)])
         df = df.merge(agg_df, on=entities, how="left")
+    from ..datastore.targets import BaseStoreTarget, get_target_from_spec
+
     fset.status.targets = []
     for target in targets:
+        if isinstance(target, (BaseStoreTarget, dict)) or \
+                (isinstance(target, str) and target not in
+                 ("parquet", "offline", "nosql", "online")):
+            tgt = get_target_from_spec(target)
+            if tgt.kind == "nosql" and fset.spec.aggregations:
+                path = f"online://{fset.fullname}"  # folded above
+            else:
+                path = tgt.write_dataframe(df, fset)
+            fset.status.targets.append(tgt.status_entry(path))
+            continue
         kind = target if isinstance(target, str) else target.get("kind")
         if kind in ("parquet", "offline"):
             path = _parquet_target_path(fset)
