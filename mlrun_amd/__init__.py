@@ -33,6 +33,11 @@ __all__ = [
     "get_run_db",
     "mlconf",
     "handler",
+    "get_secret_or_env",
+    "mount_v3io",
+    "auto_mount",
+    "v3io_cred",
+    "VolumeMount",
 ]
 
 from .config import config as mlconf  # noqa: E402
@@ -65,6 +70,13 @@ from .projects import (  # noqa: F401,E402
     run_function,
 )
 from .package import handler, ArtifactType  # noqa: F401,E402
+from .secrets import get_secret_or_env  # noqa: F401,E402
+from .platforms import (  # noqa: F401,E402
+    VolumeMount,
+    auto_mount,
+    mount_v3io,
+    v3io_cred,
+)
 
 
 def get_version() -> str:

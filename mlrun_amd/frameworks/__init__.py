@@ -22,3 +22,4 @@ from .tree import (  # noqa: F401
 )
 from .sklearn import PickleModelServer, SKLearnModelServer  # noqa: F401
 from .torch_nn import PyTorchModelServer  # noqa: F401
+from .auto import apply_mlrun, detect_framework, get_model_server_class  # noqa: F401

@@ -385,8 +385,10 @@ class LlamaServer:
         self.device = device
         self.use_graph = use_graph
         self.engine: typing.Optional[LlamaDecodeEngine] = None
-        # borrow the V2 protocol implementation
-        self._v2 = V2ModelServer.__dict__
+        # the engine owns mutable KV/cache state: serialize events
+        import threading
+
+        self._engine_lock = threading.Lock()
 
     def post_init(self, mode="sync"):
         stream = getattr(self.context, "stream", None) if self.context \
@@ -440,9 +442,10 @@ class LlamaServer:
             raise ValueError('expected {"inputs": [[token ids], ...]}')
         max_new = int(body.get("max_tokens", self.max_new_tokens))
         outputs = []
-        for chunk_start in range(0, len(inputs), self.batch_size):
-            chunk = inputs[chunk_start:chunk_start + self.batch_size]
-            outputs.extend(self._generate_chunk(chunk, max_new))
+        with self._engine_lock:
+            for chunk_start in range(0, len(inputs), self.batch_size):
+                chunk = inputs[chunk_start:chunk_start + self.batch_size]
+                outputs.extend(self._generate_chunk(chunk, max_new))
         event.body = {"id": event.id, "model_name": self.name,
                       "outputs": outputs}
         if self._model_logger:

@@ -178,3 +178,72 @@ class TestHub:
         assert fn.metadata.name == "myfunc"
         catalog = hub.get_hub_catalog("testsrc")
         assert catalog[0]["name"] == "myfunc"
+
+
+class TestFrameworksAuto:
+    def test_detect_and_server_table(self):
+        import torch
+
+        from mlrun_amd.frameworks import (
+            apply_mlrun, detect_framework, get_model_server_class)
+
+        assert detect_framework(torch.nn.Linear(2, 2)) == "pytorch"
+        from sklearn.linear_model import LinearRegression
+
+        assert detect_framework(LinearRegression()) == "sklearn"
+        assert get_model_server_class("llama").__name__ == "LlamaServer"
+        assert get_model_server_class(
+            "sklearn").__name__ == "SKLearnModelServer"
+
+    def test_sklearn_interface(self, rundb):
+        import numpy as np
+        from sklearn.linear_model import LinearRegression
+
+        import mlrun_amd
+        from mlrun_amd.frameworks import apply_mlrun
+
+        ctx = mlrun_amd.get_or_create_ctx("sk")
+        model = LinearRegression()
+        x = np.arange(10).reshape(-1, 1)
+        y = 2 * np.arange(10)
+        model.fit(x, y)
+        iface = apply_mlrun(model, context=ctx)
+        iface.log_model()
+        results = iface.evaluate_and_log(
+            x, y, [lambda a, b: float(np.abs(a - b).mean())])
+        assert list(results.values())[0] < 1e-6
+
+    def test_sklearn_gbdt_server_cpu(self):
+        """SKLearn tree ensembles export to the tree kernel path."""
+        import numpy as np
+        from sklearn.ensemble import GradientBoostingRegressor
+
+        import mlrun_amd
+        from mlrun_amd.frameworks.tree import TreeEnsembleModel
+
+        rng = np.random.default_rng(0)
+        x = rng.normal(size=(200, 4))
+        y = x[:, 0] * 2 + x[:, 1]
+        model = GradientBoostingRegressor(n_estimators=20,
+                                          max_depth=3).fit(x, y)
+        gpu_model = TreeEnsembleModel.from_sklearn(model)
+        ours = gpu_model.predict(x[:20]).numpy()
+        theirs = model.predict(x[:20])
+        assert np.abs(ours - theirs).max() < 1e-4
+
+
+class TestPlatforms:
+    def test_output_stream(self):
+        from mlrun_amd.platforms import OutputStream
+
+        stream = OutputStream("proj/events")
+        stream.push({"a": 1})
+        stream.push([{"a": 2}, {"a": 3}])
+        same = OutputStream.get_stream("proj/events")
+        assert len(same.drain()) == 3
+
+    def test_mount_noops(self):
+        import mlrun_amd
+
+        fn = mlrun_amd.new_function(name="m", kind="job")
+        assert mlrun_amd.auto_mount()(fn) is fn
