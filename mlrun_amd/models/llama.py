@@ -185,14 +185,28 @@ class LlamaDecodeEngine:
         self._ksplits = {}
 
     # ------------------------------------------------------------ gemm
-    def _gemm(self, a, w_, out):
-        """skinny GEMM into a preallocated bf16 out + f32 scratch."""
+    def _plan(self, w_):
         key = (w_.shape[0], w_.shape[1])
         if key not in self._ksplits:
-            self._ksplits[key] = ops.pick_gemm_plan(a.shape[0], *key)
-        ksplit, variant = self._ksplits[key]
+            self._ksplits[key] = ops.pick_gemm_plan(self.B, *key)
+        return self._ksplits[key]
+
+    def _gemm(self, a, w_, out):
+        """skinny GEMM into a preallocated bf16 out + f32 scratch."""
+        ksplit, variant = self._plan(w_)
         return ops.skinny_gemm(a, w_, out=out, c_f32=self.buf_c32,
                                ksplit=ksplit, variant=variant)
+
+    def _gemm_slabs(self, a, w_):
+        """split-K GEMM emitting f32 slabs (consumer folds them);
+        returns the ksplit used.  GPU-only."""
+        from mlrun_amd import _hip_ops
+
+        ksplit, variant = self._plan(w_)
+        if ksplit <= 1:
+            return 0  # caller must use the bf16 path
+        _hip_ops.skinny_gemm_slabs(self.buf_c32, a, w_, ksplit, variant)
+        return ksplit
 
     def _maybe_allreduce(self, t):
         if self.tp_size > 1:
@@ -219,33 +233,76 @@ class LlamaDecodeEngine:
         self.buf_positions.copy_(self.cache_lens)
         self.cache_lens.add_(1)
         positions = self.buf_positions
+        # slab fusion (GPU, TP=1): split-K consumers fold the f32 slabs
+        # directly (one less launch + bf16 round-trip per projection);
+        # under TP the all-reduce needs the materialized bf16 output
+        use_slabs = self.on_gpu and self.tp_size == 1
         for li, layer in enumerate(w.layers):
-            # qkv projection
-            self._gemm(self.buf_hidden, layer["wqkv"], self.buf_qkv)
-            ops.rope_kv_fused(self.buf_qkv, self.k_cache[li],
-                              self.v_cache[li], positions, self.cos_sin,
-                              w.hq)
+            # qkv projection -> rope -> caches
+            done = 0
+            if use_slabs:
+                done = self._gemm_slabs(self.buf_hidden, layer["wqkv"])
+                if done:
+                    from mlrun_amd import _hip_ops
+
+                    _hip_ops.rope_kv_slab(
+                        self.buf_qkv, self.k_cache[li], self.v_cache[li],
+                        self.buf_c32, positions, self.cos_sin, w.hq, done)
+            if not done:
+                self._gemm(self.buf_hidden, layer["wqkv"], self.buf_qkv)
+                ops.rope_kv_fused(self.buf_qkv, self.k_cache[li],
+                                  self.v_cache[li], positions,
+                                  self.cos_sin, w.hq)
             q = self.buf_qkv[:, :w.hq * d].view(B, w.hq, d)
             attn_view = self.buf_attn_out.view(B, w.hq, d)
             ops.attn_decode(q, self.k_cache[li], self.v_cache[li],
                             self.cache_lens, self.scale, out=attn_view,
                             partial_ws=self.buf_attn_ws,
                             nsplit=self.attn_nsplit)
-            self._gemm(self.buf_attn_out, layer["wo"], self.buf_proj)
-            self._maybe_allreduce(self.buf_proj)
-            ops.fused_add_rmsnorm(self.buf_proj, layer["ffn_norm"],
-                                  residual=self.buf_residual,
-                                  eps=cfg.rms_eps, out=self.buf_hidden)
+            # attn-out projection -> residual+norm
+            done = 0
+            if use_slabs:
+                done = self._gemm_slabs(self.buf_attn_out, layer["wo"])
+                if done:
+                    from mlrun_amd import _hip_ops
+
+                    _hip_ops.fused_add_rmsnorm_slab(
+                        self.buf_hidden, self.buf_residual, self.buf_c32,
+                        layer["ffn_norm"], done, cfg.rms_eps)
+            if not done:
+                self._gemm(self.buf_attn_out, layer["wo"], self.buf_proj)
+                self._maybe_allreduce(self.buf_proj)
+                ops.fused_add_rmsnorm(self.buf_proj, layer["ffn_norm"],
+                                      residual=self.buf_residual,
+                                      eps=cfg.rms_eps, out=self.buf_hidden)
             # mlp (fused gate|up projection)
-            self._gemm(self.buf_hidden, layer["wgu"], self.buf_gu)
-            ops.swiglu_fused(self.buf_gu, out=self.buf_act)
-            self._gemm(self.buf_act, layer["wdown"], self.buf_down)
-            self._maybe_allreduce(self.buf_down)
+            done = 0
+            if use_slabs:
+                done = self._gemm_slabs(self.buf_hidden, layer["wgu"])
+                if done:
+                    from mlrun_amd import _hip_ops
+
+                    _hip_ops.swiglu_slab(self.buf_act, self.buf_c32, done)
+            if not done:
+                self._gemm(self.buf_hidden, layer["wgu"], self.buf_gu)
+                ops.swiglu_fused(self.buf_gu, out=self.buf_act)
             next_norm = w.layers[li + 1]["attn_norm"] \
                 if li + 1 < cfg.num_layers else w.final_norm
-            ops.fused_add_rmsnorm(self.buf_down, next_norm,
-                                  residual=self.buf_residual,
-                                  eps=cfg.rms_eps, out=self.buf_hidden)
+            done = 0
+            if use_slabs:
+                done = self._gemm_slabs(self.buf_act, layer["wdown"])
+                if done:
+                    from mlrun_amd import _hip_ops
+
+                    _hip_ops.fused_add_rmsnorm_slab(
+                        self.buf_hidden, self.buf_residual, self.buf_c32,
+                        next_norm, done, cfg.rms_eps)
+            if not done:
+                self._gemm(self.buf_act, layer["wdown"], self.buf_down)
+                self._maybe_allreduce(self.buf_down)
+                ops.fused_add_rmsnorm(self.buf_down, next_norm,
+                                      residual=self.buf_residual,
+                                      eps=cfg.rms_eps, out=self.buf_hidden)
         self._gemm(self.buf_hidden, w.lm_head, self.buf_logits)
         torch.argmax(self.buf_logits, dim=-1, out=self.buf_tokens)
 

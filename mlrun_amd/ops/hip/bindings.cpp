@@ -112,6 +112,67 @@ void skinny_gemm(torch::Tensor out_bf16, torch::Tensor part_f32,
                      (int)variant, stream());
 }
 
+void skinny_gemm_slabs(torch::Tensor part_f32, torch::Tensor a,
+                       torch::Tensor w, int64_t ksplit, int64_t variant) {
+  CHECK_DEV(part_f32); CHECK_CONTIG(part_f32); CHECK_F32(part_f32);
+  CHECK_DEV(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  CHECK_DEV(w); CHECK_CONTIG(w); CHECK_BF16(w);
+  int M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && M <= 16 && K % 32 == 0, "bad shapes");
+  TORCH_CHECK(part_f32.numel() >= ksplit * (int64_t)M * N,
+              "part_f32 scratch too small");
+  launch_skinny_gemm_slabs(part_f32.data_ptr(), a.data_ptr(), w.data_ptr(),
+                           M, N, K, (int)ksplit, (int)variant, stream());
+}
+
+void fused_add_rmsnorm_slab(torch::Tensor out, torch::Tensor residual,
+                            torch::Tensor slabs, torch::Tensor weight,
+                            int64_t ksplit, double eps) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
+  CHECK_DEV(residual); CHECK_CONTIG(residual); CHECK_BF16(residual);
+  CHECK_DEV(slabs); CHECK_CONTIG(slabs); CHECK_F32(slabs);
+  CHECK_DEV(weight); CHECK_CONTIG(weight); CHECK_BF16(weight);
+  int64_t hidden = out.size(-1);
+  int64_t rows = out.numel() / hidden;
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  TORCH_CHECK(slabs.numel() >= ksplit * rows * hidden, "slabs too small");
+  launch_fused_add_rmsnorm_slab(out.data_ptr(), residual.data_ptr(),
+                                slabs.data_ptr(), weight.data_ptr(),
+                                (int)rows, (int)hidden, (float)eps,
+                                (int)ksplit, stream());
+}
+
+void swiglu_slab(torch::Tensor out, torch::Tensor slabs, int64_t ksplit) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
+  CHECK_DEV(slabs); CHECK_CONTIG(slabs); CHECK_F32(slabs);
+  int rows = out.size(0), inter = out.size(1);
+  TORCH_CHECK(inter % 4 == 0, "inter must be a multiple of 4");
+  TORCH_CHECK(slabs.numel() >= (int64_t)ksplit * rows * 2 * inter,
+              "slabs too small");
+  launch_swiglu_slab(out.data_ptr(), slabs.data_ptr(), rows, inter,
+                     (int)ksplit, stream());
+}
+
+void rope_kv_slab(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
+                  torch::Tensor slabs, torch::Tensor positions,
+                  torch::Tensor cos_sin, int64_t hq, int64_t ksplit) {
+  CHECK_DEV(qkv); CHECK_CONTIG(qkv); CHECK_BF16(qkv);
+  CHECK_DEV(kc); CHECK_CONTIG(kc); CHECK_BF16(kc);
+  CHECK_DEV(slabs); CHECK_CONTIG(slabs); CHECK_F32(slabs);
+  CHECK_DEV(positions); CHECK_CONTIG(positions); CHECK_I32(positions);
+  CHECK_DEV(cos_sin); CHECK_CONTIG(cos_sin); CHECK_F32(cos_sin);
+  int B = kc.size(0), Hkv = kc.size(1), Smax = kc.size(2), D = kc.size(3);
+  int qkv_row = qkv.size(1);
+  TORCH_CHECK(qkv.size(0) == B && qkv_row >= (hq + 2 * Hkv) * D,
+              "qkv buffer too small");
+  TORCH_CHECK(slabs.numel() >= (int64_t)ksplit * B * qkv_row,
+              "slabs too small");
+  launch_rope_kv_slab(qkv.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                      slabs.data_ptr(), positions.data_ptr(),
+                      cos_sin.data_ptr(), B, (int)hq, Hkv, Smax, D, qkv_row,
+                      (int)ksplit, stream());
+}
+
 void rope_kv_fused(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
                    torch::Tensor positions, torch::Tensor cos_sin,
                    int64_t hq) {
@@ -240,6 +301,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "decode GEMM C=A@W^T on MFMA (bf16 in/out, split-K f32 slabs)");
   m.def("rope_kv_fused", &rope_kv_fused,
         "fused decode rope(q,k) + KV-cache append");
+  m.def("skinny_gemm_slabs", &skinny_gemm_slabs,
+        "split-K GEMM emitting f32 slabs only");
+  m.def("fused_add_rmsnorm_slab", &fused_add_rmsnorm_slab,
+        "slab-sum + residual add + RMSNorm");
+  m.def("swiglu_slab", &swiglu_slab, "slab-sum + SwiGLU");
+  m.def("rope_kv_slab", &rope_kv_slab,
+        "slab-sum + rope(q,k) + KV append");
   m.def("cast_f32_bf16", &cast_f32_bf16, "f32 -> bf16 cast");
   m.def("attn_decode", &attn_decode, "GQA decode attention (bf16)");
   m.def("kv_append", &kv_append, "append token K/V into cache");

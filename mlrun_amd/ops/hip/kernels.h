@@ -22,6 +22,23 @@ void launch_swiglu_fused(void* out, const void* gu, int rows, int inter,
 void launch_silu_mul(void* out, const void* gate, const void* up,
                      long long n, void* stream);
 
+void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
+                              int M, int N, int K, int ksplit, int variant,
+                              void* stream);
+
+void launch_fused_add_rmsnorm_slab(void* out, void* residual,
+                                   const void* slabs, const void* weight,
+                                   int rows, int hidden, float eps,
+                                   int ksplit, void* stream);
+
+void launch_swiglu_slab(void* out, const void* slabs, int rows, int inter,
+                        int ksplit, void* stream);
+
+void launch_rope_kv_slab(void* qkv, void* Kc, void* Vc, const void* slabs,
+                         const void* positions, const void* cos_sin, int B,
+                         int Hq, int Hkv, int Smax, int D, int qkv_row,
+                         int ksplit, void* stream);
+
 void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                         const void* W, int M, int N, int K, int ksplit,
                         int variant, void* stream);

@@ -427,6 +427,30 @@ __global__ void reduce_cast_kernel(unsigned short* __restrict__ out,
   }
 }
 
+// emit split-K slabs only (consumer kernel folds them)
+void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
+                              int M, int N, int K, int ksplit, int variant,
+                              void* stream) {
+  const int nblocks = variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
+  if (variant == 2)
+    hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 16>),
+                       dim3(nblocks, ksplit), dim3(256), 0,
+                       (hipStream_t)stream, part_f32,
+                       (const unsigned short*)A, (const unsigned short*)W,
+                       M, N, K, ksplit);
+  else if (variant == 1)
+    hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 8>),
+                       dim3(nblocks, ksplit), dim3(256), 0,
+                       (hipStream_t)stream, part_f32,
+                       (const unsigned short*)A, (const unsigned short*)W,
+                       M, N, K, ksplit);
+  else
+    hipLaunchKernelGGL(skinny_gemm_kernel<true>, dim3(nblocks, ksplit),
+                       dim3(256), 0, (hipStream_t)stream, part_f32,
+                       (const unsigned short*)A, (const unsigned short*)W,
+                       M, N, K, ksplit);
+}
+
 void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                         const void* W, int M, int N, int K, int ksplit,
                         int variant, void* stream) {
@@ -518,6 +542,181 @@ void launch_zero_f32(void* p, long long n, void* stream) {
   if (blocks > 2048) blocks = 2048;
   hipLaunchKernelGGL(zero_f32_kernel, dim3((int)blocks), dim3(256), 0,
                      (hipStream_t)stream, (float*)p, n);
+}
+
+// ---------------------------------------------------------------------
+// Slab-consumer fusions: when a decode GEMM runs split-K, its f32
+// partial slabs are folded INSIDE the consuming kernel instead of a
+// separate reduce_cast pass — one less launch and one less bf16
+// round-trip per projection (profile: reduce_cast was ~8% of decode).
+// slabs layout: [ksplit, rows, cols] f32.
+// ---------------------------------------------------------------------
+
+DEV float slab_sum(const float* __restrict__ slabs, size_t idx,
+                   size_t slab_stride, int ksplit) {
+  float sum = slabs[idx];
+  for (int s = 1; s < ksplit; ++s) sum += slabs[idx + s * slab_stride];
+  return sum;
+}
+
+// residual <- sum(slabs) + residual; out <- rmsnorm(residual) * weight
+__global__ void fused_add_rmsnorm_slab_kernel(
+    unsigned short* __restrict__ out, unsigned short* __restrict__ residual,
+    const float* __restrict__ slabs,
+    const unsigned short* __restrict__ weight, int hidden, float eps,
+    int ksplit) {
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int nthreads = blockDim.x;
+  const size_t rowbase = (size_t)row * hidden;
+  const size_t slab_stride = (size_t)gridDim.x * hidden;
+  unsigned short* rr = residual + rowbase;
+  unsigned short* outr = out + rowbase;
+
+  float sumsq = 0.f;
+  for (int i = tid * 4; i < hidden; i += nthreads * 4) {
+    float z[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      z[j] = slab_sum(slabs, rowbase + i + j, slab_stride, ksplit);
+    unsigned short rv[4];
+    *reinterpret_cast<uint64_t*>(rv) =
+        *reinterpret_cast<const uint64_t*>(rr + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      z[j] += bf2f(rv[j]);
+      rv[j] = f2bf(z[j]);
+      sumsq += z[j] * z[j];
+    }
+    *reinterpret_cast<uint64_t*>(rr + i) =
+        *reinterpret_cast<const uint64_t*>(rv);
+  }
+  __shared__ float red[32];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    sumsq += __shfl_down(sumsq, off, 64);
+  if ((tid & 63) == 0) red[tid >> 6] = sumsq;
+  __syncthreads();
+  if (tid < (nthreads >> 6)) sumsq = red[tid];
+  else sumsq = 0.f;
+  if (tid < 64) {
+#pragma unroll
+    for (int off = 2; off > 0; off >>= 1)
+      sumsq += __shfl_down(sumsq, off, 64);
+  }
+  if (tid == 0) red[0] = sumsq;
+  __syncthreads();
+  const float inv = rsqrtf(red[0] / hidden + eps);
+  for (int i = tid * 8; i < hidden; i += nthreads * 8) {
+    ushort8 zv = *reinterpret_cast<const ushort8*>(rr + i);
+    ushort8 wv = *reinterpret_cast<const ushort8*>(weight + i);
+    ushort8 ov;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      ov.v[j] = f2bf(bf2f(zv.v[j]) * inv * bf2f(wv.v[j]));
+    *reinterpret_cast<ushort8*>(outr + i) = ov;
+  }
+}
+
+void launch_fused_add_rmsnorm_slab(void* out, void* residual,
+                                   const void* slabs, const void* weight,
+                                   int rows, int hidden, float eps,
+                                   int ksplit, void* stream) {
+  int threads = hidden >= 2048 ? 256 : 64;
+  hipLaunchKernelGGL(fused_add_rmsnorm_slab_kernel, dim3(rows),
+                     dim3(threads), 0, (hipStream_t)stream,
+                     (unsigned short*)out, (unsigned short*)residual,
+                     (const float*)slabs, (const unsigned short*)weight,
+                     hidden, eps, ksplit);
+}
+
+// out[r, c] = silu(sum gate_slab) * sum(up_slab); slabs [ks, rows, 2I]
+__global__ void swiglu_slab_kernel(unsigned short* __restrict__ out,
+                                   const float* __restrict__ slabs,
+                                   int rows, int inter, int ksplit) {
+  const size_t slab_stride = (size_t)rows * 2 * inter;
+  const long long total = (long long)rows * inter / 4;
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    const long long flat = i * 4;
+    const int r = (int)(flat / inter);
+    const int c = (int)(flat % inter);
+    const size_t base = (size_t)r * 2 * inter + c;
+    unsigned short o[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float g = slab_sum(slabs, base + j, slab_stride, ksplit);
+      const float u = slab_sum(slabs, base + inter + j, slab_stride,
+                               ksplit);
+      const float s = g / (1.f + __expf(-g));
+      o[j] = f2bf(s * u);
+    }
+    *reinterpret_cast<uint64_t*>(out + (size_t)r * inter + c) =
+        *reinterpret_cast<const uint64_t*>(o);
+  }
+}
+
+void launch_swiglu_slab(void* out, const void* slabs, int rows, int inter,
+                        int ksplit, void* stream) {
+  long long blocks = ((long long)rows * inter / 4 + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(swiglu_slab_kernel, dim3((int)blocks), dim3(256), 0,
+                     (hipStream_t)stream, (unsigned short*)out,
+                     (const float*)slabs, rows, inter, ksplit);
+}
+
+// qkv slab -> rope(q,k) + KV append + q written to the bf16 qkv buffer
+// slabs [ks, B, qkv_row]; qkv_row = (Hq + 2*Hkv) * D
+__global__ void rope_kv_slab_kernel(
+    unsigned short* __restrict__ qkv, unsigned short* __restrict__ Kc,
+    unsigned short* __restrict__ Vc, const float* __restrict__ slabs,
+    const int* __restrict__ positions, const float* __restrict__ cos_sin,
+    int Hq, int Hkv, int Smax, int D, int qkv_row, int ksplit) {
+  const int b = blockIdx.x;
+  const int y = blockIdx.y;
+  const int i = threadIdx.x;  // 0..D/2-1
+  const int half = D >> 1;
+  const int pos = positions[b];
+  const size_t slab_stride = (size_t)gridDim.x * qkv_row;
+  const size_t base = (size_t)b * qkv_row + (size_t)y * D;
+  const float lo_f = slab_sum(slabs, base + i, slab_stride, ksplit);
+  const float hi_f = slab_sum(slabs, base + i + half, slab_stride, ksplit);
+  if (y < Hq + Hkv) {
+    const float c = cos_sin[((size_t)pos * half + i) * 2];
+    const float s = cos_sin[((size_t)pos * half + i) * 2 + 1];
+    const unsigned short lo = f2bf(lo_f * c - hi_f * s);
+    const unsigned short hi = f2bf(hi_f * c + lo_f * s);
+    if (y < Hq) {  // q: written back for the attention kernel
+      unsigned short* qrow = qkv + (size_t)b * qkv_row + (size_t)y * D;
+      qrow[i] = lo;
+      qrow[i + half] = hi;
+    } else {       // k: straight into the cache
+      const int kvh = y - Hq;
+      unsigned short* dst =
+          Kc + (((size_t)b * Hkv + kvh) * Smax + pos) * D;
+      dst[i] = lo;
+      dst[i + half] = hi;
+    }
+  } else {         // v: cache only
+    const int kvh = y - Hq - Hkv;
+    unsigned short* dst = Vc + (((size_t)b * Hkv + kvh) * Smax + pos) * D;
+    dst[i] = f2bf(lo_f);
+    dst[i + half] = f2bf(hi_f);
+  }
+}
+
+void launch_rope_kv_slab(void* qkv, void* Kc, void* Vc, const void* slabs,
+                         const void* positions, const void* cos_sin, int B,
+                         int Hq, int Hkv, int Smax, int D, int qkv_row,
+                         int ksplit, void* stream) {
+  hipLaunchKernelGGL(rope_kv_slab_kernel, dim3(B, Hq + 2 * Hkv),
+                     dim3(D / 2), 0, (hipStream_t)stream,
+                     (unsigned short*)qkv, (unsigned short*)Kc,
+                     (unsigned short*)Vc, (const float*)slabs,
+                     (const int*)positions, (const float*)cos_sin, Hq, Hkv,
+                     Smax, D, qkv_row, ksplit);
 }
 
 // ---------------------------------------------------------------------
