@@ -234,9 +234,16 @@ class LlamaDecodeEngine:
         self.cache_lens.add_(1)
         positions = self.buf_positions
         # slab fusion (GPU, TP=1): split-K consumers fold the f32 slabs
-        # directly (one less launch + bf16 round-trip per projection);
-        # under TP the all-reduce needs the materialized bf16 output
-        use_slabs = self.on_gpu and self.tp_size == 1
+        # directly.  MEASURED: fusing into the 16-row norm kernels is a
+        # net loss (consumer grid = B rows is parallelism-starved vs the
+        # 2048-block reduce_cast), so only the rope/KV consumer (grid
+        # B x heads) fuses by default; override with MLRUN_SLAB_MODE.
+        import os as _os
+
+        mode = _os.environ.get("MLRUN_SLAB_MODE", "rope")
+        on = self.on_gpu and self.tp_size == 1 and mode != "none"
+        use_slabs = on          # qkv -> rope_kv_slab
+        use_norm_slabs = on and mode == "all" 
         for li, layer in enumerate(w.layers):
             # qkv projection -> rope -> caches
             done = 0
@@ -261,7 +268,7 @@ class LlamaDecodeEngine:
                             nsplit=self.attn_nsplit)
             # attn-out projection -> residual+norm
             done = 0
-            if use_slabs:
+            if use_norm_slabs:
                 done = self._gemm_slabs(self.buf_attn_out, layer["wo"])
                 if done:
                     from mlrun_amd import _hip_ops
@@ -277,7 +284,7 @@ class LlamaDecodeEngine:
                                       eps=cfg.rms_eps, out=self.buf_hidden)
             # mlp (fused gate|up projection)
             done = 0
-            if use_slabs:
+            if use_norm_slabs:
                 done = self._gemm_slabs(self.buf_hidden, layer["wgu"])
                 if done:
                     from mlrun_amd import _hip_ops
@@ -289,7 +296,7 @@ class LlamaDecodeEngine:
             next_norm = w.layers[li + 1]["attn_norm"] \
                 if li + 1 < cfg.num_layers else w.final_norm
             done = 0
-            if use_slabs:
+            if use_norm_slabs:
                 done = self._gemm_slabs(self.buf_act, layer["wdown"])
                 if done:
                     from mlrun_amd import _hip_ops
