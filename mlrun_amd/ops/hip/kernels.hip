@@ -262,37 +262,57 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     if (kend > K) kend = K;
   }
 
+  // K-loop notes (measured via .s dumps):
+  //  - any branch or pragma-driven unroll in the loop kept it ROLLED
+  //    with a vmcnt(0) drain per iteration -> latency-serialized;
+  //    the manual 8x unroll below issues all 24 loads of a block
+  //    before the first MFMA waits, so a wave keeps ~384B in flight
+  //  - out-of-range rows are CLAMPED, not masked: their products land
+  //    only in C cells (m>=M / n>=N) the epilogue never writes
   const int arow = lane & 15;
   const int kb = (lane >> 4) * 8;
-  const bool a_valid = arow < M;
   const int brow0 = n0 + (lane & 15);
   const int brow1 = brow0 + 16;
-  const bool b0_valid = brow0 < N;
-  const bool b1_valid = brow1 < N;
 
   f32x4v acc0 = {0.f, 0.f, 0.f, 0.f};
   f32x4v acc1 = {0.f, 0.f, 0.f, 0.f};
-  const short8v zero8 = {0, 0, 0, 0, 0, 0, 0, 0};
 
-  const unsigned short* aptr = A + (size_t)arow * K + kb;
-  const unsigned short* bptr0 = W + (size_t)brow0 * K + kb;
-  const unsigned short* bptr1 = W + (size_t)brow1 * K + kb;
+  const unsigned short* aptr = A + (size_t)min(arow, M - 1) * K + kb;
+  const unsigned short* bptr0 =
+      W + (size_t)min(brow0, N - 1) * K + kb;
+  const unsigned short* bptr1 =
+      W + (size_t)min(brow1, N - 1) * K + kb;
 
-#pragma unroll 4
-  for (int k = kbegin; k < kend; k += 32) {
-    short8v af = a_valid
-        ? *reinterpret_cast<const short8v*>(aptr + k) : zero8;
-    short8v bf0 = b0_valid
-        ? *reinterpret_cast<const short8v*>(bptr0 + k) : zero8;
-    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
-    if (b1_valid) {
-      short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+  int k = kbegin;
+  const int kend8 = kbegin + ((kend - kbegin) & ~255);
+  for (; k < kend8; k += 256) {
+    short8v af[8], bf0[8], bf1[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      af[u] = *reinterpret_cast<const short8v*>(aptr + k + u * 32);
+      bf0[u] = *reinterpret_cast<const short8v*>(bptr0 + k + u * 32);
+      bf1[u] = *reinterpret_cast<const short8v*>(bptr1 + k + u * 32);
     }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf0[u], acc0,
+                                                     0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf1[u], acc1,
+                                                     0, 0, 0);
+    }
+  }
+  for (; k < kend; k += 32) {
+    short8v af = *reinterpret_cast<const short8v*>(aptr + k);
+    short8v bf0 = *reinterpret_cast<const short8v*>(bptr0 + k);
+    short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
   }
 
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
+  const bool b0_valid = brow0 < N;
+  const bool b1_valid = brow1 < N;
   if (SPLIT) {
     float* part = (float*)out + (size_t)blockIdx.y * M * N;
 #pragma unroll
@@ -341,32 +361,51 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
   int kend = kbegin + per_wave;
   if (kend > cend) kend = cend;
 
+  // K-loop notes (measured via .s dumps):
+  //  - any branch or pragma-driven unroll in the loop kept it ROLLED
+  //    with a vmcnt(0) drain per iteration -> latency-serialized;
+  //    the manual 8x unroll below issues all 24 loads of a block
+  //    before the first MFMA waits, so a wave keeps ~384B in flight
+  //  - out-of-range rows are CLAMPED, not masked: their products land
+  //    only in C cells (m>=M / n>=N) the epilogue never writes
   const int arow = lane & 15;
   const int kb = (lane >> 4) * 8;
-  const bool a_valid = arow < M;
   const int brow0 = n0 + (lane & 15);
   const int brow1 = brow0 + 16;
-  const bool b0_valid = brow0 < N;
-  const bool b1_valid = brow1 < N;
 
   f32x4v acc0 = {0.f, 0.f, 0.f, 0.f};
   f32x4v acc1 = {0.f, 0.f, 0.f, 0.f};
-  const short8v zero8 = {0, 0, 0, 0, 0, 0, 0, 0};
-  const unsigned short* aptr = A + (size_t)arow * K + kb;
-  const unsigned short* bptr0 = W + (size_t)brow0 * K + kb;
-  const unsigned short* bptr1 = W + (size_t)brow1 * K + kb;
 
-#pragma unroll UNROLL
-  for (int k = kbegin; k < kend; k += 32) {
-    short8v af = a_valid
-        ? *reinterpret_cast<const short8v*>(aptr + k) : zero8;
-    short8v bf0 = b0_valid
-        ? *reinterpret_cast<const short8v*>(bptr0 + k) : zero8;
-    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
-    if (b1_valid) {
-      short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+  const unsigned short* aptr = A + (size_t)min(arow, M - 1) * K + kb;
+  const unsigned short* bptr0 =
+      W + (size_t)min(brow0, N - 1) * K + kb;
+  const unsigned short* bptr1 =
+      W + (size_t)min(brow1, N - 1) * K + kb;
+
+  int k = kbegin;
+  const int kend8 = kbegin + ((kend - kbegin) & ~255);
+  for (; k < kend8; k += 256) {
+    short8v af[8], bf0[8], bf1[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      af[u] = *reinterpret_cast<const short8v*>(aptr + k + u * 32);
+      bf0[u] = *reinterpret_cast<const short8v*>(bptr0 + k + u * 32);
+      bf1[u] = *reinterpret_cast<const short8v*>(bptr1 + k + u * 32);
     }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf0[u], acc0,
+                                                     0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf1[u], acc1,
+                                                     0, 0, 0);
+    }
+  }
+  for (; k < kend; k += 32) {
+    short8v af = *reinterpret_cast<const short8v*>(aptr + k);
+    short8v bf0 = *reinterpret_cast<const short8v*>(bptr0 + k);
+    short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
   }
 
   // combine the 4 waves' partials through LDS
