@@ -159,11 +159,12 @@ _EMPTY_F32 = None
 # measured on MI355X (scripts/bench_gemm.py sweep, profiles/): best
 # (ksplit, variant) per llama projection shape; variant 1/2 = wave-split
 _GEMM_PLAN_TABLE = {
-    (6144, 4096): (4, 1),     # qkv      3.06 TB/s
-    (4096, 4096): (4, 1),     # wo       2.81 TB/s
-    (28672, 4096): (1, 2),    # gate|up  4.82 TB/s (unroll16)
-    (4096, 14336): (8, 1),    # down     3.90 TB/s
-    (128256, 4096): (1, 0),   # lm_head  4.16 TB/s
+    # re-measured after the manual-unroll K-loop rewrite
+    (6144, 4096): (1, 2),     # qkv      3.49 TB/s
+    (4096, 4096): (2, 2),     # wo       3.17 TB/s
+    (28672, 4096): (1, 2),    # gate|up  5.09 TB/s
+    (4096, 14336): (4, 2),    # down     4.64 TB/s
+    (128256, 4096): (1, 0),   # lm_head  4.26 TB/s
 }
 
 
