@@ -509,9 +509,16 @@ class FlowStep(BaseStep):
     def _run_from(self, step: BaseStep, event):
         """The sync hot loop: walk the next-chain from a step
         (parity: reference states.py:1293-1320)."""
+        from ..utils import tracing
+
+        trace = tracing.is_enabled()
         while step is not None:
             try:
-                event = step.run(event)
+                if trace:
+                    with tracing.trace_step(step.fullname):
+                        event = step.run(event)
+                else:
+                    event = step.run(event)
             except Exception as exc:
                 event = self._handle_error(step, event, exc)
                 if event is None or getattr(event, "error", None):

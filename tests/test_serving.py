@@ -296,3 +296,84 @@ class TestMonitoringIntegration:
         stats = processor.endpoint_stats("tracked")
         assert stats["300"]["count"] == 5
         assert stats["300"]["avg_latency_ms"] >= 0
+
+
+class TestAPIGateway:
+    def test_canary_and_auth(self):
+        from mlrun_amd.runtimes.api_gateway import APIGateway
+
+        fn_a = _serving_fn()
+        fn_a.add_model("m", class_name=ConstModel, value=1)
+        addr_a = fn_a.deploy()
+        fn_b = _serving_fn()
+        fn_b.add_model("m", class_name=ConstModel, value=2)
+        addr_b = fn_b.deploy()
+        gateway = APIGateway(name="gw").with_canary(
+            [addr_a, addr_b], [50, 50]).with_basic_auth("user", "pw")
+        gateway.deploy()
+        try:
+            # unauthorized without credentials
+            status, _ = gateway.invoke("/v2/models/m/infer",
+                                       body={"inputs": [0]})
+            assert status == 401
+            seen = set()
+            for _ in range(20):
+                status, out = gateway.invoke(
+                    "/v2/models/m/infer", body={"inputs": [0]},
+                    credentials=("user", "pw"))
+                assert status == 200
+                seen.add(out["outputs"][0])
+            assert seen == {1, 2}  # both canary legs took traffic
+        finally:
+            gateway.stop()
+            fn_a.stop()
+            fn_b.stop()
+
+
+class TestV1Serving:
+    def test_v1_protocol(self):
+        from mlrun_amd.serving.v1_serving import MLModelServer
+
+        class V1Echo(MLModelServer):
+            def load(self):
+                self.model = True
+
+            def predict(self, request):
+                return [x + 1 for x in request["instances"]]
+
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(V1Echo(name="v1m"), name="model")
+        server = fn.to_mock_server()
+        resp = server.test("/predict", body={"instances": [1, 2]})
+        assert resp["predictions"] == [2, 3]
+        assert resp["model_name"] == "v1m"
+
+
+class TestTracing:
+    def test_spans_collected(self):
+        from mlrun_amd.utils import tracing
+
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(handler=double_handler, name="d1") \
+            .to(handler=add_one, name="a1")
+        server = fn.to_mock_server()
+        tracing.enable(True)
+        try:
+            with tracing.collect() as spans:
+                server.test("/", body={"x": 1})
+            names = [s["name"] for s in spans]
+            assert "step:d1" in names and "step:a1" in names
+            assert all(s["ms"] >= 0 for s in spans)
+        finally:
+            tracing.enable(False)
+
+    def test_disabled_no_overhead_path(self):
+        from mlrun_amd.utils import tracing
+
+        assert not tracing.is_enabled()
+        with tracing.collect() as spans:
+            with tracing.span("x"):
+                pass
+        assert spans == []
