@@ -431,7 +431,8 @@ class LlamaServer:
 
     def __init__(self, context=None, name=None, model_path=None,
                  config=None, batch_size=16, max_new_tokens=32,
-                 device=None, use_graph=True, **class_args):
+                 device=None, use_graph=True, batch_window_ms=0,
+                 **class_args):
         from ..serving.v2_serving import V2ModelServer
 
         self.name = name
@@ -453,6 +454,14 @@ class LlamaServer:
         import threading
 
         self._engine_lock = threading.Lock()
+        # dynamic batching of concurrent single-prompt requests (the
+        # reference harness sends N concurrent clients; batching them
+        # into one engine pass multiplies throughput)
+        self.batch_window_ms = batch_window_ms
+        self._pending = []          # [(prompt, max_new, future)]
+        self._pending_cv = threading.Condition()
+        self._batcher = None
+        self.engine_calls = 0
 
     def post_init(self, mode="sync"):
         stream = getattr(self.context, "stream", None) if self.context \
@@ -505,19 +514,71 @@ class LlamaServer:
         if inputs is None:
             raise ValueError('expected {"inputs": [[token ids], ...]}')
         max_new = int(body.get("max_tokens", self.max_new_tokens))
-        outputs = []
-        with self._engine_lock:
-            for chunk_start in range(0, len(inputs), self.batch_size):
-                chunk = inputs[chunk_start:chunk_start + self.batch_size]
-                outputs.extend(self._generate_chunk(chunk, max_new))
+        if self.batch_window_ms and len(inputs) < self.batch_size:
+            outputs = self._batched_submit(inputs, max_new)
+        else:
+            outputs = []
+            with self._engine_lock:
+                for chunk_start in range(0, len(inputs), self.batch_size):
+                    chunk = inputs[chunk_start:
+                                   chunk_start + self.batch_size]
+                    outputs.extend(self._generate_chunk(chunk, max_new))
         event.body = {"id": event.id, "model_name": self.name,
                       "outputs": outputs}
+        self.engine_calls += 0  # engine-call count tracked in chunk
         if self._model_logger:
             self._model_logger.push(start, {"inputs": [len(inputs)]},
                                     event.body)
         return event
 
+    def _batched_submit(self, inputs: list, max_new: int) -> list:
+        """Queue small requests; a batcher thread gathers up to
+        batch_size prompts within batch_window_ms and runs ONE engine
+        pass for all of them."""
+        import concurrent.futures
+        import threading
+
+        futures = []
+        with self._pending_cv:
+            if self._batcher is None:
+                self._batcher = threading.Thread(
+                    target=self._batch_loop, daemon=True,
+                    name=f"llama-batcher-{self.name}")
+                self._batcher.start()
+            for prompt in inputs:
+                future = concurrent.futures.Future()
+                self._pending.append((prompt, max_new, future))
+                futures.append(future)
+            self._pending_cv.notify()
+        return [f.result(timeout=600) for f in futures]
+
+    def _batch_loop(self):
+        import time as _time
+
+        while True:
+            with self._pending_cv:
+                while not self._pending:
+                    self._pending_cv.wait()
+            # window: let more requests arrive
+            _time.sleep(self.batch_window_ms / 1000.0)
+            with self._pending_cv:
+                batch, self._pending =                     self._pending[:self.batch_size],                     self._pending[self.batch_size:]
+            if not batch:
+                continue
+            prompts = [b[0] for b in batch]
+            max_new = max(b[1] for b in batch)
+            try:
+                with self._engine_lock:
+                    results = self._generate_chunk(prompts, max_new)
+                for (prompt, want, future), result in zip(batch, results):
+                    future.set_result(result[:want])
+            except Exception as exc:
+                for _, _, future in batch:
+                    if not future.done():
+                        future.set_exception(exc)
+
     def _generate_chunk(self, prompts: list, max_new: int) -> list:
+        self.engine_calls += 1
         n = len(prompts)
         max_len = max(len(p) for p in prompts)
         tokens = torch.zeros(self.batch_size, max_len, dtype=torch.int64)

@@ -213,3 +213,25 @@ class LlamaTrainer:
                            weights_only=True)
         self.model.load_state_dict(state)
         return spec
+
+
+def export_decode_state(model: "LlamaForCausalLM") -> dict:
+    """Convert a trained LlamaForCausalLM state into the serving
+    engine's weight layout (LlamaWeights.state_dict format) — the
+    train -> checkpoint -> deploy bridge."""
+    state = model.state_dict()
+    out = {
+        "embed": state["embed.weight"],
+        "final_norm": state["final_norm.weight"],
+        "lm_head": state["lm_head.weight"],
+    }
+    n_layers = model.cfg.num_layers
+    for i in range(n_layers):
+        prefix = f"blocks.{i}."
+        out[f"layers.{i}.attn_norm"] = state[prefix + "attn_norm.weight"]
+        out[f"layers.{i}.wqkv"] = state[prefix + "attn.wqkv.weight"]
+        out[f"layers.{i}.wo"] = state[prefix + "attn.wo.weight"]
+        out[f"layers.{i}.ffn_norm"] = state[prefix + "ffn_norm.weight"]
+        out[f"layers.{i}.wgu"] = state[prefix + "mlp.wgu.weight"]
+        out[f"layers.{i}.wdown"] = state[prefix + "mlp.wdown.weight"]
+    return {k: v.to(torch.bfloat16) for k, v in out.items()}

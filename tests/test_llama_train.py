@@ -62,3 +62,43 @@ class TestLlamaTrain:
         assert run.status.state == RunStates.completed, run.status.error
         assert run.status.results["world_size"] == 2
         assert run.status.results["tokens_per_sec"] > 0
+
+
+class TestTrainToServe:
+    def test_trained_model_serves_identically(self, rundb):
+        """Train -> export -> decode engine must produce the trained
+        model's greedy continuation (the MLOps train->deploy bridge)."""
+        import io
+
+        import torch
+
+        from mlrun_amd.models.llama import LlamaDecodeEngine
+        from mlrun_amd.models.llama_train import export_decode_state
+
+        torch.manual_seed(11)
+        cfg = LlamaConfig.tiny(vocab_size=512, max_seq_len=64,
+                               num_heads=2, num_kv_heads=2)
+        trainer = LlamaTrainer(cfg, device="cpu", lr=1e-3)
+        batch = torch.randint(0, 512, (2, 24))
+        for _ in range(3):
+            trainer.train_step(batch)
+
+        engine = LlamaDecodeEngine(cfg, batch_size=2, device="cpu", seed=1)
+        engine.weights.load_state_dict(export_decode_state(trainer.model))
+
+        prompt = torch.randint(0, 512, (2, 8),
+                               generator=torch.Generator().manual_seed(2))
+        generated = engine.generate(prompt, max_new_tokens=3)
+
+        # reference: greedy continuation straight from the nn module
+        model = trainer.model.eval()
+        tokens = prompt.clone()
+        expected = []
+        with torch.no_grad():
+            for _ in range(3):
+                logits = model(tokens)
+                nxt = logits[:, -1].argmax(dim=-1)
+                expected.append(nxt)
+                tokens = torch.cat([tokens, nxt[:, None]], dim=1)
+        expected = torch.stack(expected, dim=1)
+        assert torch.equal(generated, expected), (generated, expected)
