@@ -186,7 +186,7 @@ def skinny_gemm(a: torch.Tensor, w: torch.Tensor,
                 out: torch.Tensor = None,
                 c_f32: torch.Tensor = None,
                 ksplit: int = None, variant: int = None) -> torch.Tensor:
-    """C[M,N] = A[M,K] @ W[N,K]^T for decode batches (M <= 16), bf16.
+    """C[M,N] = A[M,K] @ W[N,K]^T for decode batches (M <= 32), bf16.
 
     On GPU: MFMA kernel; split-K writes per-chunk f32 slabs into c_f32
     and a reduce kernel folds them to bf16 (deterministic, no atomics).

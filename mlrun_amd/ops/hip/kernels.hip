@@ -246,7 +246,7 @@ void launch_silu_mul(void* out, const void* gate, const void* up,
 // SPLIT=true:  blockIdx.y writes its f32 partial slab part[y][M][N]
 //              (plain stores — no atomics, no pre-zero, deterministic);
 //              reduce_cast_kernel folds the slabs to bf16.
-template <bool SPLIT>
+template <bool SPLIT, int MT>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     void* __restrict__ out, const unsigned short* __restrict__ A,
     const unsigned short* __restrict__ W, int M, int N, int K, int ksplit) {
@@ -265,48 +265,69 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   // K-loop notes (measured via .s dumps):
   //  - any branch or pragma-driven unroll in the loop kept it ROLLED
   //    with a vmcnt(0) drain per iteration -> latency-serialized;
-  //    the manual 8x unroll below issues all 24 loads of a block
-  //    before the first MFMA waits, so a wave keeps ~384B in flight
+  //    the manual unrolled block issues all its loads before the
+  //    first MFMA waits, so a wave keeps ~384B in flight
   //  - out-of-range rows are CLAMPED, not masked: their products land
   //    only in C cells (m>=M / n>=N) the epilogue never writes
+  //  - MT in {1,2} A-row tiles: M<=16 or 17..32 (decode batch 32
+  //    doubles served requests per weight pass at equal HBM traffic)
   const int arow = lane & 15;
   const int kb = (lane >> 4) * 8;
   const int brow0 = n0 + (lane & 15);
   const int brow1 = brow0 + 16;
 
-  f32x4v acc0 = {0.f, 0.f, 0.f, 0.f};
-  f32x4v acc1 = {0.f, 0.f, 0.f, 0.f};
+  f32x4v acc0[MT], acc1[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t) {
+    acc0[t] = {0.f, 0.f, 0.f, 0.f};
+    acc1[t] = {0.f, 0.f, 0.f, 0.f};
+  }
 
-  const unsigned short* aptr = A + (size_t)min(arow, M - 1) * K + kb;
+  const unsigned short* aptr[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t)
+    aptr[t] = A + (size_t)min(arow + 16 * t, M - 1) * K + kb;
   const unsigned short* bptr0 =
       W + (size_t)min(brow0, N - 1) * K + kb;
   const unsigned short* bptr1 =
       W + (size_t)min(brow1, N - 1) * K + kb;
 
+  constexpr int UNR = (MT == 1) ? 8 : 4;
   int k = kbegin;
-  const int kend8 = kbegin + ((kend - kbegin) & ~255);
-  for (; k < kend8; k += 256) {
-    short8v af[8], bf0[8], bf1[8];
+  const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
+  for (; k < kend8; k += UNR * 32) {
+    short8v af[UNR][MT], bf0[UNR], bf1[UNR];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      af[u] = *reinterpret_cast<const short8v*>(aptr + k + u * 32);
+    for (int u = 0; u < UNR; ++u) {
+#pragma unroll
+      for (int t = 0; t < MT; ++t)
+        af[u][t] =
+            *reinterpret_cast<const short8v*>(aptr[t] + k + u * 32);
       bf0[u] = *reinterpret_cast<const short8v*>(bptr0 + k + u * 32);
       bf1[u] = *reinterpret_cast<const short8v*>(bptr1 + k + u * 32);
     }
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf0[u], acc0,
-                                                     0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf1[u], acc1,
-                                                     0, 0, 0);
+    for (int u = 0; u < UNR; ++u) {
+#pragma unroll
+      for (int t = 0; t < MT; ++t) {
+        acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[u][t], bf0[u], acc0[t], 0, 0, 0);
+        acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[u][t], bf1[u], acc1[t], 0, 0, 0);
+      }
     }
   }
   for (; k < kend; k += 32) {
-    short8v af = *reinterpret_cast<const short8v*>(aptr + k);
     short8v bf0 = *reinterpret_cast<const short8v*>(bptr0 + k);
     short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
-    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
-    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+#pragma unroll
+    for (int t = 0; t < MT; ++t) {
+      short8v af = *reinterpret_cast<const short8v*>(aptr[t] + k);
+      acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0[t],
+                                                        0, 0, 0);
+      acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1[t],
+                                                        0, 0, 0);
+    }
   }
 
   const int crow_base = (lane >> 4) * 4;
@@ -316,21 +337,26 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   if (SPLIT) {
     float* part = (float*)out + (size_t)blockIdx.y * M * N;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = crow_base + r;
-      if (m >= M) continue;
-      if (b0_valid) part[(size_t)m * N + n0 + ccol] = acc0[r];
-      if (b1_valid) part[(size_t)m * N + n0 + 16 + ccol] = acc1[r];
-    }
+    for (int t = 0; t < MT; ++t)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = 16 * t + crow_base + r;
+        if (m >= M) continue;
+        if (b0_valid) part[(size_t)m * N + n0 + ccol] = acc0[t][r];
+        if (b1_valid) part[(size_t)m * N + n0 + 16 + ccol] = acc1[t][r];
+      }
   } else {
     unsigned short* dst = (unsigned short*)out;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = crow_base + r;
-      if (m >= M) continue;
-      if (b0_valid) dst[(size_t)m * N + n0 + ccol] = f2bf(acc0[r]);
-      if (b1_valid) dst[(size_t)m * N + n0 + 16 + ccol] = f2bf(acc1[r]);
-    }
+    for (int t = 0; t < MT; ++t)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = 16 * t + crow_base + r;
+        if (m >= M) continue;
+        if (b0_valid) dst[(size_t)m * N + n0 + ccol] = f2bf(acc0[t][r]);
+        if (b1_valid)
+          dst[(size_t)m * N + n0 + 16 + ccol] = f2bf(acc1[t][r]);
+      }
   }
 }
 
@@ -339,7 +365,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 // LDS.  4x the workgroup count of skinny_gemm_kernel at the same
 // split-K slab traffic -> 4x the waves/SIMD for latency hiding
 // (profile: N=4096 projections were latency-bound at 2 waves/SIMD).
-template <bool SPLIT, int UNROLL>
+template <bool SPLIT, int MT>
 __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
     void* __restrict__ out, const unsigned short* __restrict__ A,
     const unsigned short* __restrict__ W, int M, int N, int K, int ksplit) {
@@ -364,65 +390,88 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
   // K-loop notes (measured via .s dumps):
   //  - any branch or pragma-driven unroll in the loop kept it ROLLED
   //    with a vmcnt(0) drain per iteration -> latency-serialized;
-  //    the manual 8x unroll below issues all 24 loads of a block
-  //    before the first MFMA waits, so a wave keeps ~384B in flight
+  //    the manual unrolled block issues all its loads before the
+  //    first MFMA waits, so a wave keeps ~384B in flight
   //  - out-of-range rows are CLAMPED, not masked: their products land
   //    only in C cells (m>=M / n>=N) the epilogue never writes
+  //  - MT in {1,2} A-row tiles: M<=16 or 17..32 (decode batch 32
+  //    doubles served requests per weight pass at equal HBM traffic)
   const int arow = lane & 15;
   const int kb = (lane >> 4) * 8;
   const int brow0 = n0 + (lane & 15);
   const int brow1 = brow0 + 16;
 
-  f32x4v acc0 = {0.f, 0.f, 0.f, 0.f};
-  f32x4v acc1 = {0.f, 0.f, 0.f, 0.f};
+  f32x4v acc0[MT], acc1[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t) {
+    acc0[t] = {0.f, 0.f, 0.f, 0.f};
+    acc1[t] = {0.f, 0.f, 0.f, 0.f};
+  }
 
-  const unsigned short* aptr = A + (size_t)min(arow, M - 1) * K + kb;
+  const unsigned short* aptr[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t)
+    aptr[t] = A + (size_t)min(arow + 16 * t, M - 1) * K + kb;
   const unsigned short* bptr0 =
       W + (size_t)min(brow0, N - 1) * K + kb;
   const unsigned short* bptr1 =
       W + (size_t)min(brow1, N - 1) * K + kb;
 
+  constexpr int UNR = (MT == 1) ? 8 : 4;
   int k = kbegin;
-  const int kend8 = kbegin + ((kend - kbegin) & ~255);
-  for (; k < kend8; k += 256) {
-    short8v af[8], bf0[8], bf1[8];
+  const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
+  for (; k < kend8; k += UNR * 32) {
+    short8v af[UNR][MT], bf0[UNR], bf1[UNR];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      af[u] = *reinterpret_cast<const short8v*>(aptr + k + u * 32);
+    for (int u = 0; u < UNR; ++u) {
+#pragma unroll
+      for (int t = 0; t < MT; ++t)
+        af[u][t] =
+            *reinterpret_cast<const short8v*>(aptr[t] + k + u * 32);
       bf0[u] = *reinterpret_cast<const short8v*>(bptr0 + k + u * 32);
       bf1[u] = *reinterpret_cast<const short8v*>(bptr1 + k + u * 32);
     }
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf0[u], acc0,
-                                                     0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf1[u], acc1,
-                                                     0, 0, 0);
+    for (int u = 0; u < UNR; ++u) {
+#pragma unroll
+      for (int t = 0; t < MT; ++t) {
+        acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[u][t], bf0[u], acc0[t], 0, 0, 0);
+        acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[u][t], bf1[u], acc1[t], 0, 0, 0);
+      }
     }
   }
   for (; k < kend; k += 32) {
-    short8v af = *reinterpret_cast<const short8v*>(aptr + k);
     short8v bf0 = *reinterpret_cast<const short8v*>(bptr0 + k);
     short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
-    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
-    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+#pragma unroll
+    for (int t = 0; t < MT; ++t) {
+      short8v af = *reinterpret_cast<const short8v*>(aptr[t] + k);
+      acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0[t],
+                                                        0, 0, 0);
+      acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1[t],
+                                                        0, 0, 0);
+    }
   }
 
   // combine the 4 waves' partials through LDS
-  __shared__ float comb[4][16][32];  // [wave][m][n] 8 KiB
+  __shared__ float comb[4][16 * MT][32];  // [wave][m][n] 8/16 KiB
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    comb[wave][crow_base + r][ccol] = acc0[r];
-    comb[wave][crow_base + r][ccol + 16] = acc1[r];
-  }
+  for (int t = 0; t < MT; ++t)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      comb[wave][16 * t + crow_base + r][ccol] = acc0[t][r];
+      comb[wave][16 * t + crow_base + r][ccol + 16] = acc1[t][r];
+    }
   __syncthreads();
   if (wave == 0) {
-    // 64 lanes fold 16x32 cells: lane covers 8 cells
+    // 64 lanes fold 16*MT x 32 cells: lane covers 8*MT cells
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
-      const int cell = lane * 8 + c;
+    for (int c = 0; c < 8 * MT; ++c) {
+      const int cell = lane * 8 * MT + c;
       const int m = cell >> 5;
       const int n = cell & 31;
       if (m >= M || n0 + n >= N) continue;
@@ -471,23 +520,30 @@ void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
                               int M, int N, int K, int ksplit, int variant,
                               void* stream) {
   const int nblocks = variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
-  if (variant == 2)
-    hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 16>),
-                       dim3(nblocks, ksplit), dim3(256), 0,
-                       (hipStream_t)stream, part_f32,
-                       (const unsigned short*)A, (const unsigned short*)W,
-                       M, N, K, ksplit);
-  else if (variant == 1)
-    hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 8>),
-                       dim3(nblocks, ksplit), dim3(256), 0,
-                       (hipStream_t)stream, part_f32,
-                       (const unsigned short*)A, (const unsigned short*)W,
-                       M, N, K, ksplit);
-  else
-    hipLaunchKernelGGL(skinny_gemm_kernel<true>, dim3(nblocks, ksplit),
-                       dim3(256), 0, (hipStream_t)stream, part_f32,
-                       (const unsigned short*)A, (const unsigned short*)W,
-                       M, N, K, ksplit);
+  const dim3 grid(nblocks, ksplit);
+  if (variant >= 1) {
+    if (M > 16)
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 2>), grid, dim3(256),
+                         0, (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+    else
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 1>), grid, dim3(256),
+                         0, (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+  } else {
+    if (M > 16)
+      hipLaunchKernelGGL((skinny_gemm_kernel<true, 2>), grid, dim3(256), 0,
+                         (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+    else
+      hipLaunchKernelGGL((skinny_gemm_kernel<true, 1>), grid, dim3(256), 0,
+                         (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+  }
 }
 
 void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
@@ -496,39 +552,34 @@ void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
   if (ksplit < 1) ksplit = 1;
   const int nblocks = variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
   if (ksplit == 1) {
-    if (variant == 2)
-      hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 16>), dim3(nblocks),
-                         dim3(256), 0, (hipStream_t)stream, out_bf16,
-                         (const unsigned short*)A, (const unsigned short*)W,
-                         M, N, K, 1);
-    else if (variant == 1)
-      hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 8>), dim3(nblocks),
-                         dim3(256), 0, (hipStream_t)stream, out_bf16,
-                         (const unsigned short*)A, (const unsigned short*)W,
-                         M, N, K, 1);
-    else
-      hipLaunchKernelGGL(skinny_gemm_kernel<false>, dim3(nblocks),
-                         dim3(256), 0, (hipStream_t)stream, out_bf16,
-                         (const unsigned short*)A, (const unsigned short*)W,
-                         M, N, K, 1);
+    if (variant >= 1) {
+      if (M > 16)
+        hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 2>),
+                           dim3(nblocks), dim3(256), 0,
+                           (hipStream_t)stream, out_bf16,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, 1);
+      else
+        hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 1>),
+                           dim3(nblocks), dim3(256), 0,
+                           (hipStream_t)stream, out_bf16,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, 1);
+    } else {
+      if (M > 16)
+        hipLaunchKernelGGL((skinny_gemm_kernel<false, 2>), dim3(nblocks),
+                           dim3(256), 0, (hipStream_t)stream, out_bf16,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, 1);
+      else
+        hipLaunchKernelGGL((skinny_gemm_kernel<false, 1>), dim3(nblocks),
+                           dim3(256), 0, (hipStream_t)stream, out_bf16,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, 1);
+    }
   } else {
-    if (variant == 2)
-      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 16>),
-                         dim3(nblocks, ksplit), dim3(256), 0,
-                         (hipStream_t)stream, part_f32,
-                         (const unsigned short*)A, (const unsigned short*)W,
-                         M, N, K, ksplit);
-    else if (variant == 1)
-      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 8>),
-                         dim3(nblocks, ksplit), dim3(256), 0,
-                         (hipStream_t)stream, part_f32,
-                         (const unsigned short*)A, (const unsigned short*)W,
-                         M, N, K, ksplit);
-    else
-      hipLaunchKernelGGL(skinny_gemm_kernel<true>, dim3(nblocks, ksplit),
-                         dim3(256), 0, (hipStream_t)stream, part_f32,
-                         (const unsigned short*)A, (const unsigned short*)W,
-                         M, N, K, ksplit);
+    launch_skinny_gemm_slabs(part_f32, A, W, M, N, K, ksplit, variant,
+                             stream);
     long long mn = (long long)M * N;
     long long blocks = (mn / 4 + 255) / 256;
     if (blocks > 2048) blocks = 2048;

@@ -95,7 +95,8 @@ class TestSiluMul:
 @requires_gpu
 class TestSkinnyGemm:
     @pytest.mark.parametrize("m,n,k", [(1, 128, 256), (16, 4096, 4096),
-                                       (13, 1000, 512), (8, 6144, 4096)])
+                                       (13, 1000, 512), (8, 6144, 4096),
+                                       (32, 4096, 4096), (24, 1024, 512)])
     def test_vs_fp32(self, m, n, k):
         from mlrun_amd import ops
 
