@@ -36,7 +36,7 @@ def main():
     parser.add_argument("--warmup", type=int, default=3)
     parser.add_argument("--model", default="llama-3-8b",
                         help="llama-3-8b | llama-3-70b | tiny")
-    parser.add_argument("--batch", type=int, default=16,
+    parser.add_argument("--batch", type=int, default=32,
                         help="requests per serving event")
     parser.add_argument("--prompt-len", type=int, default=128)
     parser.add_argument("--gen-tokens", type=int, default=32)
