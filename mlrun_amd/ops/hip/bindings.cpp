@@ -98,7 +98,7 @@ void skinny_gemm(torch::Tensor out_bf16, torch::Tensor part_f32,
   CHECK_DEV(w); CHECK_CONTIG(w); CHECK_BF16(w);
   int M = a.size(0), K = a.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "K mismatch");
-  TORCH_CHECK(M <= 32, "skinny_gemm supports M <= 32");
+  TORCH_CHECK(M <= 64, "skinny_gemm supports M <= 64");
   TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32");
   TORCH_CHECK(out_bf16.numel() == (int64_t)M * N, "out shape mismatch");
   if (ksplit > 1) {
@@ -118,7 +118,7 @@ void skinny_gemm_slabs(torch::Tensor part_f32, torch::Tensor a,
   CHECK_DEV(a); CHECK_CONTIG(a); CHECK_BF16(a);
   CHECK_DEV(w); CHECK_CONTIG(w); CHECK_BF16(w);
   int M = a.size(0), K = a.size(1), N = w.size(0);
-  TORCH_CHECK(w.size(1) == K && M <= 32 && K % 32 == 0, "bad shapes");
+  TORCH_CHECK(w.size(1) == K && M <= 64 && K % 32 == 0, "bad shapes");
   TORCH_CHECK(part_f32.numel() >= ksplit * (int64_t)M * N,
               "part_f32 scratch too small");
   launch_skinny_gemm_slabs(part_f32.data_ptr(), a.data_ptr(), w.data_ptr(),

@@ -292,7 +292,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   const unsigned short* bptr1 =
       W + (size_t)min(brow1, N - 1) * K + kb;
 
-  constexpr int UNR = (MT == 1) ? 8 : 4;
+  constexpr int UNR = (MT == 1) ? 8 : (MT == 2 ? 4 : 2);
   int k = kbegin;
   const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
   for (; k < kend8; k += UNR * 32) {
@@ -417,7 +417,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
   const unsigned short* bptr1 =
       W + (size_t)min(brow1, N - 1) * K + kb;
 
-  constexpr int UNR = (MT == 1) ? 8 : 4;
+  constexpr int UNR = (MT == 1) ? 8 : (MT == 2 ? 4 : 2);
   int k = kbegin;
   const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
   for (; k < kend8; k += UNR * 32) {
@@ -522,7 +522,12 @@ void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
   const int nblocks = variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
   const dim3 grid(nblocks, ksplit);
   if (variant >= 1) {
-    if (M > 16)
+    if (M > 32)
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 4>), grid, dim3(256),
+                         0, (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+    else if (M > 16)
       hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 2>), grid, dim3(256),
                          0, (hipStream_t)stream, part_f32,
                          (const unsigned short*)A, (const unsigned short*)W,
@@ -533,7 +538,12 @@ void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
                          (const unsigned short*)A, (const unsigned short*)W,
                          M, N, K, ksplit);
   } else {
-    if (M > 16)
+    if (M > 32)
+      hipLaunchKernelGGL((skinny_gemm_kernel<true, 4>), grid, dim3(256), 0,
+                         (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+    else if (M > 16)
       hipLaunchKernelGGL((skinny_gemm_kernel<true, 2>), grid, dim3(256), 0,
                          (hipStream_t)stream, part_f32,
                          (const unsigned short*)A, (const unsigned short*)W,
@@ -553,7 +563,13 @@ void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
   const int nblocks = variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
   if (ksplit == 1) {
     if (variant >= 1) {
-      if (M > 16)
+      if (M > 32)
+        hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 4>),
+                           dim3(nblocks), dim3(256), 0,
+                           (hipStream_t)stream, out_bf16,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, 1);
+      else if (M > 16)
         hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 2>),
                            dim3(nblocks), dim3(256), 0,
                            (hipStream_t)stream, out_bf16,
@@ -566,7 +582,12 @@ void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                            (const unsigned short*)A,
                            (const unsigned short*)W, M, N, K, 1);
     } else {
-      if (M > 16)
+      if (M > 32)
+        hipLaunchKernelGGL((skinny_gemm_kernel<false, 4>), dim3(nblocks),
+                           dim3(256), 0, (hipStream_t)stream, out_bf16,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, 1);
+      else if (M > 16)
         hipLaunchKernelGGL((skinny_gemm_kernel<false, 2>), dim3(nblocks),
                            dim3(256), 0, (hipStream_t)stream, out_bf16,
                            (const unsigned short*)A,
