@@ -112,24 +112,28 @@ class WindowRing:
         ops.window_ingest(self.ring, keys32, vals_dev, pidx)
         ops.window_ingest(self.ring_sq, keys32, vals_dev * vals_dev, pidx)
         # running aggregates over the FULL batch (storey keeps running
-        # first/last/min/max even for late events)
+        # first/last/min/max even for late events) — fully vectorized
+        # via sorted segments + reduceat (no per-key python loop)
         k = key_ids.numpy()
         v = values.to(torch.float32).numpy()
         order = np.argsort(k, kind="stable")
         k_sorted, v_sorted = k[order], v[order]
         uniq, starts = np.unique(k_sorted, return_index=True)
         ends = np.append(starts[1:], len(k_sorted))
-        for key, s, e in zip(uniq, starts, ends):
-            seg = v_sorted[s:e]
-            row = self.running[key]
-            row[0] = min(float(row[0]), float(seg.min()))
-            row[1] = max(float(row[1]), float(seg.max()))
-            if row[4] == 0:
-                row[2] = float(seg[0])
-            row[3] = float(seg[-1])
-            row[4] += len(seg)
-            row[5] += float(seg.sum())
-            row[6] += float((seg * seg).sum())
+        seg_min = np.minimum.reduceat(v_sorted, starts)
+        seg_max = np.maximum.reduceat(v_sorted, starts)
+        seg_sum = np.add.reduceat(v_sorted, starts)
+        seg_sumsq = np.add.reduceat(v_sorted * v_sorted, starts)
+        seg_count = (ends - starts).astype(np.float32)
+        run = self.running.numpy()
+        fresh_keys = run[uniq, 4] == 0
+        run[uniq, 0] = np.minimum(run[uniq, 0], seg_min)
+        run[uniq, 1] = np.maximum(run[uniq, 1], seg_max)
+        run[uniq[fresh_keys], 2] = v_sorted[starts[fresh_keys]]  # first
+        run[uniq, 3] = v_sorted[ends - 1]                        # last
+        run[uniq, 4] += seg_count
+        run[uniq, 5] += seg_sum
+        run[uniq, 6] += seg_sumsq
 
     def window_values(self, window_seconds: int, now_ts: float) -> dict:
         """Reduce the ring for one window length -> tensors keyed by op
@@ -192,18 +196,22 @@ class OnlineTable:
         return tuple(row[e] for e in entities)
 
     def _key_ids(self, keys: list) -> torch.Tensor:
-        ids = []
-        for key in keys:
+        import pandas as pd
+
+        codes, uniques = pd.factorize(pd.Index(keys), sort=False)
+        # map each unique key through (and extend) the dictionary once
+        lut = np.empty(len(uniques), dtype=np.int64)
+        for i, key in enumerate(uniques):
             idx = self.key_index.get(key)
             if idx is None:
                 idx = len(self.key_index)
                 self.key_index[key] = idx
-            ids.append(idx)
+            lut[i] = idx
         needed = len(self.key_index)
         for ring in self.rings.values():
             if needed > ring.capacity:
                 ring.grow(max(needed, ring.capacity * 2))
-        return torch.tensor(ids, dtype=torch.int64)
+        return torch.from_numpy(lut[codes])
 
     def ingest_batch(self, df):
         """Fold a dataframe batch: update latest rows + window rings."""
@@ -211,9 +219,12 @@ class OnlineTable:
 
         fset = self.feature_set
         ts_key = fset.spec.timestamp_key
+        entities = fset.entity_names()
         with self._lock:
-            keys = [self._key_of(row) for row in
-                    df.to_dict(orient="records")]
+            if len(entities) == 1:
+                keys = df[entities[0]].tolist()
+            else:
+                keys = list(zip(*(df[e] for e in entities)))
             key_ids = self._key_ids(keys)
             if ts_key and ts_key in df.columns:
                 ts = pd.to_datetime(df[ts_key]).astype("int64") // 10 ** 9
@@ -227,7 +238,18 @@ class OnlineTable:
                 values = torch.tensor(
                     df[agg.column].astype("float32").values)
                 self.rings[agg.name].ingest(key_ids, values, timestamps)
-            for key, row in zip(keys, df.to_dict(orient="records")):
+            # latest row per key: one pandas pass, not a python loop
+            if ts_key and ts_key in df.columns:
+                ordered = df.sort_values(ts_key, kind="stable")
+            else:
+                ordered = df
+            last_rows = ordered.drop_duplicates(subset=entities,
+                                                keep="last")
+            last_keys = (last_rows[entities[0]].tolist()
+                         if len(entities) == 1 else
+                         list(zip(*(last_rows[e] for e in entities))))
+            for key, row in zip(last_keys,
+                                last_rows.to_dict(orient="records")):
                 self.latest[key] = row
         return len(df)
 

@@ -363,12 +363,14 @@ def window_ingest(ring, keys, values, period_idx):
     if ring.is_cuda:
         _require_hip().window_ingest(ring, keys, values, period_idx)
         return ring
+    import numpy as np
+
     n_periods = ring.shape[1]
-    for key, value, pidx in zip(keys.tolist(), values.tolist(),
-                                period_idx.tolist()):
-        cell = ring[key, pidx % n_periods]
-        cell[0] += value
-        cell[1] += 1.0
+    flat_idx = keys.numpy().astype(np.int64) * n_periods + \
+        (period_idx.numpy().astype(np.int64) % n_periods)
+    view = ring.numpy().reshape(-1, 4)
+    np.add.at(view[:, 0], flat_idx, values.numpy())
+    np.add.at(view[:, 1], flat_idx, 1.0)
     return ring
 
 
