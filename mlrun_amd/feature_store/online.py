@@ -253,6 +253,49 @@ class OnlineTable:
                 self.latest[key] = row
         return len(df)
 
+    def get_agg_matrix(self, entity_rows: typing.List[dict],
+                       feature_names: typing.List[str],
+                       now_ts: float = None):
+        """Columnar fast path: aggregate features only -> float32
+        matrix [rows, features] (NaN for unknown keys).  One ring
+        reduce per (agg, window) serves the whole batch; rows resolve
+        by fancy indexing — no per-row dicts."""
+        import numpy as np
+
+        now_ts = now_ts or time.time()
+        fset = self.feature_set
+        with self._lock:
+            ids = np.array([self.key_index.get(self._key_of(row), -1)
+                            for row in entity_rows], dtype=np.int64)
+            known = ids >= 0
+            safe = np.where(known, ids, 0)
+            out = np.full((len(entity_rows), len(feature_names)), np.nan,
+                          dtype=np.float32)
+            reduced_cache = {}
+            for j, name in enumerate(feature_names):
+                resolved = None
+                for agg in fset.spec.aggregations:
+                    for window in agg.windows:
+                        for op in agg.operations:
+                            if f"{agg.name}_{op}_{window}" == name:
+                                resolved = (agg, window, op)
+                if resolved is None:
+                    continue
+                agg, window, op = resolved
+                ring = self.rings[agg.name]
+                ckey = (agg.name, window)
+                if ckey not in reduced_cache:
+                    reduced_cache[ckey] = ring.window_values(
+                        parse_span(window), now_ts)
+                vals = reduced_cache[ckey]
+                if op in vals:
+                    col = vals[op].numpy()[safe]
+                else:
+                    mapping = {"min": 0, "max": 1, "first": 2, "last": 3}
+                    col = ring.running.numpy()[safe, mapping[op]]
+                out[:, j] = np.where(known, col, np.nan)
+            return out
+
     def get(self, entity_rows: typing.List[dict], now_ts: float = None
             ) -> typing.List[dict]:
         """Batched online lookup: latest values + window aggregates."""

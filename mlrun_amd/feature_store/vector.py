@@ -144,6 +144,14 @@ class OnlineVectorService:
         self.impute_policy = impute_policy or {}
         self._groups = list(vector.grouped_features())
 
+    def _all_aggregates(self) -> bool:
+        for set_name, cols, _ in self._groups:
+            fset = self._tables[set_name].feature_set
+            agg_names = {f.name for f in fset.spec.features if f.aggregate}
+            if cols == ["*"] or any(c not in agg_names for c in cols):
+                return False
+        return True
+
     @property
     def status(self):
         return "ready"
@@ -152,6 +160,28 @@ class OnlineVectorService:
         """entity_rows: [{entity: value, ...}, ...] -> feature records."""
         if isinstance(entity_rows, dict):
             entity_rows = [entity_rows]
+        if as_list and self._all_aggregates():
+            # columnar fast path: fancy-indexed ring reductions
+            columns = []
+            for set_name, cols, aliases in self._groups:
+                table = self._tables[set_name]
+                matrix = table.get_agg_matrix(entity_rows, cols)
+                columns.append(matrix)
+            import numpy as np
+
+            matrix = np.concatenate(columns, axis=1)
+            out = matrix.astype(object)
+            # imputation by feature name order
+            names = [aliases.get(c, c) for _, cols, aliases in self._groups
+                     for c in cols]
+            for j, name in enumerate(names):
+                if name in self.impute_policy:
+                    col = out[:, j]
+                    col[np.isnan(matrix[:, j])] = self.impute_policy[name]
+                else:
+                    col = out[:, j]
+                    col[np.isnan(matrix[:, j])] = None
+            return out.tolist()
         results = [dict() for _ in entity_rows]
         ordered_names: typing.List[str] = []
         for set_name, columns, aliases in self._groups:

@@ -53,6 +53,19 @@ def run(device_label: str, device, n_keys=100_000, n_events=1_000_000,
         torch.cuda.synchronize()
     lookup_dt = (time.perf_counter() - t0) / lookup_iters
 
+    # columnar fast path (OnlineVectorService as_list over aggregates)
+    names = ["amount_sum_1h", "amount_avg_1h", "amount_count_1h"]
+    table.get_agg_matrix(keys[:8], names)
+    t0 = time.perf_counter()
+    for _ in range(lookup_iters):
+        table.get_agg_matrix(keys, names)
+    if device != "cpu":
+        torch.cuda.synchronize()
+    col_dt = (time.perf_counter() - t0) / lookup_iters
+    print(f"[{device_label}] columnar get: "
+          f"{lookup_batch / col_dt:,.0f} lookups/s "
+          f"({col_dt * 1000:.2f} ms per batch)", flush=True)
+
     print(f"[{device_label}] ingest: {n_events / ingest_dt:,.0f} events/s "
           f"({ingest_dt * 1000:.0f} ms for {n_events:,}); "
           f"online get: {lookup_batch / lookup_dt:,.0f} lookups/s "

@@ -184,3 +184,29 @@ class TestVectorAndServices:
         assert parse_feature_string("s.f as x") == ("s", "f", "x")
         with pytest.raises(MLRunInvalidArgumentError):
             parse_feature_string("nofset")
+
+
+class TestColumnarFastPath:
+    def test_agg_matrix_matches_dict_path(self):
+        import numpy as np
+
+        df = make_df(300)
+        fset = fstore.FeatureSet("colfp", entities=["customer"],
+                                 timestamp_key="ts")
+        fset.add_aggregation("amount", ["sum", "avg"], ["1h"], period="10m")
+        fstore.ingest(fset, df, targets=["nosql"])
+        vector = fstore.FeatureVector(
+            "vcol", features=["colfp.amount_sum_1h", "colfp.amount_avg_1h"])
+        vector.metadata.project = "default"
+        svc = fstore.get_online_feature_service(vector)
+        rows = [{"customer": "alice"}, {"customer": "bob"},
+                {"customer": "nobody"}]
+        fast = svc.get(rows, as_list=True)
+        slow = svc.get(rows, as_list=False)
+        for f_row, s_row in zip(fast, slow):
+            assert (f_row[0] is None) == (s_row["amount_sum_1h"] is None)
+            if f_row[0] is not None:
+                assert abs(f_row[0] - s_row["amount_sum_1h"]) < 1e-3
+                assert abs(f_row[1] - s_row["amount_avg_1h"]) < 1e-3
+        # unknown key imputes to None in both paths
+        assert fast[2][0] is None
