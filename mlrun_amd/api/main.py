@@ -102,11 +102,20 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     @app.get("/api/v1/runs")
     async def list_runs(project: str = "default", name: str = "",
                         state: str = "", sort: int = 1, last: int = 0,
-                        iter: int = 0, uid: typing.List[str] = Query(None),
+                        iter: int = 0, page: int = 0, page_size: int = 0,
+                        uid: typing.List[str] = Query(None),
                         label: typing.List[str] = Query(None)):
         runs = db.list_runs(name=name, uid=uid, project=project,
                             labels=label, state=state or None,
                             sort=bool(sort), last=last, iter=bool(iter))
+        total = len(runs)
+        if page_size:
+            start = max(page - 1, 0) * page_size
+            runs = runs[start:start + page_size]
+            return {"runs": runs,
+                    "pagination": {"page": max(page, 1),
+                                   "page_size": page_size,
+                                   "total": total}}
         return {"runs": runs}
 
     # ------------------------------------------------------------ logs
@@ -350,6 +359,25 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     @app.get("/api/v1/projects/{project}/background-tasks")
     async def list_background_tasks(project: str):
         return {"background_tasks": db.list_background_tasks(project)}
+
+    # -------------------------------------------------- runtime resources
+    @app.get("/api/v1/projects/{project}/runtime-resources")
+    async def runtime_resources(project: str):
+        """Node-local analog of the reference's runtime-resources
+        listing: live GPU leases + running runs."""
+        from ..parallel.scheduler import get_gpu_allocator
+
+        allocator = get_gpu_allocator()
+        running = db.list_runs(project=project, state="running")
+        return {
+            "gpu": {"total": allocator.total,
+                    "in_use": {str(k): v for k, v in
+                               allocator.usage().items()},
+                    "available": allocator.available()},
+            "runs": [{"uid": r.get("metadata", {}).get("uid"),
+                      "name": r.get("metadata", {}).get("name")}
+                     for r in running],
+        }
 
     # ------------------------------------------------------------ submit
     @app.post("/api/v1/submit_job")

@@ -184,3 +184,38 @@ class TestCron:
         assert len(runs) == 1
         sched = db.get_schedule("default", "s1")
         assert sched["last_run_uri"]
+
+
+class TestRuntimeResources:
+    def test_listing_with_fake_allocator(self, client):
+        """The reference tests k8s handlers against a mocked clientset;
+        the analog seam here is a fake GPU allocator."""
+        from mlrun_amd.parallel.scheduler import (
+            GpuAllocator, set_gpu_allocator, get_gpu_allocator)
+
+        fake = GpuAllocator(total=4)
+        set_gpu_allocator(fake)
+        try:
+            lease = fake.acquire(2, owner="run-1")
+            resp = client.get(
+                "/api/v1/projects/default/runtime-resources").json()
+            assert resp["gpu"]["total"] == 4
+            assert len(resp["gpu"]["in_use"]) == 2
+            assert resp["gpu"]["available"] == [2, 3]
+            lease.release()
+            resp = client.get(
+                "/api/v1/projects/default/runtime-resources").json()
+            assert resp["gpu"]["in_use"] == {}
+        finally:
+            set_gpu_allocator(GpuAllocator())
+
+    def test_pagination(self, client):
+        for i in range(7):
+            client.post(f"/api/v1/run/p/u{i}",
+                        json={"metadata": {"name": f"r{i}", "uid": f"u{i}"},
+                              "status": {"state": "completed"}})
+        resp = client.get("/api/v1/runs",
+                          params={"project": "p", "page": 2,
+                                  "page_size": 3}).json()
+        assert len(resp["runs"]) == 3
+        assert resp["pagination"]["total"] == 7
