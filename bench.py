@@ -41,7 +41,12 @@ def main():
     parser.add_argument("--prompt-len", type=int, default=128)
     parser.add_argument("--gen-tokens", type=int, default=32)
     parser.add_argument("--no-graph", action="store_true")
+    parser.add_argument("--replicas", type=int, default=1,
+                        help="engine replicas per GPU (own HIP streams)")
+    parser.add_argument("--inflight", type=int, default=0,
+                        help="concurrent events (default = replicas)")
     args = parser.parse_args()
+    inflight = args.inflight or args.replicas
 
     import torch
 
@@ -71,7 +76,8 @@ def main():
     fn = mlrun_amd.new_function(name="bench-serving", kind="serving")
     fn.add_model("llama", class_name=LlamaServer, config=model_cfg,
                  batch_size=batch, max_new_tokens=args.gen_tokens,
-                 device=device, use_graph=not args.no_graph)
+                 device=device, use_graph=not args.no_graph,
+                 replicas=args.replicas)
     server = fn.to_mock_server()
 
     import random
@@ -89,22 +95,24 @@ def main():
 
     path = "/v2/models/llama/infer"
 
-    # warmup (includes weight init, first prefill, hipGraph capture)
-    for _ in range(args.warmup):
+    import concurrent.futures
+
+    pool = concurrent.futures.ThreadPoolExecutor(max_workers=inflight)
+
+    def one_step():
+        s0 = time.perf_counter()
         server.test(path, body=make_body())
+        return (time.perf_counter() - s0) * 1000.0
+
+    # warmup (includes weight init, first prefill, hipGraph capture)
+    list(pool.map(lambda _: one_step(), range(max(args.warmup, inflight))))
     if on_gpu:
         torch.cuda.synchronize()
     if dist:
         dist.barrier()
 
-    latencies = []
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        s0 = time.perf_counter()
-        server.test(path, body=make_body())
-        if on_gpu:
-            torch.cuda.synchronize()
-        latencies.append((time.perf_counter() - s0) * 1000.0)
+    latencies = list(pool.map(lambda _: one_step(), range(args.steps)))
     if on_gpu:
         torch.cuda.synchronize()
     if dist:
@@ -144,7 +152,8 @@ def main():
                 "seq_len": prompt_len + args.gen_tokens,
                 "prompt_len": prompt_len,
                 "gen_tokens": args.gen_tokens,
-                "parallelism": f"dp{n_gpus} (1 serving replica/GPU)",
+                "parallelism": f"dp{n_gpus} ({args.replicas} engine "
+                               f"replica(s)/GPU, {inflight} in flight)",
                 "p50_ms": round(p50, 3),
                 "tokens_per_sec": round(
                     reqs_per_sec * args.gen_tokens, 1),

@@ -129,7 +129,7 @@ class LlamaDecodeEngine:
 
     def __init__(self, cfg: LlamaConfig, batch_size: int, device=None,
                  tp_group=None, tp_rank=0, tp_size=1, use_graph=True,
-                 seed=1234):
+                 seed=1234, weights: "LlamaWeights" = None):
         self.cfg = cfg
         self.B = batch_size
         self.device = torch.device(
@@ -139,7 +139,11 @@ class LlamaDecodeEngine:
         self.tp_rank = tp_rank
         self.tp_size = tp_size
         self.use_graph = use_graph and self.on_gpu
-        self.weights = LlamaWeights(cfg, self.device, tp_rank, tp_size, seed)
+        # weights are read-only at serving time: replica engines on the
+        # same GPU share one copy (16 GB for 8B) and keep private
+        # KV caches / buffers / streams
+        self.weights = weights if weights is not None else \
+            LlamaWeights(cfg, self.device, tp_rank, tp_size, seed)
         w = self.weights
         d, h = cfg.head_dim, cfg.hidden_size
         B, smax = batch_size, cfg.max_seq_len
@@ -425,15 +429,27 @@ class LlamaServer:
 
     Request: {"inputs": [[tok, ...], ...], "max_tokens": G}
     Response outputs: [[generated tokens], ...]
-    Events are served at the engine's fixed batch size: smaller request
-    batches are padded, larger are chunked.
+
+    Serving structure:
+    - ``replicas`` engine replicas per GPU, each on its OWN HIP stream
+      with private KV caches/buffers but SHARED weights — decode is
+      latency-bound (PMC wait/busy ~7), so concurrent replica streams
+      raise aggregate HBM utilization; requests must arrive
+      concurrently (or via the batcher) to overlap
+    - a task queue feeds the replica workers; events are chunked to
+      the engine batch size
+    - ``batch_window_ms``: small concurrent requests coalesce into one
+      engine pass (dynamic batching)
     """
 
     def __init__(self, context=None, name=None, model_path=None,
                  config=None, batch_size=16, max_new_tokens=32,
                  device=None, use_graph=True, batch_window_ms=0,
-                 **class_args):
-        from ..serving.v2_serving import V2ModelServer
+                 replicas=1, **class_args):
+        import queue as queue_mod
+        import threading
+
+        from ..serving.v2_serving import V2ModelServer  # noqa: F401
 
         self.name = name
         self.context = context
@@ -449,19 +465,19 @@ class LlamaServer:
         self.max_new_tokens = max_new_tokens
         self.device = device
         self.use_graph = use_graph
-        self.engine: typing.Optional[LlamaDecodeEngine] = None
-        # the engine owns mutable KV/cache state: serialize events
-        import threading
-
-        self._engine_lock = threading.Lock()
-        # dynamic batching of concurrent single-prompt requests (the
-        # reference harness sends N concurrent clients; batching them
-        # into one engine pass multiplies throughput)
+        self.replicas = max(int(replicas), 1)
+        self.engines: typing.List[LlamaDecodeEngine] = []
         self.batch_window_ms = batch_window_ms
+        self._tasks: "queue_mod.Queue" = queue_mod.Queue()
+        self._workers: typing.List[threading.Thread] = []
         self._pending = []          # [(prompt, max_new, future)]
         self._pending_cv = threading.Condition()
         self._batcher = None
         self.engine_calls = 0
+
+    @property
+    def engine(self):
+        return self.engines[0] if self.engines else None
 
     def post_init(self, mode="sync"):
         stream = getattr(self.context, "stream", None) if self.context \
@@ -474,7 +490,7 @@ class LlamaServer:
             self.load()
             self.ready = True
 
-    def load(self):
+    def _build_config(self) -> LlamaConfig:
         if self.cfg_name in ("llama-3-8b", "8b"):
             cfg = LlamaConfig.llama3_8b()
         elif self.cfg_name in ("llama-3-70b", "70b"):
@@ -488,20 +504,57 @@ class LlamaServer:
         for key, value in self._params.items():
             if hasattr(cfg, key):
                 setattr(cfg, key, value)
-        self.engine = LlamaDecodeEngine(cfg, self.batch_size,
-                                        device=self.device,
-                                        use_graph=self.use_graph)
+        return cfg
+
+    def load(self):
+        import threading
+
+        cfg = self._build_config()
+        if not torch.cuda.is_available():
+            self.replicas = 1
+        first = LlamaDecodeEngine(cfg, self.batch_size, device=self.device,
+                                  use_graph=self.use_graph)
+        self.engines = [first]
+        for _ in range(self.replicas - 1):
+            self.engines.append(LlamaDecodeEngine(
+                cfg, self.batch_size, device=self.device,
+                use_graph=self.use_graph, weights=first.weights))
         # load a checkpoint artifact if given (model_spec.yaml layout)
         if self.model_path:
             from ..artifacts import get_model
 
             model_file, spec, extra = get_model(self.model_path)
-            state = torch.load(model_file, map_location=self.engine.device,
+            state = torch.load(model_file, map_location=first.device,
                                weights_only=True)
-            self.engine.weights.load_state_dict(state)
+            first.weights.load_state_dict(state)
             self.model_spec = spec
+        for idx, engine in enumerate(self.engines):
+            engine._serve_stream = torch.cuda.Stream() \
+                if engine.on_gpu else None
+            worker = threading.Thread(target=self._worker_loop,
+                                      args=(engine,), daemon=True,
+                                      name=f"llama-worker-{self.name}-{idx}")
+            worker.start()
+            self._workers.append(worker)
+
+    def _worker_loop(self, engine: LlamaDecodeEngine):
+        while True:
+            prompts, max_new, future = self._tasks.get()
+            try:
+                if engine._serve_stream is not None:
+                    with torch.cuda.stream(engine._serve_stream):
+                        result = self._generate_on(engine, prompts,
+                                                   max_new)
+                    engine._serve_stream.synchronize()
+                else:
+                    result = self._generate_on(engine, prompts, max_new)
+                future.set_result(result)
+            except Exception as exc:
+                if not future.done():
+                    future.set_exception(exc)
 
     def do_event(self, event):
+        import concurrent.futures
         import time as _time
 
         start = _time.perf_counter()
@@ -517,24 +570,25 @@ class LlamaServer:
         if self.batch_window_ms and len(inputs) < self.batch_size:
             outputs = self._batched_submit(inputs, max_new)
         else:
+            futures = []
+            for chunk_start in range(0, len(inputs), self.batch_size):
+                chunk = inputs[chunk_start:chunk_start + self.batch_size]
+                future = concurrent.futures.Future()
+                self._tasks.put((chunk, max_new, future))
+                futures.append(future)
             outputs = []
-            with self._engine_lock:
-                for chunk_start in range(0, len(inputs), self.batch_size):
-                    chunk = inputs[chunk_start:
-                                   chunk_start + self.batch_size]
-                    outputs.extend(self._generate_chunk(chunk, max_new))
+            for future in futures:
+                outputs.extend(future.result(timeout=600))
         event.body = {"id": event.id, "model_name": self.name,
                       "outputs": outputs}
-        self.engine_calls += 0  # engine-call count tracked in chunk
         if self._model_logger:
             self._model_logger.push(start, {"inputs": [len(inputs)]},
                                     event.body)
         return event
 
     def _batched_submit(self, inputs: list, max_new: int) -> list:
-        """Queue small requests; a batcher thread gathers up to
-        batch_size prompts within batch_window_ms and runs ONE engine
-        pass for all of them."""
+        """Queue small requests; the batcher gathers up to batch_size
+        prompts within batch_window_ms and submits ONE engine task."""
         import concurrent.futures
         import threading
 
@@ -553,23 +607,26 @@ class LlamaServer:
         return [f.result(timeout=600) for f in futures]
 
     def _batch_loop(self):
+        import concurrent.futures
         import time as _time
 
         while True:
             with self._pending_cv:
                 while not self._pending:
                     self._pending_cv.wait()
-            # window: let more requests arrive
             _time.sleep(self.batch_window_ms / 1000.0)
             with self._pending_cv:
-                batch, self._pending =                     self._pending[:self.batch_size],                     self._pending[self.batch_size:]
+                batch, self._pending = \
+                    self._pending[:self.batch_size], \
+                    self._pending[self.batch_size:]
             if not batch:
                 continue
             prompts = [b[0] for b in batch]
             max_new = max(b[1] for b in batch)
+            task_future = concurrent.futures.Future()
+            self._tasks.put((prompts, max_new, task_future))
             try:
-                with self._engine_lock:
-                    results = self._generate_chunk(prompts, max_new)
+                results = task_future.result(timeout=600)
                 for (prompt, want, future), result in zip(batch, results):
                     future.set_result(result[:want])
             except Exception as exc:
@@ -577,7 +634,8 @@ class LlamaServer:
                     if not future.done():
                         future.set_exception(exc)
 
-    def _generate_chunk(self, prompts: list, max_new: int) -> list:
+    def _generate_on(self, engine: LlamaDecodeEngine, prompts: list,
+                     max_new: int) -> list:
         self.engine_calls += 1
         n = len(prompts)
         max_len = max(len(p) for p in prompts)
@@ -585,8 +643,8 @@ class LlamaServer:
         for i, prompt in enumerate(prompts):
             tokens[i, max_len - len(prompt):] = torch.tensor(
                 prompt, dtype=torch.int64)  # left-pad
-        self.engine.reset()
-        out = self.engine.generate(tokens, max_new_tokens=max_new)
+        engine.reset()
+        out = engine.generate(tokens, max_new_tokens=max_new)
         return out[:n].cpu().tolist()
 
     def logged_results(self, request, response, op):
