@@ -181,6 +181,7 @@ class LlamaDecodeEngine:
         self.buf_c32 = torch.empty(B * scratch, dtype=torch.float32,
                                    device=self.device)
         self.scale = 1.0 / math.sqrt(d)
+        self._capture_stream_ctx = None
         self.attn_nsplit = ops.pick_attn_nsplit(B, w.hkv)
         self.buf_attn_ws = torch.empty(
             B * w.hq * self.attn_nsplit * (d + 2), dtype=torch.float32,
@@ -317,10 +318,25 @@ class LlamaDecodeEngine:
         self._gemm(self.buf_hidden, w.lm_head, self.buf_logits)
         torch.argmax(self.buf_logits, dim=-1, out=self.buf_tokens)
 
+    _capture_lock = None
+
     def capture_graph(self):
         """Capture the decode step as one hipGraph (3 warmup runs on a
-        side stream per torch graph discipline)."""
+        side stream per torch graph discipline).  Captures are
+        serialized process-wide: concurrent stream captures on one
+        device corrupt each other."""
         if not self.use_graph or self._graph is not None:
+            return
+        import threading
+
+        cls = type(self)
+        if cls._capture_lock is None:
+            cls._capture_lock = threading.Lock()
+        with cls._capture_lock:
+            self._capture_graph_locked()
+
+    def _capture_graph_locked(self):
+        if self._graph is not None:
             return
         lens_backup = self.cache_lens.clone()
         tokens_backup = self.buf_tokens.clone()
@@ -531,6 +547,16 @@ class LlamaServer:
         for idx, engine in enumerate(self.engines):
             engine._serve_stream = torch.cuda.Stream() \
                 if engine.on_gpu else None
+            if engine.use_graph:
+                # capture sequentially before workers start (concurrent
+                # captures on one device conflict); cache state is
+                # restored by capture_graph
+                if engine._serve_stream is not None:
+                    with torch.cuda.stream(engine._serve_stream):
+                        engine.capture_graph()
+                    engine._serve_stream.synchronize()
+                else:
+                    engine.capture_graph()
             worker = threading.Thread(target=self._worker_loop,
                                       args=(engine,), daemon=True,
                                       name=f"llama-worker-{self.name}-{idx}")
