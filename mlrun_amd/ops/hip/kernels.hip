@@ -292,7 +292,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   const unsigned short* bptr1 =
       W + (size_t)min(brow1, N - 1) * K + kb;
 
-  constexpr int UNR = (MT == 1) ? 8 : (MT == 2 ? 4 : 2);
+#ifndef MLRUN_GEMM_UNR1
+#define MLRUN_GEMM_UNR1 8
+#endif
+  constexpr int UNR = (MT == 1) ? MLRUN_GEMM_UNR1 : (MT == 2 ? 4 : 2);
   int k = kbegin;
   const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
   for (; k < kend8; k += UNR * 32) {
@@ -417,7 +420,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
   const unsigned short* bptr1 =
       W + (size_t)min(brow1, N - 1) * K + kb;
 
-  constexpr int UNR = (MT == 1) ? 8 : (MT == 2 ? 4 : 2);
+#ifndef MLRUN_GEMM_UNR1
+#define MLRUN_GEMM_UNR1 8
+#endif
+  constexpr int UNR = (MT == 1) ? MLRUN_GEMM_UNR1 : (MT == 2 ? 4 : 2);
   int k = kbegin;
   const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
   for (; k < kend8; k += UNR * 32) {
