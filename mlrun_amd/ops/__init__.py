@@ -159,12 +159,12 @@ _EMPTY_F32 = None
 # measured on MI355X (scripts/bench_gemm.py sweep, profiles/): best
 # (ksplit, variant) per llama projection shape; variant 1/2 = wave-split
 _GEMM_PLAN_TABLE = {
-    # re-measured after the manual-unroll K-loop rewrite
-    (6144, 4096): (1, 2),     # qkv      3.49 TB/s
-    (4096, 4096): (2, 2),     # wo       3.17 TB/s
-    (28672, 4096): (1, 2),    # gate|up  5.09 TB/s
-    (4096, 14336): (4, 2),    # down     4.64 TB/s
-    (128256, 4096): (1, 0),   # lm_head  4.26 TB/s
+    # re-measured at manual unroll depth 16 (scripts/bench_gemm.py)
+    (6144, 4096): (1, 2),     # qkv      3.66 TB/s
+    (4096, 4096): (2, 2),     # wo       3.20 TB/s
+    (28672, 4096): (1, 0),    # gate|up  5.18 TB/s
+    (4096, 14336): (2, 2),    # down     4.73 TB/s
+    (128256, 4096): (1, 0),   # lm_head  4.52 TB/s
 }
 
 

@@ -293,7 +293,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
       W + (size_t)min(brow1, N - 1) * K + kb;
 
 #ifndef MLRUN_GEMM_UNR1
-#define MLRUN_GEMM_UNR1 8
+#define MLRUN_GEMM_UNR1 16
 #endif
   constexpr int UNR = (MT == 1) ? MLRUN_GEMM_UNR1 : (MT == 2 ? 4 : 2);
   int k = kbegin;
@@ -421,7 +421,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
       W + (size_t)min(brow1, N - 1) * K + kb;
 
 #ifndef MLRUN_GEMM_UNR1
-#define MLRUN_GEMM_UNR1 8
+#define MLRUN_GEMM_UNR1 16
 #endif
   constexpr int UNR = (MT == 1) ? MLRUN_GEMM_UNR1 : (MT == 2 ? 4 : 2);
   int k = kbegin;
