@@ -247,3 +247,17 @@ class TestPlatforms:
 
         fn = mlrun_amd.new_function(name="m", kind="job")
         assert mlrun_amd.auto_mount()(fn) is fn
+
+
+class TestMlrunAlias:
+    def test_import_mlrun(self):
+        """Reference user code (`import mlrun`) runs unchanged."""
+        import mlrun
+        import mlrun.feature_store as fstore
+        from mlrun.serving import V2ModelServer  # noqa: F401
+
+        assert mlrun.get_version()
+        run = mlrun.run_local(handler=lambda context: context.log_result(
+            "ok", 1), name="alias-run")
+        assert run.status.results["ok"] == 1
+        assert fstore.FeatureSet("a", entities=["k"]).name == "a"
