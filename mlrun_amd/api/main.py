@@ -322,6 +322,20 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         db.delete_model_endpoint(project, endpoint_id)
         return {}
 
+    @app.get("/api/v1/projects/{project}/model-endpoints/{endpoint_id}"
+             "/metrics")
+    async def model_endpoint_metrics(project: str, endpoint_id: str):
+        """Time series of window stats from the monitoring TSDB
+        (reference: grafana-proxy / TSDB connectors)."""
+        from ..model_monitoring import get_stream_processor
+
+        processor = get_stream_processor(project)
+        series = processor.tsdb_series(endpoint_id)
+        return {"endpoint_id": endpoint_id,
+                "series": [{"time": t, "stats": stats}
+                           for t, stats in series],
+                "current": processor.endpoint_stats(endpoint_id)}
+
     # ------------------------------------------------------------ alerts
     @app.put("/api/v1/projects/{project}/alerts/{name}")
     async def store_alert(project: str, name: str, request: Request):

@@ -30,7 +30,14 @@ def list_hub_sources() -> list:
     return sorted(_sources.values(), key=lambda s: s["order"])
 
 
+BUILTIN_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                           "hub_functions")
+
+
 def _ensure_default():
+    if "builtin" not in _sources:
+        _sources["builtin"] = {"name": "builtin", "path": BUILTIN_DIR,
+                               "order": 0}
     if "default" not in _sources:
         _sources["default"] = {
             "name": "default",
@@ -59,18 +66,25 @@ def get_hub_function(url_body: str):
 
     _ensure_default()
     body = url_body.strip("/")
-    source = "default"
+    source = None
     if "/" in body:
         source, body = body.split("/", 1)
     name = body.split(":")[0]
-    src = _sources.get(source)
-    if not src:
-        raise MLRunNotFoundError(f"hub source {source} not found")
-    yaml_path = os.path.join(src["path"], name, "function.yaml")
-    if not os.path.isfile(yaml_path):
+    candidates = [source] if source else \
+        [s["name"] for s in sorted(_sources.values(),
+                                   key=lambda s: s["order"])]
+    yaml_path = None
+    for cand in candidates:
+        src = _sources.get(cand)
+        if not src:
+            continue
+        path = os.path.join(src["path"], name, "function.yaml")
+        if os.path.isfile(path):
+            yaml_path = path
+            break
+    if yaml_path is None:
         raise MLRunNotFoundError(
-            f"hub function {name} not found in source {source} "
-            f"({yaml_path})")
+            f"hub function {name} not found in sources {candidates}")
     with open(yaml_path) as fp:
         struct = yaml.safe_load(fp)
     return new_function(runtime=struct)

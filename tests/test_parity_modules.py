@@ -261,3 +261,41 @@ class TestMlrunAlias:
             "ok", 1), name="alias-run")
         assert run.status.results["ok"] == 1
         assert fstore.FeatureSet("a", entities=["k"]).name == "a"
+
+
+class TestBuiltinHub:
+    def test_describe_function(self, tmp_path):
+        import pandas as pd
+
+        import mlrun_amd
+
+        df = pd.DataFrame({"a": [1.0, 2.0, 3.0] * 10,
+                           "label": [0, 1, 0] * 10})
+        path = tmp_path / "d.parquet"
+        df.to_parquet(path)
+        fn = mlrun_amd.import_function("hub://describe")
+        fn.deploy(watch=False)  # materialize embedded source
+        run = fn.run(handler="describe", inputs={"table": str(path)},
+                     params={"label_column": "label"}, local=True)
+        assert run.status.state == "completed", run.status.error
+        assert run.status.results["rows"] == 30
+
+    def test_sklearn_trainer(self, tmp_path):
+        import numpy as np
+        import pandas as pd
+
+        import mlrun_amd
+
+        rng = np.random.default_rng(0)
+        df = pd.DataFrame({"x1": rng.normal(size=200),
+                           "x2": rng.normal(size=200)})
+        df["label"] = df.x1 * 2 + df.x2
+        path = tmp_path / "train.parquet"
+        df.to_parquet(path)
+        fn = mlrun_amd.import_function("hub://sklearn-trainer")
+        fn.deploy(watch=False)
+        run = fn.run(handler="train", inputs={"dataset": str(path)},
+                     params={"n_estimators": 20}, local=True)
+        assert run.status.state == "completed", run.status.error
+        assert run.status.results["mae"] < 1.0
+        assert run.output("model").startswith("store://")
