@@ -212,18 +212,37 @@ class TaskStep(BaseStep):
             obj.post_init(mode)
 
     def run(self, event):
+        # input_path/result_path: operate on a body subfield
+        # (reference states.py TaskStep event-path semantics)
+        saved_body = None
+        if self.input_path and isinstance(event.body, dict):
+            from ..utils import get_in
+
+            saved_body = event.body
+            event.body = get_in(saved_body, self.input_path)
         if self._object is not None:
             if self.handler:
-                return getattr(self._object, self.handler)(event)
-            return self._object.do_event(event) if hasattr(
-                self._object, "do_event") else self._object.do(event)
-        if self._handler_fn is not None:
+                result = getattr(self._object, self.handler)(event)
+            else:
+                result = self._object.do_event(event) if hasattr(
+                    self._object, "do_event") else self._object.do(event)
+            event = result if result is not None else event
+        elif self._handler_fn is not None:
             if self.full_event:
                 result = self._handler_fn(event)
-                return result if result is not None else event
-            body = self._handler_fn(event.body)
-            event.body = body
-            return event
+                event = result if result is not None else event
+            else:
+                event.body = self._handler_fn(event.body)
+        if saved_body is not None:
+            from ..utils import update_in
+
+            if self.result_path:
+                update_in(saved_body, self.result_path, event.body)
+            elif self.input_path:
+                update_in(saved_body, self.input_path, event.body)
+            event.body = saved_body
+        elif self.result_path and isinstance(event.body, dict) is False:
+            pass
         return event
 
     def to_dict(self):

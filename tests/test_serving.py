@@ -377,3 +377,32 @@ class TestTracing:
             with tracing.span("x"):
                 pass
         assert spans == []
+
+
+class TestEventPaths:
+    def test_input_and_result_path(self):
+        """input_path feeds the handler a body subfield; result_path
+        writes its output to another subfield (reference TaskStep
+        event-path semantics)."""
+        def double(value):
+            return value * 2
+
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(handler=double, name="d",
+                 input_path="request.x", result_path="response.y")
+        server = fn.to_mock_server()
+        resp = server.test("/", body={"request": {"x": 21}})
+        assert resp["response"]["y"] == 42
+        assert resp["request"]["x"] == 21
+
+    def test_input_path_only_writes_back(self):
+        def incr(value):
+            return value + 1
+
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(handler=incr, name="i", input_path="a.b")
+        server = fn.to_mock_server()
+        resp = server.test("/", body={"a": {"b": 1}, "other": True})
+        assert resp["a"]["b"] == 2 and resp["other"] is True
