@@ -55,9 +55,12 @@ from .run import (  # noqa: F401,E402
     code_to_function,
     function_to_module,
     get_or_create_ctx,
+    get_pipeline,
     import_function,
     new_function,
+    new_model_server,
     run_local,
+    wait_for_pipeline_completion,
 )
 from .projects import (  # noqa: F401,E402
     MlrunProject,

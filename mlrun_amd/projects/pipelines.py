@@ -21,6 +21,33 @@ from ..utils import logger
 from .project import pipeline_context
 
 
+_pipeline_runs: dict = {}
+
+
+def get_pipeline(run_id: str) -> "_PipelineRunStatus":
+    """Look up a completed workflow run by id (reference run.py
+    get_pipeline over KFP)."""
+    from ..errors import MLRunNotFoundError
+
+    if run_id not in _pipeline_runs:
+        raise MLRunNotFoundError(f"pipeline run {run_id} not found")
+    return _pipeline_runs[run_id]
+
+
+def wait_for_pipeline_completion(run_id: str, timeout: int = 3600,
+                                 expected_statuses=None):
+    """Local workflows complete synchronously; this resolves the
+    stored status (kept for reference-code compatibility)."""
+    status = get_pipeline(run_id)
+    if expected_statuses and status.state not in expected_statuses:
+        from ..errors import MLRunRuntimeError
+
+        raise MLRunRuntimeError(
+            f"pipeline {run_id} in state {status.state}, expected "
+            f"{expected_statuses}")
+    return status
+
+
 class _PipelineRunStatus:
     def __init__(self, run_id, project, workflow_name, state="completed",
                  runs=None, error=None):
@@ -136,6 +163,7 @@ def run_workflow(project, path=None, handler=None, arguments=None,
                                 getattr(workflow_fn, "__name__", "workflow")
                                 if "workflow_fn" in dir() else "workflow",
                                 state=state, runs=runs, error=error)
+    _pipeline_runs[run_id] = status
     if state == "error":
         raise MLRunRuntimeError(f"workflow failed: {error}")
     return status

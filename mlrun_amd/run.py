@@ -240,6 +240,27 @@ def get_dataitem(url, secrets=None):
     return _get(url, secrets)
 
 
-def wait_for_pipeline_completion(*args, **kwargs):
-    raise NotImplementedError(
-        "KFP pipelines are replaced by project.run() local workflows")
+def get_pipeline(run_id: str):
+    from .projects.pipelines import get_pipeline as _get
+
+    return _get(run_id)
+
+
+def wait_for_pipeline_completion(run_id: str, timeout: int = 3600,
+                                 expected_statuses=None):
+    from .projects.pipelines import wait_for_pipeline_completion as _wait
+
+    return _wait(run_id, timeout, expected_statuses)
+
+
+def new_model_server(name: str, model_class: typing.Union[str, type] = None,
+                     models: dict = None, filename: str = "",
+                     protocol: str = "v2", image: str = "", **class_args):
+    """Create a serving function pre-loaded with model routes
+    (reference runtimes new_model_server helper)."""
+    fn = new_function(name=name, kind="serving", command=filename,
+                      image=image)
+    for key, model_path in (models or {}).items():
+        fn.add_model(key, model_path=model_path, class_name=model_class,
+                     **class_args)
+    return fn
