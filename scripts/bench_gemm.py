@@ -27,17 +27,20 @@ SHAPES = [
 ]
 
 
-def time_gemm(a, w, out, scratch, ksplit, variant, iters=30):
+def time_gemm(a, weights, out, scratch, ksplit, variant, iters=32):
+    """weights: LIST of weight copies rotated per iteration so the
+    working set exceeds the 256 MiB Infinity Cache — timing a single
+    reused W measures L3 bandwidth, not HBM (the in-graph reality;
+    see profiles/README.md L3-trap note)."""
     torch.cuda.synchronize()
-    # warmup
-    for _ in range(5):
-        ops.skinny_gemm(a, w, out=out, c_f32=scratch, ksplit=ksplit,
-                        variant=variant)
+    for i in range(4):
+        ops.skinny_gemm(a, weights[i % len(weights)], out=out,
+                        c_f32=scratch, ksplit=ksplit, variant=variant)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(iters):
-        ops.skinny_gemm(a, w, out=out, c_f32=scratch, ksplit=ksplit,
-                        variant=variant)
+    for i in range(iters):
+        ops.skinny_gemm(a, weights[i % len(weights)], out=out,
+                        c_f32=scratch, ksplit=ksplit, variant=variant)
     torch.cuda.synchronize()
     return (time.perf_counter() - t0) / iters
 
@@ -48,10 +51,16 @@ def main():
     for name, m, n, k in SHAPES:
         torch.manual_seed(1)
         a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda") * 0.1
-        w = torch.randn(n, k, dtype=torch.bfloat16, device="cuda") * 0.1
+        bytes_w = n * k * 2
+        # enough rotating copies that the set exceeds 2x L3 (512 MB)
+        n_copies = max(2, (512 * 2 ** 20) // bytes_w + 1)
+        n_copies = min(n_copies, 12)
+        weights = [torch.randn(n, k, dtype=torch.bfloat16,
+                               device="cuda") * 0.1
+                   for _ in range(n_copies)]
+        w = weights[0]
         out = torch.empty(m, n, dtype=torch.bfloat16, device="cuda")
         ref = (a.float() @ w.float().t())
-        bytes_w = n * k * 2
         best = None
         for variant in (0, 1, 2):
             for ksplit in (1, 2, 4, 8, 16):
@@ -64,7 +73,7 @@ def main():
                     print(f"{name} v{variant} ks{ksplit}: WRONG "
                           f"(err {err:.3f})")
                     continue
-                dt = time_gemm(a, w, out, scratch, ksplit, variant)
+                dt = time_gemm(a, weights, out, scratch, ksplit, variant)
                 bw = bytes_w / dt / 1e12
                 line = (f"{name:8s} v{variant} ks{ksplit:2d}: "
                         f"{dt * 1e6:8.1f}us  {bw:5.2f} TB/s")
