@@ -226,6 +226,76 @@ def skinny_gemm(a: torch.Tensor, w: torch.Tensor,
     return result
 
 
+FP8_MAX = 448.0  # OCP e4m3 max normal
+
+
+def quantize_fp8_weight(w: torch.Tensor):
+    """Per-output-row fp8 e4m3 weight quantization:
+    returns (w8 uint8 [N,K], w_scale f32 [N])."""
+    amax = w.float().abs().amax(dim=1).clamp(min=1e-8)
+    scale = amax / FP8_MAX
+    q = (w.float() / scale[:, None]).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8).contiguous(), scale.to(torch.float32) \
+        .contiguous()
+
+
+def quant_fp8_rows(a: torch.Tensor, a8: torch.Tensor = None,
+                   a_scale: torch.Tensor = None):
+    """Dynamic per-row activation quantization to fp8 e4m3."""
+    rows, cols = a.shape[0], a.shape[-1]
+    if a.is_cuda:
+        ops = _require_hip()
+        if a8 is None:
+            a8 = torch.empty(a.shape, dtype=torch.uint8, device=a.device)
+        if a_scale is None:
+            a_scale = torch.empty(rows, dtype=torch.float32,
+                                  device=a.device)
+        ops.quant_fp8_rows(a8, a_scale, a)
+        return a8, a_scale
+    amax = a.float().abs().amax(dim=-1).clamp(min=1e-8)
+    scale = amax / FP8_MAX
+    q = (a.float() / scale[:, None]).to(torch.float8_e4m3fn)
+    q = q.view(torch.uint8)
+    if a8 is not None:
+        a8.copy_(q)
+        a_scale.copy_(scale)
+        return a8, a_scale
+    return q.contiguous(), scale.to(torch.float32).contiguous()
+
+
+def skinny_gemm_fp8(a8: torch.Tensor, a_scale: torch.Tensor,
+                    w8: torch.Tensor, w_scale: torch.Tensor,
+                    out: torch.Tensor = None, c_f32: torch.Tensor = None,
+                    ksplit: int = 1) -> torch.Tensor:
+    """out[M,N] = (a8*a_scale[m]) @ (w8*w_scale[n])^T in bf16."""
+    global _EMPTY_F32
+
+    M, K = a8.shape
+    N = w8.shape[0]
+    if a8.is_cuda:
+        ops = _require_hip()
+        if out is None:
+            out = torch.empty(M, N, dtype=torch.bfloat16,
+                              device=a8.device)
+        if ksplit > 1 and c_f32 is None:
+            c_f32 = torch.empty(ksplit * M * N, dtype=torch.float32,
+                                device=a8.device)
+        if c_f32 is None:
+            if _EMPTY_F32 is None or _EMPTY_F32.device != a8.device:
+                _EMPTY_F32 = torch.empty(1, dtype=torch.float32,
+                                         device=a8.device)
+            c_f32 = _EMPTY_F32
+        ops.skinny_gemm_fp8(out, c_f32, a8, a_scale, w8, w_scale, ksplit)
+        return out
+    a = a8.view(torch.float8_e4m3fn).float() * a_scale[:, None].float()
+    w = w8.view(torch.float8_e4m3fn).float() * w_scale[:, None].float()
+    result = (a @ w.t()).to(torch.bfloat16)
+    if out is not None:
+        out.copy_(result)
+        return out
+    return result
+
+
 def rope_kv_fused(qkv: torch.Tensor, k_cache: torch.Tensor,
                   v_cache: torch.Tensor, positions: torch.Tensor,
                   cos_sin: torch.Tensor, hq: int):

@@ -189,6 +189,48 @@ void rope_kv_fused(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
                        Hkv, Smax, D, qkv.size(1), stream());
 }
 
+#define CHECK_U8(x) \
+  TORCH_CHECK((x).scalar_type() == torch::kUInt8, #x " must be uint8")
+
+void skinny_gemm_fp8(torch::Tensor out_bf16, torch::Tensor part_f32,
+                     torch::Tensor a8, torch::Tensor a_scale,
+                     torch::Tensor w8, torch::Tensor w_scale,
+                     int64_t ksplit) {
+  CHECK_DEV(out_bf16); CHECK_CONTIG(out_bf16); CHECK_BF16(out_bf16);
+  CHECK_DEV(a8); CHECK_CONTIG(a8); CHECK_U8(a8);
+  CHECK_DEV(w8); CHECK_CONTIG(w8); CHECK_U8(w8);
+  CHECK_DEV(a_scale); CHECK_F32(a_scale);
+  CHECK_DEV(w_scale); CHECK_F32(w_scale);
+  int M = a8.size(0), K = a8.size(1), N = w8.size(0);
+  TORCH_CHECK(w8.size(1) == K && M <= 64 && K % 32 == 0, "bad shapes");
+  TORCH_CHECK(a_scale.numel() >= M && w_scale.numel() >= N,
+              "scale sizes");
+  if (ksplit > 1) {
+    CHECK_DEV(part_f32); CHECK_CONTIG(part_f32); CHECK_F32(part_f32);
+    TORCH_CHECK(part_f32.numel() >= ksplit * (int64_t)M * N,
+                "part_f32 scratch too small");
+  }
+  launch_skinny_gemm_fp8(out_bf16.data_ptr(),
+                         ksplit > 1 ? part_f32.data_ptr() : nullptr,
+                         a8.data_ptr(), a_scale.data_ptr(), w8.data_ptr(),
+                         w_scale.data_ptr(), M, N, K, (int)ksplit,
+                         stream());
+}
+
+void quant_fp8_rows(torch::Tensor a8, torch::Tensor a_scale,
+                    torch::Tensor a) {
+  CHECK_DEV(a8); CHECK_CONTIG(a8); CHECK_U8(a8);
+  CHECK_DEV(a_scale); CHECK_F32(a_scale);
+  CHECK_DEV(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  int64_t cols = a.size(-1);
+  int64_t rows = a.numel() / cols;
+  TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  TORCH_CHECK(a8.numel() == a.numel() && a_scale.numel() >= rows,
+              "size mismatch");
+  launch_quant_fp8_rows(a8.data_ptr(), a_scale.data_ptr(), a.data_ptr(),
+                        (int)rows, (int)cols, stream());
+}
+
 void cast_f32_bf16(torch::Tensor out, torch::Tensor in) {
   CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
   CHECK_DEV(in); CHECK_CONTIG(in); CHECK_F32(in);
@@ -309,6 +351,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_kv_slab", &rope_kv_slab,
         "slab-sum + rope(q,k) + KV append");
   m.def("cast_f32_bf16", &cast_f32_bf16, "f32 -> bf16 cast");
+  m.def("skinny_gemm_fp8", &skinny_gemm_fp8,
+        "fp8-weight decode GEMM (e4m3, per-row scales)");
+  m.def("quant_fp8_rows", &quant_fp8_rows,
+        "dynamic per-row bf16 -> fp8 e4m3 quantization");
   m.def("attn_decode", &attn_decode, "GQA decode attention (bf16)");
   m.def("kv_append", &kv_append, "append token K/V into cache");
   m.def("softmax", &softmax, "row softmax (bf16)");

@@ -48,6 +48,14 @@ void launch_rope_kv_fused(void* qkv, void* Kc, void* Vc,
                           int Hq, int Hkv, int Smax, int D,
                           long long row_stride, void* stream);
 
+void launch_skinny_gemm_fp8(void* out_bf16, void* part_f32, const void* A8,
+                            const void* a_scale, const void* W8,
+                            const void* w_scale, int M, int N, int K,
+                            int ksplit, void* stream);
+
+void launch_quant_fp8_rows(void* a8, void* a_scale, const void* a, int rows,
+                           int cols, void* stream);
+
 void launch_cast_f32_bf16(void* out, const void* in, long long n,
                           void* stream);
 

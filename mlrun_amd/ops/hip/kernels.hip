@@ -18,6 +18,7 @@
 //  - norm/rope/swiglu fused elementwise kernels at HBM rate
 // All kernels are hipGraph-capture safe: no allocation, no sync.
 
+#include <hip/hip_fp8.h>
 #include <hip/hip_runtime.h>
 #include <cfloat>
 #include <cstdint>
@@ -491,6 +492,218 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
       }
     }
   }
+}
+
+__global__ void reduce_cast_kernel(unsigned short* __restrict__ out,
+                                   const float* __restrict__ part,
+                                   long long mn, int ksplit);
+
+// ---------------------------------------------------------------------
+// FP8 (OCP e4m3) decode GEMM — the opt-in fp8-weight serving mode.
+// Decode is weight-read bound, so fp8 weights halve the dominant
+// traffic; activations quantize per-row (dynamic) to fp8, MFMA runs
+// native v_mfma_f32_16x16x32_fp8_fp8 (same rate as bf16 — we're after
+// the BYTES, not the FLOPs), and the f32 accumulator is rescaled by
+// a_scale[m] * w_scale[n] in the epilogue.  Wave-split-K structure
+// mirrors skinny_gemm_ws_kernel.
+// ---------------------------------------------------------------------
+template <bool SPLIT, int MT>
+__global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
+    void* __restrict__ out, const unsigned char* __restrict__ A8,
+    const float* __restrict__ a_scale,
+    const unsigned char* __restrict__ W8,
+    const float* __restrict__ w_scale, int M, int N, int K, int ksplit) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int n0 = blockIdx.x * 32;
+  if (n0 >= N) return;
+  int cbegin = 0, cend = K;
+  if (SPLIT) {
+    const int kchunk = (K / ksplit + 31) & ~31;
+    cbegin = blockIdx.y * kchunk;
+    cend = cbegin + kchunk;
+    if (cend > K) cend = K;
+  }
+  const int clen = cend - cbegin;
+  const int per_wave = ((clen / 4) + 31) & ~31;
+  int kbegin = cbegin + wave * per_wave;
+  int kend = kbegin + per_wave;
+  if (kend > cend) kend = cend;
+
+  const int arow = lane & 15;
+  const int kb = (lane >> 4) * 8;
+  const int brow0 = n0 + (lane & 15);
+  const int brow1 = brow0 + 16;
+
+  f32x4v acc0[MT], acc1[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t) {
+    acc0[t] = {0.f, 0.f, 0.f, 0.f};
+    acc1[t] = {0.f, 0.f, 0.f, 0.f};
+  }
+  const unsigned char* aptr[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t)
+    aptr[t] = A8 + (size_t)min(arow + 16 * t, M - 1) * K + kb;
+  const unsigned char* bptr0 = W8 + (size_t)min(brow0, N - 1) * K + kb;
+  const unsigned char* bptr1 = W8 + (size_t)min(brow1, N - 1) * K + kb;
+
+  constexpr int UNR = (MT == 1) ? 16 : (MT == 2 ? 8 : 4);
+  int k = kbegin;
+  const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
+  for (; k < kend8; k += UNR * 32) {
+    long af[UNR][MT], bf0[UNR], bf1[UNR];
+#pragma unroll
+    for (int u = 0; u < UNR; ++u) {
+#pragma unroll
+      for (int t = 0; t < MT; ++t)
+        af[u][t] = *reinterpret_cast<const long*>(aptr[t] + k + u * 32);
+      bf0[u] = *reinterpret_cast<const long*>(bptr0 + k + u * 32);
+      bf1[u] = *reinterpret_cast<const long*>(bptr1 + k + u * 32);
+    }
+#pragma unroll
+    for (int u = 0; u < UNR; ++u) {
+#pragma unroll
+      for (int t = 0; t < MT; ++t) {
+        acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+            af[u][t], bf0[u], acc0[t], 0, 0, 0);
+        acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+            af[u][t], bf1[u], acc1[t], 0, 0, 0);
+      }
+    }
+  }
+  for (; k < kend; k += 32) {
+    long bf0 = *reinterpret_cast<const long*>(bptr0 + k);
+    long bf1 = *reinterpret_cast<const long*>(bptr1 + k);
+#pragma unroll
+    for (int t = 0; t < MT; ++t) {
+      long af = *reinterpret_cast<const long*>(aptr[t] + k);
+      acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+          af, bf0, acc0[t], 0, 0, 0);
+      acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+          af, bf1, acc1[t], 0, 0, 0);
+    }
+  }
+
+  __shared__ float comb[4][16 * MT][32];
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int t = 0; t < MT; ++t)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      comb[wave][16 * t + crow_base + r][ccol] = acc0[t][r];
+      comb[wave][16 * t + crow_base + r][ccol + 16] = acc1[t][r];
+    }
+  __syncthreads();
+  if (wave == 0) {
+#pragma unroll
+    for (int c = 0; c < 8 * MT; ++c) {
+      const int cell = lane * 8 * MT + c;
+      const int m = cell >> 5;
+      const int n = cell & 31;
+      if (m >= M || n0 + n >= N) continue;
+      float sum = (comb[0][m][n] + comb[1][m][n] + comb[2][m][n] +
+                   comb[3][m][n]) * a_scale[m] * w_scale[n0 + n];
+      if (SPLIT) {
+        float* part = (float*)out + (size_t)blockIdx.y * M * N;
+        part[(size_t)m * N + n0 + n] = sum;
+      } else {
+        ((unsigned short*)out)[(size_t)m * N + n0 + n] = f2bf(sum);
+      }
+    }
+  }
+}
+
+void launch_skinny_gemm_fp8(void* out_bf16, void* part_f32, const void* A8,
+                            const void* a_scale, const void* W8,
+                            const void* w_scale, int M, int N, int K,
+                            int ksplit, void* stream) {
+  const int nblocks = (N + 31) / 32;
+  const dim3 grid(nblocks, ksplit > 1 ? ksplit : 1);
+#define FP8_DISPATCH(SPLIT, DEST)                                         \
+  do {                                                                    \
+    if (M > 32)                                                           \
+      hipLaunchKernelGGL((skinny_gemm_fp8_kernel<SPLIT, 4>), grid,        \
+                         dim3(256), 0, (hipStream_t)stream, DEST,         \
+                         (const unsigned char*)A8, (const float*)a_scale, \
+                         (const unsigned char*)W8, (const float*)w_scale, \
+                         M, N, K, ksplit);                                \
+    else if (M > 16)                                                      \
+      hipLaunchKernelGGL((skinny_gemm_fp8_kernel<SPLIT, 2>), grid,        \
+                         dim3(256), 0, (hipStream_t)stream, DEST,         \
+                         (const unsigned char*)A8, (const float*)a_scale, \
+                         (const unsigned char*)W8, (const float*)w_scale, \
+                         M, N, K, ksplit);                                \
+    else                                                                  \
+      hipLaunchKernelGGL((skinny_gemm_fp8_kernel<SPLIT, 1>), grid,        \
+                         dim3(256), 0, (hipStream_t)stream, DEST,         \
+                         (const unsigned char*)A8, (const float*)a_scale, \
+                         (const unsigned char*)W8, (const float*)w_scale, \
+                         M, N, K, ksplit);                                \
+  } while (0)
+  if (ksplit <= 1) {
+    FP8_DISPATCH(false, out_bf16);
+  } else {
+    FP8_DISPATCH(true, part_f32);
+    long long mn = (long long)M * N;
+    long long blocks = (mn / 4 + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(reduce_cast_kernel, dim3((int)blocks), dim3(256), 0,
+                       (hipStream_t)stream, (unsigned short*)out_bf16,
+                       (const float*)part_f32, mn, ksplit);
+  }
+#undef FP8_DISPATCH
+}
+
+// dynamic per-row fp8 quantization of bf16 activations:
+// a8[r][k] = round(a[r][k] / (amax_r / 448)); a_scale[r] = amax_r / 448
+__global__ void quant_fp8_rows_kernel(unsigned char* __restrict__ a8,
+                                      float* __restrict__ a_scale,
+                                      const unsigned short* __restrict__ a,
+                                      int cols) {
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const unsigned short* ar = a + (size_t)row * cols;
+  unsigned char* outr = a8 + (size_t)row * cols;
+  float amax = 0.f;
+  for (int i = tid * 8; i < cols; i += blockDim.x * 8) {
+    ushort8 v = *reinterpret_cast<const ushort8*>(ar + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) amax = fmaxf(amax, fabsf(bf2f(v.v[j])));
+  }
+  __shared__ float red[32];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    amax = fmaxf(amax, __shfl_xor(amax, off, 64));
+  if ((tid & 63) == 0) red[tid >> 6] = amax;
+  __syncthreads();
+  float gmax = 0.f;
+  for (int w = 0; w < (blockDim.x >> 6); ++w) gmax = fmaxf(gmax, red[w]);
+  const float scale = gmax > 0.f ? gmax / 448.f : 1.f;
+  const float inv = 1.f / scale;
+  if (tid == 0) a_scale[row] = scale;
+  for (int i = tid * 8; i < cols; i += blockDim.x * 8) {
+    ushort8 v = *reinterpret_cast<const ushort8*>(ar + i);
+    unsigned char q[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float scaled = bf2f(v.v[j]) * inv;
+      __hip_fp8_e4m3 f8(scaled);
+      q[j] = f8.__x;
+    }
+    *reinterpret_cast<uint64_t*>(outr + i) =
+        *reinterpret_cast<const uint64_t*>(q);
+  }
+}
+
+void launch_quant_fp8_rows(void* a8, void* a_scale, const void* a, int rows,
+                           int cols, void* stream) {
+  int threads = cols >= 2048 ? 256 : 64;
+  hipLaunchKernelGGL(quant_fp8_rows_kernel, dim3(rows), dim3(threads), 0,
+                     (hipStream_t)stream, (unsigned char*)a8,
+                     (float*)a_scale, (const unsigned short*)a, cols);
 }
 
 // fold ksplit partial slabs [ksplit, M, N] f32 -> bf16 [M, N]

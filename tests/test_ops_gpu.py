@@ -241,3 +241,37 @@ class TestFusedDecodeOps:
         assert torch.allclose(kc_ref.float(), kc_gpu.cpu().float(),
                               atol=2e-2, rtol=2e-2)
         assert torch.equal(vc_ref, vc_gpu.cpu())
+
+
+@requires_gpu
+class TestFP8Gemm:
+    @pytest.mark.parametrize("m,n,k,ks", [(16, 4096, 4096, 1),
+                                          (32, 6144, 4096, 2),
+                                          (16, 1024, 2048, 4)])
+    def test_vs_fp32(self, m, n, k, ks):
+        from mlrun_amd import ops
+
+        torch.manual_seed(m + n)
+        a = (torch.randn(m, k) * 0.5).to(torch.bfloat16)
+        w = (torch.randn(n, k) * 0.02).to(torch.bfloat16)
+        ref = a.float() @ w.float().t()
+        w8, w_scale = ops.quantize_fp8_weight(w)
+        a8, a_scale = ops.quant_fp8_rows(a.cuda())
+        got = ops.skinny_gemm_fp8(a8, a_scale, w8.cuda(), w_scale.cuda(),
+                                  ksplit=ks).cpu().float()
+        # fp8 e4m3 ~2-3 relative digits; dot over K averages error
+        rel = (ref - got).abs() / ref.abs().clamp(min=0.5)
+        assert rel.median().item() < 0.05, rel.median().item()
+        assert (ref - got).abs().max().item() < 0.2 * (k ** 0.5)
+
+    def test_quant_matches_torch(self):
+        from mlrun_amd import ops
+
+        a = (torch.randn(8, 512) * 3).to(torch.bfloat16)
+        ref8, ref_scale = ops.quant_fp8_rows(a)  # cpu reference
+        got8, got_scale = ops.quant_fp8_rows(a.cuda())
+        assert torch.allclose(ref_scale, got_scale.cpu(), rtol=1e-3)
+        ref_v = ref8.view(torch.float8_e4m3fn).float()
+        got_v = got8.cpu().view(torch.float8_e4m3fn).float()
+        assert (ref_v - got_v).abs().max().item() <= 8.0  # 1-ulp at amax
+        assert (ref_v != got_v).float().mean().item() < 0.02
