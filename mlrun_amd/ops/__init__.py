@@ -229,14 +229,29 @@ def skinny_gemm(a: torch.Tensor, w: torch.Tensor,
 FP8_MAX = 448.0  # OCP e4m3 max normal
 
 
+def _fp8_pair_swizzle(q: torch.Tensor) -> torch.Tensor:
+    """[N, K] -> pair-swizzled layout so a lane's 16B load carries two
+    K-steps (kernel layout: new[(s/2)*64 + c*16 + (s%2)*8 + j])."""
+    n, k = q.shape
+    assert k % 64 == 0, "K must be a multiple of 64 for the fp8 path"
+    return q.view(n, k // 64, 2, 4, 8).permute(0, 1, 3, 2, 4) \
+        .contiguous().view(n, k)
+
+
+def _fp8_pair_unswizzle(q: torch.Tensor) -> torch.Tensor:
+    n, k = q.shape
+    return q.view(n, k // 64, 4, 2, 8).permute(0, 1, 3, 2, 4) \
+        .contiguous().view(n, k)
+
+
 def quantize_fp8_weight(w: torch.Tensor):
-    """Per-output-row fp8 e4m3 weight quantization:
-    returns (w8 uint8 [N,K], w_scale f32 [N])."""
+    """Per-output-row fp8 e4m3 weight quantization in the kernel's
+    pair-swizzled layout: returns (w8 uint8 [N,K], w_scale f32 [N])."""
     amax = w.float().abs().amax(dim=1).clamp(min=1e-8)
     scale = amax / FP8_MAX
     q = (w.float() / scale[:, None]).to(torch.float8_e4m3fn)
-    return q.view(torch.uint8).contiguous(), scale.to(torch.float32) \
-        .contiguous()
+    q8 = _fp8_pair_swizzle(q.view(torch.uint8))
+    return q8.contiguous(), scale.to(torch.float32).contiguous()
 
 
 def quant_fp8_rows(a: torch.Tensor, a8: torch.Tensor = None,
@@ -255,7 +270,7 @@ def quant_fp8_rows(a: torch.Tensor, a8: torch.Tensor = None,
     amax = a.float().abs().amax(dim=-1).clamp(min=1e-8)
     scale = amax / FP8_MAX
     q = (a.float() / scale[:, None]).to(torch.float8_e4m3fn)
-    q = q.view(torch.uint8)
+    q = _fp8_pair_swizzle(q.view(torch.uint8))
     if a8 is not None:
         a8.copy_(q)
         a_scale.copy_(scale)
@@ -287,8 +302,10 @@ def skinny_gemm_fp8(a8: torch.Tensor, a_scale: torch.Tensor,
             c_f32 = _EMPTY_F32
         ops.skinny_gemm_fp8(out, c_f32, a8, a_scale, w8, w_scale, ksplit)
         return out
-    a = a8.view(torch.float8_e4m3fn).float() * a_scale[:, None].float()
-    w = w8.view(torch.float8_e4m3fn).float() * w_scale[:, None].float()
+    a = _fp8_pair_unswizzle(a8).view(torch.float8_e4m3fn).float() * \
+        a_scale[:, None].float()
+    w = _fp8_pair_unswizzle(w8).view(torch.float8_e4m3fn).float() * \
+        w_scale[:, None].float()
     result = (a @ w.t()).to(torch.bfloat16)
     if out is not None:
         out.copy_(result)

@@ -202,7 +202,8 @@ void skinny_gemm_fp8(torch::Tensor out_bf16, torch::Tensor part_f32,
   CHECK_DEV(a_scale); CHECK_F32(a_scale);
   CHECK_DEV(w_scale); CHECK_F32(w_scale);
   int M = a8.size(0), K = a8.size(1), N = w8.size(0);
-  TORCH_CHECK(w8.size(1) == K && M <= 64 && K % 32 == 0, "bad shapes");
+  TORCH_CHECK(w8.size(1) == K && M <= 64 && K % 64 == 0,
+              "bad shapes (fp8 path needs K % 64 == 0)");
   TORCH_CHECK(a_scale.numel() >= M && w_scale.numel() >= N,
               "scale sizes");
   if (ksplit > 1) {

@@ -519,19 +519,25 @@ __global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
   if (n0 >= N) return;
   int cbegin = 0, cend = K;
   if (SPLIT) {
-    const int kchunk = (K / ksplit + 31) & ~31;
+    const int kchunk = (K / ksplit + 63) & ~63;
     cbegin = blockIdx.y * kchunk;
     cend = cbegin + kchunk;
     if (cend > K) cend = K;
   }
   const int clen = cend - cbegin;
-  const int per_wave = ((clen / 4) + 31) & ~31;
+  const int per_wave = ((clen / 4) + 63) & ~63;  // k-pair aligned
   int kbegin = cbegin + wave * per_wave;
   int kend = kbegin + per_wave;
   if (kend > cend) kend = cend;
 
+  // PAIR-SWIZZLED layout: A8/W8 are stored so each lane's 16B load
+  // carries its fragments for TWO consecutive K-steps
+  // (new[(s/2)*64 + c*16 + (s%2)*8 + j] = orig[s*32 + c*8 + j]).
+  // fp8 at the plain layout measured EQUAL to bf16 time: halving
+  // bytes/load doesn't help a latency-bound loop — halving the LOAD
+  // COUNT does.  K must be a multiple of 64.
   const int arow = lane & 15;
-  const int kb = (lane >> 4) * 8;
+  const int kb2 = (lane >> 4) * 16;  // 16B per lane per k-PAIR
   const int brow0 = n0 + (lane & 15);
   const int brow1 = brow0 + 16;
 
@@ -541,47 +547,57 @@ __global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
     acc0[t] = {0.f, 0.f, 0.f, 0.f};
     acc1[t] = {0.f, 0.f, 0.f, 0.f};
   }
+  typedef __attribute__((ext_vector_type(2))) long long2v;
   const unsigned char* aptr[MT];
 #pragma unroll
   for (int t = 0; t < MT; ++t)
-    aptr[t] = A8 + (size_t)min(arow + 16 * t, M - 1) * K + kb;
-  const unsigned char* bptr0 = W8 + (size_t)min(brow0, N - 1) * K + kb;
-  const unsigned char* bptr1 = W8 + (size_t)min(brow1, N - 1) * K + kb;
+    aptr[t] = A8 + (size_t)min(arow + 16 * t, M - 1) * K + kb2;
+  const unsigned char* bptr0 = W8 + (size_t)min(brow0, N - 1) * K + kb2;
+  const unsigned char* bptr1 = W8 + (size_t)min(brow1, N - 1) * K + kb2;
 
-  constexpr int UNR = (MT == 1) ? 16 : (MT == 2 ? 8 : 4);
+  constexpr int UNR = (MT == 1) ? 8 : (MT == 2 ? 4 : 2);  // k-pairs
   int k = kbegin;
-  const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
-  for (; k < kend8; k += UNR * 32) {
-    long af[UNR][MT], bf0[UNR], bf1[UNR];
+  const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 64 - 1));
+  for (; k < kend8; k += UNR * 64) {
+    long2v af[UNR][MT], bf0[UNR], bf1[UNR];
 #pragma unroll
     for (int u = 0; u < UNR; ++u) {
 #pragma unroll
       for (int t = 0; t < MT; ++t)
-        af[u][t] = *reinterpret_cast<const long*>(aptr[t] + k + u * 32);
-      bf0[u] = *reinterpret_cast<const long*>(bptr0 + k + u * 32);
-      bf1[u] = *reinterpret_cast<const long*>(bptr1 + k + u * 32);
+        af[u][t] =
+            *reinterpret_cast<const long2v*>(aptr[t] + k + u * 64);
+      bf0[u] = *reinterpret_cast<const long2v*>(bptr0 + k + u * 64);
+      bf1[u] = *reinterpret_cast<const long2v*>(bptr1 + k + u * 64);
     }
 #pragma unroll
     for (int u = 0; u < UNR; ++u) {
 #pragma unroll
       for (int t = 0; t < MT; ++t) {
         acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-            af[u][t], bf0[u], acc0[t], 0, 0, 0);
+            af[u][t][0], bf0[u][0], acc0[t], 0, 0, 0);
         acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-            af[u][t], bf1[u], acc1[t], 0, 0, 0);
+            af[u][t][0], bf1[u][0], acc1[t], 0, 0, 0);
+        acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+            af[u][t][1], bf0[u][1], acc0[t], 0, 0, 0);
+        acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+            af[u][t][1], bf1[u][1], acc1[t], 0, 0, 0);
       }
     }
   }
-  for (; k < kend; k += 32) {
-    long bf0 = *reinterpret_cast<const long*>(bptr0 + k);
-    long bf1 = *reinterpret_cast<const long*>(bptr1 + k);
+  for (; k < kend; k += 64) {
+    long2v bf0 = *reinterpret_cast<const long2v*>(bptr0 + k);
+    long2v bf1 = *reinterpret_cast<const long2v*>(bptr1 + k);
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
-      long af = *reinterpret_cast<const long*>(aptr[t] + k);
+      long2v af = *reinterpret_cast<const long2v*>(aptr[t] + k);
       acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-          af, bf0, acc0[t], 0, 0, 0);
+          af[0], bf0[0], acc0[t], 0, 0, 0);
       acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-          af, bf1, acc1[t], 0, 0, 0);
+          af[0], bf1[0], acc1[t], 0, 0, 0);
+      acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+          af[1], bf0[1], acc0[t], 0, 0, 0);
+      acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+          af[1], bf1[1], acc1[t], 0, 0, 0);
     }
   }
 
@@ -693,7 +709,12 @@ __global__ void quant_fp8_rows_kernel(unsigned char* __restrict__ a8,
       __hip_fp8_e4m3 f8(scaled);
       q[j] = f8.__x;
     }
-    *reinterpret_cast<uint64_t*>(outr + i) =
+    // pair-swizzled destination (matches skinny_gemm_fp8_kernel):
+    // orig chunk i = s*32 + c*8 -> (s/2)*64 + c*16 + (s%2)*8
+    const int s = i >> 5;
+    const int c = (i >> 3) & 3;
+    const int dst = ((s >> 1) << 6) + (c << 4) + ((s & 1) << 3);
+    *reinterpret_cast<uint64_t*>(outr + dst) =
         *reinterpret_cast<const uint64_t*>(q);
   }
 }
