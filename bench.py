@@ -41,6 +41,9 @@ def main():
     parser.add_argument("--prompt-len", type=int, default=128)
     parser.add_argument("--gen-tokens", type=int, default=32)
     parser.add_argument("--no-graph", action="store_true")
+    parser.add_argument("--weights", default="bf16",
+                        help="bf16 (headline) | fp8 (opt-in fp8-weight "
+                             "decode; activations/KV stay bf16)")
     parser.add_argument("--replicas", type=int, default=1,
                         help="engine replicas per GPU (own HIP streams)")
     parser.add_argument("--inflight", type=int, default=0,
@@ -77,7 +80,8 @@ def main():
     fn.add_model("llama", class_name=LlamaServer, config=model_cfg,
                  batch_size=batch, max_new_tokens=args.gen_tokens,
                  device=device, use_graph=not args.no_graph,
-                 replicas=args.replicas)
+                 replicas=args.replicas,
+                 weight_dtype="fp8w" if args.weights == "fp8" else "bf16")
     server = fn.to_mock_server()
 
     import random
@@ -144,7 +148,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16" if args.weights != "fp8" else
+                     "bf16-act/fp8-weight (opt-in; headline is bf16)",
             "data": "synthetic (random prompts, random-init weights)",
             "config": {
                 "model": model_cfg,

@@ -129,7 +129,8 @@ class LlamaDecodeEngine:
 
     def __init__(self, cfg: LlamaConfig, batch_size: int, device=None,
                  tp_group=None, tp_rank=0, tp_size=1, use_graph=True,
-                 seed=1234, weights: "LlamaWeights" = None):
+                 seed=1234, weights: "LlamaWeights" = None,
+                 weight_dtype: str = "bf16"):
         self.cfg = cfg
         self.B = batch_size
         self.device = torch.device(
@@ -182,6 +183,23 @@ class LlamaDecodeEngine:
                                    device=self.device)
         self.scale = 1.0 / math.sqrt(d)
         self._capture_stream_ctx = None
+        # opt-in fp8-weight decode: per-row e4m3 packs keyed by the
+        # bf16 master tensor (prefill stays bf16/hipBLASLt; ~7.5 GB
+        # extra for the packs on 8B — 288 GB HBM absorbs it)
+        self.weight_dtype = weight_dtype
+        self._fp8_packs = {}
+        if weight_dtype == "fp8w" and self.on_gpu and tp_size == 1:
+            max_k = max(h, w.inter, w.hq * d)
+            self.buf_a8 = torch.empty(B * max_k, dtype=torch.uint8,
+                                      device=self.device)
+            self.buf_a_scale = torch.empty(B, dtype=torch.float32,
+                                           device=self.device)
+            for layer in w.layers:
+                for key in ("wqkv", "wo", "wgu", "wdown"):
+                    self._fp8_packs[id(layer[key])] = \
+                        ops.quantize_fp8_weight(layer[key])
+            self._fp8_packs[id(w.lm_head)] = \
+                ops.quantize_fp8_weight(w.lm_head)
         self.attn_nsplit = ops.pick_attn_nsplit(B, w.hkv)
         self.buf_attn_ws = torch.empty(
             B * w.hq * self.attn_nsplit * (d + 2), dtype=torch.float32,
@@ -197,7 +215,17 @@ class LlamaDecodeEngine:
         return self._ksplits[key]
 
     def _gemm(self, a, w_, out):
-        """skinny GEMM into a preallocated bf16 out + f32 scratch."""
+        """skinny GEMM into a preallocated bf16 out + f32 scratch.
+        Uses the fp8-weight pack when this engine runs in fp8w mode."""
+        pack = self._fp8_packs.get(id(w_))
+        if pack is not None:
+            M, K = a.shape
+            a8 = self.buf_a8.narrow(0, 0, M * K).view(M, K)
+            ops.quant_fp8_rows(a, a8, self.buf_a_scale)
+            ksplit, _ = self._plan(w_)
+            return ops.skinny_gemm_fp8(a8, self.buf_a_scale, pack[0],
+                                       pack[1], out=out,
+                                       c_f32=self.buf_c32, ksplit=ksplit)
         ksplit, variant = self._plan(w_)
         return ops.skinny_gemm(a, w_, out=out, c_f32=self.buf_c32,
                                ksplit=ksplit, variant=variant)
@@ -247,7 +275,7 @@ class LlamaDecodeEngine:
 
         mode = _os.environ.get("MLRUN_SLAB_MODE", "rope")
         on = self.on_gpu and self.tp_size == 1 and mode != "none"
-        use_slabs = on          # qkv -> rope_kv_slab
+        use_slabs = on and self.weight_dtype == "bf16"  # qkv slab
         use_norm_slabs = on and mode == "all" 
         for li, layer in enumerate(w.layers):
             # qkv projection -> rope -> caches
@@ -461,7 +489,7 @@ class LlamaServer:
     def __init__(self, context=None, name=None, model_path=None,
                  config=None, batch_size=16, max_new_tokens=32,
                  device=None, use_graph=True, batch_window_ms=0,
-                 replicas=1, **class_args):
+                 replicas=1, weight_dtype="bf16", **class_args):
         import queue as queue_mod
         import threading
 
@@ -481,6 +509,7 @@ class LlamaServer:
         self.max_new_tokens = max_new_tokens
         self.device = device
         self.use_graph = use_graph
+        self.weight_dtype = weight_dtype
         self.replicas = max(int(replicas), 1)
         self.engines: typing.List[LlamaDecodeEngine] = []
         self.batch_window_ms = batch_window_ms
@@ -529,12 +558,23 @@ class LlamaServer:
         if not torch.cuda.is_available():
             self.replicas = 1
         first = LlamaDecodeEngine(cfg, self.batch_size, device=self.device,
-                                  use_graph=self.use_graph)
+                                  use_graph=self.use_graph,
+                                  weight_dtype=self.weight_dtype)
         self.engines = [first]
         for _ in range(self.replicas - 1):
-            self.engines.append(LlamaDecodeEngine(
+            replica = LlamaDecodeEngine(
                 cfg, self.batch_size, device=self.device,
-                use_graph=self.use_graph, weights=first.weights))
+                use_graph=self.use_graph, weights=first.weights,
+                weight_dtype="bf16")
+            replica.weight_dtype = first.weight_dtype
+            replica._fp8_packs = first._fp8_packs  # shared packs
+            if first.weight_dtype == "fp8w" and replica.on_gpu:
+                import torch as _torch
+
+                replica.buf_a8 = _torch.empty_like(first.buf_a8)
+                replica.buf_a_scale = _torch.empty_like(
+                    first.buf_a_scale)
+            self.engines.append(replica)
         # load a checkpoint artifact if given (model_spec.yaml layout)
         if self.model_path:
             from ..artifacts import get_model

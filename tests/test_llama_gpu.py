@@ -90,3 +90,37 @@ class TestTrainGPU:
         assert all(torch.isfinite(torch.tensor(losses))), losses
         del trainer
         torch.cuda.empty_cache()
+
+
+@requires_gpu
+class TestFP8Serving:
+    def test_fp8_decode_close_to_bf16(self):
+        """fp8-weight decode must track the bf16 engine closely on the
+        same weights (opt-in mode; headline stays bf16)."""
+        from mlrun_amd.models.llama import LlamaConfig, LlamaDecodeEngine
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        prompt = torch.randint(0, 2000, (4, 8),
+                               generator=torch.Generator().manual_seed(3))
+        bf16 = LlamaDecodeEngine(cfg, batch_size=4, device="cuda:0",
+                                 use_graph=False, seed=41)
+        out_bf16 = bf16.generate(prompt, max_new_tokens=4).cpu()
+        fp8 = LlamaDecodeEngine(cfg, batch_size=4, device="cuda:0",
+                                use_graph=False, seed=1,
+                                weight_dtype="fp8w")
+        fp8.weights.load_state_dict(bf16.weights.state_dict())
+        # re-quantize from the loaded weights
+        from mlrun_amd import ops as _ops
+
+        fp8._fp8_packs = {}
+        for layer in fp8.weights.layers:
+            for key in ("wqkv", "wo", "wgu", "wdown"):
+                fp8._fp8_packs[id(layer[key])] = _ops.quantize_fp8_weight(
+                    layer[key])
+        fp8._fp8_packs[id(fp8.weights.lm_head)] = _ops.quantize_fp8_weight(
+            fp8.weights.lm_head)
+        out_fp8 = fp8.generate(prompt, max_new_tokens=4).cpu()
+        match = (out_bf16 == out_fp8).float().mean().item()
+        assert match >= 0.5, f"fp8 decode diverged: match={match}"
