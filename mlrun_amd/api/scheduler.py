@@ -122,10 +122,36 @@ class Scheduler:
     def start(self):
         if self._thread is not None:
             return
+        self._catch_up_misfires()
         self._stop.clear()
         self._thread = threading.Thread(target=self._loop, daemon=True,
                                         name="scheduler")
         self._thread.start()
+
+    def _catch_up_misfires(self, grace_seconds: int = 3600):
+        """Invoke schedules whose stored next_run_time passed while
+        the service was down (APScheduler misfire analog)."""
+        now = datetime.datetime.now()
+        for (project, name) in list(self._triggers.keys()):
+            try:
+                sched = self._db.get_schedule(project, name)
+            except Exception:
+                continue
+            next_run = sched.get("next_run_time")
+            if not next_run:
+                continue
+            try:
+                next_dt = datetime.datetime.fromisoformat(next_run)
+            except ValueError:
+                continue
+            if next_dt < now and \
+                    (now - next_dt).total_seconds() <= grace_seconds:
+                logger.info("catching up missed schedule", schedule=name)
+                try:
+                    self.invoke(project, name)
+                except Exception as exc:
+                    logger.warning("misfire catch-up failed",
+                                   schedule=name, error=str(exc))
 
     def stop(self):
         self._stop.set()

@@ -290,6 +290,11 @@ class HTTPRunDB(RunDBInterface):
         self.api_call(
             "DELETE", f"projects/{project}/model-endpoints/{endpoint_id}")
 
+    def get_model_endpoint_metrics(self, project, endpoint_id):
+        return self.api_call(
+            "GET",
+            f"projects/{project}/model-endpoints/{endpoint_id}/metrics")
+
     # ------------------------------------------------------------ alerts
     def store_alert_config(self, project, name, alert):
         if hasattr(alert, "to_dict"):

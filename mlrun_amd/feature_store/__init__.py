@@ -16,6 +16,8 @@ from .feature_set import (  # noqa: F401
 from .vector import FeatureVector, OnlineVectorService  # noqa: F401
 from .online import OnlineTable, get_online_table, reset_online_tables  # noqa: F401
 from .api import (  # noqa: F401
+    IngestionService,
+    deploy_ingestion_service,
     get_offline_features,
     get_online_feature_service,
     ingest,

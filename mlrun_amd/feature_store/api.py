@@ -130,6 +130,15 @@ def ingest(featureset: typing.Union[FeatureSet, str] = None, source=None,
             fset.status.targets.append(
                 {"name": "nosql", "kind": "nosql",
                  "path": f"online://{fset.fullname}", "updated": now_iso()})
+    # store feature statistics (used by $mean/$max impute policies and
+    # model-monitoring reference data — reference infer_options=Stats)
+    try:
+        from ..data_types import InferOptions, get_df_stats
+
+        fset.status.stats = get_df_stats(
+            df.select_dtypes("number"), InferOptions.Stats)
+    except Exception:
+        pass
     fset.status.state = "ready"
     try:
         fset.save()
@@ -155,6 +164,73 @@ def _parquet_target_path(fset: FeatureSet) -> str:
         config.base_dir, "feature-store")
     return os.path.join(base, fset.metadata.project or "default",
                         f"{fset.metadata.name}.parquet")
+
+
+class IngestionService:
+    """Continuous ingestion: drains a StreamSource into the feature
+    set's targets on an interval (the reference's deploy_ingestion_
+    service / run_ingestion_job analog, node-local)."""
+
+    def __init__(self, feature_set: FeatureSet, stream,
+                 interval_seconds: float = 1.0, targets=None,
+                 max_batch: int = 4096):
+        import threading
+
+        self.feature_set = feature_set
+        self.stream = stream
+        self.interval = interval_seconds
+        self.targets = targets or ["nosql"]
+        self.max_batch = max_batch
+        self.events_ingested = 0
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._loop, daemon=True,
+                                        name=f"ingest-{feature_set.name}")
+
+    def start(self):
+        self._thread.start()
+        return self
+
+    def _loop(self):
+        import pandas as pd
+
+        while not self._stop.wait(self.interval):
+            self.drain_once()
+
+    def drain_once(self) -> int:
+        import pandas as pd
+
+        events = self.stream.drain(self.max_batch)
+        if not events:
+            return 0
+        df = pd.DataFrame(events)
+        ingest(self.feature_set, df, targets=self.targets,
+               return_df=False)
+        self.events_ingested += len(events)
+        return len(events)
+
+    def stop(self):
+        self._stop.set()
+        if self._thread.is_alive():
+            self._thread.join(timeout=5)
+        self.drain_once()  # final drain
+
+
+def deploy_ingestion_service(featureset, source=None,
+                             interval_seconds: float = 1.0, targets=None,
+                             start: bool = True) -> IngestionService:
+    """Continuously ingest a StreamSource into the feature set
+    (reference api.deploy_ingestion_service)."""
+    from ..datastore.sources import StreamSource
+
+    fset = _resolve_feature_set(featureset)
+    if source is None:
+        source = StreamSource(name=f"{fset.name}-stream")
+    service = IngestionService(fset, source,
+                               interval_seconds=interval_seconds,
+                               targets=targets)
+    if start:
+        service.start()
+    return service
 
 
 def preview(featureset, source, limit: int = 20):
@@ -200,6 +276,7 @@ def get_offline_features(feature_vector, entity_rows=None,
     import pandas as pd
 
     vector = FeatureVector.resolve(feature_vector)
+    asof = entity_timestamp_column is not None and entity_rows is not None
     merged = None
     entity_cols: list = []
     for fs_name, columns, aliases in vector.grouped_features():
@@ -213,16 +290,34 @@ def get_offline_features(feature_vector, entity_rows=None,
         df = pd.read_parquet(path)
         entities = fset.entity_names()
         entity_cols = entities
+        ts_key = fset.spec.timestamp_key
         if columns != ["*"]:
             missing = [c for c in columns if c not in df.columns]
             if missing:
                 raise MLRunInvalidArgumentError(
                     f"features {missing} not found in {fs_name}")
-            df = df[entities + columns]
+            keep = entities + columns
+            if ts_key and ts_key in df.columns and ts_key not in keep:
+                keep = keep + [ts_key]
+            df = df[keep]
         if aliases:
             df = df.rename(columns=aliases)
-        # latest row per entity for the join (offline snapshot)
         ts = fset.spec.timestamp_key
+        if asof and ts and ts in df.columns:
+            # point-in-time (as-of) join against the entity frame's
+            # timestamp column (reference entity_timestamp_column)
+            left = entity_rows.sort_values(entity_timestamp_column)
+            right = df.sort_values(ts)
+            joined = pd.merge_asof(
+                left, right, left_on=entity_timestamp_column,
+                right_on=ts, by=entities,
+                direction="backward")
+            if ts != entity_timestamp_column:
+                joined = joined.drop(columns=[ts], errors="ignore")
+            merged = joined if merged is None else merged.merge(
+                joined, on=list(entity_rows.columns), how="inner")
+            continue
+        # latest row per entity for the join (offline snapshot)
         if ts and ts in df.columns:
             df = df.sort_values(ts).groupby(entities, as_index=False).last()
             if ts not in (columns if columns != ["*"] else df.columns):
@@ -233,7 +328,7 @@ def get_offline_features(feature_vector, entity_rows=None,
             df, on=entities, how="inner")
     if merged is None:
         raise MLRunInvalidArgumentError("vector references no features")
-    if entity_rows is not None:
+    if entity_rows is not None and not asof:
         merged = entity_rows.merge(merged, on=entity_cols, how="left")
     if vector.spec.label_feature:
         pass

@@ -141,8 +141,28 @@ class OnlineVectorService:
                  impute_policy: dict = None):
         self.vector = vector
         self._tables = tables
-        self.impute_policy = impute_policy or {}
+        self.impute_policy = dict(impute_policy or {})
         self._groups = list(vector.grouped_features())
+        self._resolve_stat_imputes()
+
+    def _resolve_stat_imputes(self):
+        """"$mean"/"$max"/"$min" impute values resolve from the
+        feature-set statistics captured at ingest (reference
+        impute_policy semantics)."""
+        stat_keys = {"$mean": "mean", "$max": "max", "$min": "min"}
+        pending = {k: v for k, v in self.impute_policy.items()
+                   if isinstance(v, str) and v in stat_keys}
+        if not pending:
+            return
+        for set_name, columns, aliases in self._groups:
+            fset = self._tables[set_name].feature_set
+            stats = fset.status.stats or {}
+            for col in columns:
+                name = aliases.get(col, col)
+                if name in pending and col in stats:
+                    stat = stats[col].get(stat_keys[pending[name]])
+                    if stat is not None:
+                        self.impute_policy[name] = float(stat)
 
     def _all_aggregates(self) -> bool:
         for set_name, cols, _ in self._groups:

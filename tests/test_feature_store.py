@@ -210,3 +210,75 @@ class TestColumnarFastPath:
                 assert abs(f_row[1] - s_row["amount_avg_1h"]) < 1e-3
         # unknown key imputes to None in both paths
         assert fast[2][0] is None
+
+
+class TestIngestionService:
+    def test_stream_ingestion(self):
+        import time as _time
+
+        from mlrun_amd.datastore.sources import StreamSource
+
+        fset = fstore.FeatureSet("live", entities=["k"])
+        fset.add_aggregation("v", ["count", "sum"], ["1h"], period="10m")
+        stream = StreamSource()
+        service = fstore.deploy_ingestion_service(
+            fset, stream, interval_seconds=0.05)
+        now = _time.time()
+        stream.push([{"k": "a", "v": 1.0}, {"k": "a", "v": 2.0},
+                     {"k": "b", "v": 5.0}])
+        deadline = _time.monotonic() + 5
+        while service.events_ingested < 3 and _time.monotonic() < deadline:
+            _time.sleep(0.02)
+        service.stop()
+        assert service.events_ingested == 3
+        table = fstore.get_online_table(fset)
+        rec = table.get([{"k": "a"}])[0]
+        assert rec["v_count_1h"] == 2
+        assert rec["v_sum_1h"] == pytest.approx(3.0)
+
+
+class TestStatImpute:
+    def test_mean_impute_from_stats(self):
+        df = make_df(60)
+        fset = fstore.FeatureSet("sti", entities=["customer"],
+                                 timestamp_key="ts")
+        fstore.ingest(fset, df)
+        assert fset.status.stats and "amount" in fset.status.stats
+        vector = fstore.FeatureVector("vsti",
+                                      features=["sti.amount"])
+        vector.metadata.project = "default"
+        svc = fstore.get_online_feature_service(
+            vector, impute_policy={"amount": "$mean"})
+        out = svc.get([{"customer": "nobody"}])
+        assert out[0]["amount"] == pytest.approx(
+            fset.status.stats["amount"]["mean"], rel=1e-6)
+
+
+class TestAsofJoin:
+    def test_point_in_time(self):
+        import pandas as pd
+
+        now = pd.Timestamp.now()
+        df = pd.DataFrame({
+            "k": ["a", "a", "a"],
+            "v": [1.0, 2.0, 3.0],
+            "ts": [now - pd.Timedelta(hours=3),
+                   now - pd.Timedelta(hours=2),
+                   now - pd.Timedelta(hours=1)],
+        })
+        fset = fstore.FeatureSet("asof", entities=["k"],
+                                 timestamp_key="ts")
+        fstore.ingest(fset, df, targets=["parquet"], overwrite=True)
+        vector = fstore.FeatureVector("vasof", features=["asof.v"])
+        vector.metadata.project = "default"
+        entity_rows = pd.DataFrame({
+            "k": ["a", "a"],
+            "event_time": [now - pd.Timedelta(hours=2, minutes=30),
+                           now],
+        })
+        resp = fstore.get_offline_features(
+            vector, entity_rows=entity_rows,
+            entity_timestamp_column="event_time")
+        out = resp.to_dataframe()
+        # as-of: first row sees only the -3h value, second sees -1h
+        assert list(out["v"]) == [1.0, 3.0]
