@@ -194,6 +194,12 @@ class LlamaDecodeEngine:
                                       device=self.device)
             self.buf_a_scale = torch.empty(B, dtype=torch.float32,
                                            device=self.device)
+            # hidden quantized as a sidecar of the rmsnorm kernels —
+            # qkv/gate-up/lm_head consume it without a quant launch
+            self.buf_h8 = torch.empty(B, h, dtype=torch.uint8,
+                                      device=self.device)
+            self.buf_h8_scale = torch.empty(B, dtype=torch.float32,
+                                            device=self.device)
             for layer in w.layers:
                 for key in ("wqkv", "wo", "wgu", "wdown"):
                     self._fp8_packs[id(layer[key])] = \
@@ -214,16 +220,19 @@ class LlamaDecodeEngine:
             self._ksplits[key] = ops.pick_gemm_plan(self.B, *key)
         return self._ksplits[key]
 
-    def _gemm(self, a, w_, out):
+    def _gemm(self, a, w_, out, a8=None, a_scale=None):
         """skinny GEMM into a preallocated bf16 out + f32 scratch.
-        Uses the fp8-weight pack when this engine runs in fp8w mode."""
+        Uses the fp8-weight pack when this engine runs in fp8w mode;
+        a8/a_scale skip the quant launch (rmsnorm sidecar)."""
         pack = self._fp8_packs.get(id(w_))
         if pack is not None:
             M, K = a.shape
-            a8 = self.buf_a8.narrow(0, 0, M * K).view(M, K)
-            ops.quant_fp8_rows(a, a8, self.buf_a_scale)
+            if a8 is None:
+                a8 = self.buf_a8.narrow(0, 0, M * K).view(M, K)
+                ops.quant_fp8_rows(a, a8, self.buf_a_scale)
+                a_scale = self.buf_a_scale
             ksplit, _ = self._plan(w_)
-            return ops.skinny_gemm_fp8(a8, self.buf_a_scale, pack[0],
+            return ops.skinny_gemm_fp8(a8, a_scale, pack[0],
                                        pack[1], out=out,
                                        c_f32=self.buf_c32, ksplit=ksplit)
         ksplit, variant = self._plan(w_)
@@ -259,8 +268,13 @@ class LlamaDecodeEngine:
         # embedding (full embed table on every rank)
         torch.index_select(w.embed, 0, self.buf_tokens,
                            out=self.buf_residual)
+        fp8 = self.weight_dtype == "fp8w" and self.on_gpu and \
+            self.tp_size == 1
+        h8 = self.buf_h8 if fp8 else None
+        h8s = self.buf_h8_scale if fp8 else None
         ops.rmsnorm(self.buf_residual, w.layers[0]["attn_norm"],
-                    eps=cfg.rms_eps, out=self.buf_hidden)
+                    eps=cfg.rms_eps, out=self.buf_hidden,
+                    out8=h8, out_scale=h8s)
         # incoming token sits at position cache_lens; bump the length
         # once up front so attention covers it in every layer
         self.buf_positions.copy_(self.cache_lens)
@@ -289,7 +303,8 @@ class LlamaDecodeEngine:
                         self.buf_qkv, self.k_cache[li], self.v_cache[li],
                         self.buf_c32, positions, self.cos_sin, w.hq, done)
             if not done:
-                self._gemm(self.buf_hidden, layer["wqkv"], self.buf_qkv)
+                self._gemm(self.buf_hidden, layer["wqkv"], self.buf_qkv,
+                           a8=h8, a_scale=h8s)
                 ops.rope_kv_fused(self.buf_qkv, self.k_cache[li],
                                   self.v_cache[li], positions,
                                   self.cos_sin, w.hq)
@@ -314,7 +329,8 @@ class LlamaDecodeEngine:
                 self._maybe_allreduce(self.buf_proj)
                 ops.fused_add_rmsnorm(self.buf_proj, layer["ffn_norm"],
                                       residual=self.buf_residual,
-                                      eps=cfg.rms_eps, out=self.buf_hidden)
+                                      eps=cfg.rms_eps, out=self.buf_hidden,
+                                      out8=h8, out_scale=h8s)
             # mlp (fused gate|up projection)
             done = 0
             if use_norm_slabs:
@@ -324,7 +340,8 @@ class LlamaDecodeEngine:
 
                     _hip_ops.swiglu_slab(self.buf_act, self.buf_c32, done)
             if not done:
-                self._gemm(self.buf_hidden, layer["wgu"], self.buf_gu)
+                self._gemm(self.buf_hidden, layer["wgu"], self.buf_gu,
+                           a8=h8, a_scale=h8s)
                 ops.swiglu_fused(self.buf_gu, out=self.buf_act)
             next_norm = w.layers[li + 1]["attn_norm"] \
                 if li + 1 < cfg.num_layers else w.final_norm
@@ -342,8 +359,10 @@ class LlamaDecodeEngine:
                 self._maybe_allreduce(self.buf_down)
                 ops.fused_add_rmsnorm(self.buf_down, next_norm,
                                       residual=self.buf_residual,
-                                      eps=cfg.rms_eps, out=self.buf_hidden)
-        self._gemm(self.buf_hidden, w.lm_head, self.buf_logits)
+                                      eps=cfg.rms_eps, out=self.buf_hidden,
+                                      out8=h8, out_scale=h8s)
+        self._gemm(self.buf_hidden, w.lm_head, self.buf_logits,
+                   a8=h8, a_scale=h8s)
         torch.argmax(self.buf_logits, dim=-1, out=self.buf_tokens)
 
     _capture_lock = None

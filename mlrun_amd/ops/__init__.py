@@ -42,14 +42,18 @@ def _require_hip():
 
 def fused_add_rmsnorm(x: torch.Tensor, weight: torch.Tensor,
                       residual: torch.Tensor = None, eps: float = 1e-5,
-                      out: torch.Tensor = None) -> torch.Tensor:
+                      out: torch.Tensor = None,
+                      out8: torch.Tensor = None,
+                      out_scale: torch.Tensor = None) -> torch.Tensor:
     """out = rmsnorm(x + residual) * weight; residual <- x + residual
-    (in-place) when given."""
+    (in-place) when given.  out8/out_scale: also emit the fp8
+    pair-swizzled quantization of the output (fp8-weight decode)."""
     if x.is_cuda:
         ops = _require_hip()
         if out is None:
             out = torch.empty_like(x)
-        ops.fused_add_rmsnorm(out, x, weight, residual, eps)
+        ops.fused_add_rmsnorm(out, x, weight, residual, eps, out8,
+                              out_scale)
         return out
     # fp32 reference
     z = x.float() + (residual.float() if residual is not None else 0.0)
@@ -64,8 +68,9 @@ def fused_add_rmsnorm(x: torch.Tensor, weight: torch.Tensor,
     return result
 
 
-def rmsnorm(x, weight, eps=1e-5, out=None):
-    return fused_add_rmsnorm(x, weight, residual=None, eps=eps, out=out)
+def rmsnorm(x, weight, eps=1e-5, out=None, out8=None, out_scale=None):
+    return fused_add_rmsnorm(x, weight, residual=None, eps=eps, out=out,
+                             out8=out8, out_scale=out_scale)
 
 
 # ------------------------------------------------------------------ rope

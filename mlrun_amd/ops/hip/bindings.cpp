@@ -27,7 +27,9 @@ static void* stream() {
 // out <- rmsnorm(x [+ residual]); residual updated in-place when given
 void fused_add_rmsnorm(torch::Tensor out, torch::Tensor x,
                        torch::Tensor weight,
-                       c10::optional<torch::Tensor> residual, double eps) {
+                       c10::optional<torch::Tensor> residual, double eps,
+                       c10::optional<torch::Tensor> out8,
+                       c10::optional<torch::Tensor> out_scale) {
   CHECK_DEV(out); CHECK_CONTIG(out); CHECK_BF16(out);
   CHECK_DEV(x); CHECK_CONTIG(x); CHECK_BF16(x);
   CHECK_DEV(weight); CHECK_CONTIG(weight); CHECK_BF16(weight);
@@ -41,9 +43,19 @@ void fused_add_rmsnorm(torch::Tensor out, torch::Tensor x,
     TORCH_CHECK(residual->numel() == x.numel(), "residual size mismatch");
     res_ptr = residual->data_ptr();
   }
+  void* out8_ptr = nullptr;
+  void* scale_ptr = nullptr;
+  if (out8.has_value()) {
+    TORCH_CHECK(out_scale.has_value(), "out8 needs out_scale");
+    TORCH_CHECK(out8->numel() == x.numel() &&
+                out_scale->numel() >= rows, "fp8 sidecar sizes");
+    TORCH_CHECK(hidden % 64 == 0, "fp8 sidecar needs hidden % 64 == 0");
+    out8_ptr = out8->data_ptr();
+    scale_ptr = out_scale->data_ptr();
+  }
   launch_fused_add_rmsnorm(out.data_ptr(), res_ptr, x.data_ptr(),
                            weight.data_ptr(), (int)rows, (int)hidden,
-                           (float)eps, stream());
+                           (float)eps, out8_ptr, scale_ptr, stream());
 }
 
 // q may be a strided row-view into a fused qkv buffer:

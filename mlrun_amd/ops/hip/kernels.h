@@ -11,7 +11,8 @@ extern "C++" {
 
 void launch_fused_add_rmsnorm(void* out, void* residual, const void* x,
                               const void* weight, int rows, int hidden,
-                              float eps, void* stream);
+                              float eps, void* out8, void* out_scale,
+                              void* stream);
 
 void launch_rope(void* q, const void* positions, const void* cos_sin, int T,
                  int heads, int dim, long long row_stride, void* stream);

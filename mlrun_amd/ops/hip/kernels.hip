@@ -54,11 +54,15 @@ struct ushort2v { unsigned short x, y; };
 //   residual <- x + residual ; out <- rmsnorm(residual) * weight
 // one workgroup per row; vectorized short8 loads (guide G13).
 // ---------------------------------------------------------------------
+// out8/out_scale (optional): also emit the normalized row quantized
+// to fp8 e4m3 in the GEMM's pair-swizzled layout — feeds the
+// fp8-weight decode path without a separate quant launch.
 __global__ void fused_add_rmsnorm_kernel(
     unsigned short* __restrict__ out, unsigned short* __restrict__ residual,
     const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ weight, int hidden, float eps,
-    int has_residual) {
+    int has_residual, unsigned char* __restrict__ out8,
+    float* __restrict__ out_scale) {
   const int row = blockIdx.x;
   const int tid = threadIdx.x;
   const int nthreads = blockDim.x;
@@ -104,26 +108,61 @@ __global__ void fused_add_rmsnorm_kernel(
   const float inv = rsqrtf(red[0] / hidden + eps);
 
   const unsigned short* zsrc = has_residual ? rr : xr;
+  float amax = 0.f;
   for (int i = tid * 8; i < hidden; i += nthreads * 8) {
     ushort8 zv = *reinterpret_cast<const ushort8*>(zsrc + i);
     ushort8 wv = *reinterpret_cast<const ushort8*>(weight + i);
     ushort8 ov;
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      ov.v[j] = f2bf(bf2f(zv.v[j]) * inv * bf2f(wv.v[j]));
+    for (int j = 0; j < 8; ++j) {
+      const float value = bf2f(zv.v[j]) * inv * bf2f(wv.v[j]);
+      ov.v[j] = f2bf(value);
+      amax = fmaxf(amax, fabsf(value));
+    }
     *reinterpret_cast<ushort8*>(outr + i) = ov;
+  }
+  if (out8 == nullptr) return;
+  // fp8 sidecar: block-reduce amax, then re-emit the row quantized
+  // into the gemm's pair-swizzled layout (row data is L1/L2 hot)
+  __syncthreads();  // red[] still feeds inv on slower threads
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    amax = fmaxf(amax, __shfl_xor(amax, off, 64));
+  if ((tid & 63) == 0) red[tid >> 6] = amax;
+  __syncthreads();
+  float gmax = 0.f;
+  for (int w = 0; w < (nthreads >> 6); ++w) gmax = fmaxf(gmax, red[w]);
+  const float qscale = gmax > 0.f ? gmax / 448.f : 1.f;
+  const float qinv = 1.f / qscale;
+  if (tid == 0) out_scale[row] = qscale;
+  unsigned char* out8r = out8 + (size_t)row * hidden;
+  for (int i = tid * 8; i < hidden; i += nthreads * 8) {
+    ushort8 ov = *reinterpret_cast<const ushort8*>(outr + i);
+    unsigned char q[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __hip_fp8_e4m3 f8(bf2f(ov.v[j]) * qinv);
+      q[j] = f8.__x;
+    }
+    const int sblk = i >> 5;
+    const int c = (i >> 3) & 3;
+    const int dst = ((sblk >> 1) << 6) + (c << 4) + ((sblk & 1) << 3);
+    *reinterpret_cast<uint64_t*>(out8r + dst) =
+        *reinterpret_cast<const uint64_t*>(q);
   }
 }
 
 void launch_fused_add_rmsnorm(void* out, void* residual, const void* x,
                               const void* weight, int rows, int hidden,
-                              float eps, void* stream) {
+                              float eps, void* out8, void* out_scale,
+                              void* stream) {
   int threads = hidden >= 2048 ? 256 : 64;
   hipLaunchKernelGGL(fused_add_rmsnorm_kernel, dim3(rows), dim3(threads), 0,
                      (hipStream_t)stream, (unsigned short*)out,
                      (unsigned short*)residual, (const unsigned short*)x,
                      (const unsigned short*)weight, hidden, eps,
-                     residual != nullptr ? 1 : 0);
+                     residual != nullptr ? 1 : 0, (unsigned char*)out8,
+                     (float*)out_scale);
 }
 
 // ---------------------------------------------------------------------
