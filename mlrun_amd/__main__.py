@@ -135,6 +135,27 @@ def cmd_clean(args):
     print(f"cleaned runs in {project}")
 
 
+def cmd_build(args):
+    """Build a function "image" dir (reference `mlrun build` :485;
+    kaniko replaced by the node-local directory builder)."""
+    import mlrun_amd
+    from .utils.builder import build_runtime
+
+    if args.spec:
+        fn = mlrun_amd.import_function(args.spec)
+    else:
+        fn = mlrun_amd.new_function(name=args.name or "fn", kind="job",
+                                    command=args.command or "")
+    if args.requirements:
+        fn.spec.build.setdefault("requirements", []).extend(
+            args.requirements)
+    if args.source:
+        fn.spec.build["source"] = args.source
+    build_runtime(fn, with_mlrun=args.with_mlrun,
+                  install=args.install)
+    print(f"image built at {fn.spec.image}")
+
+
 def cmd_deploy(args):
     import mlrun_amd
 
@@ -215,6 +236,17 @@ def main(argv=None):
     p.add_argument("--project", default="")
     p.add_argument("--state", default="")
     p.set_defaults(func=cmd_clean)
+
+    p = sub.add_parser("build", help="build a function image dir")
+    p.add_argument("--spec", default="")
+    p.add_argument("--name", default="")
+    p.add_argument("--command", default="")
+    p.add_argument("--source", default="")
+    p.add_argument("--requirements", "-r", action="append")
+    p.add_argument("--with-mlrun", action="store_true")
+    p.add_argument("--install", action="store_true",
+                   help="resolve requirements from the offline wheelhouse")
+    p.set_defaults(func=cmd_build)
 
     p = sub.add_parser("deploy", help="deploy a serving function")
     p.add_argument("--spec", default="")

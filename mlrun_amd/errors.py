@@ -46,6 +46,10 @@ class MLRunPreconditionFailedError(MLRunBaseError):
     pass
 
 
+class MLRunMissingDependencyError(MLRunBaseError, ImportError):
+    """An optional integration's package is not installed."""
+
+
 class MLRunIncompatibleVersionError(MLRunBaseError):
     pass
 

@@ -13,6 +13,9 @@ subclasses).  MI355X-native set:
 - llama:    generative serving (mlrun_amd/models/llama.py LlamaServer)
 - sklearn:  pickled sklearn estimators (CPU predict in the graph,
             or exported to the tree kernel when tree-based)
+- lgbm/xgboost: portable model dumps parsed natively (no library
+            needed at serving time) onto the HIP tree kernel
+- onnx/tf_keras: import-gated interchange servers
 """
 
 from .tree import (  # noqa: F401
@@ -21,5 +24,13 @@ from .tree import (  # noqa: F401
     random_forest_nodes,
 )
 from .sklearn import PickleModelServer, SKLearnModelServer  # noqa: F401
+from .lgbm import LGBMModelServer, model_from_lgbm, nodes_from_lgbm_text  # noqa: F401
+from .xgboost import (  # noqa: F401
+    XGBoostModelServer,
+    model_from_xgboost,
+    nodes_from_xgboost_dump,
+)
+from .onnx import ONNXModelServer  # noqa: F401
+from .tf_keras import TFKerasModelServer  # noqa: F401
 from .torch_nn import PyTorchModelServer  # noqa: F401
 from .auto import apply_mlrun, detect_framework, get_model_server_class  # noqa: F401

@@ -27,6 +27,10 @@ def detect_framework(model) -> str:
         return "llama"
     if "mlrun_amd.frameworks.tree" in joined:
         return "tree"
+    if "lightgbm." in joined:
+        return "lightgbm"
+    if "xgboost." in joined:
+        return "xgboost"
     raise MLRunInvalidArgumentError(
         f"cannot auto-detect framework for {type(model).__name__}")
 
@@ -55,11 +59,14 @@ def get_model_server_class(framework: str):
     from .torch_nn import PyTorchModelServer
     from .tree import TreeEnsembleModelServer
 
+    from .lgbm import LGBMModelServer
+    from .xgboost import XGBoostModelServer
+
     table = {
         "sklearn": SKLearnModelServer,
         "pickle": PickleModelServer,
-        "xgboost": SKLearnModelServer,
-        "lightgbm": SKLearnModelServer,
+        "xgboost": XGBoostModelServer,
+        "lightgbm": LGBMModelServer,
         "tree": TreeEnsembleModelServer,
         "pytorch": PyTorchModelServer,
         "torch": PyTorchModelServer,

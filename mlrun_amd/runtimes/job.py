@@ -40,19 +40,17 @@ class KubejobRuntime(BaseRuntime):
     def deploy(self, watch=True, with_mlrun=None, skip_deployed=False,
                is_kfp=False, mlrun_version_specifier=None, builder_env=None,
                show_on_failure=False):
-        """No container builds in the node-local model — materialize the
-        function source (if embedded) to the function workdir."""
-        body = self.spec.build.get("functionSourceCode")
-        if body and not self.spec.command:
-            from ..config import config
+        """No container builds in the node-local model — the builder
+        materializes an image DIRECTORY (source + requirements +
+        run.sh; utils/builder.py, reference build_runtime :644)."""
+        if skip_deployed and self.is_deployed():
+            return True
+        build = self.spec.build or {}
+        if build.get("functionSourceCode") or build.get("source") or \
+                build.get("requirements"):
+            from ..utils.builder import build_runtime
 
-            func_dir = os.path.join(config.base_dir, "functions",
-                                    self.metadata.project or "default")
-            os.makedirs(func_dir, exist_ok=True)
-            path = os.path.join(func_dir, f"{self.metadata.name}.py")
-            with open(path, "w") as fp:
-                fp.write(body)
-            self.spec.command = path
+            return build_runtime(self, with_mlrun=with_mlrun)
         self.status.state = "ready"
         return True
 
