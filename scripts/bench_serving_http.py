@@ -1,0 +1,134 @@
+#!/usr/bin/env python3
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""HTTP-level serving load benchmark.
+
+Measures the FULL stack — uvicorn HTTP host -> serving graph ->
+V2ModelServer -> dynamic batcher -> hipGraph decode — under concurrent
+clients, the shape of the reference's serving benchmark harness
+(N concurrent clients, single-prompt requests, req/s + latency
+percentiles).  bench.py times the engine/graph path; this script adds
+the network + batching layer on top.
+
+  python scripts/bench_serving_http.py --clients 16 --requests 512 \
+      --gen-tokens 32 [--model llama-3-8b] [--weights fp8]
+"""
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import threading
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(
+    __file__))))
+
+import torch  # noqa: E402
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--clients", type=int, default=16)
+    parser.add_argument("--requests", type=int, default=256)
+    parser.add_argument("--model", default=None)
+    parser.add_argument("--prompt-len", type=int, default=128)
+    parser.add_argument("--gen-tokens", type=int, default=32)
+    parser.add_argument("--batch-window-ms", type=float, default=8.0)
+    parser.add_argument("--batch", type=int, default=32)
+    parser.add_argument("--replicas", type=int, default=1)
+    parser.add_argument("--weights", default="bf16",
+                        choices=["bf16", "fp8"])
+    parser.add_argument("--warmup", type=int, default=32)
+    args = parser.parse_args()
+
+    on_gpu = torch.cuda.is_available()
+    model = args.model or ("llama-3-8b" if on_gpu else "tiny")
+
+    import mlrun_amd
+    from mlrun_amd.models.llama import LlamaServer
+
+    fn = mlrun_amd.new_function(name="http-bench", kind="serving")
+    fn.add_model("llm", class_name=LlamaServer, config=model,
+                 batch_size=args.batch,
+                 max_new_tokens=args.gen_tokens,
+                 batch_window_ms=args.batch_window_ms,
+                 replicas=args.replicas,
+                 weight_dtype="fp8w" if args.weights == "fp8" else "bf16")
+    address = fn.deploy()
+    print(f"serving at {address}", file=sys.stderr)
+
+    import requests as http
+
+    vocab = 1000
+    rng = torch.Generator().manual_seed(0)
+    prompt = torch.randint(0, vocab, (args.prompt_len,),
+                           generator=rng).tolist()
+    url = f"{address}/v2/models/llm/infer"
+    payload = json.dumps({"inputs": [prompt],
+                          "max_tokens": args.gen_tokens})
+    headers = {"content-type": "application/json"}
+
+    latencies = []
+    lock = threading.Lock()
+    counter = {"left": args.requests}
+
+    def client():
+        session = http.Session()
+        while True:
+            with lock:
+                if counter["left"] <= 0:
+                    return
+                counter["left"] -= 1
+            start = time.perf_counter()
+            resp = session.post(url, data=payload, headers=headers,
+                                timeout=300)
+            elapsed = (time.perf_counter() - start) * 1000
+            resp.raise_for_status()
+            out = resp.json()["outputs"]
+            assert len(out[0]) == args.gen_tokens, out
+            with lock:
+                latencies.append(elapsed)
+
+    # warmup (captures hipGraphs, fills caches)
+    counter["left"] = args.warmup
+    threads = [threading.Thread(target=client)
+               for _ in range(min(args.clients, args.warmup))]
+    [t.start() for t in threads]
+    [t.join() for t in threads]
+    latencies.clear()
+
+    counter["left"] = args.requests
+    started = time.perf_counter()
+    threads = [threading.Thread(target=client)
+               for _ in range(args.clients)]
+    [t.start() for t in threads]
+    [t.join() for t in threads]
+    wall = time.perf_counter() - started
+    fn.stop()
+
+    latencies.sort()
+    result = {
+        "metric": "HTTP serving req/sec (Llama V2ModelServer, "
+                  "concurrent clients)",
+        "value": round(args.requests / wall, 2),
+        "unit": "req/s",
+        "p50_ms": round(statistics.median(latencies), 2),
+        "p99_ms": round(latencies[int(len(latencies) * 0.99) - 1], 2),
+        "clients": args.clients,
+        "requests": args.requests,
+        "gen_tokens": args.gen_tokens,
+        "tokens_per_sec": round(args.requests * args.gen_tokens / wall, 1),
+        "batch_window_ms": args.batch_window_ms,
+        "model": model,
+        "weights": args.weights,
+        "replicas": args.replicas,
+    }
+    print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()
