@@ -206,7 +206,8 @@ class LlamaDecodeEngine:
                         ops.quantize_fp8_weight(layer[key])
             self._fp8_packs[id(w.lm_head)] = \
                 ops.quantize_fp8_weight(w.lm_head)
-        self.attn_nsplit = ops.pick_attn_nsplit(B, w.hkv)
+        self.attn_nsplit = ops.pick_attn_nsplit(B, w.hkv,
+                                        seq_len=smax)
         self.buf_attn_ws = torch.empty(
             B * w.hq * self.attn_nsplit * (d + 2), dtype=torch.float32,
             device=self.device) if self.on_gpu else None
