@@ -206,8 +206,16 @@ class LlamaDecodeEngine:
                         ops.quantize_fp8_weight(layer[key])
             self._fp8_packs[id(w.lm_head)] = \
                 ops.quantize_fp8_weight(w.lm_head)
-        self.attn_nsplit = ops.pick_attn_nsplit(B, w.hkv,
-                                        seq_len=smax)
+        # fill-based (ns4 at B=32): the graph bakes ONE nsplit for all
+        # context lengths, and decode mostly runs at S << max_seq_len —
+        # measured (scripts/bench_attn_longctx.py): chunk-aware ns16
+        # wins >=10% only at S>=1024 but costs 2.4% on the headline
+        # (S~160).  Long-context deployments: pass seq_len to
+        # pick_attn_nsplit / set MLRUN_ATTN_NSPLIT.
+        import os as _os
+        _ns = _os.environ.get("MLRUN_ATTN_NSPLIT")
+        self.attn_nsplit = int(_ns) if _ns else \
+            ops.pick_attn_nsplit(B, w.hkv)
         self.buf_attn_ws = torch.empty(
             B * w.hq * self.attn_nsplit * (d + 2), dtype=torch.float32,
             device=self.device) if self.on_gpu else None

@@ -18,16 +18,36 @@ the network + batching layer on top.
 
 import argparse
 import json
+import multiprocessing
 import os
 import statistics
 import sys
-import threading
 import time
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(
     __file__))))
 
 import torch  # noqa: E402
+
+
+def _client_proc(url, payload, gen_tokens, remaining, results):
+    import requests as http
+
+    session = http.Session()
+    headers = {"content-type": "application/json"}
+    while True:
+        with remaining.get_lock():
+            if remaining.value <= 0:
+                return
+            remaining.value -= 1
+        start = time.perf_counter()
+        resp = session.post(url, data=payload, headers=headers,
+                            timeout=300)
+        elapsed = (time.perf_counter() - start) * 1000
+        resp.raise_for_status()
+        out = resp.json()["outputs"]
+        assert len(out[0]) == gen_tokens, out
+        results.put(elapsed)
 
 
 def main():
@@ -61,8 +81,6 @@ def main():
     address = fn.deploy()
     print(f"serving at {address}", file=sys.stderr)
 
-    import requests as http
-
     vocab = 1000
     rng = torch.Generator().manual_seed(0)
     prompt = torch.randint(0, vocab, (args.prompt_len,),
@@ -70,44 +88,31 @@ def main():
     url = f"{address}/v2/models/llm/infer"
     payload = json.dumps({"inputs": [prompt],
                           "max_tokens": args.gen_tokens})
-    headers = {"content-type": "application/json"}
 
-    latencies = []
-    lock = threading.Lock()
-    counter = {"left": args.requests}
+    # clients are separate PROCESSES — one client process running
+    # dozens of threads is GIL-bound and under-drives the server
+    ctx = multiprocessing.get_context("fork")
+    remaining = ctx.Value("i", 0)
+    results: multiprocessing.Queue = ctx.Queue()
 
-    def client():
-        session = http.Session()
-        while True:
-            with lock:
-                if counter["left"] <= 0:
-                    return
-                counter["left"] -= 1
-            start = time.perf_counter()
-            resp = session.post(url, data=payload, headers=headers,
-                                timeout=300)
-            elapsed = (time.perf_counter() - start) * 1000
-            resp.raise_for_status()
-            out = resp.json()["outputs"]
-            assert len(out[0]) == args.gen_tokens, out
-            with lock:
-                latencies.append(elapsed)
+    def run_wave(n_requests):
+        with remaining.get_lock():
+            remaining.value = n_requests
+        procs = [ctx.Process(target=_client_proc,
+                             args=(url, payload, args.gen_tokens,
+                                   remaining, results))
+                 for _ in range(args.clients)]
+        start = time.perf_counter()
+        [p.start() for p in procs]
+        [p.join() for p in procs]
+        elapsed = time.perf_counter() - start
+        lat = []
+        while not results.empty():
+            lat.append(results.get())
+        return elapsed, lat
 
-    # warmup (captures hipGraphs, fills caches)
-    counter["left"] = args.warmup
-    threads = [threading.Thread(target=client)
-               for _ in range(min(args.clients, args.warmup))]
-    [t.start() for t in threads]
-    [t.join() for t in threads]
-    latencies.clear()
-
-    counter["left"] = args.requests
-    started = time.perf_counter()
-    threads = [threading.Thread(target=client)
-               for _ in range(args.clients)]
-    [t.start() for t in threads]
-    [t.join() for t in threads]
-    wall = time.perf_counter() - started
+    run_wave(args.warmup)  # captures hipGraphs, fills caches
+    wall, latencies = run_wave(args.requests)
     fn.stop()
 
     latencies.sort()
