@@ -1171,12 +1171,86 @@ void launch_rope_kv_fused(void* qkv, void* Kc, void* Vc,
 // Decode attention (single new token per sequence, GQA):
 //   O[b,h,:] = softmax(Q[b,h,:] K[b,kvh,:len,:]^T * scale) V[b,kvh,:len,:]
 // One workgroup per (b, kv-head); one wave per grouped q-head.
-// Memory-bound on the KV cache; K reads 16 lanes x 16B contiguous per
-// position, V reads 64 lanes x 4B contiguous per position.
-// dim must be 128 (llama-family head_dim).
+//
+// K/V chunks are staged through LDS by ALL waves cooperatively: the G
+// grouped q-heads attend to the SAME kv rows, so per-wave global reads
+// would issue the identical K and V bytes G times (L2 absorbs the
+// reuse but the kernel becomes issue/latency-bound — measured 2.5
+// TB/s apparent).  One cooperative load + G waves reading LDS cuts
+// vector-memory traffic G-fold.  dim must be 128 (llama head_dim).
 // ---------------------------------------------------------------------
-#define ATTN_SCHUNK 256
+#define ATTN_SCHUNK 64  // rows staged per LDS buffer (2 x 16 KB)
 #define ATTN_MAXG 8
+
+struct AttnState {
+  float m, l, acc0, acc1;
+};
+
+// shared chunk loop: stage [s_begin, s_final) through kbuf/vbuf and
+// accumulate the online-softmax state for this wave's head
+__device__ __forceinline__ AttnState attn_chunk_loop(
+    const unsigned short* __restrict__ kbase,
+    const unsigned short* __restrict__ vbase,
+    const float* qf, float scale, int s_begin, int s_final,
+    int wave, int lane, int tid, int nthreads,
+    unsigned short (*kbuf)[128], unsigned short (*vbuf)[128],
+    float (*scores)[ATTN_SCHUNK]) {
+  constexpr int D = 128;
+  AttnState st = {-FLT_MAX, 0.f, 0.f, 0.f};
+  const int d0 = lane * 2;
+  for (int s0 = s_begin; s0 < s_final; s0 += ATTN_SCHUNK) {
+    const int cnt = min(ATTN_SCHUNK, s_final - s0);
+    // --- cooperative stage: every thread loads 16B vectors of K and V
+    const int total_vec = cnt * (D / 8);
+    for (int i = tid; i < total_vec; i += nthreads) {
+      const int row = i >> 4;
+      const int col = (i & 15) * 8;
+      const size_t off = (size_t)(s0 + row) * D + col;
+      *reinterpret_cast<ushort8*>(&kbuf[row][col]) =
+          *reinterpret_cast<const ushort8*>(kbase + off);
+      *reinterpret_cast<ushort8*>(&vbuf[row][col]) =
+          *reinterpret_cast<const ushort8*>(vbase + off);
+    }
+    __syncthreads();
+    // --- scores from LDS (16-lane groups each cover all 128 dims)
+    for (int si = lane >> 4; si < cnt; si += 4) {
+      const unsigned short* kp = &kbuf[si][(lane & 15) * 8];
+      ushort8 kv = *reinterpret_cast<const ushort8*>(kp);
+      float dot = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dot += qf[j] * bf2f(kv.v[j]);
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        dot += __shfl_xor(dot, off, 64);
+      if ((lane & 15) == 0) scores[wave][si] = dot * scale;
+    }
+    __syncthreads();
+    // --- chunk max (wave-wide) ---
+    float cm = -FLT_MAX;
+    for (int si = lane; si < cnt; si += 64)
+      cm = fmaxf(cm, scores[wave][si]);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      cm = fmaxf(cm, __shfl_xor(cm, off, 64));
+    const float m_new = fmaxf(st.m, cm);
+    const float rescale = (st.m == -FLT_MAX) ? 0.f : __expf(st.m - m_new);
+    st.acc0 *= rescale;
+    st.acc1 *= rescale;
+    st.l *= rescale;
+    // --- p * V from LDS (lane owns dims d0, d0+1) ---
+    for (int si = 0; si < cnt; ++si) {
+      const float p = __expf(scores[wave][si] - m_new);
+      st.l += p;
+      const ushort2v vv =
+          *reinterpret_cast<const ushort2v*>(&vbuf[si][d0]);
+      st.acc0 += p * bf2f(vv.x);
+      st.acc1 += p * bf2f(vv.y);
+    }
+    st.m = m_new;
+    __syncthreads();
+  }
+  return st;
+}
 
 __global__ void attn_decode_kernel(
     unsigned short* __restrict__ O, const unsigned short* __restrict__ Q,
@@ -1194,6 +1268,8 @@ __global__ void attn_decode_kernel(
   const int len = seq_lens[b];
 
   __shared__ float scores[ATTN_MAXG][ATTN_SCHUNK];
+  __shared__ __align__(16) unsigned short kbuf[ATTN_SCHUNK][D];
+  __shared__ __align__(16) unsigned short vbuf[ATTN_SCHUNK][D];
 
   // q fragment: lane holds 8 consecutive dims at (lane&15)*8 for the
   // K-dot phase (16-lane groups each cover all 128 dims)
@@ -1207,59 +1283,14 @@ __global__ void attn_decode_kernel(
   }
 
   const size_t kv_base = ((size_t)b * Hkv + kvh) * Smax * D;
-  const unsigned short* kbase = Kc + kv_base;
-  const unsigned short* vbase = Vc + kv_base;
+  const AttnState st = attn_chunk_loop(
+      Kc + kv_base, Vc + kv_base, qf, scale, 0, len, wave, lane,
+      threadIdx.x, G * 64, kbuf, vbuf, scores);
 
-  float m = -FLT_MAX, l = 0.f;
-  float acc0 = 0.f, acc1 = 0.f;  // lane owns dims lane*2, lane*2+1
-
-  for (int s0 = 0; s0 < len; s0 += ATTN_SCHUNK) {
-    const int cnt = min(ATTN_SCHUNK, len - s0);
-    // --- scores for this chunk ---
-    for (int si = lane >> 4; si < cnt; si += 4) {
-      const unsigned short* kp = kbase + (size_t)(s0 + si) * D +
-                                 (lane & 15) * 8;
-      ushort8 kv = *reinterpret_cast<const ushort8*>(kp);
-      float dot = 0.f;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dot += qf[j] * bf2f(kv.v[j]);
-      // reduce across the 16-lane group
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        dot += __shfl_xor(dot, off, 64);
-      if ((lane & 15) == 0) scores[wave][si] = dot * scale;
-    }
-    __syncthreads();
-    // --- chunk max (wave-wide) ---
-    float cm = -FLT_MAX;
-    for (int si = lane; si < cnt; si += 64)
-      cm = fmaxf(cm, scores[wave][si]);
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1)
-      cm = fmaxf(cm, __shfl_xor(cm, off, 64));
-    const float m_new = fmaxf(m, cm);
-    const float rescale = (m == -FLT_MAX) ? 0.f : __expf(m - m_new);
-    acc0 *= rescale;
-    acc1 *= rescale;
-    l *= rescale;
-    // --- p * V accumulate ---
-    const int d0 = lane * 2;
-    for (int si = 0; si < cnt; ++si) {
-      const float p = __expf(scores[wave][si] - m_new);
-      l += p;
-      const unsigned short* vp = vbase + (size_t)(s0 + si) * D + d0;
-      ushort2v vv = *reinterpret_cast<const ushort2v*>(vp);
-      acc0 += p * bf2f(vv.x);
-      acc1 += p * bf2f(vv.y);
-    }
-    m = m_new;
-    __syncthreads();
-  }
-
-  const float inv = (l > 0.f) ? 1.f / l : 0.f;
+  const float inv = (st.l > 0.f) ? 1.f / st.l : 0.f;
   unsigned short* op = O + ((size_t)b * Hq + h) * D + lane * 2;
-  op[0] = f2bf(acc0 * inv);
-  op[1] = f2bf(acc1 * inv);
+  op[0] = f2bf(st.acc0 * inv);
+  op[1] = f2bf(st.acc1 * inv);
 }
 
 // ---------------------------------------------------------------------
@@ -1291,6 +1322,8 @@ __global__ void attn_decode_split_kernel(
   float* prow = partial + (((size_t)b * Hq + h) * nsplit + split) * (D + 2);
 
   __shared__ float scores[ATTN_MAXG][ATTN_SCHUNK];
+  __shared__ __align__(16) unsigned short kbuf[ATTN_SCHUNK][D];
+  __shared__ __align__(16) unsigned short vbuf[ATTN_SCHUNK][D];
 
   const unsigned short* qp = Q + (size_t)b * q_row_stride +
                              (size_t)h * D + (lane & 15) * 8;
@@ -1301,54 +1334,15 @@ __global__ void attn_decode_split_kernel(
     for (int j = 0; j < 8; ++j) qf[j] = bf2f(qv.v[j]);
   }
   const size_t kv_base = ((size_t)b * Hkv + kvh) * Smax * D;
-  const unsigned short* kbase = Kc + kv_base;
-  const unsigned short* vbase = Vc + kv_base;
+  const AttnState st = attn_chunk_loop(
+      Kc + kv_base, Vc + kv_base, qf, scale, s_begin, s_final, wave,
+      lane, threadIdx.x, G * 64, kbuf, vbuf, scores);
 
-  float m = -FLT_MAX, l = 0.f;
-  float acc0 = 0.f, acc1 = 0.f;
-  for (int s0 = s_begin; s0 < s_final; s0 += ATTN_SCHUNK) {
-    const int cnt = min(ATTN_SCHUNK, s_final - s0);
-    for (int si = lane >> 4; si < cnt; si += 4) {
-      const unsigned short* kp = kbase + (size_t)(s0 + si) * D +
-                                 (lane & 15) * 8;
-      ushort8 kv = *reinterpret_cast<const ushort8*>(kp);
-      float dot = 0.f;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dot += qf[j] * bf2f(kv.v[j]);
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        dot += __shfl_xor(dot, off, 64);
-      if ((lane & 15) == 0) scores[wave][si] = dot * scale;
-    }
-    __syncthreads();
-    float cm = -FLT_MAX;
-    for (int si = lane; si < cnt; si += 64)
-      cm = fmaxf(cm, scores[wave][si]);
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1)
-      cm = fmaxf(cm, __shfl_xor(cm, off, 64));
-    const float m_new = fmaxf(m, cm);
-    const float rescale = (m == -FLT_MAX) ? 0.f : __expf(m - m_new);
-    acc0 *= rescale;
-    acc1 *= rescale;
-    l *= rescale;
-    const int d0 = lane * 2;
-    for (int si = 0; si < cnt; ++si) {
-      const float p = __expf(scores[wave][si] - m_new);
-      l += p;
-      const unsigned short* vp = vbase + (size_t)(s0 + si) * D + d0;
-      ushort2v vv = *reinterpret_cast<const ushort2v*>(vp);
-      acc0 += p * bf2f(vv.x);
-      acc1 += p * bf2f(vv.y);
-    }
-    m = m_new;
-    __syncthreads();
-  }
-  prow[lane * 2] = acc0;
-  prow[lane * 2 + 1] = acc1;
+  prow[lane * 2] = st.acc0;
+  prow[lane * 2 + 1] = st.acc1;
   if (lane == 0) {
-    prow[D] = m;
-    prow[D + 1] = l;
+    prow[D] = st.m;
+    prow[D + 1] = st.l;
   }
 }
 
