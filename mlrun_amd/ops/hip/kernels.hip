@@ -1187,11 +1187,7 @@ struct AttnState {
 };
 
 // shared chunk loop: stage [s_begin, s_final) through kbuf/vbuf and
-// accumulate the online-softmax state for this wave's head.
-// Software-pipelined: each thread prefetches its next chunk's K/V
-// vectors into registers right after the LDS buffers are published,
-// so the global-load latency overlaps the score/softmax/V compute of
-// the current chunk (single LDS buffer, reg double-buffer).
+// accumulate the online-softmax state for this wave's head
 __device__ __forceinline__ AttnState attn_chunk_loop(
     const unsigned short* __restrict__ kbase,
     const unsigned short* __restrict__ vbase,
@@ -1200,74 +1196,22 @@ __device__ __forceinline__ AttnState attn_chunk_loop(
     unsigned short (*kbuf)[128], unsigned short (*vbuf)[128],
     float (*scores)[ATTN_SCHUNK]) {
   constexpr int D = 128;
-  constexpr int VPC = ATTN_SCHUNK * (D / 8);  // 16B vectors per chunk
   AttnState st = {-FLT_MAX, 0.f, 0.f, 0.f};
-  if (s_begin >= s_final) return st;
   const int d0 = lane * 2;
-  // reg-prefetch needs 4 slots/thread to cover a chunk; smaller
-  // thread counts (G < 4) fall back to direct cooperative staging
-  const bool pipelined = nthreads * 4 >= VPC;
-  // this thread's staging slots (fixed across chunks)
-  int rows[4], cols[4];
-  int nslots = 0;
-#pragma unroll
-  for (int v = 0; v < 4; ++v) {
-    const int i = tid + v * nthreads;
-    if (i < VPC) {
-      rows[v] = i >> 4;
-      cols[v] = (i & 15) * 8;
-      nslots = v + 1;
-    }
-  }
-  const int last = s_final - 1;
-  ushort8 kreg[4], vreg[4];
-  // prologue: fetch chunk 0 (rows clamped to the slice end — clamped
-  // rows land in LDS but the compute loop never reads past cnt)
-  if (pipelined) {
-#pragma unroll
-    for (int v = 0; v < 4; ++v)
-      if (v < nslots) {
-        const size_t off = (size_t)min(s_begin + rows[v], last) * D +
-                           cols[v];
-        kreg[v] = *reinterpret_cast<const ushort8*>(kbase + off);
-        vreg[v] = *reinterpret_cast<const ushort8*>(vbase + off);
-      }
-  }
   for (int s0 = s_begin; s0 < s_final; s0 += ATTN_SCHUNK) {
     const int cnt = min(ATTN_SCHUNK, s_final - s0);
-    if (pipelined) {
-      // publish the prefetched chunk
-#pragma unroll
-      for (int v = 0; v < 4; ++v)
-        if (v < nslots) {
-          *reinterpret_cast<ushort8*>(&kbuf[rows[v]][cols[v]]) = kreg[v];
-          *reinterpret_cast<ushort8*>(&vbuf[rows[v]][cols[v]]) = vreg[v];
-        }
-    } else {
-      const int total_vec = cnt * (D / 8);
-      for (int i = tid; i < total_vec; i += nthreads) {
-        const int row = i >> 4;
-        const int col = (i & 15) * 8;
-        const size_t off = (size_t)(s0 + row) * D + col;
-        *reinterpret_cast<ushort8*>(&kbuf[row][col]) =
-            *reinterpret_cast<const ushort8*>(kbase + off);
-        *reinterpret_cast<ushort8*>(&vbuf[row][col]) =
-            *reinterpret_cast<const ushort8*>(vbase + off);
-      }
+    // --- cooperative stage: every thread loads 16B vectors of K and V
+    const int total_vec = cnt * (D / 8);
+    for (int i = tid; i < total_vec; i += nthreads) {
+      const int row = i >> 4;
+      const int col = (i & 15) * 8;
+      const size_t off = (size_t)(s0 + row) * D + col;
+      *reinterpret_cast<ushort8*>(&kbuf[row][col]) =
+          *reinterpret_cast<const ushort8*>(kbase + off);
+      *reinterpret_cast<ushort8*>(&vbuf[row][col]) =
+          *reinterpret_cast<const ushort8*>(vbase + off);
     }
     __syncthreads();
-    // issue next chunk's loads NOW — latency hides under the compute
-    const int s_next = s0 + ATTN_SCHUNK;
-    if (pipelined && s_next < s_final) {
-#pragma unroll
-      for (int v = 0; v < 4; ++v)
-        if (v < nslots) {
-          const size_t off =
-              (size_t)min(s_next + rows[v], last) * D + cols[v];
-          kreg[v] = *reinterpret_cast<const ushort8*>(kbase + off);
-          vreg[v] = *reinterpret_cast<const ushort8*>(vbase + off);
-        }
-    }
     // --- scores from LDS (16-lane groups each cover all 128 dims)
     for (int si = lane >> 4; si < cnt; si += 4) {
       const unsigned short* kp = &kbuf[si][(lane & 15) * 8];
