@@ -1179,7 +1179,7 @@ void launch_rope_kv_fused(void* qkv, void* Kc, void* Vc,
 // TB/s apparent).  One cooperative load + G waves reading LDS cuts
 // vector-memory traffic G-fold.  dim must be 128 (llama head_dim).
 // ---------------------------------------------------------------------
-#define ATTN_SCHUNK 64  // rows staged per LDS buffer (2 x 16 KB)
+#define ATTN_SCHUNK 64  // default rows staged per LDS buffer
 #define ATTN_MAXG 8
 
 struct AttnState {
@@ -1188,18 +1188,19 @@ struct AttnState {
 
 // shared chunk loop: stage [s_begin, s_final) through kbuf/vbuf and
 // accumulate the online-softmax state for this wave's head
+template <int SCHUNK>
 __device__ __forceinline__ AttnState attn_chunk_loop(
     const unsigned short* __restrict__ kbase,
     const unsigned short* __restrict__ vbase,
     const float* qf, float scale, int s_begin, int s_final,
     int wave, int lane, int tid, int nthreads,
     unsigned short (*kbuf)[128], unsigned short (*vbuf)[128],
-    float (*scores)[ATTN_SCHUNK]) {
+    float (*scores)[SCHUNK]) {
   constexpr int D = 128;
   AttnState st = {-FLT_MAX, 0.f, 0.f, 0.f};
   const int d0 = lane * 2;
-  for (int s0 = s_begin; s0 < s_final; s0 += ATTN_SCHUNK) {
-    const int cnt = min(ATTN_SCHUNK, s_final - s0);
+  for (int s0 = s_begin; s0 < s_final; s0 += SCHUNK) {
+    const int cnt = min(SCHUNK, s_final - s0);
     // --- cooperative stage: every thread loads 16B vectors of K and V
     const int total_vec = cnt * (D / 8);
     for (int i = tid; i < total_vec; i += nthreads) {
@@ -1252,6 +1253,7 @@ __device__ __forceinline__ AttnState attn_chunk_loop(
   return st;
 }
 
+template <int SCHUNK>
 __global__ void attn_decode_kernel(
     unsigned short* __restrict__ O, const unsigned short* __restrict__ Q,
     const unsigned short* __restrict__ Kc,
@@ -1267,9 +1269,9 @@ __global__ void attn_decode_kernel(
   const int h = kvh * G + wave;
   const int len = seq_lens[b];
 
-  __shared__ float scores[ATTN_MAXG][ATTN_SCHUNK];
-  __shared__ __align__(16) unsigned short kbuf[ATTN_SCHUNK][D];
-  __shared__ __align__(16) unsigned short vbuf[ATTN_SCHUNK][D];
+  __shared__ float scores[ATTN_MAXG][SCHUNK];
+  __shared__ __align__(16) unsigned short kbuf[SCHUNK][D];
+  __shared__ __align__(16) unsigned short vbuf[SCHUNK][D];
 
   // q fragment: lane holds 8 consecutive dims at (lane&15)*8 for the
   // K-dot phase (16-lane groups each cover all 128 dims)
@@ -1283,7 +1285,7 @@ __global__ void attn_decode_kernel(
   }
 
   const size_t kv_base = ((size_t)b * Hkv + kvh) * Smax * D;
-  const AttnState st = attn_chunk_loop(
+  const AttnState st = attn_chunk_loop<SCHUNK>(
       Kc + kv_base, Vc + kv_base, qf, scale, 0, len, wave, lane,
       threadIdx.x, G * 64, kbuf, vbuf, scores);
 
@@ -1301,6 +1303,7 @@ __global__ void attn_decode_kernel(
 // peaked at 128 workgroups on 256 CUs).
 // partial layout: [B, Hq, nsplit, D+2] f32
 // ---------------------------------------------------------------------
+template <int SCHUNK>
 __global__ void attn_decode_split_kernel(
     float* __restrict__ partial, const unsigned short* __restrict__ Q,
     const unsigned short* __restrict__ Kc,
@@ -1321,9 +1324,9 @@ __global__ void attn_decode_split_kernel(
   const int s_final = min(len, s_begin + per);
   float* prow = partial + (((size_t)b * Hq + h) * nsplit + split) * (D + 2);
 
-  __shared__ float scores[ATTN_MAXG][ATTN_SCHUNK];
-  __shared__ __align__(16) unsigned short kbuf[ATTN_SCHUNK][D];
-  __shared__ __align__(16) unsigned short vbuf[ATTN_SCHUNK][D];
+  __shared__ float scores[ATTN_MAXG][SCHUNK];
+  __shared__ __align__(16) unsigned short kbuf[SCHUNK][D];
+  __shared__ __align__(16) unsigned short vbuf[SCHUNK][D];
 
   const unsigned short* qp = Q + (size_t)b * q_row_stride +
                              (size_t)h * D + (lane & 15) * 8;
@@ -1334,7 +1337,7 @@ __global__ void attn_decode_split_kernel(
     for (int j = 0; j < 8; ++j) qf[j] = bf2f(qv.v[j]);
   }
   const size_t kv_base = ((size_t)b * Hkv + kvh) * Smax * D;
-  const AttnState st = attn_chunk_loop(
+  const AttnState st = attn_chunk_loop<SCHUNK>(
       Kc + kv_base, Vc + kv_base, qf, scale, s_begin, s_final, wave,
       lane, threadIdx.x, G * 64, kbuf, vbuf, scores);
 
@@ -1373,21 +1376,36 @@ __global__ void attn_decode_combine_kernel(
   op[1] = f2bf(a1 * inv);
 }
 
+// staged-chunk rows: 32 halves LDS/WG (16.5 KB -> ~8 WGs/CU = 8
+// waves/SIMD) vs 64 (33 KB -> 4 waves/SIMD); selectable for sweeps
+// via MLRUN_ATTN_SCHUNK, default chosen by measurement.
+static int attn_schunk_env() {
+  static int v = [] {
+    const char* e = getenv("MLRUN_ATTN_SCHUNK");
+    return e ? atoi(e) : 32;
+  }();
+  return v;
+}
+
 void launch_attn_decode(void* O, const void* Q, const void* Kc,
                         const void* Vc, const void* seq_lens, int B, int Hq,
                         int Hkv, int Smax, float scale,
                         long long q_row_stride, float* partial_ws,
                         int nsplit, void* stream) {
   const int G = Hq / Hkv;
+  const bool c64 = attn_schunk_env() >= 64;
   if (nsplit <= 1 || partial_ws == nullptr) {
-    hipLaunchKernelGGL(attn_decode_kernel, dim3(B * Hkv), dim3(G * 64), 0,
+    auto* kern = c64 ? attn_decode_kernel<64> : attn_decode_kernel<32>;
+    hipLaunchKernelGGL(kern, dim3(B * Hkv), dim3(G * 64), 0,
                        (hipStream_t)stream, (unsigned short*)O,
                        (const unsigned short*)Q, (const unsigned short*)Kc,
                        (const unsigned short*)Vc, (const int*)seq_lens, B,
                        Hq, Hkv, Smax, scale, q_row_stride);
     return;
   }
-  hipLaunchKernelGGL(attn_decode_split_kernel, dim3(B * Hkv, nsplit),
+  auto* kern = c64 ? attn_decode_split_kernel<64>
+                   : attn_decode_split_kernel<32>;
+  hipLaunchKernelGGL(kern, dim3(B * Hkv, nsplit),
                      dim3(G * 64), 0, (hipStream_t)stream, partial_ws,
                      (const unsigned short*)Q, (const unsigned short*)Kc,
                      (const unsigned short*)Vc, (const int*)seq_lens, B, Hq,
