@@ -340,12 +340,13 @@ def rope_kv_fused(qkv: torch.Tensor, k_cache: torch.Tensor,
 # ------------------------------------------------------------- attention
 
 
-def pick_attn_nsplit(B: int, Hkv: int, target: int = 1024,
+def pick_attn_nsplit(B: int, Hkv: int, target: int = 2048,
                      seq_len: int = None) -> int:
-    """Split-S factor for decode attention: fill 256 CUs AND keep each
-    split's chunk <= ~128 tokens (measured, scripts/
-    bench_attn_longctx.py: at B=32 ns16 beats the fill-only ns4 by
-    10-26% for S>=1024; chunk ~128 is the sweet spot across S)."""
+    """Split-S factor for decode attention.  Measured (scripts/
+    bench_attn_longctx.py, 32-row LDS chunks): ~2048 workgroups is the
+    sweet spot — ns8 at B=32 ties ns4 on the S=160 headline and wins
+    8-14% for S>=1024; the seq_len hint additionally keeps each
+    split's chunk <= ~128 tokens for very long contexts."""
     base = max(B * Hkv, 1)
     fill = (target + base - 1) // base
     chunk = (seq_len + 127) // 128 if seq_len else 1
