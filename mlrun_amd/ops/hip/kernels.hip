@@ -408,7 +408,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 // LDS.  4x the workgroup count of skinny_gemm_kernel at the same
 // split-K slab traffic -> 4x the waves/SIMD for latency hiding
 // (profile: N=4096 projections were latency-bound at 2 waves/SIMD).
-template <bool SPLIT, int MT>
+template <bool SPLIT, int MT, int UNR2 = 4>
 __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
     void* __restrict__ out, const unsigned short* __restrict__ A,
     const unsigned short* __restrict__ W, int M, int N, int K, int ksplit) {
@@ -463,7 +463,8 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
 #ifndef MLRUN_GEMM_UNR1
 #define MLRUN_GEMM_UNR1 16
 #endif
-  constexpr int UNR = (MT == 1) ? MLRUN_GEMM_UNR1 : (MT == 2 ? 4 : 2);
+  constexpr int UNR = (MT == 1) ? MLRUN_GEMM_UNR1
+                                : (MT == 2 ? UNR2 : 2);
   int k = kbegin;
   const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
   for (; k < kend8; k += UNR * 32) {
@@ -795,6 +796,16 @@ __global__ void reduce_cast_kernel(unsigned short* __restrict__ out,
 }
 
 // emit split-K slabs only (consumer kernel folds them)
+// MT=2 wave-split K-loop unroll depth (4 default; 8 doubles the
+// in-flight bytes per wave at +64 VGPRs) — env knob for e2e A/B.
+static int gemm_unr2_env() {
+  static int v = [] {
+    const char* e = getenv("MLRUN_GEMM_UNR2");
+    return e ? atoi(e) : 4;
+  }();
+  return v;
+}
+
 void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
                               int M, int N, int K, int ksplit, int variant,
                               void* stream) {
@@ -807,7 +818,10 @@ void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
                          (const unsigned short*)A, (const unsigned short*)W,
                          M, N, K, ksplit);
     else if (M > 16)
-      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 2>), grid, dim3(256),
+      hipLaunchKernelGGL(gemm_unr2_env() >= 8
+                             ? (skinny_gemm_ws_kernel<true, 2, 8>)
+                             : (skinny_gemm_ws_kernel<true, 2, 4>),
+                         grid, dim3(256),
                          0, (hipStream_t)stream, part_f32,
                          (const unsigned short*)A, (const unsigned short*)W,
                          M, N, K, ksplit);
@@ -849,7 +863,9 @@ void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                            (const unsigned short*)A,
                            (const unsigned short*)W, M, N, K, 1);
       else if (M > 16)
-        hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 2>),
+        hipLaunchKernelGGL(gemm_unr2_env() >= 8
+                               ? (skinny_gemm_ws_kernel<false, 2, 8>)
+                               : (skinny_gemm_ws_kernel<false, 2, 4>),
                            dim3(nblocks), dim3(256), 0,
                            (hipStream_t)stream, out_bf16,
                            (const unsigned short*)A,
