@@ -41,6 +41,8 @@ def main():
     parser.add_argument("--prompt-len", type=int, default=128)
     parser.add_argument("--gen-tokens", type=int, default=32)
     parser.add_argument("--no-graph", action="store_true")
+    parser.add_argument("--kv", default="bf16", choices=["bf16", "fp8"],
+                        help="KV-cache dtype (fp8 = e4m3 + row scales)")
     parser.add_argument("--weights", default="bf16",
                         help="bf16 (headline) | fp8 (opt-in fp8-weight "
                              "decode; activations/KV stay bf16)")
@@ -81,7 +83,8 @@ def main():
                  batch_size=batch, max_new_tokens=args.gen_tokens,
                  device=device, use_graph=not args.no_graph,
                  replicas=args.replicas,
-                 weight_dtype="fp8w" if args.weights == "fp8" else "bf16")
+                 weight_dtype="fp8w" if args.weights == "fp8" else "bf16",
+                 kv_dtype=args.kv)
     server = fn.to_mock_server()
 
     import random
@@ -148,8 +151,9 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if args.weights != "fp8" else
-                     "bf16-act/fp8-weight (opt-in; headline is bf16)",
+            "dtype": ("bf16" if args.weights != "fp8" else
+                      "bf16-act/fp8-weight (opt-in; headline is bf16)") +
+                     ("" if args.kv != "fp8" else " + fp8-kv (opt-in)"),
             "data": "synthetic (random prompts, random-init weights)",
             "config": {
                 "model": model_cfg,

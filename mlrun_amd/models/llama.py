@@ -130,7 +130,7 @@ class LlamaDecodeEngine:
     def __init__(self, cfg: LlamaConfig, batch_size: int, device=None,
                  tp_group=None, tp_rank=0, tp_size=1, use_graph=True,
                  seed=1234, weights: "LlamaWeights" = None,
-                 weight_dtype: str = "bf16"):
+                 weight_dtype: str = "bf16", kv_dtype: str = "bf16"):
         self.cfg = cfg
         self.B = batch_size
         self.device = torch.device(
@@ -150,9 +150,28 @@ class LlamaDecodeEngine:
         B, smax = batch_size, cfg.max_seq_len
         bf16 = dict(dtype=torch.bfloat16, device=self.device)
 
-        # per-layer KV caches (per-rank kv heads): [L, B, Hkv, Smax, D]
-        self.k_cache = torch.zeros(cfg.num_layers, B, w.hkv, smax, d, **bf16)
-        self.v_cache = torch.zeros(cfg.num_layers, B, w.hkv, smax, d, **bf16)
+        # per-layer KV caches (per-rank kv heads): [L, B, Hkv, Smax, D].
+        # kv_dtype="fp8": OCP e4m3 bytes + per-row f32 scales — halves
+        # KV memory AND decode-attention HBM traffic (dequant happens
+        # while staging chunks through LDS).
+        self.kv_dtype = kv_dtype
+        if kv_dtype == "fp8":
+            u8 = dict(dtype=torch.uint8, device=self.device)
+            self.k_cache = torch.zeros(cfg.num_layers, B, w.hkv, smax, d,
+                                       **u8)
+            self.v_cache = torch.zeros(cfg.num_layers, B, w.hkv, smax, d,
+                                       **u8)
+            f32 = dict(dtype=torch.float32, device=self.device)
+            self.k_scale = torch.ones(cfg.num_layers, B, w.hkv, smax,
+                                      **f32)
+            self.v_scale = torch.ones(cfg.num_layers, B, w.hkv, smax,
+                                      **f32)
+        else:
+            self.k_cache = torch.zeros(cfg.num_layers, B, w.hkv, smax, d,
+                                       **bf16)
+            self.v_cache = torch.zeros(cfg.num_layers, B, w.hkv, smax, d,
+                                       **bf16)
+            self.k_scale = self.v_scale = None
         self.cache_lens = torch.zeros(B, dtype=torch.int32,
                                       device=self.device)
         self.buf_positions = torch.zeros(B, dtype=torch.int32,
@@ -298,7 +317,8 @@ class LlamaDecodeEngine:
 
         mode = _os.environ.get("MLRUN_SLAB_MODE", "rope")
         on = self.on_gpu and self.tp_size == 1 and mode != "none"
-        use_slabs = on and self.weight_dtype == "bf16"  # qkv slab
+        use_slabs = on and self.weight_dtype == "bf16" and \
+            self.kv_dtype == "bf16"  # qkv slab (bf16 caches only)
         use_norm_slabs = on and mode == "all" 
         for li, layer in enumerate(w.layers):
             # qkv projection -> rope -> caches
@@ -316,13 +336,17 @@ class LlamaDecodeEngine:
                            a8=h8, a_scale=h8s)
                 ops.rope_kv_fused(self.buf_qkv, self.k_cache[li],
                                   self.v_cache[li], positions,
-                                  self.cos_sin, w.hq)
+                                  self.cos_sin, w.hq,
+                                  k_scale=self._kscale(li),
+                                  v_scale=self._vscale(li))
             q = self.buf_qkv[:, :w.hq * d].view(B, w.hq, d)
             attn_view = self.buf_attn_out.view(B, w.hq, d)
             ops.attn_decode(q, self.k_cache[li], self.v_cache[li],
                             self.cache_lens, self.scale, out=attn_view,
                             partial_ws=self.buf_attn_ws,
-                            nsplit=self.attn_nsplit)
+                            nsplit=self.attn_nsplit,
+                            k_scale=self._kscale(li),
+                            v_scale=self._vscale(li))
             # attn-out projection -> residual+norm
             done = 0
             if use_norm_slabs:
@@ -422,6 +446,12 @@ class LlamaDecodeEngine:
             self._decode_step_body()
 
     # --------------------------------------------------------- prefill
+    def _kscale(self, li):
+        return None if self.k_scale is None else self.k_scale[li]
+
+    def _vscale(self, li):
+        return None if self.v_scale is None else self.v_scale[li]
+
     @torch.no_grad()
     def prefill(self, tokens: torch.Tensor) -> torch.Tensor:
         """Process prompts [B, S]; fill KV caches; return last-token
@@ -446,8 +476,16 @@ class LlamaDecodeEngine:
             ops.rope_inplace(k, positions, self.cos_sin)
             kc = k.view(B, S, w.hkv, d).transpose(1, 2).contiguous()
             vc = v.view(B, S, w.hkv, d).transpose(1, 2).contiguous()
-            self.k_cache[li, :, :, :S] = kc
-            self.v_cache[li, :, :, :S] = vc
+            if self.kv_dtype == "fp8":
+                k8, ks = ops.quantize_kv_rows(kc)
+                v8, vs = ops.quantize_kv_rows(vc)
+                self.k_cache[li, :, :, :S] = k8
+                self.v_cache[li, :, :, :S] = v8
+                self.k_scale[li, :, :, :S] = ks
+                self.v_scale[li, :, :, :S] = vs
+            else:
+                self.k_cache[li, :, :, :S] = kc
+                self.v_cache[li, :, :, :S] = vc
             qh = q.view(B, S, w.hq, d).transpose(1, 2)
             attn = torch.nn.functional.scaled_dot_product_attention(
                 qh, kc, vc, is_causal=True, enable_gqa=True)
@@ -517,7 +555,8 @@ class LlamaServer:
     def __init__(self, context=None, name=None, model_path=None,
                  config=None, batch_size=16, max_new_tokens=32,
                  device=None, use_graph=True, batch_window_ms=0,
-                 replicas=1, weight_dtype="bf16", **class_args):
+                 replicas=1, weight_dtype="bf16", kv_dtype="bf16",
+                 **class_args):
         import queue as queue_mod
         import threading
 
@@ -538,6 +577,7 @@ class LlamaServer:
         self.device = device
         self.use_graph = use_graph
         self.weight_dtype = weight_dtype
+        self.kv_dtype = kv_dtype
         self.replicas = max(int(replicas), 1)
         self.engines: typing.List[LlamaDecodeEngine] = []
         self.batch_window_ms = batch_window_ms
@@ -587,13 +627,14 @@ class LlamaServer:
             self.replicas = 1
         first = LlamaDecodeEngine(cfg, self.batch_size, device=self.device,
                                   use_graph=self.use_graph,
-                                  weight_dtype=self.weight_dtype)
+                                  weight_dtype=self.weight_dtype,
+                                  kv_dtype=self.kv_dtype)
         self.engines = [first]
         for _ in range(self.replicas - 1):
             replica = LlamaDecodeEngine(
                 cfg, self.batch_size, device=self.device,
                 use_graph=self.use_graph, weights=first.weights,
-                weight_dtype="bf16")
+                weight_dtype="bf16", kv_dtype=self.kv_dtype)
             replica.weight_dtype = first.weight_dtype
             replica._fp8_packs = first._fp8_packs  # shared packs
             if first.weight_dtype == "fp8w" and replica.on_gpu:

@@ -320,21 +320,51 @@ def skinny_gemm_fp8(a8: torch.Tensor, a_scale: torch.Tensor,
 
 def rope_kv_fused(qkv: torch.Tensor, k_cache: torch.Tensor,
                   v_cache: torch.Tensor, positions: torch.Tensor,
-                  cos_sin: torch.Tensor, hq: int):
+                  cos_sin: torch.Tensor, hq: int,
+                  k_scale: torch.Tensor = None,
+                  v_scale: torch.Tensor = None):
     """Fused decode rope(q,k) + KV append from the fused qkv buffer
-    [B, (hq+2*hkv)*d]."""
+    [B, (hq+2*hkv)*d].  With k_scale/v_scale the caches are fp8
+    (uint8 e4m3 + per-row f32 scales)."""
     B = qkv.shape[0]
     hkv, d = k_cache.shape[1], k_cache.shape[3]
     if qkv.is_cuda:
         _require_hip().rope_kv_fused(qkv, k_cache, v_cache, positions,
-                                     cos_sin, hq)
+                                     cos_sin, hq, k_scale, v_scale)
         return
     q = qkv[:, :hq * d].view(B, hq, d)
     k = qkv[:, hq * d:(hq + hkv) * d].view(B, hkv, d)
     v = qkv[:, (hq + hkv) * d:(hq + 2 * hkv) * d].view(B, hkv, d)
     rope_inplace(q, positions, cos_sin)
     rope_inplace(k, positions, cos_sin)
+    if k_scale is not None:
+        for b in range(B):
+            pos = int(positions[b])
+            k8, ks = quantize_kv_rows(k[b])
+            v8, vs = quantize_kv_rows(v[b])
+            k_cache[b, :, pos] = k8
+            v_cache[b, :, pos] = v8
+            k_scale.view(B, hkv, -1)[b, :, pos] = ks
+            v_scale.view(B, hkv, -1)[b, :, pos] = vs
+        return
     kv_append(k_cache, v_cache, k, v, positions)
+
+
+def quantize_kv_rows(x: torch.Tensor):
+    """Per-row (last-dim) OCP e4m3 quantization -> (uint8, f32 scales).
+    Matches the rope_kv_fused_q8 kernel: scale = amax/448."""
+    amax = x.float().abs().amax(dim=-1, keepdim=True)
+    scales = torch.where(amax > 0, amax / 448.0,
+                         torch.ones_like(amax))
+    q8 = (x.float() / scales).to(torch.float8_e4m3fn).view(torch.uint8)
+    return q8, scales.squeeze(-1)
+
+
+def dequantize_kv_rows(q8: torch.Tensor, scales: torch.Tensor,
+                       dtype=torch.float32):
+    """Inverse of quantize_kv_rows (reference/tests)."""
+    x = q8.view(torch.float8_e4m3fn).float()
+    return (x * scales.unsqueeze(-1)).to(dtype)
 
 
 # ------------------------------------------------------------- attention
@@ -361,9 +391,12 @@ def attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                 scale: float = None,
                 out: torch.Tensor = None,
                 partial_ws: torch.Tensor = None,
-                nsplit: int = None) -> torch.Tensor:
+                nsplit: int = None,
+                k_scale: torch.Tensor = None,
+                v_scale: torch.Tensor = None) -> torch.Tensor:
     """GQA decode attention (flash-decode split-S on GPU).
-    q [B,Hq,D], caches [B,Hkv,Smax,D]."""
+    q [B,Hq,D], caches [B,Hkv,Smax,D] bf16 — or uint8 e4m3 with
+    per-row k_scale/v_scale [B,Hkv,Smax] (fp8 KV mode)."""
     B, Hq, D = q.shape
     Hkv = k_cache.shape[1]
     scale = scale if scale is not None else 1.0 / math.sqrt(D)
@@ -377,11 +410,16 @@ def attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
             partial_ws = torch.empty(B * Hq * nsplit * (D + 2),
                                      dtype=torch.float32, device=q.device)
         ops.attn_decode(out, q, k_cache, v_cache, seq_lens, scale,
-                        partial_ws, nsplit)
+                        partial_ws, nsplit, k_scale, v_scale)
         return out
     # fp32 reference
     if out is None:
         out = torch.empty_like(q)
+    if k_scale is not None:
+        k_cache = dequantize_kv_rows(k_cache,
+                                     k_scale.view(B, Hkv, -1))
+        v_cache = dequantize_kv_rows(v_cache,
+                                     v_scale.view(B, Hkv, -1))
     G = Hq // Hkv
     for b in range(B):
         length = int(seq_lens[b])

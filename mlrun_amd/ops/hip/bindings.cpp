@@ -17,6 +17,10 @@
   TORCH_CHECK((x).scalar_type() == torch::kBFloat16, #x " must be bf16")
 #define CHECK_F32(x) \
   TORCH_CHECK((x).scalar_type() == torch::kFloat32, #x " must be f32")
+
+#define CHECK_U8(x) \
+  TORCH_CHECK((x).scalar_type() == torch::kUInt8, #x " must be uint8")
+
 #define CHECK_I32(x) \
   TORCH_CHECK((x).scalar_type() == torch::kInt32, #x " must be int32")
 
@@ -187,22 +191,38 @@ void rope_kv_slab(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
 
 void rope_kv_fused(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
                    torch::Tensor positions, torch::Tensor cos_sin,
-                   int64_t hq) {
+                   int64_t hq,
+                   c10::optional<torch::Tensor> k_scale = c10::nullopt,
+                   c10::optional<torch::Tensor> v_scale = c10::nullopt) {
   CHECK_DEV(qkv); CHECK_CONTIG(qkv); CHECK_BF16(qkv);
-  CHECK_DEV(kc); CHECK_CONTIG(kc); CHECK_BF16(kc);
-  CHECK_DEV(vc); CHECK_CONTIG(vc); CHECK_BF16(vc);
+  CHECK_DEV(kc); CHECK_CONTIG(kc);
+  CHECK_DEV(vc); CHECK_CONTIG(vc);
   CHECK_DEV(positions); CHECK_CONTIG(positions); CHECK_I32(positions);
   CHECK_DEV(cos_sin); CHECK_CONTIG(cos_sin); CHECK_F32(cos_sin);
   int B = kc.size(0), Hkv = kc.size(1), Smax = kc.size(2), D = kc.size(3);
   TORCH_CHECK(qkv.dim() == 2 && qkv.size(0) == B, "qkv must be [B, rows]");
   TORCH_CHECK(qkv.size(1) >= (hq + 2 * Hkv) * D, "qkv row too small");
+  if (k_scale.has_value()) {
+    // fp8 KV mode: caches are uint8 e4m3 + per-row f32 scales
+    TORCH_CHECK(v_scale.has_value(), "fp8 KV needs both scales");
+    CHECK_U8(kc); CHECK_U8(vc);
+    CHECK_DEV(*k_scale); CHECK_CONTIG(*k_scale); CHECK_F32(*k_scale);
+    CHECK_DEV(*v_scale); CHECK_CONTIG(*v_scale); CHECK_F32(*v_scale);
+    TORCH_CHECK(k_scale->numel() >= (int64_t)B * Hkv * Smax &&
+                v_scale->numel() >= (int64_t)B * Hkv * Smax,
+                "scale tensors too small");
+    TORCH_CHECK(D == 128, "fp8 KV requires head_dim 128");
+    launch_rope_kv_fused_q8(qkv.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                            k_scale->data_ptr(), v_scale->data_ptr(),
+                            positions.data_ptr(), cos_sin.data_ptr(), B,
+                            (int)hq, Hkv, Smax, D, qkv.size(1), stream());
+    return;
+  }
+  CHECK_BF16(kc); CHECK_BF16(vc);
   launch_rope_kv_fused(qkv.data_ptr(), kc.data_ptr(), vc.data_ptr(),
                        positions.data_ptr(), cos_sin.data_ptr(), B, (int)hq,
                        Hkv, Smax, D, qkv.size(1), stream());
 }
-
-#define CHECK_U8(x) \
-  TORCH_CHECK((x).scalar_type() == torch::kUInt8, #x " must be uint8")
 
 void skinny_gemm_fp8(torch::Tensor out_bf16, torch::Tensor part_f32,
                      torch::Tensor a8, torch::Tensor a_scale,
@@ -253,11 +273,22 @@ void cast_f32_bf16(torch::Tensor out, torch::Tensor in) {
 
 void attn_decode(torch::Tensor o, torch::Tensor q, torch::Tensor kc,
                  torch::Tensor vc, torch::Tensor seq_lens, double scale,
-                 c10::optional<torch::Tensor> partial_ws, int64_t nsplit) {
+                 c10::optional<torch::Tensor> partial_ws, int64_t nsplit,
+                 c10::optional<torch::Tensor> k_scale = c10::nullopt,
+                 c10::optional<torch::Tensor> v_scale = c10::nullopt) {
   CHECK_DEV(o); CHECK_CONTIG(o); CHECK_BF16(o);
   CHECK_DEV(q); CHECK_BF16(q);
-  CHECK_DEV(kc); CHECK_CONTIG(kc); CHECK_BF16(kc);
-  CHECK_DEV(vc); CHECK_CONTIG(vc); CHECK_BF16(vc);
+  CHECK_DEV(kc); CHECK_CONTIG(kc);
+  CHECK_DEV(vc); CHECK_CONTIG(vc);
+  const bool kvq = k_scale.has_value();
+  if (kvq) {
+    TORCH_CHECK(v_scale.has_value(), "fp8 KV needs both scales");
+    CHECK_U8(kc); CHECK_U8(vc);
+    CHECK_DEV(*k_scale); CHECK_CONTIG(*k_scale); CHECK_F32(*k_scale);
+    CHECK_DEV(*v_scale); CHECK_CONTIG(*v_scale); CHECK_F32(*v_scale);
+  } else {
+    CHECK_BF16(kc); CHECK_BF16(vc);
+  }
   CHECK_DEV(seq_lens); CHECK_CONTIG(seq_lens); CHECK_I32(seq_lens);
   TORCH_CHECK(q.dim() == 3 && kc.dim() == 4, "q [B,Hq,D], kc [B,Hkv,S,D]");
   long long q_row_stride = check_head_view(q);
@@ -277,8 +308,15 @@ void attn_decode(torch::Tensor o, torch::Tensor q, torch::Tensor kc,
                 "partial_ws too small");
     ws = (float*)partial_ws->data_ptr();
   }
+  if (kvq)
+    TORCH_CHECK(k_scale->numel() >= (int64_t)B * Hkv * Smax &&
+                v_scale->numel() >= (int64_t)B * Hkv * Smax,
+                "scale tensors too small");
   launch_attn_decode(o.data_ptr(), q.data_ptr(), kc.data_ptr(),
-                     vc.data_ptr(), seq_lens.data_ptr(), B, Hq, Hkv, Smax,
+                     vc.data_ptr(),
+                     kvq ? k_scale->data_ptr() : nullptr,
+                     kvq ? v_scale->data_ptr() : nullptr,
+                     seq_lens.data_ptr(), B, Hq, Hkv, Smax,
                      (float)scale, q_row_stride, ws, (int)nsplit, stream());
 }
 
@@ -354,7 +392,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "silu(gu[:, :I]) * gu[:, I:] over a fused gate|up buffer");
   m.def("skinny_gemm", &skinny_gemm,
         "decode GEMM C=A@W^T on MFMA (bf16 in/out, split-K f32 slabs)");
-  m.def("rope_kv_fused", &rope_kv_fused,
+  m.def("rope_kv_fused", &rope_kv_fused, py::arg("qkv"), py::arg("kc"),
+        py::arg("vc"), py::arg("positions"), py::arg("cos_sin"),
+        py::arg("hq"), py::arg("k_scale") = py::none(),
+        py::arg("v_scale") = py::none(),
         "fused decode rope(q,k) + KV-cache append");
   m.def("skinny_gemm_slabs", &skinny_gemm_slabs,
         "split-K GEMM emitting f32 slabs only");
@@ -368,7 +409,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fp8-weight decode GEMM (e4m3, per-row scales)");
   m.def("quant_fp8_rows", &quant_fp8_rows,
         "dynamic per-row bf16 -> fp8 e4m3 quantization");
-  m.def("attn_decode", &attn_decode, "GQA decode attention (bf16)");
+  m.def("attn_decode", &attn_decode, py::arg("o"), py::arg("q"),
+        py::arg("kc"), py::arg("vc"), py::arg("seq_lens"),
+        py::arg("scale"), py::arg("partial_ws") = py::none(),
+        py::arg("nsplit") = 1, py::arg("k_scale") = py::none(),
+        py::arg("v_scale") = py::none(),
+        "GQA decode attention (bf16 or fp8 KV cache)");
   m.def("kv_append", &kv_append, "append token K/V into cache");
   m.def("softmax", &softmax, "row softmax (bf16)");
   m.def("tree_ensemble", &tree_ensemble, "GBDT ensemble inference (f32)");

@@ -49,6 +49,12 @@ void launch_rope_kv_fused(void* qkv, void* Kc, void* Vc,
                           int Hq, int Hkv, int Smax, int D,
                           long long row_stride, void* stream);
 
+void launch_rope_kv_fused_q8(void* qkv, void* Kc8, void* Vc8,
+                             void* kscale, void* vscale,
+                             const void* positions, const void* cos_sin,
+                             int B, int Hq, int Hkv, int Smax, int D,
+                             long long row_stride, void* stream);
+
 void launch_skinny_gemm_fp8(void* out_bf16, void* part_f32, const void* A8,
                             const void* a_scale, const void* W8,
                             const void* w_scale, int M, int N, int K,
@@ -63,8 +69,9 @@ void launch_cast_f32_bf16(void* out, const void* in, long long n,
 void launch_zero_f32(void* p, long long n, void* stream);
 
 void launch_attn_decode(void* O, const void* Q, const void* Kc,
-                        const void* Vc, const void* seq_lens, int B, int Hq,
-                        int Hkv, int Smax, float scale,
+                        const void* Vc, const void* kscale,
+                        const void* vscale, const void* seq_lens, int B,
+                        int Hq, int Hkv, int Smax, float scale,
                         long long q_row_stride, float* partial_ws,
                         int nsplit, void* stream);
 

@@ -1172,6 +1172,56 @@ __global__ void rope_kv_fused_kernel(
   }
 }
 
+// fp8-KV variant: same rope + append, but the caches hold OCP e4m3
+// bytes with one f32 scale per row (token): scale = row amax / 448.
+// Halves KV-cache HBM traffic and memory; decode attention dequants
+// while staging through LDS.
+__global__ void rope_kv_fused_q8_kernel(
+    unsigned short* __restrict__ qkv, unsigned char* __restrict__ Kc8,
+    unsigned char* __restrict__ Vc8, float* __restrict__ kscale,
+    float* __restrict__ vscale, const int* __restrict__ positions,
+    const float* __restrict__ cos_sin, int Hq, int Hkv, int Smax, int D,
+    long long row_stride) {
+  const int b = blockIdx.x;
+  const int y = blockIdx.y;
+  const int i = threadIdx.x;  // 0..D/2-1 (one wave when D=128)
+  const int half = D >> 1;
+  const int pos = positions[b];
+  unsigned short* row = qkv + (size_t)b * row_stride;
+  float lo_f = 0.f, hi_f = 0.f;
+  if (y < Hq + Hkv) {
+    unsigned short* head = row + (size_t)y * D;
+    const float c = cos_sin[((size_t)pos * half + i) * 2];
+    const float sn = cos_sin[((size_t)pos * half + i) * 2 + 1];
+    const float a = bf2f(head[i]);
+    const float bvf = bf2f(head[i + half]);
+    lo_f = a * c - bvf * sn;
+    hi_f = bvf * c + a * sn;
+    head[i] = f2bf(lo_f);
+    head[i + half] = f2bf(hi_f);
+    if (y < Hq) return;
+  } else {
+    const unsigned short* src = row + (size_t)y * D;
+    lo_f = bf2f(src[i]);
+    hi_f = bf2f(src[i + half]);
+  }
+  // wave-reduce the row amax (D/2 lanes x 2 values)
+  float amax = fmaxf(fabsf(lo_f), fabsf(hi_f));
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    amax = fmaxf(amax, __shfl_xor(amax, off, 64));
+  const float qs = (amax > 0.f) ? amax / 448.f : 1.f;
+  const float qinv = 1.f / qs;
+  const bool is_k = (y < Hq + Hkv);
+  const int kvh = is_k ? (y - Hq) : (y - Hq - Hkv);
+  const size_t rowi = ((size_t)b * Hkv + kvh) * Smax + pos;
+  unsigned char* dst = (is_k ? Kc8 : Vc8) + rowi * D;
+  __hip_fp8_e4m3 lo8(lo_f * qinv), hi8(hi_f * qinv);
+  dst[i] = lo8.__x;
+  dst[i + half] = hi8.__x;
+  if (i == 0) (is_k ? kscale : vscale)[rowi] = qs;
+}
+
 void launch_rope_kv_fused(void* qkv, void* Kc, void* Vc,
                           const void* positions, const void* cos_sin, int B,
                           int Hq, int Hkv, int Smax, int D,
@@ -1181,6 +1231,19 @@ void launch_rope_kv_fused(void* qkv, void* Kc, void* Vc,
                      (unsigned short*)qkv, (unsigned short*)Kc,
                      (unsigned short*)Vc, (const int*)positions,
                      (const float*)cos_sin, Hq, Hkv, Smax, D, row_stride);
+}
+
+void launch_rope_kv_fused_q8(void* qkv, void* Kc8, void* Vc8,
+                             void* kscale, void* vscale,
+                             const void* positions, const void* cos_sin,
+                             int B, int Hq, int Hkv, int Smax, int D,
+                             long long row_stride, void* stream) {
+  hipLaunchKernelGGL(rope_kv_fused_q8_kernel, dim3(B, Hq + 2 * Hkv),
+                     dim3(D / 2), 0, (hipStream_t)stream,
+                     (unsigned short*)qkv, (unsigned char*)Kc8,
+                     (unsigned char*)Vc8, (float*)kscale, (float*)vscale,
+                     (const int*)positions, (const float*)cos_sin, Hq,
+                     Hkv, Smax, D, row_stride);
 }
 
 // ---------------------------------------------------------------------
@@ -1204,10 +1267,12 @@ struct AttnState {
 
 // shared chunk loop: stage [s_begin, s_final) through kbuf/vbuf and
 // accumulate the online-softmax state for this wave's head
-template <int SCHUNK>
+struct uchar16v { unsigned char v[16]; };
+
+template <int SCHUNK, bool KVQ>
 __device__ __forceinline__ AttnState attn_chunk_loop(
-    const unsigned short* __restrict__ kbase,
-    const unsigned short* __restrict__ vbase,
+    const void* __restrict__ kbase_, const void* __restrict__ vbase_,
+    const float* __restrict__ kscale, const float* __restrict__ vscale,
     const float* qf, float scale, int s_begin, int s_final,
     int wave, int lane, int tid, int nthreads,
     unsigned short (*kbuf)[128], unsigned short (*vbuf)[128],
@@ -1217,16 +1282,42 @@ __device__ __forceinline__ AttnState attn_chunk_loop(
   const int d0 = lane * 2;
   for (int s0 = s_begin; s0 < s_final; s0 += SCHUNK) {
     const int cnt = min(SCHUNK, s_final - s0);
-    // --- cooperative stage: every thread loads 16B vectors of K and V
-    const int total_vec = cnt * (D / 8);
-    for (int i = tid; i < total_vec; i += nthreads) {
-      const int row = i >> 4;
-      const int col = (i & 15) * 8;
-      const size_t off = (size_t)(s0 + row) * D + col;
-      *reinterpret_cast<ushort8*>(&kbuf[row][col]) =
-          *reinterpret_cast<const ushort8*>(kbase + off);
-      *reinterpret_cast<ushort8*>(&vbuf[row][col]) =
-          *reinterpret_cast<const ushort8*>(vbase + off);
+    if (!KVQ) {
+      // --- cooperative stage: every thread loads 16B K/V vectors
+      const unsigned short* kbase = (const unsigned short*)kbase_;
+      const unsigned short* vbase = (const unsigned short*)vbase_;
+      const int total_vec = cnt * (D / 8);
+      for (int i = tid; i < total_vec; i += nthreads) {
+        const int row = i >> 4;
+        const int col = (i & 15) * 8;
+        const size_t off = (size_t)(s0 + row) * D + col;
+        *reinterpret_cast<ushort8*>(&kbuf[row][col]) =
+            *reinterpret_cast<const ushort8*>(kbase + off);
+        *reinterpret_cast<ushort8*>(&vbuf[row][col]) =
+            *reinterpret_cast<const ushort8*>(vbase + off);
+      }
+    } else {
+      // --- fp8 cache: 16B loads carry 16 dims; dequant into LDS bf16
+      const unsigned char* k8 = (const unsigned char*)kbase_;
+      const unsigned char* v8 = (const unsigned char*)vbase_;
+      const int total_vec = cnt * (D / 16);
+      for (int i = tid; i < total_vec; i += nthreads) {
+        const int row = i >> 3;
+        const int col = (i & 7) * 16;
+        const size_t off = (size_t)(s0 + row) * D + col;
+        const uchar16v kq = *reinterpret_cast<const uchar16v*>(k8 + off);
+        const uchar16v vq = *reinterpret_cast<const uchar16v*>(v8 + off);
+        const float ks = kscale[s0 + row];
+        const float vs = vscale[s0 + row];
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          __hip_fp8_e4m3 kf8, vf8;
+          kf8.__x = kq.v[j];
+          vf8.__x = vq.v[j];
+          kbuf[row][col + j] = f2bf(float(kf8) * ks);
+          vbuf[row][col + j] = f2bf(float(vf8) * vs);
+        }
+      }
     }
     __syncthreads();
     // --- scores from LDS (16-lane groups each cover all 128 dims)
@@ -1269,11 +1360,11 @@ __device__ __forceinline__ AttnState attn_chunk_loop(
   return st;
 }
 
-template <int SCHUNK>
+template <int SCHUNK, bool KVQ = false>
 __global__ void attn_decode_kernel(
     unsigned short* __restrict__ O, const unsigned short* __restrict__ Q,
-    const unsigned short* __restrict__ Kc,
-    const unsigned short* __restrict__ Vc,
+    const void* __restrict__ Kc, const void* __restrict__ Vc,
+    const float* __restrict__ kscale, const float* __restrict__ vscale,
     const int* __restrict__ seq_lens, int B, int Hq, int Hkv, int Smax,
     float scale, long long q_row_stride) {
   constexpr int D = 128;
@@ -1300,9 +1391,12 @@ __global__ void attn_decode_kernel(
     for (int j = 0; j < 8; ++j) qf[j] = bf2f(qv.v[j]);
   }
 
-  const size_t kv_base = ((size_t)b * Hkv + kvh) * Smax * D;
-  const AttnState st = attn_chunk_loop<SCHUNK>(
-      Kc + kv_base, Vc + kv_base, qf, scale, 0, len, wave, lane,
+  const size_t kv_row = ((size_t)b * Hkv + kvh) * Smax;
+  const AttnState st = attn_chunk_loop<SCHUNK, KVQ>(
+      (const char*)Kc + kv_row * D * (KVQ ? 1 : 2),
+      (const char*)Vc + kv_row * D * (KVQ ? 1 : 2),
+      kscale ? kscale + kv_row : nullptr,
+      vscale ? vscale + kv_row : nullptr, qf, scale, 0, len, wave, lane,
       threadIdx.x, G * 64, kbuf, vbuf, scores);
 
   const float inv = (st.l > 0.f) ? 1.f / st.l : 0.f;
@@ -1319,11 +1413,11 @@ __global__ void attn_decode_kernel(
 // peaked at 128 workgroups on 256 CUs).
 // partial layout: [B, Hq, nsplit, D+2] f32
 // ---------------------------------------------------------------------
-template <int SCHUNK>
+template <int SCHUNK, bool KVQ = false>
 __global__ void attn_decode_split_kernel(
     float* __restrict__ partial, const unsigned short* __restrict__ Q,
-    const unsigned short* __restrict__ Kc,
-    const unsigned short* __restrict__ Vc,
+    const void* __restrict__ Kc, const void* __restrict__ Vc,
+    const float* __restrict__ kscale, const float* __restrict__ vscale,
     const int* __restrict__ seq_lens, int B, int Hq, int Hkv, int Smax,
     float scale, long long q_row_stride, int nsplit) {
   constexpr int D = 128;
@@ -1352,10 +1446,13 @@ __global__ void attn_decode_split_kernel(
 #pragma unroll
     for (int j = 0; j < 8; ++j) qf[j] = bf2f(qv.v[j]);
   }
-  const size_t kv_base = ((size_t)b * Hkv + kvh) * Smax * D;
-  const AttnState st = attn_chunk_loop<SCHUNK>(
-      Kc + kv_base, Vc + kv_base, qf, scale, s_begin, s_final, wave,
-      lane, threadIdx.x, G * 64, kbuf, vbuf, scores);
+  const size_t kv_row = ((size_t)b * Hkv + kvh) * Smax;
+  const AttnState st = attn_chunk_loop<SCHUNK, KVQ>(
+      (const char*)Kc + kv_row * D * (KVQ ? 1 : 2),
+      (const char*)Vc + kv_row * D * (KVQ ? 1 : 2),
+      kscale ? kscale + kv_row : nullptr,
+      vscale ? vscale + kv_row : nullptr, qf, scale, s_begin, s_final,
+      wave, lane, threadIdx.x, G * 64, kbuf, vbuf, scores);
 
   prow[lane * 2] = st.acc0;
   prow[lane * 2 + 1] = st.acc1;
@@ -1404,27 +1501,36 @@ static int attn_schunk_env() {
 }
 
 void launch_attn_decode(void* O, const void* Q, const void* Kc,
-                        const void* Vc, const void* seq_lens, int B, int Hq,
-                        int Hkv, int Smax, float scale,
+                        const void* Vc, const void* kscale,
+                        const void* vscale, const void* seq_lens, int B,
+                        int Hq, int Hkv, int Smax, float scale,
                         long long q_row_stride, float* partial_ws,
                         int nsplit, void* stream) {
   const int G = Hq / Hkv;
   const bool c64 = attn_schunk_env() >= 64;
+  const bool kvq = kscale != nullptr;
   if (nsplit <= 1 || partial_ws == nullptr) {
-    auto* kern = c64 ? attn_decode_kernel<64> : attn_decode_kernel<32>;
+    auto* kern = kvq ? (c64 ? attn_decode_kernel<64, true>
+                            : attn_decode_kernel<32, true>)
+                     : (c64 ? attn_decode_kernel<64, false>
+                            : attn_decode_kernel<32, false>);
     hipLaunchKernelGGL(kern, dim3(B * Hkv), dim3(G * 64), 0,
                        (hipStream_t)stream, (unsigned short*)O,
-                       (const unsigned short*)Q, (const unsigned short*)Kc,
-                       (const unsigned short*)Vc, (const int*)seq_lens, B,
+                       (const unsigned short*)Q, Kc, Vc,
+                       (const float*)kscale, (const float*)vscale,
+                       (const int*)seq_lens, B,
                        Hq, Hkv, Smax, scale, q_row_stride);
     return;
   }
-  auto* kern = c64 ? attn_decode_split_kernel<64>
-                   : attn_decode_split_kernel<32>;
+  auto* kern = kvq ? (c64 ? attn_decode_split_kernel<64, true>
+                          : attn_decode_split_kernel<32, true>)
+                   : (c64 ? attn_decode_split_kernel<64, false>
+                          : attn_decode_split_kernel<32, false>);
   hipLaunchKernelGGL(kern, dim3(B * Hkv, nsplit),
                      dim3(G * 64), 0, (hipStream_t)stream, partial_ws,
-                     (const unsigned short*)Q, (const unsigned short*)Kc,
-                     (const unsigned short*)Vc, (const int*)seq_lens, B, Hq,
+                     (const unsigned short*)Q, Kc, Vc,
+                     (const float*)kscale, (const float*)vscale,
+                     (const int*)seq_lens, B, Hq,
                      Hkv, Smax, scale, q_row_stride, nsplit);
   hipLaunchKernelGGL(attn_decode_combine_kernel, dim3(B * Hq), dim3(64), 0,
                      (hipStream_t)stream, (unsigned short*)O, partial_ws,

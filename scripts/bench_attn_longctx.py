@@ -40,6 +40,7 @@ def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--batch", type=int, default=32)
     parser.add_argument("--iters", type=int, default=50)
+    parser.add_argument("--kv", default="bf16", choices=["bf16", "fp8"])
     args = parser.parse_args()
     assert torch.cuda.is_available()
     device = "cuda:0"
@@ -56,16 +57,24 @@ def main():
         k = torch.randn(B, Hkv, Smax, D, device=device,
                         dtype=torch.bfloat16)
         v = torch.randn_like(k)
+        ks = vs = None
+        if args.kv == "fp8":
+            k, ks = ops.quantize_kv_rows(k)
+            v, vs = ops.quantize_kv_rows(v)
+            k, v = k.contiguous(), v.contiguous()
+            ks, vs = ks.contiguous(), vs.contiguous()
         seq_lens = torch.full((B,), S, device=device, dtype=torch.int32)
-        out = torch.empty_like(q)
-        kv_bytes = 2 * B * Hkv * S * D * 2
+        out = torch.empty(B, Hq, D, device=device, dtype=torch.bfloat16)
+        bytes_per = 1 if args.kv == "fp8" else 2
+        kv_bytes = 2 * B * Hkv * S * D * bytes_per
         times = {}
         for nsplit in [1, 2, 4, 8, 16]:
             ws = torch.empty(B, Hq, nsplit, D + 2, device=device,
                              dtype=torch.float32) if nsplit > 1 else None
             times[nsplit] = time_call(
                 lambda: ops.attn_decode(q, k, v, seq_lens, out=out,
-                                        partial_ws=ws, nsplit=nsplit),
+                                        partial_ws=ws, nsplit=nsplit,
+                                        k_scale=ks, v_scale=vs),
                 args.iters)
         best = min(times, key=times.get)
         heur = ops.pick_attn_nsplit(B, Hkv)

@@ -125,3 +125,27 @@ class TestDynamicBatching:
             graph.steps else list(graph.steps.values())[0].routes["gen"]
         model = route.object
         assert model.engine_calls <= 4, model.engine_calls
+
+
+class TestFP8KVCache:
+    def test_quantize_roundtrip(self):
+        from mlrun_amd import ops
+
+        x = torch.randn(4, 8, 128, dtype=torch.bfloat16)
+        q8, s = ops.quantize_kv_rows(x)
+        assert q8.dtype == torch.uint8 and s.shape == (4, 8)
+        back = ops.dequantize_kv_rows(q8, s)
+        rel = (back - x.float()).abs().max() / x.float().abs().max()
+        assert rel < 0.06
+
+    def test_tiny_engine_fp8_kv_matches_bf16(self):
+        cfg = LlamaConfig.tiny()
+        e1 = LlamaDecodeEngine(cfg, 2, device="cpu", use_graph=False,
+                               seed=7)
+        e2 = LlamaDecodeEngine(cfg, 2, device="cpu", use_graph=False,
+                               seed=7, kv_dtype="fp8")
+        toks = torch.randint(0, cfg.vocab_size, (2, 12),
+                             generator=torch.Generator().manual_seed(2))
+        o1 = e1.generate(toks, max_new_tokens=6)
+        o2 = e2.generate(toks, max_new_tokens=6)
+        assert (o1 == o2).float().mean().item() >= 0.5

@@ -124,3 +124,43 @@ class TestFP8Serving:
         out_fp8 = fp8.generate(prompt, max_new_tokens=4).cpu()
         match = (out_bf16 == out_fp8).float().mean().item()
         assert match >= 0.5, f"fp8 decode diverged: match={match}"
+
+
+@requires_gpu
+class TestFP8KVServing:
+    def test_fp8_kv_decode_close_to_bf16(self):
+        """fp8 KV cache must track the bf16-cache engine closely on
+        identical weights (opt-in mode; headline stays bf16)."""
+        from mlrun_amd.models.llama import LlamaConfig, LlamaDecodeEngine
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        prompt = torch.randint(0, 2000, (4, 8),
+                               generator=torch.Generator().manual_seed(5))
+        bf16 = LlamaDecodeEngine(cfg, batch_size=4, device="cuda:0",
+                                 use_graph=False, seed=41)
+        out_bf16 = bf16.generate(prompt, max_new_tokens=4).cpu()
+        q8 = LlamaDecodeEngine(cfg, batch_size=4, device="cuda:0",
+                               use_graph=False, seed=41, kv_dtype="fp8")
+        out_q8 = q8.generate(prompt, max_new_tokens=4).cpu()
+        match = (out_bf16 == out_q8).float().mean().item()
+        assert match >= 0.5, f"fp8 KV decode diverged: match={match}"
+
+    def test_fp8_kv_under_hipgraph(self):
+        from mlrun_amd.models.llama import LlamaConfig, LlamaDecodeEngine
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        prompt = torch.randint(0, 2000, (4, 8),
+                               generator=torch.Generator().manual_seed(9))
+        eager = LlamaDecodeEngine(cfg, batch_size=4, device="cuda:0",
+                                  use_graph=False, seed=17,
+                                  kv_dtype="fp8")
+        graphed = LlamaDecodeEngine(cfg, batch_size=4, device="cuda:0",
+                                    use_graph=True, seed=17,
+                                    kv_dtype="fp8")
+        out_e = eager.generate(prompt, max_new_tokens=4).cpu()
+        out_g = graphed.generate(prompt, max_new_tokens=4).cpu()
+        assert torch.equal(out_e, out_g)

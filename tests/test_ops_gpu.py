@@ -275,3 +275,67 @@ class TestFP8Gemm:
         got_v = got8.cpu().view(torch.float8_e4m3fn).float()
         assert (ref_v - got_v).abs().max().item() <= 8.0  # 1-ulp at amax
         assert (ref_v != got_v).float().mean().item() < 0.02
+
+
+@requires_gpu
+class TestFP8KV:
+    """fp8 KV cache: e4m3 bytes + per-row scales (quantized append +
+    dequant-while-staging attention)."""
+
+    def test_rope_kv_fused_q8_matches_python_quant(self):
+        from mlrun_amd import ops
+
+        B, Hq, Hkv, D, Smax = 4, 8, 2, 128, 64
+        qkv = _rand_bf16(B, (Hq + 2 * Hkv) * D, seed=33)
+        positions = torch.tensor([0, 3, 10, 63], dtype=torch.int32)
+        table = ops.build_rope_cos_sin(Smax, D)
+
+        # CPU reference path (python rope + quantize_kv_rows)
+        kc_ref = torch.zeros(B, Hkv, Smax, D, dtype=torch.uint8)
+        vc_ref = torch.zeros_like(kc_ref)
+        ks_ref = torch.ones(B, Hkv, Smax)
+        vs_ref = torch.ones(B, Hkv, Smax)
+        qkv_ref = qkv.clone()
+        ops.rope_kv_fused(qkv_ref, kc_ref, vc_ref, positions, table, Hq,
+                          k_scale=ks_ref, v_scale=vs_ref)
+
+        kc = torch.zeros_like(kc_ref).cuda()
+        vc = torch.zeros_like(vc_ref).cuda()
+        ks = torch.ones(B, Hkv, Smax).cuda()
+        vs = torch.ones(B, Hkv, Smax).cuda()
+        qkv_gpu = qkv.cuda()
+        ops.rope_kv_fused(qkv_gpu, kc, vc, positions.cuda(), table.cuda(),
+                          Hq, k_scale=ks, v_scale=vs)
+        # compare DEQUANTIZED rows (encodings may differ by 1 ulp)
+        deq_ref = ops.dequantize_kv_rows(kc_ref, ks_ref)
+        deq_gpu = ops.dequantize_kv_rows(kc.cpu(), ks.cpu())
+        assert torch.allclose(deq_ref, deq_gpu, atol=3e-2, rtol=3e-2)
+        deq_ref = ops.dequantize_kv_rows(vc_ref, vs_ref)
+        deq_gpu = ops.dequantize_kv_rows(vc.cpu(), vs.cpu())
+        assert torch.allclose(deq_ref, deq_gpu, atol=3e-2, rtol=3e-2)
+
+    @pytest.mark.parametrize("b,hq,hkv,s", [(2, 8, 2, 64),
+                                            (4, 32, 8, 300),
+                                            (16, 32, 8, 1024)])
+    def test_attn_decode_q8_vs_fp32(self, b, hq, hkv, s):
+        from mlrun_amd import ops
+
+        D = 128
+        torch.manual_seed(b * 7 + s)
+        q = _rand_bf16(b, hq, D)
+        smax = s + 17
+        kc = _rand_bf16(b, hkv, smax, D)
+        vc = _rand_bf16(b, hkv, smax, D)
+        k8, ks = ops.quantize_kv_rows(kc)
+        v8, vs = ops.quantize_kv_rows(vc)
+        seq_lens = torch.randint(1, s + 1, (b,), dtype=torch.int32)
+        # fp32 reference over the DEQUANTIZED caches (isolates the
+        # kernel's staging/dequant from the quantization error)
+        ref = ops.attn_decode(q, k8, v8, seq_lens,
+                              k_scale=ks, v_scale=vs)
+        got = ops.attn_decode(q.cuda(), k8.contiguous().cuda(),
+                              v8.contiguous().cuda(), seq_lens.cuda(),
+                              k_scale=ks.contiguous().cuda(),
+                              v_scale=vs.contiguous().cuda()).cpu()
+        assert torch.allclose(ref.float(), got.float(), atol=4e-2,
+                              rtol=4e-2)
