@@ -13,6 +13,7 @@ Dispatch policy:
 """
 
 import math
+import os
 import typing
 
 import torch
@@ -175,6 +176,15 @@ _GEMM_PLAN_TABLE = {
     (4096, 14336): (4, 2),    # down
     (128256, 4096): (1, 0),   # lm_head
 }
+
+# e2e A/B override: MLRUN_GEMM_PLAN="6144x4096=2,2;4096x4096=4,2"
+_plan_env = os.environ.get("MLRUN_GEMM_PLAN", "")
+if _plan_env:
+    for _cell in _plan_env.split(";"):
+        _shape, _plan = _cell.split("=")
+        _n, _k = _shape.split("x")
+        _ks, _var = _plan.split(",")
+        _GEMM_PLAN_TABLE[(int(_n), int(_k))] = (int(_ks), int(_var))
 
 
 def pick_gemm_plan(M: int, N: int, K: int) -> tuple:
