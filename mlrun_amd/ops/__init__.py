@@ -170,7 +170,8 @@ _GEMM_PLAN_TABLE = {
     # picks cells that regress the bench by ~6% (in-graph execution
     # state differs: back-to-back kernels, dirty L2, sustained clocks).
     # Three e2e A/Bs confirmed this table; treat e2e as ground truth.
-    (6144, 4096): (1, 2),     # qkv
+    (6144, 4096): (4, 2),     # qkv (ks4: 768 WGs vs 192 — re-tuned
+    #   after the attention/LDS round; e2e 126.4 -> 127.9 req/s)
     (4096, 4096): (2, 2),     # wo
     (28672, 4096): (1, 2),    # gate|up
     (4096, 14336): (4, 2),    # down
