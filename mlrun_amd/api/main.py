@@ -336,6 +336,42 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
                            for t, stats in series],
                 "current": processor.endpoint_stats(endpoint_id)}
 
+    @app.get("/api/v1/monitoring/memory")
+    async def memory_report():
+        """Process + GPU memory report (reference:
+        server/api/utils/memory_reports.py)."""
+        import resource
+
+        import psutil
+
+        proc = psutil.Process()
+        report = {
+            "rss_bytes": proc.memory_info().rss,
+            "vms_bytes": proc.memory_info().vms,
+            "peak_rss_bytes": resource.getrusage(
+                resource.RUSAGE_SELF).ru_maxrss * 1024,
+            "open_files": len(proc.open_files()),
+            "threads": proc.num_threads(),
+            "gpus": [],
+        }
+        try:
+            import torch
+
+            if torch.cuda.is_available():
+                for i in range(torch.cuda.device_count()):
+                    free, total = torch.cuda.mem_get_info(i)
+                    report["gpus"].append({
+                        "device": i, "free_bytes": free,
+                        "total_bytes": total,
+                        "allocated_bytes":
+                            torch.cuda.memory_allocated(i),
+                        "reserved_bytes":
+                            torch.cuda.memory_reserved(i),
+                    })
+        except Exception:
+            pass
+        return report
+
     # ------------------------------------------------------------ alerts
     @app.put("/api/v1/projects/{project}/alerts/{name}")
     async def store_alert(project: str, name: str, request: Request):

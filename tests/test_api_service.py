@@ -219,3 +219,10 @@ class TestRuntimeResources:
                                   "page_size": 3}).json()
         assert len(resp["runs"]) == 3
         assert resp["pagination"]["total"] == 7
+
+
+def test_memory_report(client):
+    resp = client.get("/api/v1/monitoring/memory")
+    assert resp.status_code == 200
+    body = resp.json()
+    assert body["rss_bytes"] > 0 and body["threads"] >= 1
