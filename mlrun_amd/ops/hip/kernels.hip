@@ -408,13 +408,14 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 // LDS.  4x the workgroup count of skinny_gemm_kernel at the same
 // split-K slab traffic -> 4x the waves/SIMD for latency hiding
 // (profile: N=4096 projections were latency-bound at 2 waves/SIMD).
-template <bool SPLIT, int MT, int UNR2 = 4>
+template <bool SPLIT, int MT, int UNR2 = 4, int NB = 2>
 __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
     void* __restrict__ out, const unsigned short* __restrict__ A,
     const unsigned short* __restrict__ W, int M, int N, int K, int ksplit) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int n0 = blockIdx.x * 32;  // one 32-wide tile per block
+  constexpr int NT = 16 * NB;      // n-tile width (16 or 32)
+  const int n0 = blockIdx.x * NT;
   if (n0 >= N) return;
   int cbegin = 0, cend = K;
   if (SPLIT) {
@@ -439,6 +440,9 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
   //    only in C cells (m>=M / n>=N) the epilogue never writes
   //  - MT in {1,2} A-row tiles: M<=16 or 17..32 (decode batch 32
   //    doubles served requests per weight pass at equal HBM traffic)
+  //  - NB=1 halves the n-tile to 16: double the workgroup count at
+  //    UNCHANGED per-wave K length (grid-starved shapes like qkv) at
+  //    the cost of re-reading A once more (L2-resident, tiny)
   const int arow = lane & 15;
   const int kb = (lane >> 4) * 8;
   const int brow0 = n0 + (lane & 15);
@@ -458,13 +462,13 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
   const unsigned short* bptr0 =
       W + (size_t)min(brow0, N - 1) * K + kb;
   const unsigned short* bptr1 =
-      W + (size_t)min(brow1, N - 1) * K + kb;
+      W + (size_t)min(NB == 2 ? brow1 : brow0, N - 1) * K + kb;
 
 #ifndef MLRUN_GEMM_UNR1
 #define MLRUN_GEMM_UNR1 16
 #endif
   constexpr int UNR = (MT == 1) ? MLRUN_GEMM_UNR1
-                                : (MT == 2 ? UNR2 : 2);
+                                : (MT == 2 ? (NB == 1 ? 8 : UNR2) : 2);
   int k = kbegin;
   const int kend8 = kbegin + ((kend - kbegin) & ~(UNR * 32 - 1));
   for (; k < kend8; k += UNR * 32) {
@@ -476,7 +480,8 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
         af[u][t] =
             *reinterpret_cast<const short8v*>(aptr[t] + k + u * 32);
       bf0[u] = *reinterpret_cast<const short8v*>(bptr0 + k + u * 32);
-      bf1[u] = *reinterpret_cast<const short8v*>(bptr1 + k + u * 32);
+      if (NB == 2)
+        bf1[u] = *reinterpret_cast<const short8v*>(bptr1 + k + u * 32);
     }
 #pragma unroll
     for (int u = 0; u < UNR; ++u) {
@@ -484,26 +489,30 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
       for (int t = 0; t < MT; ++t) {
         acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[u][t], bf0[u], acc0[t], 0, 0, 0);
-        acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[u][t], bf1[u], acc1[t], 0, 0, 0);
+        if (NB == 2)
+          acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[u][t], bf1[u], acc1[t], 0, 0, 0);
       }
     }
   }
   for (; k < kend; k += 32) {
     short8v bf0 = *reinterpret_cast<const short8v*>(bptr0 + k);
-    short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
       short8v af = *reinterpret_cast<const short8v*>(aptr[t] + k);
       acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0[t],
                                                         0, 0, 0);
-      acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1[t],
-                                                        0, 0, 0);
+      if (NB == 2) {
+        short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
+        acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1,
+                                                          acc1[t], 0, 0,
+                                                          0);
+      }
     }
   }
 
   // combine the 4 waves' partials through LDS
-  __shared__ float comb[4][16 * MT][32];  // [wave][m][n] 8/16 KiB
+  __shared__ float comb[4][16 * MT][NT];  // [wave][m][n] 4-16 KiB
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
 #pragma unroll
@@ -511,16 +520,18 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       comb[wave][16 * t + crow_base + r][ccol] = acc0[t][r];
-      comb[wave][16 * t + crow_base + r][ccol + 16] = acc1[t][r];
+      if (NB == 2)
+        comb[wave][16 * t + crow_base + r][ccol + 16] = acc1[t][r];
     }
   __syncthreads();
   if (wave == 0) {
-    // 64 lanes fold 16*MT x 32 cells: lane covers 8*MT cells
+    // 64 lanes fold 16*MT x NT cells: lane covers 4*MT*NB cells
+    constexpr int CPL = 4 * MT * NB;
 #pragma unroll
-    for (int c = 0; c < 8 * MT; ++c) {
-      const int cell = lane * 8 * MT + c;
-      const int m = cell >> 5;
-      const int n = cell & 31;
+    for (int c = 0; c < CPL; ++c) {
+      const int cell = lane * CPL + c;
+      const int m = cell / NT;
+      const int n = cell % NT;
       if (m >= M || n0 + n >= N) continue;
       float sum = comb[0][m][n] + comb[1][m][n] + comb[2][m][n] +
                   comb[3][m][n];
@@ -809,9 +820,27 @@ static int gemm_unr2_env() {
 void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
                               int M, int N, int K, int ksplit, int variant,
                               void* stream) {
-  const int nblocks = variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
+  const int nblocks = variant == 3 ? (N + 15) / 16
+                      : variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
   const dim3 grid(nblocks, ksplit);
-  if (variant >= 1) {
+  if (variant == 3) {
+    // 16-wide n-tiles: 2x the workgroups at unchanged per-wave K
+    if (M > 32)
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 4, 4, 1>), grid,
+                         dim3(256), 0, (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+    else if (M > 16)
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 2, 4, 1>), grid,
+                         dim3(256), 0, (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+    else
+      hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 1, 4, 1>), grid,
+                         dim3(256), 0, (hipStream_t)stream, part_f32,
+                         (const unsigned short*)A, (const unsigned short*)W,
+                         M, N, K, ksplit);
+  } else if (variant >= 1) {
     if (M > 32)
       hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 4>), grid, dim3(256),
                          0, (hipStream_t)stream, part_f32,
@@ -853,9 +882,29 @@ void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                         const void* W, int M, int N, int K, int ksplit,
                         int variant, void* stream) {
   if (ksplit < 1) ksplit = 1;
-  const int nblocks = variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
+  const int nblocks = variant == 3 ? (N + 15) / 16
+                      : variant >= 1 ? (N + 31) / 32 : (N + 127) / 128;
   if (ksplit == 1) {
-    if (variant >= 1) {
+    if (variant == 3) {
+      if (M > 32)
+        hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 4, 4, 1>),
+                           dim3(nblocks), dim3(256), 0,
+                           (hipStream_t)stream, out_bf16,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, 1);
+      else if (M > 16)
+        hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 2, 4, 1>),
+                           dim3(nblocks), dim3(256), 0,
+                           (hipStream_t)stream, out_bf16,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, 1);
+      else
+        hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 1, 4, 1>),
+                           dim3(nblocks), dim3(256), 0,
+                           (hipStream_t)stream, out_bf16,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, 1);
+    } else if (variant >= 1) {
       if (M > 32)
         hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 4>),
                            dim3(nblocks), dim3(256), 0,
