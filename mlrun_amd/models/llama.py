@@ -643,6 +643,9 @@ class LlamaServer:
                 replica.buf_a8 = _torch.empty_like(first.buf_a8)
                 replica.buf_a_scale = _torch.empty_like(
                     first.buf_a_scale)
+                replica.buf_h8 = _torch.empty_like(first.buf_h8)
+                replica.buf_h8_scale = _torch.empty_like(
+                    first.buf_h8_scale)
             self.engines.append(replica)
         # load a checkpoint artifact if given (model_spec.yaml layout)
         if self.model_path:
