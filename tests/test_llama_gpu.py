@@ -164,3 +164,23 @@ class TestFP8KVServing:
         out_e = eager.generate(prompt, max_new_tokens=4).cpu()
         out_g = graphed.generate(prompt, max_new_tokens=4).cpu()
         assert torch.equal(out_e, out_g)
+
+
+@requires_gpu
+class TestFP8Replicas:
+    def test_fp8_server_with_replicas(self):
+        """Replica engines must get their own quant-sidecar buffers
+        (regression: shared-weight replicas crashed in fp8w mode)."""
+        from mlrun_amd.models.llama import LlamaConfig, LlamaServer
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        server = LlamaServer(name="m", config=cfg, batch_size=4,
+                             max_new_tokens=4, weight_dtype="fp8w",
+                             replicas=2, use_graph=False)
+        server.load()
+        out = server.predict({"inputs": [[1, 2, 3]] * 6,
+                              "max_tokens": 4})
+        assert len(out) == 6 and all(len(o) == 4 for o in out)
+        server.shutdown() if hasattr(server, "shutdown") else None
