@@ -180,7 +180,11 @@ class TestFP8Replicas:
                              max_new_tokens=4, weight_dtype="fp8w",
                              replicas=2, use_graph=False)
         server.load()
-        out = server.predict({"inputs": [[1, 2, 3]] * 6,
-                              "max_tokens": 4})
+
+        class _Ev:
+            body = {"inputs": [[1, 2, 3]] * 6, "max_tokens": 4}
+            path = "/infer"
+            id = "t"
+
+        out = server.do_event(_Ev()).body["outputs"]
         assert len(out) == 6 and all(len(o) == 4 for o in out)
-        server.shutdown() if hasattr(server, "shutdown") else None
