@@ -719,6 +719,8 @@ class LlamaServer:
                 outputs.extend(future.result(timeout=600))
         event.body = {"id": event.id, "model_name": self.name,
                       "outputs": outputs}
+        if "retrieval" in body:  # RAG upstream step metadata
+            event.body["retrieval"] = body["retrieval"]
         if self._model_logger:
             self._model_logger.push(start, {"inputs": [len(inputs)]},
                                     event.body)

@@ -30,3 +30,4 @@ from .routers import (  # noqa: F401
     VotingEnsemble,
 )
 from .remote import BatchHttpRequests, RemoteStep  # noqa: F401
+from .rag import RetrievalStep, TokenMeanEmbedder, VectorIndex  # noqa: F401

@@ -406,3 +406,81 @@ class TestEventPaths:
         server = fn.to_mock_server()
         resp = server.test("/", body={"a": {"b": 1}, "other": True})
         assert resp["a"]["b"] == 2 and resp["other"] is True
+
+
+class TestRAG:
+    def test_vector_index_topk_matches_reference(self):
+        import torch
+
+        from mlrun_amd.serving import VectorIndex
+
+        index = VectorIndex(dim=64)
+        gen = torch.Generator().manual_seed(0)
+        emb = torch.randn(32, 64, generator=gen)
+        index.add(emb, [{"id": i} for i in range(32)])
+        q = torch.randn(64, generator=gen)
+        hits = index.search(q, k=3)
+        qn = torch.nn.functional.normalize(q.unsqueeze(0), dim=-1)
+        en = torch.nn.functional.normalize(emb, dim=-1)
+        # reference via bf16-rounded fp32 matmul (index stores bf16)
+        scores = (qn.to(torch.bfloat16).float() @
+                  en.to(torch.bfloat16).float().t())[0]
+        expect = torch.topk(scores, 3).indices.tolist()
+        assert [h["payload"]["id"] for h in hits] == expect
+        assert hits[0]["score"] >= hits[1]["score"] >= hits[2]["score"]
+
+    def test_retrieval_step_enriches_prompts(self):
+        import torch
+
+        from mlrun_amd.serving import (RetrievalStep, TokenMeanEmbedder,
+                                       VectorIndex)
+
+        embedder = TokenMeanEmbedder(vocab_size=100, dim=32, seed=3)
+        index = VectorIndex(dim=32)
+        docs = [{"id": f"d{i}", "tokens": [i * 10 + j for j in range(4)]}
+                for i in range(8)]
+        index.add(torch.stack([embedder(d["tokens"])[0] for d in docs]),
+                  docs)
+        step = RetrievalStep(index=index, embedder=embedder, top_k=2)
+
+        class _Ev:
+            body = {"inputs": [[5, 6, 7]]}
+            path = "/infer"
+
+        out = step.do_event(_Ev()).body
+        assert len(out["inputs"][0]) == 3 + 2 * 4  # 2 docs x 4 tokens
+        assert out["inputs"][0][-3:] == [5, 6, 7]  # prompt preserved
+        assert len(out["retrieval"][0]) == 2
+        assert out["retrieval"][0][0]["doc_id"].startswith("d")
+
+    def test_rag_graph_end_to_end(self):
+        import torch
+
+        import mlrun_amd
+        from mlrun_amd.models.llama import LlamaServer
+        from mlrun_amd.serving import (RetrievalStep, TokenMeanEmbedder,
+                                       VectorIndex)
+
+        vocab = 200
+        embedder = TokenMeanEmbedder(vocab_size=vocab, dim=32, seed=1)
+        index = VectorIndex(dim=32)
+        gen = torch.Generator().manual_seed(7)
+        docs = [{"id": f"doc{i}",
+                 "tokens": torch.randint(0, vocab, (6,),
+                                         generator=gen).tolist()}
+                for i in range(16)]
+        index.add(torch.stack([embedder(d["tokens"])[0] for d in docs]),
+                  docs)
+        fn = mlrun_amd.new_function(name="rag", kind="serving")
+        graph = fn.set_topology("flow", engine="sync")
+        graph.add_step(RetrievalStep, name="retrieve", index=index,
+                       embedder=embedder, top_k=2)
+        graph.add_step(LlamaServer, name="llm", after="retrieve",
+                       config="tiny", batch_size=4, max_new_tokens=4,
+                       respond=True)
+        server = fn.to_mock_server()
+        prompt = torch.randint(0, vocab, (5,), generator=gen).tolist()
+        resp = server.test("/v2/models/llm/infer",
+                           body={"inputs": [prompt], "max_tokens": 4})
+        assert len(resp["outputs"][0]) == 4
+        assert len(resp["retrieval"][0]) == 2
