@@ -203,6 +203,54 @@ class MemoryStore(DataStore):
         return super().as_df(key, columns, df_module, format, **kwargs)
 
 
+class FsspecStore(DataStore):
+    """Object-store adapter over fsspec (reference: the s3/az/gcs/
+    dbfs/hdfs/http store classes of mlrun/datastore — one adapter
+    here; each scheme works when its fsspec protocol package is
+    installed, with a clear error otherwise)."""
+
+    def __init__(self, parent, name, scheme, endpoint=""):
+        super().__init__(parent, name, scheme, endpoint)
+        import fsspec
+
+        try:
+            self._fs = fsspec.filesystem(scheme)
+        except (ImportError, ValueError) as exc:
+            from ..errors import MLRunMissingDependencyError
+
+            raise MLRunMissingDependencyError(
+                f"fsspec protocol {scheme!r} is unavailable in this "
+                f"image: {exc}")
+
+    def _full(self, key):
+        base = f"{self.endpoint}/" if self.endpoint else ""
+        return f"{base}{key.lstrip('/')}" if base else key
+
+    def get(self, key, size=None, offset=0):
+        with self._fs.open(self._full(key), "rb") as stream:
+            if offset:
+                stream.seek(offset)
+            return stream.read(size) if size else stream.read()
+
+    def put(self, key, data, append=False):
+        if isinstance(data, str):
+            data = data.encode()
+        mode = "ab" if append else "wb"
+        with self._fs.open(self._full(key), mode) as stream:
+            stream.write(data)
+
+    def stat(self, key):
+        info = self._fs.info(self._full(key))
+        return {"size": info.get("size", 0),
+                "modified": info.get("mtime", 0)}
+
+    def listdir(self, key):
+        return [str(p) for p in self._fs.ls(self._full(key))]
+
+    def rm(self, key, recursive=False):
+        self._fs.rm(self._full(key), recursive=recursive)
+
+
 _schemes: dict = {}
 
 
@@ -213,6 +261,12 @@ def register_store(scheme: str, cls):
 register_store("file", FileStore)
 register_store("", FileStore)
 register_store("memory", MemoryStore)
+# object-store schemes ride the fsspec adapter (protocol packages
+# optional; errors are explicit when missing)
+for _scheme in ("s3", "gs", "gcs", "az", "abfs", "http", "https",
+                "hdfs", "webhdfs", "dbfs", "oss", "ftp", "sftp",
+                "redis"):
+    register_store(_scheme, FsspecStore)
 
 
 def schema_to_store(scheme: str):

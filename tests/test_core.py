@@ -242,3 +242,33 @@ class TestArtifacts:
         assert spec.framework == "torch"
         assert os.path.basename(model_file) == "mymodel.bin"
         assert "config.json" in extra
+
+
+class TestFsspecStore:
+    def test_local_protocol_roundtrip(self, tmp_path):
+        from mlrun_amd.datastore import FsspecStore
+
+        store = FsspecStore(None, "t", "file")
+        path = str(tmp_path / "blob.bin")
+        store.put(path, b"hello fsspec")
+        assert store.get(path) == b"hello fsspec"
+        assert store.stat(path)["size"] == 12
+        store.put(path, b"!", append=True)
+        assert store.get(path).endswith(b"fsspec!")
+        store.rm(path)
+        import os
+
+        assert not os.path.exists(path)
+
+    def test_memory_protocol(self):
+        from mlrun_amd.datastore import FsspecStore
+
+        store = FsspecStore(None, "t", "memory")
+        store.put("/m/x", b"abc")
+        assert store.get("/m/x") == b"abc"
+
+    def test_scheme_registry_covers_object_stores(self):
+        from mlrun_amd.datastore import FsspecStore, schema_to_store
+
+        for scheme in ("s3", "gcs", "az", "http", "hdfs"):
+            assert schema_to_store(scheme) is FsspecStore
