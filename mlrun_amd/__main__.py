@@ -15,7 +15,6 @@ import sys
 
 from . import __version__
 from .config import config
-from .utils import logger
 
 
 def cmd_version(args):

@@ -14,8 +14,6 @@ informers.
 """
 
 import asyncio
-import json
-import traceback
 import typing
 
 from fastapi import FastAPI, Query, Request, Response
@@ -23,9 +21,8 @@ from fastapi.responses import JSONResponse
 
 from ..config import config
 from ..db.sqldb import SQLRunDB
-from ..errors import MLRunBaseError, MLRunNotFoundError, err_to_status
-from ..model import RunStates
-from ..utils import logger, now_iso
+from ..errors import MLRunBaseError, err_to_status
+from ..utils import logger
 from .scheduler import Scheduler
 
 

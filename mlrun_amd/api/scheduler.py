@@ -16,7 +16,7 @@ import typing
 
 from ..config import config
 from ..errors import MLRunInvalidArgumentError
-from ..utils import logger, now_iso
+from ..utils import logger
 
 
 def _parse_field(field: str, low: int, high: int) -> typing.Set[int]:

@@ -12,7 +12,6 @@ dataset.py:144 DatasetArtifact.
 
 import os
 import pathlib
-import tempfile
 import typing
 
 import yaml

@@ -9,7 +9,6 @@ infer.py schema+stats inference, used by feature-store ingest and
 dataset artifacts).
 """
 
-import typing
 
 
 class ValueType:

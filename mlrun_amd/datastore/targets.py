@@ -10,7 +10,6 @@ Parity target: reference mlrun/datastore/targets.py (BaseStoreTarget
 """
 
 import os
-import typing
 
 from ..config import config
 from ..errors import MLRunInvalidArgumentError

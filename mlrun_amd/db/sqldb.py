@@ -22,8 +22,6 @@ import json
 import os
 import sqlite3
 import threading
-import time
-import typing
 
 from ..errors import MLRunConflictError, MLRunNotFoundError
 from ..model import RunStates

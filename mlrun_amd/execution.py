@@ -14,14 +14,12 @@ launcher instead of OpenMPI env).
 
 import os
 import threading
-import traceback
-import typing
 
 from .artifacts import ArtifactManager, ArtifactProducer
 from .config import config
 from .errors import MLRunInvalidArgumentError
-from .model import ModelObj, RunStates, generate_uid
-from .utils import logger, now_date, now_iso, to_date_str
+from .model import RunStates, generate_uid
+from .utils import logger, now_date, to_date_str
 
 
 class MLClientCtx:

@@ -11,15 +11,14 @@ retrieval/local_merger.py (pandas offline merge).
 """
 
 import os
-import time
 import typing
 
 from ..config import config
 from ..errors import MLRunInvalidArgumentError, MLRunNotFoundError
 from ..utils import logger, now_iso
 from .feature_set import FeatureSet
-from .online import OnlineTable, get_online_table
-from .vector import FeatureVector, OnlineVectorService, parse_feature_string
+from .online import get_online_table
+from .vector import FeatureVector, OnlineVectorService
 
 
 def _resolve_feature_set(ref) -> FeatureSet:

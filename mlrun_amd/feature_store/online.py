@@ -21,8 +21,6 @@ import typing
 import numpy as np
 import torch
 
-from ..errors import MLRunInvalidArgumentError
-from ..utils import logger
 from .feature_set import FeatureSet, parse_span
 
 

@@ -11,9 +11,7 @@ FeaturesetValidator :94, MapValues :152, Imputer :377, OneHotEncoder
 :427, DateExtractor :516, SetEventMetadata :635, DropFeatures :699).
 """
 
-import typing
 
-from ..errors import MLRunInvalidArgumentError
 
 
 class MLRunStep:

@@ -13,7 +13,7 @@ asyncio emits.
 
 import typing
 
-from ..errors import MLRunInvalidArgumentError, MLRunNotFoundError
+from ..errors import MLRunInvalidArgumentError
 from ..model import ModelObj
 from ..utils import now_iso
 

@@ -10,7 +10,6 @@ torch modules, sklearn estimators (incl. tree ensembles -> HIP
 kernel), Llama engines, plain callables.
 """
 
-import typing
 
 from ..errors import MLRunInvalidArgumentError
 

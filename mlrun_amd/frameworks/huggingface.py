@@ -11,10 +11,8 @@ serves it on the GPU in bf16; generative models route through the
 native LlamaDecodeEngine when the architecture matches.
 """
 
-import typing
 
 from ..serving.v2_serving import V2ModelServer
-from ..utils import logger
 
 
 class HuggingFaceModelServer(V2ModelServer):

@@ -11,7 +11,6 @@ source maps to {base_dir}/hub.
 """
 
 import os
-import typing
 
 import yaml
 

@@ -17,11 +17,10 @@ import os
 import typing
 
 from .config import config
-from .errors import MLRunRuntimeError
 from .execution import MLClientCtx
 from .model import RunObject, RunStates, generate_uid
 from .runtimes.generators import get_generator, selector
-from .utils import logger, now_iso
+from .utils import logger
 
 
 class BaseLauncher:

@@ -12,8 +12,6 @@ background thread).
 """
 
 import threading
-import time
-import typing
 
 from ..utils import logger, now_iso
 from .drift import compute_feature_drift, drift_status

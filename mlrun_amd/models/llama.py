@@ -21,7 +21,7 @@ Design (MI355X-first, not a port — the reference has no model code):
 
 import math
 import typing
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 

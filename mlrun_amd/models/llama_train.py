@@ -17,8 +17,6 @@ Adam moments fp32 (64 GB) + activations fits a single MI355X with
 room; DP=8 shards nothing and still fits.
 """
 
-import math
-import typing
 
 import torch
 import torch.nn as nn

@@ -14,7 +14,6 @@ Dispatch policy:
 
 import math
 import os
-import typing
 
 import torch
 

@@ -13,7 +13,6 @@ results/artifacts by the ``outputs`` spec.
 import functools
 import inspect
 import json
-import typing
 
 
 class ArtifactType:

@@ -11,8 +11,6 @@ with_limits(gpus=...)).  File-lock based so multiple processes on the
 node cooperate.
 """
 
-import contextlib
-import json
 import os
 import threading
 import time

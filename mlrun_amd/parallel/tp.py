@@ -14,13 +14,10 @@ The reference has NO tensor parallelism anywhere —
 this is a new MI355X-native capability.
 """
 
-import os
-import typing
 
 import torch
 import torch.distributed as dist
 
-from ..utils import logger
 
 
 def init_tp_group(tp_size: int = None, backend: str = None):

@@ -13,13 +13,9 @@ volume).
 """
 
 import json
-import os
 import queue
 import threading
-import typing
 
-from .config import config
-from .utils import logger, now_iso
 
 
 class OutputStream:

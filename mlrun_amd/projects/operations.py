@@ -8,7 +8,6 @@ Parity target: reference mlrun/projects/operations.py (run_function
 :60, build_function :256, deploy_function :372).
 """
 
-import typing
 
 from ..errors import MLRunInvalidArgumentError
 from .project import pipeline_context

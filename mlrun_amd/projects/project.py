@@ -10,13 +10,11 @@ set_function :2325, run :3055, log_model :1735).  Workflows run with
 the local runner (no KFP).
 """
 
-import glob
 import os
-import typing
 
 from ..artifacts import ArtifactManager, ArtifactProducer
 from ..config import config
-from ..errors import MLRunInvalidArgumentError, MLRunNotFoundError
+from ..errors import MLRunNotFoundError
 from ..model import ModelObj, generate_uid
 from ..utils import logger, normalize_name, now_iso
 

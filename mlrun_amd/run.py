@@ -9,7 +9,6 @@ code_to_function :581, get_or_create_ctx :198, import_function :330,
 function_to_module :77, run_local wrapper).
 """
 
-import importlib
 import json
 import os
 import typing
@@ -17,10 +16,10 @@ import typing
 from .config import config
 from .errors import MLRunInvalidArgumentError
 from .execution import MLClientCtx
-from .model import RunObject, RunTemplate
+from .model import RunObject
 from .runtimes import get_runtime_class, RuntimeKinds
 from .runtimes.local import META_TMPFILE_ENV, load_module
-from .utils import logger, normalize_name
+from .utils import normalize_name
 
 
 def new_function(name: str = "", project: str = "", tag: str = "",

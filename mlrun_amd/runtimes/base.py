@@ -17,9 +17,8 @@ import os
 import typing
 
 from ..config import config
-from ..errors import MLRunInvalidArgumentError, MLRunRuntimeError
-from ..model import ModelObj, RunObject, RunTemplate, new_task
-from ..utils import logger, normalize_name, now_iso
+from ..model import ModelObj, RunObject, RunTemplate
+from ..utils import logger, normalize_name
 
 
 class FunctionMetadata(ModelObj):

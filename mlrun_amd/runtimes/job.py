@@ -11,17 +11,11 @@ HIP_VISIBLE_DEVICES, scheduled by the node-local GPU allocator
 (mlrun_amd/parallel/scheduler.py).
 """
 
-import json
 import os
-import subprocess
-import sys
-import tempfile
 
-from ..errors import MLRunInvalidArgumentError
-from ..model import RunObject, RunStates
-from ..utils import logger
+from ..model import RunObject
 from .base import BaseRuntime
-from .local import LocalRuntime, load_module, exec_from_params, run_exec_command
+from .local import LocalRuntime
 
 
 class KubejobRuntime(BaseRuntime):

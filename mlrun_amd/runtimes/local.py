@@ -20,7 +20,7 @@ import tempfile
 import traceback
 import typing
 
-from ..errors import MLRunInvalidArgumentError, MLRunRuntimeError
+from ..errors import MLRunInvalidArgumentError
 from ..execution import MLClientCtx
 from ..model import RunObject, RunStates
 from ..utils import logger

@@ -18,7 +18,6 @@ via ctx.is_logging_worker().
 
 import json
 import os
-import signal
 import socket
 import subprocess
 import sys
@@ -29,7 +28,7 @@ import typing
 from ..config import config
 from ..errors import MLRunRuntimeError
 from ..model import RunObject, RunStates
-from ..utils import logger, now_iso
+from ..utils import logger
 from .base import BaseRuntime
 
 

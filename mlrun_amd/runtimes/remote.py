@@ -18,7 +18,7 @@ import typing
 
 from ..errors import MLRunInvalidArgumentError
 from ..serving.server import Event, GraphServer, GraphServerHost
-from ..serving.states import TaskStep, RootFlowStep
+from ..serving.states import RootFlowStep
 from ..utils import logger
 from .base import BaseRuntime
 

@@ -16,7 +16,6 @@ queue steps.
 import typing
 
 from ..errors import MLRunInvalidArgumentError
-from ..model import ModelObj
 from ..serving.server import GraphServer, GraphServerHost, create_graph_server
 from ..serving.states import (
     FlowStep,

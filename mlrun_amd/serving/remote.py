@@ -9,7 +9,6 @@ part of a graph.  Parity: reference mlrun/serving/remote.py
 
 import concurrent.futures
 import json
-import typing
 
 from ..errors import MLRunRuntimeError
 

@@ -14,10 +14,8 @@ process executors :391-398, VotingEnsemble :480, EnrichmentModelRouter
 import concurrent.futures
 import copy
 import json
-import typing
 
 from ..errors import MLRunInvalidArgumentError
-from ..utils import logger
 
 
 class BaseModelRouter:

@@ -16,18 +16,10 @@ import traceback
 import typing
 import uuid
 
-from ..config import config
 from ..errors import MLRunInvalidArgumentError
 from ..model import ModelObj
 from ..utils import logger
-from .states import (
-    FlowStep,
-    RootFlowStep,
-    RouterStep,
-    TaskStep,
-    graph_root_setter,
-    step_from_dict,
-)
+from .states import (RootFlowStep, RouterStep, graph_root_setter, step_from_dict)
 
 
 class Event:

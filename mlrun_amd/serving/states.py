@@ -309,7 +309,7 @@ class RouterStep(TaskStep):
     def init_object(self, context, namespace, mode="sync", reset=False):
         self.class_name = self.class_name or "ModelRouter"
         if not self._object:
-            from .routers import BaseModelRouter, router_classes
+            from .routers import router_classes
 
             cls = None
             if self.class_name in router_classes:

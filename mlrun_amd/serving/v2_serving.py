@@ -19,10 +19,9 @@ device and capture their per-batch kernel chain in a hipGraph
 import threading
 import time
 import traceback
-import typing
 
 from ..errors import MLRunInvalidArgumentError
-from ..utils import logger, now_iso
+from ..utils import logger
 
 
 class V2ModelServer:

@@ -15,7 +15,6 @@ import typing
 
 from ..config import config
 from ..errors import MLRunRuntimeError
-from . import logger
 
 NATIVE_DIR = os.path.join(os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))), "native")
