@@ -344,3 +344,25 @@ class TestFP8KV:
                               v_scale=vs.contiguous().cuda()).cpu()
         assert torch.allclose(ref.float(), got.float(), atol=4e-2,
                               rtol=4e-2)
+
+
+@requires_gpu
+class TestVectorIndexGPU:
+    def test_search_uses_gemm_and_matches_cpu(self):
+        import torch
+
+        from mlrun_amd.serving import VectorIndex
+
+        gen = torch.Generator().manual_seed(4)
+        emb = torch.randn(512, 256, generator=gen)
+        payloads = [{"id": i} for i in range(512)]
+        gpu = VectorIndex(dim=256, device="cuda:0")
+        gpu.add(emb, payloads)
+        cpu = VectorIndex(dim=256, device="cpu")
+        cpu.add(emb, payloads)
+        q = torch.randn(4, 256, generator=gen)
+        got = gpu.search(q.cuda(), k=5)
+        ref = cpu.search(q, k=5)
+        for row_got, row_ref in zip(got, ref):
+            assert [h["payload"]["id"] for h in row_got] == \
+                   [h["payload"]["id"] for h in row_ref]
