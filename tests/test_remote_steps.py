@@ -1,0 +1,87 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""RemoteStep / BatchHttpRequests against a real local HTTP host, and
+the ApplicationRuntime process lifecycle."""
+
+import sys
+import time
+
+import mlrun_amd
+from mlrun_amd.serving import GraphServer, GraphServerHost, MockEvent
+from mlrun_amd.serving.remote import BatchHttpRequests, RemoteStep
+from mlrun_amd.serving.states import RootFlowStep
+
+
+class _Echo:
+    def __init__(self, context=None, name=None):
+        pass
+
+    def do_event(self, event):
+        body = event.body if isinstance(event.body, dict) else {}
+        body["echoed"] = True
+        event.body = body
+        return event
+
+
+def _make_host():
+    graph = RootFlowStep()
+    graph.add_step(_Echo, name="echo")
+    server = GraphServer(graph=graph)
+    server.init_states(namespace={"_Echo": _Echo})
+    host = GraphServerHost(server, port=0)
+    host.start()
+    return host
+
+
+class TestRemoteStep:
+    def test_calls_local_graph_host(self):
+        host = _make_host()
+        try:
+            step = RemoteStep(url=host.address, name="r")
+            event = MockEvent(body={"x": 1})
+            out = step.do_event(event)
+            assert out.body["echoed"] is True and out.body["x"] == 1
+        finally:
+            host.stop()
+
+    def test_url_expression_and_subpath(self):
+        host = _make_host()
+        try:
+            step = RemoteStep(url=host.address, subpath="$route",
+                              name="r")
+            out = step.do_event(MockEvent(body={"route": "sub/path",
+                                                "y": 2}))
+            assert out.body["y"] == 2
+        finally:
+            host.stop()
+
+    def test_batch_requests_fan_out(self):
+        host = _make_host()
+        try:
+            step = BatchHttpRequests(url=host.address, name="b",
+                                     max_in_flight=4)
+            out = step.do_event(MockEvent(body=[{"i": i}
+                                                for i in range(6)]))
+            assert len(out.body) == 6
+            assert all(item["echoed"] for item in out.body)
+            assert sorted(item["i"] for item in out.body) == list(range(6))
+        finally:
+            host.stop()
+
+
+class TestApplicationRuntime:
+    def test_deploy_and_stop(self, tmp_path):
+        app = tmp_path / "app.py"
+        app.write_text("import time\nwhile True:\n    time.sleep(0.2)\n")
+        fn = mlrun_amd.new_function(name="app", kind="application",
+                                    command=str(app))
+        address = fn.deploy()
+        assert address.startswith("pid://")
+        assert fn.is_running()
+        fn.stop()
+        deadline = time.time() + 5
+        while fn.is_running() and time.time() < deadline:
+            time.sleep(0.1)
+        assert not fn.is_running()
