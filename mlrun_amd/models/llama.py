@@ -304,9 +304,13 @@ class LlamaDecodeEngine:
                     eps=cfg.rms_eps, out=self.buf_hidden,
                     out8=h8, out_scale=h8s)
         # incoming token sits at position cache_lens; bump the length
-        # once up front so attention covers it in every layer
-        self.buf_positions.copy_(self.cache_lens)
-        self.cache_lens.add_(1)
+        # once up front so attention covers it in every layer.  Clamp
+        # to the cache end: under continuous batching idle/finished
+        # slots keep stepping until reused — they must park on the
+        # last row instead of running off the cache (graph-safe ops)
+        torch.clamp(self.cache_lens, max=self.cfg.max_seq_len - 1,
+                    out=self.buf_positions)
+        self.cache_lens.add_(1).clamp_(max=self.cfg.max_seq_len)
         positions = self.buf_positions
         # slab fusion (GPU, TP=1): split-K consumers fold the f32 slabs
         # directly.  MEASURED: fusing into the 16-row norm kernels is a
@@ -509,6 +513,72 @@ class LlamaDecodeEngine:
         logits = last @ w.lm_head.t()
         return logits
 
+    @torch.no_grad()
+    def prefill_slots(self, tokens: torch.Tensor,
+                      slots: torch.Tensor) -> torch.Tensor:
+        """Continuous-batching admission: prefill B' <= B prompts and
+        scatter their KV/state into cache rows ``slots``, leaving the
+        other slots' caches and lengths untouched.  Returns last-token
+        logits [B'].  (Runs the same prefill math on a temporary
+        B'-sized view; decode then advances ALL slots each step.)"""
+        cfg, w = self.cfg, self.weights
+        Bp, S = tokens.shape
+        slots = torch.as_tensor(slots, dtype=torch.long,
+                                device=self.device)
+        assert Bp == slots.numel() and Bp <= self.B
+        assert S < cfg.max_seq_len
+        d = cfg.head_dim
+        tokens = tokens.to(self.device)
+        x = w.embed[tokens.reshape(-1)]
+        residual = x.contiguous()
+        positions = torch.arange(S, dtype=torch.int32,
+                                 device=self.device).repeat(Bp)
+        hidden = ops.rmsnorm(residual, w.layers[0]["attn_norm"],
+                             eps=cfg.rms_eps)
+        for li, layer in enumerate(w.layers):
+            qkv = hidden @ layer["wqkv"].t()
+            q = qkv[:, :w.hq * d].reshape(Bp * S, w.hq, d).contiguous()
+            k = qkv[:, w.hq * d:(w.hq + w.hkv) * d].reshape(
+                Bp * S, w.hkv, d).contiguous()
+            v = qkv[:, (w.hq + w.hkv) * d:].reshape(Bp * S, w.hkv, d)
+            ops.rope_inplace(q, positions, self.cos_sin)
+            ops.rope_inplace(k, positions, self.cos_sin)
+            kc = k.view(Bp, S, w.hkv, d).transpose(1, 2).contiguous()
+            vc = v.view(Bp, S, w.hkv, d).transpose(1, 2).contiguous()
+            if self.kv_dtype == "fp8":
+                k8, ks = ops.quantize_kv_rows(kc)
+                v8, vs = ops.quantize_kv_rows(vc)
+                self.k_cache[li, slots, :, :S] = k8
+                self.v_cache[li, slots, :, :S] = v8
+                self.k_scale[li, slots, :, :S] = ks
+                self.v_scale[li, slots, :, :S] = vs
+            else:
+                self.k_cache[li, slots, :, :S] = kc
+                self.v_cache[li, slots, :, :S] = vc
+            qh = q.view(Bp, S, w.hq, d).transpose(1, 2)
+            attn = torch.nn.functional.scaled_dot_product_attention(
+                qh, kc, vc, is_causal=True, enable_gqa=True)
+            attn = attn.transpose(1, 2).reshape(Bp * S, w.hq * d)
+            proj = attn @ layer["wo"].t()
+            self._maybe_allreduce(proj)
+            hidden = ops.fused_add_rmsnorm(proj, layer["ffn_norm"],
+                                           residual=residual,
+                                           eps=cfg.rms_eps)
+            gu = hidden @ layer["wgu"].t()
+            act = ops.swiglu_fused(gu.contiguous())
+            down = act @ layer["wdown"].t()
+            self._maybe_allreduce(down)
+            next_norm = w.layers[li + 1]["attn_norm"] \
+                if li + 1 < cfg.num_layers else w.final_norm
+            hidden = ops.fused_add_rmsnorm(down, next_norm,
+                                           residual=residual,
+                                           eps=cfg.rms_eps)
+        self.cache_lens[slots] = S
+        last = hidden.view(Bp, S, -1)[:, -1]
+        logits = last @ w.lm_head.t()
+        self.buf_tokens[slots] = logits.argmax(dim=-1)
+        return logits
+
     # -------------------------------------------------------- generate
     @torch.no_grad()
     def generate(self, tokens: torch.Tensor, max_new_tokens: int = 32
@@ -556,7 +626,7 @@ class LlamaServer:
                  config=None, batch_size=16, max_new_tokens=32,
                  device=None, use_graph=True, batch_window_ms=0,
                  replicas=1, weight_dtype="bf16", kv_dtype="bf16",
-                 **class_args):
+                 scheduling="batch", **class_args):
         import queue as queue_mod
         import threading
 
@@ -578,6 +648,7 @@ class LlamaServer:
         self.use_graph = use_graph
         self.weight_dtype = weight_dtype
         self.kv_dtype = kv_dtype
+        self.scheduling = scheduling  # "batch" | "continuous"
         self.replicas = max(int(replicas), 1)
         self.engines: typing.List[LlamaDecodeEngine] = []
         self.batch_window_ms = batch_window_ms
@@ -669,7 +740,9 @@ class LlamaServer:
                     engine._serve_stream.synchronize()
                 else:
                     engine.capture_graph()
-            worker = threading.Thread(target=self._worker_loop,
+            target = self._continuous_loop \
+                if self.scheduling == "continuous" else self._worker_loop
+            worker = threading.Thread(target=target,
                                       args=(engine,), daemon=True,
                                       name=f"llama-worker-{self.name}-{idx}")
             worker.start()
@@ -691,6 +764,92 @@ class LlamaServer:
                 if not future.done():
                     future.set_exception(exc)
 
+    def _continuous_loop(self, engine: LlamaDecodeEngine):
+        """Token-level continuous batching: the engine decodes ALL
+        slots every step; finished slots free up and new prompts are
+        admitted at token boundaries via ``prefill_slots`` — no
+        generate-pass barrier (vLLM-style scheduling on the fixed-B
+        hipGraph).  Each request is one queue item (prompt, max_new,
+        future)."""
+        import contextlib
+        import queue as queue_mod
+
+        B = self.batch_size
+        ring_len = engine.cfg.max_seq_len
+        out_ring = torch.empty(B, ring_len, dtype=torch.int64,
+                               device=engine.device)
+        slots: list = [None] * B
+        step = 0
+
+        def stream_ctx():
+            return torch.cuda.stream(engine._serve_stream) \
+                if engine._serve_stream is not None else \
+                contextlib.nullcontext()
+
+        if engine.use_graph and engine._graph is None:
+            with stream_ctx():
+                engine.capture_graph()
+        while True:
+            free = [i for i in range(B) if slots[i] is None]
+            have_active = len(free) < B
+            taken = []
+            if free:
+                try:
+                    taken.append(self._tasks.get(block=not have_active))
+                except queue_mod.Empty:
+                    pass
+                while len(taken) < len(free):
+                    try:
+                        taken.append(self._tasks.get_nowait())
+                    except queue_mod.Empty:
+                        break
+            if taken:
+                self.engine_calls += 1
+                prompts, admit_slots = [], []
+                for (prompt, max_new, future), slot in zip(taken, free):
+                    keep = engine.cfg.max_seq_len - max_new - 1
+                    prompt = list(prompt)[-keep:]
+                    prompts.append(prompt)
+                    admit_slots.append(slot)
+                    slots[slot] = {"future": future, "max_new": max_new,
+                                   "start": step, "produced": 1}
+                max_len = max(len(p) for p in prompts)
+                tokens = torch.zeros(len(prompts), max_len,
+                                     dtype=torch.int64)
+                for i, prompt in enumerate(prompts):
+                    tokens[i, max_len - len(prompt):] = torch.tensor(
+                        prompt, dtype=torch.int64)
+                slot_ids = torch.tensor(admit_slots, dtype=torch.long)
+                with stream_ctx():
+                    engine.prefill_slots(tokens, slot_ids)
+                    out_ring[slot_ids, step % ring_len] = \
+                        engine.buf_tokens[slot_ids]
+            active = [i for i in range(B) if slots[i] is not None]
+            if not active:
+                continue
+            with stream_ctx():
+                engine.decode_step()
+                step += 1
+                out_ring[:, step % ring_len].copy_(engine.buf_tokens)
+            finished = []
+            for i in active:
+                state = slots[i]
+                state["produced"] += 1
+                if state["produced"] >= state["max_new"]:
+                    finished.append(i)
+            if finished:
+                if engine._serve_stream is not None:
+                    engine._serve_stream.synchronize()
+                for i in finished:
+                    state = slots[i]
+                    cols = torch.tensor(
+                        [(state["start"] + j) % ring_len
+                         for j in range(state["max_new"])],
+                        dtype=torch.long)
+                    result = out_ring[i, cols.to(out_ring.device)]
+                    state["future"].set_result(result.cpu().tolist())
+                    slots[i] = None
+
     def do_event(self, event):
         import concurrent.futures
         import time as _time
@@ -705,7 +864,14 @@ class LlamaServer:
         if inputs is None:
             raise ValueError('expected {"inputs": [[token ids], ...]}')
         max_new = int(body.get("max_tokens", self.max_new_tokens))
-        if self.batch_window_ms and len(inputs) < self.batch_size:
+        if self.scheduling == "continuous":
+            futures = []
+            for prompt in inputs:
+                future = concurrent.futures.Future()
+                self._tasks.put((prompt, max_new, future))
+                futures.append(future)
+            outputs = [f.result(timeout=600) for f in futures]
+        elif self.batch_window_ms and len(inputs) < self.batch_size:
             outputs = self._batched_submit(inputs, max_new)
         else:
             futures = []

@@ -63,6 +63,8 @@ def main():
     parser.add_argument("--weights", default="bf16",
                         choices=["bf16", "fp8"])
     parser.add_argument("--warmup", type=int, default=32)
+    parser.add_argument("--scheduling", default="batch",
+                        choices=["batch", "continuous"])
     args = parser.parse_args()
 
     on_gpu = torch.cuda.is_available()
@@ -77,7 +79,8 @@ def main():
                  max_new_tokens=args.gen_tokens,
                  batch_window_ms=args.batch_window_ms,
                  replicas=args.replicas,
-                 weight_dtype="fp8w" if args.weights == "fp8" else "bf16")
+                 weight_dtype="fp8w" if args.weights == "fp8" else "bf16",
+                 scheduling=args.scheduling)
     address = fn.deploy()
     print(f"serving at {address}", file=sys.stderr)
 
@@ -128,6 +131,7 @@ def main():
         "gen_tokens": args.gen_tokens,
         "tokens_per_sec": round(args.requests * args.gen_tokens / wall, 1),
         "batch_window_ms": args.batch_window_ms,
+        "scheduling": args.scheduling,
         "model": model,
         "weights": args.weights,
         "replicas": args.replicas,

@@ -149,3 +149,82 @@ class TestFP8KVCache:
         o1 = e1.generate(toks, max_new_tokens=6)
         o2 = e2.generate(toks, max_new_tokens=6)
         assert (o1 == o2).float().mean().item() >= 0.5
+
+
+class TestContinuousBatching:
+    def _servers(self):
+        cfg = LlamaConfig.tiny()
+        cont = LlamaServer(name="c", config=cfg, batch_size=4,
+                           max_new_tokens=6, scheduling="continuous",
+                           use_graph=False)
+        cont.load()
+        batch = LlamaServer(name="b", config=cfg, batch_size=4,
+                            max_new_tokens=6, use_graph=False)
+        batch.load()
+        batch.engines[0].weights.load_state_dict(
+            cont.engines[0].weights.state_dict())
+        return cont, batch
+
+    def test_matches_batch_mode(self):
+        cont, batch = self._servers()
+
+        class _Ev:
+            body = {"inputs": [[1, 2, 3], [4, 5, 6, 7], [8, 9]],
+                    "max_tokens": 5}
+            path = "/infer"
+            id = "t"
+
+        out_c = cont.do_event(_Ev()).body["outputs"]
+        out_b = batch.do_event(_Ev()).body["outputs"]
+        assert out_c == out_b
+
+    def test_slot_reuse_beyond_batch_size(self):
+        """More requests than slots: admission must recycle freed
+        slots at token boundaries."""
+        import concurrent.futures
+        import threading
+
+        cont, batch = self._servers()
+        prompts = [[i + 1, i + 2, i + 3] for i in range(10)]
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        results = {}
+
+        def one(i):
+            ev = _Ev()
+            ev.body = {"inputs": [prompts[i]], "max_tokens": 4}
+            results[i] = cont.do_event(ev).body["outputs"][0]
+
+        threads = [threading.Thread(target=one, args=(i,))
+                   for i in range(10)]
+        [t.start() for t in threads]
+        [t.join(timeout=60) for t in threads]
+        assert len(results) == 10
+        # every result must equal the batch-mode output for its prompt
+        for i in range(10):
+            ev = _Ev()
+            ev.body = {"inputs": [prompts[i]], "max_tokens": 4}
+            expect = batch.do_event(ev).body["outputs"][0]
+            assert results[i] == expect, i
+
+    def test_staggered_lengths_finish_independently(self):
+        cont, _ = self._servers()
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        import concurrent.futures
+
+        futures = []
+        with concurrent.futures.ThreadPoolExecutor(4) as pool:
+            for want in (2, 5, 3, 7):
+                ev = _Ev()
+                ev.body = {"inputs": [[1, 2, 3]], "max_tokens": want}
+                futures.append((want, pool.submit(
+                    lambda e=ev: cont.do_event(e).body["outputs"][0])))
+            for want, future in futures:
+                assert len(future.result(timeout=60)) == want

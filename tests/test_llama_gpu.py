@@ -188,3 +188,62 @@ class TestFP8Replicas:
 
         out = server.do_event(_Ev()).body["outputs"]
         assert len(out) == 6 and all(len(o) == 4 for o in out)
+
+
+@requires_gpu
+class TestContinuousBatchingGPU:
+    def test_continuous_matches_batch_under_hipgraph(self):
+        from mlrun_amd.models.llama import LlamaConfig, LlamaServer
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        cont = LlamaServer(name="c", config=cfg, batch_size=4,
+                           max_new_tokens=6, scheduling="continuous",
+                           use_graph=True, device="cuda:0")
+        cont.load()
+        batch = LlamaServer(name="b", config=cfg, batch_size=4,
+                            max_new_tokens=6, use_graph=True,
+                            device="cuda:0")
+        batch.load()
+        batch.engines[0].weights.load_state_dict(
+            cont.engines[0].weights.state_dict())
+
+        class _Ev:
+            body = {"inputs": [[1, 2, 3], [9, 8, 7, 6]],
+                    "max_tokens": 5}
+            path = "/infer"
+            id = "t"
+
+        out_c = cont.do_event(_Ev()).body["outputs"]
+        out_b = batch.do_event(_Ev()).body["outputs"]
+        assert out_c == out_b
+
+    def test_slot_reuse_on_gpu(self):
+        import threading
+
+        from mlrun_amd.models.llama import LlamaConfig, LlamaServer
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        cont = LlamaServer(name="c2", config=cfg, batch_size=2,
+                           max_new_tokens=4, scheduling="continuous",
+                           use_graph=True, device="cuda:0")
+        cont.load()
+        results = {}
+
+        def one(i):
+            class _Ev:
+                body = {"inputs": [[i + 1, i + 2]], "max_tokens": 3}
+                path = "/infer"
+                id = "t"
+
+            results[i] = cont.do_event(_Ev()).body["outputs"][0]
+
+        threads = [threading.Thread(target=one, args=(i,))
+                   for i in range(6)]
+        [t.start() for t in threads]
+        [t.join(timeout=120) for t in threads]
+        assert len(results) == 6
+        assert all(len(r) == 3 for r in results.values())
