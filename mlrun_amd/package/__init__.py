@@ -120,6 +120,10 @@ class Packager:
             isinstance(obj, cls.PACKABLE_OBJECT_TYPE)
 
     @classmethod
+    def handles_type(cls, hint) -> bool:
+        return cls.PACKABLE_OBJECT_TYPE is hint
+
+    @classmethod
     def pack(cls, obj, key: str, context, artifact_type=None):
         raise NotImplementedError
 
@@ -140,3 +144,13 @@ class DefaultPackager(Packager):
     @classmethod
     def unpack(cls, data_item, artifact_type=None):
         return data_item
+
+
+from .packagers import (  # noqa: E402,F401
+    NumPyPackager,
+    PackagersManager,
+    PandasPackager,
+    PythonObjectPackager,
+    TorchTensorPackager,
+    default_packagers_manager,
+)

@@ -172,3 +172,63 @@ class TestScheduleCreation:
         scheds = rundb.list_schedules("default")
         assert len(scheds) == 1
         assert scheds[0]["cron_trigger"] == "*/5 * * * *"
+
+
+class TestPackagers:
+    def test_roundtrip_all_types(self, rundb):
+        import io
+
+        import numpy as np
+        import pandas as pd
+        import torch
+
+        import mlrun_amd
+        from mlrun_amd.package import default_packagers_manager
+
+        manager = default_packagers_manager()
+        fn = mlrun_amd.new_function(name="pack", kind="local")
+        run = fn.run(handler=lambda context: None, local=True)
+        from mlrun_amd.execution import MLClientCtx
+        from mlrun_amd.model import RunObject
+
+        ctx = MLClientCtx.from_dict(run.to_dict())
+
+        arr = np.arange(12, dtype=np.float32).reshape(3, 4)
+        manager.pack(arr, "arr", ctx)
+        df = pd.DataFrame({"a": [1, 2], "b": [3.0, 4.0]})
+        manager.pack(df, "df", ctx)
+        tensor = torch.randn(4, 4)
+        manager.pack(tensor, "tensor", ctx)
+        manager.pack({"k": 1}, "plain", ctx)
+        assert ctx.results.get("plain") == {"k": 1}
+
+        class _Item:
+            def __init__(self, data):
+                self._data = data
+
+            def get(self):
+                return self._data
+
+        buf = io.BytesIO()
+        np.save(buf, arr, allow_pickle=False)
+        back = manager.unpack(_Item(buf.getvalue()), np.ndarray)
+        assert np.array_equal(back, arr)
+        buf = io.BytesIO()
+        torch.save(tensor, buf)
+        back_t = manager.unpack(_Item(buf.getvalue()), torch.Tensor)
+        assert torch.equal(back_t, tensor)
+
+    def test_resolve_order_and_errors(self):
+        import pytest
+
+        from mlrun_amd.package import (NumPyPackager, PandasPackager,
+                                       default_packagers_manager)
+
+        manager = default_packagers_manager()
+        import numpy as np
+        import pandas as pd
+
+        assert manager.resolve(np.zeros(2)) is NumPyPackager
+        assert manager.resolve(pd.DataFrame()) is PandasPackager
+        with pytest.raises(TypeError):
+            manager.resolve(object())
