@@ -26,6 +26,7 @@ __all__ = [
     "new_task",
     "new_project",
     "load_project",
+    "get_current_project",
     "get_or_create_project",
     "run_function",
     "build_function",
@@ -62,6 +63,7 @@ from .run import (  # noqa: F401,E402
     run_local,
     wait_for_pipeline_completion,
 )
+from .projects import pipeline_context  # noqa: F401,E402
 from .projects import (  # noqa: F401,E402
     MlrunProject,
     ProjectMetadata,
@@ -121,3 +123,14 @@ def _load_env_file(path: str):
     from .config import _populate
 
     _populate()
+
+
+def get_current_project(silent: bool = False):
+    """The project active in the current pipeline/workflow context
+    (reference mlrun/__init__.py:167)."""
+    if pipeline_context.project is None and not silent:
+        from .errors import MLRunInvalidArgumentError
+
+        raise MLRunInvalidArgumentError(
+            "no project is active; load/create one first")
+    return pipeline_context.project
