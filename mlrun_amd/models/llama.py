@@ -764,6 +764,39 @@ class LlamaServer:
                 if not future.done():
                     future.set_exception(exc)
 
+    def _stream_generate(self, inputs, max_new, event):
+        """Token streaming (continuous scheduling only): yields one
+        json line per generated token as the slots produce them."""
+        import concurrent.futures
+        import json as _json
+        import queue as queue_mod
+
+        streams = []
+        for prompt in inputs:
+            future = concurrent.futures.Future()
+            stream_q: "queue_mod.Queue" = queue_mod.Queue()
+            self._tasks.put((prompt, max_new, future, stream_q))
+            streams.append(stream_q)
+
+        def gen():
+            done = 0
+            indexes = {id(q): i for i, q in enumerate(streams)}
+            active = list(streams)
+            while active:
+                for stream_q in list(active):
+                    try:
+                        token = stream_q.get(timeout=0.05)
+                    except queue_mod.Empty:
+                        continue
+                    if token is None:
+                        active.remove(stream_q)
+                        continue
+                    yield _json.dumps(
+                        {"index": indexes[id(stream_q)],
+                         "token": token}) + "\n"
+
+        return gen()
+
     def _continuous_loop(self, engine: LlamaDecodeEngine):
         """Token-level continuous batching: the engine decodes ALL
         slots every step; finished slots free up and new prompts are
@@ -806,13 +839,16 @@ class LlamaServer:
             if taken:
                 self.engine_calls += 1
                 prompts, admit_slots = [], []
-                for (prompt, max_new, future), slot in zip(taken, free):
+                for item, slot in zip(taken, free):
+                    prompt, max_new, future = item[:3]
+                    stream_q = item[3] if len(item) > 3 else None
                     keep = engine.cfg.max_seq_len - max_new - 1
                     prompt = list(prompt)[-keep:]
                     prompts.append(prompt)
                     admit_slots.append(slot)
                     slots[slot] = {"future": future, "max_new": max_new,
-                                   "start": step, "produced": 1}
+                                   "start": step, "produced": 1,
+                                   "stream": stream_q}
                 max_len = max(len(p) for p in prompts)
                 tokens = torch.zeros(len(prompts), max_len,
                                      dtype=torch.int64)
@@ -827,10 +863,28 @@ class LlamaServer:
             active = [i for i in range(B) if slots[i] is not None]
             if not active:
                 continue
+            streaming = [i for i in active if slots[i]["stream"]]
+            if streaming:
+                # streamers need the admission token on the host too
+                for i in streaming:
+                    state = slots[i]
+                    if state["produced"] == 1 and \
+                            not state.get("_first_sent"):
+                        if engine._serve_stream is not None:
+                            engine._serve_stream.synchronize()
+                        state["stream"].put(int(
+                            out_ring[i, state["start"] % ring_len]))
+                        state["_first_sent"] = True
             with stream_ctx():
                 engine.decode_step()
                 step += 1
                 out_ring[:, step % ring_len].copy_(engine.buf_tokens)
+            if streaming:
+                if engine._serve_stream is not None:
+                    engine._serve_stream.synchronize()
+                host_tokens = engine.buf_tokens.cpu()
+                for i in streaming:
+                    slots[i]["stream"].put(int(host_tokens[i]))
             finished = []
             for i in active:
                 state = slots[i]
@@ -848,6 +902,8 @@ class LlamaServer:
                         dtype=torch.long)
                     result = out_ring[i, cols.to(out_ring.device)]
                     state["future"].set_result(result.cpu().tolist())
+                    if state["stream"]:
+                        state["stream"].put(None)  # end-of-stream
                     slots[i] = None
 
     def do_event(self, event):
@@ -864,6 +920,9 @@ class LlamaServer:
         if inputs is None:
             raise ValueError('expected {"inputs": [[token ids], ...]}')
         max_new = int(body.get("max_tokens", self.max_new_tokens))
+        if self.scheduling == "continuous" and body.get("stream"):
+            event.body = self._stream_generate(inputs, max_new, event)
+            return event
         if self.scheduling == "continuous":
             futures = []
             for prompt in inputs:

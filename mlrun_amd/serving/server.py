@@ -9,6 +9,7 @@ init_states :150, test :196, run :252 — the hot path, GraphContext
 :493, MockEvent :445, create_graph_server :412).
 """
 
+import inspect
 import json
 import socket
 import time
@@ -320,6 +321,13 @@ class GraphServerHost:
                 return Response(json.dumps(response.body), status_code=500,
                                 media_type="application/json")
             body_out = response.body if response is not None else ""
+            if inspect.isgenerator(body_out) or (
+                    hasattr(body_out, "__next__") and
+                    not isinstance(body_out, (bytes, str))):
+                from fastapi.responses import StreamingResponse
+
+                return StreamingResponse(
+                    body_out, media_type="application/x-ndjson")
             if isinstance(body_out, (dict, list)):
                 return Response(json.dumps(body_out, default=str),
                                 media_type="application/json")

@@ -228,3 +228,57 @@ class TestContinuousBatching:
                     lambda e=ev: cont.do_event(e).body["outputs"][0])))
             for want, future in futures:
                 assert len(future.result(timeout=60)) == want
+
+
+class TestTokenStreaming:
+    def test_stream_tokens_match_nonstream(self):
+        import json
+
+        cfg = LlamaConfig.tiny()
+        srv = LlamaServer(name="s", config=cfg, batch_size=4,
+                          max_new_tokens=8, scheduling="continuous",
+                          use_graph=False)
+        srv.load()
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        ev = _Ev()
+        ev.body = {"inputs": [[1, 2, 3], [4, 5]], "max_tokens": 5,
+                   "stream": True}
+        gen = srv.do_event(ev).body
+        per = {}
+        for line in gen:
+            msg = json.loads(line)
+            per.setdefault(msg["index"], []).append(msg["token"])
+        assert len(per[0]) == 5 and len(per[1]) == 5
+        ev2 = _Ev()
+        ev2.body = {"inputs": [[1, 2, 3], [4, 5]], "max_tokens": 5}
+        out = srv.do_event(ev2).body["outputs"]
+        assert per[0] == out[0] and per[1] == out[1]
+
+    def test_http_host_streams_ndjson(self):
+        import json
+
+        import requests
+
+        import mlrun_amd
+
+        fn = mlrun_amd.new_function(name="stream-fn", kind="serving")
+        fn.add_model("llm", class_name=LlamaServer, config="tiny",
+                     batch_size=2, max_new_tokens=6,
+                     scheduling="continuous", use_graph=False)
+        addr = fn.deploy()
+        try:
+            resp = requests.post(
+                addr + "/v2/models/llm/infer",
+                json={"inputs": [[7, 8, 9]], "max_tokens": 4,
+                      "stream": True}, stream=True, timeout=120)
+            assert resp.headers["content-type"].startswith(
+                "application/x-ndjson")
+            tokens = [json.loads(l)["token"]
+                      for l in resp.iter_lines() if l]
+            assert len(tokens) == 4
+        finally:
+            fn.stop()
