@@ -152,8 +152,19 @@ class LlamaTrainer:
     tokens/s accounting, artifact checkpointing."""
 
     def __init__(self, cfg: LlamaConfig, device=None, lr: float = 1e-4,
-                 bucket_cap_mb: int = None, context=None):
+                 bucket_cap_mb: int = None, context=None,
+                 grad_accum_steps: int = 1, warmup_steps: int = 0,
+                 total_steps: int = 0, min_lr_ratio: float = 0.1,
+                 grad_clip: float = 1.0):
         self.cfg = cfg
+        self.grad_accum_steps = max(int(grad_accum_steps), 1)
+        self.warmup_steps = warmup_steps
+        self.total_steps = total_steps
+        self.min_lr_ratio = min_lr_ratio
+        self.grad_clip = grad_clip
+        self.base_lr = lr
+        self.step_count = 0
+        self._accum = 0
         self.device = torch.device(
             device or ("cuda:0" if torch.cuda.is_available() else "cpu"))
         dtype = torch.bfloat16 if self.device.type == "cuda" \
@@ -176,17 +187,45 @@ class LlamaTrainer:
                                            betas=(0.9, 0.95),
                                            weight_decay=0.1, foreach=True)
 
+    def current_lr(self) -> float:
+        """Linear warmup + cosine decay to min_lr_ratio (standard
+        llama fine-tune schedule; constant lr when total_steps=0)."""
+        import math
+
+        step = self.step_count
+        if self.warmup_steps and step < self.warmup_steps:
+            return self.base_lr * (step + 1) / self.warmup_steps
+        if not self.total_steps:
+            return self.base_lr
+        progress = min(1.0, (step - self.warmup_steps) /
+                       max(1, self.total_steps - self.warmup_steps))
+        floor = self.base_lr * self.min_lr_ratio
+        return floor + 0.5 * (self.base_lr - floor) * (
+            1.0 + math.cos(math.pi * progress))
+
     def train_step(self, tokens: torch.Tensor) -> float:
-        """One step: forward, backward (+overlapped all-reduce), step."""
+        """One micro-step: forward, backward (+overlapped all-reduce);
+        the optimizer steps every ``grad_accum_steps`` micro-steps
+        with the scheduled lr."""
         tokens = tokens.to(self.device)
         model = self.ddp or self.model
-        self.optimizer.zero_grad(set_to_none=True)
+        if self._accum == 0:
+            self.optimizer.zero_grad(set_to_none=True)
         _, loss = model(tokens, labels=tokens)
-        loss.backward()
-        if self.ddp is not None:
-            self.ddp.finalize_backward()
-        torch.nn.utils.clip_grad_norm_(self.model.parameters(), 1.0)
-        self.optimizer.step()
+        (loss / self.grad_accum_steps).backward()
+        self._accum += 1
+        if self._accum >= self.grad_accum_steps:
+            self._accum = 0
+            if self.ddp is not None:
+                self.ddp.finalize_backward()
+            if self.grad_clip:
+                torch.nn.utils.clip_grad_norm_(self.model.parameters(),
+                                               self.grad_clip)
+            lr = self.current_lr()
+            for group in self.optimizer.param_groups:
+                group["lr"] = lr
+            self.optimizer.step()
+            self.step_count += 1
         return float(loss.detach())
 
     def save_checkpoint(self, key: str = "model"):
