@@ -247,3 +247,34 @@ class TestContinuousBatchingGPU:
         [t.join(timeout=120) for t in threads]
         assert len(results) == 6
         assert all(len(r) == 3 for r in results.values())
+
+
+@requires_gpu
+class TestStreamingGPU:
+    def test_stream_matches_nonstream_under_hipgraph(self):
+        import json
+
+        from mlrun_amd.models.llama import LlamaConfig, LlamaServer
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        srv = LlamaServer(name="sg", config=cfg, batch_size=2,
+                          max_new_tokens=6, scheduling="continuous",
+                          use_graph=True, device="cuda:0")
+        srv.load()
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        ev = _Ev()
+        ev.body = {"inputs": [[3, 1, 4]], "max_tokens": 5,
+                   "stream": True}
+        per = []
+        for line in srv.do_event(ev).body:
+            per.append(json.loads(line)["token"])
+        ev2 = _Ev()
+        ev2.body = {"inputs": [[3, 1, 4]], "max_tokens": 5}
+        out = srv.do_event(ev2).body["outputs"][0]
+        assert per == out
