@@ -333,6 +333,19 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
                            for t, stats in series],
                 "current": processor.endpoint_stats(endpoint_id)}
 
+    @app.get("/metrics")
+    async def prometheus_metrics():
+        """Prometheus scrape endpoint for the API service."""
+        try:
+            from prometheus_client import (CONTENT_TYPE_LATEST,
+                                           generate_latest)
+            from fastapi import Response as _Resp
+
+            return _Resp(generate_latest(),
+                         media_type=CONTENT_TYPE_LATEST)
+        except ImportError:
+            return {"error": "prometheus_client not installed"}
+
     @app.get("/api/v1/monitoring/memory")
     async def memory_report():
         """Process + GPU memory report (reference:

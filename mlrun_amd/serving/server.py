@@ -305,9 +305,39 @@ class GraphServerHost:
         async def healthz():
             return {"status": "ok"}
 
+        # Prometheus scrape endpoint (reference: `scrape_metrics` run
+        # label + grafana integration; here native counters/latency
+        # histograms per graph host)
+        try:
+            from prometheus_client import (CONTENT_TYPE_LATEST,
+                                           CollectorRegistry, Counter,
+                                           Histogram, generate_latest)
+
+            registry = CollectorRegistry()
+            self._metric_requests = Counter(
+                "mlrun_serving_requests_total",
+                "Requests handled by this graph host",
+                ["path", "status"], registry=registry)
+            self._metric_latency = Histogram(
+                "mlrun_serving_request_seconds",
+                "End-to-end request latency", ["path"],
+                registry=registry,
+                buckets=(.005, .01, .025, .05, .1, .25, .5, 1, 2.5, 5,
+                         10, 30))
+
+            @app.get("/metrics")
+            async def metrics():
+                return Response(generate_latest(registry),
+                                media_type=CONTENT_TYPE_LATEST)
+        except ImportError:
+            self._metric_requests = self._metric_latency = None
+
         @app.api_route("/{full_path:path}",
                        methods=["GET", "POST", "PUT", "DELETE"])
         async def handle(full_path: str, request: Request):
+            import time as _time
+
+            handle_start = _time.perf_counter()
             body = await request.body()
             event = Event(body=body, path="/" + full_path,
                           method=request.method,
@@ -317,6 +347,14 @@ class GraphServerHost:
 
             response = await anyio.to_thread.run_sync(
                 lambda: graph_server.run(event, get_body=False))
+            status = "500" if isinstance(response, _ErrorResponse) \
+                else "200"
+            if self._metric_requests is not None:
+                route = "/" + full_path.split("/")[0]
+                self._metric_requests.labels(path=route,
+                                             status=status).inc()
+                self._metric_latency.labels(path=route).observe(
+                    _time.perf_counter() - handle_start)
             if isinstance(response, _ErrorResponse):
                 return Response(json.dumps(response.body), status_code=500,
                                 media_type="application/json")

@@ -484,3 +484,23 @@ class TestRAG:
                            body={"inputs": [prompt], "max_tokens": 4})
         assert len(resp["outputs"][0]) == 4
         assert len(resp["retrieval"][0]) == 2
+
+
+class TestPrometheusMetrics:
+    def test_host_exposes_request_counters(self):
+        import requests
+
+        import mlrun_amd
+
+        fn = mlrun_amd.new_function(name="prom", kind="remote")
+        fn.spec.build["functionSourceCode"] = (
+            "def handler(ctx, event):\n    return {'ok': 1}\n")
+        addr = fn.deploy()
+        try:
+            requests.post(addr + "/infer", json={}, timeout=10)
+            body = requests.get(addr + "/metrics", timeout=10).text
+            assert 'mlrun_serving_requests_total{path="/infer",' \
+                   'status="200"} 1.0' in body
+            assert "mlrun_serving_request_seconds_bucket" in body
+        finally:
+            fn.stop()
