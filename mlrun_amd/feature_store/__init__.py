@@ -13,7 +13,7 @@ from .feature_set import (  # noqa: F401
     FeatureSet,
     parse_span,
 )
-from .vector import FeatureVector, OnlineVectorService  # noqa: F401
+from .vector import FeatureVector, JoinGraph, OnlineVectorService  # noqa: F401
 from .online import OnlineTable, get_online_table, reset_online_tables  # noqa: F401
 from .api import (  # noqa: F401
     IngestionService,

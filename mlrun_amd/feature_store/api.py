@@ -269,7 +269,8 @@ class OfflineVectorResponse:
 def get_offline_features(feature_vector, entity_rows=None,
                          entity_timestamp_column=None, target=None,
                          drop_columns=None, with_indexes=False,
-                         update_stats=False) -> OfflineVectorResponse:
+                         update_stats=False,
+                         join_graph=None) -> OfflineVectorResponse:
     """Join features from the parquet targets of the referenced sets
     (pandas merger — reference retrieval/local_merger.py)."""
     import pandas as pd
@@ -278,7 +279,13 @@ def get_offline_features(feature_vector, entity_rows=None,
     asof = entity_timestamp_column is not None and entity_rows is not None
     merged = None
     entity_cols: list = []
-    for fs_name, columns, aliases in vector.grouped_features():
+    grouped = vector.grouped_features()
+    if join_graph is not None:
+        order = join_graph.order()
+        grouped = sorted(
+            grouped, key=lambda g: order.index(g[0])
+            if g[0] in order else len(order))
+    for fs_name, columns, aliases in grouped:
         fset = _resolve_feature_set(
             f"{vector.metadata.project or 'default'}/{fs_name}")
         path = _parquet_target_path(fset)
@@ -323,8 +330,10 @@ def get_offline_features(feature_vector, entity_rows=None,
                 df = df.drop(columns=[ts], errors="ignore")
         else:
             df = df.groupby(entities, as_index=False).last()
+        how = join_graph.how_for(fs_name) if join_graph is not None \
+            else "inner"
         merged = df if merged is None else merged.merge(
-            df, on=entities, how="inner")
+            df, on=entities, how=how)
     if merged is None:
         raise MLRunInvalidArgumentError("vector references no features")
     if entity_rows is not None and not asof:

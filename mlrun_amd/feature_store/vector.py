@@ -133,6 +133,48 @@ class FeatureVector(ModelObj):
         raise MLRunInvalidArgumentError("cannot resolve feature vector")
 
 
+class JoinGraph:
+    """Per-feature-set join specification for offline merges
+    (reference feature_vector.py:204): chain ``inner/left/outer``
+    calls in merge order; ``get_offline_features(join_graph=...)``
+    applies the requested join type per set instead of the default
+    inner join."""
+
+    def __init__(self, first_feature_set: str = None):
+        self.steps: list = []
+        if first_feature_set:
+            self.steps.append((_name_of(first_feature_set), "first"))
+
+    def _join(self, other, how: str) -> "JoinGraph":
+        self.steps.append((_name_of(other), how))
+        return self
+
+    def inner(self, other) -> "JoinGraph":
+        return self._join(other, "inner")
+
+    def left(self, other) -> "JoinGraph":
+        return self._join(other, "left")
+
+    def outer(self, other) -> "JoinGraph":
+        return self._join(other, "outer")
+
+    def how_for(self, feature_set_name: str, default: str = "inner"):
+        for name, how in self.steps:
+            if name == feature_set_name and how != "first":
+                return how
+        return default
+
+    def order(self) -> list:
+        return [name for name, _ in self.steps]
+
+
+def _name_of(feature_set) -> str:
+    name = getattr(feature_set, "metadata", None)
+    if name is not None:
+        return feature_set.metadata.name
+    return str(feature_set).split("/")[-1].split(":")[0]
+
+
 class OnlineVectorService:
     """Online lookups over the vector's feature sets (reference
     OnlineVectorService.get :975 — here batched)."""

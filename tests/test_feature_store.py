@@ -282,3 +282,35 @@ class TestAsofJoin:
         out = resp.to_dataframe()
         # as-of: first row sees only the -3h value, second sees -1h
         assert list(out["v"]) == [1.0, 3.0]
+
+
+class TestJoinGraph:
+    def test_left_join_keeps_unmatched_entities(self, rundb):
+        import pandas as pd
+
+        from mlrun_amd import feature_store as fstore
+
+        a = fstore.FeatureSet("jga", entities=["id"])
+        fstore.ingest(a, pd.DataFrame({"id": [1, 2, 3],
+                                       "x": [10, 20, 30]}))
+        b = fstore.FeatureSet("jgb", entities=["id"])
+        fstore.ingest(b, pd.DataFrame({"id": [1, 2], "y": [7, 8]}))
+
+        vector = fstore.FeatureVector("jgv", ["jga.x", "jgb.y"])
+        inner = fstore.get_offline_features(vector).to_dataframe()
+        assert len(inner) == 2  # default inner drops id=3
+
+        graph = fstore.JoinGraph(first_feature_set="jga").left("jgb")
+        left = fstore.get_offline_features(
+            vector, join_graph=graph).to_dataframe()
+        assert len(left) == 3
+        assert left[left["id"] == 3]["y"].isna().all()
+
+    def test_join_graph_orders_merge(self, rundb):
+        from mlrun_amd.feature_store import JoinGraph
+
+        graph = JoinGraph(first_feature_set="b").inner("a").outer("c")
+        assert graph.order() == ["b", "a", "c"]
+        assert graph.how_for("c") == "outer"
+        assert graph.how_for("a") == "inner"
+        assert graph.how_for("unknown") == "inner"
