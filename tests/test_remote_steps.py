@@ -85,3 +85,47 @@ class TestApplicationRuntime:
         while fn.is_running() and time.time() < deadline:
             time.sleep(0.1)
         assert not fn.is_running()
+
+
+class TestWebhookNotificationLive:
+    def test_webhook_posts_to_local_host(self):
+        """Webhook + slack notifications against a real local HTTP
+        endpoint (no external network)."""
+        received = []
+
+        class _Sink:
+            def __init__(self, context=None, name=None):
+                pass
+
+            def do_event(self, event):
+                received.append(event.body)
+                event.body = {"ok": True}
+                return event
+
+        from mlrun_amd.serving.states import RootFlowStep
+
+        graph = RootFlowStep()
+        graph.add_step(_Sink, name="sink")
+        server = GraphServer(graph=graph)
+        server.init_states(namespace={"_Sink": _Sink})
+        host = GraphServerHost(server, port=0)
+        host.start()
+        try:
+            from mlrun_amd.utils.notifications import (
+                get_notification_class)
+
+            hook = get_notification_class("webhook")(
+                params={"url": host.address + "/hook"})
+            hook.push("run done", "info",
+                      [{"metadata": {"name": "r1"},
+                        "status": {"state": "completed"}}])
+            slack = get_notification_class("slack")(
+                params={"webhook": host.address + "/slack"})
+            slack.push("run failed", "error",
+                       [{"metadata": {"name": "r2"},
+                         "status": {"state": "error"}}])
+        finally:
+            host.stop()
+        assert received[0]["message"] == "run done"
+        assert received[0]["runs"][0]["metadata"]["name"] == "r1"
+        assert "*error* run failed" in received[1]["text"]
