@@ -1318,13 +1318,15 @@ struct AttnState {
 // accumulate the online-softmax state for this wave's head
 struct uchar16v { unsigned char v[16]; };
 
+#define ATTN_LDS_PAD 8  // halves; rows start on different banks
 template <int SCHUNK, bool KVQ>
 __device__ __forceinline__ AttnState attn_chunk_loop(
     const void* __restrict__ kbase_, const void* __restrict__ vbase_,
     const float* __restrict__ kscale, const float* __restrict__ vscale,
     const float* qf, float scale, int s_begin, int s_final,
     int wave, int lane, int tid, int nthreads,
-    unsigned short (*kbuf)[128], unsigned short (*vbuf)[128],
+    unsigned short (*kbuf)[128 + ATTN_LDS_PAD],
+    unsigned short (*vbuf)[128 + ATTN_LDS_PAD],
     float (*scores)[SCHUNK]) {
   constexpr int D = 128;
   AttnState st = {-FLT_MAX, 0.f, 0.f, 0.f};
@@ -1426,8 +1428,8 @@ __global__ void attn_decode_kernel(
   const int len = seq_lens[b];
 
   __shared__ float scores[ATTN_MAXG][SCHUNK];
-  __shared__ __align__(16) unsigned short kbuf[SCHUNK][D];
-  __shared__ __align__(16) unsigned short vbuf[SCHUNK][D];
+  __shared__ __align__(16) unsigned short kbuf[SCHUNK][D + ATTN_LDS_PAD];
+  __shared__ __align__(16) unsigned short vbuf[SCHUNK][D + ATTN_LDS_PAD];
 
   // q fragment: lane holds 8 consecutive dims at (lane&15)*8 for the
   // K-dot phase (16-lane groups each cover all 128 dims)
@@ -1484,8 +1486,8 @@ __global__ void attn_decode_split_kernel(
   float* prow = partial + (((size_t)b * Hq + h) * nsplit + split) * (D + 2);
 
   __shared__ float scores[ATTN_MAXG][SCHUNK];
-  __shared__ __align__(16) unsigned short kbuf[SCHUNK][D];
-  __shared__ __align__(16) unsigned short vbuf[SCHUNK][D];
+  __shared__ __align__(16) unsigned short kbuf[SCHUNK][D + ATTN_LDS_PAD];
+  __shared__ __align__(16) unsigned short vbuf[SCHUNK][D + ATTN_LDS_PAD];
 
   const unsigned short* qp = Q + (size_t)b * q_row_stride +
                              (size_t)h * D + (lane & 15) * 8;
