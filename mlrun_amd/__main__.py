@@ -25,6 +25,21 @@ def cmd_config(args):
     print(config.dump_yaml())
 
 
+def _parse_hyperparams(spec: str):
+    """Accept JSON ({"p1": [1,2]}) or key=json-list (p1=[1,2];p2=[..])
+    like the reference's -x arguments."""
+    if not spec:
+        return None
+    spec = spec.strip()
+    if spec.startswith("{"):
+        return json.loads(spec)
+    params = {}
+    for part in spec.split(";"):
+        key, _, value = part.partition("=")
+        params[key.strip()] = json.loads(value)
+    return params
+
+
 def cmd_run(args):
     import mlrun_amd
 
@@ -50,8 +65,7 @@ def cmd_run(args):
                  params=params, inputs=inputs, project=args.project or "",
                  artifact_path=args.artifact_path or "",
                  schedule=args.schedule, watch=not args.no_wait,
-                 hyperparams=json.loads(args.hyperparam)
-                 if args.hyperparam else None)
+                 hyperparams=_parse_hyperparams(args.hyperparam))
     state = run.status.state
     print(f"run {run.metadata.uid} finished: {state}")
     if run.status.results:

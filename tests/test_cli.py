@@ -52,3 +52,35 @@ class TestCLI:
                          "status": {"state": "completed"}}, "u9", "default")
         main(["clean"])
         assert rundb.list_runs(project="default") == []
+
+    def test_build_and_run_image(self, tmp_path, capsys, rundb):
+        code = tmp_path / "fn.py"
+        code.write_text("def handler(context):\n"
+                        "    context.log_result('v', 3)\n")
+        main(["build", "--name", "cli-img", "--command", str(code)])
+        out = capsys.readouterr().out
+        assert "image built at" in out
+        import os
+
+        from mlrun_amd.config import config
+
+        image = os.path.join(config.base_dir, "images", "default",
+                             "cli-img", "latest")
+        assert os.path.isfile(os.path.join(image, "run.sh"))
+
+    def test_get_schedules_and_projects(self, capsys, rundb):
+        import mlrun_amd
+
+        mlrun_amd.new_project("cli-proj")
+        main(["get", "projects"])
+        out = capsys.readouterr().out
+        assert "cli-proj" in out
+
+    def test_hyperparam_run(self, tmp_path, capsys, rundb):
+        code = tmp_path / "hp.py"
+        code.write_text("def handler(context, p1=0):\n"
+                        "    context.log_result('r', p1 * 2)\n")
+        main(["run", str(code), "--handler", "handler", "--name", "hp",
+              "--hyperparam", "p1=[1,2,3]", "--local"])
+        out = capsys.readouterr().out
+        assert "hp" in out
