@@ -83,4 +83,8 @@ class TestCLI:
         main(["run", str(code), "--handler", "handler", "--name", "hp",
               "--hyperparam", "p1=[1,2,3]", "--local"])
         out = capsys.readouterr().out
-        assert "hp" in out
+        assert "finished: completed" in out
+        runs = rundb.list_runs(project="default", name="hp")
+        iters = runs[0]["status"].get("iterations") or []
+        results = runs[0]["status"].get("results", {})
+        assert results.get("best_iteration") or iters
