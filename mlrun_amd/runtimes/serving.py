@@ -36,7 +36,8 @@ class ServingSpec(FunctionSpec):
                  graph_initializer=None, error_stream=None, track_models=None,
                  function_refs=None, default_content_type=None,
                  secret_sources=None, disable_auto_mount=None,
-                 allow_empty_resources=None, resources=None):
+                 allow_empty_resources=None, resources=None,
+                 min_replicas=None, max_replicas=None):
         super().__init__(command, args, image, mode, build, entry_points,
                          description, workdir, default_handler, pythonpath,
                          disable_auto_mount, allow_empty_resources, resources)
@@ -51,6 +52,9 @@ class ServingSpec(FunctionSpec):
         self.function_refs = function_refs or {}
         self.default_content_type = default_content_type
         self.secret_sources = secret_sources or []
+        # worker-process replicas (nuclio min/max_replicas analog)
+        self.min_replicas = min_replicas
+        self.max_replicas = max_replicas
 
     @property
     def graph(self):
@@ -194,8 +198,23 @@ class ServingRuntime(BaseRuntime):
 
     def deploy(self, project="", tag="", verbose=False, auth_info=None,
                builder_env=None, force_build=False, with_mlrun=None,
-               namespace=None) -> str:
-        """Start the node-local HTTP serving host; returns its address."""
+               namespace=None, workers: int = 0) -> str:
+        """Start the node-local HTTP serving host; returns its address.
+        workers > 1: N worker PROCESSES behind an L4 round-robin proxy
+        (nuclio min_replicas analog; removes the single-process GIL
+        cap — serving/workers.py)."""
+        workers = workers or int(self.spec.min_replicas or 0)
+        if workers > 1:
+            from ..serving.workers import WorkerPool
+
+            self._worker_pool = WorkerPool(self.to_dict(), workers)
+            address = self._worker_pool.start()
+            self.status.state = "ready"
+            self.status.address = address
+            self.status.external_invocation_urls = [address]
+            logger.info("serving function deployed (worker pool)",
+                        address=address, workers=workers)
+            return address
         server = self._build_server(namespace)
         self._server = server
         self._host = GraphServerHost(server)
@@ -256,12 +275,22 @@ class ServingRuntime(BaseRuntime):
                                       headers=headers)
 
     def stop(self):
+        if getattr(self, "_worker_pool", None) is not None:
+            self._worker_pool.stop()
+            self._worker_pool = None
         if self._host is not None:
             self._host.stop()
             self._host = None
 
     def with_secrets(self, kind, source):
         self.spec.secret_sources.append({"kind": kind, "source": source})
+        return self
+
+    def with_replicas(self, min_replicas: int, max_replicas: int = None):
+        """Worker-process count (reference nuclio min/max_replicas;
+        the node-local pool is fixed-size = min_replicas)."""
+        self.spec.min_replicas = min_replicas
+        self.spec.max_replicas = max_replicas or min_replicas
         return self
 
 

@@ -65,6 +65,8 @@ def main():
     parser.add_argument("--warmup", type=int, default=32)
     parser.add_argument("--scheduling", default="batch",
                         choices=["batch", "continuous"])
+    parser.add_argument("--workers", type=int, default=0,
+                        help="serving worker processes (L4 round-robin)")
     args = parser.parse_args()
 
     on_gpu = torch.cuda.is_available()
@@ -81,7 +83,7 @@ def main():
                  replicas=args.replicas,
                  weight_dtype="fp8w" if args.weights == "fp8" else "bf16",
                  scheduling=args.scheduling)
-    address = fn.deploy()
+    address = fn.deploy(workers=args.workers)
     print(f"serving at {address}", file=sys.stderr)
 
     vocab = 1000
@@ -132,6 +134,7 @@ def main():
         "tokens_per_sec": round(args.requests * args.gen_tokens / wall, 1),
         "batch_window_ms": args.batch_window_ms,
         "scheduling": args.scheduling,
+        "workers": args.workers,
         "model": model,
         "weights": args.weights,
         "replicas": args.replicas,

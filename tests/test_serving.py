@@ -504,3 +504,32 @@ class TestPrometheusMetrics:
             assert "mlrun_serving_request_seconds_bucket" in body
         finally:
             fn.stop()
+
+
+class TestWorkerPool:
+    def test_multi_process_workers_roundrobin(self):
+        import requests
+
+        import mlrun_amd
+
+        fn = mlrun_amd.new_function(name="pool", kind="serving")
+        fn.add_model("llm",
+                     class_name="mlrun_amd.models.llama.LlamaServer",
+                     config="tiny", batch_size=2, max_new_tokens=6)
+        fn.with_replicas(2)
+        addr = fn.deploy()
+        try:
+            outs = []
+            for _ in range(4):
+                resp = requests.post(
+                    addr + "/v2/models/llm/infer",
+                    json={"inputs": [[5, 6, 7]], "max_tokens": 4},
+                    timeout=120)
+                assert resp.status_code == 200
+                outs.append(resp.json()["outputs"][0])
+            assert all(len(o) == 4 for o in outs)
+            # both workers run the same weights (same seed) — outputs
+            # must agree across backends
+            assert all(o == outs[0] for o in outs)
+        finally:
+            fn.stop()
