@@ -148,6 +148,23 @@ class ServingRuntime(BaseRuntime):
         if model_path:
             class_args = dict(class_args)
             class_args["model_path"] = model_path
+        if isinstance(class_name, type):
+            # store the dotted path when it resolves back — so the
+            # spec survives serialization (worker processes /
+            # save-load); locally-defined classes stay as objects
+            dotted = f"{class_name.__module__}." \
+                     f"{class_name.__qualname__}"
+            try:
+                import importlib
+
+                module = importlib.import_module(class_name.__module__)
+                resolved = module
+                for part in class_name.__qualname__.split("."):
+                    resolved = getattr(resolved, part)
+                if resolved is class_name:
+                    class_name = dotted
+            except (ImportError, AttributeError):
+                pass
         route = TaskStep(class_name, class_args, handler=handler, name=key,
                          function=child_function)
         return router.add_route(key, route=route)
