@@ -842,8 +842,10 @@ class LlamaServer:
                 for item, slot in zip(taken, free):
                     prompt, max_new, future = item[:3]
                     stream_q = item[3] if len(item) > 3 else None
+                    max_new = max(1, min(max_new,
+                                         engine.cfg.max_seq_len - 2))
                     keep = engine.cfg.max_seq_len - max_new - 1
-                    prompt = list(prompt)[-keep:]
+                    prompt = list(prompt)[-keep:] or [0]
                     prompts.append(prompt)
                     admit_slots.append(slot)
                     slots[slot] = {"future": future, "max_new": max_new,
@@ -856,10 +858,22 @@ class LlamaServer:
                     tokens[i, max_len - len(prompt):] = torch.tensor(
                         prompt, dtype=torch.int64)
                 slot_ids = torch.tensor(admit_slots, dtype=torch.long)
-                with stream_ctx():
-                    engine.prefill_slots(tokens, slot_ids)
-                    out_ring[slot_ids, step % ring_len] = \
-                        engine.buf_tokens[slot_ids]
+                try:
+                    with stream_ctx():
+                        engine.prefill_slots(tokens, slot_ids)
+                        out_ring[slot_ids, step % ring_len] = \
+                            engine.buf_tokens[slot_ids]
+                except Exception as exc:  # admission failed: fail the
+                    for slot in admit_slots:  # requests, free the slots
+                        state = slots[slot]
+                        if not state["future"].done():
+                            state["future"].set_exception(exc)
+                        if state["stream"]:
+                            state["stream"].put(None)
+                        slots[slot] = None
+                    logger.error("continuous admission failed",
+                                 error=str(exc))
+                    continue
             active = [i for i in range(B) if slots[i] is not None]
             if not active:
                 continue

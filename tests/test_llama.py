@@ -282,3 +282,28 @@ class TestTokenStreaming:
             assert len(tokens) == 4
         finally:
             fn.stop()
+
+    def test_admission_failure_fails_request_not_loop(self):
+        """A broken prompt must fail ITS request and free the slot;
+        later requests still serve."""
+        cfg = LlamaConfig.tiny()
+        srv = LlamaServer(name="hf", config=cfg, batch_size=2,
+                          max_new_tokens=4, scheduling="continuous",
+                          use_graph=False)
+        srv.load()
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        bad = _Ev()
+        # out-of-range token ids -> embedding index error at prefill
+        bad.body = {"inputs": [[10**9]], "max_tokens": 3}
+        import pytest as _pytest
+
+        with _pytest.raises(Exception):
+            srv.do_event(bad)
+        good = _Ev()
+        good.body = {"inputs": [[1, 2, 3]], "max_tokens": 3}
+        out = srv.do_event(good).body["outputs"]
+        assert len(out[0]) == 3
