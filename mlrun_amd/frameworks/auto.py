@@ -47,6 +47,14 @@ def apply_mlrun(model=None, context=None, model_name: str = "model",
         return _SKLearnInterface(model, context, model_name)
     if framework in ("llama", "tree"):
         return model
+    if framework == "lightgbm":
+        from .lgbm import model_from_lgbm
+
+        return model_from_lgbm(model)
+    if framework == "xgboost":
+        from .xgboost import model_from_xgboost
+
+        return model_from_xgboost(model)
     raise MLRunInvalidArgumentError(f"unsupported framework {framework}")
 
 

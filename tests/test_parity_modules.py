@@ -299,3 +299,51 @@ class TestBuiltinHub:
         assert run.status.state == "completed", run.status.error
         assert run.status.results["mae"] < 1.0
         assert run.output("model").startswith("store://")
+
+
+class TestHubBuiltins:
+    def test_batch_infer_end_to_end(self, rundb, tmp_path):
+        import numpy as np
+        import pandas as pd
+        from sklearn.ensemble import GradientBoostingRegressor
+
+        import mlrun_amd
+        from mlrun_amd.frameworks import TreeEnsembleModel
+
+        rng = np.random.default_rng(0)
+        x = rng.normal(size=(64, 4)).astype(np.float32)
+        y = (x[:, 0] * 2 + x[:, 1]).astype(np.float32)
+        skl = GradientBoostingRegressor(n_estimators=10,
+                                        max_depth=3).fit(x, y)
+        model = TreeEnsembleModel.from_sklearn(skl)
+        path = tmp_path / "model.npz"
+        model.save(str(path))
+
+        log_fn = mlrun_amd.new_function(name="log-model", kind="local")
+
+        def log_handler(context):
+            context.log_model("gbdt", body=open(path, "rb").read(),
+                              model_file="model.npz",
+                              framework="tree")
+
+        run = log_fn.run(handler=log_handler, local=True)
+        model_uri = run.outputs["gbdt"]
+
+        fn = mlrun_amd.import_function("hub://batch-infer")
+        df = pd.DataFrame(x, columns=[f"f{i}" for i in range(4)])
+        namespace = {}
+        exec(compile(fn.spec.build["functionSourceCode"],
+                     "<hub>", "exec"), namespace)
+        ctx_run = log_fn.run(handler=lambda context: namespace[
+            "batch_infer"](context, model_uri=model_uri, dataset=df),
+            local=True, name="bi")
+        assert ctx_run.outputs["count"] == 64
+        assert "predictions" in ctx_run.outputs
+
+    def test_llm_serving_hub_spec_loads(self, rundb):
+        import mlrun_amd
+
+        fn = mlrun_amd.import_function("hub://llm-serving")
+        assert fn.kind == "serving"
+        graph = fn.spec.graph
+        assert "llm" in graph.routes
