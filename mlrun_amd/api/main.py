@@ -16,7 +16,8 @@ informers.
 import asyncio
 import typing
 
-from fastapi import FastAPI, Query, Request, Response
+from fastapi import (FastAPI, HTTPException, Query, Request,
+                     Response)
 from fastapi.responses import JSONResponse
 
 from ..config import config
@@ -332,6 +333,68 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
                 "series": [{"time": t, "stats": stats}
                            for t, stats in series],
                 "current": processor.endpoint_stats(endpoint_id)}
+
+    # ------------------------------------------------------------ hub
+    @app.get("/api/v1/hub/sources")
+    async def hub_sources():
+        from ..hub import list_hub_sources
+
+        return {"sources": list_hub_sources()}
+
+    @app.get("/api/v1/hub/sources/{source}/items")
+    async def hub_catalog(source: str):
+        from ..hub import get_hub_catalog
+
+        return {"catalog": get_hub_catalog(source)}
+
+    @app.get("/api/v1/hub/sources/{source}/items/{name}")
+    async def hub_item(source: str, name: str):
+        import yaml as _yaml
+
+        from ..hub import get_hub_catalog
+
+        for item in get_hub_catalog(source):
+            if item["name"] == name:
+                with open(item["path"]) as stream:
+                    return {"item": item,
+                            "spec": _yaml.safe_load(stream)}
+        raise HTTPException(status_code=404,
+                            detail=f"hub item {name} not found")
+
+    # -------------------------------------------------------- secrets
+    @app.post("/api/v1/projects/{project}/secrets")
+    async def store_project_secrets(project: str, body: dict):
+        db.store_project_secrets(project, body.get("secrets") or {})
+        return {}
+
+    @app.get("/api/v1/projects/{project}/secret-keys")
+    async def list_secret_keys(project: str):
+        return {"secret_keys": db.list_project_secret_keys(project)}
+
+    @app.delete("/api/v1/projects/{project}/secrets")
+    async def delete_project_secrets(project: str,
+                                     secrets: str = ""):
+        keys = [k for k in secrets.split(",") if k] or None
+        db.delete_project_secrets(project, keys)
+        return {}
+
+    # ------------------------------------------------------ pipelines
+    @app.get("/api/v1/projects/{project}/pipelines")
+    async def list_pipelines(project: str):
+        from ..projects.pipelines import list_pipeline_runs
+
+        return {"runs": [r.to_dict()
+                         for r in list_pipeline_runs(project)]}
+
+    @app.get("/api/v1/projects/{project}/pipelines/{run_id}")
+    async def get_pipeline_run(project: str, run_id: str):
+        from ..projects.pipelines import get_pipeline
+
+        try:
+            return get_pipeline(run_id).to_dict()
+        except Exception:
+            raise HTTPException(status_code=404,
+                                detail=f"pipeline {run_id} not found")
 
     @app.get("/metrics")
     async def prometheus_metrics():

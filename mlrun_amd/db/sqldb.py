@@ -96,6 +96,10 @@ CREATE TABLE IF NOT EXISTS alert_states (
     last_updated TEXT, active INTEGER DEFAULT 0,
     PRIMARY KEY (project, name)
 );
+CREATE TABLE IF NOT EXISTS project_secrets (
+    project TEXT NOT NULL, key TEXT NOT NULL, value TEXT,
+    PRIMARY KEY (project, key)
+);
 CREATE TABLE IF NOT EXISTS hub_sources (
     name TEXT PRIMARY KEY, idx INTEGER, updated TEXT, body TEXT
 );
@@ -474,6 +478,42 @@ class SQLRunDB(RunDBInterface):
         if not rows:
             raise MLRunNotFoundError(f"project {name} not found")
         return json.loads(rows[0]["body"])
+
+    # ------------------------------------------------------- secrets
+    def store_project_secrets(self, project: str, secrets: dict,
+                              provider: str = "kubernetes"):
+        """Store project secrets (values retrievable server-side only
+        via get_project_secret; the list endpoint exposes KEYS, like
+        the reference k8s-secret flow)."""
+        for key, value in (secrets or {}).items():
+            self._execute(
+                "INSERT OR REPLACE INTO project_secrets "
+                "(project, key, value) VALUES (?, ?, ?)",
+                (project, key, str(value)))
+
+    def list_project_secret_keys(self, project: str,
+                                 provider: str = "kubernetes") -> list:
+        return [row["key"] for row in self._query(
+            "SELECT key FROM project_secrets WHERE project = ? "
+            "ORDER BY key", (project,))]
+
+    def get_project_secret(self, project: str, key: str):
+        rows = self._query(
+            "SELECT value FROM project_secrets WHERE project = ? "
+            "AND key = ?", (project, key))
+        return rows[0]["value"] if rows else None
+
+    def delete_project_secrets(self, project: str, keys: list = None,
+                               provider: str = "kubernetes"):
+        if keys:
+            for key in keys:
+                self._execute(
+                    "DELETE FROM project_secrets WHERE project = ? "
+                    "AND key = ?", (project, key))
+        else:
+            self._execute(
+                "DELETE FROM project_secrets WHERE project = ?",
+                (project,))
 
     def list_projects(self, owner=None, format_=None, labels=None, state=None):
         out = []

@@ -24,6 +24,17 @@ from .project import pipeline_context
 _pipeline_runs: dict = {}
 
 
+def list_pipeline_runs(project: str = None) -> list:
+    """All recorded workflow runs (newest last), optionally filtered
+    by project name (reference: pipelines list endpoint)."""
+    runs = list(_pipeline_runs.values())
+    if project:
+        runs = [r for r in runs
+                if getattr(r._project, "metadata", None) is not None
+                and r._project.metadata.name == project]
+    return runs
+
+
 def get_pipeline(run_id: str) -> "_PipelineRunStatus":
     """Look up a completed workflow run by id (reference run.py
     get_pipeline over KFP)."""
@@ -57,6 +68,14 @@ class _PipelineRunStatus:
         self.state = state
         self.runs = runs or []
         self.error = error
+
+    def to_dict(self) -> dict:
+        return {"run_id": self.run_id,
+                "workflow_name": self.workflow_name,
+                "state": self.state,
+                "project": self._project.metadata.name
+                if getattr(self._project, "metadata", None) else "",
+                "runs": [r.metadata.uid for r in (self.runs or [])]}
 
     def wait_for_completion(self, timeout=None, expected_statuses=None):
         return self.state

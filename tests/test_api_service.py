@@ -226,3 +226,60 @@ def test_memory_report(client):
     assert resp.status_code == 200
     body = resp.json()
     assert body["rss_bytes"] > 0 and body["threads"] >= 1
+
+
+class TestHubSecretsPipelinesAPI:
+    def test_hub_endpoints(self, client):
+        resp = client.get("/api/v1/hub/sources")
+        assert resp.status_code == 200
+        names = [s["name"] for s in resp.json()["sources"]]
+        assert "builtin" in names
+        resp = client.get("/api/v1/hub/sources/builtin/items")
+        items = [i["name"] for i in resp.json()["catalog"]]
+        assert "describe" in items and "batch-infer" in items
+        resp = client.get("/api/v1/hub/sources/builtin/items/describe")
+        assert resp.json()["spec"]["kind"] == "job"
+        resp = client.get("/api/v1/hub/sources/builtin/items/nope")
+        assert resp.status_code == 404
+
+    def test_secrets_endpoints(self, client):
+        resp = client.post("/api/v1/projects/p1/secrets",
+                           json={"secrets": {"TOKEN": "abc",
+                                             "KEY2": "v"}})
+        assert resp.status_code == 200
+        keys = client.get("/api/v1/projects/p1/secret-keys"
+                          ).json()["secret_keys"]
+        assert keys == ["KEY2", "TOKEN"]
+        # values are never exposed over the list API
+        assert "abc" not in client.get(
+            "/api/v1/projects/p1/secret-keys").text
+        client.delete("/api/v1/projects/p1/secrets?secrets=TOKEN")
+        keys = client.get("/api/v1/projects/p1/secret-keys"
+                          ).json()["secret_keys"]
+        assert keys == ["KEY2"]
+
+    def test_pipelines_endpoints(self, client, tmp_path):
+        import mlrun_amd
+
+        project = mlrun_amd.new_project("pipeproj",
+                                        context=str(tmp_path))
+        code = tmp_path / "wf_fn.py"
+        code.write_text("def handler(context):\n"
+                        "    context.log_result('ok', 1)\n")
+        project.set_function(str(code), name="step1", kind="job")
+
+        def workflow(project=None, **kw):
+            from mlrun_amd.projects.operations import run_function
+
+            run_function("step1", handler="handler",
+                         project_object=project)
+
+        status = project.run(workflow_handler=workflow)
+        resp = client.get("/api/v1/projects/pipeproj/pipelines")
+        runs = resp.json()["runs"]
+        assert any(r["run_id"] == status.run_id for r in runs)
+        resp = client.get(
+            f"/api/v1/projects/pipeproj/pipelines/{status.run_id}")
+        assert resp.json()["state"] == "completed"
+        assert client.get(
+            "/api/v1/projects/pipeproj/pipelines/zzz").status_code == 404
