@@ -67,6 +67,29 @@ def new_function(name: str = "", project: str = "", tag: str = "",
     return fn
 
 
+def _notebook_to_code(path: str) -> str:
+    """Extract the code cells of a .ipynb (cells starting with
+    `# mlrun: ignore` and ipython magics are skipped)."""
+    import json as _json
+
+    with open(path) as fp:
+        nb = _json.load(fp)
+    chunks = []
+    for cell in nb.get("cells", []):
+        if cell.get("cell_type") != "code":
+            continue
+        lines = cell.get("source") or []
+        if isinstance(lines, str):
+            lines = lines.splitlines(keepends=True)
+        if lines and lines[0].strip().lower().startswith("# mlrun: ignore"):
+            continue
+        kept = [ln for ln in lines
+                if not ln.lstrip().startswith(("%", "!"))]
+        if kept:
+            chunks.append("".join(kept).rstrip() + "\n")
+    return "\n".join(chunks)
+
+
 def code_to_function(name: str = "", project: str = "", tag: str = "",
                      filename: str = "", handler: str = "", kind: str = "",
                      image: str = "", code_output: str = "", embed_code=True,
@@ -81,11 +104,20 @@ def code_to_function(name: str = "", project: str = "", tag: str = "",
             "the node-local build)")
     if not os.path.isfile(filename):
         raise MLRunInvalidArgumentError(f"file {filename} not found")
+    source = None
+    command = filename
+    if filename.endswith(".ipynb"):
+        # notebook -> code (reference: nuclio-jupyter export); code
+        # cells concatenate, `# mlrun: ignore` cells are dropped
+        source = _notebook_to_code(filename)
+        command = ""
     fn = new_function(name=name or os.path.splitext(
         os.path.basename(filename))[0], project=project, tag=tag,
-        kind=kind or "job", command=filename, image=image, handler=handler)
+        kind=kind or "job", command=command, image=image, handler=handler)
     fn.spec.description = description
-    if embed_code:
+    if source is not None:
+        fn.spec.build["functionSourceCode"] = source
+    elif embed_code:
         with open(filename) as fp:
             fn.spec.build["functionSourceCode"] = fp.read()
     if requirements:

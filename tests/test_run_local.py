@@ -232,3 +232,31 @@ class TestPackagers:
         assert manager.resolve(pd.DataFrame()) is PandasPackager
         with pytest.raises(TypeError):
             manager.resolve(object())
+
+
+class TestNotebookToFunction:
+    def test_code_to_function_from_ipynb(self, rundb, tmp_path):
+        import json
+
+        import mlrun_amd
+
+        nb = {"cells": [
+            {"cell_type": "markdown", "source": ["# title\n"]},
+            {"cell_type": "code",
+             "source": ["%matplotlib inline\n",
+                        "import math\n"]},
+            {"cell_type": "code",
+             "source": ["# mlrun: ignore\n", "print('skipped')\n"]},
+            {"cell_type": "code",
+             "source": ["def handler(context, x=2):\n",
+                        "    context.log_result('sq', x * x)\n"]},
+        ], "metadata": {}, "nbformat": 4, "nbformat_minor": 5}
+        path = tmp_path / "train.ipynb"
+        path.write_text(json.dumps(nb))
+        fn = mlrun_amd.code_to_function(name="nb-fn",
+                                        filename=str(path), kind="job")
+        code = fn.spec.build["functionSourceCode"]
+        assert "def handler" in code and "import math" in code
+        assert "skipped" not in code and "%matplotlib" not in code
+        run = fn.run(handler="handler", params={"x": 5}, local=True)
+        assert run.outputs["sq"] == 25
