@@ -278,3 +278,60 @@ class TestStreamingGPU:
         ev2.body = {"inputs": [[3, 1, 4]], "max_tokens": 5}
         out = srv.do_event(ev2).body["outputs"][0]
         assert per == out
+
+
+@requires_gpu
+class TestGoldenPathGPU:
+    def test_train_then_serve_from_artifact_on_gpu(self, tmp_path):
+        """Tiny llama fine-tune on cuda:0 -> model artifact -> decode
+        engine serves from the artifact bit-exactly (the train->serve
+        bridge on real hardware)."""
+        import mlrun_amd
+        from mlrun_amd.models.llama import LlamaConfig
+        from mlrun_amd.models.llama_train import (LlamaTrainer,
+                                                  export_decode_state)
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        fn = mlrun_amd.new_function(name="gtrain", kind="local")
+
+        state = {}
+
+        def train_handler(context):
+            trainer = LlamaTrainer(cfg, device="cuda:0", lr=1e-3,
+                                   context=context)
+            tokens = torch.randint(
+                0, cfg.vocab_size, (2, 32),
+                generator=torch.Generator().manual_seed(0)).cuda()
+            for _ in range(3):
+                loss = trainer.train_step(tokens)
+            context.log_result("final_loss", loss)
+            state["decode"] = export_decode_state(trainer.model)
+            trainer.save_checkpoint("model")
+
+        run = fn.run(handler=train_handler, local=True)
+        model_uri = run.outputs["model"]
+
+        from mlrun_amd.models.llama import LlamaServer
+
+        server = LlamaServer(name="g", config=cfg, batch_size=2,
+                             max_new_tokens=4, model_path=model_uri,
+                             device="cuda:0", use_graph=True)
+        server.load()
+
+        # the served engine must match a direct export of the trained
+        # weights
+        from mlrun_amd.models.llama import LlamaDecodeEngine
+
+        direct = LlamaDecodeEngine(cfg, 2, device="cuda:0",
+                                   use_graph=True, seed=99)
+        direct.weights.load_state_dict(state["decode"])
+        prompt = torch.randint(0, cfg.vocab_size, (2, 8),
+                               generator=torch.Generator().manual_seed(4))
+        out_direct = direct.generate(prompt, max_new_tokens=4).cpu()
+        out_served = torch.tensor(server.do_event(type(
+            "E", (), {"body": {"inputs": prompt.tolist(),
+                               "max_tokens": 4},
+                      "path": "/infer", "id": "t"})()).body["outputs"])
+        assert torch.equal(out_direct, out_served)
