@@ -106,7 +106,12 @@ class LlamaWeights:
         self.hq, self.hkv, self.inter = hq, hkv, inter
 
     def load_state_dict(self, state: dict):
-        """Load a checkpoint saved by state_dict() (model artifacts)."""
+        """Load a checkpoint saved by state_dict() — or a TRAINING
+        checkpoint (LlamaForCausalLM.state_dict(), keys like
+        "embed.weight"/"blocks.N..."), converted in place: artifacts
+        logged by LlamaTrainer.save_checkpoint serve directly."""
+        if "embed" not in state and "embed.weight" in state:
+            state = _training_state_to_decode(state)
         self.embed.copy_(state["embed"])
         for i, layer in enumerate(self.layers):
             for key in layer:
@@ -121,6 +126,28 @@ class LlamaWeights:
             for key, value in layer.items():
                 out[f"layers.{i}.{key}"] = value
         return out
+
+
+def _training_state_to_decode(state: dict) -> dict:
+    """Map a LlamaForCausalLM (training) state dict onto the decode
+    weight layout (same mapping as llama_train.export_decode_state,
+    duplicated here so serving never imports the training stack)."""
+    out = {
+        "embed": state["embed.weight"],
+        "final_norm": state["final_norm.weight"],
+        "lm_head": state["lm_head.weight"],
+    }
+    i = 0
+    while f"blocks.{i}.attn_norm.weight" in state:
+        prefix = f"blocks.{i}."
+        out[f"layers.{i}.attn_norm"] = state[prefix + "attn_norm.weight"]
+        out[f"layers.{i}.wqkv"] = state[prefix + "attn.wqkv.weight"]
+        out[f"layers.{i}.wo"] = state[prefix + "attn.wo.weight"]
+        out[f"layers.{i}.ffn_norm"] = state[prefix + "ffn_norm.weight"]
+        out[f"layers.{i}.wgu"] = state[prefix + "mlp.wgu.weight"]
+        out[f"layers.{i}.wdown"] = state[prefix + "mlp.wdown.weight"]
+        i += 1
+    return {k: v.to(torch.bfloat16) for k, v in out.items()}
 
 
 class LlamaDecodeEngine:
