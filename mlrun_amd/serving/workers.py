@@ -176,6 +176,7 @@ class WorkerPool:
                             f"serving worker on port {port} did not "
                             f"become ready")
                     time.sleep(0.5)
+        os.unlink(spec_file.name)
         self.proxy = TcpRoundRobinProxy(
             [("127.0.0.1", p) for p in self.ports])
         self.proxy.start()
