@@ -653,7 +653,8 @@ class LlamaServer:
                  config=None, batch_size=16, max_new_tokens=32,
                  device=None, use_graph=True, batch_window_ms=0,
                  replicas=1, weight_dtype="bf16", kv_dtype="bf16",
-                 scheduling="batch", **class_args):
+                 scheduling="batch", stop_token: int = None,
+                 **class_args):
         import queue as queue_mod
         import threading
 
@@ -676,6 +677,7 @@ class LlamaServer:
         self.weight_dtype = weight_dtype
         self.kv_dtype = kv_dtype
         self.scheduling = scheduling  # "batch" | "continuous"
+        self.stop_token = stop_token  # finish a request at this token
         self.replicas = max(int(replicas), 1)
         self.engines: typing.List[LlamaDecodeEngine] = []
         self.batch_window_ms = batch_window_ms
@@ -904,6 +906,7 @@ class LlamaServer:
             active = [i for i in range(B) if slots[i] is not None]
             if not active:
                 continue
+            stop = self.stop_token
             streaming = [i for i in active if slots[i]["stream"]]
             if streaming:
                 # streamers need the admission token on the host too
@@ -920,7 +923,8 @@ class LlamaServer:
                 engine.decode_step()
                 step += 1
                 out_ring[:, step % ring_len].copy_(engine.buf_tokens)
-            if streaming:
+            host_tokens = None
+            if streaming or stop is not None:
                 if engine._serve_stream is not None:
                     engine._serve_stream.synchronize()
                 host_tokens = engine.buf_tokens.cpu()
@@ -930,7 +934,10 @@ class LlamaServer:
             for i in active:
                 state = slots[i]
                 state["produced"] += 1
-                if state["produced"] >= state["max_new"]:
+                hit_stop = stop is not None and \
+                    int(host_tokens[i]) == stop
+                if state["produced"] >= state["max_new"] or hit_stop:
+                    state["max_new"] = state["produced"]  # actual len
                     finished.append(i)
             if finished:
                 if engine._serve_stream is not None:
@@ -1051,7 +1058,15 @@ class LlamaServer:
                 prompt, dtype=torch.int64)  # left-pad
         engine.reset()
         out = engine.generate(tokens, max_new_tokens=max_new)
-        return out[:n].cpu().tolist()
+        rows = out[:n].cpu().tolist()
+        if self.stop_token is not None:
+            trimmed = []
+            for row in rows:
+                if self.stop_token in row:
+                    row = row[:row.index(self.stop_token) + 1]
+                trimmed.append(row)
+            return trimmed
+        return rows
 
     def logged_results(self, request, response, op):
         return request.get("inputs"), None

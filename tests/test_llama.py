@@ -307,3 +307,43 @@ class TestTokenStreaming:
         good.body = {"inputs": [[1, 2, 3]], "max_tokens": 3}
         out = srv.do_event(good).body["outputs"]
         assert len(out[0]) == 3
+
+
+class TestStopToken:
+    def test_continuous_stops_early_and_batch_trims(self):
+        cfg = LlamaConfig.tiny()
+        probe = LlamaServer(name="p", config=cfg, batch_size=2,
+                            max_new_tokens=8, use_graph=False)
+        probe.load()
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        ev = _Ev()
+        ev.body = {"inputs": [[1, 2, 3]], "max_tokens": 8}
+        full = probe.do_event(ev).body["outputs"][0]
+        stop = full[2]  # make a generated token the stop token
+        expect = full[:full.index(stop) + 1]  # up to FIRST occurrence
+
+        cont = LlamaServer(name="c", config=cfg, batch_size=2,
+                           max_new_tokens=8, scheduling="continuous",
+                           use_graph=False, stop_token=stop)
+        cont.load()
+        cont.engines[0].weights.load_state_dict(
+            probe.engines[0].weights.state_dict())
+        ev2 = _Ev()
+        ev2.body = {"inputs": [[1, 2, 3]], "max_tokens": 8}
+        out_c = cont.do_event(ev2).body["outputs"][0]
+        assert out_c == expect  # ends AT the stop token
+
+        batch = LlamaServer(name="b", config=cfg, batch_size=2,
+                            max_new_tokens=8, use_graph=False,
+                            stop_token=stop)
+        batch.load()
+        batch.engines[0].weights.load_state_dict(
+            probe.engines[0].weights.state_dict())
+        ev3 = _Ev()
+        ev3.body = {"inputs": [[1, 2, 3]], "max_tokens": 8}
+        out_b = batch.do_event(ev3).body["outputs"][0]
+        assert out_b == expect
