@@ -157,7 +157,8 @@ class LlamaDecodeEngine:
     def __init__(self, cfg: LlamaConfig, batch_size: int, device=None,
                  tp_group=None, tp_rank=0, tp_size=1, use_graph=True,
                  seed=1234, weights: "LlamaWeights" = None,
-                 weight_dtype: str = "bf16", kv_dtype: str = "bf16"):
+                 weight_dtype: str = "bf16", kv_dtype: str = "bf16",
+                 temperature: float = 0.0, top_k: int = 0):
         self.cfg = cfg
         self.B = batch_size
         self.device = torch.device(
@@ -167,6 +168,11 @@ class LlamaDecodeEngine:
         self.tp_rank = tp_rank
         self.tp_size = tp_size
         self.use_graph = use_graph and self.on_gpu
+        # sampling: 0 = greedy argmax; otherwise Gumbel-max over
+        # logits/T (graph-safe: torch captures RNG ops so every
+        # replay draws fresh noise), optionally top-k restricted
+        self.temperature = float(temperature)
+        self.top_k = int(top_k)
         # weights are read-only at serving time: replica engines on the
         # same GPU share one copy (16 GB for 8B) and keep private
         # KV caches / buffers / streams
@@ -427,7 +433,32 @@ class LlamaDecodeEngine:
                                       out8=h8, out_scale=h8s)
         self._gemm(self.buf_hidden, w.lm_head, self.buf_logits,
                    a8=h8, a_scale=h8s)
-        torch.argmax(self.buf_logits, dim=-1, out=self.buf_tokens)
+        self._select_tokens(self.buf_logits, out=self.buf_tokens)
+
+    def _select_tokens(self, logits: torch.Tensor,
+                       out: torch.Tensor = None) -> torch.Tensor:
+        """Greedy argmax, or temperature/top-k sampling via the
+        Gumbel-max trick: argmax(logits/T + G), G = -log(-log(U)).
+        One argmax either way — capture-safe and branch-free on
+        device."""
+        if self.temperature <= 0.0:
+            if out is None:
+                return logits.argmax(dim=-1)
+            torch.argmax(logits, dim=-1, out=out)
+            return out
+        scores = logits.float() / self.temperature
+        if self.top_k > 0 and self.top_k < scores.shape[-1]:
+            kth = torch.topk(scores, self.top_k, dim=-1
+                             ).values[..., -1:]
+            scores = torch.where(scores < kth,
+                                 torch.full_like(scores, float("-inf")),
+                                 scores)
+        u = torch.rand_like(scores).clamp_(1e-9, 1.0 - 1e-9)
+        gumbel = -torch.log(-torch.log(u))
+        if out is None:
+            return (scores + gumbel).argmax(dim=-1)
+        torch.argmax(scores + gumbel, dim=-1, out=out)
+        return out
 
     _capture_lock = None
 
@@ -603,7 +634,7 @@ class LlamaDecodeEngine:
         self.cache_lens[slots] = S
         last = hidden.view(Bp, S, -1)[:, -1]
         logits = last @ w.lm_head.t()
-        self.buf_tokens[slots] = logits.argmax(dim=-1)
+        self.buf_tokens[slots] = self._select_tokens(logits)
         return logits
 
     # -------------------------------------------------------- generate
@@ -615,7 +646,7 @@ class LlamaDecodeEngine:
         assert B == self.B, f"engine built for batch {self.B}, got {B}"
         assert S + max_new_tokens <= self.cfg.max_seq_len
         logits = self.prefill(tokens)
-        next_tokens = logits.argmax(dim=-1)
+        next_tokens = self._select_tokens(logits)
         self.buf_tokens.copy_(next_tokens)
         generated = [next_tokens.clone()]
         if self.use_graph and self._graph is None:
@@ -654,6 +685,7 @@ class LlamaServer:
                  device=None, use_graph=True, batch_window_ms=0,
                  replicas=1, weight_dtype="bf16", kv_dtype="bf16",
                  scheduling="batch", stop_token: int = None,
+                 temperature: float = 0.0, top_k: int = 0,
                  **class_args):
         import queue as queue_mod
         import threading
@@ -678,6 +710,8 @@ class LlamaServer:
         self.kv_dtype = kv_dtype
         self.scheduling = scheduling  # "batch" | "continuous"
         self.stop_token = stop_token  # finish a request at this token
+        self.temperature = float(temperature)
+        self.top_k = int(top_k)
         self.replicas = max(int(replicas), 1)
         self.engines: typing.List[LlamaDecodeEngine] = []
         self.batch_window_ms = batch_window_ms
@@ -728,13 +762,16 @@ class LlamaServer:
         first = LlamaDecodeEngine(cfg, self.batch_size, device=self.device,
                                   use_graph=self.use_graph,
                                   weight_dtype=self.weight_dtype,
-                                  kv_dtype=self.kv_dtype)
+                                  kv_dtype=self.kv_dtype,
+                                  temperature=self.temperature,
+                                  top_k=self.top_k)
         self.engines = [first]
         for _ in range(self.replicas - 1):
             replica = LlamaDecodeEngine(
                 cfg, self.batch_size, device=self.device,
                 use_graph=self.use_graph, weights=first.weights,
-                weight_dtype="bf16", kv_dtype=self.kv_dtype)
+                weight_dtype="bf16", kv_dtype=self.kv_dtype,
+                temperature=self.temperature, top_k=self.top_k)
             replica.weight_dtype = first.weight_dtype
             replica._fp8_packs = first._fp8_packs  # shared packs
             if first.weight_dtype == "fp8w" and replica.on_gpu:
