@@ -335,3 +335,30 @@ class TestGoldenPathGPU:
                                "max_tokens": 4},
                       "path": "/infer", "id": "t"})()).body["outputs"])
         assert torch.equal(out_direct, out_served)
+
+
+@requires_gpu
+class TestSamplingGPU:
+    def test_sampling_under_hipgraph(self):
+        """RNG ops captured in the hipGraph must draw fresh noise per
+        replay (token diversity) while T=0 stays greedy-exact."""
+        from mlrun_amd.models.llama import LlamaConfig, LlamaDecodeEngine
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        prompt = torch.randint(0, 2000, (2, 8),
+                               generator=torch.Generator().manual_seed(1))
+        sampler = LlamaDecodeEngine(cfg, 2, device="cuda:0",
+                                    use_graph=True, seed=31,
+                                    temperature=1.0)
+        out = sampler.generate(prompt, max_new_tokens=8).cpu()
+        # fresh noise per replay: a high-T sample should not be one
+        # token repeated 8 times for both rows
+        assert not all(row.eq(row[0]).all() for row in out)
+        greedy_g = LlamaDecodeEngine(cfg, 2, device="cuda:0",
+                                     use_graph=True, seed=31)
+        greedy_e = LlamaDecodeEngine(cfg, 2, device="cuda:0",
+                                     use_graph=False, seed=31)
+        assert torch.equal(greedy_g.generate(prompt, 6).cpu(),
+                           greedy_e.generate(prompt, 6).cpu())
