@@ -170,9 +170,17 @@ class LlamaDecodeEngine:
         self.use_graph = use_graph and self.on_gpu
         # sampling: 0 = greedy argmax; otherwise Gumbel-max over
         # logits/T (graph-safe: torch captures RNG ops so every
-        # replay draws fresh noise), optionally top-k restricted
+        # replay draws fresh noise), optionally top-k restricted.
+        # temperature < 0 enables PER-SLOT temperatures via buf_temp
+        # (one graph serves mixed greedy/sampled requests: greedy
+        # slots get a tiny T so the scaled logit gap swamps the noise)
         self.temperature = float(temperature)
         self.top_k = int(top_k)
+        self.per_slot_temp = self.temperature < 0
+        self.buf_temp = torch.full((batch_size, 1), 1e-4,
+                                   dtype=torch.float32,
+                                   device=self.device) \
+            if self.per_slot_temp else None
         # weights are read-only at serving time: replica engines on the
         # same GPU share one copy (16 GB for 8B) and keep private
         # KV caches / buffers / streams
@@ -441,6 +449,14 @@ class LlamaDecodeEngine:
         Gumbel-max trick: argmax(logits/T + G), G = -log(-log(U)).
         One argmax either way — capture-safe and branch-free on
         device."""
+        if self.per_slot_temp and logits.shape[0] == self.B:
+            scores = logits.float() / self.buf_temp
+            u = torch.rand_like(scores).clamp_(1e-9, 1.0 - 1e-9)
+            gumbel = -torch.log(-torch.log(u))
+            if out is None:
+                return (scores + gumbel).argmax(dim=-1)
+            torch.argmax(scores + gumbel, dim=-1, out=out)
+            return out
         if self.temperature <= 0.0:
             if out is None:
                 return logits.argmax(dim=-1)
@@ -634,7 +650,13 @@ class LlamaDecodeEngine:
         self.cache_lens[slots] = S
         last = hidden.view(Bp, S, -1)[:, -1]
         logits = last @ w.lm_head.t()
-        self.buf_tokens[slots] = self._select_tokens(logits)
+        if self.per_slot_temp:
+            scores = logits.float() / self.buf_temp[slots]
+            u = torch.rand_like(scores).clamp_(1e-9, 1.0 - 1e-9)
+            self.buf_tokens[slots] = (
+                scores - torch.log(-torch.log(u))).argmax(dim=-1)
+        else:
+            self.buf_tokens[slots] = self._select_tokens(logits)
         return logits
 
     # -------------------------------------------------------- generate
@@ -830,7 +852,7 @@ class LlamaServer:
                 if not future.done():
                     future.set_exception(exc)
 
-    def _stream_generate(self, inputs, max_new, event):
+    def _stream_generate(self, inputs, max_new, event, req_temp=None):
         """Token streaming (continuous scheduling only): yields one
         json line per generated token as the slots produce them."""
         import concurrent.futures
@@ -841,7 +863,8 @@ class LlamaServer:
         for prompt in inputs:
             future = concurrent.futures.Future()
             stream_q: "queue_mod.Queue" = queue_mod.Queue()
-            self._tasks.put((prompt, max_new, future, stream_q))
+            self._tasks.put((prompt, max_new, future, stream_q,
+                             req_temp))
             streams.append(stream_q)
 
         def gen():
@@ -908,6 +931,10 @@ class LlamaServer:
                 for item, slot in zip(taken, free):
                     prompt, max_new, future = item[:3]
                     stream_q = item[3] if len(item) > 3 else None
+                    temp = item[4] if len(item) > 4 else None
+                    if engine.per_slot_temp:
+                        engine.buf_temp[slot] = max(
+                            float(temp or 0.0), 1e-4)
                     max_new = max(1, min(max_new,
                                          engine.cfg.max_seq_len - 2))
                     keep = engine.cfg.max_seq_len - max_new - 1
@@ -1005,14 +1032,17 @@ class LlamaServer:
         if inputs is None:
             raise ValueError('expected {"inputs": [[token ids], ...]}')
         max_new = int(body.get("max_tokens", self.max_new_tokens))
+        req_temp = body.get("temperature")
         if self.scheduling == "continuous" and body.get("stream"):
-            event.body = self._stream_generate(inputs, max_new, event)
+            event.body = self._stream_generate(inputs, max_new, event,
+                                               req_temp)
             return event
         if self.scheduling == "continuous":
             futures = []
             for prompt in inputs:
                 future = concurrent.futures.Future()
-                self._tasks.put((prompt, max_new, future))
+                self._tasks.put((prompt, max_new, future, None,
+                                 req_temp))
                 futures.append(future)
             outputs = [f.result(timeout=600) for f in futures]
         elif self.batch_window_ms and len(inputs) < self.batch_size:

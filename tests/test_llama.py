@@ -400,3 +400,38 @@ class TestSampling:
         match = (engine.generate(toks, 8) ==
                  greedy.generate(toks, 8)).float().mean()
         assert match > 0.7
+
+    def test_per_request_temperature_mixed_batch(self):
+        """temperature=-1 engines serve mixed greedy/sampled requests
+        on ONE graph: a temperature=0 request reproduces the greedy
+        output while a high-T request diverges."""
+        cfg = LlamaConfig.tiny()
+        greedy = LlamaServer(name="g", config=cfg, batch_size=2,
+                             max_new_tokens=8, use_graph=False)
+        greedy.load()
+        mixed = LlamaServer(name="m", config=cfg, batch_size=2,
+                            max_new_tokens=8, scheduling="continuous",
+                            use_graph=False, temperature=-1)
+        mixed.load()
+        mixed.engines[0].weights.load_state_dict(
+            greedy.engines[0].weights.state_dict())
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        base = _Ev()
+        base.body = {"inputs": [[1, 2, 3]], "max_tokens": 6}
+        expect = greedy.do_event(base).body["outputs"][0]
+
+        cold = _Ev()
+        cold.body = {"inputs": [[1, 2, 3]], "max_tokens": 6,
+                     "temperature": 0}
+        assert mixed.do_event(cold).body["outputs"][0] == expect
+
+        torch.manual_seed(3)
+        hot = _Ev()
+        hot.body = {"inputs": [[1, 2, 3]], "max_tokens": 6,
+                    "temperature": 50.0}
+        sampled = mixed.do_event(hot).body["outputs"][0]
+        assert sampled != expect
