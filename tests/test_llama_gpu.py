@@ -362,3 +362,37 @@ class TestSamplingGPU:
                                      use_graph=False, seed=31)
         assert torch.equal(greedy_g.generate(prompt, 6).cpu(),
                            greedy_e.generate(prompt, 6).cpu())
+
+    def test_per_request_temperature_under_hipgraph(self):
+        from mlrun_amd.models.llama import LlamaConfig, LlamaServer
+
+        cfg = LlamaConfig.tiny(num_layers=2, num_heads=4, num_kv_heads=2,
+                               hidden_size=512, intermediate_size=1024,
+                               vocab_size=2048)
+        greedy = LlamaServer(name="pg", config=cfg, batch_size=2,
+                             max_new_tokens=6, use_graph=True,
+                             device="cuda:0")
+        greedy.load()
+        mixed = LlamaServer(name="pm", config=cfg, batch_size=2,
+                            max_new_tokens=6, scheduling="continuous",
+                            use_graph=True, device="cuda:0",
+                            temperature=-1)
+        mixed.load()
+        mixed.engines[0].weights.load_state_dict(
+            greedy.engines[0].weights.state_dict())
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        base = _Ev()
+        base.body = {"inputs": [[7, 8, 9]], "max_tokens": 5}
+        expect = greedy.do_event(base).body["outputs"][0]
+        cold = _Ev()
+        cold.body = {"inputs": [[7, 8, 9]], "max_tokens": 5,
+                     "temperature": 0}
+        assert mixed.do_event(cold).body["outputs"][0] == expect
+        hot = _Ev()
+        hot.body = {"inputs": [[7, 8, 9]], "max_tokens": 5,
+                    "temperature": 100.0}
+        assert mixed.do_event(hot).body["outputs"][0] != expect
