@@ -75,3 +75,21 @@ class TestLogCollector:
         with pytest.raises(Exception):
             daemon.get_logs("nope", "p")
         assert daemon.get_log_size("nope", "p") == -1
+
+    def test_list_runs_in_progress(self, daemon, tmp_path):
+        src = tmp_path / "live.log"
+        src.write_text("streaming\n")
+        daemon.start_log("uid-live", "proj-x", str(src))
+        import time
+
+        deadline = time.time() + 5
+        while time.time() < deadline:
+            runs = daemon.list_runs_in_progress() \
+                if hasattr(daemon, "list_runs_in_progress") else \
+                daemon._call({"op": "list_runs_in_progress"}).get(
+                    "runs", [])
+            if any("uid-live" in str(r) for r in runs):
+                break
+            time.sleep(0.2)
+        assert any("uid-live" in str(r) for r in runs)
+        daemon.stop_logs("proj-x", "uid-live")
