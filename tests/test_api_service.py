@@ -336,3 +336,22 @@ class TestSchedulerMisfire:
             assert not sched.get("last_run_uri")  # beyond grace window
         finally:
             scheduler.stop()
+
+
+def test_runs_pagination(client, rundb):
+    for i in range(7):
+        rundb.store_run({"metadata": {"name": f"r{i}", "uid": f"u{i}"},
+                         "status": {"state": "completed"}},
+                        f"u{i}", "default")
+    resp = client.get("/api/v1/runs?page=1&page_size=3")
+    body = resp.json()
+    assert len(body["runs"]) == 3
+    assert body["pagination"]["page"] == 1
+    resp2 = client.get("/api/v1/runs?page=3&page_size=3")
+    assert len(resp2.json()["runs"]) == 1  # 7 = 3 + 3 + 1
+    uids = {r["metadata"]["uid"] for r in body["runs"]} | \
+        {r["metadata"]["uid"]
+         for r in client.get("/api/v1/runs?page=2&page_size=3"
+                             ).json()["runs"]} | \
+        {r["metadata"]["uid"] for r in resp2.json()["runs"]}
+    assert len(uids) == 7  # pages partition the set
