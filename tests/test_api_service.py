@@ -338,11 +338,11 @@ class TestSchedulerMisfire:
             scheduler.stop()
 
 
-def test_runs_pagination(client, rundb):
+def test_runs_pagination(client):
     for i in range(7):
-        rundb.store_run({"metadata": {"name": f"r{i}", "uid": f"u{i}"},
-                         "status": {"state": "completed"}},
-                        f"u{i}", "default")
+        client.db.store_run(
+            {"metadata": {"name": f"r{i}", "uid": f"u{i}"},
+             "status": {"state": "completed"}}, f"u{i}", "default")
     resp = client.get("/api/v1/runs?page=1&page_size=3")
     body = resp.json()
     assert len(body["runs"]) == 3
