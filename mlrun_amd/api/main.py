@@ -378,6 +378,58 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         db.delete_project_secrets(project, keys)
         return {}
 
+    # ------------------------------------------------------ workflows
+    @app.get("/api/v1/projects/{project}/workflows")
+    async def list_workflows(project: str):
+        body = db.get_project(project)
+        spec = (body or {}).get("spec", {})
+        return {"workflows": spec.get("workflows", [])}
+
+    @app.post("/api/v1/projects/{project}/workflows/{name}/submit")
+    async def submit_workflow(project: str, name: str,
+                              request: Request):
+        """Server-side workflow execution (reference _RemoteRunner /
+        workflows endpoint): loads the stored project and runs the
+        named workflow in a background task."""
+        import threading
+
+        try:
+            payload = await request.json()
+        except Exception:
+            payload = {}
+        body = db.get_project(project)
+        if not body:
+            raise HTTPException(status_code=404,
+                                detail=f"project {project} not found")
+        from ..model import generate_uid
+        from ..projects import MlrunProject
+
+        proj = MlrunProject.from_dict(body)
+        task_name = f"workflow-{name}-{generate_uid()[:8]}"
+        db.store_background_task(project, {
+            "name": task_name, "status": {"state": "running"}})
+
+        def run_it():
+            try:
+                status = proj.run(
+                    name=name,
+                    arguments=payload.get("arguments") or {})
+                db.store_background_task(project, {
+                    "name": task_name,
+                    "status": {"state": "succeeded"
+                               if status.state == "completed"
+                               else "failed"}})
+            except Exception as exc:
+                logger.error("workflow run failed", error=str(exc))
+                db.store_background_task(project, {
+                    "name": task_name,
+                    "status": {"state": "failed"}})
+
+        threading.Thread(target=run_it, daemon=True,
+                         name=task_name).start()
+        return {"name": name, "project": project,
+                "background_task": task_name}
+
     # ------------------------------------------------------ pipelines
     @app.get("/api/v1/projects/{project}/pipelines")
     async def list_pipelines(project: str):

@@ -355,3 +355,42 @@ def test_runs_pagination(client):
                              ).json()["runs"]} | \
         {r["metadata"]["uid"] for r in resp2.json()["runs"]}
     assert len(uids) == 7  # pages partition the set
+
+
+def test_workflow_submit_endpoint(client, tmp_path):
+    import time
+
+    import mlrun_amd
+
+    project = mlrun_amd.new_project("wfapi", context=str(tmp_path),
+                                    save=False)
+    code = tmp_path / "wf_step.py"
+    code.write_text("def handler(context):\n"
+                    "    context.log_result('done', 1)\n")
+    project.set_function(str(code), name="s1", kind="job")
+    wf = tmp_path / "flow.py"
+    wf.write_text(
+        "def pipeline(project=None, **kw):\n"
+        "    from mlrun_amd.projects.operations import run_function\n"
+        "    run_function('s1', handler='handler',\n"
+        "                 project_object=project)\n")
+    project.set_workflow("main", str(wf), handler="pipeline")
+    client.db.store_project("wfapi", project.to_dict())
+
+    resp = client.get("/api/v1/projects/wfapi/workflows")
+    assert resp.json()["workflows"][0]["name"] == "main"
+
+    resp = client.post("/api/v1/projects/wfapi/workflows/main/submit",
+                       json={})
+    assert resp.status_code == 200
+    task = resp.json()["background_task"]
+    deadline = time.time() + 30
+    state = "running"
+    while time.time() < deadline:
+        state = client.get(
+            f"/api/v1/projects/wfapi/background-tasks/{task}"
+        ).json().get("status", {}).get("state", "running")
+        if state in ("succeeded", "failed"):
+            break
+        time.sleep(0.3)
+    assert state == "succeeded"
