@@ -326,6 +326,39 @@ class HTTPRunDB(RunDBInterface):
         return resp.get("data", resp)
 
     # -------------------------------------------------- background tasks
+    # --------------------------------------------------- new surfaces
+    def submit_workflow(self, project, name, arguments=None):
+        return self.api_call(
+            "POST", f"projects/{project}/workflows/{name}/submit",
+            json_body={"arguments": arguments or {}})
+
+    def list_workflows(self, project):
+        return self.api_call(
+            "GET", f"projects/{project}/workflows").get("workflows", [])
+
+    def store_project_secrets(self, project, secrets: dict,
+                              provider="kubernetes"):
+        self.api_call("POST", f"projects/{project}/secrets",
+                      json_body={"secrets": secrets})
+
+    def list_project_secret_keys(self, project, provider="kubernetes"):
+        return self.api_call(
+            "GET", f"projects/{project}/secret-keys"
+        ).get("secret_keys", [])
+
+    def delete_project_secrets(self, project, keys=None,
+                               provider="kubernetes"):
+        self.api_call("DELETE", f"projects/{project}/secrets",
+                      params={"secrets": ",".join(keys or [])})
+
+    def get_hub_catalog(self, source="builtin"):
+        return self.api_call(
+            "GET", f"hub/sources/{source}/items").get("catalog", [])
+
+    def list_pipelines(self, project):
+        return self.api_call(
+            "GET", f"projects/{project}/pipelines").get("runs", [])
+
     def get_background_task(self, project, name):
         return self.api_call(
             "GET", f"projects/{project}/background-tasks/{name}")

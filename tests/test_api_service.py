@@ -394,3 +394,31 @@ def test_workflow_submit_endpoint(client, tmp_path):
             break
         time.sleep(0.3)
     assert state == "succeeded"
+
+
+def test_httpdb_new_surfaces_roundtrip(client, tmp_path, monkeypatch):
+    """HTTPRunDB client methods against the live app (secrets, hub,
+    workflows, pipelines)."""
+    from mlrun_amd.db.httpdb import HTTPRunDB
+
+    db = HTTPRunDB("http://testserver")
+    # route the client's HTTP through the TestClient
+    def api_call(method, path, params=None, json_body=None, **kw):
+        url = f"/api/v1/{path}"
+        resp = client.request(method, url, params=params,
+                              json=json_body)
+        assert resp.status_code < 500, resp.text
+        try:
+            return resp.json()
+        except ValueError:
+            return {}
+    monkeypatch.setattr(db, "api_call", api_call)
+
+    db.store_project_secrets("hp", {"A": "1", "B": "2"})
+    assert db.list_project_secret_keys("hp") == ["A", "B"]
+    db.delete_project_secrets("hp", ["A"])
+    assert db.list_project_secret_keys("hp") == ["B"]
+    items = [i["name"] for i in db.get_hub_catalog("builtin")]
+    assert "llm-serving" in items
+    assert db.list_pipelines("hp") == []
+    assert db.list_workflows("hp") == []
