@@ -134,3 +134,8 @@ def get_current_project(silent: bool = False):
         raise MLRunInvalidArgumentError(
             "no project is active; load/create one first")
     return pipeline_context.project
+
+# expose common submodules as attributes (reference: mlrun.feature_store
+# etc. are importable directly off the package)
+from . import feature_store, serving  # noqa: F401,E402
+from .artifacts import get_model  # noqa: F401,E402
