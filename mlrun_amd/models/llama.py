@@ -967,6 +967,22 @@ class LlamaServer:
                     logger.error("continuous admission failed",
                                  error=str(exc))
                     continue
+                if self.stop_token is not None:
+                    # the ADMISSION token can already be the stop token
+                    if engine._serve_stream is not None:
+                        engine._serve_stream.synchronize()
+                    first = engine.buf_tokens.cpu()
+                    for slot in admit_slots:
+                        if int(first[slot]) != self.stop_token:
+                            continue
+                        state = slots[slot]
+                        if state["stream"]:
+                            state["stream"].put(int(first[slot]))
+                            state["stream"].put(None)
+                            state["_first_sent"] = True
+                        state["future"].set_result(
+                            [int(first[slot])])
+                        slots[slot] = None
             active = [i for i in range(B) if slots[i] is not None]
             if not active:
                 continue

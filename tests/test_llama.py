@@ -435,3 +435,59 @@ class TestSampling:
                     "temperature": 50.0}
         sampled = mixed.do_event(hot).body["outputs"][0]
         assert sampled != expect
+
+
+class TestServingModeInterplay:
+    def test_stream_plus_stop_token_plus_mixed_temp(self):
+        """All serving controls together: streaming request with a
+        stop token and per-request temperature, alongside a plain
+        request, on one continuous engine."""
+        import json
+        import threading
+
+        cfg = LlamaConfig.tiny()
+        probe = LlamaServer(name="p", config=cfg, batch_size=2,
+                            max_new_tokens=8, use_graph=False)
+        probe.load()
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        ev = _Ev()
+        ev.body = {"inputs": [[4, 5, 6]], "max_tokens": 8}
+        full = probe.do_event(ev).body["outputs"][0]
+        stop = full[3]
+        expect = full[:full.index(stop) + 1]
+
+        srv = LlamaServer(name="mix", config=cfg, batch_size=2,
+                          max_new_tokens=8, scheduling="continuous",
+                          use_graph=False, stop_token=stop,
+                          temperature=-1)
+        srv.load()
+        srv.engines[0].weights.load_state_dict(
+            probe.engines[0].weights.state_dict())
+
+        results = {}
+
+        def plain():
+            ev2 = _Ev()
+            ev2.body = {"inputs": [[4, 5, 6]], "max_tokens": 8,
+                        "temperature": 0}
+            results["plain"] = srv.do_event(ev2).body["outputs"][0]
+
+        def stream():
+            ev3 = _Ev()
+            ev3.body = {"inputs": [[4, 5, 6]], "max_tokens": 8,
+                        "temperature": 0, "stream": True}
+            gen = srv.do_event(ev3).body
+            results["stream"] = [json.loads(line)["token"]
+                                 for line in gen]
+
+        threads = [threading.Thread(target=plain),
+                   threading.Thread(target=stream)]
+        [t.start() for t in threads]
+        [t.join(timeout=60) for t in threads]
+        assert results["plain"] == expect  # stop token honored
+        # the stream also ends at the stop token
+        assert results["stream"][:len(expect)] == expect
