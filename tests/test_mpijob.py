@@ -135,3 +135,69 @@ class TestMpiRuntime:
         env = fn.rccl_env()
         assert env["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
         assert "NCCL_MIN_NCHANNELS" in env
+
+
+class TestGpuAllocatorContention:
+    def test_blocking_waiters_get_released_devices(self):
+        import threading
+        import time
+
+        from mlrun_amd.parallel.scheduler import GpuAllocator
+
+        allocator = GpuAllocator(total=2)
+        lease = allocator.acquire(2, owner="first")
+        got = {}
+
+        def waiter():
+            got["lease"] = allocator.acquire(1, owner="second",
+                                             timeout=10)
+
+        thread = threading.Thread(target=waiter)
+        thread.start()
+        time.sleep(0.3)
+        assert "lease" not in got  # still blocked
+        lease.release()
+        thread.join(timeout=10)
+        assert got["lease"].devices[0] in (0, 1)
+        assert allocator.usage() == {got["lease"].devices[0]: "second"}
+        got["lease"].release()
+        assert allocator.available() == [0, 1]
+
+    def test_nonblocking_raises_and_overask_rejected(self):
+        import pytest as _pytest
+
+        from mlrun_amd.errors import MLRunRuntimeError
+        from mlrun_amd.parallel.scheduler import GpuAllocator
+
+        allocator = GpuAllocator(total=1)
+        lease = allocator.acquire(1)
+        with _pytest.raises(MLRunRuntimeError):
+            allocator.acquire(1, block=False)
+        with _pytest.raises(MLRunRuntimeError):
+            allocator.acquire(2)  # more than the node has
+        lease.release()
+
+    def test_concurrent_acquire_release_consistent(self):
+        import threading
+
+        from mlrun_amd.parallel.scheduler import GpuAllocator
+
+        allocator = GpuAllocator(total=4)
+        errors = []
+
+        def churn(i):
+            try:
+                for _ in range(25):
+                    lease = allocator.acquire(1, owner=f"w{i}",
+                                              timeout=30)
+                    assert len(set(lease.devices)) == 1
+                    lease.release()
+            except Exception as exc:
+                errors.append(exc)
+
+        threads = [threading.Thread(target=churn, args=(i,))
+                   for i in range(8)]
+        [t.start() for t in threads]
+        [t.join(timeout=60) for t in threads]
+        assert not errors
+        assert allocator.available() == [0, 1, 2, 3]
