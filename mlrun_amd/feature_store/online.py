@@ -143,6 +143,9 @@ class WindowRing:
         # the oldest partial bucket are included — bucket-quantized
         # sliding window, storey-compatible)
         window_periods = max(window_seconds // self.period_seconds, 1) + 1
+        # a window longer than the ring span would re-count wrapped
+        # buckets — clamp to one pass over the ring
+        window_periods = min(window_periods, self.n_periods)
         current_period = int(now_ts // self.period_seconds)
         self._expire_old_periods(current_period)
         out = ops.window_reduce(self.ring, window_periods,

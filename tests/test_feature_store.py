@@ -314,3 +314,39 @@ class TestJoinGraph:
         assert graph.how_for("c") == "outer"
         assert graph.how_for("a") == "inner"
         assert graph.how_for("unknown") == "inner"
+
+
+class TestRingGrowth:
+    def test_capacity_grows_preserving_state(self):
+        import torch
+
+        from mlrun_amd.feature_store.online import WindowRing
+
+        ring = WindowRing(60, 4, capacity=4)
+        ring.ingest(torch.tensor([0, 1, 2]),
+                    torch.tensor([1.0, 2.0, 3.0]),
+                    torch.tensor([30.0, 30.0, 30.0]))
+        before = ring.window_values(60, 30.0)["sum"][:3].clone()
+        ring.grow(64)
+        assert ring.capacity == 64
+        after = ring.window_values(60, 30.0)["sum"][:3]
+        assert torch.equal(before, after)
+        # new keys land in grown space
+        ring.ingest(torch.tensor([50]), torch.tensor([9.0]),
+                    torch.tensor([31.0]))
+        assert ring.window_values(60, 31.0)["sum"][50].item() == 9.0
+
+    def test_full_ring_wrap_expires_everything(self):
+        import torch
+
+        from mlrun_amd.feature_store.online import WindowRing
+
+        ring = WindowRing(60, 4, capacity=4)
+        ring.ingest(torch.tensor([0]), torch.tensor([5.0]),
+                    torch.tensor([30.0]))
+        # jump forward past the whole ring span: old partials expire
+        ring.ingest(torch.tensor([1]), torch.tensor([7.0]),
+                    torch.tensor([30.0 + 60 * 10]))
+        vals = ring.window_values(600, 30.0 + 60 * 10)
+        assert vals["sum"][0].item() == 0.0
+        assert vals["sum"][1].item() == 7.0
