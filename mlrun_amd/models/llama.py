@@ -944,21 +944,29 @@ class LlamaServer:
                     slots[slot] = {"future": future, "max_new": max_new,
                                    "start": step, "produced": 1,
                                    "stream": stream_q}
-                max_len = max(len(p) for p in prompts)
-                tokens = torch.zeros(len(prompts), max_len,
-                                     dtype=torch.int64)
-                for i, prompt in enumerate(prompts):
-                    tokens[i, max_len - len(prompt):] = torch.tensor(
-                        prompt, dtype=torch.int64)
-                slot_ids = torch.tensor(admit_slots, dtype=torch.long)
+                # prefill per EXACT prompt length: left-padding a
+                # mixed-length admission group would make a request's
+                # output depend on co-arriving traffic (pad tokens are
+                # attended) — measured by the scheduler property test
+                by_len: dict = {}
+                for prompt, slot in zip(prompts, admit_slots):
+                    by_len.setdefault(len(prompt), []).append(
+                        (prompt, slot))
                 try:
-                    with stream_ctx():
-                        engine.prefill_slots(tokens, slot_ids)
-                        out_ring[slot_ids, step % ring_len] = \
-                            engine.buf_tokens[slot_ids]
+                    for length, group in by_len.items():
+                        tokens = torch.tensor(
+                            [p for p, _ in group], dtype=torch.int64)
+                        slot_ids = torch.tensor(
+                            [sl for _, sl in group], dtype=torch.long)
+                        with stream_ctx():
+                            engine.prefill_slots(tokens, slot_ids)
+                            out_ring[slot_ids, step % ring_len] = \
+                                engine.buf_tokens[slot_ids]
                 except Exception as exc:  # admission failed: fail the
                     for slot in admit_slots:  # requests, free the slots
                         state = slots[slot]
+                        if state is None:
+                            continue
                         if not state["future"].done():
                             state["future"].set_exception(exc)
                         if state["stream"]:
@@ -967,21 +975,26 @@ class LlamaServer:
                     logger.error("continuous admission failed",
                                  error=str(exc))
                     continue
-                if self.stop_token is not None:
-                    # the ADMISSION token can already be the stop token
+                done_at_admit = [
+                    slot for slot in admit_slots
+                    if slots[slot]["max_new"] <= 1]
+                if self.stop_token is not None or done_at_admit:
+                    # the ADMISSION token can already be the stop
+                    # token, and max_tokens=1 requests finish here
                     if engine._serve_stream is not None:
                         engine._serve_stream.synchronize()
                     first = engine.buf_tokens.cpu()
                     for slot in admit_slots:
-                        if int(first[slot]) != self.stop_token:
-                            continue
                         state = slots[slot]
+                        token = int(first[slot])
+                        if token != self.stop_token and \
+                                state["max_new"] > 1:
+                            continue
                         if state["stream"]:
-                            state["stream"].put(int(first[slot]))
+                            state["stream"].put(token)
                             state["stream"].put(None)
                             state["_first_sent"] = True
-                        state["future"].set_result(
-                            [int(first[slot])])
+                        state["future"].set_result([token])
                         slots[slot] = None
             active = [i for i in range(B) if slots[i] is not None]
             if not active:
@@ -1132,15 +1145,28 @@ class LlamaServer:
 
     def _generate_on(self, engine: LlamaDecodeEngine, prompts: list,
                      max_new: int) -> list:
+        """Batch generation with per-LENGTH prefill groups: padding a
+        mixed-length batch would let pad tokens be attended, making a
+        prompt's output depend on its co-batch (see the continuous
+        scheduler property test)."""
         self.engine_calls += 1
         n = len(prompts)
-        max_len = max(len(p) for p in prompts)
-        tokens = torch.zeros(self.batch_size, max_len, dtype=torch.int64)
+        by_len: dict = {}
         for i, prompt in enumerate(prompts):
-            tokens[i, max_len - len(prompt):] = torch.tensor(
-                prompt, dtype=torch.int64)  # left-pad
+            by_len.setdefault(len(prompt), []).append(i)
         engine.reset()
-        out = engine.generate(tokens, max_new_tokens=max_new)
+        for length, idxs in by_len.items():
+            tokens = torch.tensor([prompts[i] for i in idxs],
+                                  dtype=torch.int64)
+            engine.prefill_slots(tokens, torch.tensor(idxs,
+                                                      dtype=torch.long))
+        if engine.use_graph and engine._graph is None:
+            engine.capture_graph()
+        generated = [engine.buf_tokens.clone()]
+        for _ in range(max_new - 1):
+            engine.decode_step()
+            generated.append(engine.buf_tokens.clone())
+        out = torch.stack(generated, dim=1)
         rows = out[:n].cpu().tolist()
         if self.stop_token is not None:
             trimmed = []

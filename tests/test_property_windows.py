@@ -106,3 +106,55 @@ class TestDriftMetricProperties:
                             abs_tol=1e-9)
         assert math.isclose(hell, hellinger_distance(q, p),
                             abs_tol=1e-7)
+
+
+class TestContinuousSchedulerProperty:
+    @settings(max_examples=10, deadline=None)
+    @given(st.lists(
+        st.tuples(st.lists(st.integers(min_value=1, max_value=900),
+                           min_size=1, max_size=12),
+                  st.integers(min_value=1, max_value=7)),
+        min_size=1, max_size=9))
+    def test_continuous_matches_batch_for_any_workload(self, requests):
+        """For ANY set of (prompt, max_new) requests, token-level
+        continuous scheduling produces exactly the greedy outputs of
+        isolated batch generation."""
+        from mlrun_amd.models.llama import LlamaConfig, LlamaServer
+
+        cfg = LlamaConfig.tiny()
+        if not hasattr(self, "_servers"):
+            cont = LlamaServer(name="pc", config=cfg, batch_size=3,
+                               max_new_tokens=8,
+                               scheduling="continuous",
+                               use_graph=False)
+            cont.load()
+            batch = LlamaServer(name="pb", config=cfg, batch_size=3,
+                                max_new_tokens=8, use_graph=False)
+            batch.load()
+            batch.engines[0].weights.load_state_dict(
+                cont.engines[0].weights.state_dict())
+            self.__class__._servers = (cont, batch)
+        cont, batch = self._servers
+
+        class _Ev:
+            path = "/infer"
+            id = "t"
+
+        import threading
+
+        results = {}
+
+        def one(i, prompt, max_new):
+            ev = _Ev()
+            ev.body = {"inputs": [prompt], "max_tokens": max_new}
+            results[i] = cont.do_event(ev).body["outputs"][0]
+
+        threads = [threading.Thread(target=one, args=(i, p, m))
+                   for i, (p, m) in enumerate(requests)]
+        [t.start() for t in threads]
+        [t.join(timeout=60) for t in threads]
+        for i, (prompt, max_new) in enumerate(requests):
+            ev = _Ev()
+            ev.body = {"inputs": [prompt], "max_tokens": max_new}
+            expect = batch.do_event(ev).body["outputs"][0]
+            assert results[i] == expect, (i, prompt, max_new)
