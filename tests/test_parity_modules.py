@@ -347,3 +347,39 @@ class TestHubBuiltins:
         assert fn.kind == "serving"
         graph = fn.spec.graph
         assert "llm" in graph.routes
+
+
+class TestSecretsSources:
+    def test_inline_env_file_layering(self, tmp_path, monkeypatch):
+        from mlrun_amd.secrets import SecretsStore
+
+        store = SecretsStore()
+        store.add_source("inline", {"A": "1"})
+        monkeypatch.setenv("FROM_ENV", "2")
+        store.add_source("env", "FROM_ENV")
+        secrets_file = tmp_path / "secrets.env"
+        secrets_file.write_text("B=3\n# comment=skip\nC=4\n")
+        store.add_source("file", str(secrets_file))
+        assert store.get("A") == "1"
+        assert store.get("FROM_ENV") == "2"
+        assert store.get("B") == "3" and store.get("C") == "4"
+        monkeypatch.setenv("MLRUN_SECRET_D", "5")
+        assert store.get("D") == "5"  # env-prefix fallback
+        assert store.get("missing", "dflt") == "dflt"
+
+    def test_unavailable_providers_raise(self):
+        import pytest as _pytest
+
+        from mlrun_amd.errors import MLRunInvalidArgumentError
+        from mlrun_amd.secrets import SecretsStore
+
+        for kind in ("vault", "azure_vault", "kubernetes", "bogus"):
+            with _pytest.raises(MLRunInvalidArgumentError):
+                SecretsStore().add_source(kind, {})
+
+    def test_get_secret_or_env(self, monkeypatch):
+        from mlrun_amd.secrets import get_secret_or_env
+
+        monkeypatch.setenv("SOME_TOKEN", "tok")
+        assert get_secret_or_env("SOME_TOKEN") == "tok"
+        assert get_secret_or_env("NOPE_X", default="d") == "d"
