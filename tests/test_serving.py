@@ -533,3 +533,4 @@ class TestWorkerPool:
             assert all(o == outs[0] for o in outs)
         finally:
             fn.stop()
+
