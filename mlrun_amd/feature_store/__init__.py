@@ -22,6 +22,7 @@ from .api import (  # noqa: F401
     get_online_feature_service,
     ingest,
     preview,
+    run_ingestion_job,
 )
 from .steps import (  # noqa: F401
     DateExtractor,

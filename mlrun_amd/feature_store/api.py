@@ -214,6 +214,36 @@ class IngestionService:
         self.drain_once()  # final drain
 
 
+def run_ingestion_job(featureset, source_path: str, name: str = "",
+                      schedule: str = None, run_config=None):
+    """Batch-ingest as a (schedulable) JOB run (reference api.py
+    run_ingestion_job): materializes a wrapper function that calls
+    ``ingest`` on the stored feature set, then runs it now or stores
+    a cron schedule for the service scheduler."""
+    import mlrun_amd
+
+    fset = featureset if isinstance(featureset, FeatureSet) else \
+        _resolve_feature_set(featureset)
+    fset.save()
+    project = fset.metadata.project or "default"
+    name = name or f"{fset.metadata.name}-ingest"
+    code = (
+        "from mlrun_amd import feature_store as fstore\n\n\n"
+        "def handler(context, featureset_uri: str = "
+        f"{project + '/' + fset.metadata.name!r}, "
+        f"source_path: str = {source_path!r}):\n"
+        "    import pandas as pd\n\n"
+        "    read = pd.read_parquet if source_path.endswith("
+        "'.parquet') else pd.read_csv\n"
+        "    df = fstore.ingest(featureset_uri, read(source_path))\n"
+        "    context.log_result('rows', int(len(df)))\n")
+    fn = mlrun_amd.new_function(name=name, kind="job",
+                                project=project)
+    fn.with_code(body=code)
+    return fn.run(handler="handler", name=name, schedule=schedule,
+                  local=schedule is None, watch=schedule is None)
+
+
 def deploy_ingestion_service(featureset, source=None,
                              interval_seconds: float = 1.0, targets=None,
                              start: bool = True) -> IngestionService:

@@ -350,3 +350,29 @@ class TestRingGrowth:
         vals = ring.window_values(600, 30.0 + 60 * 10)
         assert vals["sum"][0].item() == 0.0
         assert vals["sum"][1].item() == 7.0
+
+
+class TestIngestionJob:
+    def test_run_now(self, rundb, tmp_path):
+        df = make_df(40)
+        src = tmp_path / "events.parquet"
+        df.to_parquet(src)
+        fset = fstore.FeatureSet("jobfs", entities=["customer"],
+                                 timestamp_key="ts")
+        run = fstore.run_ingestion_job(fset, str(src))
+        assert run.status.state == "completed"
+        assert run.outputs["rows"] == 40
+        table = fstore.get_online_table(fset)
+        assert table is not None
+
+    def test_scheduled_creates_cron_entry(self, rundb, tmp_path):
+        df = make_df(10)
+        src = tmp_path / "sched.parquet"
+        df.to_parquet(src)
+        fset = fstore.FeatureSet("schedfs", entities=["customer"],
+                                 timestamp_key="ts")
+        fstore.run_ingestion_job(fset, str(src),
+                                 schedule="*/30 * * * *")
+        schedules = rundb.list_schedules("default")
+        names = [s.get("name") for s in schedules]
+        assert any("schedfs-ingest" in str(n) for n in names)
