@@ -188,6 +188,27 @@ class BaseRuntime(ModelObj):
     def with_requests(self, mem=None, cpu=None):
         return self.with_limits(mem=mem, cpu=cpu)
 
+    def with_node_selection(self, node_name=None, node_selector=None,
+                            affinity=None, tolerations=None):
+        """Node placement (reference pod.py:1156).  Single-node
+        deployment: recorded as labels for parity; device placement
+        happens via the GPU allocator instead of k8s scheduling."""
+        if node_name:
+            self.set_label("node-name", node_name)
+        for key, value in (node_selector or {}).items():
+            self.set_label(f"node-selector/{key}", value)
+        return self
+
+    def with_preemption_mode(self, mode):
+        """Preemptible-node policy (reference pod.py:1207): recorded
+        only — there are no spot nodes in the node-local deployment."""
+        self.set_label("preemption-mode", str(mode))
+        return self
+
+    def with_priority_class(self, name=""):
+        self.set_label("priority-class", name)
+        return self
+
     def set_env(self, name, value):
         self.spec.build.setdefault("env", {})[name] = str(value)
         return self
