@@ -61,6 +61,11 @@ def process_event(project: str, event_kind: str, event: dict,
             continue
         fired.append(name)
         _push_alert_notifications(alert, event_kind, event)
+        if (alert.get("reset_policy") or "auto") == "auto" and \
+                hasattr(db, "reset_alert_state"):
+            # auto: firing rearms the alert (another `count` events
+            # must arrive before it fires again)
+            db.reset_alert_state(project, name)
     return fired
 
 

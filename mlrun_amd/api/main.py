@@ -334,6 +334,11 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
                            for t, stats in series],
                 "current": processor.endpoint_stats(endpoint_id)}
 
+    @app.post("/api/v1/projects/{project}/alerts/{name}/reset")
+    async def reset_alert(project: str, name: str):
+        db.reset_alert_state(project, name)
+        return {}
+
     # ------------------------------------------------------------ hub
     @app.get("/api/v1/hub/sources")
     async def hub_sources():

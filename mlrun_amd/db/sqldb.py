@@ -780,6 +780,14 @@ class SQLRunDB(RunDBInterface):
             (project, name))
         return rows[0]["count"] if rows else 0
 
+    def reset_alert_state(self, project, name):
+        """Reset an alert's event counter (manual reset API, and the
+        auto reset applied after an alert fires)."""
+        self._execute(
+            "UPDATE alert_states SET count=0, active=0, last_updated=? "
+            "WHERE project=? AND name=?",
+            (now_iso(), project or "default", name))
+
     def get_alert_state(self, project, name):
         rows = self._query(
             "SELECT count, last_updated, active FROM alert_states "

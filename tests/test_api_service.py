@@ -422,3 +422,27 @@ def test_httpdb_new_surfaces_roundtrip(client, tmp_path, monkeypatch):
     assert "llm-serving" in items
     assert db.list_pipelines("hp") == []
     assert db.list_workflows("hp") == []
+
+
+def test_alert_reset_policies(client):
+    db = client.db
+    db.store_alert_config("rp", "auto-a", {
+        "name": "auto-a", "trigger": {"events": ["e"]},
+        "criteria": {"count": 2}, "reset_policy": "auto",
+        "notifications": [{"kind": "console"}]})
+    db.store_alert_config("rp", "manual-a", {
+        "name": "manual-a", "trigger": {"events": ["e"]},
+        "criteria": {"count": 2}, "reset_policy": "manual",
+        "notifications": [{"kind": "console"}]})
+    from mlrun_amd.api.events import process_event
+
+    assert process_event("rp", "e", {}, db=db) == []
+    assert sorted(process_event("rp", "e", {}, db=db)) == \
+        ["auto-a", "manual-a"]
+    # auto rearmed (count reset); manual stays above threshold
+    assert process_event("rp", "e", {}, db=db) == ["manual-a"]
+    resp = client.post("/api/v1/projects/rp/alerts/manual-a/reset")
+    assert resp.status_code == 200
+    # 4th event: auto-a reaches its threshold again (1 -> 2); the
+    # freshly reset manual-a is back at 1
+    assert process_event("rp", "e", {}, db=db) == ["auto-a"]
