@@ -531,7 +531,13 @@ class FlowStep(BaseStep):
         from ..utils import tracing
 
         trace = tracing.is_enabled()
+        hops = 0
         while step is not None:
+            hops += 1
+            if hops > 10000:  # cycle guard: a mis-wired graph must
+                raise GraphError(  # error, not hang the server
+                    f"flow exceeded 10000 hops at step "
+                    f"{step.fullname} — graph cycle?")
             try:
                 if trace:
                     with tracing.trace_step(step.fullname):

@@ -534,3 +534,27 @@ class TestWorkerPool:
         finally:
             fn.stop()
 
+
+
+class TestCycleGuard:
+    def test_cyclic_graph_errors_instead_of_hanging(self):
+        import mlrun_amd
+        from mlrun_amd.serving.states import RootFlowStep
+
+        class Pass:
+            def __init__(self, context=None, name=None):
+                pass
+
+            def do_event(self, event):
+                return event
+
+        fn = mlrun_amd.new_function(name="cycfn", kind="serving")
+        graph = fn.set_topology("flow", engine="sync")
+        graph.add_step(Pass, name="a")
+        graph.add_step(Pass, name="b", after="a")
+        server = fn.to_mock_server(namespace={"Pass": Pass})
+        # wire the cycle after build: b -> a
+        flow = server.graph
+        flow.steps["b"]._next = ["a"]
+        out = server.test("/x", body={})
+        assert "cycle" in str(out).lower() or "error" in str(out).lower()
