@@ -574,52 +574,54 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     return app
 
 
-async def _runs_monitor(db: SQLRunDB):
-    """Periodic run monitoring: abort runs stuck past state thresholds
-    (parity: reference main.py:608 + runtime_handlers/base.py:1387)."""
+def check_stuck_runs(db: SQLRunDB, now=None) -> list:
+    """One monitoring sweep: abort runs stuck past the configured
+    state thresholds (parity: reference runtime_handlers/base.py:1387
+    state-threshold logic).  Returns the aborted run uids."""
     import datetime
 
+    from ..utils import parse_time
+
+    thresholds = config.runs.state_thresholds.to_dict()
+    now = now or datetime.datetime.now(datetime.timezone.utc)
+    aborted = []
+    projects = [p.get("metadata", {}).get("name", "default")
+                for p in db.list_projects()] or ["default"]
+    for project in projects:
+        for state in ("running", "pending"):
+            limit = thresholds.get(state)
+            if not limit:
+                continue
+            for run in db.list_runs(project=project, state=state):
+                started = run.get("status", {}).get("start_time")
+                if not started:
+                    continue
+                start_dt = parse_time(started)
+                if start_dt is None:
+                    continue
+                if start_dt.tzinfo is None:
+                    start_dt = start_dt.replace(
+                        tzinfo=datetime.timezone.utc)
+                age = (now - start_dt).total_seconds()
+                if age > float(limit):
+                    uid = run.get("metadata", {}).get("uid")
+                    logger.warning("aborting run past state threshold",
+                                   uid=uid, state=state, age=age)
+                    db.abort_run(
+                        uid, project,
+                        status_text=f"aborted by monitor: {state} "
+                                    f"for {int(age)}s > {limit}s")
+                    aborted.append(uid)
+    return aborted
+
+
+async def _runs_monitor(db: SQLRunDB):
+    """Periodic run monitoring loop (reference main.py:608)."""
     interval = int(config.runs.monitoring_interval)
     while True:
         try:
             await asyncio.sleep(interval)
-            thresholds = config.runs.state_thresholds.to_dict()
-            now = datetime.datetime.now(datetime.timezone.utc)
-            for state, limit in (("running", thresholds.get("running")),
-                                 ("pending", thresholds.get("pending"))):
-                for run in db.list_runs(project="", state=state):
-                    pass  # project-scoped listing below
-            for project in [p.get("metadata", {}).get("name", "default")
-                            for p in db.list_projects()] or ["default"]:
-                for state in ("running", "pending"):
-                    limit = thresholds.get(state)
-                    if not limit:
-                        continue
-                    for run in db.list_runs(project=project, state=state):
-                        started = run.get("status", {}).get("start_time")
-                        if not started:
-                            continue
-                        try:
-                            from ..utils import parse_time
-
-                            start_dt = parse_time(started)
-                            if start_dt is None:
-                                continue
-                            if start_dt.tzinfo is None:
-                                start_dt = start_dt.replace(
-                                    tzinfo=datetime.timezone.utc)
-                            age = (now - start_dt).total_seconds()
-                        except Exception:
-                            continue
-                        if age > float(limit):
-                            uid = run.get("metadata", {}).get("uid")
-                            logger.warning(
-                                "aborting run past state threshold",
-                                uid=uid, state=state, age=age)
-                            db.abort_run(
-                                uid, project,
-                                status_text=f"aborted by monitor: {state} "
-                                            f"for {int(age)}s > {limit}s")
+            check_stuck_runs(db)
         except asyncio.CancelledError:
             return
         except Exception as exc:

@@ -446,3 +446,25 @@ def test_alert_reset_policies(client):
     # 4th event: auto-a reaches its threshold again (1 -> 2); the
     # freshly reset manual-a is back at 1
     assert process_event("rp", "e", {}, db=db) == ["auto-a"]
+
+
+def test_runs_monitor_aborts_stuck_runs(client, monkeypatch):
+    import datetime
+
+    from mlrun_amd.api.main import check_stuck_runs
+    from mlrun_amd.config import config as cfg
+
+    db = client.db
+    old = (datetime.datetime.now(datetime.timezone.utc) -
+           datetime.timedelta(hours=30)).isoformat()
+    fresh = datetime.datetime.now(datetime.timezone.utc).isoformat()
+    db.store_run({"metadata": {"name": "stuck", "uid": "s1"},
+                  "status": {"state": "running", "start_time": old}},
+                 "s1", "default")
+    db.store_run({"metadata": {"name": "live", "uid": "s2"},
+                  "status": {"state": "running", "start_time": fresh}},
+                 "s2", "default")
+    aborted = check_stuck_runs(db)
+    assert aborted == ["s1"]
+    assert db.read_run("s1", "default")["status"]["state"] == "aborted"
+    assert db.read_run("s2", "default")["status"]["state"] == "running"
