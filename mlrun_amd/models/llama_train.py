@@ -211,6 +211,10 @@ class LlamaTrainer:
         model = self.ddp or self.model
         if self._accum == 0:
             self.optimizer.zero_grad(set_to_none=True)
+        if self.ddp is not None:
+            # reduce only on the accumulation boundary (no_sync analog)
+            self.ddp.require_backward_grad_sync = \
+                self._accum + 1 >= self.grad_accum_steps
         _, loss = model(tokens, labels=tokens)
         (loss / self.grad_accum_steps).backward()
         self._accum += 1

@@ -96,6 +96,11 @@ class DistributedModel(torch.nn.Module):
             next(module.parameters()).is_cuda
         if self._use_comm_stream:
             self._comm_stream = torch.cuda.Stream()
+        # grad-accumulation contract (torch-DDP no_sync analog):
+        # reduction fires only when require_backward_grad_sync is True
+        # — set it False for non-boundary micro-steps, True for the
+        # final micro-step of the accumulation window
+        self.require_backward_grad_sync = True
         if self.world_size > 1:
             broadcast_module(module)
             self._build_buckets()
@@ -126,6 +131,8 @@ class DistributedModel(torch.nn.Module):
             self._hooks.append(hook)
 
     def _on_grad_ready(self, param):
+        if not self.require_backward_grad_sync:
+            return  # accumulating: reduce on the boundary step only
         bucket = self._param_bucket[param]
         bucket.ready += 1
         if bucket.ready == len(bucket.params):

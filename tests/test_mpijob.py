@@ -40,6 +40,48 @@ TRAIN_SCRIPT = textwrap.dedent("""
     dist.destroy_process_group()
 """)
 
+ACCUM_SCRIPT = textwrap.dedent("""
+    import os
+    import torch
+    import torch.distributed as dist
+    import mlrun_amd
+    from mlrun_amd.parallel.ddp import DistributedModel, init_process_group
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    rank, world = init_process_group(backend="gloo")
+    torch.manual_seed(5)
+    model = torch.nn.Linear(8, 1, bias=False)
+    reference = torch.nn.Linear(8, 1, bias=False)
+    reference.load_state_dict(model.state_dict())
+    ddp = DistributedModel(model, bucket_cap_mb=1)
+
+    # two micro-batches per rank; DDP must reduce ONLY at the boundary
+    torch.manual_seed(200 + rank)
+    micro = [(torch.randn(4, 8), torch.randn(4, 1)) for _ in range(2)]
+    loss_fn = torch.nn.MSELoss()
+
+    for i, (x, y) in enumerate(micro):
+        ddp.require_backward_grad_sync = (i == 1)
+        loss = loss_fn(ddp(x), y) / 2
+        loss.backward()
+    ddp.finalize_backward()
+
+    # expected grad = mean over ranks of the per-rank ACCUMULATED grad
+    for x, y in micro:
+        (loss_fn(reference(x), y) / 2).backward()
+    local = reference.weight.grad.reshape(-1)
+    gathered = [torch.zeros_like(local) for _ in range(world)]
+    dist.all_gather(gathered, local)
+    expect = torch.stack(gathered).mean(0)
+    got = model.weight.grad.reshape(-1)
+    assert torch.allclose(got, expect, atol=1e-6), (got, expect)
+    if rank == 0:
+        ctx = mlrun_amd.get_or_create_ctx("accum")
+        ctx.log_result("grad_err", float((got - expect).abs().max()))
+        ctx.commit(completed=True)
+    dist.destroy_process_group()
+""")
+
 DDP_SCRIPT = textwrap.dedent("""
     import os
     import torch
@@ -201,3 +243,17 @@ class TestGpuAllocatorContention:
         [t.join(timeout=60) for t in threads]
         assert not errors
         assert allocator.available() == [0, 1, 2, 3]
+
+
+    def test_ddp_grad_accumulation_no_sync(self, tmp_path):
+        """Accumulated micro-steps must reduce once, at the boundary
+        (regression: hooks fired every micro-step and reduced stale
+        partial gradients)."""
+        script = tmp_path / "accum.py"
+        script.write_text(ACCUM_SCRIPT)
+        fn = mlrun_amd.new_function(name="accum", kind="mpijob",
+                                    command=str(script))
+        fn.with_replicas(2)
+        run = fn.run(local=True)
+        assert run.status.state == "completed"
+        assert run.status.results["grad_err"] < 1e-6
