@@ -560,3 +560,29 @@ class TestCycleGuard:
 
         with _pytest.raises(RuntimeError, match="cycle"):
             server.test("/x", body={})
+
+
+class TestWorkerPoolStreaming:
+    def test_ndjson_streams_through_the_l4_proxy(self):
+        import json
+
+        import requests
+
+        import mlrun_amd
+
+        fn = mlrun_amd.new_function(name="pool-stream", kind="serving")
+        fn.add_model("llm",
+                     class_name="mlrun_amd.models.llama.LlamaServer",
+                     config="tiny", batch_size=2, max_new_tokens=6,
+                     scheduling="continuous")
+        addr = fn.deploy(workers=2)
+        try:
+            with requests.post(addr + "/v2/models/llm/infer",
+                               json={"inputs": [[1, 2, 3]],
+                                     "max_tokens": 4, "stream": True},
+                               stream=True, timeout=120) as resp:
+                toks = [json.loads(line)["token"]
+                        for line in resp.iter_lines() if line]
+            assert len(toks) == 4
+        finally:
+            fn.stop()
