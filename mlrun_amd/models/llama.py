@@ -260,12 +260,7 @@ class LlamaDecodeEngine:
                                       device=self.device)
             self.buf_h8_scale = torch.empty(B, dtype=torch.float32,
                                             device=self.device)
-            for layer in w.layers:
-                for key in ("wqkv", "wo", "wgu", "wdown"):
-                    self._fp8_packs[id(layer[key])] = \
-                        ops.quantize_fp8_weight(layer[key])
-            self._fp8_packs[id(w.lm_head)] = \
-                ops.quantize_fp8_weight(w.lm_head)
+            self.rebuild_fp8_packs()
         # fill-based (ns4 at B=32): the graph bakes ONE nsplit for all
         # context lengths, and decode mostly runs at S << max_seq_len —
         # measured (scripts/bench_attn_longctx.py): chunk-aware ns16
@@ -524,6 +519,22 @@ class LlamaDecodeEngine:
             self._decode_step_body()
 
     # --------------------------------------------------------- prefill
+    def rebuild_fp8_packs(self):
+        """(Re)quantize the fp8 weight packs from the CURRENT weights.
+        Must be called again after load_state_dict — otherwise serving
+        would run on packs quantized from the discarded init weights."""
+        if self.weight_dtype != "fp8w" or not (self.on_gpu and
+                                               self.tp_size == 1):
+            return
+        w = self.weights
+        self._fp8_packs = {}
+        for layer in w.layers:
+            for key in ("wqkv", "wo", "wgu", "wdown"):
+                self._fp8_packs[id(layer[key])] = \
+                    ops.quantize_fp8_weight(layer[key])
+        self._fp8_packs[id(w.lm_head)] = \
+            ops.quantize_fp8_weight(w.lm_head)
+
     def _kscale(self, li):
         return None if self.k_scale is None else self.k_scale[li]
 
@@ -814,7 +825,10 @@ class LlamaServer:
             state = torch.load(model_file, map_location=first.device,
                                weights_only=True)
             first.weights.load_state_dict(state)
-            self.model_spec = spec
+            first.rebuild_fp8_packs()  # packs quantized pre-load are
+            self.model_spec = spec     # stale (fp8w + model_path)
+            for replica in self.engines[1:]:
+                replica._fp8_packs = first._fp8_packs
         for idx, engine in enumerate(self.engines):
             engine._serve_stream = torch.cuda.Stream() \
                 if engine.on_gpu else None

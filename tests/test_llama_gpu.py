@@ -111,16 +111,7 @@ class TestFP8Serving:
                                 use_graph=False, seed=1,
                                 weight_dtype="fp8w")
         fp8.weights.load_state_dict(bf16.weights.state_dict())
-        # re-quantize from the loaded weights
-        from mlrun_amd import ops as _ops
-
-        fp8._fp8_packs = {}
-        for layer in fp8.weights.layers:
-            for key in ("wqkv", "wo", "wgu", "wdown"):
-                fp8._fp8_packs[id(layer[key])] = _ops.quantize_fp8_weight(
-                    layer[key])
-        fp8._fp8_packs[id(fp8.weights.lm_head)] = _ops.quantize_fp8_weight(
-            fp8.weights.lm_head)
+        fp8.rebuild_fp8_packs()  # re-quantize from the loaded weights
         out_fp8 = fp8.generate(prompt, max_new_tokens=4).cpu()
         match = (out_bf16 == out_fp8).float().mean().item()
         assert match >= 0.5, f"fp8 decode diverged: match={match}"
