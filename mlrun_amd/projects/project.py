@@ -335,6 +335,24 @@ class MlrunProject(ModelObj):
         return filepath
 
     def export(self, filepath=None):
+        """Export the project: yaml spec, or a .zip/.tar.gz archive of
+        the whole context directory (reference project.export)."""
+        if filepath and str(filepath).endswith((".zip", ".tar.gz",
+                                                ".tgz")):
+            import shutil
+            import tempfile
+
+            context = self.context or "."
+            self.save(os.path.join(context, "project.yaml"),
+                      store=False)
+            base, fmt = (filepath[:-4], "zip") \
+                if filepath.endswith(".zip") else \
+                (filepath[:-7] if filepath.endswith(".tar.gz")
+                 else filepath[:-4], "gztar")
+            archive = shutil.make_archive(base, fmt, root_dir=context)
+            if archive != filepath:
+                shutil.move(archive, filepath)
+            return filepath
         return self.save(filepath, store=False)
 
     def set_secrets(self, secrets: dict = None, file_path=None, provider=None):
@@ -401,6 +419,14 @@ def load_project(context="./", url=None, name=None, secrets=None,
     import yaml
 
     path = url or os.path.join(context, "project.yaml")
+    if url and str(url).endswith((".zip", ".tar.gz", ".tgz")) and \
+            os.path.isfile(url):
+        # archive export -> unpack into the context dir and load
+        import shutil
+
+        os.makedirs(context, exist_ok=True)
+        shutil.unpack_archive(url, context)
+        path = os.path.join(context, "project.yaml")
     if os.path.isfile(path):
         with open(path) as fp:
             struct = yaml.safe_load(fp)

@@ -260,3 +260,29 @@ class TestNotebookToFunction:
         assert "skipped" not in code and "%matplotlib" not in code
         run = fn.run(handler="handler", params={"x": 5}, local=True)
         assert run.outputs["sq"] == 25
+
+
+class TestProjectArchive:
+    def test_export_zip_and_load(self, rundb, tmp_path):
+        import os
+
+        import mlrun_amd
+
+        src_ctx = tmp_path / "src"
+        src_ctx.mkdir()
+        (src_ctx / "fn.py").write_text(
+            "def handler(context):\n    context.log_result('v', 9)\n")
+        project = mlrun_amd.new_project("arch", context=str(src_ctx))
+        project.set_function(str(src_ctx / "fn.py"), name="f1",
+                             kind="job")
+        archive = str(tmp_path / "proj.zip")
+        out = project.export(archive)
+        assert out == archive and os.path.isfile(archive)
+
+        dst_ctx = str(tmp_path / "dst")
+        loaded = mlrun_amd.load_project(context=dst_ctx, url=archive,
+                                        save=False)
+        assert loaded.name == "arch"
+        assert os.path.isfile(os.path.join(dst_ctx, "fn.py"))
+        fn = loaded.get_function("f1")
+        assert fn is not None
