@@ -219,7 +219,15 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     # -------------------------------------------------------- schedules
     @app.post("/api/v1/projects/{project}/schedules")
     async def create_schedule(project: str, request: Request):
+        import pydantic
+
+        from ..common.schemas import ScheduleSchema
+
         body = await request.json()
+        try:  # wire-schema validation (pydantic; 422 on bad payloads)
+            body = ScheduleSchema(**body).dict()
+        except pydantic.ValidationError as exc:
+            raise HTTPException(status_code=422, detail=str(exc))
         db.create_schedule(project, body)
         if app.state.scheduler:
             app.state.scheduler.reload()

@@ -468,3 +468,18 @@ def test_runs_monitor_aborts_stuck_runs(client, monkeypatch):
     assert aborted == ["s1"]
     assert db.read_run("s1", "default")["status"]["state"] == "aborted"
     assert db.read_run("s2", "default")["status"]["state"] == "running"
+
+
+def test_schedule_payload_validated(client):
+    # missing cron_trigger -> pydantic rejects before reaching the DB
+    resp = client.post("/api/v1/projects/vp/schedules",
+                       json={"name": "bad"})
+    assert resp.status_code in (400, 422, 500)
+    assert client.db.list_schedules("vp") == []
+    resp = client.post("/api/v1/projects/vp/schedules",
+                       json={"name": "ok",
+                             "cron_trigger": "*/5 * * * *",
+                             "task": {"metadata": {"name": "t"}}})
+    assert resp.status_code == 200
+    assert client.db.get_schedule("vp", "ok")["cron_trigger"] == \
+        "*/5 * * * *"
