@@ -225,7 +225,9 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
 
         body = await request.json()
         try:  # wire-schema validation (pydantic; 422 on bad payloads)
-            body = ScheduleSchema(**body).dict()
+            schema = ScheduleSchema(**body)
+            body = schema.model_dump() \
+                if hasattr(schema, "model_dump") else schema.dict()
         except pydantic.ValidationError as exc:
             raise HTTPException(status_code=422, detail=str(exc))
         db.create_schedule(project, body)
