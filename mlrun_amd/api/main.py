@@ -29,7 +29,24 @@ from .scheduler import Scheduler
 
 def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     db = db or SQLRunDB(config.httpdb.dsn or "")
-    app = FastAPI(title="mlrun-amd-api", version="v1")
+    import contextlib
+
+    @contextlib.asynccontextmanager
+    async def lifespan(app_):
+        if app_.state.scheduler:
+            app_.state.scheduler.start()
+        app_.state.monitor_task = asyncio.create_task(_runs_monitor(db))
+        try:
+            yield
+        finally:
+            if app_.state.scheduler:
+                app_.state.scheduler.stop()
+            task = getattr(app_.state, "monitor_task", None)
+            if task:
+                task.cancel()
+
+    app = FastAPI(title="mlrun-amd-api", version="v1",
+                  lifespan=lifespan)
     app.state.db = db
     app.state.scheduler = Scheduler(db) if with_scheduler else None
 
@@ -37,20 +54,6 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     async def mlrun_error_handler(request: Request, exc: MLRunBaseError):
         return JSONResponse(status_code=err_to_status(exc),
                             content={"detail": str(exc)})
-
-    @app.on_event("startup")
-    async def startup():
-        if app.state.scheduler:
-            app.state.scheduler.start()
-        app.state.monitor_task = asyncio.create_task(_runs_monitor(db))
-
-    @app.on_event("shutdown")
-    async def shutdown():
-        if app.state.scheduler:
-            app.state.scheduler.stop()
-        task = getattr(app.state, "monitor_task", None)
-        if task:
-            task.cancel()
 
     # ------------------------------------------------------------ misc
     @app.get("/api/v1/healthz")
