@@ -586,3 +586,23 @@ class TestWorkerPoolStreaming:
             assert len(toks) == 4
         finally:
             fn.stop()
+
+
+class TestV2ModelListing:
+    def test_list_models_endpoint(self):
+        import mlrun_amd
+
+        fn = mlrun_amd.new_function(name="lst", kind="serving")
+        fn.add_model("m1",
+                     class_name="mlrun_amd.frameworks.tree."
+                                "TreeEnsembleModelServer",
+                     n_trees=3, depth=2, n_features=4)
+        fn.add_model("m2",
+                     class_name="mlrun_amd.frameworks.tree."
+                                "TreeEnsembleModelServer",
+                     n_trees=3, depth=2, n_features=4)
+        server = fn.to_mock_server()
+        out = server.test("/v2/models/", body=None, method="GET")
+        assert sorted(out["models"]) == ["m1", "m2"]
+        meta = server.test("/v2/models/m1", body=None, method="GET")
+        assert meta["name"] == "m1"
