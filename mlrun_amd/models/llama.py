@@ -850,9 +850,22 @@ class LlamaServer:
             worker.start()
             self._workers.append(worker)
 
+    def shutdown(self):
+        """Stop the worker/batcher threads (one sentinel per engine).
+        In-flight requests finish first; daemon threads would
+        otherwise live until process exit (test hygiene)."""
+        for _ in self.engines:
+            self._tasks.put(None)
+        for worker in self._workers:
+            worker.join(timeout=10)
+        self._workers = []
+
     def _worker_loop(self, engine: LlamaDecodeEngine):
         while True:
-            prompts, max_new, future = self._tasks.get()
+            item = self._tasks.get()
+            if item is None:
+                return
+            prompts, max_new, future = item
             try:
                 if engine._serve_stream is not None:
                     with torch.cuda.stream(engine._serve_stream):
@@ -931,12 +944,18 @@ class LlamaServer:
             taken = []
             if free:
                 try:
-                    taken.append(self._tasks.get(block=not have_active))
+                    item = self._tasks.get(block=not have_active)
+                    if item is None:
+                        return  # shutdown sentinel
+                    taken.append(item)
                 except queue_mod.Empty:
                     pass
                 while len(taken) < len(free):
                     try:
-                        taken.append(self._tasks.get_nowait())
+                        item = self._tasks.get_nowait()
+                        if item is None:
+                            return
+                        taken.append(item)
                     except queue_mod.Empty:
                         break
             if taken:

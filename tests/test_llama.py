@@ -491,3 +491,32 @@ class TestServingModeInterplay:
         assert results["plain"] == expect  # stop token honored
         # the stream also ends at the stop token
         assert results["stream"][:len(expect)] == expect
+
+
+class TestServerShutdown:
+    def test_shutdown_stops_workers_after_inflight(self):
+        cfg = LlamaConfig.tiny()
+        srv = LlamaServer(name="sd", config=cfg, batch_size=2,
+                          max_new_tokens=4, use_graph=False)
+        srv.load()
+
+        class _Ev:
+            body = {"inputs": [[1, 2]], "max_tokens": 3}
+            path = "/infer"
+            id = "t"
+
+        out = srv.do_event(_Ev()).body["outputs"]
+        assert len(out[0]) == 3
+        workers = list(srv._workers)
+        srv.shutdown()
+        assert all(not w.is_alive() for w in workers)
+
+    def test_shutdown_continuous(self):
+        cfg = LlamaConfig.tiny()
+        srv = LlamaServer(name="sdc", config=cfg, batch_size=2,
+                          max_new_tokens=4, use_graph=False,
+                          scheduling="continuous")
+        srv.load()
+        workers = list(srv._workers)
+        srv.shutdown()
+        assert all(not w.is_alive() for w in workers)
