@@ -854,11 +854,17 @@ class LlamaServer:
         """Stop the worker/batcher threads (one sentinel per engine).
         In-flight requests finish first; daemon threads would
         otherwise live until process exit (test hygiene)."""
+        self._stopping = True
+        with self._pending_cv:
+            self._pending_cv.notify_all()
         for _ in self.engines:
             self._tasks.put(None)
         for worker in self._workers:
             worker.join(timeout=10)
         self._workers = []
+        if self._batcher is not None:
+            self._batcher.join(timeout=10)
+            self._batcher = None
 
     def _worker_loop(self, engine: LlamaDecodeEngine):
         while True:
@@ -1152,10 +1158,12 @@ class LlamaServer:
         import concurrent.futures
         import time as _time
 
-        while True:
+        while not getattr(self, "_stopping", False):
             with self._pending_cv:
                 while not self._pending:
                     self._pending_cv.wait()
+                    if getattr(self, "_stopping", False):
+                        return
             _time.sleep(self.batch_window_ms / 1000.0)
             with self._pending_cv:
                 batch, self._pending = \
