@@ -88,3 +88,50 @@ class TestCLI:
         iters = runs[0]["status"].get("iterations") or []
         results = runs[0]["status"].get("results", {})
         assert results.get("best_iteration") or iters
+
+    def test_db_service_boots_over_http(self, tmp_path):
+        """`mlrun_amd db` boots the real uvicorn service; check
+        healthz + one CRUD roundtrip over actual HTTP."""
+        import os
+        import socket
+        import subprocess
+        import sys
+        import time
+
+        import requests
+
+        with socket.socket() as sock:
+            sock.bind(("127.0.0.1", 0))
+            port = sock.getsockname()[1]
+        env = dict(os.environ)
+        env["MLRUN_BASE_DIR"] = str(tmp_path)
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(
+            __file__)))
+        env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "mlrun_amd", "db", "--port",
+             str(port)], env=env)
+        base = f"http://127.0.0.1:{port}"
+        try:
+            deadline = time.time() + 60
+            while time.time() < deadline:
+                try:
+                    resp = requests.get(base + "/api/v1/healthz",
+                                        timeout=2)
+                    if resp.status_code == 200:
+                        break
+                except Exception:
+                    time.sleep(0.3)
+            else:
+                raise TimeoutError("service did not become healthy")
+            run = {"metadata": {"name": "httprun", "uid": "h1"},
+                   "status": {"state": "completed"}}
+            assert requests.post(base + "/api/v1/run/default/h1",
+                                 json=run, timeout=10
+                                 ).status_code == 200
+            got = requests.get(base + "/api/v1/run/default/h1",
+                               timeout=10).json()["data"]
+            assert got["metadata"]["name"] == "httprun"
+        finally:
+            proc.terminate()
+            proc.wait(timeout=10)
