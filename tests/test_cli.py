@@ -135,3 +135,60 @@ class TestCLI:
         finally:
             proc.terminate()
             proc.wait(timeout=10)
+
+    def test_remote_submit_through_live_service(self, tmp_path):
+        """Reference stack 3.2: client fn.run(local=False) ->
+        HTTPRunDB.submit_job over real HTTP -> ServerSideLauncher
+        executes -> client reads the run back."""
+        import os
+        import socket
+        import subprocess
+        import sys
+        import time
+
+        import requests
+
+        with socket.socket() as sock:
+            sock.bind(("127.0.0.1", 0))
+            port = sock.getsockname()[1]
+        env = dict(os.environ)
+        env["MLRUN_BASE_DIR"] = str(tmp_path / "server")
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(
+            __file__)))
+        env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "mlrun_amd", "db", "--port",
+             str(port)], env=env)
+        base = f"http://127.0.0.1:{port}"
+        code = tmp_path / "remote_fn.py"
+        code.write_text("def handler(context, x=1):\n"
+                        "    context.log_result('doubled', x * 2)\n")
+        client_code = tmp_path / "client.py"
+        client_code.write_text(f"""
+import mlrun_amd
+from mlrun_amd.config import config
+config.dbpath = {base!r}
+fn = mlrun_amd.new_function(name="remote-job", kind="job",
+                            command={str(code)!r})
+run = fn.run(handler="handler", params={{"x": 21}}, local=False,
+             watch=True)
+assert run.status.state == "completed", run.status.state
+assert run.outputs["doubled"] == 42, run.outputs
+print("REMOTE_OK", run.metadata.uid)
+""")
+        try:
+            deadline = time.time() + 60
+            while time.time() < deadline:
+                try:
+                    if requests.get(base + "/api/v1/healthz",
+                                    timeout=2).status_code == 200:
+                        break
+                except Exception:
+                    time.sleep(0.3)
+            out = subprocess.run(
+                [sys.executable, str(client_code)], env=env,
+                capture_output=True, text=True, timeout=120)
+            assert "REMOTE_OK" in out.stdout, out.stdout + out.stderr
+        finally:
+            proc.terminate()
+            proc.wait(timeout=10)
