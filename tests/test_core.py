@@ -272,3 +272,30 @@ class TestFsspecStore:
 
         for scheme in ("s3", "gcs", "az", "http", "hdfs"):
             assert schema_to_store(scheme) is FsspecStore
+
+
+class TestSQLRunDBConcurrency:
+    def test_concurrent_writers_and_readers(self, rundb):
+        import threading
+
+        errors = []
+
+        def writer(tid):
+            try:
+                for i in range(30):
+                    uid = f"c{tid}-{i}"
+                    rundb.store_run(
+                        {"metadata": {"name": f"r{tid}", "uid": uid},
+                         "status": {"state": "completed"}},
+                        uid, "default")
+                    rundb.read_run(uid, "default")
+            except Exception as exc:
+                errors.append(exc)
+
+        threads = [threading.Thread(target=writer, args=(t,))
+                   for t in range(8)]
+        [t.start() for t in threads]
+        [t.join(timeout=60) for t in threads]
+        assert not errors, errors[:2]
+        assert len(rundb.list_runs(project="default",
+                                   last=1000)) >= 240
