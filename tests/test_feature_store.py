@@ -376,3 +376,40 @@ class TestIngestionJob:
         schedules = rundb.list_schedules("default")
         names = [s.get("name") for s in schedules]
         assert any("schedfs-ingest" in str(n) for n in names)
+
+
+class TestOnlineConcurrency:
+    def test_gets_during_ingest(self, rundb):
+        """Online lookups racing batched ingests must never raise or
+        return malformed vectors (torch tensor reads are safe against
+        concurrent writes; values may be slightly stale)."""
+        import threading
+
+        fset = fstore.FeatureSet("conc", entities=["customer"],
+                                 timestamp_key="ts")
+        fset.add_aggregation("amount", ["sum", "count"], ["1h"], "10m")
+        fstore.ingest(fset, make_df(200))
+        service = fstore.get_online_feature_service(
+            fstore.FeatureVector("concv", ["conc.amount_sum_1h"]))
+        errors = []
+        stop = threading.Event()
+
+        def reader():
+            try:
+                while not stop.is_set():
+                    rows = service.get([{"customer": "c1"},
+                                        {"customer": "c2"}])
+                    assert isinstance(rows, list) and len(rows) == 2
+            except Exception as exc:
+                errors.append(exc)
+
+        threads = [threading.Thread(target=reader) for _ in range(3)]
+        [t.start() for t in threads]
+        try:
+            for _ in range(10):
+                fstore.ingest(fset, make_df(100))
+        finally:
+            stop.set()
+            [t.join(timeout=30) for t in threads]
+        assert not errors, errors[:1]
+        service.close()
