@@ -97,6 +97,11 @@ default_config = {
         "sample_percent": 100,
         "parquet_batching_max_events": 1024,
     },
+    "model_monitoring": {
+        # drift classification thresholds (reference
+        # histogram_data_drift defaults)
+        "drift_thresholds": {"detected": 0.7, "possible": 0.5},
+    },
     "notifications": {
         "smtp": {"server": "", "sender": ""},
     },

@@ -69,13 +69,26 @@ def compute_feature_drift(reference_samples: typing.Sequence,
     return histogram_drift_metrics(ref_hist, cur_hist)
 
 
-DRIFT_DETECTED_THRESHOLD = 0.7
-POSSIBLE_DRIFT_THRESHOLD = 0.5
+DRIFT_DETECTED_THRESHOLD = 0.7   # defaults; overridable via config
+POSSIBLE_DRIFT_THRESHOLD = 0.5   # model_monitoring.drift_thresholds
 
 
 def drift_status(drift_score: float) -> str:
-    if drift_score >= DRIFT_DETECTED_THRESHOLD:
+    """Classify a drift score (reference histogram_data_drift
+    thresholds, overridable via
+    config.model_monitoring.drift_thresholds.{detected,possible})."""
+    detected, possible = (DRIFT_DETECTED_THRESHOLD,
+                          POSSIBLE_DRIFT_THRESHOLD)
+    try:
+        from ..config import config
+
+        thresholds = config.model_monitoring.drift_thresholds
+        detected = float(thresholds.detected)
+        possible = float(thresholds.possible)
+    except Exception:
+        pass
+    if drift_score >= detected:
         return "drift_detected"
-    if drift_score >= POSSIBLE_DRIFT_THRESHOLD:
+    if drift_score >= possible:
         return "possible_drift"
     return "no_drift"
