@@ -230,11 +230,14 @@ class BaseRuntime(ModelObj):
             hyperparams: dict = None, hyper_param_options=None, verbose=None,
             scrape_metrics=None, local: bool = None, local_code_path=None,
             auto_build=None, param_file_secrets=None, notifications=None,
-            returns=None, state_thresholds=None, **launcher_kwargs
-            ) -> RunObject:
+            returns=None, state_thresholds=None, selector: str = None,
+            **launcher_kwargs) -> RunObject:
         """Run this function (locally or submitted to the service)."""
         from ..launcher import LauncherFactory
 
+        if selector:  # reference API: run(selector="max.accuracy")
+            hyper_param_options = dict(hyper_param_options or {})
+            hyper_param_options.setdefault("selector", selector)
         run = self._enrich_run_template(
             runspec, handler=handler, name=name, project=project,
             params=params, inputs=inputs, out_path=out_path,

@@ -286,3 +286,44 @@ class TestProjectArchive:
         assert os.path.isfile(os.path.join(dst_ctx, "fn.py"))
         fn = loaded.get_function("f1")
         assert fn is not None
+
+
+class TestGeneratorProperties:
+    def test_grid_covers_full_product(self, rundb):
+        import itertools
+
+        import mlrun_amd
+
+        seen = []
+
+        def handler(context, a=0, b=""):
+            seen.append((a, b))
+            context.log_result("score", a)
+
+        fn = mlrun_amd.new_function(name="grid", kind="local")
+        run = fn.run(handler=handler, local=True,
+                     hyperparams={"a": [1, 2, 3], "b": ["x", "y"]},
+                     selector="max.score")
+        expect = set(itertools.product([1, 2, 3], ["x", "y"]))
+        assert set(seen) == expect and len(seen) == 6
+        assert run.status.results["best_iteration"] in range(1, 7)
+        # the max selector must pick an a=3 iteration
+        best = run.status.results
+        assert best.get("score") == 3
+
+    def test_random_generator_samples_within_space(self, rundb):
+        import mlrun_amd
+
+        seen = []
+
+        def handler(context, p=0):
+            seen.append(p)
+            context.log_result("r", p)
+
+        fn = mlrun_amd.new_function(name="rand", kind="local")
+        fn.run(handler=handler, local=True,
+               hyperparams={"p": [1, 2, 3, 4, 5]},
+               hyper_param_options={"strategy": "random",
+                                    "max_iterations": 4})
+        assert len(seen) == 4
+        assert all(p in [1, 2, 3, 4, 5] for p in seen)
