@@ -246,10 +246,14 @@ class BaseRuntime(ModelObj):
             verbose=verbose, scrape_metrics=scrape_metrics,
             notifications=notifications, returns=returns,
             state_thresholds=state_thresholds)
+        if launcher_kwargs:
+            from ..utils import logger
+
+            logger.warning("unknown run() arguments ignored",
+                           arguments=sorted(launcher_kwargs))
         launcher = LauncherFactory.create_launcher(
             is_remote=self._is_remote and not local, local=local)
-        return launcher.launch(self, run, schedule=schedule, watch=watch,
-                               **launcher_kwargs)
+        return launcher.launch(self, run, schedule=schedule, watch=watch)
 
     def _enrich_run_template(self, runspec=None, handler=None, name="",
                              project="", params=None, inputs=None,
