@@ -200,7 +200,7 @@ class Scheduler:
                 nxt = trigger.next_fire_time(datetime.datetime.now())
                 self._db.update_schedule(project, name,
                                          {"next_run_time": nxt.isoformat()})
-        except Exception:
-            pass
+        except Exception as exc:
+            logger.debug("next_run_time update failed", error=str(exc))
         logger.info("schedule invoked", schedule=name, run=uid)
         return result

@@ -162,8 +162,8 @@ class MlrunProject(ModelObj):
                                if f.get("name") != name] + [entry]
         try:
             fn.save()
-        except Exception:
-            pass
+        except Exception as exc:
+            logger.debug("function save skipped", error=str(exc))
         return fn
 
     def get_function(self, key: str, sync=False, enrich=False,
