@@ -327,3 +327,21 @@ class TestGeneratorProperties:
                                     "max_iterations": 4})
         assert len(seen) == 4
         assert all(p in [1, 2, 3, 4, 5] for p in seen)
+
+
+class TestFunctionDBRoundtrip:
+    def test_save_then_import_db_uri(self, rundb, tmp_path):
+        import mlrun_amd
+
+        code = tmp_path / "saved_fn.py"
+        code.write_text("def handler(context):\n"
+                        "    context.log_result('ok', 1)\n")
+        fn = mlrun_amd.new_function(name="saved-fn", kind="job",
+                                    project="dbproj",
+                                    command=str(code))
+        fn.save()
+        loaded = mlrun_amd.import_function("db://dbproj/saved-fn")
+        assert loaded.metadata.name == "saved-fn"
+        assert loaded.kind == "job"
+        run = loaded.run(handler="handler", local=True)
+        assert run.outputs["ok"] == 1
