@@ -396,6 +396,54 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         db.delete_project_secrets(project, keys)
         return {}
 
+    # ----------------------------------------------------------- tags
+    @app.get("/api/v1/projects/{project}/tags")
+    async def list_tags(project: str, key: str = ""):
+        return {"tags": db.list_artifact_tags(project, key)}
+
+    @app.put("/api/v1/projects/{project}/tags/{tag}")
+    async def tag_objects(project: str, tag: str, body: dict):
+        """Attach a tag to artifact versions (reference tags.py):
+        body = {"identifiers": [{"key", "tree"|"uid", "iter"?}]}."""
+        for ident in body.get("identifiers") or []:
+            db.tag_artifact(project, ident.get("key"),
+                            ident.get("tree") or ident.get("uid"),
+                            tag, ident.get("iter", 0))
+        return {}
+
+    @app.delete("/api/v1/projects/{project}/tags/{tag}")
+    async def delete_tag(project: str, tag: str, key: str):
+        db.delete_artifact_tag(project, key, tag)
+        return {}
+
+    # ---------------------------------------------------------- files
+    @app.get("/api/v1/files")
+    async def get_file(path: str, size: int = 0, offset: int = 0):
+        """Serve object bytes through the datastore layer (reference
+        files.py GET /files?path=...)."""
+        from ..datastore import store_manager
+
+        try:
+            data = store_manager.object(path).get(size=size or None,
+                                                  offset=offset)
+        except FileNotFoundError:
+            raise HTTPException(status_code=404,
+                                detail=f"{path} not found")
+        if isinstance(data, str):
+            data = data.encode()
+        return Response(content=data,
+                        media_type="application/octet-stream")
+
+    @app.get("/api/v1/filestat")
+    async def file_stat(path: str):
+        from ..datastore import store_manager
+
+        try:
+            return store_manager.object(path).stat()
+        except FileNotFoundError:
+            raise HTTPException(status_code=404,
+                                detail=f"{path} not found")
+
     # ------------------------------------------------------ workflows
     @app.get("/api/v1/projects/{project}/workflows")
     async def list_workflows(project: str):

@@ -335,6 +335,31 @@ class SQLRunDB(RunDBInterface):
                 (project, key, a_tag, tree, iter))
         return artifact
 
+    def tag_artifact(self, project, key, tree, tag, iteration=0):
+        """Attach a tag to a stored artifact version (reference tags
+        endpoint)."""
+        self._execute(
+            "INSERT OR REPLACE INTO artifact_tags "
+            "(project, key, tag, tree, iteration) VALUES (?,?,?,?,?)",
+            (project or "default", key, tag, tree or "", iteration))
+
+    def delete_artifact_tag(self, project, key, tag):
+        self._execute(
+            "DELETE FROM artifact_tags WHERE project=? AND key=? AND "
+            "tag=?", (project or "default", key, tag))
+
+    def list_artifact_tags(self, project, key=""):
+        if key:
+            rows = self._query(
+                "SELECT DISTINCT tag FROM artifact_tags WHERE "
+                "project=? AND key=? ORDER BY tag",
+                (project or "default", key))
+        else:
+            rows = self._query(
+                "SELECT DISTINCT tag FROM artifact_tags WHERE "
+                "project=? ORDER BY tag", (project or "default",))
+        return [r["tag"] for r in rows]
+
     def read_artifact(self, key, tag="", iter=None, project="", tree=None,
                       uid=None):
         project = project or "default"

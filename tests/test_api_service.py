@@ -483,3 +483,33 @@ def test_schedule_payload_validated(client):
     assert resp.status_code == 200
     assert client.db.get_schedule("vp", "ok")["cron_trigger"] == \
         "*/5 * * * *"
+
+
+def test_tags_and_files_endpoints(client, tmp_path):
+    db = client.db
+    db.store_artifact("modelA", {"kind": "model",
+                                 "metadata": {"key": "modelA",
+                                              "tree": "t1"},
+                                 "spec": {}},
+                      tree="t1", project="tp")
+    resp = client.put("/api/v1/projects/tp/tags/prod",
+                      json={"identifiers": [{"key": "modelA",
+                                             "tree": "t1"}]})
+    assert resp.status_code == 200
+    tags = client.get("/api/v1/projects/tp/tags?key=modelA"
+                      ).json()["tags"]
+    assert "prod" in tags
+    client.delete("/api/v1/projects/tp/tags/prod?key=modelA")
+    tags = client.get("/api/v1/projects/tp/tags?key=modelA"
+                      ).json()["tags"]
+    assert "prod" not in tags
+
+    blob = tmp_path / "blob.bin"
+    blob.write_bytes(b"0123456789")
+    resp = client.get(f"/api/v1/files?path={blob}")
+    assert resp.content == b"0123456789"
+    resp = client.get(f"/api/v1/files?path={blob}&offset=3&size=4")
+    assert resp.content == b"3456"
+    assert client.get(f"/api/v1/filestat?path={blob}"
+                      ).json()["size"] == 10
+    assert client.get("/api/v1/files?path=/nope/x").status_code == 404
