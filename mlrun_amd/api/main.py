@@ -396,6 +396,29 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         db.delete_project_secrets(project, keys)
         return {}
 
+    # ------------------------------------------------ alert templates
+    @app.put("/api/v1/alert-templates/{name}")
+    async def store_alert_template(name: str, body: dict):
+        """System-wide reusable alert templates (reference
+        alert_template.py); instantiate by merging into an
+        AlertConfig."""
+        body["template_name"] = name
+        db.store_alert_template(name, body)
+        return body
+
+    @app.get("/api/v1/alert-templates")
+    async def list_alert_templates():
+        return {"templates": db.list_alert_templates()}
+
+    @app.get("/api/v1/alert-templates/{name}")
+    async def get_alert_template(name: str):
+        return db.get_alert_template(name)
+
+    @app.delete("/api/v1/alert-templates/{name}")
+    async def delete_alert_template(name: str):
+        db.delete_alert_template(name)
+        return {}
+
     # ----------------------------------------------------------- tags
     @app.get("/api/v1/projects/{project}/tags")
     async def list_tags(project: str, key: str = ""):

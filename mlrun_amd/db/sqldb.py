@@ -96,6 +96,9 @@ CREATE TABLE IF NOT EXISTS alert_states (
     last_updated TEXT, active INTEGER DEFAULT 0,
     PRIMARY KEY (project, name)
 );
+CREATE TABLE IF NOT EXISTS alert_templates (
+    name TEXT PRIMARY KEY, updated TEXT, body TEXT
+);
 CREATE TABLE IF NOT EXISTS project_secrets (
     project TEXT NOT NULL, key TEXT NOT NULL, value TEXT,
     PRIMARY KEY (project, key)
@@ -334,6 +337,27 @@ class SQLRunDB(RunDBInterface):
                 "(project, key, tag, tree, iteration) VALUES (?,?,?,?,?)",
                 (project, key, a_tag, tree, iter))
         return artifact
+
+    def store_alert_template(self, name, template: dict):
+        self._execute(
+            "INSERT OR REPLACE INTO alert_templates "
+            "(name, updated, body) VALUES (?,?,?)",
+            (name, now_iso(), json.dumps(template, default=str)))
+
+    def get_alert_template(self, name):
+        rows = self._query(
+            "SELECT body FROM alert_templates WHERE name=?", (name,))
+        if not rows:
+            raise MLRunNotFoundError(f"alert template {name} not found")
+        return json.loads(rows[0]["body"])
+
+    def list_alert_templates(self):
+        return [json.loads(r["body"]) for r in self._query(
+            "SELECT body FROM alert_templates ORDER BY name")]
+
+    def delete_alert_template(self, name):
+        self._execute("DELETE FROM alert_templates WHERE name=?",
+                      (name,))
 
     def tag_artifact(self, project, key, tree, tag, iteration=0):
         """Attach a tag to a stored artifact version (reference tags

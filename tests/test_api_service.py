@@ -513,3 +513,20 @@ def test_tags_and_files_endpoints(client, tmp_path):
     assert client.get(f"/api/v1/filestat?path={blob}"
                       ).json()["size"] == 10
     assert client.get("/api/v1/files?path=/nope/x").status_code == 404
+
+
+def test_alert_templates(client):
+    body = {"summary": "high drift on {{endpoint}}",
+            "severity": "high",
+            "trigger": {"events": ["model-drift"]},
+            "criteria": {"count": 3}}
+    assert client.put("/api/v1/alert-templates/drift-high",
+                      json=body).status_code == 200
+    got = client.get("/api/v1/alert-templates/drift-high").json()
+    assert got["criteria"]["count"] == 3
+    names = [t["template_name"] for t in
+             client.get("/api/v1/alert-templates").json()["templates"]]
+    assert "drift-high" in names
+    client.delete("/api/v1/alert-templates/drift-high")
+    assert client.get("/api/v1/alert-templates/drift-high"
+                      ).status_code in (404, 500)
