@@ -396,6 +396,28 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         db.delete_project_secrets(project, keys)
         return {}
 
+    # ---------------------------------------------- datastore profiles
+    @app.put("/api/v1/projects/{project}/datastore-profiles")
+    async def store_datastore_profile(project: str, body: dict):
+        """Named datastore configs (reference datastore_profile.py —
+        connection parameters for s3/redis/... targets, minus the
+        secrets which live in the project secret store)."""
+        db.store_datastore_profile(project, body)
+        return body
+
+    @app.get("/api/v1/projects/{project}/datastore-profiles")
+    async def list_datastore_profiles(project: str):
+        return {"profiles": db.list_datastore_profiles(project)}
+
+    @app.get("/api/v1/projects/{project}/datastore-profiles/{name}")
+    async def get_datastore_profile(project: str, name: str):
+        return db.get_datastore_profile(project, name)
+
+    @app.delete("/api/v1/projects/{project}/datastore-profiles/{name}")
+    async def delete_datastore_profile(project: str, name: str):
+        db.delete_datastore_profile(project, name)
+        return {}
+
     # ------------------------------------------------ alert templates
     @app.put("/api/v1/alert-templates/{name}")
     async def store_alert_template(name: str, body: dict):

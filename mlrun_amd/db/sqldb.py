@@ -96,6 +96,11 @@ CREATE TABLE IF NOT EXISTS alert_states (
     last_updated TEXT, active INTEGER DEFAULT 0,
     PRIMARY KEY (project, name)
 );
+CREATE TABLE IF NOT EXISTS datastore_profiles (
+    project TEXT NOT NULL, name TEXT NOT NULL, type TEXT, updated TEXT,
+    body TEXT,
+    PRIMARY KEY (project, name)
+);
 CREATE TABLE IF NOT EXISTS alert_templates (
     name TEXT PRIMARY KEY, updated TEXT, body TEXT
 );
@@ -337,6 +342,33 @@ class SQLRunDB(RunDBInterface):
                 "(project, key, tag, tree, iteration) VALUES (?,?,?,?,?)",
                 (project, key, a_tag, tree, iter))
         return artifact
+
+    def store_datastore_profile(self, project, profile: dict):
+        name = profile.get("name")
+        self._execute(
+            "INSERT OR REPLACE INTO datastore_profiles "
+            "(project, name, type, updated, body) VALUES (?,?,?,?,?)",
+            (project or "default", name, profile.get("type", ""),
+             now_iso(), json.dumps(profile, default=str)))
+
+    def get_datastore_profile(self, project, name):
+        rows = self._query(
+            "SELECT body FROM datastore_profiles WHERE project=? AND "
+            "name=?", (project or "default", name))
+        if not rows:
+            raise MLRunNotFoundError(
+                f"datastore profile {name} not found")
+        return json.loads(rows[0]["body"])
+
+    def list_datastore_profiles(self, project):
+        return [json.loads(r["body"]) for r in self._query(
+            "SELECT body FROM datastore_profiles WHERE project=? "
+            "ORDER BY name", (project or "default",))]
+
+    def delete_datastore_profile(self, project, name):
+        self._execute(
+            "DELETE FROM datastore_profiles WHERE project=? AND name=?",
+            (project or "default", name))
 
     def store_alert_template(self, name, template: dict):
         self._execute(

@@ -530,3 +530,19 @@ def test_alert_templates(client):
     client.delete("/api/v1/alert-templates/drift-high")
     assert client.get("/api/v1/alert-templates/drift-high"
                       ).status_code in (404, 500)
+
+
+def test_datastore_profiles(client):
+    prof = {"name": "my-s3", "type": "s3",
+            "public": {"endpoint_url": "http://minio:9000",
+                       "bucket": "data"}}
+    assert client.put("/api/v1/projects/dp/datastore-profiles",
+                      json=prof).status_code == 200
+    got = client.get("/api/v1/projects/dp/datastore-profiles/my-s3"
+                     ).json()
+    assert got["public"]["bucket"] == "data"
+    assert len(client.get("/api/v1/projects/dp/datastore-profiles"
+                          ).json()["profiles"]) == 1
+    client.delete("/api/v1/projects/dp/datastore-profiles/my-s3")
+    assert client.get("/api/v1/projects/dp/datastore-profiles"
+                      ).json()["profiles"] == []
