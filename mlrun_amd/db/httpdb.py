@@ -359,6 +359,39 @@ class HTTPRunDB(RunDBInterface):
         return self.api_call(
             "GET", f"projects/{project}/pipelines").get("runs", [])
 
+    def tag_artifact(self, project, key, tree, tag, iteration=0):
+        self.api_call("PUT", f"projects/{project}/tags/{tag}",
+                      json_body={"identifiers": [
+                          {"key": key, "tree": tree,
+                           "iter": iteration}]})
+
+    def list_artifact_tags(self, project, key=""):
+        return self.api_call(
+            "GET", f"projects/{project}/tags",
+            params={"key": key}).get("tags", [])
+
+    def get_file(self, path, size=0, offset=0):
+        return self.api_call("GET", "files",
+                             params={"path": path, "size": size,
+                                     "offset": offset}, raw=True)
+
+    def store_datastore_profile(self, project, profile: dict):
+        self.api_call("PUT", f"projects/{project}/datastore-profiles",
+                      json_body=profile)
+
+    def list_datastore_profiles(self, project):
+        return self.api_call(
+            "GET", f"projects/{project}/datastore-profiles"
+        ).get("profiles", [])
+
+    def store_alert_template(self, name, template: dict):
+        self.api_call("PUT", f"alert-templates/{name}",
+                      json_body=template)
+
+    def list_alert_templates(self):
+        return self.api_call("GET", "alert-templates"
+                             ).get("templates", [])
+
     def get_background_task(self, project, name):
         return self.api_call(
             "GET", f"projects/{project}/background-tasks/{name}")
