@@ -192,3 +192,22 @@ print("REMOTE_OK", run.metadata.uid)
         finally:
             proc.terminate()
             proc.wait(timeout=10)
+
+    def test_get_artifacts_and_schedules(self, capsys, rundb, tmp_path):
+        rundb.store_artifact("art1", {"kind": "file",
+                                      "metadata": {"key": "art1"},
+                                      "spec": {}}, tree="t1",
+                            project="default")
+        code = tmp_path / "s.py"
+        code.write_text("def handler(context):\n    pass\n")
+        rundb.create_schedule("default", {
+            "name": "nightly", "kind": "job",
+            "cron_trigger": "0 3 * * *",
+            "task": {"metadata": {"name": "nightly"},
+                     "spec": {"function": str(code),
+                              "handler": "handler"}}})
+        main(["get", "artifacts"])
+        assert "art1" in capsys.readouterr().out
+        main(["get", "schedules"])
+        out = capsys.readouterr().out
+        assert "nightly" in out and "0 3 * * *" in out
