@@ -383,3 +383,23 @@ class TestSecretsSources:
         monkeypatch.setenv("SOME_TOKEN", "tok")
         assert get_secret_or_env("SOME_TOKEN") == "tok"
         assert get_secret_or_env("NOPE_X", default="d") == "d"
+
+
+class TestAliasEquivalence:
+    def test_mlrun_is_mlrun_amd(self):
+        """`import mlrun` must expose the SAME objects (drop-in)."""
+        import mlrun
+        import mlrun_amd
+
+        assert mlrun.new_function is mlrun_amd.new_function
+        assert mlrun.code_to_function is mlrun_amd.code_to_function
+        from mlrun.feature_store import FeatureSet as A
+        from mlrun_amd.feature_store import FeatureSet as B
+
+        assert A is B
+        from mlrun.serving import V2ModelServer as SA
+        from mlrun_amd.serving import V2ModelServer as SB
+
+        assert SA is SB
+        fn = mlrun.new_function(name="alias-fn", kind="local")
+        assert type(fn).__module__.startswith("mlrun_amd")
