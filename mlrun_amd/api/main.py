@@ -14,6 +14,7 @@ informers.
 """
 
 import asyncio
+import os
 import typing
 
 from fastapi import (FastAPI, HTTPException, Query, Request,
@@ -462,12 +463,46 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         return {}
 
     # ---------------------------------------------------------- files
+    def _authorize_file_path(path: str):
+        """Restrict /files and /filestat to configured data prefixes
+        (reference server routes files.py through per-path
+        authorization; here the allowlist is base_dir + artifact_path +
+        the service's own dirpath + any ``httpdb.files_allowed_paths``
+        entries)."""
+        from ..config import config as _cfg
+
+        allowed = [p for p in (
+            _cfg.httpdb.files_allowed_paths or "").split(",") if p]
+        for candidate in (_cfg.base_dir, _cfg.artifact_path,
+                          _cfg.httpdb.dirpath, _cfg.httpdb.logs_path):
+            if candidate:
+                allowed.append(candidate)
+        # non-file schemes (memory://, store://) resolve through their
+        # own stores and never touch the host filesystem directly
+        if "://" in path and not path.startswith("file://"):
+            return
+        local = path[len("file://"):] if path.startswith("file://") else path
+        real = os.path.realpath(local)
+        for prefix in allowed:
+            if "://" in prefix:
+                if path.startswith(prefix):
+                    return
+                continue
+            rp = os.path.realpath(os.path.expanduser(prefix))
+            if real == rp or real.startswith(rp.rstrip("/") + "/"):
+                return
+        raise HTTPException(
+            status_code=403,
+            detail=f"path {path!r} is outside the allowed data prefixes "
+                   f"(configure httpdb.files_allowed_paths)")
+
     @app.get("/api/v1/files")
     async def get_file(path: str, size: int = 0, offset: int = 0):
         """Serve object bytes through the datastore layer (reference
         files.py GET /files?path=...)."""
         from ..datastore import store_manager
 
+        _authorize_file_path(path)
         try:
             data = store_manager.object(path).get(size=size or None,
                                                   offset=offset)
@@ -483,6 +518,7 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     async def file_stat(path: str):
         from ..datastore import store_manager
 
+        _authorize_file_path(path)
         try:
             return store_manager.object(path).stat()
         except FileNotFoundError:

@@ -34,10 +34,18 @@ def _parse_field(field: str, low: int, high: int) -> typing.Set[int]:
             start, end = int(start_s), int(end_s)
         else:
             start = end = int(part)
-        for value in range(start, end + 1, step):
-            if low <= value <= high:
-                values.add(value)
+        if start < low or end > high or start > end:
+            raise MLRunInvalidArgumentError(
+                f"cron field value {part!r} out of range [{low},{high}]")
+        values.update(range(start, end + 1, step))
     return values
+
+
+def _parse_weekday_field(field: str) -> typing.Set[int]:
+    """Standard-cron day-of-week (Sun=0 or 7, Mon=1 .. Sat=6) mapped to
+    Python ``dt.weekday()`` numbering (Mon=0 .. Sun=6)."""
+    cron_values = _parse_field(field, 0, 7)
+    return {(v - 1) % 7 for v in cron_values}
 
 
 class CronTrigger:
@@ -53,7 +61,7 @@ class CronTrigger:
         self.hours = _parse_field(fields[1], 0, 23)
         self.days = _parse_field(fields[2], 1, 31)
         self.months = _parse_field(fields[3], 1, 12)
-        self.weekdays = _parse_field(fields[4], 0, 6)
+        self.weekdays = _parse_weekday_field(fields[4])
 
     def matches(self, dt: datetime.datetime) -> bool:
         return (dt.minute in self.minutes and dt.hour in self.hours and

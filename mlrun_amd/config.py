@@ -44,6 +44,11 @@ default_config = {
         "max_log_size_bytes": 1024 * 1024 * 8,
         "retry_api_call_on_exception": "enabled",
         "http_connection_timeout": 20,
+        # comma-separated path/scheme prefixes the /files and /filestat
+        # endpoints may serve; empty -> derived from base_dir +
+        # artifact_path (reference routes files.py through per-path
+        # authorization — this is the node-local analog)
+        "files_allowed_paths": "",
     },
     "runs": {
         "monitoring_interval": 30,  # seconds between run-monitor sweeps

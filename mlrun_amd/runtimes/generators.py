@@ -90,8 +90,10 @@ class TaskGenerator:
         condition = self.options.stop_condition
         if not condition:
             return False
+        from ..utils.safe_eval import safe_eval
+
         try:
-            return bool(eval(condition, {"__builtins__": {}}, dict(results)))
+            return bool(safe_eval(condition, dict(results)))
         except Exception:
             return False
 

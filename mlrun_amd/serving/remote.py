@@ -47,8 +47,9 @@ class RemoteStep:
 
     def _resolve_url(self, event) -> str:
         if self.url_expression:
-            return eval(self.url_expression, {"__builtins__": {}},
-                        {"event": event})
+            from ..utils.safe_eval import safe_eval
+
+            return safe_eval(self.url_expression, {"event": event})
         url = self.url or ""
         if self.subpath:
             subpath = self.subpath
@@ -61,8 +62,9 @@ class RemoteStep:
         url = self._resolve_url(event)
         body = event.body
         if self.body_expression:
-            body = eval(self.body_expression, {"__builtins__": {}},
-                        {"event": event})
+            from ..utils.safe_eval import safe_eval
+
+            body = safe_eval(self.body_expression, {"event": event})
         data = json.dumps(body, default=str) if isinstance(
             body, (dict, list)) else body
         last_error = None

@@ -146,9 +146,11 @@ class NotificationPusher:
                     continue
                 condition = spec.get("condition")
                 if condition:
+                    from .safe_eval import safe_eval
+
                     try:
-                        ok = bool(eval(condition, {"__builtins__": {}},
-                                       {"run": run.to_dict()}))
+                        ok = bool(safe_eval(condition,
+                                            {"run": run.to_dict()}))
                     except Exception:
                         ok = True
                     if not ok:
