@@ -504,15 +504,26 @@ def test_tags_and_files_endpoints(client, tmp_path):
                       ).json()["tags"]
     assert "prod" not in tags
 
+    from mlrun_amd.config import config
+
     blob = tmp_path / "blob.bin"
     blob.write_bytes(b"0123456789")
-    resp = client.get(f"/api/v1/files?path={blob}")
-    assert resp.content == b"0123456789"
-    resp = client.get(f"/api/v1/files?path={blob}&offset=3&size=4")
-    assert resp.content == b"3456"
-    assert client.get(f"/api/v1/filestat?path={blob}"
-                      ).json()["size"] == 10
-    assert client.get("/api/v1/files?path=/nope/x").status_code == 404
+    saved = config.httpdb.files_allowed_paths
+    config.httpdb.files_allowed_paths = str(tmp_path)
+    try:
+        resp = client.get(f"/api/v1/files?path={blob}")
+        assert resp.content == b"0123456789"
+        resp = client.get(f"/api/v1/files?path={blob}&offset=3&size=4")
+        assert resp.content == b"3456"
+        assert client.get(f"/api/v1/filestat?path={blob}"
+                          ).json()["size"] == 10
+        assert client.get(f"/api/v1/files?path={tmp_path}/nope"
+                          ).status_code == 404
+        # outside every allowed prefix -> denied, not served
+        assert client.get("/api/v1/files?path=/nope/x"
+                          ).status_code == 403
+    finally:
+        config.httpdb.files_allowed_paths = saved
 
 
 def test_alert_templates(client):
