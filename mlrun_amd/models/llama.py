@@ -491,8 +491,15 @@ class LlamaDecodeEngine:
     def _capture_graph_locked(self):
         if self._graph is not None:
             return
+        import torch.distributed as dist
+
         lens_backup = self.cache_lens.clone()
         tokens_backup = self.buf_tokens.clone()
+        if self.tp_size > 1 and dist.is_available() and \
+                dist.is_initialized():
+            # warmup runs REAL collectives — align ranks first so the
+            # three eager steps pair up even if ranks arrive staggered
+            dist.barrier(group=self.tp_group)
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
@@ -502,7 +509,15 @@ class LlamaDecodeEngine:
         self.cache_lens.copy_(lens_backup)
         self.buf_tokens.copy_(tokens_backup)
         graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        # RCCL collectives ARE capturable (the all-reduce kernel is
+        # recorded into the graph and pairs up across ranks at replay),
+        # but the NCCL watchdog thread polls events during capture —
+        # "global" capture mode would see those as stray captures, so
+        # distributed captures use thread_local error mode (the decode
+        # body launches everything from this thread).
+        err_mode = "thread_local" if dist.is_available() and \
+            dist.is_initialized() else "global"
+        with torch.cuda.graph(graph, capture_error_mode=err_mode):
             self._decode_step_body()
         # stream capture records without executing, but restore anyway
         # in case a backend executed eagerly during capture

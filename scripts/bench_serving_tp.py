@@ -61,12 +61,14 @@ def main():
         cfg = LlamaConfig.llama3_8b()
     batch = args.batch if on_gpu else 2
     prompt_len = args.prompt_len if on_gpu else 8
-    # TP decode runs eager (RCCL collectives inside hipGraph capture
-    # are exercised in round 2); per-rank random shards — throughput
-    # measurement only
+    # TP decode is hipGraph-captured: the per-layer RCCL all-reduces
+    # are recorded into the graph and pair up across ranks at replay
+    # (capture_error_mode=thread_local dodges the NCCL watchdog).
+    # MLRUN_TP_GRAPH=0 falls back to eager for A/B.
+    use_graph = on_gpu and os.environ.get("MLRUN_TP_GRAPH", "1") != "0"
     engine = LlamaDecodeEngine(cfg, batch, device=device, tp_group=group,
                                tp_rank=tp_rank, tp_size=tp_size,
-                               use_graph=False, seed=7)
+                               use_graph=use_graph, seed=7)
 
     gen = torch.Generator().manual_seed(77)
 
@@ -110,7 +112,8 @@ def main():
             "data": "synthetic",
             "config": {"model": cfg.name, "global_batch": batch,
                        "seq_len": prompt_len + args.gen_tokens,
-                       "parallelism": f"tp{tp_size}"},
+                       "parallelism": f"tp{tp_size}",
+                       "hipgraph": use_graph},
         }))
     if dist.is_initialized():
         dist.destroy_process_group()
