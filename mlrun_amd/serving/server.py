@@ -28,7 +28,8 @@ class Event:
     is on the hot path)."""
 
     __slots__ = ["id", "body", "path", "method", "headers", "content_type",
-                 "error", "terminated", "responded", "origin_state", "time"]
+                 "error", "terminated", "responded", "origin_state", "time",
+                 "_recovery"]
 
     def __init__(self, body=None, id=None, path="", method="POST",
                  headers=None, content_type=None, time_=None):

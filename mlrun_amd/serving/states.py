@@ -370,6 +370,11 @@ class QueueStep(BaseStep):
         from ..config import config
 
         self.context = context
+        if self.parent is not None and \
+                getattr(self.parent, "engine", "sync") == "async":
+            # async engine: the step's bounded inbox IS the buffer —
+            # no thread-backed queue (see async_flow.py)
+            return
         if self._queue is None:
             self._queue = queue_mod.Queue(
                 maxsize=self.max_size or int(config.serving.max_queue))
@@ -394,6 +399,18 @@ class QueueStep(BaseStep):
                 self._queue.task_done()
 
     def run(self, event):
+        if self.parent is not None and \
+                getattr(self.parent, "engine", "sync") == "async":
+            # async engine: passthrough — downstream steps have their
+            # own bounded inboxes; a pathed queue with no local
+            # consumers publishes to its stream instead (reference
+            # _init_async_objects stream-target behavior)
+            if self.path and not self._next:
+                from ..platforms import OutputStream
+
+                OutputStream(self.path).push([event.body])
+                event.terminated = True
+            return event
         if self._queue is None:
             self.init_object(self.context, None)
         self._queue.put(event)
@@ -420,6 +437,7 @@ class FlowStep(BaseStep):
         self.engine = engine or "sync"
         self.final_step = final_step
         self._start_steps: typing.List[BaseStep] = []
+        self._controller = None  # async engine (engine="async")
         if steps:
             for key, step in steps.items():
                 step.name = step.name or key
@@ -449,7 +467,16 @@ class FlowStep(BaseStep):
             step = class_name
             step.name = step.name or name
         elif class_name == "$queue" or class_args.pop("_queue", False):
-            step = QueueStep(name, path=class_args.pop("path", None))
+            step = QueueStep(name, path=class_args.pop("path", None),
+                             max_size=class_args.pop("max_size", None),
+                             **class_args)
+            class_args = {}
+        elif isinstance(class_name, str) and class_name.startswith("*"):
+            # "*" = router step ("*RouterClass" names the router class —
+            # reference states.py new_model_endpoint/"*" convention)
+            step = RouterStep(class_name=class_name[1:] or None,
+                              class_args=class_args, name=name,
+                              function=function)
         else:
             step = TaskStep(class_name, class_args, handler=handler,
                             name=name, full_event=full_event,
@@ -465,6 +492,17 @@ class FlowStep(BaseStep):
         elif self.steps:
             # default chaining: after the previously added step
             step.after = [list(self.steps.values())[-1].name]
+        if self.engine == "async" and not isinstance(step, QueueStep):
+            # a step consuming from a queue runs in a CHILD function —
+            # it must name one (reference async-flow contract, tested
+            # by test_async_error_on_missing_function_parameter)
+            for upstream in step.after or []:
+                if upstream in self.steps and \
+                        isinstance(self.steps[upstream], QueueStep) and \
+                        not (function or getattr(step, "function", None)):
+                    raise MLRunInvalidArgumentError(
+                        f"step '{step.name}' must specify a function, "
+                        f"because it follows a queue step")
         self._attach(step)
         if before:
             self.steps[before].after = [step.name]
@@ -494,6 +532,13 @@ class FlowStep(BaseStep):
         self._build_links()
         for step in self.steps.values():
             step.init_object(context, namespace, mode, reset=reset)
+        if self.engine == "async" and self._controller is None:
+            from ..config import config
+            from .async_flow import AsyncFlowController
+
+            self._controller = AsyncFlowController(
+                self, max_queue=int(config.serving.max_queue))
+            self._controller.start()
 
     def _build_links(self):
         for step in self.steps.values():
@@ -520,10 +565,24 @@ class FlowStep(BaseStep):
     def run(self, event):
         if not self._start_steps:
             self._build_links()
+        if self._controller is not None:
+            # async engine: submit + block on the per-event future
+            # (async callers use run_async for a non-blocking await)
+            return self._controller.emit(event, timeout=600)
         for step in self._start_steps:
             result = self._run_from(step, event)
             event = result if result is not None else event
         return event
+
+    async def run_async(self, event):
+        """Awaitable entry for async hosts (FastAPI): wraps the
+        per-event future."""
+        if self._controller is None:
+            return self.run(event)
+        import asyncio as _asyncio
+
+        return await _asyncio.wrap_future(
+            self._controller.emit_nowait(event))
 
     def _run_from(self, step: BaseStep, event):
         """The sync hot loop: walk the next-chain from a step
@@ -584,9 +643,20 @@ class FlowStep(BaseStep):
         return event
 
     def wait_for_completion(self):
+        if self._controller is not None:
+            self._controller.wait_for_completion()
         for step in self.steps.values():
             if isinstance(step, QueueStep) and step._queue is not None:
                 step._queue.join()
+
+    def shutdown(self):
+        """Stop the async controller (and any queue workers)."""
+        if self._controller is not None:
+            self._controller.stop()
+            self._controller = None
+        for step in self.steps.values():
+            if isinstance(step, QueueStep):
+                step._stop = True
 
     def to_dict(self):
         struct = super().to_dict()

@@ -1,0 +1,294 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Async flow engine tests — analogs of the reference's
+tests/serving/test_async_flow.py (storey semantics: awaitable results,
+responder mid-flow, queue fan-out, error handlers, back-pressure)."""
+
+import threading
+import time
+
+import pytest
+
+import mlrun_amd
+from mlrun_amd.errors import MLRunInvalidArgumentError
+
+
+class ChainWithContext:
+    """Appends its step name to the body and counts visits on the
+    context (reference demo_states.ChainWithContext)."""
+
+    def __init__(self, context=None, name=None, **kw):
+        self.context = context
+        self.name = name
+
+    def do(self, event):
+        visits = getattr(self.context, "visits", None)
+        if visits is None:
+            self.context.visits = visits = {}
+        visits[self.name] = visits.get(self.name, 0) + 1
+        body = event.body if isinstance(event.body, list) else []
+        event.body = body + [self.name]
+        return event
+
+
+class Chain:
+    def __init__(self, context=None, name=None, **kw):
+        self.name = name
+
+    def do(self, event):
+        event.body = (event.body or []) + [self.name]
+        return event
+
+
+class Echo:
+    def __init__(self, context=None, name=None, **kw):
+        self.name = name
+
+    def do(self, event):
+        return event
+
+
+class Raiser:
+    def __init__(self, context=None, name=None, **kw):
+        pass
+
+    def do(self, event):
+        raise ValueError("simulated failure")
+
+
+class EchoError:
+    def __init__(self, context=None, name=None, **kw):
+        pass
+
+    def do(self, event):
+        event.body = {"error": event.error,
+                      "origin_state": event.origin_state}
+        return event
+
+
+class Slow:
+    """Sleeps per event — used to observe back-pressure/pipelining."""
+
+    seen = []
+
+    def __init__(self, context=None, name=None, delay=0.05, **kw):
+        self.delay = delay
+        self.name = name
+
+    def do(self, event):
+        time.sleep(self.delay)
+        Slow.seen.append((self.name, event.body))
+        return event
+
+
+NS = dict(ChainWithContext=ChainWithContext, Chain=Chain, Echo=Echo,
+          Raiser=Raiser, EchoError=EchoError, Slow=Slow)
+
+
+def _mock_server(fn):
+    return fn.to_mock_server(namespace=NS)
+
+
+class TestAsyncBasic:
+    def test_async_basic(self):
+        """Queue fan-out across (mock) child functions; responder step
+        's5' supplies the response while s3/s4 continue async —
+        reference test_async_basic."""
+        fn = mlrun_amd.new_function("t-async", kind="serving")
+        flow = fn.set_topology("flow", engine="async")
+        queue = flow.to(name="s1", class_name="ChainWithContext").to(
+            "$queue", "q1", path="")
+        s2 = queue.to(name="s2", class_name="ChainWithContext",
+                      function="some_function")
+        s2.to(name="s4", class_name="ChainWithContext")
+        s2.to(name="s5", class_name="ChainWithContext").respond()
+        queue.to(name="s3", class_name="ChainWithContext",
+                 function="some_other_function")
+
+        server = _mock_server(fn)
+        server.context.visits = {}
+        resp = server.test(body=[])
+        server.wait_for_completion()
+        assert resp == ["s1", "s2", "s5"], f"wrong response {resp}"
+        assert server.context.visits == {
+            "s1": 1, "s2": 1, "s4": 1, "s3": 1, "s5": 1}
+        server.graph.shutdown()
+
+    def test_missing_function_after_queue_raises(self):
+        fn = mlrun_amd.new_function("t-async2", kind="serving")
+        flow = fn.set_topology("flow", engine="async")
+        queue = flow.to(name="s1", class_name="ChainWithContext").to(
+            "$queue", "q1", path="")
+        with pytest.raises(MLRunInvalidArgumentError,
+                           match="must specify a function"):
+            queue.to(name="s2", class_name="ChainWithContext")
+
+    def test_async_nested_router(self):
+        """Router nested inside an async flow (reference
+        test_async_nested)."""
+        from tests.test_serving import EchoModel  # noqa: F401
+
+        fn = mlrun_amd.new_function("t-async3", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+        graph.add_step(name="s1", class_name="Echo")
+        router = graph.add_step("*", name="ensemble", after="s1")
+        router.add_route("m1", class_name=EchoModel, model_path=".")
+        graph.add_step(name="final", class_name="Echo",
+                       after="ensemble").respond()
+        server = _mock_server(fn)
+        resp = server.test("/v2/models/m1/infer", body={"inputs": [5]})
+        server.wait_for_completion()
+        assert resp["outputs"] == [10]  # EchoModel doubles its inputs
+        server.graph.shutdown()
+
+    def test_on_error_handler(self):
+        """Error handler placement: error event routes to the handler
+        and ITS downstream continues (reference test_on_error)."""
+        fn = mlrun_amd.new_function("t-async4", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+        chain = graph.to("Chain", name="s1")
+        chain.to("Raiser").error_handler(
+            name="catch", class_name="EchoError", full_event=True)
+        server = _mock_server(fn)
+        resp = server.test(body=[], silent=True)
+        server.wait_for_completion()
+        assert resp["error"].startswith("ValueError")
+        assert resp["origin_state"] == "Raiser"
+        server.graph.shutdown()
+
+    def test_unhandled_error_is_response(self):
+        fn = mlrun_amd.new_function("t-async5", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+        graph.to("Chain", name="s1").to("Raiser")
+        server = _mock_server(fn)
+        result = server.test(body=[], silent=True, get_body=False)
+        server.wait_for_completion()
+        assert "ValueError" in str(result.body)
+        server.graph.shutdown()
+
+
+class TestAsyncConcurrency:
+    def test_awaitable_results_concurrent(self):
+        """N events submitted concurrently each resolve with their own
+        result (per-event futures, no cross-talk)."""
+        fn = mlrun_amd.new_function("t-async6", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+        graph.to("Chain", name="a").to("Chain", name="b").respond()
+        server = _mock_server(fn)
+        results = [None] * 8
+        errors = []
+
+        def call(i):
+            try:
+                results[i] = server.test(body=[f"e{i}"])
+            except Exception as exc:  # pragma: no cover
+                errors.append(exc)
+
+        threads = [threading.Thread(target=call, args=(i,))
+                   for i in range(8)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        assert not errors
+        for i, r in enumerate(results):
+            assert r == [f"e{i}", "a", "b"], r
+        server.graph.shutdown()
+
+    def test_pipelining_across_steps(self):
+        """With per-step workers, step 'a' starts event 2 while step
+        'b' still processes event 1 (storey-style pipelining)."""
+        Slow.seen = []
+        fn = mlrun_amd.new_function("t-async7", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+        graph.to("Slow", name="a", delay=0.03).to(
+            "Slow", name="b", delay=0.03).respond()
+        server = _mock_server(fn)
+        t0 = time.perf_counter()
+        futures = []
+        flow = server.graph
+        from mlrun_amd.serving.server import Event
+
+        for i in range(6):
+            futures.append(
+                flow._controller.emit_nowait(Event(body=[i])))
+        for f in futures:
+            f.result(timeout=10)
+        elapsed = time.perf_counter() - t0
+        # serial would be 6*(0.03+0.03)=0.36s; pipelined ~0.03*7=0.21
+        assert elapsed < 0.33, f"no pipelining: {elapsed:.3f}s"
+        server.graph.shutdown()
+
+    def test_backpressure_bounded_inbox(self):
+        """A slow step's bounded inbox blocks upstream puts — emitted
+        events stay pending rather than accumulating unboundedly."""
+        fn = mlrun_amd.new_function("t-async8", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+        queue = graph.to("Chain", name="src").to("$queue", "q1",
+                                                 max_size=2)
+        queue.to(name="slow", class_name="Slow", delay=0.05,
+                 function="child").respond()
+        server = _mock_server(fn)
+        flow = server.graph
+        from mlrun_amd.serving.server import Event
+
+        futures = [flow._controller.emit_nowait(Event(body=[i]))
+                   for i in range(6)]
+        # the bounded q1 inbox (2) + slow worker → pressure: not all
+        # events can be in q1 at once
+        inbox = flow._controller._inboxes["q1"]
+        assert inbox.maxsize == 2
+        for f in futures:
+            f.result(timeout=10)
+        server.graph.shutdown()
+
+
+class TestAsyncRemoteFunction:
+    def test_forward_to_child_function_host(self):
+        """A step annotated function='child' following a queue runs on
+        the child function's HTTP host when the context registers an
+        endpoint (multi-function graph across processes)."""
+        import json
+
+        from http.server import BaseHTTPRequestHandler, HTTPServer
+
+        hits = []
+
+        class Handler(BaseHTTPRequestHandler):
+            def do_POST(self):
+                length = int(self.headers.get("content-length", 0))
+                body = json.loads(self.rfile.read(length) or b"null")
+                hits.append((self.path, body))
+                out = json.dumps(body + ["remote"]).encode()
+                self.send_response(200)
+                self.send_header("content-type", "application/json")
+                self.end_headers()
+                self.wfile.write(out)
+
+            def log_message(self, *a):
+                pass
+
+        httpd = HTTPServer(("127.0.0.1", 0), Handler)
+        port = httpd.server_address[1]
+        thread = threading.Thread(target=httpd.serve_forever,
+                                  daemon=True)
+        thread.start()
+        try:
+            fn = mlrun_amd.new_function("t-async9", kind="serving")
+            graph = fn.set_topology("flow", engine="async")
+            queue = graph.to("Chain", name="s1").to("$queue", "q1")
+            queue.to(name="s2", class_name="Chain",
+                     function="child").respond()
+            server = _mock_server(fn)
+            server.context.get_remote_endpoint = \
+                lambda name, external=False: \
+                f"http://127.0.0.1:{port}" if name == "child" else ""
+            resp = server.test("/do", body=["x"])
+            server.wait_for_completion()
+            assert resp == ["x", "s1", "remote"], resp
+            assert hits and hits[0][0] == "/do"
+            server.graph.shutdown()
+        finally:
+            httpd.shutdown()
