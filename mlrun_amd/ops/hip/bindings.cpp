@@ -7,7 +7,7 @@
 // Thin validation + launch layer; kernels live in kernels.hip.
 
 #include <torch/extension.h>
-#include <ATen/cuda/CUDAContext.h>
+#include <c10/hip/HIPStream.h>
 
 #include "kernels.h"
 
@@ -24,8 +24,10 @@
 #define CHECK_I32(x) \
   TORCH_CHECK((x).scalar_type() == torch::kInt32, #x " must be int32")
 
+// native ROCm stream accessor (c10::hip — no hipify pass needed; the
+// kernels in kernels.hip are pure CDNA4 HIP already)
 static void* stream() {
-  return (void*)at::cuda::getCurrentCUDAStream().stream();
+  return (void*)c10::hip::getCurrentHIPStream().stream();
 }
 
 // out <- rmsnorm(x [+ residual]); residual updated in-place when given
