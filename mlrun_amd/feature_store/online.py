@@ -27,28 +27,54 @@ from .feature_set import FeatureSet, parse_span
 class WindowRing:
     """Ring of per-period partials for one aggregated column.
 
-    ring[key, period, 0] = sum, [1] = count; min/max/first/last are
-    tracked in auxiliary per-key tensors (they don't decompose over
-    ring periods for sliding windows; kept as running values like the
-    reference's running aggregates)."""
+    All TEN reference ops (feature_set.py:715 — count/sum/sqr/max/min/
+    first/last/avg/stdvar/stddev) are computed over the SAME bucket-
+    quantized sliding window from per-period cells:
+    - ring[key, period] = (sum, count) f32; ring_sq mirrors sum-of-
+      squares for stdvar/stddev
+    - ring_mm[key, period] = (min, max) as ordered-f32 bits (integer
+      atomics give float order)
+    - ring_fl[key, period] = (first, last) as packed (ts<<32)|ordered
+      u64 — per-period first/last with intra-period ordering, so
+      window first = first of the oldest covered event, window last =
+      last of the newest (true sliding-window semantics, not the
+      running-value approximation round 1 fell back to)."""
 
     def __init__(self, period_seconds: int, n_periods: int, device="cpu",
                  capacity: int = 1024):
+        from .. import ops
+
         self.period_seconds = period_seconds
         self.n_periods = n_periods
         self.device = device
         self.capacity = capacity
         self.ring = torch.zeros(capacity, n_periods, 4, dtype=torch.float32,
                                 device=device)
-        # running (non-windowed) aggregates per key:
-        # [min, max, first, last, count, sum, sumsq]
-        self.running = torch.zeros(capacity, 7, dtype=torch.float32)
-        self.running[:, 0] = float("inf")
-        self.running[:, 1] = float("-inf")
-        # per-(key, period) sum of squares for stdvar/stddev windows
+        # per-(key, period) sum of squares for stdvar/stddev windows —
+        # f64: sumsq/n - mean^2 cancels catastrophically in f32
         self.ring_sq = torch.zeros(capacity, n_periods, 4,
-                                   dtype=torch.float32, device=device)
+                                   dtype=torch.float64, device=device)
+        self.ring_mm = self._empty_mm(capacity)
+        self.ring_fl = self._empty_fl(capacity)
         self.last_period = -1
+
+    def _empty_mm(self, capacity):
+        from .. import ops
+
+        mm = torch.empty(capacity, self.n_periods, 2, dtype=torch.int32,
+                         device=self.device)
+        mm[:, :, 0] = ops.MM_MIN_EMPTY
+        mm[:, :, 1] = ops.MM_MAX_EMPTY
+        return mm
+
+    def _empty_fl(self, capacity):
+        from .. import ops
+
+        fl = torch.empty(capacity, self.n_periods, 2, dtype=torch.int64,
+                         device=self.device)
+        fl[:, :, 0] = ops.FL_FIRST_EMPTY
+        fl[:, :, 1] = ops.FL_LAST_EMPTY
+        return fl
 
     def grow(self, capacity: int):
         if capacity <= self.capacity:
@@ -58,18 +84,29 @@ class WindowRing:
         new_ring[:self.capacity] = self.ring
         self.ring = new_ring
         new_sq = torch.zeros(capacity, self.n_periods, 4,
-                             dtype=torch.float32, device=self.device)
+                             dtype=torch.float64, device=self.device)
         new_sq[:self.capacity] = self.ring_sq
         self.ring_sq = new_sq
-        new_running = torch.zeros(capacity, 7, dtype=torch.float32)
-        new_running[:, 0] = float("inf")
-        new_running[:, 1] = float("-inf")
-        new_running[:self.capacity] = self.running
-        self.running = new_running
+        new_mm = self._empty_mm(capacity)
+        new_mm[:self.capacity] = self.ring_mm
+        self.ring_mm = new_mm
+        new_fl = self._empty_fl(capacity)
+        new_fl[:self.capacity] = self.ring_fl
+        self.ring_fl = new_fl
         self.capacity = capacity
 
+    def _reset_period(self, idx: int):
+        from .. import ops
+
+        self.ring[:, idx].zero_()
+        self.ring_sq[:, idx].zero_()
+        self.ring_mm[:, idx, 0] = ops.MM_MIN_EMPTY
+        self.ring_mm[:, idx, 1] = ops.MM_MAX_EMPTY
+        self.ring_fl[:, idx, 0] = ops.FL_FIRST_EMPTY
+        self.ring_fl[:, idx, 1] = ops.FL_LAST_EMPTY
+
     def _expire_old_periods(self, current_period: int):
-        """Zero ring cells for periods that wrapped since last ingest."""
+        """Reset ring cells for periods that wrapped since last ingest."""
         if self.last_period < 0:
             self.last_period = current_period
             return
@@ -77,18 +114,16 @@ class WindowRing:
         if gap <= 0:
             return
         if gap >= self.n_periods:
-            self.ring.zero_()
-            self.ring_sq.zero_()
+            for idx in range(self.n_periods):
+                self._reset_period(idx)
         else:
             for p in range(self.last_period + 1, current_period + 1):
-                idx = p % self.n_periods
-                self.ring[:, idx].zero_()
-                self.ring_sq[:, idx].zero_()
+                self._reset_period(p % self.n_periods)
         self.last_period = current_period
 
     def ingest(self, key_ids: torch.Tensor, values: torch.Tensor,
                timestamps: torch.Tensor):
-        """Batched fold of events into the ring (HIP kernel on GPU)."""
+        """Batched fold of events into the ring (HIP kernels on GPU)."""
         from .. import ops
 
         period_idx_abs = (timestamps.long() //
@@ -101,41 +136,21 @@ class WindowRing:
         keys32 = key_ids[fresh].to(torch.int32)
         pidx = (period_idx_abs[fresh] % self.n_periods).to(torch.int32)
         vals = values[fresh].to(torch.float32)
+        ts32 = timestamps[fresh].to(torch.int32)
         if self.ring.is_cuda:
-            keys32 = keys32.cuda(self.ring.device)
+            keys32 = keys32.to(self.ring.device)
             pidx = pidx.to(self.ring.device)
-            vals_dev = vals.to(self.ring.device)
-        else:
-            vals_dev = vals
-        ops.window_ingest(self.ring, keys32, vals_dev, pidx)
-        ops.window_ingest(self.ring_sq, keys32, vals_dev * vals_dev, pidx)
-        # running aggregates over the FULL batch (storey keeps running
-        # first/last/min/max even for late events) — fully vectorized
-        # via sorted segments + reduceat (no per-key python loop)
-        k = key_ids.numpy()
-        v = values.to(torch.float32).numpy()
-        order = np.argsort(k, kind="stable")
-        k_sorted, v_sorted = k[order], v[order]
-        uniq, starts = np.unique(k_sorted, return_index=True)
-        ends = np.append(starts[1:], len(k_sorted))
-        seg_min = np.minimum.reduceat(v_sorted, starts)
-        seg_max = np.maximum.reduceat(v_sorted, starts)
-        seg_sum = np.add.reduceat(v_sorted, starts)
-        seg_sumsq = np.add.reduceat(v_sorted * v_sorted, starts)
-        seg_count = (ends - starts).astype(np.float32)
-        run = self.running.numpy()
-        fresh_keys = run[uniq, 4] == 0
-        run[uniq, 0] = np.minimum(run[uniq, 0], seg_min)
-        run[uniq, 1] = np.maximum(run[uniq, 1], seg_max)
-        run[uniq[fresh_keys], 2] = v_sorted[starts[fresh_keys]]  # first
-        run[uniq, 3] = v_sorted[ends - 1]                        # last
-        run[uniq, 4] += seg_count
-        run[uniq, 5] += seg_sum
-        run[uniq, 6] += seg_sumsq
+            vals = vals.to(self.ring.device)
+            ts32 = ts32.to(self.ring.device)
+        ops.window_ingest(self.ring, keys32, vals, pidx)
+        ops.window_ingest(self.ring_sq, keys32, vals, pidx)  # squares
+        ops.window_ingest_mm(self.ring_mm, keys32, vals, pidx)
+        ops.window_ingest_fl(self.ring_fl, keys32, vals, ts32, pidx)
 
     def window_values(self, window_seconds: int, now_ts: float) -> dict:
         """Reduce the ring for one window length -> tensors keyed by op
-        (sum/count/avg/stdvar/stddev), each [capacity]."""
+        (all 10 reference ops), each [capacity]; min/max/first/last are
+        NaN where the window holds no events."""
         from .. import ops
 
         # cover every bucket intersecting [now - window, now]: the
@@ -148,19 +163,28 @@ class WindowRing:
         window_periods = min(window_periods, self.n_periods)
         current_period = int(now_ts // self.period_seconds)
         self._expire_old_periods(current_period)
-        out = ops.window_reduce(self.ring, window_periods,
-                                current_period % self.n_periods)
-        out_sq = ops.window_reduce(self.ring_sq, window_periods,
-                                   current_period % self.n_periods)
+        cur = current_period % self.n_periods
+        out = ops.window_reduce(self.ring, window_periods, cur)
+        out_sq = ops.window_reduce(self.ring_sq, window_periods, cur)
+        mmfl = ops.window_reduce_mmfl(self.ring_mm, self.ring_fl,
+                                      window_periods, cur)
         if out.is_cuda:
             out = out.cpu()
             out_sq = out_sq.cpu()
+            mmfl = mmfl.cpu()
         total, count, avg = out[:, 0], out[:, 1], out[:, 2]
+        # stdvar/stddev entirely in f64: sumsq, count AND sum come from
+        # the f64 ring (f32 mean² cancels against sumsq/n)
         sumsq = out_sq[:, 0]
-        n = count.clamp(min=1.0)
-        var = (sumsq / n - (total / n) ** 2).clamp(min=0.0)
-        unbiased = torch.where(count > 1, var * count / (count - 1).clamp(
-            min=1.0), torch.zeros_like(var))
+        n = out_sq[:, 1].clamp(min=1.0)
+        mean = out_sq[:, 2] / n
+        var = (sumsq / n - mean * mean).clamp(min=0.0)
+        unbiased = torch.where(
+            count > 1,
+            (var * n / (n - 1.0).clamp(min=1.0)).to(torch.float32),
+            torch.zeros_like(total))
+        empty = count == 0
+        nan = torch.full_like(total, float("nan"))
         return {
             "sum": total,
             "count": count,
@@ -168,6 +192,10 @@ class WindowRing:
             "sqr": sumsq,
             "stdvar": unbiased,
             "stddev": unbiased.sqrt(),
+            "min": torch.where(empty, nan, mmfl[:, 0]),
+            "max": torch.where(empty, nan, mmfl[:, 1]),
+            "first": torch.where(empty, nan, mmfl[:, 2]),
+            "last": torch.where(empty, nan, mmfl[:, 3]),
         }
 
 
@@ -288,12 +316,7 @@ class OnlineTable:
                 if ckey not in reduced_cache:
                     reduced_cache[ckey] = ring.window_values(
                         parse_span(window), now_ts)
-                vals = reduced_cache[ckey]
-                if op in vals:
-                    col = vals[op].numpy()[safe]
-                else:
-                    mapping = {"min": 0, "max": 1, "first": 2, "last": 3}
-                    col = ring.running.numpy()[safe, mapping[op]]
+                col = reduced_cache[ckey][op].numpy()[safe]
                 out[:, j] = np.where(known, col, np.nan)
             return out
 
@@ -319,24 +342,16 @@ class OnlineTable:
                 if latest:
                     record.update(latest)
                 for agg in fset.spec.aggregations:
-                    ring = self.rings[agg.name]
                     for window in agg.windows:
                         vals = reduced[(agg.name, window)]
                         for op in agg.operations:
                             name = f"{agg.name}_{op}_{window}"
                             if idx is None:
                                 record[name] = None
-                            elif op in vals:
-                                record[name] = float(vals[op][idx])
-                            else:  # min/max/first/last: running values
-                                run = ring.running[idx]
-                                mapping = {"min": 0, "max": 1, "first": 2,
-                                           "last": 3}
-                                value = float(run[mapping[op]])
-                                if op in ("min", "max") and \
-                                        float(run[4]) == 0:
-                                    value = None
-                                record[name] = value
+                            else:
+                                value = float(vals[op][idx])
+                                record[name] = None \
+                                    if value != value else value
                 out.append(record)
             return out
 

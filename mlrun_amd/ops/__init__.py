@@ -514,6 +514,9 @@ def tree_ensemble_predict(features: torch.Tensor, nodes: dict,
 
 
 def window_ingest(ring, keys, values, period_idx):
+    """Fold (sum, count) period cells.  An f64 ring accumulates the
+    SQUARES of the values instead (the stdvar/stddev ring — f32
+    sum-of-squares cancels catastrophically)."""
     if ring.is_cuda:
         _require_hip().window_ingest(ring, keys, values, period_idx)
         return ring
@@ -523,9 +526,123 @@ def window_ingest(ring, keys, values, period_idx):
     flat_idx = keys.numpy().astype(np.int64) * n_periods + \
         (period_idx.numpy().astype(np.int64) % n_periods)
     view = ring.numpy().reshape(-1, 4)
-    np.add.at(view[:, 0], flat_idx, values.numpy())
+    vals = values.numpy()
+    if ring.dtype == torch.float64:
+        vals = vals.astype(np.float64)
+        np.add.at(view[:, 0], flat_idx, vals * vals)
+        np.add.at(view[:, 1], flat_idx, 1.0)
+        np.add.at(view[:, 2], flat_idx, vals)  # f64 sum
+        return ring
+    np.add.at(view[:, 0], flat_idx, vals)
     np.add.at(view[:, 1], flat_idx, 1.0)
     return ring
+
+
+# ordered-f32 transform (monotone f32 -> u32 so integer min/max give
+# float order) — numpy mirror of the kernel's f32_to_ordered
+def _f32_to_ordered_np(values):
+    import numpy as np
+
+    bits = values.astype(np.float32).view(np.uint32)
+    return np.where(bits & 0x80000000, ~bits, bits | 0x80000000)
+
+
+def _ordered_to_f32_np(keys):
+    import numpy as np
+
+    keys = keys.astype(np.uint32)
+    bits = np.where(keys & 0x80000000, keys & 0x7FFFFFFF, ~keys)
+    return bits.astype(np.uint32).view(np.float32)
+
+
+MM_MIN_EMPTY = -1          # int32 view of 0xFFFFFFFF
+MM_MAX_EMPTY = 0
+FL_FIRST_EMPTY = -1        # int64 view of 0xFFFFFFFFFFFFFFFF
+FL_LAST_EMPTY = 0
+
+
+def window_ingest_mm(ring_mm, keys, values, period_idx):
+    """Fold per-period MIN/MAX cells (ring_mm int32 = ordered-f32
+    bits; empty = MM_MIN_EMPTY/MM_MAX_EMPTY)."""
+    if ring_mm.is_cuda:
+        _require_hip().window_ingest_mm(ring_mm, keys, values, period_idx)
+        return ring_mm
+    import numpy as np
+
+    n_periods = ring_mm.shape[1]
+    flat = keys.numpy().astype(np.int64) * n_periods + \
+        (period_idx.numpy().astype(np.int64) % n_periods)
+    ov = _f32_to_ordered_np(values.numpy())
+    view = ring_mm.numpy().view(np.uint32).reshape(-1, 2)
+    np.minimum.at(view[:, 0], flat, ov)
+    np.maximum.at(view[:, 1], flat, ov)
+    return ring_mm
+
+
+def window_ingest_fl(ring_fl, keys, values, timestamps, period_idx):
+    """Fold per-period FIRST/LAST cells (ring_fl int64 packed
+    (ts << 32) | ordered-f32; empty = FL_FIRST_EMPTY/FL_LAST_EMPTY)."""
+    if ring_fl.is_cuda:
+        _require_hip().window_ingest_fl(ring_fl, keys, values,
+                                        timestamps, period_idx)
+        return ring_fl
+    import numpy as np
+
+    n_periods = ring_fl.shape[1]
+    flat = keys.numpy().astype(np.int64) * n_periods + \
+        (period_idx.numpy().astype(np.int64) % n_periods)
+    pack = (timestamps.numpy().astype(np.uint64) << np.uint64(32)) | \
+        _f32_to_ordered_np(values.numpy()).astype(np.uint64)
+    view = ring_fl.numpy().view(np.uint64).reshape(-1, 2)
+    np.minimum.at(view[:, 0], flat, pack)
+    np.maximum.at(view[:, 1], flat, pack)
+    return ring_fl
+
+
+def window_reduce_mmfl(ring_mm, ring_fl, window_periods: int,
+                       current_period: int, out: torch.Tensor = None):
+    """Reduce min/max/first/last over the covered window cells ->
+    [keys, 4] f32 (min, max, first, last); empty windows yield 0 —
+    callers mask by the count from window_reduce."""
+    if ring_mm.is_cuda:
+        ops = _require_hip()
+        if out is None:
+            out = torch.empty(ring_mm.shape[0], 4, dtype=torch.float32,
+                              device=ring_mm.device)
+        ops.window_reduce_mmfl(out, ring_mm, ring_fl, window_periods,
+                               current_period)
+        return out
+    import numpy as np
+
+    n_keys, n_periods, _ = ring_mm.shape
+    cols = [(current_period - w) % n_periods
+            for w in range(window_periods)]
+    mm = ring_mm.numpy().view(np.uint32)
+    fl = ring_fl.numpy().view(np.uint64)
+    omin = mm[:, cols, 0].min(axis=1)
+    omax = mm[:, cols, 1].max(axis=1)
+    first_pack = fl[:, cols, 0].min(axis=1)
+    last_pack = fl[:, cols, 1].max(axis=1)
+    result = np.zeros((n_keys, 4), dtype=np.float32)
+    has_min = omin != 0xFFFFFFFF
+    has_max = omax != 0
+    result[:, 0] = np.where(has_min, _ordered_to_f32_np(omin), 0.0)
+    result[:, 1] = np.where(has_max, _ordered_to_f32_np(omax), 0.0)
+    has_first = first_pack != np.uint64(0xFFFFFFFFFFFFFFFF)
+    has_last = last_pack != 0
+    result[:, 2] = np.where(
+        has_first,
+        _ordered_to_f32_np((first_pack & np.uint64(0xFFFFFFFF)
+                            ).astype(np.uint32)), 0.0)
+    result[:, 3] = np.where(
+        has_last,
+        _ordered_to_f32_np((last_pack & np.uint64(0xFFFFFFFF)
+                            ).astype(np.uint32)), 0.0)
+    tensor = torch.from_numpy(result)
+    if out is not None:
+        out.copy_(tensor)
+        return out
+    return tensor
 
 
 def window_reduce(ring, window_periods: int, current_period: int,
@@ -533,20 +650,23 @@ def window_reduce(ring, window_periods: int, current_period: int,
     if ring.is_cuda:
         ops = _require_hip()
         if out is None:
-            out = torch.empty(ring.shape[0], 4, dtype=torch.float32,
+            out = torch.empty(ring.shape[0], 4, dtype=ring.dtype,
                               device=ring.device)
         ops.window_reduce(out, ring, window_periods, current_period)
         return out
     n_keys, n_periods, _ = ring.shape
-    result = torch.zeros(n_keys, 4, dtype=torch.float32)
+    result = torch.zeros(n_keys, 4, dtype=ring.dtype)
     for w in range(window_periods):
         p = (current_period - w) % n_periods
         result[:, 0] += ring[:, p, 0]
         result[:, 1] += ring[:, p, 1]
-    counts = result[:, 1].clamp(min=1.0)
-    result[:, 2] = result[:, 0] / counts
-    result[:, 2] = torch.where(result[:, 1] > 0, result[:, 2],
-                               torch.zeros_like(result[:, 2]))
+        if ring.dtype == torch.float64:
+            result[:, 2] += ring[:, p, 2]  # f64 sum slot
+    if ring.dtype != torch.float64:
+        counts = result[:, 1].clamp(min=1.0)
+        result[:, 2] = result[:, 0] / counts
+        result[:, 2] = torch.where(result[:, 1] > 0, result[:, 2],
+                                   torch.zeros_like(result[:, 2]))
     if out is not None:
         out.copy_(result)
         return out

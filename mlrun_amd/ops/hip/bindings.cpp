@@ -367,10 +367,18 @@ void tree_ensemble(torch::Tensor out, torch::Tensor features,
 
 void window_ingest(torch::Tensor ring, torch::Tensor keys,
                    torch::Tensor values, torch::Tensor period_idx) {
-  CHECK_DEV(ring); CHECK_CONTIG(ring); CHECK_F32(ring);
+  CHECK_DEV(ring); CHECK_CONTIG(ring);
   CHECK_I32(keys); CHECK_F32(values); CHECK_I32(period_idx);
   TORCH_CHECK(ring.dim() == 3 && ring.size(2) == 4,
               "ring must be [keys, periods, 4]");
+  if (ring.scalar_type() == torch::kFloat64) {
+    // f64 ring: sum-of-SQUARES accumulation (stdvar precision)
+    launch_window_ingest64(ring.data_ptr(), keys.data_ptr(),
+                           values.data_ptr(), period_idx.data_ptr(),
+                           keys.numel(), (int)ring.size(1), stream());
+    return;
+  }
+  CHECK_F32(ring);
   launch_window_ingest(ring.data_ptr(), keys.data_ptr(), values.data_ptr(),
                        period_idx.data_ptr(), keys.numel(),
                        (int)ring.size(1), stream());
@@ -378,11 +386,64 @@ void window_ingest(torch::Tensor ring, torch::Tensor keys,
 
 void window_reduce(torch::Tensor out, torch::Tensor ring,
                    int64_t window_periods, int64_t current_period) {
-  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_F32(out);
-  CHECK_DEV(ring); CHECK_CONTIG(ring); CHECK_F32(ring);
+  CHECK_DEV(out); CHECK_CONTIG(out);
+  CHECK_DEV(ring); CHECK_CONTIG(ring);
+  if (ring.scalar_type() == torch::kFloat64) {
+    TORCH_CHECK(out.scalar_type() == torch::kFloat64,
+                "out must be f64 for an f64 ring");
+    launch_window_reduce64(out.data_ptr(), ring.data_ptr(),
+                           (int)ring.size(0), (int)ring.size(1),
+                           (int)window_periods, (int)current_period,
+                           stream());
+    return;
+  }
+  CHECK_F32(out); CHECK_F32(ring);
   launch_window_reduce(out.data_ptr(), ring.data_ptr(), (int)ring.size(0),
                        (int)ring.size(1), (int)window_periods,
                        (int)current_period, stream());
+}
+
+void window_ingest_mm(torch::Tensor ring_mm, torch::Tensor keys,
+                      torch::Tensor values, torch::Tensor period_idx) {
+  CHECK_DEV(ring_mm); CHECK_CONTIG(ring_mm);
+  TORCH_CHECK(ring_mm.scalar_type() == torch::kInt32,
+              "ring_mm must be int32 (ordered-f32 bits)");
+  CHECK_I32(keys); CHECK_F32(values); CHECK_I32(period_idx);
+  TORCH_CHECK(ring_mm.dim() == 3 && ring_mm.size(2) == 2,
+              "ring_mm must be [keys, periods, 2]");
+  launch_window_ingest_mm(ring_mm.data_ptr(), keys.data_ptr(),
+                          values.data_ptr(), period_idx.data_ptr(),
+                          keys.numel(), (int)ring_mm.size(1), stream());
+}
+
+void window_ingest_fl(torch::Tensor ring_fl, torch::Tensor keys,
+                      torch::Tensor values, torch::Tensor timestamps,
+                      torch::Tensor period_idx) {
+  CHECK_DEV(ring_fl); CHECK_CONTIG(ring_fl);
+  TORCH_CHECK(ring_fl.scalar_type() == torch::kInt64,
+              "ring_fl must be int64 (packed ts|ordered-f32)");
+  CHECK_I32(keys); CHECK_F32(values); CHECK_I32(timestamps);
+  CHECK_I32(period_idx);
+  TORCH_CHECK(ring_fl.dim() == 3 && ring_fl.size(2) == 2,
+              "ring_fl must be [keys, periods, 2]");
+  launch_window_ingest_fl(ring_fl.data_ptr(), keys.data_ptr(),
+                          values.data_ptr(), timestamps.data_ptr(),
+                          period_idx.data_ptr(), keys.numel(),
+                          (int)ring_fl.size(1), stream());
+}
+
+void window_reduce_mmfl(torch::Tensor out, torch::Tensor ring_mm,
+                        torch::Tensor ring_fl, int64_t window_periods,
+                        int64_t current_period) {
+  CHECK_DEV(out); CHECK_CONTIG(out); CHECK_F32(out);
+  CHECK_DEV(ring_mm); CHECK_CONTIG(ring_mm);
+  CHECK_DEV(ring_fl); CHECK_CONTIG(ring_fl);
+  TORCH_CHECK(out.dim() == 2 && out.size(1) == 4,
+              "out must be [keys, 4]");
+  launch_window_reduce_mmfl(out.data_ptr(), ring_mm.data_ptr(),
+                            ring_fl.data_ptr(), (int)ring_mm.size(0),
+                            (int)ring_mm.size(1), (int)window_periods,
+                            (int)current_period, stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -424,4 +485,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "feature-store window ring ingest (f32)");
   m.def("window_reduce", &window_reduce,
         "feature-store window ring reduce (f32)");
+  m.def("window_ingest_mm", &window_ingest_mm,
+        "per-period min/max ingest (ordered-f32 atomics)");
+  m.def("window_ingest_fl", &window_ingest_fl,
+        "per-period first/last ingest (packed ts|value u64 atomics)");
+  m.def("window_reduce_mmfl", &window_reduce_mmfl,
+        "window min/max/first/last reduce over period cells");
 }

@@ -93,6 +93,25 @@ void launch_window_ingest(void* ring, const void* keys, const void* values,
                           const void* period_idx, long long n_events,
                           int n_periods, void* stream);
 
+void launch_window_ingest64(void* ring, const void* keys,
+                            const void* values, const void* period_idx,
+                            long long n_events, int n_periods,
+                            void* stream);
+void launch_window_reduce64(void* out, const void* ring, int n_keys,
+                            int n_periods, int window_periods,
+                            int current_period, void* stream);
+void launch_window_ingest_mm(void* ring_mm, const void* keys,
+                             const void* values, const void* period_idx,
+                             long long n_events, int n_periods,
+                             void* stream);
+void launch_window_ingest_fl(void* ring_fl, const void* keys,
+                             const void* values, const void* timestamps,
+                             const void* period_idx, long long n_events,
+                             int n_periods, void* stream);
+void launch_window_reduce_mmfl(void* out, const void* ring_mm,
+                               const void* ring_fl, int n_keys,
+                               int n_periods, int window_periods,
+                               int current_period, void* stream);
 void launch_window_reduce(void* out, const void* ring, int n_keys,
                           int n_periods, int window_periods,
                           int current_period, void* stream);

@@ -1770,6 +1770,204 @@ __global__ void window_reduce_kernel(float* __restrict__ out,
   out[key * 4 + 3] = 0.f;
 }
 
+// ---------------------------------------------------------------------
+// Per-period min/max cells: float atomic min/max via the ordered-int
+// transform (monotone map f32 -> u32 so unsigned atomics give float
+// order).  Per-period cells make min/max DECOMPOSABLE for the
+// bucket-quantized sliding windows served by window_reduce — unlike
+// the running aggregates the round-1 path fell back to.
+// ring_mm: [n_keys, n_periods, 2] u32 (ordered-min, ordered-max),
+// empty cells hold 0xFFFFFFFF / 0x00000000.
+// ---------------------------------------------------------------------
+__device__ __forceinline__ unsigned int f32_to_ordered(float v) {
+  unsigned int bits = __float_as_uint(v);
+  return (bits & 0x80000000u) ? ~bits : (bits | 0x80000000u);
+}
+
+__device__ __forceinline__ float ordered_to_f32(unsigned int key) {
+  unsigned int bits = (key & 0x80000000u) ? (key & 0x7FFFFFFFu) : ~key;
+  return __uint_as_float(bits);
+}
+
+__global__ void window_ingest_mm_kernel(unsigned int* __restrict__ ring_mm,
+                                        const int* __restrict__ keys,
+                                        const float* __restrict__ values,
+                                        const int* __restrict__ period_idx,
+                                        long long n_events, int n_periods) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n_events; i += stride) {
+    const int key = keys[i];
+    const int p = period_idx[i] % n_periods;
+    unsigned int* cell = ring_mm + ((size_t)key * n_periods + p) * 2;
+    const unsigned int ov = f32_to_ordered(values[i]);
+    atomicMin(cell + 0, ov);
+    atomicMax(cell + 1, ov);
+  }
+}
+
+// first/last per period: packed u64 = (ts << 32) | ordered(value);
+// atomicMin gives the earliest event (ties broken by value order),
+// atomicMax the latest.  ts fits 31 bits until 2038 (unix seconds).
+// ring_fl: [n_keys, n_periods, 2] u64, empty = (ULLONG_MAX, 0).
+__global__ void window_ingest_fl_kernel(
+    unsigned long long* __restrict__ ring_fl,
+    const int* __restrict__ keys, const float* __restrict__ values,
+    const int* __restrict__ timestamps, const int* __restrict__ period_idx,
+    long long n_events, int n_periods) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n_events; i += stride) {
+    const int key = keys[i];
+    const int p = period_idx[i] % n_periods;
+    unsigned long long* cell = ring_fl + ((size_t)key * n_periods + p) * 2;
+    const unsigned long long pack =
+        ((unsigned long long)(unsigned int)timestamps[i] << 32) |
+        f32_to_ordered(values[i]);
+    atomicMin(cell + 0, pack);  // first = earliest pack
+    atomicMax(cell + 1, pack);  // last = latest pack
+  }
+}
+
+// reduce min/max + first/last over the covered window cells.
+// out_mmfl: [n_keys, 4] f32 (min, max, first, last); count==0 rows are
+// left as written (caller masks by count from window_reduce).
+__global__ void window_reduce_mmfl_kernel(
+    float* __restrict__ out, const unsigned int* __restrict__ ring_mm,
+    const unsigned long long* __restrict__ ring_fl, int n_keys,
+    int n_periods, int window_periods, int current_period) {
+  int key = blockIdx.x * blockDim.x + threadIdx.x;
+  if (key >= n_keys) return;
+  unsigned int omin = 0xFFFFFFFFu, omax = 0u;
+  unsigned long long first_pack = 0xFFFFFFFFFFFFFFFFull, last_pack = 0ull;
+  for (int w = 0; w < window_periods; ++w) {
+    int p = (current_period - w) % n_periods;
+    if (p < 0) p += n_periods;
+    const unsigned int* mm = ring_mm + ((size_t)key * n_periods + p) * 2;
+    omin = min(omin, mm[0]);
+    omax = max(omax, mm[1]);
+    const unsigned long long* fl =
+        ring_fl + ((size_t)key * n_periods + p) * 2;
+    first_pack = min(first_pack, fl[0]);
+    last_pack = max(last_pack, fl[1]);
+  }
+  float* row = out + (size_t)key * 4;
+  row[0] = omin == 0xFFFFFFFFu ? 0.f : ordered_to_f32(omin);
+  row[1] = omax == 0u ? 0.f : ordered_to_f32(omax);
+  row[2] = first_pack == 0xFFFFFFFFFFFFFFFFull
+               ? 0.f
+               : ordered_to_f32((unsigned int)(first_pack & 0xFFFFFFFFull));
+  row[3] = last_pack == 0ull
+               ? 0.f
+               : ordered_to_f32((unsigned int)(last_pack & 0xFFFFFFFFull));
+}
+
+// f64 variants for the sum-of-squares ring: the stdvar formula
+// sumsq/n - mean^2 cancels catastrophically in f32 when values are
+// large and var is small — f64 accumulation keeps ~16 digits
+// (atomicAdd(double) is native on CDNA4 HBM).
+__global__ void window_ingest64_kernel(double* __restrict__ ring,
+                                       const int* __restrict__ keys,
+                                       const float* __restrict__ values,
+                                       const int* __restrict__ period_idx,
+                                       long long n_events, int n_periods) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i < n_events; i += stride) {
+    const int key = keys[i];
+    const int p = period_idx[i] % n_periods;
+    double* cell = ring + ((size_t)key * n_periods + p) * 4;
+    const double v = (double)values[i];
+    atomicAdd(cell + 0, v * v);  // sum of squares
+    atomicAdd(cell + 1, 1.0);    // count
+    atomicAdd(cell + 2, v);      // f64 sum (variance needs f64 mean)
+  }
+}
+
+__global__ void window_reduce64_kernel(double* __restrict__ out,
+                                       const double* __restrict__ ring,
+                                       int n_keys, int n_periods,
+                                       int window_periods,
+                                       int current_period) {
+  int key = blockIdx.x * blockDim.x + threadIdx.x;
+  if (key >= n_keys) return;
+  double sumsq = 0.0, count = 0.0, sum = 0.0;
+  for (int w = 0; w < window_periods; ++w) {
+    int p = (current_period - w) % n_periods;
+    if (p < 0) p += n_periods;
+    const double* cell = ring + ((size_t)key * n_periods + p) * 4;
+    sumsq += cell[0];
+    count += cell[1];
+    sum += cell[2];
+  }
+  out[key * 4 + 0] = sumsq;
+  out[key * 4 + 1] = count;
+  out[key * 4 + 2] = sum;
+  out[key * 4 + 3] = 0.0;
+}
+
+void launch_window_ingest64(void* ring, const void* keys,
+                            const void* values, const void* period_idx,
+                            long long n_events, int n_periods,
+                            void* stream) {
+  long long blocks = (n_events + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(window_ingest64_kernel, dim3((int)blocks), dim3(256),
+                     0, (hipStream_t)stream, (double*)ring,
+                     (const int*)keys, (const float*)values,
+                     (const int*)period_idx, n_events, n_periods);
+}
+
+void launch_window_reduce64(void* out, const void* ring, int n_keys,
+                            int n_periods, int window_periods,
+                            int current_period, void* stream) {
+  int blocks = (n_keys + 255) / 256;
+  hipLaunchKernelGGL(window_reduce64_kernel, dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, (double*)out,
+                     (const double*)ring, n_keys, n_periods,
+                     window_periods, current_period);
+}
+
+void launch_window_ingest_mm(void* ring_mm, const void* keys,
+                             const void* values, const void* period_idx,
+                             long long n_events, int n_periods,
+                             void* stream) {
+  long long blocks = (n_events + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(window_ingest_mm_kernel, dim3((int)blocks), dim3(256),
+                     0, (hipStream_t)stream, (unsigned int*)ring_mm,
+                     (const int*)keys, (const float*)values,
+                     (const int*)period_idx, n_events, n_periods);
+}
+
+void launch_window_ingest_fl(void* ring_fl, const void* keys,
+                             const void* values, const void* timestamps,
+                             const void* period_idx, long long n_events,
+                             int n_periods, void* stream) {
+  long long blocks = (n_events + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(window_ingest_fl_kernel, dim3((int)blocks), dim3(256),
+                     0, (hipStream_t)stream,
+                     (unsigned long long*)ring_fl, (const int*)keys,
+                     (const float*)values, (const int*)timestamps,
+                     (const int*)period_idx, n_events, n_periods);
+}
+
+void launch_window_reduce_mmfl(void* out, const void* ring_mm,
+                               const void* ring_fl, int n_keys,
+                               int n_periods, int window_periods,
+                               int current_period, void* stream) {
+  int blocks = (n_keys + 255) / 256;
+  hipLaunchKernelGGL(window_reduce_mmfl_kernel, dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, (float*)out,
+                     (const unsigned int*)ring_mm,
+                     (const unsigned long long*)ring_fl, n_keys, n_periods,
+                     window_periods, current_period);
+}
+
 void launch_window_ingest(void* ring, const void* keys, const void* values,
                           const void* period_idx, long long n_events,
                           int n_periods, void* stream) {

@@ -210,6 +210,93 @@ class TestWindowAgg:
         assert torch.allclose(ref[:, :2], got[:, :2], atol=1e-2, rtol=1e-3)
         assert torch.allclose(ref[:, 2], got[:, 2], atol=1e-3, rtol=1e-3)
 
+    def test_ingest_reduce_f64_sq(self):
+        """f64 sumsq/count/sum ring (stdvar precision path)."""
+        from mlrun_amd import ops
+
+        torch.manual_seed(14)
+        n_keys, n_periods = 32, 6
+        n_events = 3000
+        keys = torch.randint(0, n_keys, (n_events,), dtype=torch.int32)
+        values = torch.randn(n_events) * 50 + 60  # large mean, small var
+        period_idx = torch.randint(0, n_periods, (n_events,),
+                                   dtype=torch.int32)
+        ring_ref = torch.zeros(n_keys, n_periods, 4, dtype=torch.float64)
+        ops.window_ingest(ring_ref, keys, values, period_idx)
+        ref = ops.window_reduce(ring_ref, window_periods=3,
+                                current_period=4)
+        ring_gpu = torch.zeros(n_keys, n_periods, 4, dtype=torch.float64,
+                               device="cuda")
+        ops.window_ingest(ring_gpu, keys.cuda(), values.cuda(),
+                          period_idx.cuda())
+        got = ops.window_reduce(ring_gpu, window_periods=3,
+                                current_period=4).cpu()
+        assert torch.allclose(ref, got, atol=1e-6, rtol=1e-9)
+
+    def test_min_max_first_last_gpu_matches_cpu(self):
+        """GPU per-period min/max + packed first/last cells vs the
+        numpy fallback — all four windowed ops bit-match."""
+        from mlrun_amd import ops
+
+        torch.manual_seed(15)
+        n_keys, n_periods = 48, 8
+        n_events = 4000
+        keys = torch.randint(0, n_keys, (n_events,), dtype=torch.int32)
+        values = torch.randn(n_events) * 10
+        ts = torch.randint(1_000_000, 2_000_000, (n_events,),
+                           dtype=torch.int32)
+        period_idx = torch.randint(0, n_periods, (n_events,),
+                                   dtype=torch.int32)
+
+        mm_ref = torch.empty(n_keys, n_periods, 2, dtype=torch.int32)
+        mm_ref[:, :, 0] = ops.MM_MIN_EMPTY
+        mm_ref[:, :, 1] = ops.MM_MAX_EMPTY
+        fl_ref = torch.empty(n_keys, n_periods, 2, dtype=torch.int64)
+        fl_ref[:, :, 0] = ops.FL_FIRST_EMPTY
+        fl_ref[:, :, 1] = ops.FL_LAST_EMPTY
+        ops.window_ingest_mm(mm_ref, keys, values, period_idx)
+        ops.window_ingest_fl(fl_ref, keys, values, ts, period_idx)
+        ref = ops.window_reduce_mmfl(mm_ref, fl_ref, 5, 6)
+
+        mm_gpu = mm_ref.clone().zero_().cuda()
+        mm_gpu[:, :, 0] = ops.MM_MIN_EMPTY
+        mm_gpu[:, :, 1] = ops.MM_MAX_EMPTY
+        fl_gpu = fl_ref.clone().zero_().cuda()
+        fl_gpu[:, :, 0] = ops.FL_FIRST_EMPTY
+        fl_gpu[:, :, 1] = ops.FL_LAST_EMPTY
+        ops.window_ingest_mm(mm_gpu, keys.cuda(), values.cuda(),
+                             period_idx.cuda())
+        ops.window_ingest_fl(fl_gpu, keys.cuda(), values.cuda(),
+                             ts.cuda(), period_idx.cuda())
+        got = ops.window_reduce_mmfl(mm_gpu, fl_gpu, 5, 6).cpu()
+        assert torch.equal(ref, got)
+
+    def test_window_ring_all_ops_gpu_matches_cpu(self):
+        """End-to-end WindowRing on GPU vs CPU: all 10 ops agree."""
+        import math
+
+        from mlrun_amd.feature_store.online import WindowRing
+
+        torch.manual_seed(16)
+        n = 2000
+        keys = torch.randint(0, 20, (n,))
+        values = torch.randn(n) * 30 + 100
+        ts = torch.sort(torch.randint(0, 3600, (n,)).float()).values
+        cpu_ring = WindowRing(60, 61, device="cpu", capacity=32)
+        gpu_ring = WindowRing(60, 61, device="cuda:0", capacity=32)
+        cpu_ring.ingest(keys, values, ts)
+        gpu_ring.ingest(keys, values, ts)
+        now = float(ts.max())
+        got_cpu = cpu_ring.window_values(1800, now)
+        got_gpu = gpu_ring.window_values(1800, now)
+        for op in ("count", "sum", "avg", "sqr", "stdvar", "stddev",
+                   "min", "max", "first", "last"):
+            a, b = got_cpu[op], got_gpu[op]
+            mask = ~torch.isnan(a)
+            assert torch.equal(torch.isnan(a), torch.isnan(b)), op
+            assert torch.allclose(a[mask], b[mask], atol=1e-2,
+                                  rtol=1e-4), (op, a[mask], b[mask])
+
 
 @requires_gpu
 class TestFusedDecodeOps:

@@ -25,19 +25,19 @@ EVENT = st.tuples(
 def brute_force(events, period, n_periods, window_seconds, now_ts):
     """Bucket-quantized sliding window over raw events (the semantics
     window_values documents): include every bucket intersecting
-    [now - window, now], drop events older than the ring span."""
+    [now - window, now], drop events older than the ring span.
+    Returns per-key lists of (ts, value) inside the window."""
     current = now_ts // period
     window_periods = max(window_seconds // period, 1) + 1
-    sums, counts = {}, {}
+    covered = {}
     for key, value, ts in events:
         bucket = ts // period
         if bucket <= current - n_periods:
             continue  # outside the ring span
         if bucket > current or bucket <= current - window_periods:
             continue
-        sums[key] = sums.get(key, 0.0) + value
-        counts[key] = counts.get(key, 0) + 1
-    return sums, counts
+        covered.setdefault(key, []).append((ts, value))
+    return covered
 
 
 class TestWindowRingProperties:
@@ -58,15 +58,52 @@ class TestWindowRingProperties:
         ring.ingest(keys, vals, ts)
         now_ts = float(max(e[2] for e in events))
         got = ring.window_values(window_seconds, now_ts)
-        sums, counts = brute_force(events, period, n_periods,
-                                   window_seconds, int(now_ts))
+        covered = brute_force(events, period, n_periods,
+                              window_seconds, int(now_ts))
         for key in range(8):
-            expect_sum = sums.get(key, 0.0)
-            expect_count = counts.get(key, 0)
+            inside = covered.get(key, [])
+            values = [v for _, v in inside]
+            expect_count = len(values)
             assert got["count"][key].item() == expect_count, (
                 key, events)
-            assert math.isclose(got["sum"][key].item(), expect_sum,
+            assert math.isclose(got["sum"][key].item(), sum(values),
                                 rel_tol=1e-4, abs_tol=1e-3), (key, events)
+            if not values:
+                for op in ("min", "max", "first", "last"):
+                    assert math.isnan(got[op][key].item()), (op, key)
+                continue
+            # ALL TEN reference ops against the brute-force oracle
+            assert math.isclose(got["min"][key].item(), min(values),
+                                rel_tol=1e-6, abs_tol=1e-6)
+            assert math.isclose(got["max"][key].item(), max(values),
+                                rel_tol=1e-6, abs_tol=1e-6)
+            assert math.isclose(
+                got["sqr"][key].item(), sum(v * v for v in values),
+                rel_tol=1e-4, abs_tol=1e-2)
+            assert math.isclose(
+                got["avg"][key].item(), sum(values) / len(values),
+                rel_tol=1e-4, abs_tol=1e-3)
+            # first/last by event time (stable ties resolved by the
+            # packed value order — pick matching oracle values)
+            first_ts = min(t for t, _ in inside)
+            last_ts = max(t for t, _ in inside)
+            first_candidates = [v for t, v in inside if t == first_ts]
+            last_candidates = [v for t, v in inside if t == last_ts]
+            assert any(math.isclose(got["first"][key].item(), v,
+                                    rel_tol=1e-6, abs_tol=1e-6)
+                       for v in first_candidates), (key, events)
+            assert any(math.isclose(got["last"][key].item(), v,
+                                    rel_tol=1e-6, abs_tol=1e-6)
+                       for v in last_candidates), (key, events)
+            if len(values) > 1:
+                mean = sum(values) / len(values)
+                var = sum((v - mean) ** 2 for v in values) / \
+                    (len(values) - 1)
+                assert math.isclose(got["stdvar"][key].item(), var,
+                                    rel_tol=1e-3, abs_tol=1e-2)
+                assert math.isclose(got["stddev"][key].item(),
+                                    math.sqrt(var),
+                                    rel_tol=1e-3, abs_tol=1e-2)
 
     @settings(max_examples=30, deadline=None)
     @given(st.lists(st.floats(min_value=-50, max_value=50,
