@@ -210,6 +210,14 @@ class OnlineTable:
         self.latest: typing.Dict[typing.Any, dict] = {}
         self.rings: typing.Dict[str, WindowRing] = {}
         self._lock = threading.Lock()
+        # lookup fast path: key->id via a pandas Index (C hash table;
+        # rebuilt lazily when ingest adds keys) + ring-reduce cache
+        # keyed by (agg, window, period, version) — reduces happen once
+        # per ingest generation, not once per get batch
+        self._key_lookup = None
+        self._version = 0
+        self._reduce_cache: dict = {}
+        self._name_map: typing.Optional[dict] = None
         for agg in feature_set.spec.aggregations:
             period = agg.period or agg.windows[0]
             period_s = parse_span(period)
@@ -230,17 +238,55 @@ class OnlineTable:
         codes, uniques = pd.factorize(pd.Index(keys), sort=False)
         # map each unique key through (and extend) the dictionary once
         lut = np.empty(len(uniques), dtype=np.int64)
+        grew = False
         for i, key in enumerate(uniques):
             idx = self.key_index.get(key)
             if idx is None:
                 idx = len(self.key_index)
                 self.key_index[key] = idx
+                grew = True
             lut[i] = idx
+        if grew:
+            self._key_lookup = None  # pandas Index rebuilt lazily
         needed = len(self.key_index)
         for ring in self.rings.values():
             if needed > ring.capacity:
                 ring.grow(max(needed, ring.capacity * 2))
         return torch.from_numpy(lut[codes])
+
+    def _ids_for_keys(self, keys: list) -> np.ndarray:
+        """Vectorized key -> dense id (-1 for unknown): one C-level
+        hash probe per key instead of a python dict.get loop."""
+        import pandas as pd
+
+        if self._key_lookup is None or \
+                len(self._key_lookup) != len(self.key_index):
+            self._key_lookup = pd.Index(list(self.key_index.keys()))
+        if not len(self._key_lookup):
+            return np.full(len(keys), -1, dtype=np.int64)
+        return self._key_lookup.get_indexer(pd.Index(keys))
+
+    def _window_values_cached(self, agg, window, now_ts: float) -> dict:
+        ring = self.rings[agg.name]
+        period = int(now_ts // ring.period_seconds)
+        key = (agg.name, window, period, self._version)
+        hit = self._reduce_cache.get(key)
+        if hit is None:
+            if len(self._reduce_cache) > 64:
+                self._reduce_cache.clear()
+            hit = {op: t.numpy() for op, t in ring.window_values(
+                parse_span(window), now_ts).items()}
+            self._reduce_cache[key] = hit
+        return hit
+
+    def _feature_name_map(self) -> dict:
+        if self._name_map is None:
+            self._name_map = {
+                f"{agg.name}_{op}_{window}": (agg, window, op)
+                for agg in self.feature_set.spec.aggregations
+                for window in agg.windows
+                for op in agg.operations}
+        return self._name_map
 
     def ingest_batch(self, df):
         """Fold a dataframe batch: update latest rows + window rings."""
@@ -250,6 +296,8 @@ class OnlineTable:
         ts_key = fset.spec.timestamp_key
         entities = fset.entity_names()
         with self._lock:
+            self._version += 1
+            self._reduce_cache.clear()
             if len(entities) == 1:
                 keys = df[entities[0]].tolist()
             else:
@@ -286,72 +334,73 @@ class OnlineTable:
                        feature_names: typing.List[str],
                        now_ts: float = None):
         """Columnar fast path: aggregate features only -> float32
-        matrix [rows, features] (NaN for unknown keys).  One ring
-        reduce per (agg, window) serves the whole batch; rows resolve
-        by fancy indexing — no per-row dicts."""
-        import numpy as np
-
+        matrix [rows, features] (NaN for unknown keys).  Ring reduces
+        are cached per ingest generation, key->id mapping is one
+        vectorized Index probe, and rows resolve by fancy indexing —
+        no per-row python at all."""
         now_ts = now_ts or time.time()
-        fset = self.feature_set
         with self._lock:
-            ids = np.array([self.key_index.get(self._key_of(row), -1)
-                            for row in entity_rows], dtype=np.int64)
+            entities = self.feature_set.entity_names()
+            if len(entities) == 1:
+                ename = entities[0]
+                keys = [row[ename] for row in entity_rows]
+            else:
+                keys = [tuple(row[e] for e in entities)
+                        for row in entity_rows]
+            ids = self._ids_for_keys(keys)
             known = ids >= 0
             safe = np.where(known, ids, 0)
             out = np.full((len(entity_rows), len(feature_names)), np.nan,
                           dtype=np.float32)
-            reduced_cache = {}
+            name_map = self._feature_name_map()
             for j, name in enumerate(feature_names):
-                resolved = None
-                for agg in fset.spec.aggregations:
-                    for window in agg.windows:
-                        for op in agg.operations:
-                            if f"{agg.name}_{op}_{window}" == name:
-                                resolved = (agg, window, op)
+                resolved = name_map.get(name)
                 if resolved is None:
                     continue
                 agg, window, op = resolved
-                ring = self.rings[agg.name]
-                ckey = (agg.name, window)
-                if ckey not in reduced_cache:
-                    reduced_cache[ckey] = ring.window_values(
-                        parse_span(window), now_ts)
-                col = reduced_cache[ckey][op].numpy()[safe]
-                out[:, j] = np.where(known, col, np.nan)
+                vals = self._window_values_cached(agg, window, now_ts)
+                out[:, j] = np.where(known, vals[op][safe], np.nan)
             return out
 
     def get(self, entity_rows: typing.List[dict], now_ts: float = None
             ) -> typing.List[dict]:
-        """Batched online lookup: latest values + window aggregates."""
+        """Batched online lookup: latest values + window aggregates.
+        Aggregates come from the cached per-generation ring reduces +
+        one vectorized id probe; only the output-dict assembly is
+        per-row."""
         now_ts = now_ts or time.time()
         fset = self.feature_set
         with self._lock:
-            # one ring reduce per (agg, window) serves the whole batch
-            reduced = {}
+            entities = fset.entity_names()
+            if len(entities) == 1:
+                ename = entities[0]
+                keys = [row[ename] for row in entity_rows]
+            else:
+                keys = [tuple(row[e] for e in entities)
+                        for row in entity_rows]
+            ids = self._ids_for_keys(keys)
+            columns = []  # (name, values np array)
             for agg in fset.spec.aggregations:
-                ring = self.rings[agg.name]
                 for window in agg.windows:
-                    reduced[(agg.name, window)] = ring.window_values(
-                        parse_span(window), now_ts)
+                    vals = self._window_values_cached(agg, window,
+                                                      now_ts)
+                    for op in agg.operations:
+                        columns.append((f"{agg.name}_{op}_{window}",
+                                        vals[op]))
+            latest_get = self.latest.get
             out = []
-            for row in entity_rows:
-                key = self._key_of(row)
-                idx = self.key_index.get(key)
+            for key, idx in zip(keys, ids):
                 record: dict = {}
-                latest = self.latest.get(key)
+                latest = latest_get(key)
                 if latest:
                     record.update(latest)
-                for agg in fset.spec.aggregations:
-                    for window in agg.windows:
-                        vals = reduced[(agg.name, window)]
-                        for op in agg.operations:
-                            name = f"{agg.name}_{op}_{window}"
-                            if idx is None:
-                                record[name] = None
-                            else:
-                                value = float(vals[op][idx])
-                                record[name] = None \
-                                    if value != value else value
+                if idx < 0:
+                    for name, _ in columns:
+                        record[name] = None
+                else:
+                    for name, col in columns:
+                        value = float(col[idx])
+                        record[name] = None if value != value else value
                 out.append(record)
             return out
 

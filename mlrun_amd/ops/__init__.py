@@ -527,14 +527,21 @@ def window_ingest(ring, keys, values, period_idx):
         (period_idx.numpy().astype(np.int64) % n_periods)
     view = ring.numpy().reshape(-1, 4)
     vals = values.numpy()
+    # sort + reduceat segment sums (np.add.at is per-element)
+    order = np.argsort(flat_idx, kind="stable")
+    flat_s = flat_idx[order]
+    vals_s = vals[order]
+    uniq, starts = np.unique(flat_s, return_index=True)
+    ends = np.append(starts[1:], len(flat_s))
+    counts = (ends - starts).astype(view.dtype)
     if ring.dtype == torch.float64:
-        vals = vals.astype(np.float64)
-        np.add.at(view[:, 0], flat_idx, vals * vals)
-        np.add.at(view[:, 1], flat_idx, 1.0)
-        np.add.at(view[:, 2], flat_idx, vals)  # f64 sum
+        vals_s = vals_s.astype(np.float64)
+        view[uniq, 0] += np.add.reduceat(vals_s * vals_s, starts)
+        view[uniq, 1] += counts
+        view[uniq, 2] += np.add.reduceat(vals_s, starts)  # f64 sum
         return ring
-    np.add.at(view[:, 0], flat_idx, vals)
-    np.add.at(view[:, 1], flat_idx, 1.0)
+    view[uniq, 0] += np.add.reduceat(vals_s, starts)
+    view[uniq, 1] += counts
     return ring
 
 
@@ -574,8 +581,16 @@ def window_ingest_mm(ring_mm, keys, values, period_idx):
         (period_idx.numpy().astype(np.int64) % n_periods)
     ov = _f32_to_ordered_np(values.numpy())
     view = ring_mm.numpy().view(np.uint32).reshape(-1, 2)
-    np.minimum.at(view[:, 0], flat, ov)
-    np.maximum.at(view[:, 1], flat, ov)
+    # sort + reduceat: one segment min/max per touched cell (ufunc.at
+    # is a per-element python-level loop — too slow for 1M-event
+    # batches)
+    order = np.argsort(flat, kind="stable")
+    flat_s, ov_s = flat[order], ov[order]
+    uniq, starts = np.unique(flat_s, return_index=True)
+    seg_min = np.minimum.reduceat(ov_s, starts)
+    seg_max = np.maximum.reduceat(ov_s, starts)
+    view[uniq, 0] = np.minimum(view[uniq, 0], seg_min)
+    view[uniq, 1] = np.maximum(view[uniq, 1], seg_max)
     return ring_mm
 
 
@@ -594,8 +609,13 @@ def window_ingest_fl(ring_fl, keys, values, timestamps, period_idx):
     pack = (timestamps.numpy().astype(np.uint64) << np.uint64(32)) | \
         _f32_to_ordered_np(values.numpy()).astype(np.uint64)
     view = ring_fl.numpy().view(np.uint64).reshape(-1, 2)
-    np.minimum.at(view[:, 0], flat, pack)
-    np.maximum.at(view[:, 1], flat, pack)
+    order = np.argsort(flat, kind="stable")
+    flat_s, pack_s = flat[order], pack[order]
+    uniq, starts = np.unique(flat_s, return_index=True)
+    seg_min = np.minimum.reduceat(pack_s, starts)
+    seg_max = np.maximum.reduceat(pack_s, starts)
+    view[uniq, 0] = np.minimum(view[uniq, 0], seg_min)
+    view[uniq, 1] = np.maximum(view[uniq, 1], seg_max)
     return ring_fl
 
 
