@@ -80,11 +80,35 @@ class MLRunTorchInterface:
         self._ddp = DistributedModel(self.model, bucket_cap_mb=bucket_cap_mb)
         return self._ddp
 
+    def _shard_loader(self, loader):
+        """Shard a DataLoader across ranks with a DistributedSampler
+        (reference mlrun_interface.py:617-631 _add_distributed_sampler);
+        non-DataLoader iterables are returned unchanged (the caller
+        already shards, e.g. synthetic per-rank batches)."""
+        if self.world_size <= 1 or \
+                not isinstance(loader, torch.utils.data.DataLoader) or \
+                isinstance(loader.sampler,
+                           torch.utils.data.distributed.DistributedSampler):
+            return loader
+        sampler = torch.utils.data.distributed.DistributedSampler(
+            loader.dataset, num_replicas=self.world_size, rank=self.rank)
+        return torch.utils.data.DataLoader(
+            loader.dataset, batch_size=loader.batch_size,
+            sampler=sampler, num_workers=loader.num_workers,
+            collate_fn=loader.collate_fn, drop_last=loader.drop_last)
+
     def train(self, train_loader, loss_fn, optimizer, epochs: int = 1,
               validation_loader=None, metric_fns: list = None,
-              scheduler=None, use_amp: bool = False):
-        """The epoch loop (reference mlrun_interface.py:657 _train)."""
+              scheduler=None, use_amp: bool = False,
+              auto_log_model: bool = True, model_key: str = "model"):
+        """The epoch loop (reference mlrun_interface.py:657 _train):
+        auto DDP (if apply_mlrun set it up), per-rank shards via
+        DistributedSampler, allreduce-mean epoch metrics, rank-0
+        logging, and final model artifact logging."""
         model = self._ddp or self.model
+        train_loader = self._shard_loader(train_loader)
+        if validation_loader is not None:
+            validation_loader = self._shard_loader(validation_loader)
         metric_fns = metric_fns or []
         history: typing.Dict[str, list] = {}
         for epoch in range(epochs):
@@ -112,6 +136,9 @@ class MLRunTorchInterface:
             if self.context is not None and self.rank == 0:
                 self.context.log_results(results)
             logger.info("epoch done", **results)
+        if auto_log_model and self.context is not None:
+            self.log_model(model_key,
+                           metrics={k: v[-1] for k, v in history.items()})
         return history
 
     @torch.no_grad()
@@ -150,6 +177,8 @@ class MLRunTorchInterface:
         return float(tensor.item()) / self.world_size
 
     def log_model(self, key="model", **kwargs):
+        """Rank-0-only model artifact (reference is_logging_worker gate,
+        execution.py:1040)."""
         if self.context is None or self.rank != 0:
             return None
         import io
@@ -160,7 +189,26 @@ class MLRunTorchInterface:
                                       framework="pytorch", **kwargs)
 
 
-def apply_mlrun(model: torch.nn.Module, context=None) -> MLRunTorchInterface:
+def apply_mlrun(model: torch.nn.Module, context=None, auto_ddp=True,
+                bucket_cap_mb: int = None) -> MLRunTorchInterface:
     """Attach MLRun training instrumentation to a torch module
-    (reference: frameworks/pytorch apply_mlrun/train)."""
-    return MLRunTorchInterface(model, context=context)
+    (reference frameworks/pytorch mlrun_interface.py:555 _setup):
+    when launched under WORLD_SIZE > 1, this auto-initializes the
+    RCCL/gloo process group, pins the local GPU, broadcasts rank-0
+    parameters and wraps the model in the bucketed-allreduce
+    DistributedModel — the Horovod-equivalent, own engine."""
+    interface = MLRunTorchInterface(model, context=context)
+    if auto_ddp and interface.world_size > 1:
+        import torch.distributed as dist
+
+        from ..parallel.ddp import init_process_group
+
+        if not dist.is_initialized():
+            init_process_group()
+        if torch.cuda.is_available():
+            local_rank = int(os.environ.get("LOCAL_RANK",
+                                            interface.rank))
+            torch.cuda.set_device(local_rank)
+            model.cuda(local_rank)
+        interface.setup_distributed(bucket_cap_mb=bucket_cap_mb)
+    return interface
