@@ -18,3 +18,14 @@ from .stream import (  # noqa: F401
 )
 from .drift import histogram_drift_metrics  # noqa: F401
 from .controller import enable_model_monitoring, MonitoringController  # noqa: F401
+from .apps import (  # noqa: F401
+    HistogramDataDriftApplication,
+    LatencyPerformanceApplication,
+    ModelMonitoringApplicationBase,
+    ModelMonitoringApplicationMetric,
+    ModelMonitoringApplicationResult,
+    MonitoringApplicationContext,
+    ResultKindApp,
+    ResultStatusApp,
+)
+from .writer import ModelMonitoringWriter  # noqa: F401

@@ -46,48 +46,22 @@ class ModelMonitoringEvent:
         return {slot: getattr(self, slot) for slot in self.__slots__}
 
 
-class _WindowStats:
-    """Tumbling-window aggregate for one (endpoint, window)."""
-
-    def __init__(self, window_seconds: int):
-        self.window_seconds = window_seconds
-        self.window_start = 0.0
-        self.count = 0
-        self.error_count = 0
-        self.latency_sum = 0.0
-        self.latency_max = 0.0
-
-    def add(self, event: ModelMonitoringEvent):
-        bucket = event.timestamp - (event.timestamp % self.window_seconds)
-        if bucket != self.window_start:
-            self.window_start = bucket
-            self.count = self.error_count = 0
-            self.latency_sum = self.latency_max = 0.0
-        self.count += 1
-        if event.error:
-            self.error_count += 1
-        self.latency_sum += event.latency_ms
-        self.latency_max = max(self.latency_max, event.latency_ms)
-
-    def snapshot(self) -> dict:
-        return {
-            "window_seconds": self.window_seconds,
-            "window_start": self.window_start,
-            "count": self.count,
-            "error_count": self.error_count,
-            "avg_latency_ms": (self.latency_sum / self.count)
-            if self.count else 0.0,
-            "max_latency_ms": self.latency_max,
-        }
-
-
 class EventStreamProcessor:
-    """In-process monitoring stream: push() from serving steps."""
+    """In-process monitoring stream: push() from serving steps.
+
+    Window stats are SLIDING (5m/1h over 60s period buckets), computed
+    from the feature-store WindowRing — the same per-period-cell HIP
+    kernels that power the online feature store serve the monitoring
+    latency/error windows (GPU-resident when the endpoint runs on
+    GPU)."""
 
     WINDOWS = (300, 3600)  # 5m / 1h, matching the reference defaults
+    RING_PERIOD = 60       # one bucket per minute; ring spans 1h + 1
 
     def __init__(self, project: str = "", parquet_dir: str = "",
-                 parquet_batch: int = None, db=None):
+                 parquet_batch: int = None, db=None, device="cpu"):
+        from ..feature_store.online import WindowRing
+
         self.project = project or "default"
         self.parquet_dir = parquet_dir or os.path.join(
             config.base_dir, "monitoring", self.project)
@@ -95,11 +69,43 @@ class EventStreamProcessor:
             config.model_endpoint_monitoring.parquet_batching_max_events)
         self._db = db
         self._lock = threading.Lock()
-        self._stats: dict = defaultdict(
-            lambda: {w: _WindowStats(w) for w in self.WINDOWS})
+        n_periods = max(self.WINDOWS) // self.RING_PERIOD + 1
+        self._latency_ring = WindowRing(self.RING_PERIOD, n_periods,
+                                        device=device, capacity=64)
+        self._error_ring = WindowRing(self.RING_PERIOD, n_periods,
+                                      device=device, capacity=64)
+        self._endpoint_ids: dict = {}
+        self._ring_buffer: list = []   # (endpoint_idx, latency, err, ts)
         self._pending: list = []
         self._feature_samples: dict = defaultdict(lambda: deque(maxlen=4096))
         self._tsdb: dict = defaultdict(list)  # endpoint -> [(t, snapshot)]
+
+    def _endpoint_idx(self, endpoint_id) -> int:
+        idx = self._endpoint_ids.get(endpoint_id)
+        if idx is None:
+            idx = len(self._endpoint_ids)
+            self._endpoint_ids[endpoint_id] = idx
+            if idx >= self._latency_ring.capacity:
+                self._latency_ring.grow(self._latency_ring.capacity * 2)
+                self._error_ring.grow(self._error_ring.capacity * 2)
+        return idx
+
+    def _fold_rings_locked(self):
+        """Batch-ingest buffered events into the window rings."""
+        if not self._ring_buffer:
+            return
+        import torch
+
+        idxs = torch.tensor([b[0] for b in self._ring_buffer])
+        lats = torch.tensor([b[1] for b in self._ring_buffer])
+        errs = torch.tensor([b[2] for b in self._ring_buffer])
+        # f64: unix seconds in a float32 tensor quantize to ~128 s and
+        # can land events in FUTURE buckets the reduce never covers
+        ts = torch.tensor([b[3] for b in self._ring_buffer],
+                          dtype=torch.float64)
+        self._ring_buffer = []
+        self._latency_ring.ingest(idxs, lats, ts)
+        self._error_ring.ingest(idxs, errs, ts)
 
     def _get_db(self):
         if self._db is None:
@@ -112,8 +118,12 @@ class EventStreamProcessor:
         if isinstance(event, dict):
             event = ModelMonitoringEvent(**event)
         with self._lock:
-            for win in self._stats[event.endpoint_id].values():
-                win.add(event)
+            idx = self._endpoint_idx(event.endpoint_id)
+            self._ring_buffer.append(
+                (idx, float(event.latency_ms),
+                 1.0 if event.error else 0.0, float(event.timestamp)))
+            if len(self._ring_buffer) >= 256:
+                self._fold_rings_locked()
             self._pending.append(event.to_dict())
             if event.inputs is not None:
                 self._feature_samples[event.endpoint_id].append(event.inputs)
@@ -121,10 +131,38 @@ class EventStreamProcessor:
         if flush:
             self.flush()
 
-    def endpoint_stats(self, endpoint_id: str) -> dict:
+    def endpoint_stats(self, endpoint_id: str,
+                       now_ts: float = None) -> dict:
+        """Sliding-window serving stats from the rings: count, error
+        count, avg/max/min latency per 5m/1h window."""
+        now_ts = now_ts or time.time()
         with self._lock:
-            return {str(w): s.snapshot()
-                    for w, s in self._stats[endpoint_id].items()}
+            self._fold_rings_locked()
+            idx = self._endpoint_ids.get(endpoint_id)
+            out = {}
+            for window in self.WINDOWS:
+                if idx is None:
+                    out[str(window)] = {"window_seconds": window,
+                                        "count": 0, "error_count": 0,
+                                        "avg_latency_ms": 0.0,
+                                        "max_latency_ms": 0.0}
+                    continue
+                lat = self._latency_ring.window_values(window, now_ts)
+                err = self._error_ring.window_values(window, now_ts)
+
+                def _val(tensor):
+                    value = float(tensor[idx])
+                    return 0.0 if value != value else value
+
+                out[str(window)] = {
+                    "window_seconds": window,
+                    "count": int(_val(lat["count"])),
+                    "error_count": int(_val(err["sum"])),
+                    "avg_latency_ms": _val(lat["avg"]),
+                    "max_latency_ms": _val(lat["max"]),
+                    "min_latency_ms": _val(lat["min"]),
+                }
+            return out
 
     def feature_samples(self, endpoint_id: str) -> list:
         with self._lock:
@@ -133,11 +171,10 @@ class EventStreamProcessor:
     def record_tsdb(self):
         """Snapshot all endpoint windows into the in-memory TSDB."""
         now = time.time()
-        with self._lock:
-            for endpoint_id, windows in self._stats.items():
-                self._tsdb[endpoint_id].append(
-                    (now, {str(w): s.snapshot()
-                           for w, s in windows.items()}))
+        for endpoint_id in list(self._endpoint_ids):
+            snapshot = self.endpoint_stats(endpoint_id, now)
+            with self._lock:
+                self._tsdb[endpoint_id].append((now, snapshot))
 
     def tsdb_series(self, endpoint_id: str) -> list:
         with self._lock:
@@ -167,13 +204,23 @@ class EventStreamProcessor:
     def update_endpoint_record(self, endpoint_id: str, model: str = "",
                                function_uri: str = "", extra: dict = None):
         stats = self.endpoint_stats(endpoint_id)
-        record = {
-            "kind": "model-endpoint",
-            "metadata": {"project": self.project, "uid": endpoint_id},
-            "spec": {"model": model, "function_uri": function_uri},
-            "status": {"state": "ready", "last_request": now_iso(),
-                       "stats": stats, **(extra or {})},
-        }
+        record = {}
+        try:  # merge into the stored record (the writer's app_results
+            record = self._get_db().get_model_endpoint(  # must survive)
+                self.project, endpoint_id) or {}
+        except Exception:
+            record = {}
+        record.setdefault("kind", "model-endpoint")
+        record.setdefault("metadata", {}).update(
+            {"project": self.project, "uid": endpoint_id})
+        spec = record.setdefault("spec", {})
+        if model:
+            spec["model"] = model
+        if function_uri:
+            spec["function_uri"] = function_uri
+        record.setdefault("status", {}).update(
+            {"state": "ready", "last_request": now_iso(),
+             "stats": stats, **(extra or {})})
         try:
             self._get_db().store_model_endpoint(self.project, endpoint_id,
                                                 record)

@@ -46,6 +46,11 @@ def config_test_base(tmp_path, monkeypatch):
 
     store_manager._stores.clear()
     store_manager._db = None
+    # fresh monitoring stream processors (the "default" processor is a
+    # process-global: ring stats would leak between tests otherwise)
+    from mlrun_amd.model_monitoring import stream as _stream_mod
+
+    _stream_mod._processors.clear()
     yield config
     dbmod._run_db = None
     dbmod._run_db_url = None

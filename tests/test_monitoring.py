@@ -68,3 +68,99 @@ class TestDriftMetrics:
         assert low["drift_score"] < 0.1
         assert high["drift_score"] > 0.5
         assert high["tvd"] > low["tvd"]
+
+
+class TestMonitoringPipeline:
+    """End-to-end: serve traffic -> sliding-window stats -> app
+    framework -> writer -> drift detection -> alert + notification
+    (VERDICT round-1 item 7)."""
+
+    def test_serve_shift_alert(self, rundb, tmp_path, monkeypatch):
+        import numpy as np
+
+        import mlrun_amd
+        from mlrun_amd.config import config
+        from mlrun_amd.model_monitoring import (
+            MonitoringController,
+            ModelMonitoringWriter,
+            get_stream_processor,
+        )
+        from mlrun_amd.model_monitoring import stream as stream_mod
+
+        monkeypatch.setattr(config, "base_dir", str(tmp_path))
+        stream_mod._processors.clear()
+        project = "mon-e2e"
+        # alert config: model-drift events push a console notification
+        pushed = []
+
+        from mlrun_amd.utils import notifications as notif_mod
+
+        class _Capture:
+            def __init__(self, name="", params=None):
+                pass
+
+            def push(self, message, severity="info", runs=None):
+                pushed.append((message, severity))
+
+        monkeypatch.setitem(notif_mod._kinds, "capture", _Capture)
+        rundb.store_alert_config(project, "drift-alert", {
+            "name": "drift-alert", "project": project,
+            "summary": "drift on {{endpoint}}",
+            "severity": "high",
+            "trigger": {"events": ["model-drift"]},
+            "criteria": {"count": 1},
+            "notifications": [{"kind": "capture"}],
+        })
+
+        # a serving function with model tracking enabled
+        from tests.test_serving import EchoModel
+
+        fn = mlrun_amd.new_function("mon-fn", kind="serving", project=project)
+        fn.set_topology("router")
+        fn.add_model("m1", class_name=EchoModel, model_path=".")
+        fn.set_tracking()
+        server = fn.to_mock_server(track_models=True)
+        processor = get_stream_processor(project)
+
+        rng = np.random.default_rng(0)
+        reference = rng.normal(0, 1, 2000)
+
+        def serve(dist_mean, n):
+            for _ in range(n):
+                values = rng.normal(dist_mean, 1, 8).tolist()
+                server.test("/v2/models/m1/infer",
+                            body={"inputs": values})
+
+        controller = MonitoringController(project, db=rundb)
+        endpoint_id = None
+        serve(0.0, 40)
+        # the endpoint id under which the server pushed events
+        assert processor._endpoint_ids, "no monitoring events pushed"
+        endpoint_id = next(iter(processor._endpoint_ids))
+        controller.set_reference(endpoint_id, reference)
+
+        results = controller.run_iteration()
+        assert results[endpoint_id]["status"] != "drift_detected"
+        assert not pushed
+
+        # sliding-window serving stats exist (ring-backed)
+        stats = processor.endpoint_stats(endpoint_id)
+        assert stats["300"]["count"] >= 40
+        assert stats["3600"]["count"] >= 40
+
+        # inject a distribution shift and sweep again
+        serve(6.0, 200)
+        results = controller.run_iteration()
+        assert results[endpoint_id]["status"] == "drift_detected"
+        assert pushed, "alert notification not fired"
+
+        # writer persisted app results (parquet TSDB analog + record)
+        writer: ModelMonitoringWriter = controller.writer
+        df = writer.read_results(endpoint_id)
+        assert len(df) > 0
+        assert "general_drift" in set(df.get("result_name", []))
+        record = rundb.get_model_endpoint(project, endpoint_id)
+        app_results = record["status"]["app_results"]
+        assert "histogram-data-drift" in app_results
+        assert app_results["histogram-data-drift"][
+            "general_drift_status"] == 2  # detected
