@@ -23,7 +23,8 @@ import os
 import sqlite3
 import threading
 
-from ..errors import MLRunConflictError, MLRunNotFoundError
+from ..errors import (MLRunConflictError, MLRunInvalidArgumentError,
+                      MLRunNotFoundError)
 from ..model import RunStates
 from ..utils import now_iso
 from .base import RunDBInterface
@@ -116,6 +117,55 @@ CREATE INDEX IF NOT EXISTS idx_runs_state ON runs(project, state);
 CREATE INDEX IF NOT EXISTS idx_artifacts_key ON artifacts(project, key);
 """
 
+# ---------------------------------------------------------------------
+# Versioned schema migrations (reference: server/api/db/sqldb alembic
+# tree, 2,532 LoC).  The version lives in SQLite's PRAGMA user_version;
+# a round-1 DB file reports 0 and every migration is applied in order,
+# so old on-disk DBs open cleanly after an upgrade.  Each migration
+# must be idempotent (guard ALTERs by inspecting the live schema) —
+# _SCHEMA above only creates missing objects and never alters.
+# ---------------------------------------------------------------------
+SCHEMA_VERSION = 3
+
+
+def _has_column(conn, table: str, column: str) -> bool:
+    return any(r[1] == column for r in
+               conn.execute(f"PRAGMA table_info({table})"))
+
+
+def _migration_1(conn):
+    """Run bookkeeping columns the round-1 schema lacked:
+    requested_logs (log-collector pull tracking, reference models.py
+    Run.requested_logs) + updated-time indexes for list pagination."""
+    if not _has_column(conn, "runs", "requested_logs"):
+        conn.execute("ALTER TABLE runs ADD COLUMN "
+                     "requested_logs INTEGER DEFAULT 0")
+    conn.execute("CREATE INDEX IF NOT EXISTS idx_runs_updated "
+                 "ON runs(project, updated)")
+    conn.execute("CREATE INDEX IF NOT EXISTS idx_artifacts_updated "
+                 "ON artifacts(project, updated)")
+
+
+def _migration_2(conn):
+    """Pagination cache (reference models.py PaginationCache): stores
+    paginated-query cursors so clients can page with a token."""
+    conn.execute("""
+        CREATE TABLE IF NOT EXISTS pagination_cache (
+            key TEXT PRIMARY KEY, method TEXT, current_page INTEGER,
+            page_size INTEGER, kwargs TEXT, last_accessed TEXT)""")
+
+
+def _migration_3(conn):
+    """Time-window trackers (reference models.py TimeWindowTracker):
+    high-water marks for periodic sweeps (runs monitor, summaries)."""
+    conn.execute("""
+        CREATE TABLE IF NOT EXISTS time_window_trackers (
+            key TEXT PRIMARY KEY, timestamp TEXT, max_allowed_period
+            INTEGER)""")
+
+
+_MIGRATIONS = {1: _migration_1, 2: _migration_2, 3: _migration_3}
+
 
 def _match_labels(body: dict, labels) -> bool:
     if not labels:
@@ -171,8 +221,35 @@ class SQLRunDB(RunDBInterface):
 
     def _init_schema(self):
         with self._write_lock:
-            self._conn().executescript(_SCHEMA)
-            self._conn().commit()
+            conn = self._conn()
+            conn.executescript(_SCHEMA)
+            conn.commit()
+            self._migrate_locked(conn)
+
+    def _migrate_locked(self, conn) -> list:
+        """Apply pending schema migrations; returns applied versions."""
+        current = conn.execute("PRAGMA user_version").fetchone()[0]
+        applied = []
+        for version in range(current + 1, SCHEMA_VERSION + 1):
+            _MIGRATIONS[version](conn)
+            conn.execute(f"PRAGMA user_version = {version}")
+            conn.commit()
+            applied.append(version)
+        if applied:
+            from ..utils import logger
+
+            logger.info("schema migrated", dsn=self.dsn,
+                        from_version=current, to_version=SCHEMA_VERSION)
+        return applied
+
+    def trigger_migrations(self) -> dict:
+        """Explicit migration entry (reference httpdb
+        trigger_migrations -> operations/migrations endpoint)."""
+        with self._write_lock:
+            conn = self._conn()
+            applied = self._migrate_locked(conn)
+            current = conn.execute("PRAGMA user_version").fetchone()[0]
+        return {"schema_version": current, "applied": applied}
 
     def _execute(self, sql, params=()):
         with self._write_lock:
@@ -223,7 +300,15 @@ class SQLRunDB(RunDBInterface):
 
     def list_runs(self, name="", uid=None, project="", labels=None, state=None,
                   sort=True, last=0, iter=False, start_time_from=None,
-                  start_time_to=None):
+                  start_time_to=None, partition_by=None,
+                  rows_per_partition: int = 1, partition_sort_by="updated",
+                  partition_order="desc", max_partitions: int = 0,
+                  offset: int = 0, limit: int = 0):
+        """List runs with the reference's richer query surface
+        (httpdb.py list_runs): label selectors, iteration expansion,
+        time windows, OFFSET/LIMIT pagination and partition-by
+        ("name"/"project"): per-partition top-N by the sort field —
+        the reference's newest-run-per-name query shape."""
         project = project or "default"
         sql = "SELECT body FROM runs WHERE project=?"
         params: list = [project]
@@ -249,15 +334,52 @@ class SQLRunDB(RunDBInterface):
         if start_time_to:
             sql += " AND start_time<=?"
             params.append(str(start_time_to))
-        if sort:
+        if partition_by:
+            # window-function partition query (SQLite >= 3.25): top
+            # rows_per_partition per name/project by partition_sort_by
+            field = {"updated": "updated",
+                     "created": "start_time",
+                     "start_time": "start_time"}.get(
+                partition_sort_by or "updated", "updated")
+            order = "DESC" if (partition_order or "desc").lower() == \
+                "desc" else "ASC"
+            part_col = {"name": "name", "project": "project"}.get(
+                partition_by)
+            if part_col is None:
+                raise MLRunInvalidArgumentError(
+                    f"unsupported partition_by {partition_by!r}")
+            inner = sql.replace(
+                "SELECT body FROM runs",
+                f"SELECT body, ROW_NUMBER() OVER (PARTITION BY "
+                f"{part_col} ORDER BY {field} {order}) AS rn FROM runs",
+                1)
+            sql = (f"SELECT body FROM ({inner}) "
+                   f"WHERE rn <= {int(rows_per_partition)}")
+        elif sort:
             sql += " ORDER BY start_time DESC"
         if last:
             sql += f" LIMIT {int(last)}"
+        elif limit:
+            sql += f" LIMIT {int(limit)} OFFSET {int(offset)}"
+        elif offset:
+            sql += f" LIMIT -1 OFFSET {int(offset)}"
         out = []
         for row in self._query(sql, params):
             body = json.loads(row["body"])
             if _match_labels(body, labels):
                 out.append(body)
+        if partition_by and max_partitions:
+            seen: dict = {}
+            capped = []
+            for body in out:
+                part = (body.get("metadata", {}) or {}).get(
+                    "name" if partition_by == "name" else "project", "")
+                bucket = seen.setdefault(part, [])
+                if len(seen) > int(max_partitions) and not bucket:
+                    continue
+                bucket.append(body)
+                capped.append(body)
+            out = capped
         return out
 
     def del_run(self, uid, project="", iter=0):
@@ -440,9 +562,21 @@ class SQLRunDB(RunDBInterface):
             raise MLRunNotFoundError(f"artifact {project}/{key} not found")
         return json.loads(rows[0]["body"])
 
+    _ARTIFACT_CATEGORIES = {
+        # reference common/schemas ArtifactCategories kind groupings
+        "model": ("model",),
+        "dataset": ("dataset",),
+        "document": ("document",),
+        "other": ("", "artifact", "plot", "chart", "table", "link"),
+    }
+
     def list_artifacts(self, name="", project="", tag="", labels=None,
                        since=None, until=None, kind=None, category=None,
-                       iter=None, tree=None):
+                       iter=None, tree=None, limit: int = 0,
+                       offset: int = 0):
+        """Artifact listing with the v2 query surface (reference
+        endpoints/artifacts_v2.py): kind/category filters, time
+        window, tree (producer id), OFFSET/LIMIT pagination."""
         project = project or "default"
         sql = "SELECT body FROM artifacts WHERE project=?"
         params: list = [project]
@@ -452,10 +586,34 @@ class SQLRunDB(RunDBInterface):
         if kind:
             sql += " AND kind=?"
             params.append(kind)
+        if category:
+            kinds = self._ARTIFACT_CATEGORIES.get(category)
+            if kinds is None:
+                raise MLRunInvalidArgumentError(
+                    f"unknown artifact category {category!r}")
+            sql += f" AND kind IN ({','.join('?' * len(kinds))})"
+            params.extend(kinds)
         if tree:
             sql += " AND tree=?"
             params.append(tree)
+        if iter is not None:
+            sql += " AND iteration=?"
+            params.append(int(iter))
+        if since:
+            sql += " AND updated>=?"
+            params.append(str(since))
+        if until:
+            sql += " AND updated<=?"
+            params.append(str(until))
+        if tag and tag not in ("*", "latest"):
+            sql += (" AND key IN (SELECT key FROM artifact_tags "
+                    "WHERE project=? AND tag=?)")
+            params.extend([project, tag])
         sql += " ORDER BY updated DESC"
+        if limit:
+            sql += f" LIMIT {int(limit)} OFFSET {int(offset)}"
+        elif offset:
+            sql += f" LIMIT -1 OFFSET {int(offset)}"
         out = []
         for row in self._query(sql, params):
             body = json.loads(row["body"])
@@ -893,3 +1051,65 @@ class SQLRunDB(RunDBInterface):
             return {"schedule": sched}
         launcher = ServerSideLauncher(db=self)
         return launcher.launch_task(runspec)
+
+    # ------------------------------------------- pagination / trackers
+    def paginated_list(self, method: str, page_token: str = None,
+                       page: int = 1, page_size: int = 20, **kwargs):
+        """Token-based pagination over any list_* method (reference
+        utils/pagination.py + PaginationCache model).  First call:
+        pass method/page_size/filters -> returns (items, token).
+        Subsequent calls: pass page_token alone to get the next page;
+        token is None when exhausted."""
+        import hashlib
+
+        if page_token:
+            rows = self._query(
+                "SELECT method, current_page, page_size, kwargs "
+                "FROM pagination_cache WHERE key=?", (page_token,))
+            if not rows:
+                raise MLRunNotFoundError(
+                    f"pagination token {page_token!r} expired")
+            method = rows[0]["method"]
+            page = rows[0]["current_page"] + 1
+            page_size = rows[0]["page_size"]
+            kwargs = json.loads(rows[0]["kwargs"])
+        else:
+            token_src = json.dumps([method, page_size, kwargs],
+                                   sort_keys=True, default=str)
+            page_token = hashlib.sha1(token_src.encode()).hexdigest()[:20]
+        lister = getattr(self, method)
+        items = lister(limit=page_size + 1,
+                       offset=(page - 1) * page_size, **kwargs)
+        has_more = len(items) > page_size
+        items = items[:page_size]
+        if has_more:
+            self._execute(
+                "INSERT OR REPLACE INTO pagination_cache "
+                "(key, method, current_page, page_size, kwargs, "
+                "last_accessed) VALUES (?,?,?,?,?,?)",
+                (page_token, method, page, page_size,
+                 json.dumps(kwargs, default=str), now_iso()))
+            return items, page_token
+        self._execute("DELETE FROM pagination_cache WHERE key=?",
+                      (page_token,))
+        return items, None
+
+    def clean_pagination_cache(self, older_than_iso: str = None):
+        if older_than_iso:
+            self._execute("DELETE FROM pagination_cache WHERE "
+                          "last_accessed < ?", (older_than_iso,))
+        else:
+            self._execute("DELETE FROM pagination_cache")
+
+    def get_time_window_tracker(self, key: str):
+        rows = self._query(
+            "SELECT timestamp FROM time_window_trackers WHERE key=?",
+            (key,))
+        return rows[0]["timestamp"] if rows else None
+
+    def store_time_window_tracker(self, key: str, timestamp: str = None):
+        """High-water mark for periodic sweeps (reference
+        TimeWindowTracker model)."""
+        self._execute(
+            "INSERT OR REPLACE INTO time_window_trackers "
+            "(key, timestamp) VALUES (?,?)", (key, timestamp or now_iso()))
