@@ -14,6 +14,7 @@ informers.
 """
 
 import asyncio
+import json
 import os
 import typing
 
@@ -24,7 +25,17 @@ from fastapi.responses import JSONResponse
 from ..config import config
 from ..db.sqldb import SQLRunDB
 from ..errors import MLRunBaseError, err_to_status
-from ..utils import logger
+from ..utils import logger, now_iso
+
+
+def _deep_update(target: dict, patch: dict):
+    """Recursive dict merge (PATCH semantics — reference strategies)."""
+    for key, value in patch.items():
+        if isinstance(value, dict) and isinstance(target.get(key), dict):
+            _deep_update(target[key], value)
+        else:
+            target[key] = value
+    return target
 from .scheduler import Scheduler
 
 
@@ -105,11 +116,36 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     async def list_runs(project: str = "default", name: str = "",
                         state: str = "", sort: int = 1, last: int = 0,
                         iter: int = 0, page: int = 0, page_size: int = 0,
+                        page_token: str = "",
+                        partition_by: str = "",
+                        rows_per_partition: int = 1,
+                        partition_sort_by: str = "updated",
+                        partition_order: str = "desc",
+                        max_partitions: int = 0,
                         uid: typing.List[str] = Query(None),
                         label: typing.List[str] = Query(None)):
+        if page_token or (page_size and not page):
+            items, token = db.paginated_list(
+                "list_runs", page_token=page_token or None,
+                page_size=page_size or 20,
+                **({} if page_token else dict(
+                    name=name, project=project, labels=label,
+                    state=state or None, iter=bool(iter),
+                    partition_by=partition_by or None,
+                    rows_per_partition=rows_per_partition,
+                    partition_sort_by=partition_sort_by,
+                    partition_order=partition_order)))
+            return {"runs": items,
+                    "pagination": {"page_token": token,
+                                   "page_size": page_size}}
         runs = db.list_runs(name=name, uid=uid, project=project,
                             labels=label, state=state or None,
-                            sort=bool(sort), last=last, iter=bool(iter))
+                            sort=bool(sort), last=last, iter=bool(iter),
+                            partition_by=partition_by or None,
+                            rows_per_partition=rows_per_partition,
+                            partition_sort_by=partition_sort_by,
+                            partition_order=partition_order,
+                            max_partitions=max_partitions)
         total = len(runs)
         if page_size:
             start = max(page - 1, 0) * page_size
@@ -162,10 +198,33 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     @app.get("/api/v1/artifacts")
     async def list_artifacts(project: str = "default", name: str = "",
                              tag: str = "", kind: str = "", tree: str = "",
+                             category: str = "", since: str = "",
+                             until: str = "", limit: int = 0,
+                             offset: int = 0, page_token: str = "",
+                             page_size: int = 0,
                              label: typing.List[str] = Query(None)):
+        """Artifacts listing with the v2 query surface (reference
+        endpoints/artifacts_v2.py): category/time filters + token
+        pagination."""
+        if page_token or page_size:
+            items, token = db.paginated_list(
+                "list_artifacts", page_token=page_token or None,
+                page_size=page_size or 20,
+                **({} if page_token else dict(
+                    name=name, project=project, tag=tag, labels=label,
+                    kind=kind or None, category=category or None,
+                    tree=tree or None, since=since or None,
+                    until=until or None)))
+            return {"artifacts": items,
+                    "pagination": {"page_token": token,
+                                   "page_size": page_size}}
         artifacts = db.list_artifacts(name=name, project=project, tag=tag,
                                       labels=label, kind=kind or None,
-                                      tree=tree or None)
+                                      category=category or None,
+                                      tree=tree or None,
+                                      since=since or None,
+                                      until=until or None,
+                                      limit=limit, offset=offset)
         return {"artifacts": artifacts}
 
     # -------------------------------------------------------- functions
@@ -702,6 +761,241 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         }
 
     # ------------------------------------------------------------ submit
+    # ------------------------------------------------ function deploy
+    def _load_function(project: str, name: str, tag: str = ""):
+        from ..run import new_function
+
+        body = db.get_function(name, project, tag=tag)
+        return new_function(runtime=body)
+
+    @app.post("/api/v1/build/function")
+    async def build_function(request: Request):
+        """Server-side function build (reference build/function +
+        kaniko): node-locally a venv/wheel build via utils/builder,
+        tracked as a background task."""
+        payload = await request.json()
+        func = payload.get("function") or {}
+        project = func.get("metadata", {}).get("project", "default")
+        name = func.get("metadata", {}).get("name", "")
+        with_mlrun = payload.get("with_mlrun", False)
+        task_name = f"build-{name}"
+        db.store_background_task(project, {
+            "metadata": {"name": task_name, "project": project},
+            "status": {"state": "running"}})
+
+        def _build():
+            try:
+                from ..run import new_function
+                from ..utils.builder import build_runtime
+
+                fn = new_function(runtime=func)
+                build_runtime(fn, with_mlrun=with_mlrun, install=False)
+                db.store_function(fn.to_dict(), name, project)
+                db.store_background_task(project, {
+                    "metadata": {"name": task_name, "project": project},
+                    "status": {"state": "succeeded"}})
+            except Exception as exc:
+                db.store_background_task(project, {
+                    "metadata": {"name": task_name, "project": project},
+                    "status": {"state": "failed", "error": str(exc)}})
+
+        import threading as _threading
+
+        _threading.Thread(target=_build, daemon=True).start()
+        return {"data": func, "ready": False,
+                "background_task": task_name}
+
+    @app.get("/api/v1/build/status")
+    async def build_status(project: str = "default", name: str = "",
+                           tag: str = ""):
+        task = db.get_background_task(project, f"build-{name}")
+        state = (task.get("status") or {}).get("state", "unknown")
+        return {"ready": state == "succeeded", "state": state,
+                "error": (task.get("status") or {}).get("error", "")}
+
+    @app.post("/api/v1/start/function")
+    async def start_function(request: Request):
+        """Deploy a STORED function as a live local host (the node-
+        local analog of the reference's nuclio deploy): serving/remote
+        kinds get a FastAPI host; the address lands in the function
+        status."""
+        payload = await request.json()
+        func = payload.get("function") or {}
+        project = func.get("metadata", {}).get("project", "default")
+        name = func.get("metadata", {}).get("name", "")
+
+        def _deploy():
+            from ..run import new_function
+
+            fn = new_function(runtime=func)
+            address = fn.deploy()
+            body = fn.to_dict()
+            body.setdefault("status", {})["address"] = address
+            body["status"]["state"] = "ready"
+            db.store_function(body, name, project)
+            return address
+
+        address = await asyncio.to_thread(_deploy)
+        return {"data": {"address": address, "name": name,
+                         "state": "ready"}}
+
+    @app.get("/api/v1/projects/{project}/functions/{name}/status")
+    async def function_status(project: str, name: str, tag: str = ""):
+        body = db.get_function(name, project, tag=tag)
+        return {"status": body.get("status", {})}
+
+    # ------------------------------------------ migrations/operations
+    @app.post("/api/v1/operations/migrations")
+    async def trigger_migrations():
+        """Apply pending schema migrations (reference
+        operations.py trigger_migrations)."""
+        return db.trigger_migrations()
+
+    # ------------------------------------------------ features search
+    @app.get("/api/v1/projects/{project}/features")
+    async def list_features(project: str, name: str = "",
+                            entity: typing.List[str] = Query(None),
+                            label: typing.List[str] = Query(None)):
+        return {"features": db.list_features(
+            project, name=name or None, entities=entity, labels=label)}
+
+    @app.get("/api/v1/projects/{project}/entities")
+    async def list_entities(project: str, name: str = "",
+                            label: typing.List[str] = Query(None)):
+        return {"entities": db.list_entities(project, name=name or None,
+                                             labels=label)}
+
+    # ------------------------------------------------- patch variants
+    @app.patch("/api/v1/projects/{name}")
+    async def patch_project(name: str, body: dict):
+        current = db.get_project(name)
+        from ..utils import update_in
+
+        for key, value in (body or {}).items():
+            if isinstance(value, dict) and isinstance(
+                    current.get(key), dict):
+                current[key].update(value)
+            else:
+                update_in(current, key, value)
+        db.store_project(name, current)
+        return current
+
+    @app.patch("/api/v1/projects/{project}/feature-sets/{name}")
+    async def patch_feature_set(project: str, name: str, body: dict,
+                                tag: str = "latest"):
+        current = db.get_feature_set(name, project, tag)
+        _deep_update(current, body or {})
+        db.store_feature_set(current, name, project, tag=tag)
+        return current
+
+    @app.patch("/api/v1/projects/{project}/feature-vectors/{name}")
+    async def patch_feature_vector(project: str, name: str, body: dict,
+                                   tag: str = "latest"):
+        current = db.get_feature_vector(name, project, tag)
+        _deep_update(current, body or {})
+        db.store_feature_vector(current, name, project, tag=tag)
+        return current
+
+    @app.patch("/api/v1/projects/{project}/model-endpoints/{endpoint_id}")
+    async def patch_model_endpoint(project: str, endpoint_id: str,
+                                   body: dict):
+        current = db.get_model_endpoint(project, endpoint_id)
+        _deep_update(current, body or {})
+        db.store_model_endpoint(project, endpoint_id, current)
+        return current
+
+    # ------------------------------------------------------ bulk dels
+    @app.delete("/api/v1/runs")
+    async def del_runs(project: str = "", name: str = "",
+                       state: str = ""):
+        db.del_runs(name=name, project=project, state=state or None)
+        return {}
+
+    @app.delete("/api/v1/artifacts")
+    async def del_artifacts(project: str = "", name: str = "",
+                            tag: str = ""):
+        db.del_artifacts(name=name, project=project, tag=tag)
+        return {}
+
+    # --------------------------------------------------- hub source CRUD
+    @app.put("/api/v1/hub/sources/{name}")
+    async def store_hub_source(name: str, body: dict):
+        """Register a hub source (reference hub.py store_source): body
+        {"path": ..., "order": N}; registered in the hub registry AND
+        persisted."""
+        from .. import hub
+
+        path = (body.get("spec", body) or {}).get("path", "")
+        order = int((body.get("spec", body) or {}).get("order", -1))
+        hub.add_hub_source(name, path, order)
+        db._execute(
+            "INSERT OR REPLACE INTO hub_sources "
+            "(name, idx, updated, body) VALUES (?,?,?,?)",
+            (name, order, now_iso(),
+             json.dumps(body, default=str)))
+        return body
+
+    @app.get("/api/v1/hub/sources/{name}")
+    async def get_hub_source(name: str):
+        rows = db._query("SELECT body FROM hub_sources WHERE name=?",
+                         (name,))
+        if rows:
+            return json.loads(rows[0]["body"])
+        from ..hub import list_hub_sources
+
+        for source in list_hub_sources():
+            if source["name"] == name:
+                return source
+        raise HTTPException(status_code=404,
+                            detail=f"hub source {name} not found")
+
+    @app.delete("/api/v1/hub/sources/{name}")
+    async def delete_hub_source(name: str):
+        from .. import hub
+
+        hub._sources.pop(name, None)
+        db._execute("DELETE FROM hub_sources WHERE name=?", (name,))
+        return {}
+
+    # --------------------------------------------------- api gateways
+    @app.put("/api/v1/projects/{project}/api-gateways/{name}")
+    async def store_api_gateway(project: str, name: str, body: dict):
+        return db.store_api_gateway(project, name, body)
+
+    @app.get("/api/v1/projects/{project}/api-gateways/{name}")
+    async def get_api_gateway(project: str, name: str):
+        return db.get_api_gateway(project, name)
+
+    @app.get("/api/v1/projects/{project}/api-gateways")
+    async def list_api_gateways(project: str):
+        return {"api_gateways": db.list_api_gateways(project)}
+
+    @app.delete("/api/v1/projects/{project}/api-gateways/{name}")
+    async def delete_api_gateway(project: str, name: str):
+        db.delete_api_gateway(project, name)
+        return {}
+
+    # -------------------------------------------------- notifications
+    @app.put("/api/v1/projects/{project}/runs/{uid}/notifications")
+    async def set_run_notifications(project: str, uid: str, body: dict):
+        """Replace a run's notification list (reference
+        set_run_notifications)."""
+        run = db.read_run(uid, project)
+        run.setdefault("spec", {})["notifications"] = \
+            body.get("notifications", [])
+        db.store_run(run, uid, project)
+        return {}
+
+    @app.put("/api/v1/projects/{project}/schedules/{name}/notifications")
+    async def set_schedule_notifications(project: str, name: str,
+                                         body: dict):
+        sched = db.get_schedule(project, name)
+        task = sched.get("task") or {}
+        task.setdefault("spec", {})["notifications"] = \
+            body.get("notifications", [])
+        db.update_schedule(project, name, {"task": task})
+        return {}
+
     @app.post("/api/v1/submit_job")
     async def submit_job(request: Request):
         body = await request.json()

@@ -125,7 +125,7 @@ CREATE INDEX IF NOT EXISTS idx_artifacts_key ON artifacts(project, key);
 # must be idempotent (guard ALTERs by inspecting the live schema) —
 # _SCHEMA above only creates missing objects and never alters.
 # ---------------------------------------------------------------------
-SCHEMA_VERSION = 3
+SCHEMA_VERSION = 4
 
 
 def _has_column(conn, table: str, column: str) -> bool:
@@ -164,7 +164,18 @@ def _migration_3(conn):
             INTEGER)""")
 
 
-_MIGRATIONS = {1: _migration_1, 2: _migration_2, 3: _migration_3}
+def _migration_4(conn):
+    """API-gateway configs (reference models.py has no table — the
+    reference stores gateways in nuclio; node-locally they are DB
+    documents served by /projects/{p}/api-gateways)."""
+    conn.execute("""
+        CREATE TABLE IF NOT EXISTS api_gateways (
+            project TEXT NOT NULL, name TEXT NOT NULL, updated TEXT,
+            body TEXT, PRIMARY KEY (project, name))""")
+
+
+_MIGRATIONS = {1: _migration_1, 2: _migration_2, 3: _migration_3,
+               4: _migration_4}
 
 
 def _match_labels(body: dict, labels) -> bool:
@@ -1113,3 +1124,80 @@ class SQLRunDB(RunDBInterface):
         self._execute(
             "INSERT OR REPLACE INTO time_window_trackers "
             "(key, timestamp) VALUES (?,?)", (key, timestamp or now_iso()))
+
+    # -------------------------------------------------- feature search
+    def list_features(self, project, name=None, tag=None, entities=None,
+                      labels=None):
+        """Search features ACROSS feature sets (reference
+        endpoints/feature_store.py list_features): returns
+        feature + its owning feature-set digest."""
+        out = []
+        for fset in self.list_feature_sets(project):
+            meta = fset.get("metadata", {})
+            spec = fset.get("spec", {})
+            if not _match_labels(fset, labels):
+                continue
+            set_entities = [e.get("name") for e in
+                            spec.get("entities", [])]
+            if entities and not set(entities) & set(set_entities):
+                continue
+            for feature in spec.get("features", []):
+                fname = feature.get("name", "")
+                if name and name.strip("~") not in fname:
+                    continue
+                out.append({
+                    "feature": feature,
+                    "feature_set_digest": {
+                        "metadata": {"name": meta.get("name"),
+                                     "tag": meta.get("tag", "latest")},
+                        "spec": {"entities": spec.get("entities", [])},
+                    }})
+        return out
+
+    def list_entities(self, project, name=None, tag=None, labels=None):
+        """Search entities across feature sets (reference
+        list_entities)."""
+        out = []
+        for fset in self.list_feature_sets(project):
+            meta = fset.get("metadata", {})
+            if not _match_labels(fset, labels):
+                continue
+            for entity in fset.get("spec", {}).get("entities", []):
+                ename = entity.get("name", "")
+                if name and name.strip("~") not in ename:
+                    continue
+                out.append({
+                    "entity": entity,
+                    "feature_set_digest": {
+                        "metadata": {"name": meta.get("name"),
+                                     "tag": meta.get("tag", "latest")},
+                    }})
+        return out
+
+    # ----------------------------------------------------- api gateways
+    def store_api_gateway(self, project, name, body: dict):
+        self._execute(
+            "INSERT OR REPLACE INTO api_gateways "
+            "(project, name, updated, body) VALUES (?,?,?,?)",
+            (project or "default", name, now_iso(),
+             json.dumps(body, default=str)))
+        return body
+
+    def get_api_gateway(self, project, name):
+        rows = self._query(
+            "SELECT body FROM api_gateways WHERE project=? AND name=?",
+            (project or "default", name))
+        if not rows:
+            raise MLRunNotFoundError(f"api gateway {name} not found")
+        return json.loads(rows[0]["body"])
+
+    def list_api_gateways(self, project):
+        rows = self._query(
+            "SELECT body FROM api_gateways WHERE project=?",
+            (project or "default",))
+        return [json.loads(r["body"]) for r in rows]
+
+    def delete_api_gateway(self, project, name):
+        self._execute(
+            "DELETE FROM api_gateways WHERE project=? AND name=?",
+            (project or "default", name))

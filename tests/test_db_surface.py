@@ -1,0 +1,192 @@
+# Copyright 2026 mlrun_amd authors
+#
+# Licensed under the Apache License, Version 2.0 (the "License");
+# you may not use this file except in compliance with the License.
+"""Round-2 HTTPRunDB surface tests: the method families added to close
+the client gap vs reference httpdb.py."""
+
+import pytest
+
+from mlrun_amd.db.httpdb import HTTPRunDB
+from mlrun_amd.db.sqldb import SQLRunDB
+
+
+@pytest.fixture()
+def client(tmp_path):
+    from fastapi.testclient import TestClient
+
+    from mlrun_amd.api.main import create_app
+
+    db = SQLRunDB(str(tmp_path / "api.db"))
+    app = create_app(db, with_scheduler=False)
+    with TestClient(app) as c:
+        c.db = db
+        yield c
+
+
+@pytest.fixture()
+def httpdb(client):
+    db = HTTPRunDB("http://testserver")
+    orig_request = client.request
+
+    def request(method, url, params=None, data=None, json=None,
+                headers=None, timeout=None):
+        return orig_request(method, url, params=params, content=data,
+                            json=json, headers=headers)
+
+    db.session = client
+    db.session.request = request
+    return db
+
+
+class TestOperations:
+    def test_trigger_migrations(self, httpdb):
+        from mlrun_amd.db.sqldb import SCHEMA_VERSION
+
+        result = httpdb.trigger_migrations()
+        assert result["schema_version"] == SCHEMA_VERSION
+
+    def test_url_helpers(self, httpdb):
+        assert httpdb.get_api_path_prefix() == "api/v1"
+        assert httpdb.verify_authorization() is True
+
+
+class TestRunsSurface:
+    def test_del_runs_and_notifications(self, httpdb):
+        httpdb.store_run({"metadata": {"name": "r1", "uid": "u1"},
+                          "status": {"state": "completed"}}, "u1", "p")
+        httpdb.set_run_notifications("p", "u1", [{"kind": "console"}])
+        run = httpdb.read_run("u1", "p")
+        assert run["spec"]["notifications"] == [{"kind": "console"}]
+        httpdb.del_runs(project="p")
+        assert httpdb.list_runs(project="p") == []
+
+    def test_runs_partition_over_http(self, httpdb):
+        for i in range(4):
+            for name in ("a", "b"):
+                httpdb.store_run(
+                    {"metadata": {"name": name, "uid": f"{name}{i}"},
+                     "status": {"state": "completed"}},
+                    f"{name}{i}", "p")
+        runs = httpdb.api_call(
+            "GET", "runs", params={"project": "p",
+                                   "partition_by": "name",
+                                   "rows_per_partition": 1})["runs"]
+        assert len(runs) == 2
+
+    def test_paginated_api_call(self, httpdb):
+        for i in range(25):
+            httpdb.store_artifact(f"k{i:02d}",
+                                  {"kind": "model",
+                                   "metadata": {"key": f"k{i:02d}"}},
+                                  uid=f"t{i}", project="p")
+        pages = list(httpdb.paginated_api_call(
+            "GET", "artifacts", params={"project": "p",
+                                        "page_size": 10}))
+        items = httpdb.process_paginated_responses(pages, "artifacts")
+        assert len(items) == 25
+        assert len(pages) == 3
+
+
+class TestFeatureSurface:
+    def test_create_patch_list_features(self, httpdb):
+        fset = {"metadata": {"name": "fs1"},
+                "spec": {"entities": [{"name": "uid"}],
+                         "features": [{"name": "spend"},
+                                      {"name": "clicks"}]}}
+        httpdb.create_feature_set(fset, project="p")
+        httpdb.patch_feature_set(
+            "fs1", {"spec": {"description": "patched"}}, project="p")
+        got = httpdb.get_feature_set("fs1", "p")
+        assert got["spec"]["description"] == "patched"
+        assert got["spec"]["features"]  # additive patch kept features
+        features = httpdb.list_features("p", name="spend")
+        assert len(features) == 1
+        assert features[0]["feature"]["name"] == "spend"
+        entities = httpdb.list_entities("p")
+        assert entities[0]["entity"]["name"] == "uid"
+
+    def test_feature_vector_patch(self, httpdb):
+        httpdb.create_feature_vector(
+            {"metadata": {"name": "v1"},
+             "spec": {"features": ["fs1.spend"]}}, project="p")
+        httpdb.patch_feature_vector("v1", {"spec": {"label_feature":
+                                                    "y"}}, project="p")
+        got = httpdb.get_feature_vector("v1", "p")
+        assert got["spec"]["label_feature"] == "y"
+        assert got["spec"]["features"] == ["fs1.spend"]
+
+
+class TestHubAndGateways:
+    def test_hub_source_crud(self, httpdb, tmp_path):
+        httpdb.store_hub_source("extra", {"spec": {"path": str(tmp_path),
+                                                   "order": 5}})
+        got = httpdb.get_hub_source("extra")
+        assert got["spec"]["path"] == str(tmp_path)
+        names = [s["name"] for s in httpdb.list_hub_sources()]
+        assert "extra" in names
+        httpdb.delete_hub_source("extra")
+        with pytest.raises(Exception):
+            httpdb.get_hub_source("extra")
+
+    def test_api_gateway_crud(self, httpdb):
+        httpdb.store_api_gateway(
+            {"metadata": {"name": "gw1"},
+             "spec": {"functions": ["f1"], "project": "p"}},
+            project="p")
+        got = httpdb.get_api_gateway("gw1", "p")
+        assert got["spec"]["functions"] == ["f1"]
+        assert len(httpdb.list_api_gateways("p")) == 1
+        httpdb.delete_api_gateway("gw1", "p")
+        assert httpdb.list_api_gateways("p") == []
+
+
+class TestModelEndpointSurface:
+    def test_create_and_patch(self, httpdb):
+        httpdb.create_model_endpoint("p", "ep1", {
+            "kind": "model-endpoint",
+            "metadata": {"uid": "ep1"},
+            "spec": {"model": "m"}, "status": {"state": "ready"}})
+        httpdb.patch_model_endpoint("p", "ep1",
+                                    {"status": {"drift_status":
+                                                "no_drift"}})
+        got = httpdb.get_model_endpoint("p", "ep1")
+        assert got["status"]["drift_status"] == "no_drift"
+        assert got["status"]["state"] == "ready"
+
+
+class TestFunctionLifecycle:
+    def test_remote_builder_and_status(self, httpdb):
+        func = {"kind": "job", "metadata": {"name": "bf",
+                                            "project": "p"},
+                "spec": {}}
+        httpdb.store_function(func, "bf", "p")
+        resp = httpdb.remote_builder(func, with_mlrun=False)
+        assert resp["background_task"] == "build-bf"
+        import time
+
+        for _ in range(50):
+            status = httpdb.get_builder_status(func)
+            if status["state"] in ("succeeded", "failed"):
+                break
+            time.sleep(0.1)
+        assert status["state"] == "succeeded", status
+
+    def test_function_status(self, httpdb):
+        httpdb.store_function(
+            {"kind": "job", "metadata": {"name": "fs", "project": "p"},
+             "status": {"state": "ready"}}, "fs", "p")
+        status = httpdb.function_status("p", "fs")
+        assert status["status"]["state"] == "ready"
+
+
+class TestSecretsAndBackgroundAliases:
+    def test_secrets_aliases(self, httpdb):
+        httpdb.create_project_secrets("p", secrets={"KEY": "VAL"})
+        assert "KEY" in httpdb.list_project_secrets("p")["secrets"]
+        with pytest.raises(NotImplementedError):
+            httpdb.create_user_secrets("me", secrets={"a": "b"})
+
+    def test_runtime_resources(self, httpdb):
+        resources = httpdb.list_runtime_resources("default")
+        assert "gpu" in resources
