@@ -545,6 +545,157 @@ __global__ __launch_bounds__(256) void skinny_gemm_ws_kernel(
   }
 }
 
+// Software-pipelined wave-split variant (round-2 design doc §1, built
+// in C++ instead of inline asm): a DEPTH-stage rotating register ring
+// of one 32-K step each (MT A-vecs + NB B-rows).  The fully-unrolled
+// ring gives LLVM's waitcnt pass compile-time register stages, so it
+// emits PARTIAL vmcnt waits (retire the oldest stage while
+// (MT+NB)*(DEPTH-1) newer loads stay in flight) — steady-state loads
+// in flight instead of the base kernel's issue-burst-then-drain
+// blocks.  Staging VGPRs = DEPTH * (MT+NB) * 4; DEPTH<=4 stays under
+// the round-1 occupancy backfire threshold (64 staging VGPRs).
+template <bool SPLIT, int MT, int DEPTH, int NB = 2>
+__global__ __launch_bounds__(256) void skinny_gemm_ws_pipe_kernel(
+    void* __restrict__ out, const unsigned short* __restrict__ A,
+    const unsigned short* __restrict__ W, int M, int N, int K, int ksplit) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  constexpr int NT = 16 * NB;
+  const int n0 = blockIdx.x * NT;
+  if (n0 >= N) return;
+  int cbegin = 0, cend = K;
+  if (SPLIT) {
+    const int kchunk = (K / ksplit + 31) & ~31;
+    cbegin = blockIdx.y * kchunk;
+    cend = cbegin + kchunk;
+    if (cend > K) cend = K;
+  }
+  const int clen = cend - cbegin;
+  const int per_wave = ((clen / 4) + 31) & ~31;
+  int kbegin = cbegin + wave * per_wave;
+  int kend = kbegin + per_wave;
+  if (kend > cend) kend = cend;
+
+  const int arow = lane & 15;
+  const int kb = (lane >> 4) * 8;
+  const int brow0 = n0 + (lane & 15);
+  const int brow1 = brow0 + 16;
+
+  f32x4v acc0[MT], acc1[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t) {
+    acc0[t] = {0.f, 0.f, 0.f, 0.f};
+    acc1[t] = {0.f, 0.f, 0.f, 0.f};
+  }
+  const unsigned short* aptr[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t)
+    aptr[t] = A + (size_t)min(arow + 16 * t, M - 1) * K + kb;
+  const unsigned short* bptr0 =
+      W + (size_t)min(brow0, N - 1) * K + kb;
+  const unsigned short* bptr1 =
+      W + (size_t)min(NB == 2 ? brow1 : brow0, N - 1) * K + kb;
+
+  short8v a_ring[DEPTH][MT], b0_ring[DEPTH], b1_ring[DEPTH];
+  const int nsteps_total = (kend - kbegin) / 32;
+  // main region: a multiple of DEPTH steps; the rest runs in the tail
+  const int ngroups = nsteps_total / DEPTH;
+  const int nsteps = ngroups > 0 ? ngroups * DEPTH : 0;
+  int kload = kbegin;
+  int k = kbegin + nsteps * 32;
+  if (nsteps) {
+#pragma unroll
+    for (int s = 0; s < DEPTH; ++s) {
+#pragma unroll
+      for (int t = 0; t < MT; ++t)
+        a_ring[s][t] =
+            *reinterpret_cast<const short8v*>(aptr[t] + kload);
+      b0_ring[s] = *reinterpret_cast<const short8v*>(bptr0 + kload);
+      if (NB == 2)
+        b1_ring[s] = *reinterpret_cast<const short8v*>(bptr1 + kload);
+      kload += 32;
+    }
+    for (int g = DEPTH; g < nsteps; g += DEPTH) {
+#pragma unroll
+      for (int s = 0; s < DEPTH; ++s) {
+#pragma unroll
+        for (int t = 0; t < MT; ++t) {
+          acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_ring[s][t], b0_ring[s], acc0[t], 0, 0, 0);
+          if (NB == 2)
+            acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_ring[s][t], b1_ring[s], acc1[t], 0, 0, 0);
+        }
+#pragma unroll
+        for (int t = 0; t < MT; ++t)
+          a_ring[s][t] =
+              *reinterpret_cast<const short8v*>(aptr[t] + kload);
+        b0_ring[s] = *reinterpret_cast<const short8v*>(bptr0 + kload);
+        if (NB == 2)
+          b1_ring[s] = *reinterpret_cast<const short8v*>(bptr1 + kload);
+        kload += 32;
+      }
+    }
+#pragma unroll
+    for (int s = 0; s < DEPTH; ++s) {
+#pragma unroll
+      for (int t = 0; t < MT; ++t) {
+        acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_ring[s][t], b0_ring[s], acc0[t], 0, 0, 0);
+        if (NB == 2)
+          acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_ring[s][t], b1_ring[s], acc1[t], 0, 0, 0);
+      }
+    }
+  }
+  for (; k < kend; k += 32) {
+    short8v bf0 = *reinterpret_cast<const short8v*>(bptr0 + k);
+#pragma unroll
+    for (int t = 0; t < MT; ++t) {
+      short8v af = *reinterpret_cast<const short8v*>(aptr[t] + k);
+      acc0[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0[t],
+                                                        0, 0, 0);
+      if (NB == 2) {
+        short8v bf1 = *reinterpret_cast<const short8v*>(bptr1 + k);
+        acc1[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1,
+                                                          acc1[t], 0, 0,
+                                                          0);
+      }
+    }
+  }
+
+  __shared__ float comb[4][16 * MT][NT];
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int t = 0; t < MT; ++t)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      comb[wave][16 * t + crow_base + r][ccol] = acc0[t][r];
+      if (NB == 2)
+        comb[wave][16 * t + crow_base + r][ccol + 16] = acc1[t][r];
+    }
+  __syncthreads();
+  if (wave == 0) {
+    constexpr int CPL = 4 * MT * NB;
+#pragma unroll
+    for (int c = 0; c < CPL; ++c) {
+      const int cell = lane * CPL + c;
+      const int m = cell / NT;
+      const int n = cell % NT;
+      if (m >= M || n0 + n >= N) continue;
+      float sum = comb[0][m][n] + comb[1][m][n] + comb[2][m][n] +
+                  comb[3][m][n];
+      if (SPLIT) {
+        float* part = (float*)out + (size_t)blockIdx.y * M * N;
+        part[(size_t)m * N + n0 + n] = sum;
+      } else {
+        ((unsigned short*)out)[(size_t)m * N + n0 + n] = f2bf(sum);
+      }
+    }
+  }
+}
+
 __global__ void reduce_cast_kernel(unsigned short* __restrict__ out,
                                    const float* __restrict__ part,
                                    long long mn, int ksplit);
@@ -817,6 +968,32 @@ static int gemm_unr2_env() {
   return v;
 }
 
+// MLRUN_GEMM_PIPE=3|4|6 selects the software-pipelined ring kernel
+// (ring depth) for the M in (16,32] decode shapes; 0 (default) keeps
+// the burst-unrolled kernel.
+static int gemm_pipe_env() {
+  static int v = [] {
+    const char* e = getenv("MLRUN_GEMM_PIPE");
+    return e ? atoi(e) : 0;
+  }();
+  return v;
+}
+
+template <bool SPLIT>
+static bool launch_pipe_mt2(void* out, const void* A, const void* W,
+                            int M, int N, int K, int ksplit,
+                            const dim3& grid, void* stream) {
+  const int depth = gemm_pipe_env();
+  if (depth < 2) return false;
+  auto kern = depth >= 6 ? (skinny_gemm_ws_pipe_kernel<SPLIT, 2, 6>)
+              : depth >= 4 ? (skinny_gemm_ws_pipe_kernel<SPLIT, 2, 4>)
+                           : (skinny_gemm_ws_pipe_kernel<SPLIT, 2, 3>);
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, (hipStream_t)stream, out,
+                     (const unsigned short*)A, (const unsigned short*)W,
+                     M, N, K, ksplit);
+  return true;
+}
+
 void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
                               int M, int N, int K, int ksplit, int variant,
                               void* stream) {
@@ -846,14 +1023,17 @@ void launch_skinny_gemm_slabs(void* part_f32, const void* A, const void* W,
                          0, (hipStream_t)stream, part_f32,
                          (const unsigned short*)A, (const unsigned short*)W,
                          M, N, K, ksplit);
-    else if (M > 16)
-      hipLaunchKernelGGL(gemm_unr2_env() >= 8
-                             ? (skinny_gemm_ws_kernel<true, 2, 8>)
-                             : (skinny_gemm_ws_kernel<true, 2, 4>),
-                         grid, dim3(256),
-                         0, (hipStream_t)stream, part_f32,
-                         (const unsigned short*)A, (const unsigned short*)W,
-                         M, N, K, ksplit);
+    else if (M > 16) {
+      if (!launch_pipe_mt2<true>(part_f32, A, W, M, N, K, ksplit, grid,
+                                 stream))
+        hipLaunchKernelGGL(gemm_unr2_env() >= 8
+                               ? (skinny_gemm_ws_kernel<true, 2, 8>)
+                               : (skinny_gemm_ws_kernel<true, 2, 4>),
+                           grid, dim3(256),
+                           0, (hipStream_t)stream, part_f32,
+                           (const unsigned short*)A,
+                           (const unsigned short*)W, M, N, K, ksplit);
+    }
     else
       hipLaunchKernelGGL((skinny_gemm_ws_kernel<true, 1>), grid, dim3(256),
                          0, (hipStream_t)stream, part_f32,
@@ -911,14 +1091,17 @@ void launch_skinny_gemm(void* out_bf16, void* part_f32, const void* A,
                            (hipStream_t)stream, out_bf16,
                            (const unsigned short*)A,
                            (const unsigned short*)W, M, N, K, 1);
-      else if (M > 16)
-        hipLaunchKernelGGL(gemm_unr2_env() >= 8
-                               ? (skinny_gemm_ws_kernel<false, 2, 8>)
-                               : (skinny_gemm_ws_kernel<false, 2, 4>),
-                           dim3(nblocks), dim3(256), 0,
-                           (hipStream_t)stream, out_bf16,
-                           (const unsigned short*)A,
-                           (const unsigned short*)W, M, N, K, 1);
+      else if (M > 16) {
+        if (!launch_pipe_mt2<false>(out_bf16, A, W, M, N, K, 1,
+                                    dim3(nblocks), stream))
+          hipLaunchKernelGGL(gemm_unr2_env() >= 8
+                                 ? (skinny_gemm_ws_kernel<false, 2, 8>)
+                                 : (skinny_gemm_ws_kernel<false, 2, 4>),
+                             dim3(nblocks), dim3(256), 0,
+                             (hipStream_t)stream, out_bf16,
+                             (const unsigned short*)A,
+                             (const unsigned short*)W, M, N, K, 1);
+      }
       else
         hipLaunchKernelGGL((skinny_gemm_ws_kernel<false, 1>),
                            dim3(nblocks), dim3(256), 0,
