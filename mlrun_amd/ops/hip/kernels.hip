@@ -968,13 +968,16 @@ static int gemm_unr2_env() {
   return v;
 }
 
-// MLRUN_GEMM_PIPE=3|4|6 selects the software-pipelined ring kernel
-// (ring depth) for the M in (16,32] decode shapes; 0 (default) keeps
-// the burst-unrolled kernel.
+// MLRUN_GEMM_PIPE=0|3|4|6 selects the software-pipelined ring kernel
+// (ring depth) for the M in (16,32] decode shapes.  Default 4:
+// measured +1.1% e2e over the burst-unrolled kernel at B=32
+// (gpurun_out/pipe_ab.log — the per-wave outstanding-bytes wall caps
+// the gain, as docs/round2_kernel_designs.md §1 predicted); numerics
+// verified vs fp32 on all decode shapes (scripts/check_gemm_pipe.py).
 static int gemm_pipe_env() {
   static int v = [] {
     const char* e = getenv("MLRUN_GEMM_PIPE");
-    return e ? atoi(e) : 0;
+    return e ? atoi(e) : 4;
   }();
   return v;
 }

@@ -221,10 +221,13 @@ class ServingRuntime(BaseRuntime):
         (nuclio min_replicas analog; removes the single-process GIL
         cap — serving/workers.py)."""
         workers = workers or int(self.spec.min_replicas or 0)
-        if workers > 1:
+        max_workers = int(self.spec.max_replicas or 0)
+        if workers > 1 or max_workers > max(workers, 1):
             from ..serving.workers import WorkerPool
 
-            self._worker_pool = WorkerPool(self.to_dict(), workers)
+            self._worker_pool = WorkerPool(self.to_dict(),
+                                           max(workers, 1),
+                                           max_workers=max_workers)
             address = self._worker_pool.start()
             self.status.state = "ready"
             self.status.address = address
@@ -304,8 +307,10 @@ class ServingRuntime(BaseRuntime):
         return self
 
     def with_replicas(self, min_replicas: int, max_replicas: int = None):
-        """Worker-process count (reference nuclio min/max_replicas;
-        the node-local pool is fixed-size = min_replicas)."""
+        """Worker-process count (reference nuclio min/max_replicas):
+        the pool starts min_replicas workers and AUTOSCALES up to
+        max_replicas when live connections exceed the per-worker
+        budget (serving/workers.py WorkerPool._scale_loop)."""
         self.spec.min_replicas = min_replicas
         self.spec.max_replicas = max_replicas or min_replicas
         return self

@@ -45,6 +45,12 @@ class TcpRoundRobinProxy:
         self._thread = None
         self._loop = None
         self._next = 0
+        self.active = 0  # live proxied connections (autoscale signal)
+
+    def add_backend(self, host: str, port: int):
+        """Register a new worker (autoscale-up); next connections
+        round-robin over the larger set."""
+        self.backends.append((host, port))
 
     @property
     def address(self) -> str:
@@ -87,8 +93,12 @@ class TcpRoundRobinProxy:
                 except OSError:
                     client_w.close()
                     return
-                await asyncio.gather(pipe(client_r, backend_w),
-                                     pipe(backend_r, client_w))
+                self.active += 1
+                try:
+                    await asyncio.gather(pipe(client_r, backend_w),
+                                         pipe(backend_r, client_w))
+                finally:
+                    self.active -= 1
 
             async def main():
                 server = await asyncio.start_server(handle, self.host,
@@ -126,13 +136,23 @@ class WorkerPool:
     """N serving-host processes + proxy, built from a function spec."""
 
     def __init__(self, function_spec: dict, workers: int,
-                 env: dict = None):
+                 env: dict = None, max_workers: int = 0,
+                 scale_connections_per_worker: int = 16,
+                 scale_interval: float = 2.0):
         self.spec = function_spec
         self.workers = workers
+        self.max_workers = max(max_workers, workers)
+        # scale-up trigger: live proxied connections per worker above
+        # this -> add a worker (nuclio min/max_replicas analog)
+        self.scale_connections_per_worker = scale_connections_per_worker
+        self.scale_interval = scale_interval
         self.env = env or {}
         self.processes: typing.List[subprocess.Popen] = []
         self.ports: typing.List[int] = []
         self.proxy: typing.Optional[TcpRoundRobinProxy] = None
+        self._spec_path = ""
+        self._scaler = None
+        self._stopping = False
 
     @property
     def address(self) -> str:
@@ -153,13 +173,10 @@ class WorkerPool:
         env["PYTHONPATH"] = repo_root + os.pathsep + \
             env.get("PYTHONPATH", "")
         env.update({k: str(v) for k, v in self.env.items()})
+        self._env = env
+        self._spec_path = spec_file.name
         for _ in range(self.workers):
-            port = _free_port()
-            self.ports.append(port)
-            proc = subprocess.Popen(
-                [sys.executable, "-m", "mlrun_amd.serving.workers",
-                 spec_file.name, str(port)], env=env)
-            self.processes.append(proc)
+            self._spawn_worker()
         deadline = time.time() + timeout
         import requests
 
@@ -176,17 +193,70 @@ class WorkerPool:
                             f"serving worker on port {port} did not "
                             f"become ready")
                     time.sleep(0.5)
-        os.unlink(spec_file.name)
         self.proxy = TcpRoundRobinProxy(
             [("127.0.0.1", p) for p in self.ports])
         self.proxy.start()
+        if self.max_workers > self.workers:
+            import threading
+
+            self._scaler = threading.Thread(target=self._scale_loop,
+                                            daemon=True,
+                                            name="worker-autoscaler")
+            self._scaler.start()
         logger.info("serving worker pool started",
-                    workers=self.workers, address=self.address)
+                    workers=self.workers, max_workers=self.max_workers,
+                    address=self.address)
         return self.address
 
+    def _spawn_worker(self) -> int:
+        port = _free_port()
+        self.ports.append(port)
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "mlrun_amd.serving.workers",
+             self._spec_path, str(port)], env=self._env)
+        self.processes.append(proc)
+        return port
+
+    def _scale_loop(self):
+        """Scale-up monitor (nuclio max_replicas analog): when live
+        proxied connections exceed the per-worker budget, start
+        another worker and add it to the round-robin set.  No
+        scale-down — decode engines hold captured graphs and HBM is
+        plentiful (288 GB); idle workers cost nothing hot."""
+        import requests
+
+        while not self._stopping and \
+                len(self.ports) < self.max_workers:
+            time.sleep(self.scale_interval)
+            if self.proxy is None:
+                continue
+            threshold = self.scale_connections_per_worker * \
+                len(self.ports)
+            if self.proxy.active <= threshold:
+                continue
+            port = self._spawn_worker()
+            deadline = time.time() + 180
+            while not self._stopping:
+                try:
+                    requests.get(f"http://127.0.0.1:{port}/healthz",
+                                 timeout=2)
+                    break
+                except Exception:
+                    if time.time() > deadline:
+                        logger.warning("autoscaled worker never became "
+                                       "ready", port=port)
+                        return
+                    time.sleep(0.5)
+            self.proxy.add_backend("127.0.0.1", port)
+            logger.info("worker pool scaled up", workers=len(self.ports),
+                        active_connections=self.proxy.active)
+
     def stop(self):
+        self._stopping = True
         if self.proxy is not None:
             self.proxy.stop()
+        if self._spec_path and os.path.exists(self._spec_path):
+            os.unlink(self._spec_path)
         for proc in self.processes:
             if proc.poll() is None:
                 proc.terminate()

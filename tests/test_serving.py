@@ -536,6 +536,61 @@ class TestWorkerPool:
 
 
 
+class TestWorkerAutoscale:
+    def test_pool_scales_up_under_load(self):
+        """min_replicas=1, max_replicas=2: sustained concurrent
+        connections trigger a scale-up (nuclio max_replicas analog)."""
+        import threading
+
+        import requests
+
+        import mlrun_amd
+
+        fn = mlrun_amd.new_function(name="auto-pool", kind="serving")
+        fn.add_model("llm",
+                     class_name="mlrun_amd.models.llama.LlamaServer",
+                     config="tiny", batch_size=2, max_new_tokens=6)
+        fn.with_replicas(1, max_replicas=2)
+        pool = None
+        addr = fn.deploy()
+        try:
+            pool = fn._worker_pool
+            pool.scale_connections_per_worker = 2  # low trigger
+            pool.scale_interval = 0.2
+            assert len(pool.ports) == 1
+            stop = threading.Event()
+
+            def hammer():
+                session = requests.Session()
+                while not stop.is_set():
+                    try:
+                        session.post(addr + "/v2/models/llm/infer",
+                                     json={"inputs": [[1, 2, 3]],
+                                           "max_tokens": 3},
+                                     timeout=60)
+                    except Exception:
+                        return
+
+            threads = [threading.Thread(target=hammer, daemon=True)
+                       for _ in range(6)]
+            [t.start() for t in threads]
+            import time as _time
+
+            deadline = _time.time() + 60
+            while len(pool.ports) < 2 and _time.time() < deadline:
+                _time.sleep(0.3)
+            stop.set()
+            [t.join(timeout=10) for t in threads]
+            assert len(pool.ports) == 2, "pool did not scale up"
+            # the scaled-up backend serves traffic
+            resp = requests.post(addr + "/v2/models/llm/infer",
+                                 json={"inputs": [[1, 2, 3]],
+                                       "max_tokens": 3}, timeout=120)
+            assert resp.status_code == 200
+        finally:
+            fn.stop()
+
+
 class TestCycleGuard:
     def test_cyclic_graph_errors_instead_of_hanging(self):
         import mlrun_amd
