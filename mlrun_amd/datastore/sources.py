@@ -130,9 +130,78 @@ class StreamSource(BaseSource):
         return df
 
 
+class KafkaSource(BaseSource):
+    """Consume from a node-local kafka-analog stream (the topic
+    written by KafkaTarget; reference KafkaSource, sources.py:1052).
+    Records arrive json-encoded on the OutputStream queue."""
+
+    kind = "kafka"
+
+    def __init__(self, path="", key_field=None, time_field=None,
+                 attributes=None, **kwargs):
+        super().__init__(kwargs.get("name", ""), path, attributes,
+                         key_field, time_field)
+
+    def to_dataframe(self, columns=None, df_module=None, start_time=None,
+                     end_time=None, time_field=None):
+        import json as _json
+
+        import pandas as pd
+
+        from ..platforms import OutputStream
+
+        topic = (self.path or "").replace("kafka://", "")
+        stream = OutputStream.get_stream(f"kafka://{topic}")
+        records = []
+        for raw in stream.drain():
+            if isinstance(raw, (str, bytes)):
+                try:
+                    raw = _json.loads(raw)
+                except ValueError:
+                    continue
+            records.append(raw)
+        df = pd.DataFrame(records)
+        if columns:
+            df = df[columns]
+        return df
+
+
+class SQLSource(BaseSource):
+    """Read a SQL table (reference SQLSource): sqlite path or
+    SQLAlchemy-style sqlite:/// DSN."""
+
+    kind = "sql"
+
+    def __init__(self, path="", table=None, attributes=None, **kwargs):
+        attributes = dict(attributes or {})
+        if table:
+            attributes["table"] = table
+        super().__init__(kwargs.get("name", ""), path, attributes,
+                         kwargs.get("key_field"),
+                         kwargs.get("time_field"))
+
+    def to_dataframe(self, columns=None, df_module=None, start_time=None,
+                     end_time=None, time_field=None):
+        import sqlite3
+
+        import pandas as pd
+
+        path = (self.path or "").replace("sqlite:///", "")
+        table = self.attributes.get("table")
+        conn = sqlite3.connect(path)
+        try:
+            df = pd.read_sql_query(f"SELECT * FROM {table}", conn)
+        finally:
+            conn.close()
+        if columns:
+            df = df[columns]
+        return df
+
+
 def get_source_from_dict(struct: dict) -> BaseSource:
     kinds = {"csv": CSVSource, "parquet": ParquetSource,
-             "dataframe": DataFrameSource, "stream": StreamSource}
+             "dataframe": DataFrameSource, "stream": StreamSource,
+             "kafka": KafkaSource, "sql": SQLSource}
     kind = struct.get("kind", "csv")
     cls = kinds.get(kind)
     if cls is None:

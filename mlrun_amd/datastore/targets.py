@@ -144,9 +144,102 @@ class SQLTarget(BaseStoreTarget):
         return path
 
 
+class KafkaTarget(BaseStoreTarget):
+    """Stream-fan-out target (reference KafkaTarget,
+    targets.py:1634).  Node-locally a kafka topic is an OutputStream
+    (platforms.py) keyed by the topic path — consumers attach via
+    StreamSource/get_stream; when the ``kafka`` client IS importable
+    and brokers are configured, rows publish to real Kafka instead."""
+
+    kind = "kafka"
+    is_online = True
+
+    def __init__(self, path=None, brokers=None, **kwargs):
+        super().__init__(path=path, **kwargs)
+        self.brokers = brokers
+
+    def write_dataframe(self, df, feature_set) -> str:
+        topic = (self.path or feature_set.metadata.name).replace(
+            "kafka://", "")
+        records = df.to_dict(orient="records")
+        if self.brokers:
+            try:
+                from kafka import KafkaProducer  # optional client
+            except ImportError as exc:
+                raise ImportError(
+                    "kafka-python is not installed; omit `brokers` to "
+                    "use the node-local stream analog") from exc
+            import json as _json
+
+            producer = KafkaProducer(bootstrap_servers=self.brokers)
+            for record in records:
+                producer.send(topic, _json.dumps(
+                    record, default=str).encode())
+            producer.flush()
+        else:
+            from ..platforms import OutputStream
+
+            OutputStream(f"kafka://{topic}").push(records)
+        self.path = f"kafka://{topic}"
+        return self.path
+
+
+class RedisNoSqlTarget(BaseStoreTarget):
+    """Online KV rows in Redis (reference RedisNoSqlTarget,
+    targets.py:1482).  Requires the ``redis`` client (not shipped in
+    this image): fails loudly with the node-local alternative
+    (NoSqlTarget = GPU OnlineTable) rather than degrading."""
+
+    kind = "redisnosql"
+    is_online = True
+
+    def write_dataframe(self, df, feature_set) -> str:
+        try:
+            import redis
+        except ImportError as exc:
+            raise ImportError(
+                "redis client not installed — use NoSqlTarget (the "
+                "HBM-resident online table) for node-local online "
+                "storage") from exc
+        url = (self.path or "redis://localhost:6379").replace(
+            "rediss://", "redis://")
+        client = redis.Redis.from_url(url)
+        entities = feature_set.entity_names()
+        prefix = f"{feature_set.fullname}:"
+        import json as _json
+
+        for record in df.to_dict(orient="records"):
+            key = prefix + "|".join(str(record.get(e)) for e in entities)
+            client.hset(key, mapping={
+                k: _json.dumps(v, default=str)
+                for k, v in record.items()})
+        self.path = url
+        return url
+
+
+class TSDBTarget(BaseStoreTarget):
+    """Append-only time-series parquet log (the node-local TSDB the
+    reference writes through v3io-frames/TDEngine — model monitoring's
+    results log uses the same layout)."""
+
+    kind = "tsdb"
+    is_offline = True
+
+    def write_dataframe(self, df, feature_set) -> str:
+        import time as _time
+
+        base = self.path or self.default_path(feature_set) + "_tsdb"
+        os.makedirs(base, exist_ok=True)
+        path = os.path.join(base, f"ts-{int(_time.time() * 1000)}"
+                            f".parquet")
+        df.to_parquet(path)
+        self.path = base
+        return base
+
+
 _target_kinds = {cls.kind: cls for cls in
                  [ParquetTarget, CSVTarget, NoSqlTarget, StreamTarget,
-                  SQLTarget]}
+                  SQLTarget, KafkaTarget, RedisNoSqlTarget, TSDBTarget]}
 
 
 def get_target_from_spec(spec) -> BaseStoreTarget:
