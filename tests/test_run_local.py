@@ -230,8 +230,16 @@ class TestPackagers:
 
         assert manager.resolve(np.zeros(2)) is NumPyPackager
         assert manager.resolve(pd.DataFrame()) is PandasPackager
+        # arbitrary objects fall through to the cloudpickle catch-all
+        # (reference DefaultPackager object artifact-type)
+        from mlrun_amd.package.packagers import PicklePackager
+
+        assert manager.resolve(object()) is PicklePackager
+        # an empty manager still errors
+        from mlrun_amd.package.packagers import PackagersManager
+
         with pytest.raises(TypeError):
-            manager.resolve(object())
+            PackagersManager().resolve(object())
 
 
 class TestNotebookToFunction:
