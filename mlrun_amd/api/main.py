@@ -844,6 +844,19 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         body = db.get_function(name, project, tag=tag)
         return {"status": body.get("status", {})}
 
+    # ---------------------------------------------- project summaries
+    @app.get("/api/v1/project-summaries")
+    async def list_project_summaries():
+        """Per-project entity counts (reference projects summaries
+        endpoint + the periodic summary loop, server main.py:630 —
+        node-locally computed on demand; counts are indexed SQL)."""
+        return {"project_summaries": db.list_project_summaries()}
+
+    @app.get("/api/v1/project-summaries/{project}")
+    async def get_project_summary(project: str):
+        db.get_project(project)  # 404 on unknown project
+        return db.compute_project_summary(project)
+
     # ------------------------------------------ migrations/operations
     @app.post("/api/v1/operations/migrations")
     async def trigger_migrations():
@@ -1051,13 +1064,49 @@ def check_stuck_runs(db: SQLRunDB, now=None) -> list:
     return aborted
 
 
+def sweep_service_caches(db: SQLRunDB, now=None) -> dict:
+    """Periodic hygiene (reference cleanup loops, server main.py:599):
+    expire idle pagination-cache rows and terminal background tasks
+    past their TTL.  Returns what was swept (tested directly)."""
+    import datetime
+
+    now = now or datetime.datetime.now()
+    ttl = int(config.pagination.cache_ttl_seconds)
+    cutoff = (now - datetime.timedelta(seconds=ttl)).isoformat()
+    db.clean_pagination_cache(older_than_iso=cutoff)
+    task_ttl = int(config.background_tasks.ttl_seconds)
+    task_cutoff = (now - datetime.timedelta(
+        seconds=task_ttl)).isoformat()
+    swept = []
+    projects = [r[0] for r in db._query(
+        "SELECT DISTINCT project FROM background_tasks")]
+    for project in projects:
+        for task in db.list_background_tasks(project):
+            state = (task.get("status") or {}).get("state", "")
+            updated = (task.get("metadata") or {}).get("updated") or \
+                (task.get("status") or {}).get("updated", "")
+            if state in ("succeeded", "failed") and updated and \
+                    updated < task_cutoff:
+                name = (task.get("metadata") or {}).get("name", "")
+                db._execute(
+                    "DELETE FROM background_tasks WHERE project=? AND "
+                    "name=?", (project, name))
+                swept.append(f"{project}/{name}")
+    return {"background_tasks": swept, "pagination_cutoff": cutoff}
+
+
 async def _runs_monitor(db: SQLRunDB):
-    """Periodic run monitoring loop (reference main.py:608)."""
+    """Periodic run monitoring loop (reference main.py:608) + service
+    cache hygiene sweeps."""
     interval = int(config.runs.monitoring_interval)
+    ticks = 0
     while True:
         try:
             await asyncio.sleep(interval)
             check_stuck_runs(db)
+            ticks += 1
+            if ticks % 10 == 0:  # hygiene at 1/10th cadence
+                sweep_service_caches(db)
         except asyncio.CancelledError:
             return
         except Exception as exc:

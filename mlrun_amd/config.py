@@ -109,12 +109,79 @@ default_config = {
     },
     "notifications": {
         "smtp": {"server": "", "sender": ""},
+        # default pushers applied when a run declares none
+        "default_kinds": "console",
     },
     "secret_stores": {
         "env_file": "",
     },
-    "packagers": {"enabled": True},
-    "background_tasks": {"default_timeout": 600},
+    "packagers": {"enabled": True, "pack_results": True},
+    "background_tasks": {
+        "default_timeout": 600,
+        # terminal tasks older than this are swept (reference
+        # background-task cleanup loop)
+        "ttl_seconds": 6 * 3600,
+    },
+    "artifacts": {
+        # mirror of the reference artifacts block (config.py
+        # "artifacts"): hashing + target-path generation behavior
+        "calculate_hash": True,
+        "generate_target_path_from_artifact_hash": False,
+        "artifact_max_size_mb": 1024,
+        # artifact tags accepted by the tag endpoints
+        "allowed_tag_chars": "a-zA-Z0-9-_.",
+    },
+    "pagination": {
+        "default_page_size": 20,
+        # pagination_cache rows idle longer than this are cleaned
+        "cache_ttl_seconds": 3600,
+    },
+    "alerts": {
+        "max_per_project": 1000,
+        # reset criteria count after this many seconds without events
+        "event_window_seconds": 600,
+    },
+    "projects": {
+        # summary recompute cadence for the /project-summaries loop
+        "summaries_interval": 30,
+        "default_owner": "",
+    },
+    "workflows": {
+        "default_timeout": 3600,
+        "engine": "local",  # the node-local runner (KFP out of scope)
+    },
+    "logs": {
+        "decode": {"errors": "replace"},
+        "pull_state_interval": 3,   # run-log poll cadence (watch_log)
+        "pipelines_redirect": True,
+    },
+    "function": {
+        # per-run spec guards (reference function block: limits the
+        # server enforces at submit time)
+        "spec": {
+            "max_parameters": 1024,
+            "max_notifications": 16,
+        },
+    },
+    "runtimes": {
+        # node-local runtime behavior knobs
+        "job": {"venv_cache": True},
+        "mpijob": {
+            "gang_abort_grace_seconds": 5,
+            "heartbeat_interval": 5,
+        },
+        "serving": {
+            "worker_ready_timeout": 180,
+            "autoscale_connections_per_worker": 16,
+        },
+    },
+    "hub": {
+        "default_source": "default",
+    },
+    "tracing": {
+        # rocTX range emission for serving steps (utils/tracing.py)
+        "enabled": "auto",
+    },
 }
 
 

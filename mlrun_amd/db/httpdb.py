@@ -855,3 +855,10 @@ class HTTPRunDB(RunDBInterface):
         self.api_call(
             "DELETE",
             f"projects/{project or 'default'}/datastore-profiles/{name}")
+
+    def get_project_summary(self, project):
+        return self.api_call("GET", f"project-summaries/{project}")
+
+    def list_project_summaries(self):
+        return self.api_call("GET", "project-summaries").get(
+            "project_summaries", [])

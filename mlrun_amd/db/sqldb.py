@@ -1201,3 +1201,47 @@ class SQLRunDB(RunDBInterface):
         self._execute(
             "DELETE FROM api_gateways WHERE project=? AND name=?",
             (project or "default", name))
+
+    # ------------------------------------------------ project summaries
+    def compute_project_summary(self, project: str) -> dict:
+        """Per-project counts (reference ProjectSummary model +
+        the periodic summary-calculation loop, server main.py:630)."""
+        def _count(sql, params):
+            return self._query(sql, params)[0][0]
+
+        counts = {
+            "runs_completed_recent_count": _count(
+                "SELECT COUNT(*) FROM runs WHERE project=? AND "
+                "state='completed'", (project,)),
+            "runs_failed_recent_count": _count(
+                "SELECT COUNT(*) FROM runs WHERE project=? AND "
+                "state IN ('error','failed','aborted')", (project,)),
+            "runs_running_count": _count(
+                "SELECT COUNT(*) FROM runs WHERE project=? AND "
+                "state='running'", (project,)),
+            "files_count": _count(
+                "SELECT COUNT(*) FROM artifacts WHERE project=? AND "
+                "(kind NOT IN ('model','dataset') OR kind IS NULL)",
+                (project,)),
+            "models_count": _count(
+                "SELECT COUNT(*) FROM artifacts WHERE project=? AND "
+                "kind='model'", (project,)),
+            "datasets_count": _count(
+                "SELECT COUNT(*) FROM artifacts WHERE project=? AND "
+                "kind='dataset'", (project,)),
+            "feature_sets_count": _count(
+                "SELECT COUNT(*) FROM feature_sets WHERE project=?",
+                (project,)),
+            "schedules_count": _count(
+                "SELECT COUNT(*) FROM schedules WHERE project=?",
+                (project,)),
+            "model_endpoints_count": _count(
+                "SELECT COUNT(*) FROM model_endpoints WHERE project=?",
+                (project,)),
+        }
+        return {"name": project, "updated": now_iso(), **counts}
+
+    def list_project_summaries(self) -> list:
+        names = [p.get("metadata", {}).get("name", p.get("name", ""))
+                 for p in self.list_projects()]
+        return [self.compute_project_summary(n) for n in names if n]

@@ -137,8 +137,13 @@ class WorkerPool:
 
     def __init__(self, function_spec: dict, workers: int,
                  env: dict = None, max_workers: int = 0,
-                 scale_connections_per_worker: int = 16,
+                 scale_connections_per_worker: int = None,
                  scale_interval: float = 2.0):
+        from ..config import config
+
+        if scale_connections_per_worker is None:
+            scale_connections_per_worker = int(
+                config.runtimes.serving.autoscale_connections_per_worker)
         self.spec = function_spec
         self.workers = workers
         self.max_workers = max(max_workers, workers)

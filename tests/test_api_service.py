@@ -557,3 +557,40 @@ def test_datastore_profiles(client):
     client.delete("/api/v1/projects/dp/datastore-profiles/my-s3")
     assert client.get("/api/v1/projects/dp/datastore-profiles"
                       ).json()["profiles"] == []
+
+
+def test_sweep_service_caches(tmp_path):
+    """Pagination-cache + background-task TTL hygiene (reference
+    cleanup loops)."""
+    import datetime
+
+    from mlrun_amd.api.main import sweep_service_caches
+    from mlrun_amd.db.sqldb import SQLRunDB
+
+    db = SQLRunDB(str(tmp_path / "h.db"))
+    for i in range(30):
+        db.store_artifact(f"a{i}", {"kind": "model",
+                                    "metadata": {"key": f"a{i}"}},
+                          tree=f"t{i}", project="h")
+    _, token = db.paginated_list("list_artifacts", project="h",
+                                 page_size=10)
+    assert token
+    db.store_background_task("h", {
+        "metadata": {"name": "old-task", "project": "h",
+                     "updated": "2020-01-01T00:00:00"},
+        "status": {"state": "succeeded"}})
+    db.store_background_task("h", {
+        "metadata": {"name": "live-task", "project": "h",
+                     "updated": "2020-01-01T00:00:00"},
+        "status": {"state": "running"}})
+    future = datetime.datetime.now() + datetime.timedelta(days=400)
+    result = sweep_service_caches(db, now=future)
+    assert "h/old-task" in result["background_tasks"]
+    assert "h/live-task" not in result["background_tasks"]
+    # the idle pagination token is gone
+    import pytest as _pytest
+
+    from mlrun_amd.errors import MLRunNotFoundError
+
+    with _pytest.raises(MLRunNotFoundError):
+        db.paginated_list("list_artifacts", page_token=token)

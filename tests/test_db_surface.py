@@ -190,3 +190,22 @@ class TestSecretsAndBackgroundAliases:
     def test_runtime_resources(self, httpdb):
         resources = httpdb.list_runtime_resources("default")
         assert "gpu" in resources
+
+
+class TestProjectSummaries:
+    def test_summary_counts(self, httpdb):
+        httpdb.create_project({"metadata": {"name": "sums"}})
+        httpdb.store_run({"metadata": {"name": "r", "uid": "u1"},
+                          "status": {"state": "completed"}}, "u1",
+                         "sums")
+        httpdb.store_run({"metadata": {"name": "r", "uid": "u2"},
+                          "status": {"state": "error"}}, "u2", "sums")
+        httpdb.store_artifact("m1", {"kind": "model",
+                                     "metadata": {"key": "m1"}},
+                              uid="t1", project="sums")
+        summary = httpdb.get_project_summary("sums")
+        assert summary["runs_completed_recent_count"] == 1
+        assert summary["runs_failed_recent_count"] == 1
+        assert summary["models_count"] == 1
+        all_summaries = httpdb.list_project_summaries()
+        assert any(s["name"] == "sums" for s in all_summaries)
