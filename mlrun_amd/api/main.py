@@ -844,6 +844,50 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         body = db.get_function(name, project, tag=tag)
         return {"status": body.get("status", {})}
 
+    # ------------------------------------------------- grafana proxy
+    @app.get("/api/v1/grafana-proxy/model-endpoints")
+    async def grafana_proxy_health():
+        return {"status": "ok"}  # datasource test endpoint
+
+    @app.post("/api/v1/grafana-proxy/model-endpoints/search")
+    async def grafana_search(body: dict = None):
+        """Metric-name discovery (reference grafana_proxy.py /search):
+        returns the monitoring metric names recorded per endpoint."""
+        from ..model_monitoring import get_stream_processor
+
+        names = set()
+        for project in [p.get("metadata", {}).get("name", "default")
+                        for p in db.list_projects()] + ["default"]:
+            processor = get_stream_processor(project)
+            for endpoint_id in processor._endpoint_ids:
+                names.add(endpoint_id)
+        return sorted(names)
+
+    @app.post("/api/v1/grafana-proxy/model-endpoints/query")
+    async def grafana_query(body: dict):
+        """Grafana simple-json timeseries query (reference
+        grafana_proxy.py): targets name endpoint ids; datapoints come
+        from the in-memory monitoring TSDB snapshots."""
+        from ..model_monitoring import get_stream_processor
+
+        out = []
+        project = body.get("project", "default")
+        processor = get_stream_processor(project)
+        for target in body.get("targets") or []:
+            endpoint_id = target.get("target") if isinstance(
+                target, dict) else str(target)
+            metric = (target.get("metric", "count")
+                      if isinstance(target, dict) else "count")
+            window = str(target.get("window", "300")) if isinstance(
+                target, dict) else "300"
+            series = []
+            for ts, snapshot in processor.tsdb_series(endpoint_id):
+                value = (snapshot.get(window) or {}).get(metric, 0)
+                series.append([value, int(ts * 1000)])
+            out.append({"target": f"{endpoint_id}.{metric}.{window}",
+                        "datapoints": series})
+        return out
+
     # ---------------------------------------------- project summaries
     @app.get("/api/v1/project-summaries")
     async def list_project_summaries():

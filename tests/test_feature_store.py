@@ -493,3 +493,64 @@ class TestRoundTwoTargets:
 
         for kind in ("kafka", "redisnosql", "tsdb", "nosql", "sql"):
             assert get_target_from_spec(kind).kind == kind
+
+
+class TestAsofMergeProperty:
+    """Property check of the point-in-time join: for ANY event set the
+    merged value equals the latest feature row at-or-before the entity
+    timestamp (the guarantee the reference's 818-LoC merge logic
+    provides — retrieval/base.py)."""
+
+    def test_asof_matches_bruteforce(self, rundb):
+        import numpy as np
+        import pandas as pd
+
+        from hypothesis import given, settings
+        from hypothesis import strategies as st
+
+        from mlrun_amd import feature_store as fstore
+
+        EV = st.tuples(st.integers(0, 3),              # key
+                       st.integers(0, 1000),           # value
+                       st.integers(0, 100))            # ts (sec)
+
+        @settings(max_examples=25, deadline=None)
+        @given(st.lists(EV, min_size=1, max_size=30),
+               st.lists(st.tuples(st.integers(0, 3),
+                                  st.integers(0, 100)),
+                        min_size=1, max_size=10))
+        def check(events, queries):
+            fstore.reset_online_tables()
+            fset = fstore.FeatureSet("asofp", entities=["k"],
+                                     timestamp_key="ts")
+            base = pd.Timestamp("2026-01-01")
+            df = pd.DataFrame({
+                "k": [e[0] for e in events],
+                "v": [float(e[1]) for e in events],
+                "ts": [base + pd.Timedelta(seconds=e[2])
+                       for e in events]})
+            # drop exact (k, ts) duplicates: their asof pick is
+            # order-dependent in any engine
+            df = df.drop_duplicates(subset=["k", "ts"], keep="last")
+            fstore.ingest(fset, df, targets=["parquet"],
+                          overwrite=True)  # fresh file per example
+            vector = fstore.FeatureVector("vasofp",
+                                          features=["asofp.v"])
+            entity = pd.DataFrame({
+                "k": [q[0] for q in queries],
+                "event_time": [base + pd.Timedelta(seconds=q[1])
+                               for q in queries]})
+            entity = entity.drop_duplicates()
+            out = fstore.get_offline_features(
+                vector, entity_rows=entity,
+                entity_timestamp_column="event_time").to_dataframe()
+            for _, row in out.iterrows():
+                past = df[(df.k == row.k) &
+                          (df.ts <= row.event_time)]
+                if past.empty:
+                    assert pd.isna(row.v)
+                else:
+                    expect = past.sort_values("ts").iloc[-1].v
+                    assert row.v == expect, (row, past)
+
+        check()

@@ -164,3 +164,38 @@ class TestMonitoringPipeline:
         assert "histogram-data-drift" in app_results
         assert app_results["histogram-data-drift"][
             "general_drift_status"] == 2  # detected
+
+
+class TestGrafanaProxy:
+    def test_search_and_query(self, tmp_path):
+        from fastapi.testclient import TestClient
+
+        from mlrun_amd.api.main import create_app
+        from mlrun_amd.db.sqldb import SQLRunDB
+        from mlrun_amd.model_monitoring import (
+            ModelMonitoringEvent,
+            get_stream_processor,
+        )
+
+        processor = get_stream_processor("default")
+        for i in range(5):
+            processor.push(ModelMonitoringEvent(
+                endpoint_id="g-ep", model="m", latency_ms=10.0 + i))
+        processor.record_tsdb()
+        db = SQLRunDB(str(tmp_path / "g.db"))
+        app = create_app(db, with_scheduler=False)
+        with TestClient(app) as client:
+            assert client.get(
+                "/api/v1/grafana-proxy/model-endpoints"
+            ).json()["status"] == "ok"
+            found = client.post(
+                "/api/v1/grafana-proxy/model-endpoints/search",
+                json={}).json()
+            assert "g-ep" in found
+            series = client.post(
+                "/api/v1/grafana-proxy/model-endpoints/query",
+                json={"targets": [{"target": "g-ep",
+                                   "metric": "count",
+                                   "window": "300"}]}).json()
+            assert series[0]["target"] == "g-ep.count.300"
+            assert series[0]["datapoints"][-1][0] == 5
