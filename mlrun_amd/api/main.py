@@ -28,6 +28,16 @@ from ..errors import MLRunBaseError, err_to_status
 from ..utils import logger, now_iso
 
 
+
+def _validate_or_422(validator, body):
+    """Run a pydantic wire-schema validator; malformed shapes 422
+    (reference: every endpoint validates its pydantic body)."""
+    try:
+        validator(body)
+    except Exception as exc:
+        raise HTTPException(status_code=422, detail=str(exc))
+
+
 def _deep_update(target: dict, patch: dict):
     """Recursive dict merge (PATCH semantics — reference strategies)."""
     for key, value in patch.items():
@@ -332,6 +342,9 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     async def store_feature_set(project: str, name: str, request: Request,
                                 tag: str = ""):
         body = await request.json()
+        from ..common.schemas import validate_feature_set
+
+        _validate_or_422(validate_feature_set, body)
         return db.store_feature_set(body, name=name, project=project,
                                     tag=tag or None)
 
@@ -375,6 +388,9 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     async def store_model_endpoint(project: str, endpoint_id: str,
                                    request: Request):
         body = await request.json()
+        from ..common.schemas import validate_model_endpoint
+
+        _validate_or_422(validate_model_endpoint, body)
         db.store_model_endpoint(project, endpoint_id, body)
         return {}
 
@@ -462,6 +478,9 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
         """Named datastore configs (reference datastore_profile.py —
         connection parameters for s3/redis/... targets, minus the
         secrets which live in the project secret store)."""
+        from ..common.schemas import validate_datastore_profile
+
+        _validate_or_422(validate_datastore_profile, body)
         db.store_datastore_profile(project, body)
         return body
 
@@ -707,6 +726,9 @@ def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
     @app.put("/api/v1/projects/{project}/alerts/{name}")
     async def store_alert(project: str, name: str, request: Request):
         body = await request.json()
+        from ..common.schemas import validate_alert_config
+
+        _validate_or_422(validate_alert_config, body)
         db.store_alert_config(project, name, body)
         return {}
 

@@ -594,3 +594,52 @@ def test_sweep_service_caches(tmp_path):
 
     with _pytest.raises(MLRunNotFoundError):
         db.paginated_list("list_artifacts", page_token=token)
+
+
+class TestWireValidation:
+    """Expanded pydantic wire schemas: malformed shapes 422
+    (reference: per-endpoint pydantic bodies)."""
+
+    def test_feature_set_validation(self, client):
+        bad = {"metadata": {"name": "fs"},
+               "spec": {"entities": [{"no_name": True}]}}
+        assert client.put("/api/v1/projects/p/feature-sets/fs",
+                          json=bad).status_code == 422
+        good = {"metadata": {"name": "fs"},
+                "spec": {"entities": [{"name": "k"}],
+                         "aggregations": [
+                             {"name": "a", "column": "v",
+                              "operations": ["sum"],
+                              "windows": ["1h"]}]}}
+        assert client.put("/api/v1/projects/p/feature-sets/fs",
+                          json=good).status_code == 200
+
+    def test_aggregation_requires_ops_and_windows(self, client):
+        bad = {"metadata": {"name": "fs2"},
+               "spec": {"aggregations": [{"name": "a",
+                                          "column": "v"}]}}
+        assert client.put("/api/v1/projects/p/feature-sets/fs2",
+                          json=bad).status_code == 422
+
+    def test_alert_validation(self, client):
+        bad = {"name": "a2", "severity": "catastrophic"}
+        assert client.put("/api/v1/projects/p/alerts/a2",
+                          json=bad).status_code == 422
+        bad_notification = {"name": "a2", "severity": "high",
+                            "notifications": [{"kind": "carrier-pigeon"}]}
+        assert client.put("/api/v1/projects/p/alerts/a2",
+                          json=bad_notification).status_code == 422
+
+    def test_datastore_profile_validation(self, client):
+        assert client.put("/api/v1/projects/p/datastore-profiles",
+                          json={"type": "s3"}).status_code == 422
+        assert client.put("/api/v1/projects/p/datastore-profiles",
+                          json={"name": "s3main",
+                                "type": "s3"}).status_code == 200
+
+    def test_model_endpoint_validation(self, client):
+        bad = {"metadata": {"name": "ep"},
+               "spec": {"monitoring_mode": ["not-a-string"]}}
+        assert client.put(
+            "/api/v1/projects/p/model-endpoints/ep",
+            json=bad).status_code == 422
