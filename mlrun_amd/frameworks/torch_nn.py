@@ -59,6 +59,83 @@ class PyTorchModelServer(V2ModelServer):
         return outputs.float().cpu().tolist()
 
 
+class TrainingCallback:
+    """Epoch-hook protocol (reference frameworks/pytorch/callbacks
+    base): subclass and override; train() drives the hooks."""
+
+    def on_train_begin(self, interface):
+        pass
+
+    def on_epoch_end(self, interface, epoch: int, results: dict):
+        pass
+
+    def on_train_end(self, interface, history: dict):
+        pass
+
+
+class MLRunLoggingCallback(TrainingCallback):
+    """Per-epoch child-iteration results + final model logging
+    (reference mlrun_logging_callback.py:30) — rank-0 gated."""
+
+    def __init__(self, model_key: str = "model", log_model: bool = True):
+        self.model_key = model_key
+        self.log_model = log_model
+
+    def on_epoch_end(self, interface, epoch, results):
+        if interface.context is not None and interface.rank == 0:
+            for key, value in results.items():
+                interface.context.log_result(f"epoch_{epoch}_{key}",
+                                             value)
+
+    def on_train_end(self, interface, history):
+        if self.log_model:
+            interface.log_model(
+                self.model_key,
+                metrics={k: v[-1] for k, v in history.items() if v})
+
+
+class CheckpointCallback(TrainingCallback):
+    """Save a model checkpoint artifact every N epochs (reference
+    ModelHandler.save cadence; restartable-runs story)."""
+
+    def __init__(self, every: int = 1, key: str = "checkpoint"):
+        self.every = max(int(every), 1)
+        self.key = key
+
+    def on_epoch_end(self, interface, epoch, results):
+        if (epoch + 1) % self.every == 0:
+            interface.log_model(f"{self.key}-epoch-{epoch}")
+
+
+class EarlyStoppingCallback(TrainingCallback):
+    """Stop when a monitored metric fails to improve for `patience`
+    epochs (reference callbacks; train() checks `should_stop`)."""
+
+    def __init__(self, monitor: str = "loss", patience: int = 3,
+                 mode: str = "min"):
+        self.monitor = monitor
+        self.patience = patience
+        self.mode = mode
+        self.best = None
+        self.bad_epochs = 0
+        self.should_stop = False
+
+    def on_epoch_end(self, interface, epoch, results):
+        value = results.get(self.monitor)
+        if value is None:
+            return
+        better = self.best is None or (
+            value < self.best if self.mode == "min" else
+            value > self.best)
+        if better:
+            self.best = value
+            self.bad_epochs = 0
+        else:
+            self.bad_epochs += 1
+            if self.bad_epochs >= self.patience:
+                self.should_stop = True
+
+
 class MLRunTorchInterface:
     """Training-loop instrumentation: epoch loop + metric logging +
     distributed data parallel over RCCL (the Horovod-equivalent)."""
@@ -100,16 +177,21 @@ class MLRunTorchInterface:
     def train(self, train_loader, loss_fn, optimizer, epochs: int = 1,
               validation_loader=None, metric_fns: list = None,
               scheduler=None, use_amp: bool = False,
-              auto_log_model: bool = True, model_key: str = "model"):
+              auto_log_model: bool = True, model_key: str = "model",
+              callbacks: list = None):
         """The epoch loop (reference mlrun_interface.py:657 _train):
         auto DDP (if apply_mlrun set it up), per-rank shards via
         DistributedSampler, allreduce-mean epoch metrics, rank-0
-        logging, and final model artifact logging."""
+        logging, callbacks (epoch hooks / checkpoints / early stop),
+        and final model artifact logging."""
         model = self._ddp or self.model
         train_loader = self._shard_loader(train_loader)
         if validation_loader is not None:
             validation_loader = self._shard_loader(validation_loader)
         metric_fns = metric_fns or []
+        callbacks = callbacks or []
+        for callback in callbacks:
+            callback.on_train_begin(self)
         history: typing.Dict[str, list] = {}
         for epoch in range(epochs):
             model.train()
@@ -136,6 +218,15 @@ class MLRunTorchInterface:
             if self.context is not None and self.rank == 0:
                 self.context.log_results(results)
             logger.info("epoch done", **results)
+            stop = False
+            for callback in callbacks:
+                callback.on_epoch_end(self, epoch, results)
+                stop = stop or getattr(callback, "should_stop", False)
+            if stop:
+                logger.info("early stopping", epoch=epoch)
+                break
+        for callback in callbacks:
+            callback.on_train_end(self, history)
         if auto_log_model and self.context is not None:
             self.log_model(model_key,
                            metrics={k: v[-1] for k, v in history.items()})

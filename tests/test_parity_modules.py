@@ -403,3 +403,56 @@ class TestAliasEquivalence:
         assert SA is SB
         fn = mlrun.new_function(name="alias-fn", kind="local")
         assert type(fn).__module__.startswith("mlrun_amd")
+
+
+class TestTorchCallbacks:
+    """Reference pytorch callbacks analog: epoch hooks, checkpoints,
+    early stopping (frameworks/pytorch/callbacks)."""
+
+    def _train(self, ctx, callbacks, epochs=6):
+        import torch
+
+        from mlrun_amd.frameworks.torch_nn import apply_mlrun
+
+        torch.manual_seed(3)
+        model = torch.nn.Linear(4, 1)
+        iface = apply_mlrun(model, context=ctx, auto_ddp=False)
+        x, y = torch.randn(16, 4), torch.randn(16, 1)
+        return iface.train([(x, y)], torch.nn.MSELoss(),
+                           torch.optim.SGD(model.parameters(), lr=0.1),
+                           epochs=epochs, callbacks=callbacks,
+                           auto_log_model=False)
+
+    def test_logging_and_checkpoint_callbacks(self, rundb):
+        import mlrun_amd
+        from mlrun_amd.frameworks.torch_nn import (
+            CheckpointCallback,
+            MLRunLoggingCallback,
+        )
+
+        ctx = mlrun_amd.get_or_create_ctx("cb-test")
+        self._train(ctx, [MLRunLoggingCallback(log_model=True),
+                          CheckpointCallback(every=3)], epochs=6)
+        assert "epoch_0_loss" in ctx.results
+        assert "epoch_5_loss" in ctx.results
+        keys = {a.get("metadata", {}).get("key")
+                for a in rundb.list_artifacts(
+                    project=ctx.project or "default")}
+        assert "model" in keys
+        assert "checkpoint-epoch-2" in keys
+        assert "checkpoint-epoch-5" in keys
+
+    def test_early_stopping(self, rundb):
+        import mlrun_amd
+        from mlrun_amd.frameworks.torch_nn import EarlyStoppingCallback
+
+        ctx = mlrun_amd.get_or_create_ctx("es-test")
+
+        class _AlwaysWorse(EarlyStoppingCallback):
+            def on_epoch_end(self, interface, epoch, results):
+                results = dict(results, loss=1.0 + epoch)  # worsening
+                super().on_epoch_end(interface, epoch, results)
+
+        history = self._train(ctx, [_AlwaysWorse(patience=2)],
+                              epochs=50)
+        assert len(history["loss"]) < 50  # stopped early
