@@ -184,11 +184,13 @@ def main():
     # stdout — this prints to STDERR and a JSON file only, and a
     # watchdog hard-exits if a collective wedges so the DP record is
     # never lost.
-    if dist and on_gpu and world_size > 1 and \
+    # only at the full-node run (world 8): at N=2/4 the rider would
+    # eat the driver's scaling-sweep budget for a non-headline point
+    if dist and on_gpu and world_size >= 8 and \
             os.environ.get("MLRUN_BENCH_TP", "1") != "0":
         import threading
 
-        watchdog = threading.Timer(600.0, lambda: os._exit(0))
+        watchdog = threading.Timer(420.0, lambda: os._exit(0))
         watchdog.daemon = True
         watchdog.start()
         try:
