@@ -292,3 +292,39 @@ class TestAsyncRemoteFunction:
             server.graph.shutdown()
         finally:
             httpd.shutdown()
+
+
+class TestAsyncOverHttp:
+    def test_async_graph_served_over_http(self):
+        """engine="async" behind the real HTTP host: responder output
+        returns over the wire while the rest of the DAG continues."""
+        import requests
+
+        fn = mlrun_amd.new_function("t-async-http", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+        graph.to("tests.test_async_flow.Chain", name="s1").to(
+            "tests.test_async_flow.Chain", name="s2").respond().to(
+            "tests.test_async_flow.Slow", name="bg", delay=0.01)
+        addr = fn.deploy()
+        try:
+            resp = requests.post(addr + "/run", json=["x"], timeout=30)
+            assert resp.status_code == 200
+            assert resp.json() == ["x", "s1", "s2"]
+        finally:
+            fn.stop()
+
+    def test_queue_path_publishes_to_stream(self):
+        """A pathed queue step with no local consumers publishes to
+        the node-local stream (reference stream-target behavior)."""
+        from mlrun_amd.platforms import OutputStream
+
+        fn = mlrun_amd.new_function("t-async-q", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+        graph.to("Chain", name="s1").to("$queue", "qout",
+                                        path="monitoring-stream")
+        server = _mock_server(fn)
+        server.test(body=["e"], silent=True)
+        server.wait_for_completion()
+        records = OutputStream.get_stream("monitoring-stream").drain()
+        assert records, "nothing published to the stream"
+        server.graph.shutdown()
