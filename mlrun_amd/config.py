@@ -101,6 +101,10 @@ default_config = {
     "model_endpoint_monitoring": {
         "sample_percent": 100,
         "parquet_batching_max_events": 1024,
+        # window-ring placement: cpu | auto (GPU when available) |
+        # explicit device — the rings reuse the feature-store HIP
+        # window kernels
+        "device": "cpu",
     },
     "model_monitoring": {
         # drift classification thresholds (reference

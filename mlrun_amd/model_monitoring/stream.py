@@ -233,8 +233,22 @@ _processors: dict = {}
 _processors_lock = threading.Lock()
 
 
+def _monitoring_device() -> str:
+    """Ring placement: config model_endpoint_monitoring.device —
+    "cpu" (default; stats are tiny), "auto" (GPU when available —
+    shares the feature-store HIP window kernels), or an explicit
+    device string."""
+    device = str(config.model_endpoint_monitoring.get("device", "cpu"))
+    if device == "auto":
+        import torch
+
+        return "cuda:0" if torch.cuda.is_available() else "cpu"
+    return device or "cpu"
+
+
 def get_stream_processor(project: str = "default") -> EventStreamProcessor:
     with _processors_lock:
         if project not in _processors:
-            _processors[project] = EventStreamProcessor(project)
+            _processors[project] = EventStreamProcessor(
+                project, device=_monitoring_device())
         return _processors[project]

@@ -450,3 +450,34 @@ class TestRcclInGraphCapture:
                               timeout=600)
         assert proc.returncode == 0, proc.stderr[-3000:]
         assert "RCCL_IN_GRAPH_OK" in proc.stdout
+
+
+@requires_gpu
+class TestMonitoringRingsOnGPU:
+    def test_stream_stats_on_gpu_rings(self):
+        """model_endpoint_monitoring.device=auto places the 5m/1h
+        serving-stat rings in HBM and serves identical stats."""
+        import time
+
+        from mlrun_amd.model_monitoring.stream import (
+            EventStreamProcessor,
+            ModelMonitoringEvent,
+        )
+
+        gpu = EventStreamProcessor("gpu-mon", device="cuda:0")
+        cpu = EventStreamProcessor("cpu-mon", device="cpu")
+        now = time.time()
+        for i in range(64):
+            event = dict(endpoint_id="ep", model="m",
+                         latency_ms=float(i), timestamp=now - i,
+                         error="boom" if i % 8 == 0 else None)
+            gpu.push(ModelMonitoringEvent(**event))
+            cpu.push(ModelMonitoringEvent(**event))
+        a = gpu.endpoint_stats("ep", now_ts=now)
+        b = cpu.endpoint_stats("ep", now_ts=now)
+        assert gpu._latency_ring.ring.is_cuda
+        for window in ("300", "3600"):
+            for key in ("count", "error_count", "avg_latency_ms",
+                        "max_latency_ms", "min_latency_ms"):
+                assert abs(a[window][key] - b[window][key]) < 1e-3, (
+                    window, key, a[window], b[window])
