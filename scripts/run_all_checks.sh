@@ -9,6 +9,11 @@ echo "== CPU test tier =="
 python -m pytest tests -q -m "not gpu"
 echo "== ASan lane (C++ log collector) =="
 bash scripts/asan_check.sh
+echo "== examples =="
+for ex in examples/*.py; do
+  echo "-- $ex"
+  python "$ex" >/dev/null
+done
 echo "== bench contract (CPU tiny) =="
 python bench.py --steps 2 --warmup 1 | python -c "
 import json, sys
