@@ -203,6 +203,28 @@ def cmd_watch_stream(args):
     print(json.dumps(stats, indent=2, default=str))
 
 
+def cmd_migrate(args):
+    """Apply pending run-DB schema migrations (reference
+    `mlrun migrate` / trigger_migrations)."""
+    from .db import get_run_db
+
+    db = get_run_db(args.dbpath or None)
+    result = db.trigger_migrations()
+    print(json.dumps(result, indent=2))
+
+
+def cmd_summary(args):
+    from .db import get_run_db
+
+    db = get_run_db()
+    if args.project:
+        print(json.dumps(db.compute_project_summary(args.project),
+                         indent=2, default=str))
+    else:
+        print(json.dumps(db.list_project_summaries(), indent=2,
+                         default=str))
+
+
 def main(argv=None):
     parser = argparse.ArgumentParser(
         prog="mlrun_amd",
@@ -276,6 +298,15 @@ def main(argv=None):
     p.add_argument("endpoint")
     p.add_argument("--project", default="")
     p.set_defaults(func=cmd_watch_stream)
+
+    p = sub.add_parser("migrate",
+                       help="apply pending run-DB schema migrations")
+    p.add_argument("--dbpath", default="")
+    p.set_defaults(func=cmd_migrate)
+
+    p = sub.add_parser("summary", help="per-project entity counts")
+    p.add_argument("project", nargs="?", default="")
+    p.set_defaults(func=cmd_summary)
 
     sub.add_parser("version").set_defaults(func=cmd_version)
     sub.add_parser("config").set_defaults(func=cmd_config)

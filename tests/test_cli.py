@@ -211,3 +211,26 @@ print("REMOTE_OK", run.metadata.uid)
         main(["get", "schedules"])
         out = capsys.readouterr().out
         assert "nightly" in out and "0 3 * * *" in out
+
+
+def test_cli_migrate_and_summary(tmp_path, monkeypatch, capsys):
+    import json as _json
+
+    from mlrun_amd.__main__ import main
+    from mlrun_amd.config import config
+
+    monkeypatch.setattr(config, "base_dir", str(tmp_path))
+    main(["migrate"])
+    out = capsys.readouterr().out
+    assert _json.loads(out)["schema_version"] >= 4
+
+    from mlrun_amd.db import get_run_db
+
+    db = get_run_db()
+    db.store_project("cliproj", {"metadata": {"name": "cliproj"}})
+    db.store_run({"metadata": {"name": "r", "uid": "u"},
+                  "status": {"state": "completed"}}, "u", "cliproj")
+    main(["summary", "cliproj"])
+    out = capsys.readouterr().out
+    summary = _json.loads(out)
+    assert summary["runs_completed_recent_count"] == 1
