@@ -406,6 +406,9 @@ class GraphServerHost:
             self._uvicorn_server.should_exit = True
         if self._thread is not None:
             self._thread.join(timeout=5)
+        graph = getattr(self.server, "graph", None)
+        if graph is not None and hasattr(graph, "shutdown"):
+            graph.shutdown()  # async controller / queue workers
 
 
 def _free_port() -> int:
