@@ -209,3 +209,14 @@ class TestProjectSummaries:
         assert summary["models_count"] == 1
         all_summaries = httpdb.list_project_summaries()
         assert any(s["name"] == "sums" for s in all_summaries)
+
+
+class TestLogHelpers:
+    def test_watch_log_and_size(self, httpdb, capsys):
+        httpdb.store_run({"metadata": {"name": "wl", "uid": "w1"},
+                          "status": {"state": "completed"}}, "w1", "p")
+        httpdb.store_log("w1", "p", b"hello logs")
+        assert httpdb.get_log_size("w1", "p") == len(b"hello logs")
+        state = httpdb.watch_log("w1", project="p")
+        assert state == "completed"
+        assert "hello logs" in capsys.readouterr().out
