@@ -215,18 +215,23 @@ def main():
 
 
 def _bench_tp70b(dist, rank, world_size, local_rank, steps=3, warmup=2,
-                 batch=16, prompt_len=128, gen_tokens=32):
+                 batch=16, prompt_len=128, gen_tokens=32, cfg=None,
+                 device=None, use_graph=True):
     """Llama-3-70B decode at TP=world_size over xGMI, hipGraph-captured
     (BASELINE config 5).  Reuses the already-initialized process
-    group."""
+    group.  cfg/device overrides exist so the exact code path also
+    runs as a 2-rank gloo CPU test (tests/test_tensor_parallel.py)."""
     import torch
 
     from mlrun_amd.models.llama import LlamaConfig, LlamaDecodeEngine
 
-    cfg = LlamaConfig.llama3_70b()
-    engine = LlamaDecodeEngine(cfg, batch, device=f"cuda:{local_rank}",
+    cfg = cfg or LlamaConfig.llama3_70b()
+    device = device or f"cuda:{local_rank}"
+    on_gpu = device.startswith("cuda")
+    engine = LlamaDecodeEngine(cfg, batch, device=device,
                                tp_group=None, tp_rank=rank,
-                               tp_size=world_size, use_graph=True,
+                               tp_size=world_size,
+                               use_graph=use_graph and on_gpu,
                                seed=7)
     gen = torch.Generator().manual_seed(77)
 
@@ -237,16 +242,20 @@ def _bench_tp70b(dist, rank, world_size, local_rank, steps=3, warmup=2,
     for _ in range(warmup):
         engine.reset()
         engine.generate(make_prompts(), max_new_tokens=gen_tokens)
-    torch.cuda.synchronize()
+    if on_gpu:
+        torch.cuda.synchronize()
     dist.barrier()
     t0 = time.perf_counter()
     for _ in range(steps):
         engine.reset()
         engine.generate(make_prompts(), max_new_tokens=gen_tokens)
-    torch.cuda.synchronize()
+    if on_gpu:
+        torch.cuda.synchronize()
     dist.barrier()
     elapsed = time.perf_counter() - t0
-    t = torch.tensor([elapsed], dtype=torch.float64).cuda()
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if on_gpu:
+        t = t.cuda()
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t.item())
     if rank != 0:

@@ -113,3 +113,57 @@ class TestTensorParallel:
         assert run.status.results["prefill_max_err"] < 0.25, \
             run.status.results
         assert run.status.results["token_match"] >= 0.5
+
+
+RIDER_SCRIPT = textwrap.dedent("""
+    import os, sys
+    import torch
+    import torch.distributed as dist
+
+    sys.path.insert(0, os.getcwd())
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    from bench import _bench_tp70b
+    from mlrun_amd.models.llama import LlamaConfig
+
+    cfg = LlamaConfig.tiny(num_heads=4, num_kv_heads=2, hidden_size=512,
+                           intermediate_size=1024, vocab_size=2048)
+    result = _bench_tp70b(dist, rank, world, 0, steps=1, warmup=1,
+                          batch=2, prompt_len=6, gen_tokens=3,
+                          cfg=cfg, device="cpu")
+    if rank == 0:
+        assert result["value"] > 0, result
+        assert result["config"]["parallelism"] == f"tp{world}"
+        print("RIDER_OK", result["value"])
+    dist.destroy_process_group()
+""")
+
+
+class TestTpRiderPath:
+    def test_bench_tp70b_code_path_gloo(self, tmp_path):
+        """The TP rider in bench.py runs exactly once, unattended, on
+        the driver's 8-GPU box — exercise the EXACT function over
+        2-rank gloo with a tiny config so API breaks are caught
+        here."""
+        import subprocess
+        import sys
+
+        script = tmp_path / "rider.py"
+        script.write_text(RIDER_SCRIPT)
+        procs = []
+        import os as _os
+
+        for rank in range(2):
+            env = dict(_os.environ, RANK=str(rank), WORLD_SIZE="2",
+                       MASTER_ADDR="127.0.0.1", MASTER_PORT="29433",
+                       LOCAL_RANK=str(rank))
+            procs.append(subprocess.Popen(
+                [sys.executable, str(script)], env=env,
+                cwd=_os.getcwd(), stdout=subprocess.PIPE,
+                stderr=subprocess.STDOUT))
+        outs = [p.communicate(timeout=180)[0].decode() for p in procs]
+        assert all(p.returncode == 0 for p in procs), outs
+        assert "RIDER_OK" in outs[0] + outs[1], outs
