@@ -31,14 +31,36 @@ class HTTPRunDB(RunDBInterface):
 
     def api_call(self, method: str, path: str, params: dict = None,
                  body: typing.Any = None, json_body: dict = None,
-                 raw: bool = False, timeout: int = 45):
+                 raw: bool = False, timeout: int = 45,
+                 retries: int = None):
+        """One REST call; idempotent methods retry on transport errors
+        with exponential backoff when
+        ``httpdb.retry_api_call_on_exception`` is enabled (reference
+        httpdb.api_call retry semantics)."""
+        import time as _time
+
+        from ..config import config
+
         url = f"{self.base_url}/api/v1/{path.lstrip('/')}"
         headers = {}
         if self.token:
             headers["Authorization"] = f"Bearer {self.token}"
-        resp = self.session.request(
-            method, url, params=params, data=body, json=json_body,
-            headers=headers, timeout=timeout)
+        if retries is None:
+            retry_on = str(config.httpdb.retry_api_call_on_exception
+                           ) == "enabled"
+            retries = 2 if retry_on and method in ("GET", "HEAD") else 0
+        attempt = 0
+        while True:
+            try:
+                resp = self.session.request(
+                    method, url, params=params, data=body,
+                    json=json_body, headers=headers, timeout=timeout)
+                break
+            except (requests.ConnectionError, requests.Timeout):
+                if attempt >= retries:
+                    raise
+                _time.sleep(0.2 * (2 ** attempt))
+                attempt += 1
         if resp.status_code >= 400:
             try:
                 detail = resp.json().get("detail", resp.text)

@@ -220,3 +220,44 @@ class TestLogHelpers:
         state = httpdb.watch_log("w1", project="p")
         assert state == "completed"
         assert "hello logs" in capsys.readouterr().out
+
+
+class TestApiCallRetry:
+    def test_get_retries_on_connection_error(self, httpdb):
+        calls = {"n": 0}
+        orig = httpdb.session.request
+
+        def flaky(method, url, **kw):
+            calls["n"] += 1
+            if calls["n"] < 3:
+                import requests as _requests
+
+                raise _requests.ConnectionError("transient")
+            return orig(method, url, **kw)
+
+        httpdb.store_project("rp", {"metadata": {"name": "rp"}})
+        httpdb.session.request = flaky
+        try:
+            result = httpdb.get_project("rp")
+            assert result["metadata"]["name"] == "rp"
+            assert calls["n"] == 3  # two retries then success
+        finally:
+            httpdb.session.request = orig
+
+    def test_post_does_not_retry(self, httpdb):
+        import requests as _requests
+
+        calls = {"n": 0}
+
+        def always_fail(method, url, **kw):
+            calls["n"] += 1
+            raise _requests.ConnectionError("down")
+
+        orig = httpdb.session.request
+        httpdb.session.request = always_fail
+        try:
+            with pytest.raises(_requests.ConnectionError):
+                httpdb.api_call("POST", "runs")
+            assert calls["n"] == 1  # non-idempotent: no retry
+        finally:
+            httpdb.session.request = orig
