@@ -67,6 +67,8 @@ def main():
                         choices=["batch", "continuous"])
     parser.add_argument("--workers", type=int, default=0,
                         help="serving worker processes (L4 round-robin)")
+    parser.add_argument("--max-workers", type=int, default=0,
+                        help="autoscale ceiling (with_replicas max)")
     args = parser.parse_args()
 
     on_gpu = torch.cuda.is_available()
@@ -83,7 +85,11 @@ def main():
                  replicas=args.replicas,
                  weight_dtype="fp8w" if args.weights == "fp8" else "bf16",
                  scheduling=args.scheduling)
-    address = fn.deploy(workers=args.workers)
+    if args.max_workers:
+        fn.with_replicas(max(args.workers, 1), args.max_workers)
+        address = fn.deploy()
+    else:
+        address = fn.deploy(workers=args.workers)
     print(f"serving at {address}", file=sys.stderr)
 
     vocab = 1000
@@ -135,6 +141,9 @@ def main():
         "batch_window_ms": args.batch_window_ms,
         "scheduling": args.scheduling,
         "workers": args.workers,
+        "max_workers": args.max_workers,
+        "workers_final": len(fn._worker_pool.ports)
+        if getattr(fn, "_worker_pool", None) else args.workers or 1,
         "model": model,
         "weights": args.weights,
         "replicas": args.replicas,
