@@ -32,10 +32,61 @@ class RemoteRuntime(BaseRuntime):
         super().__init__(metadata, spec)
         self._host: typing.Optional[GraphServerHost] = None
         self._handler_fn = None
+        self._triggers: list = []
+        self._trigger_threads: list = []
+        self._trigger_stop = None
 
     def with_http(self, workers=None, port=0, host=None, **kwargs):
         self.spec.build["port"] = port
         return self
+
+    def add_trigger(self, name: str, spec) -> "RemoteRuntime":
+        """Attach a trigger (reference RemoteRuntime.add_trigger):
+        cron triggers invoke the deployed handler on schedule — the
+        nuclio cron-trigger analog, node-local."""
+        if isinstance(spec, str):
+            spec = {"kind": "cron", "interval": spec}
+        self._triggers.append({"name": name, **spec})
+        return self
+
+    def add_cron_trigger(self, name: str, interval: str = "",
+                         body=None, path: str = "/"):
+        """Sugar: interval like "10s"/"5m" (nuclio cron interval
+        form)."""
+        return self.add_trigger(name, {"kind": "cron",
+                                       "interval": interval,
+                                       "body": body, "path": path})
+
+    def _start_triggers(self):
+        import threading
+
+        from ..feature_store.feature_set import parse_span
+
+        if not self._triggers:
+            return
+        self._trigger_stop = threading.Event()
+
+        def runner(trigger):
+            interval = trigger.get("interval", "1m")
+            seconds = parse_span(interval) if not str(
+                interval).replace(".", "").isdigit() else float(interval)
+            while not self._trigger_stop.wait(seconds):
+                try:
+                    self.invoke(trigger.get("path", "/"),
+                                body=trigger.get("body"))
+                except Exception as exc:
+                    logger.warning("cron trigger invoke failed",
+                                   trigger=trigger["name"],
+                                   error=str(exc))
+
+        for trigger in self._triggers:
+            if trigger.get("kind") != "cron":
+                continue
+            thread = threading.Thread(target=runner, args=(trigger,),
+                                      daemon=True,
+                                      name=f"trigger-{trigger['name']}")
+            thread.start()
+            self._trigger_threads.append(thread)
 
     def _resolve_handler(self):
         if self._handler_fn is not None:
@@ -82,6 +133,7 @@ class RemoteRuntime(BaseRuntime):
         self.status.state = "ready"
         self.status.address = self._host.address
         self.status.external_invocation_urls = [self._host.address]
+        self._start_triggers()
         logger.info("remote function deployed", address=self._host.address)
         return self._host.address
 
@@ -104,6 +156,11 @@ class RemoteRuntime(BaseRuntime):
             return resp.content
 
     def stop(self):
+        if self._trigger_stop is not None:
+            self._trigger_stop.set()
+            for thread in self._trigger_threads:
+                thread.join(timeout=3)
+            self._trigger_threads = []
         if self._host is not None:
             self._host.stop()
             self._host = None

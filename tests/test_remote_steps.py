@@ -129,3 +129,34 @@ class TestWebhookNotificationLive:
         assert received[0]["message"] == "run done"
         assert received[0]["runs"][0]["metadata"]["name"] == "r1"
         assert "*error* run failed" in received[1]["text"]
+
+
+class TestCronTrigger:
+    def test_cron_trigger_invokes_handler(self):
+        """Nuclio cron-trigger analog: the deployed handler fires on
+        the configured interval (reference RemoteRuntime.add_trigger)."""
+        import time
+
+        import mlrun_amd
+
+        hits = []
+
+        def handler(context, event):
+            hits.append(event.body)
+            return {"n": len(hits)}
+
+        fn = mlrun_amd.new_function("cronfn", kind="remote")
+        fn._handler_fn = handler
+        fn.add_cron_trigger("tick", interval="1", body={"from": "cron"})
+        fn.deploy()
+        try:
+            deadline = time.time() + 15
+            while len(hits) < 2 and time.time() < deadline:
+                time.sleep(0.2)
+            assert len(hits) >= 2, hits
+            assert {"from": "cron"} in hits
+        finally:
+            fn.stop()
+        count = len(hits)
+        time.sleep(1.5)
+        assert len(hits) == count  # stopped cleanly
