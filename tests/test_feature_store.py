@@ -554,3 +554,113 @@ class TestAsofMergeProperty:
                     assert row.v == expect, (row, past)
 
         check()
+
+
+class TestTransformStepBreadth:
+    """Reference tests/feature-store step-semantics analogs."""
+
+    def test_mapvalues_with_original(self):
+        import pandas as pd
+
+        from mlrun_amd import feature_store as fstore
+
+        df = pd.DataFrame({"a": [1, 7, 12]})
+        out = fstore.MapValues(
+            mapping={"a": {"ranges": {"low": [0, 5],
+                                      "mid": [5, 10],
+                                      "high": [10, "inf"]}}},
+            with_original_features=True).do(df)
+        assert list(out["a"]) == [1, 7, 12]  # original kept
+        assert list(out["a_mapped"]) == ["low", "mid", "high"]
+
+    def test_mapvalues_negative_infinity_range(self):
+        import pandas as pd
+
+        from mlrun_amd import feature_store as fstore
+
+        df = pd.DataFrame({"t": [-40.0, 15.0, 60.0]})
+        out = fstore.MapValues(mapping={"t": {"ranges": {
+            "cold": ["-inf", 0], "warm": [0, 30],
+            "hot": [30, "inf"]}}}).do(df)
+        assert list(out["t"]) == ["cold", "warm", "hot"]
+
+    def test_imputer_per_column_mapping_beats_method(self):
+        import numpy as np
+        import pandas as pd
+
+        from mlrun_amd import feature_store as fstore
+
+        df = pd.DataFrame({"a": [1.0, np.nan, 3.0],
+                           "b": [np.nan, 2.0, 4.0]})
+        out = fstore.Imputer(method="avg",
+                             mapping={"a": -1.0}).do(df)
+        assert list(out["a"]) == [1.0, -1.0, 3.0]  # mapping wins
+        assert list(out["b"]) == [3.0, 2.0, 4.0]   # avg fallback
+
+    def test_onehot_unseen_category_all_zero(self):
+        import pandas as pd
+
+        from mlrun_amd import feature_store as fstore
+
+        df = pd.DataFrame({"cat": ["x", "z"]})
+        out = fstore.OneHotEncoder(
+            mapping={"cat": ["x", "y"]}).do(df)
+        assert list(out["cat_x"]) == [1, 0]
+        assert list(out["cat_y"]) == [0, 0]  # unseen 'z' -> all zeros
+        assert "cat" not in out.columns
+
+    def test_onehot_sanitizes_category_names(self):
+        import pandas as pd
+
+        from mlrun_amd import feature_store as fstore
+
+        df = pd.DataFrame({"c": ["a b", "c-d"]})
+        out = fstore.OneHotEncoder(mapping={"c": ["a b", "c-d"]}).do(df)
+        assert "c_a_b" in out.columns and "c_c_d" in out.columns
+
+    def test_date_extractor_parts(self):
+        import pandas as pd
+
+        from mlrun_amd import feature_store as fstore
+
+        df = pd.DataFrame({"ts": pd.to_datetime(
+            ["2026-09-12 13:45", "2026-01-01 00:10"])})
+        out = fstore.DateExtractor(
+            parts=["day_of_week", "hour", "month", "year",
+                   "week_of_year"],
+            timestamp_col="ts").do(df)
+        assert list(out["ts_day_of_week"]) == [5, 3]  # Sat, Thu
+        assert list(out["ts_hour"]) == [13, 0]
+        assert list(out["ts_month"]) == [9, 1]
+        assert list(out["ts_year"]) == [2026, 2026]
+        assert list(out["ts_week_of_year"]) == [37, 1]
+
+    def test_validator_drops_non_numeric(self):
+        import pandas as pd
+
+        from mlrun_amd import feature_store as fstore
+
+        df = pd.DataFrame({"v": ["1", "oops", "3"]})
+        out = fstore.FeaturesetValidator(columns=["v"]).do(df)
+        assert len(out) == 2
+
+    def test_steps_chain_in_ingestion_graph(self, rundb):
+        import numpy as np
+        import pandas as pd
+
+        from mlrun_amd import feature_store as fstore
+
+        fstore.reset_online_tables()
+        fset = fstore.FeatureSet("chain", entities=["k"])
+        fset.graph.to(fstore.Imputer(method="avg")).to(
+            fstore.OneHotEncoder(mapping={"cat": ["x", "y"]})).to(
+            fstore.DropFeatures(features=["junk"]))
+        df = pd.DataFrame({"k": ["a", "b"],
+                           "v": [1.0, np.nan],
+                           "cat": ["x", "y"],
+                           "junk": [0, 0]})
+        out = fstore.ingest(fset, df, targets=["parquet"],
+                            overwrite=True)
+        assert "junk" not in out.columns
+        assert list(out["cat_x"]) == [1, 0]
+        assert out["v"].notna().all()
