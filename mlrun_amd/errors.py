@@ -14,8 +14,13 @@ class MLRunBaseError(Exception):
     """Base for all framework errors."""
 
 
-class MLRunInvalidArgumentError(MLRunBaseError, ValueError):
+class MLRunBadRequestError(MLRunBaseError):
     pass
+
+
+class MLRunInvalidArgumentError(MLRunBadRequestError, ValueError):
+    """Invalid user input (maps to HTTP 400, like the reference's
+    MLRunInvalidArgumentError -> BadRequest)."""
 
 
 class MLRunNotFoundError(MLRunBaseError):
@@ -27,10 +32,6 @@ class MLRunConflictError(MLRunBaseError):
 
 
 class MLRunAccessDeniedError(MLRunBaseError):
-    pass
-
-
-class MLRunBadRequestError(MLRunBaseError):
     pass
 
 
@@ -51,10 +52,6 @@ class MLRunMissingDependencyError(MLRunBaseError, ImportError):
 
 
 class MLRunIncompatibleVersionError(MLRunBaseError):
-    pass
-
-
-class MLRunMissingDependencyError(MLRunBaseError):
     pass
 
 

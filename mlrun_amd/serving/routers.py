@@ -15,7 +15,7 @@ import concurrent.futures
 import copy
 import json
 
-from ..errors import MLRunInvalidArgumentError
+from ..errors import MLRunInvalidArgumentError, MLRunNotFoundError
 
 
 class BaseModelRouter:
@@ -79,7 +79,7 @@ class BaseModelRouter:
             return event
         route = self.routes.get(model)
         if route is None:
-            raise MLRunInvalidArgumentError(
+            raise MLRunNotFoundError(  # -> HTTP 404 at the host
                 f"model {model} not found in router "
                 f"(available: {list(self.routes)})")
         return route.run(event)

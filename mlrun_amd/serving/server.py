@@ -257,10 +257,28 @@ class GraphServer(ModelObj):
 
 
 class _ErrorResponse:
-    def __init__(self, error, origin=None, status_code=500):
+    def __init__(self, error, origin=None, status_code=None):
         self.body = {"error": error, "origin_state": origin}
         self.error = error
-        self.status_code = status_code
+        self.status_code = status_code or _status_from_error(error)
+
+
+def _status_from_error(error_text: str) -> int:
+    """Map a step error (formatted "ExcClass: message") back to an
+    HTTP status (reference: unknown model -> 404, bad request ->
+    400).  The class is resolved from the errors module so subclasses
+    (e.g. MLRunInvalidArgumentError -> 400) map correctly."""
+    from .. import errors as errors_mod
+
+    cls_name = str(error_text).split(":", 1)[0]
+    exc_cls = getattr(errors_mod, cls_name, None)
+    if isinstance(exc_cls, type) and issubclass(exc_cls, BaseException):
+        for code, base in errors_mod.STATUS_ERRORS.items():
+            if issubclass(exc_cls, base):
+                return code
+    if cls_name in ("ValueError", "KeyError", "TypeError"):
+        return 400
+    return 500
 
     def __repr__(self):
         return f"ErrorResponse({self.error!r})"
@@ -348,7 +366,8 @@ class GraphServerHost:
 
             response = await anyio.to_thread.run_sync(
                 lambda: graph_server.run(event, get_body=False))
-            status = "500" if isinstance(response, _ErrorResponse) \
+            status = str(response.status_code) \
+                if isinstance(response, _ErrorResponse) \
                 else "200"
             if self._metric_requests is not None:
                 route = "/" + full_path.split("/")[0]
@@ -357,7 +376,8 @@ class GraphServerHost:
                 self._metric_latency.labels(path=route).observe(
                     _time.perf_counter() - handle_start)
             if isinstance(response, _ErrorResponse):
-                return Response(json.dumps(response.body), status_code=500,
+                return Response(json.dumps(response.body),
+                                status_code=response.status_code,
                                 media_type="application/json")
             body_out = response.body if response is not None else ""
             if inspect.isgenerator(body_out) or (

@@ -661,3 +661,37 @@ class TestV2ModelListing:
         assert sorted(out["models"]) == ["m1", "m2"]
         meta = server.test("/v2/models/m1", body=None, method="GET")
         assert meta["name"] == "m1"
+
+
+class TestErrorStatusMapping:
+    def test_unknown_model_404_over_http(self):
+        import requests
+
+        import mlrun_amd
+
+        fn = _serving_fn()
+        fn.add_model("m", class_name=EchoModel)
+        addr = fn.deploy()
+        try:
+            resp = requests.post(addr + "/v2/models/nope/infer",
+                                 json={"inputs": [1]}, timeout=30)
+            assert resp.status_code == 404
+            assert "not found" in resp.json()["error"]
+            # bad input shape -> 400 family, not 500
+            resp = requests.post(addr + "/v2/models/m/infer",
+                                 data="definitely-not-json{{{",
+                                 headers={"content-type":
+                                          "application/json"},
+                                 timeout=30)
+            assert resp.status_code in (200, 400)
+        finally:
+            fn.stop()
+
+    def test_mock_server_status_codes(self):
+        fn = _serving_fn()
+        fn.add_model("m", class_name=EchoModel)
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/ghost/infer",
+                           body={"inputs": [1]}, silent=True,
+                           get_body=False)
+        assert resp.status_code == 404
