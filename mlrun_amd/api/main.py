@@ -26,7 +26,7 @@ from ..config import config
 from ..db.sqldb import SQLRunDB
 from ..errors import MLRunBaseError, err_to_status
 from ..utils import logger, now_iso
-
+from .scheduler import Scheduler
 
 
 def _validate_or_422(validator, body):
@@ -46,7 +46,6 @@ def _deep_update(target: dict, patch: dict):
         else:
             target[key] = value
     return target
-from .scheduler import Scheduler
 
 
 def create_app(db: SQLRunDB = None, with_scheduler: bool = True) -> FastAPI:
