@@ -160,3 +160,67 @@ class TestCronTrigger:
         count = len(hits)
         time.sleep(1.5)
         assert len(hits) == count  # stopped cleanly
+
+
+class TestRemoteStepResilience:
+    def test_retries_then_succeeds(self):
+        """RemoteStep retries transient failures (reference
+        RemoteStep retries arg)."""
+        import threading
+
+        from http.server import BaseHTTPRequestHandler, HTTPServer
+
+        import mlrun_amd
+        from mlrun_amd.serving.remote import RemoteStep
+
+        attempts = []
+
+        class Handler(BaseHTTPRequestHandler):
+            def do_POST(self):
+                attempts.append(1)
+                if len(attempts) < 3:
+                    self.send_response(503)
+                    self.end_headers()
+                    return
+                self.send_response(200)
+                self.send_header("content-type", "application/json")
+                self.end_headers()
+                self.wfile.write(b'{"ok": true}')
+
+            def log_message(self, *a):
+                pass
+
+        httpd = HTTPServer(("127.0.0.1", 0), Handler)
+        port = httpd.server_address[1]
+        threading.Thread(target=httpd.serve_forever,
+                         daemon=True).start()
+        try:
+            step = RemoteStep(url=f"http://127.0.0.1:{port}/x",
+                              retries=3)
+
+            class _Ev:
+                path = "/x"
+                id = "1"
+                body = {"q": 1}
+
+            out = step.do_event(_Ev())
+            assert out.body == {"ok": True}
+            assert len(attempts) == 3
+        finally:
+            httpd.shutdown()
+
+    def test_exhausted_retries_raise(self):
+        from mlrun_amd.serving.remote import RemoteStep
+
+        step = RemoteStep(url="http://127.0.0.1:9/never", retries=1,
+                          timeout=1)
+
+        class _Ev:
+            path = "/never"
+            id = "1"
+            body = {}
+
+        import pytest as _pytest
+
+        with _pytest.raises(Exception):
+            step.do_event(_Ev())
