@@ -97,8 +97,31 @@ class GraphContext:
         return ""
 
     def push_error(self, event, message, source=None, **kwargs):
+        """Log AND publish failed events to the configured error
+        stream (reference GraphContext.push_error -> error stream;
+        node-locally an OutputStream queue)."""
         self.logger.error(f"graph error from {source}: {message}",
                           event_id=getattr(event, "id", None))
+        stream = getattr(self.server, "_error_stream_object", None) \
+            if self.server else None
+        if stream is None and self.server and \
+                getattr(self.server, "error_stream", None):
+            from ..platforms import OutputStream
+
+            stream = OutputStream(self.server.error_stream)
+            self.server._error_stream_object = stream
+        if stream is not None:
+            try:
+                stream.push([{
+                    "error": message, "source": source,
+                    "event_id": getattr(event, "id", None),
+                    "path": getattr(event, "path", ""),
+                    "body": getattr(event, "body", None)}])
+            except Exception as exc:
+                # a broken error stream must never mask the original
+                # failure (reference test_push_error contract)
+                self.logger.error("error-stream push failed",
+                                  error=str(exc))
 
 
 class GraphServer(ModelObj):
@@ -200,6 +223,10 @@ class GraphServer(ModelObj):
         if response is None:
             return None
         if getattr(response, "error", None):
+            if self.error_stream and server_context:
+                server_context.push_error(
+                    event, response.error,
+                    source=getattr(response, "origin_state", None))
             return _ErrorResponse(response.error,
                                   origin=getattr(response, "origin_state",
                                                  None))

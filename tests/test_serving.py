@@ -695,3 +695,48 @@ class TestErrorStatusMapping:
                            body={"inputs": [1]}, silent=True,
                            get_body=False)
         assert resp.status_code == 404
+
+
+class TestErrorStream:
+    def test_failed_events_pushed_to_error_stream(self):
+        """server.error_stream receives failed events (reference
+        test_push_error / _init_async_objects error-stream wiring)."""
+        import mlrun_amd
+        from mlrun_amd.platforms import OutputStream
+
+        def boom(body):
+            raise ValueError("nope")
+
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(handler=boom, name="boom")
+        server = fn.to_mock_server(namespace={"boom": boom})
+        server.error_stream = "errors-stream"
+        server.test(body={"x": 1}, silent=True)
+        records = OutputStream.get_stream("errors-stream").drain()
+        assert records, "error stream got nothing"
+        import json as _json
+
+        record = _json.loads(records[0]) if isinstance(
+            records[0], (str, bytes)) else records[0]
+        assert "nope" in record["error"]
+
+    def test_raising_error_stream_does_not_mask(self):
+        """A broken error-stream object must not crash serving
+        (reference test_push_error _DummyStreamRaiser)."""
+        def boom(body):
+            raise ValueError("original")
+
+        fn = _serving_fn()
+        graph = fn.set_topology("flow")
+        graph.to(handler=boom, name="boom")
+        server = fn.to_mock_server(namespace={"boom": boom})
+        server.error_stream = "dummy:///nothing"
+
+        class _Raiser:
+            def push(self, data):
+                raise RuntimeError("stream down")
+
+        server._error_stream_object = _Raiser()
+        resp = server.test(body={}, silent=True, get_body=False)
+        assert "original" in str(resp.body)
