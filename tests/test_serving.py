@@ -740,3 +740,21 @@ class TestErrorStream:
         server._error_stream_object = _Raiser()
         resp = server.test(body={}, silent=True, get_body=False)
         assert "original" in str(resp.body)
+
+
+class TestAsyncLoadMode:
+    def test_lazy_model_load(self):
+        """load_mode="async": models load on FIRST event, not at init
+        (reference v2 async load mode)."""
+        import mlrun_amd
+
+        fn = _serving_fn()
+        fn.spec.load_mode = "async"
+        fn.set_topology("router")
+        fn.add_model("m", class_name=EchoModel, model_path=".")
+        server = fn.to_mock_server()
+        step = server.graph.steps["router"].routes["m"]
+        assert step._object.ready is False  # not loaded yet
+        resp = server.test("/v2/models/m/infer", body={"inputs": [2]})
+        assert resp["outputs"] == [4]
+        assert step._object.ready is True   # loaded on demand
