@@ -481,3 +481,43 @@ class TestMonitoringRingsOnGPU:
                         "max_latency_ms", "min_latency_ms"):
                 assert abs(a[window][key] - b[window][key]) < 1e-3, (
                     window, key, a[window], b[window])
+
+
+@requires_gpu
+class TestAsyncEngineWithGpuModel:
+    def test_async_flow_serves_gpu_model(self):
+        """engine="async" with a CUDA model step: step bodies run on
+        per-step executor threads — decode must still work and the
+        responder contract must hold."""
+        import mlrun_amd
+        from mlrun_amd.models.llama import LlamaServer
+
+        fn = mlrun_amd.new_function(name="async-gpu", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+
+        class Tag:
+            def __init__(self, context=None, name=None):
+                pass
+
+            def do(self, event):
+                event.body = dict(event.body)
+                event.body.setdefault("tags", []).append("pre")
+                return event
+
+        graph.to(Tag, name="pre").to(
+            LlamaServer, name="gen", config="tiny", batch_size=2,
+            max_new_tokens=4, device="cuda:0").respond()
+        server = fn.to_mock_server(namespace={"Tag": Tag})
+        try:
+            resp = server.test("/infer",
+                               body={"inputs": [[1, 2, 3]],
+                                     "max_tokens": 4})
+            assert len(resp["outputs"]) == 1
+            assert len(resp["outputs"][0]) == 4
+            # a second event reuses the loaded engine
+            resp2 = server.test("/infer",
+                                body={"inputs": [[1, 2, 3]],
+                                      "max_tokens": 4})
+            assert resp2["outputs"] == resp["outputs"]  # greedy
+        finally:
+            server.graph.shutdown()
