@@ -236,13 +236,11 @@ class OnlineVectorService:
             # imputation by feature name order
             names = [aliases.get(c, c) for _, cols, aliases in self._groups
                      for c in cols]
+            default = self.impute_policy.get("*")
             for j, name in enumerate(names):
-                if name in self.impute_policy:
-                    col = out[:, j]
-                    col[np.isnan(matrix[:, j])] = self.impute_policy[name]
-                else:
-                    col = out[:, j]
-                    col[np.isnan(matrix[:, j])] = None
+                fill = self.impute_policy.get(name, default)
+                col = out[:, j]
+                col[np.isnan(matrix[:, j])] = fill
             return out.tolist()
         results = [dict() for _ in entity_rows]
         ordered_names: typing.List[str] = []
@@ -261,11 +259,15 @@ class OnlineVectorService:
                 name = aliases.get(col, col)
                 if name not in ordered_names:
                     ordered_names.append(name)
-        # imputation
+        # imputation ("*" = default for every feature — reference
+        # impute_policy wildcard)
+        default = self.impute_policy.get("*")
         for out in results:
             for key, value in list(out.items()):
                 if value is None and key in self.impute_policy:
                     out[key] = self.impute_policy[key]
+                elif value is None and default is not None:
+                    out[key] = default
         if as_list:
             return [[out.get(name) for name in ordered_names]
                     for out in results]

@@ -664,3 +664,30 @@ class TestTransformStepBreadth:
         assert "junk" not in out.columns
         assert list(out["cat_x"]) == [1, 0]
         assert out["v"].notna().all()
+
+
+class TestVectorServiceEdges:
+    def test_as_list_order_and_imputation(self):
+        """as_list rows follow the vector's feature order; impute
+        policy fills unknown-entity cells in BOTH output shapes."""
+        df = pd.DataFrame({
+            "customer": ["alice", "bob"],
+            "amount": [10.0, 20.0],
+            "ts": pd.to_datetime([time.time() - 10] * 2, unit="s"),
+        })
+        fset = fstore.FeatureSet("edge", entities=["customer"],
+                                 timestamp_key="ts")
+        fset.add_aggregation("amount", ["sum", "count"], ["1h"], "10m")
+        fstore.ingest(fset, df)
+        vector = fstore.FeatureVector(
+            "vedge", features=["edge.amount_count_1h",
+                               "edge.amount_sum_1h"])
+        vector.metadata.project = "default"
+        svc = fstore.get_online_feature_service(
+            vector, impute_policy={"*": -9.0})
+        rows = svc.get([{"customer": "alice"},
+                        {"customer": "ghost"}], as_list=True)
+        assert rows[0] == [1.0, 10.0]     # count first, sum second
+        assert rows[1] == [-9.0, -9.0]    # imputed for unknown entity
+        dicts = svc.get([{"customer": "ghost"}])
+        assert dicts[0]["amount_sum_1h"] == -9.0
