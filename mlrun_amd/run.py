@@ -95,9 +95,22 @@ def code_to_function(name: str = "", project: str = "", tag: str = "",
                      image: str = "", code_output: str = "", embed_code=True,
                      description: str = "", requirements=None,
                      categories=None, labels=None,
-                     with_doc=True, ignored_tags=None) -> "BaseRuntime":
+                     with_doc=True, ignored_tags=None,
+                     code_: str = None) -> "BaseRuntime":
     """Package a python file (or the calling notebook's code) into a
-    function object."""
+    function object.  ``code_`` embeds source directly (reference
+    code_to_function code_ param) instead of reading a file."""
+    if code_:
+        fn = new_function(name=name or "fn", project=project, tag=tag,
+                          kind=kind or "job", image=image,
+                          handler=handler)
+        fn.spec.description = description
+        fn.spec.build["functionSourceCode"] = code_
+        if requirements:
+            fn.spec.build["requirements"] = requirements
+        if labels:
+            fn.metadata.labels.update(labels)
+        return fn
     if not filename:
         raise MLRunInvalidArgumentError(
             "filename is required (notebook capture is not supported in "
@@ -150,7 +163,8 @@ def _extract_entry_points(fn, filename):
 
 
 def import_function(url: str = "", project: str = "", name: str = "",
-                    new_name: str = "") -> "BaseRuntime":
+                    new_name: str = "", secrets=None,
+                    db="") -> "BaseRuntime":
     """Load a function object from a yaml file, db:// reference, or
     hub://  (parity: reference run.py:330)."""
     import yaml
@@ -163,7 +177,8 @@ def import_function(url: str = "", project: str = "", name: str = "",
         fn_name, _, tag = rest.partition(":")
         from .db import get_run_db
 
-        struct = get_run_db().get_function(fn_name, proj, tag=tag or "latest")
+        struct = get_run_db(db or None).get_function(
+            fn_name, proj, tag=tag or "latest")
         fn = new_function(runtime=struct)
     elif url.startswith("hub://"):
         from .hub import get_hub_function
@@ -203,7 +218,8 @@ def function_to_module(code: str, workdir=None, secrets=None, silent=False):
 
 def get_or_create_ctx(name: str, event=None, spec=None, with_env: bool = True,
                       rundb: str = "", project: str = "",
-                      upload_artifacts=False) -> MLClientCtx:
+                      upload_artifacts=False,
+                      labels: dict = None) -> MLClientCtx:
     """Entry point for user scripts: returns the active run context.
 
     Inside a framework-launched run (local subprocess, job process, or
@@ -231,6 +247,8 @@ def get_or_create_ctx(name: str, event=None, spec=None, with_env: bool = True,
         db = create_run_db(rundb)
     ctx = MLClientCtx.from_dict(spec_struct, rundb=db, autocommit=True,
                                 tmp=tmpfile)
+    for key, value in (labels or {}).items():
+        ctx.set_label(key, value)
     return ctx
 
 
