@@ -430,11 +430,14 @@ class ArtifactManager:
                   framework=None, algorithm=None, metrics=None,
                   parameters=None, inputs=None, outputs=None, tag="",
                   extra_data=None, labels=None, upload=None,
+                  feature_vector=None, feature_weights=None,
                   **kwargs) -> ModelArtifact:
         model = ModelArtifact(
             key, body=body, model_file=model_file, model_dir=model_dir,
             framework=framework, algorithm=algorithm, metrics=metrics,
             parameters=parameters, inputs=inputs, outputs=outputs,
+            feature_vector=feature_vector,
+            feature_weights=feature_weights,
             project=producer.project)
         if model_dir and not model_file:
             # pick first file in dir as the model file

@@ -256,7 +256,10 @@ class MLClientCtx:
 
     def log_artifact(self, item, body=None, local_path="", artifact_path=None,
                      tag="", viewer=None, target_path="", format=None,
-                     upload=None, labels=None, db_key=None, **kwargs):
+                     upload=None, labels=None, db_key=None,
+                     src_path: str = None, **kwargs):
+        # src_path: deprecated reference alias of local_path
+        local_path = local_path or src_path or ""
         if not self.is_logging_worker():
             return None
         self._init_db()
@@ -285,9 +288,30 @@ class MLClientCtx:
                   model_file=None, algorithm=None, metrics=None,
                   parameters=None, artifact_path=None, upload=None,
                   labels=None, inputs=None, outputs=None, extra_data=None,
-                  db_key=None, **kwargs):
+                  db_key=None, training_set=None, label_column=None,
+                  feature_vector: str = None, feature_weights: list = None,
+                  **kwargs):
+        """training_set/label_column derive the model's input/output
+        feature lists (reference log_model behavior); feature_vector/
+        feature_weights record the serving enrichment source."""
         if not self.is_logging_worker():
             return None
+        if training_set is not None and hasattr(training_set, "columns"):
+            label_columns = [label_column] if isinstance(
+                label_column, str) else list(label_column or [])
+            inputs = inputs or [
+                {"name": col, "value_type": str(dtype)}
+                for col, dtype in zip(training_set.columns,
+                                      training_set.dtypes)
+                if col not in label_columns]
+            outputs = outputs or [
+                {"name": col,
+                 "value_type": str(training_set[col].dtype)}
+                for col in label_columns if col in training_set.columns]
+        if feature_vector:
+            kwargs.setdefault("feature_vector", feature_vector)
+        if feature_weights:
+            kwargs.setdefault("feature_weights", feature_weights)
         self._init_db()
         producer = self._producer()
         model = self._artifacts_manager.log_model(

@@ -42,7 +42,8 @@ def _resolve_feature_set(ref) -> FeatureSet:
 
 def ingest(featureset: typing.Union[FeatureSet, str] = None, source=None,
            targets: list = None, namespace=None, return_df: bool = True,
-           infer_options=None, overwrite=None):
+           infer_options=None, overwrite=None, mlrun_context=None,
+           run_config=None, spark_context=None):
     """Batch-ingest a source (DataFrame / csv / parquet path) through
     the featureset's transform graph into its targets (parquet offline
     + the online window/KV table)."""
@@ -50,6 +51,16 @@ def ingest(featureset: typing.Union[FeatureSet, str] = None, source=None,
 
     from ..datastore.sources import BaseSource
 
+    if spark_context is not None:
+        raise MLRunInvalidArgumentError(
+            "spark ingestion is replaced by the node-local batch/GPU "
+            "engine (SURVEY §7); drop spark_context")
+    if run_config is not None:
+        # run-as-job (reference RunConfig): route through the job
+        # runtime wrapper
+        return run_ingestion_job(featureset,
+                                 source if isinstance(source, str)
+                                 else "")
     fset = _resolve_feature_set(featureset)
     if source is None:
         source = fset.spec.source
@@ -299,11 +310,40 @@ class OfflineVectorResponse:
 def get_offline_features(feature_vector, entity_rows=None,
                          entity_timestamp_column=None, target=None,
                          drop_columns=None, with_indexes=False,
-                         update_stats=False,
-                         join_graph=None) -> OfflineVectorResponse:
+                         update_stats=False, join_graph=None,
+                         start_time=None, end_time=None,
+                         timestamp_for_filtering: str = None,
+                         query: str = None, order_by=None,
+                         additional_filters: list = None,
+                         engine: str = None, engine_args: dict = None,
+                         run_config=None,
+                         spark_service=None) -> OfflineVectorResponse:
     """Join features from the parquet targets of the referenced sets
-    (pandas merger — reference retrieval/local_merger.py)."""
+    (pandas merger — reference retrieval/local_merger.py).
+
+    Reference query surface: start_time/end_time window on
+    ``timestamp_for_filtering`` (or each set's timestamp key), a
+    pandas ``query`` expression, ``order_by`` columns, and
+    ``additional_filters`` as (column, op, value) tuples with op in
+    =/==/!=/in/not in/>/>=/</<=.  dask/spark engines are replaced by
+    the node-local pandas merger: any other ``engine``
+    raises; ``run_config`` (run-as-job) is not needed node-locally."""
     import pandas as pd
+
+    if engine not in (None, "", "local", "pandas"):
+        raise MLRunInvalidArgumentError(
+            f"engine {engine!r} is not available in the node-local "
+            f"build — the pandas merger runs in-process (dask/spark "
+            f"engines are replaced; SURVEY §7 design)")
+    if spark_service:
+        raise MLRunInvalidArgumentError(
+            "spark_service is not supported (spark engines are "
+            "replaced by the node-local design)")
+    if run_config is not None:
+        raise MLRunInvalidArgumentError(
+            "run_config (run-as-job) is unnecessary node-locally — "
+            "call get_offline_features in-process, or wrap it in a "
+            "job runtime yourself")
 
     vector = FeatureVector.resolve(feature_vector)
     asof = entity_timestamp_column is not None and entity_rows is not None
@@ -370,6 +410,45 @@ def get_offline_features(feature_vector, entity_rows=None,
         merged = entity_rows.merge(merged, on=entity_cols, how="left")
     if vector.spec.label_feature:
         pass
+    # reference query surface: time window, row filters, query, order
+    ts_col = timestamp_for_filtering or entity_timestamp_column
+    if (start_time is not None or end_time is not None) and ts_col and \
+            ts_col in merged.columns:
+        ts_series = pd.to_datetime(merged[ts_col])
+        if start_time is not None:
+            merged = merged[ts_series >= pd.Timestamp(start_time)]
+            ts_series = ts_series[ts_series >= pd.Timestamp(start_time)]
+        if end_time is not None:
+            merged = merged[ts_series <= pd.Timestamp(end_time)]
+    for filt in additional_filters or []:
+        col, op, value = filt
+        if col not in merged.columns:
+            continue
+        if op in ("=", "=="):
+            merged = merged[merged[col] == value]
+        elif op == "!=":
+            merged = merged[merged[col] != value]
+        elif op == "in":
+            merged = merged[merged[col].isin(value)]
+        elif op == "not in":
+            merged = merged[~merged[col].isin(value)]
+        elif op == ">":
+            merged = merged[merged[col] > value]
+        elif op == ">=":
+            merged = merged[merged[col] >= value]
+        elif op == "<":
+            merged = merged[merged[col] < value]
+        elif op == "<=":
+            merged = merged[merged[col] <= value]
+        else:
+            raise MLRunInvalidArgumentError(
+                f"unsupported filter op {op!r}")
+    if query:
+        merged = merged.query(query)
+    if order_by:
+        cols = [order_by] if isinstance(order_by, str) else list(order_by)
+        merged = merged.sort_values([c for c in cols
+                                     if c in merged.columns])
     if drop_columns:
         merged = merged.drop(columns=[c for c in drop_columns
                                       if c in merged.columns])
@@ -382,10 +461,33 @@ def get_offline_features(feature_vector, entity_rows=None,
 
 def get_online_feature_service(feature_vector, impute_policy: dict = None,
                                fixed_window_type=None,
-                               entity_keys=None) -> OnlineVectorService:
+                               entity_keys=None, update_stats=False,
+                               run_config=None) -> OnlineVectorService:
     """Start an online lookup service over the vector's feature sets
-    (reference api.py:296)."""
+    (reference api.py:296).  update_stats refreshes the vector's
+    feature-stats from the referenced sets; run_config (deploy-as-
+    nuclio) is unnecessary node-locally — the service runs
+    in-process."""
+    if run_config is not None:
+        raise MLRunInvalidArgumentError(
+            "run_config is not supported: the online service is "
+            "in-process (node-local design); use fn.deploy() for an "
+            "HTTP-fronted service")
     vector = FeatureVector.resolve(feature_vector)
+    if update_stats:
+        stats = {}
+        for fs_name, _, _ in vector.grouped_features():
+            fset = _resolve_feature_set(
+                f"{vector.metadata.project or 'default'}/{fs_name}")
+            fset_stats = (fset.status.to_dict() or {}).get("stats") \
+                if hasattr(fset.status, "to_dict") else None
+            if fset_stats:
+                stats.update(fset_stats)
+        vector.status.stats = stats
+        try:
+            vector.save()
+        except Exception:
+            pass  # vector may be ad-hoc (not stored)
     tables = {}
     for fs_name, _, _ in vector.grouped_features():
         fset = _resolve_feature_set(

@@ -231,7 +231,7 @@ class BaseRuntime(ModelObj):
             scrape_metrics=None, local: bool = None, local_code_path=None,
             auto_build=None, param_file_secrets=None, notifications=None,
             returns=None, state_thresholds=None, selector: str = None,
-            **launcher_kwargs) -> RunObject:
+            reset_on_run: bool = None, **launcher_kwargs) -> RunObject:
         """Run this function (locally or submitted to the service)."""
         from ..launcher import LauncherFactory
 

@@ -691,3 +691,60 @@ class TestVectorServiceEdges:
         assert rows[1] == [-9.0, -9.0]    # imputed for unknown entity
         dicts = svc.get([{"customer": "ghost"}])
         assert dicts[0]["amount_sum_1h"] == -9.0
+
+
+class TestOfflineQuerySurface:
+    """Reference get_offline_features query params: time window,
+    filters, query, order_by, engine guards."""
+
+    def _setup(self):
+        df = pd.DataFrame({
+            "k": ["a", "b", "c", "d"],
+            "v": [1.0, 5.0, 9.0, 13.0],
+            "ts": pd.to_datetime(["2026-01-01", "2026-02-01",
+                                  "2026-03-01", "2026-04-01"]),
+        })
+        fset = fstore.FeatureSet("q", entities=["k"],
+                                 timestamp_key="ts")
+        fstore.ingest(fset, df, targets=["parquet"], overwrite=True)
+        vector = fstore.FeatureVector("vq", features=["q.v", "q.ts"])
+        vector.metadata.project = "default"
+        return vector
+
+    def test_filters_query_order(self, rundb):
+        vector = self._setup()
+        out = fstore.get_offline_features(
+            vector, additional_filters=[("v", ">", 2.0)],
+            query="v < 10", order_by="v").to_dataframe()
+        assert list(out["v"]) == [5.0, 9.0]
+
+    def test_in_and_not_in(self, rundb):
+        vector = self._setup()
+        out = fstore.get_offline_features(
+            vector,
+            additional_filters=[("k", "in", ["a", "d"])]).to_dataframe()
+        assert sorted(out["k"]) == ["a", "d"]
+        out = fstore.get_offline_features(
+            vector,
+            additional_filters=[("k", "not in",
+                                 ["a", "d"])]).to_dataframe()
+        assert sorted(out["k"]) == ["b", "c"]
+
+    def test_time_window(self, rundb):
+        vector = self._setup()
+        out = fstore.get_offline_features(
+            vector, start_time="2026-01-15", end_time="2026-03-15",
+            timestamp_for_filtering="ts").to_dataframe()
+        assert sorted(out["k"]) == ["b", "c"]
+
+    def test_replaced_engines_raise(self, rundb):
+        vector = self._setup()
+        with pytest.raises(MLRunInvalidArgumentError, match="engine"):
+            fstore.get_offline_features(vector, engine="spark")
+        with pytest.raises(MLRunInvalidArgumentError,
+                           match="spark_service"):
+            fstore.get_offline_features(vector, spark_service="x")
+        with pytest.raises(MLRunInvalidArgumentError,
+                           match="run_config"):
+            fstore.get_online_feature_service(vector,
+                                              run_config=object())
