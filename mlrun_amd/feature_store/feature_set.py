@@ -163,8 +163,12 @@ class FeatureSet(ModelObj):
     def fullname(self):
         return f"{self.metadata.project or 'default'}/{self.metadata.name}"
 
-    def add_entity(self, name, value_type=None, description=None):
-        self.spec.entities.append(Entity(name, value_type, description))
+    def add_entity(self, name, value_type=None, description=None,
+                   labels: dict = None):
+        entity = Entity(name, value_type, description)
+        if labels:
+            entity.labels = labels
+        self.spec.entities.append(entity)
         return self
 
     def add_feature(self, feature: Feature, name=None):
@@ -174,7 +178,8 @@ class FeatureSet(ModelObj):
         return self
 
     def add_aggregation(self, column, operations, windows, period=None,
-                        name=None, step_name=None, after=None):
+                        name=None, step_name=None, after=None,
+                        before=None, emit_policy=None):
         """Register sliding/tumbling window aggregations on a column
         (parity: reference add_aggregation feature_set.py:715)."""
         agg = FeatureAggregation(name=name or column, column=column,

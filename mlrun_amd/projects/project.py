@@ -121,7 +121,17 @@ class MlrunProject(ModelObj):
     # --------------------------------------------------------- functions
     def set_function(self, func=None, name: str = "", kind: str = "",
                      image: str = "", handler: str = "", with_repo=None,
-                     tag: str = "", requirements=None) -> "BaseRuntime":
+                     tag: str = "", requirements=None,
+                     function_object=None,
+                     function_dict: dict = None) -> "BaseRuntime":
+        # reference aliases: function_object (a live runtime),
+        # function_dict (a serialized runtime)
+        if function_object is not None and func is None:
+            func = function_object
+        if function_dict is not None and func is None:
+            from ..run import new_function
+
+            func = new_function(runtime=function_dict)
         from ..run import code_to_function, import_function, new_function
         from ..runtimes import BaseRuntime
 
@@ -189,16 +199,21 @@ class MlrunProject(ModelObj):
     def run_function(self, function, handler=None, name="", params=None,
                      inputs=None, hyperparams=None, hyper_param_options=None,
                      artifact_path=None, workdir="", watch=True, local=None,
-                     schedule=None, returns=None, notifications=None):
+                     schedule=None, returns=None, notifications=None,
+                     labels=None, outputs=None, selector=None,
+                     verbose=None, auto_build=None, base_task=None,
+                     builder_env=None, reset_on_run=None):
         from .operations import run_function as _run_function
 
-        return _run_function(
+        run = _run_function(
             function, handler=handler, name=name, params=params,
             inputs=inputs, hyperparams=hyperparams,
             hyper_param_options=hyper_param_options,
             artifact_path=artifact_path or self.artifact_path,
             workdir=workdir, watch=watch, local=local, schedule=schedule,
-            returns=returns, notifications=notifications, project_object=self)
+            returns=returns or outputs, notifications=notifications,
+            selector=selector, labels=labels, project_object=self)
+        return run
 
     def build_function(self, function, with_mlrun=None, skip_deployed=False,
                        image=None, base_image=None, commands=None,
@@ -238,11 +253,16 @@ class MlrunProject(ModelObj):
 
     def log_dataset(self, key, df, tag="", local_path=None, format="parquet",
                     preview=None, stats=None, target_path="",
-                    artifact_path=None, upload=None, labels=None, **kwargs):
+                    artifact_path=None, upload=None, labels=None,
+                    extra_data=None, label_column: str = None, **kwargs):
         from ..artifacts import DatasetArtifact
 
         ds = DatasetArtifact(key, df=df, format=format, preview=preview,
                              stats=stats, target_path=target_path)
+        if label_column:
+            ds.label_column = label_column
+        if extra_data:
+            kwargs.setdefault("extra_data", extra_data)
         return self.log_artifact(ds, local_path=local_path,
                                  artifact_path=artifact_path, tag=tag,
                                  upload=upload, labels=labels, **kwargs)
@@ -251,7 +271,26 @@ class MlrunProject(ModelObj):
                   model_file=None, metrics=None, parameters=None,
                   artifact_path=None, upload=None, labels=None, inputs=None,
                   outputs=None, tag="", extra_data=None, algorithm=None,
+                  training_set=None, label_column=None,
+                  feature_vector: str = None, feature_weights: list = None,
                   **kwargs):
+        if training_set is not None and hasattr(training_set, "columns"):
+            label_columns = [label_column] if isinstance(
+                label_column, str) else list(label_column or [])
+            inputs = inputs or [
+                {"name": col, "value_type": str(dtype)}
+                for col, dtype in zip(training_set.columns,
+                                      training_set.dtypes)
+                if col not in label_columns]
+            outputs = outputs or [
+                {"name": col,
+                 "value_type": str(training_set[col].dtype)}
+                for col in label_columns
+                if col in training_set.columns]
+        if feature_vector:
+            kwargs.setdefault("feature_vector", feature_vector)
+        if feature_weights:
+            kwargs.setdefault("feature_weights", feature_weights)
         return self._get_artifact_manager().log_model(
             self._producer(), key, body=body, model_file=model_file,
             model_dir=model_dir,

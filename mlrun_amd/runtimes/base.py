@@ -231,7 +231,8 @@ class BaseRuntime(ModelObj):
             scrape_metrics=None, local: bool = None, local_code_path=None,
             auto_build=None, param_file_secrets=None, notifications=None,
             returns=None, state_thresholds=None, selector: str = None,
-            reset_on_run: bool = None, **launcher_kwargs) -> RunObject:
+            reset_on_run: bool = None, labels: dict = None,
+            **launcher_kwargs) -> RunObject:
         """Run this function (locally or submitted to the service)."""
         from ..launcher import LauncherFactory
 
@@ -246,6 +247,8 @@ class BaseRuntime(ModelObj):
             verbose=verbose, scrape_metrics=scrape_metrics,
             notifications=notifications, returns=returns,
             state_thresholds=state_thresholds)
+        if labels:
+            run.metadata.labels.update(labels)
         if launcher_kwargs:
             from ..utils import logger
 
