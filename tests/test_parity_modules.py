@@ -456,3 +456,100 @@ class TestTorchCallbacks:
         history = self._train(ctx, [_AlwaysWorse(patience=2)],
                               epochs=50)
         assert len(history["loss"]) < 50  # stopped early
+
+
+class TestSignatureParity:
+    """Lock in drop-in signature parity: every parameter of the
+    reference's core entrypoints exists on ours (extra params are
+    allowed; the reference source is parsed textually)."""
+
+    CHECKS = [
+        ("run.py", "new_function", None),
+        ("run.py", "code_to_function", None),
+        ("run.py", "get_or_create_ctx", None),
+        ("run.py", "import_function", None),
+        ("projects/project.py", "new_project", None),
+        ("projects/project.py", "get_or_create_project", None),
+    ]
+    METHOD_CHECKS = [
+        ("runtimes/base.py", "run",
+         "mlrun_amd.runtimes.base.BaseRuntime"),
+        ("projects/project.py", "set_function",
+         "mlrun_amd.projects.project.MlrunProject"),
+        ("projects/project.py", "run_function",
+         "mlrun_amd.projects.project.MlrunProject"),
+        ("projects/project.py", "log_model",
+         "mlrun_amd.projects.project.MlrunProject"),
+        ("feature_store/feature_set.py", "add_aggregation",
+         "mlrun_amd.feature_store.feature_set.FeatureSet"),
+        ("feature_store/api.py", "get_offline_features", None),
+        ("feature_store/api.py", "get_online_feature_service", None),
+        ("feature_store/api.py", "ingest", None),
+        ("execution.py", "log_artifact",
+         "mlrun_amd.execution.MLClientCtx"),
+        ("execution.py", "log_model",
+         "mlrun_amd.execution.MLClientCtx"),
+    ]
+    REF = "/root/reference/mlrun/"
+
+    @staticmethod
+    def _ref_params(path, name, method):
+        import os
+        import re
+
+        ref_file = os.path.join("/root/reference/mlrun", path)
+        if not os.path.isfile(ref_file):
+            return None
+        src = open(ref_file).read()
+        pattern = (rf'\n    def {name}\(\s*self,?(.*?)\)( ->|:)'
+                   if method else rf'\ndef {name}\((.*?)\)( ->|:)')
+        match = re.search(pattern, src, re.S)
+        if not match:
+            return None
+        text = re.sub(r"\s+", " ", match.group(1))
+        out, depth, cur = set(), 0, ""
+        for ch in text + ",":
+            if ch in "([{":
+                depth += 1
+            if ch in ")]}":
+                depth -= 1
+            if ch == "," and depth == 0:
+                token = cur.split("=")[0].split(":")[0].strip().lstrip("*")
+                if token:
+                    out.add(token)
+                cur = ""
+            else:
+                cur += ch
+        return out - {"self", "", "kwargs", "class_args"}
+
+    def _ours(self, dotted, name):
+        import importlib
+        import inspect
+
+        if dotted is None:
+            import mlrun_amd
+
+            target = getattr(mlrun_amd, name, None)
+            if target is None:
+                module = importlib.import_module(
+                    "mlrun_amd.feature_store.api")
+                target = getattr(module, name)
+        else:
+            module_name, _, cls_name = dotted.rpartition(".")
+            cls = getattr(importlib.import_module(module_name), cls_name)
+            target = getattr(cls, name)
+        return set(inspect.signature(target).parameters)
+
+    def test_core_entrypoints(self):
+        import pytest as _pytest
+
+        failures = []
+        for path, name, method in self.CHECKS + self.METHOD_CHECKS:
+            ref = self._ref_params(path, name, method)
+            if ref is None:
+                continue  # reference moved; skip silently
+            ours = self._ours(method, name) | {"kwargs"}
+            missing = ref - ours
+            if missing:
+                failures.append((name, sorted(missing)))
+        assert not failures, failures
