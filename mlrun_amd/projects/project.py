@@ -427,7 +427,9 @@ class MlrunProject(ModelObj):
 def new_project(name, context="./", init_git=False, user_project=False,
                 remote=None, from_template=None, secrets=None,
                 description=None, subpath=None, save=True,
-                overwrite=False, parameters=None) -> MlrunProject:
+                overwrite=False, parameters=None,
+                default_function_node_selector: dict = None
+                ) -> MlrunProject:
     name = normalize_name(name)
     project = MlrunProject(context=context)
     project.metadata.name = name
@@ -490,7 +492,9 @@ def load_project(context="./", url=None, name=None, secrets=None,
 def get_or_create_project(name, context="./", url=None, secrets=None,
                           init_git=False, subpath=None, clone=False,
                           user_project=False, from_template=None, save=True,
-                          parameters=None) -> MlrunProject:
+                          parameters=None,
+                          allow_cross_project: bool = None
+                          ) -> MlrunProject:
     from ..db import get_run_db
 
     name = normalize_name(name)

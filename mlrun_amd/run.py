@@ -25,7 +25,8 @@ from .utils import normalize_name
 def new_function(name: str = "", project: str = "", tag: str = "",
                  kind: str = "", command: str = "", image: str = "",
                  args: list = None, mode=None, handler=None, source=None,
-                 requirements=None, kfp=None, runtime=None) -> "BaseRuntime":
+                 requirements=None, kfp=None, runtime=None,
+                 requirements_file: str = "") -> "BaseRuntime":
     """Create a function object of the given runtime kind."""
     if runtime:
         if hasattr(runtime, "to_dict"):
@@ -62,6 +63,11 @@ def new_function(name: str = "", project: str = "", tag: str = "",
             fn.spec.default_handler = handler
     if source:
         fn.spec.build["source"] = source
+    if requirements_file:
+        with open(requirements_file) as fp:
+            requirements = (requirements or []) + [
+                line.strip() for line in fp
+                if line.strip() and not line.startswith("#")]
     if requirements:
         fn.spec.build["requirements"] = requirements
     return fn
@@ -96,7 +102,8 @@ def code_to_function(name: str = "", project: str = "", tag: str = "",
                      description: str = "", requirements=None,
                      categories=None, labels=None,
                      with_doc=True, ignored_tags=None,
-                     code_: str = None) -> "BaseRuntime":
+                     code_: str = None,
+                     requirements_file: str = "") -> "BaseRuntime":
     """Package a python file (or the calling notebook's code) into a
     function object.  ``code_`` embeds source directly (reference
     code_to_function code_ param) instead of reading a file."""
@@ -133,6 +140,11 @@ def code_to_function(name: str = "", project: str = "", tag: str = "",
     elif embed_code:
         with open(filename) as fp:
             fn.spec.build["functionSourceCode"] = fp.read()
+    if requirements_file:
+        with open(requirements_file) as fp:
+            requirements = (requirements or []) + [
+                line.strip() for line in fp
+                if line.strip() and not line.startswith("#")]
     if requirements:
         fn.spec.build["requirements"] = requirements
     if categories:
