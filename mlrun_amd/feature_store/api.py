@@ -346,6 +346,8 @@ def get_offline_features(feature_vector, entity_rows=None,
             "job runtime yourself")
 
     vector = FeatureVector.resolve(feature_vector)
+    if join_graph is None:
+        join_graph = getattr(vector.spec, "join_graph", None)
     asof = entity_timestamp_column is not None and entity_rows is not None
     merged = None
     entity_cols: list = []

@@ -54,12 +54,26 @@ class FeatureVector(ModelObj):
     kind = "FeatureVector"
 
     def __init__(self, name=None, features=None, label_feature=None,
-                 description=None, with_indexes=None, project=None):
+                 description=None, with_indexes=None, project=None,
+                 join_graph=None, relations: dict = None,
+                 entity_fields=None, entity_source=None,
+                 timestamp_field: str = None, function=None,
+                 analysis=None, graph=None):
         self.metadata = FeatureVectorMetadata(name=name, project=project)
         self.spec = FeatureVectorSpec(features=features,
                                       description=description,
                                       label_feature=label_feature,
                                       with_indexes=with_indexes)
+        # reference spec extras: join_graph/relations drive the
+        # offline merger; the rest are enrichment metadata
+        self.spec.join_graph = join_graph
+        self.spec.relations = relations or {}
+        self.spec.entity_fields = entity_fields or []
+        self.spec.entity_source = entity_source
+        self.spec.timestamp_field = timestamp_field
+        self.spec.function = function
+        self.spec.analysis = analysis
+        self.spec.graph = graph
         self.status = ModelObj()
 
     @property

@@ -170,8 +170,12 @@ class ServingRuntime(BaseRuntime):
         return router.add_route(key, route=route)
 
     def set_tracking(self, stream_path=None, batch=None, sample=None,
-                     tracking_policy=None):
-        self.spec.track_models = True
+                     tracking_policy=None, enable_tracking: bool = True,
+                     stream_args: dict = None):
+        self.spec.track_models = bool(enable_tracking)
+        if stream_args:
+            self.spec.parameters = {**(self.spec.parameters or {}),
+                                    "tracking_stream_args": stream_args}
         return self
 
     def add_child_function(self, name, url=None, image=None, requirements=None,
