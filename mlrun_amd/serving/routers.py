@@ -35,6 +35,17 @@ class BaseModelRouter:
     def post_init(self, mode="sync"):
         pass
 
+    # user-overridable hooks (reference routers.py preprocess/
+    # postprocess/validate around _handle_event)
+    def preprocess(self, event):
+        return event
+
+    def postprocess(self, event):
+        return event
+
+    def validate(self, event):
+        return event
+
     def parse_event(self, event):
         if isinstance(event.body, (str, bytes)) and event.body:
             try:
@@ -70,7 +81,10 @@ class BaseModelRouter:
             # list models
             event.body = {"models": list(self.routes.keys())}
             return event
-        return self._handle_event(event)
+        event = self.preprocess(event)
+        event = self.validate(event)
+        event = self._handle_event(event)
+        return self.postprocess(event)
 
     def _handle_event(self, event):
         model = self._resolve_route(event)

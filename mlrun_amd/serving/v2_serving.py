@@ -78,6 +78,11 @@ class V2ModelServer:
                 raise
 
     # ------------------------------------------------------------- hooks
+    def set_metric(self, name: str, value):
+        """Set a real-time metric (rides the monitoring push —
+        reference v2_serving.py:162)."""
+        self.metrics[name] = value
+
     def get_param(self, key: str, default=None):
         if key in self._params:
             return self._params[key]
