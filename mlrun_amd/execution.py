@@ -349,6 +349,41 @@ class MLClientCtx:
     def set_hostname(self, host: str):
         self._host = host
 
+    def get_dataitem(self, url: str, secrets: dict = None):
+        """Resolve a data URL to a DataItem handle (reference
+        execution.py get_dataitem — the common handler entry for
+        reading inputs by URL)."""
+        from .datastore import store_manager
+
+        return store_manager.object(url, project=self._project)
+
+    def get_store_resource(self, url: str):
+        """Resolve a store:// URI to its artifact/feature object."""
+        from .datastore import store_manager
+
+        return store_manager.object(url, project=self._project)
+
+    def to_yaml(self) -> str:
+        import yaml
+
+        return yaml.safe_dump(self.to_dict(), default_flow_style=False)
+
+    @property
+    def artifacts(self) -> list:
+        """Artifacts logged by this run (reference ctx.artifacts)."""
+        return list(self._artifact_uris.values()) if isinstance(
+            getattr(self, "_artifact_uris", None), dict) else \
+            list(getattr(self, "_artifact_uris", []) or [])
+
+    def update_artifact(self, artifact):
+        """Re-log an updated artifact object (reference
+        update_artifact)."""
+        return self.log_artifact(artifact)
+
+    def store_run(self):
+        """Force-store the run document to the DB now."""
+        self.commit()
+
     def commit(self, message: str = "", completed=False):
         self._commit_text = message
         if completed:

@@ -433,6 +433,36 @@ class RunObject(RunTemplate):
             self.status = updated.get("status", {})
         return self
 
+    @property
+    def error(self) -> str:
+        """Error string of a failed run (reference RunObject.error)."""
+        if self.status and self.status.state in ("error", "failed"):
+            return (getattr(self.status, "error", "") or
+                    getattr(self.status, "status_text", "") or
+                    "unknown error")
+        return ""
+
+    @property
+    def ui_url(self) -> str:
+        """UI link (node-local: the run's REST resource URL)."""
+        from .config import config
+
+        base = str(config.ui_url or
+                   f"http://127.0.0.1:{config.httpdb.port}")
+        project = self.metadata.project or "default"
+        return f"{base}/api/v1/run/{project}/{self.metadata.uid}"
+
+    def is_failed(self) -> bool:
+        return self.status is not None and \
+            self.status.state in ("error", "failed", "aborted")
+
+    def show(self):
+        """Render the run as an HTML table in notebooks (reference
+        RunObject.show -> render)."""
+        from .render import run_to_html
+
+        return run_to_html(self.to_dict())
+
     def wait_for_completion(self, timeout: int = 600, sleep: float = 0.5,
                             raise_on_failure: bool = True) -> str:
         deadline = time.monotonic() + timeout
