@@ -370,10 +370,11 @@ class MLClientCtx:
 
     @property
     def artifacts(self) -> list:
-        """Artifacts logged by this run (reference ctx.artifacts)."""
-        return list(self._artifact_uris.values()) if isinstance(
-            getattr(self, "_artifact_uris", None), dict) else \
-            list(getattr(self, "_artifact_uris", []) or [])
+        """Artifact documents logged by this run (reference
+        ctx.artifacts — run status artifact list)."""
+        if self._artifacts_manager is None:
+            return []
+        return self._artifacts_manager.artifact_list()
 
     def update_artifact(self, artifact):
         """Re-log an updated artifact object (reference
