@@ -358,6 +358,21 @@ class DataItem:
     def listdir(self):
         return self._store.listdir(self._path)
 
+    def ls(self):
+        """Alias of listdir (reference DataItem.ls)."""
+        return self.listdir()
+
+    def open(self, mode: str = "rb"):
+        """Open the (localized) item as a file object (reference
+        DataItem.open)."""
+        return open(self.local(), mode)
+
+    def get_artifact_type(self):
+        """Kind of the backing artifact, if this item resolves one."""
+        if self._meta and isinstance(self._meta, dict):
+            return self._meta.get("kind")
+        return None
+
     def local(self) -> str:
         """Download to a local temp file (or return the path if local)."""
         if self.kind == "file":
