@@ -28,6 +28,7 @@ from ..errors import (MLRunConflictError, MLRunInvalidArgumentError,
 from ..model import RunStates
 from ..utils import now_iso
 from .base import RunDBInterface
+from .extras import RunDBExtras
 
 _SCHEMA = """
 CREATE TABLE IF NOT EXISTS runs (
@@ -199,7 +200,7 @@ def _match_labels(body: dict, labels) -> bool:
     return True
 
 
-class SQLRunDB(RunDBInterface):
+class SQLRunDB(RunDBExtras, RunDBInterface):
     """Node-local SQLite run DB (kind="local")."""
 
     kind = "local"

@@ -261,3 +261,38 @@ class TestApiCallRetry:
             assert calls["n"] == 1  # non-idempotent: no retry
         finally:
             httpdb.session.request = orig
+
+
+class TestLocalDbSurfaceParity:
+    """The LOCAL SQLRunDB exposes the same client surface as the HTTP
+    client (drop-in local mode — RunDBExtras mixin)."""
+
+    def test_no_method_gap(self):
+        from mlrun_amd.db.httpdb import HTTPRunDB
+        from mlrun_amd.db.sqldb import SQLRunDB
+
+        http_methods = {m for m in dir(HTTPRunDB)
+                        if not m.startswith("_")}
+        local_methods = {m for m in dir(SQLRunDB)
+                         if not m.startswith("_")}
+        assert not (http_methods - local_methods)
+
+    def test_local_extras_behave(self, tmp_path):
+        from mlrun_amd.db.sqldb import SQLRunDB
+
+        db = SQLRunDB(str(tmp_path / "x.db"))
+        db.store_feature_set({"metadata": {"name": "fs"},
+                              "spec": {"features": []}},
+                             name="fs", project="p")
+        db.patch_feature_set("fs", {"spec": {"description": "d"}},
+                             project="p")
+        assert db.get_feature_set("fs", "p")["spec"][
+            "description"] == "d"
+        db.store_run({"metadata": {"name": "r", "uid": "u"},
+                      "status": {"state": "completed"}}, "u", "p")
+        db.set_run_notifications("p", "u", [{"kind": "console"}])
+        assert db.read_run("u", "p")["spec"]["notifications"]
+        db.store_log("u", "p", b"12345")
+        assert db.get_log_size("u", "p") == 5
+        summary = db.get_project_summary("p")
+        assert summary["runs_completed_recent_count"] == 1
