@@ -184,6 +184,44 @@ class ServingRuntime(BaseRuntime):
                                          "kind": kind or "serving"}
         return self
 
+    def list_child_functions(self) -> list:
+        """Distinct child-function names referenced by graph steps
+        (reference states.py list_child_functions)."""
+        graph = self.spec.graph
+        names = []
+        if graph is None or not hasattr(graph, "steps"):
+            return names
+        for step in graph.steps.values():
+            function = getattr(step, "function", None)
+            if function and function != "*" and function not in names:
+                names.append(function)
+        return names
+
+    def _deploy_child_functions(self, parent_server, namespace=None):
+        """Start one host per child function named by graph steps and
+        register the addresses (reference _deploy_function_refs + the
+        queue/stream links, serving.py:512): forwarded events enter
+        the child's graph AT the annotated step (x-mlrun-step)."""
+        names = self.list_child_functions()
+        if not names:
+            return
+        self._child_hosts = getattr(self, "_child_hosts", [])
+        endpoints = {}
+        for name in names:
+            child = self._build_server(namespace, current_function=name,
+                                       fresh_graph=True)
+            host = GraphServerHost(child)
+            host.start()
+            self._child_hosts.append(host)
+            endpoints[name] = host.address
+            logger.info("child function deployed", function=name,
+                        address=host.address)
+        # every server (parent + children) can reach every child;
+        # get_remote_endpoint refuses self-forwarding by name
+        parent_server.child_endpoints.update(endpoints)
+        for host in self._child_hosts:
+            host.server.child_endpoints.update(endpoints)
+
     def remove_states(self, keys: list):
         graph = self.spec.graph
         if isinstance(graph, RouterStep):
@@ -193,17 +231,25 @@ class ServingRuntime(BaseRuntime):
                 graph.steps.pop(key, None)
 
     # ---------------------------------------------------------- serving
-    def _build_server(self, namespace=None) -> GraphServer:
+    def _build_server(self, namespace=None, current_function=None,
+                      fresh_graph=False) -> GraphServer:
+        graph = self.spec.graph
+        if fresh_graph and graph is not None:
+            # child hosts need their OWN step instances/controller —
+            # rebuild from the serialized spec (class objects stored
+            # as dotted paths survive; add_model guarantees that)
+            graph = step_from_dict(graph.to_dict())
         server = create_graph_server(
             parameters=self.spec.parameters,
             load_mode=self.spec.load_mode,
-            graph=self.spec.graph,
+            graph=graph,
             verbose=bool(self.spec.verbose),
             graph_initializer=self.spec.graph_initializer,
             error_stream=self.spec.error_stream,
             track_models=self.spec.track_models,
             function_uri=f"{self.metadata.project or 'default'}/"
                          f"{self.metadata.name}",
+            current_function=current_function,
         )
         server.init_states(namespace=namespace or _caller_namespace())
         return server
@@ -243,6 +289,7 @@ class ServingRuntime(BaseRuntime):
         self._server = server
         self._host = GraphServerHost(server)
         self._host.start()
+        self._deploy_child_functions(server, namespace)
         self.status.state = "ready"
         self.status.address = self._host.address
         self.status.external_invocation_urls = [self._host.address]
@@ -302,6 +349,9 @@ class ServingRuntime(BaseRuntime):
         if getattr(self, "_worker_pool", None) is not None:
             self._worker_pool.stop()
             self._worker_pool = None
+        for host in getattr(self, "_child_hosts", []):
+            host.stop()
+        self._child_hosts = []
         if self._host is not None:
             self._host.stop()
             self._host = None

@@ -138,20 +138,25 @@ class AsyncFlowController:
         self._executors = {}
 
     # ---------------------------------------------------------- submit
-    def emit(self, event, timeout: float = None):
+    def emit(self, event, timeout: float = None, start_step: str = None):
         """Submit one event; block until its future resolves (sync
         callers — GraphServer.run)."""
-        return self.emit_nowait(event).result(timeout=timeout)
+        return self.emit_nowait(event,
+                                start_step=start_step).result(
+            timeout=timeout)
 
-    def emit_nowait(self, event) -> concurrent.futures.Future:
+    def emit_nowait(self, event,
+                    start_step: str = None) -> concurrent.futures.Future:
         """Submit one event; return its result future (the awaitable
-        per-event result contract)."""
+        per-event result contract).  start_step: enter mid-graph
+        (forwarded child-function events)."""
         self.start()
         future: concurrent.futures.Future = concurrent.futures.Future()
         envelope = _Envelope(future)
 
         async def _inject():
-            starts = self.flow._start_steps
+            starts = ([self.flow.steps[start_step]] if start_step
+                      else self.flow._start_steps)
             if not starts:
                 envelope.finish_branch(event)
                 return
@@ -304,7 +309,8 @@ class AsyncFlowController:
                 body, (dict, list)) else body
             resp = requests.post(url, data=data, timeout=60,
                                  headers={"content-type":
-                                          "application/json"})
+                                          "application/json",
+                                          "x-mlrun-step": step.name})
             resp.raise_for_status()
             try:
                 return resp.json()

@@ -94,7 +94,13 @@ class GraphContext:
         return store_manager.object(uri, project=self.project)
 
     def get_remote_endpoint(self, name, external=False):
-        return ""
+        """Address of a deployed child-function host ("" = run
+        locally).  A function never forwards to itself."""
+        if self.server is None:
+            return ""
+        if name == (self.server._current_function or ""):
+            return ""
+        return (self.server.child_endpoints or {}).get(name, "")
 
     def push_error(self, event, message, source=None, **kwargs):
         """Log AND publish failed events to the configured error
@@ -149,6 +155,9 @@ class GraphServer(ModelObj):
         self.http_trigger = True
         self._namespace = None
         self._current_function = None
+        # child-function name -> deployed host address (multi-function
+        # graphs; reference _deploy_function_refs + queue links)
+        self.child_endpoints: dict = {}
 
     @property
     def graph(self) -> typing.Union[RootFlowStep, RouterStep]:
@@ -199,9 +208,12 @@ class GraphServer(ModelObj):
     def init_object(self, namespace):
         self._graph.init_object(self.context, namespace, self.load_mode)
 
-    def run(self, event: Event, context=None, get_body=False):
+    def run(self, event: Event, context=None, get_body=False,
+            start_step: str = None):
         """Feed one event through the graph (HOT PATH — parity:
-        reference server.py:252)."""
+        reference server.py:252).  start_step: enter the flow at a
+        named step (child-function hosts receiving forwarded
+        events)."""
         server_context = self.context
         body = event.body
         if isinstance(body, (str, bytes)) and body and \
@@ -211,7 +223,10 @@ class GraphServer(ModelObj):
             except (ValueError, TypeError):
                 pass
         try:
-            response = self._graph.run(event)
+            if start_step and hasattr(self._graph, "run_from_step"):
+                response = self._graph.run_from_step(start_step, event)
+            else:
+                response = self._graph.run(event)
         except Exception as exc:
             if server_context and server_context.verbose:
                 logger.error("graph run failed",
@@ -391,8 +406,10 @@ class GraphServerHost:
                           content_type=request.headers.get("content-type"))
             import anyio
 
+            start_step = request.headers.get("x-mlrun-step")
             response = await anyio.to_thread.run_sync(
-                lambda: graph_server.run(event, get_body=False))
+                lambda: graph_server.run(event, get_body=False,
+                                         start_step=start_step))
             status = str(response.status_code) \
                 if isinstance(response, _ErrorResponse) \
                 else "200"

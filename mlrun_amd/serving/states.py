@@ -574,6 +574,19 @@ class FlowStep(BaseStep):
             event = result if result is not None else event
         return event
 
+    def run_from_step(self, step_name: str, event):
+        """Run starting at a NAMED step (child-function hosts entering
+        mid-graph — reference queue/stream links deliver into the
+        consumer step)."""
+        if step_name not in self.steps:
+            raise GraphError(f"unknown start step {step_name}")
+        if not self._start_steps:
+            self._build_links()
+        if self._controller is not None:
+            return self._controller.emit(event, timeout=600,
+                                         start_step=step_name)
+        return self._run_from(self.steps[step_name], event)
+
     async def run_async(self, event):
         """Awaitable entry for async hosts (FastAPI): wraps the
         per-event future."""

@@ -431,3 +431,47 @@ class TestAsyncFlowProperties:
             server.graph.shutdown()
 
         check()
+
+
+class MarkFn:
+    """Records which function host executed it."""
+
+    def __init__(self, context=None, name=None, **kw):
+        self.name = name
+        self.context = context
+
+    def do(self, event):
+        current = ""
+        if self.context and self.context.server:
+            current = self.context.server._current_function or "parent"
+        event.body = dict(event.body)
+        event.body.setdefault("chain", []).append(
+            f"{self.name}@{current}")
+        return event
+
+
+class TestChildFunctionDeploy:
+    def test_deploy_starts_child_hosts_and_forwards(self):
+        """fn.deploy() auto-deploys one host per child function named
+        by graph steps (reference _deploy_function_refs): the queue
+        hand-off crosses hosts over HTTP and enters the child AT the
+        annotated step (x-mlrun-step)."""
+        import requests
+
+        fn = mlrun_amd.new_function("t-multi", kind="serving")
+        graph = fn.set_topology("flow", engine="async")
+        queue = graph.to("MarkFn", name="pre").to("$queue", "q1")
+        queue.to(name="enrich", class_name="MarkFn",
+                 function="enricher").respond()
+        addr = fn.deploy(namespace={"MarkFn": MarkFn})
+        try:
+            assert fn.list_child_functions() == ["enricher"]
+            assert len(fn._child_hosts) == 1
+            resp = requests.post(addr + "/score", json={"x": 1},
+                                 timeout=60)
+            assert resp.status_code == 200
+            assert resp.json()["chain"] == ["pre@parent",
+                                            "enrich@enricher"]
+        finally:
+            fn.stop()
+        assert fn._child_hosts == []  # stop() tears children down
