@@ -141,6 +141,16 @@ class ServingRuntime(BaseRuntime):
                     "no router step found in flow topology")
         else:
             raise MLRunInvalidArgumentError("graph has no router")
+        if model_url:
+            # remote model endpoint route (reference
+            # new_remote_endpoint): proxy infer calls over HTTP
+            from ..serving.remote import RemoteStep
+
+            route = RemoteStep(url=model_url, name=key, **class_args)
+            from ..serving.states import TaskStep as _TaskStep
+
+            step = _TaskStep(route, name=key)
+            return router.add_route(key, route=step)
         if class_name is None and handler is None:
             raise MLRunInvalidArgumentError(
                 "class_name (a V2ModelServer subclass) or handler is "

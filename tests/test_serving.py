@@ -758,3 +758,44 @@ class TestAsyncLoadMode:
         resp = server.test("/v2/models/m/infer", body={"inputs": [2]})
         assert resp["outputs"] == [4]
         assert step._object.ready is True   # loaded on demand
+
+
+class TestRemoteModelUrl:
+    def test_add_model_with_model_url_proxies(self):
+        """add_model(model_url=...) routes infers to a remote model
+        endpoint (reference new_remote_endpoint route)."""
+        import json
+        import threading
+
+        from http.server import BaseHTTPRequestHandler, HTTPServer
+
+        import mlrun_amd
+
+        class Handler(BaseHTTPRequestHandler):
+            def do_POST(self):
+                n = int(self.headers.get("content-length", 0))
+                body = json.loads(self.rfile.read(n))
+                out = json.dumps(
+                    {"outputs": [sum(body["inputs"])]}).encode()
+                self.send_response(200)
+                self.send_header("content-type", "application/json")
+                self.end_headers()
+                self.wfile.write(out)
+
+            def log_message(self, *a):
+                pass
+
+        httpd = HTTPServer(("127.0.0.1", 0), Handler)
+        port = httpd.server_address[1]
+        threading.Thread(target=httpd.serve_forever,
+                         daemon=True).start()
+        try:
+            fn = _serving_fn()
+            fn.add_model("ext",
+                         model_url=f"http://127.0.0.1:{port}/infer")
+            server = fn.to_mock_server()
+            resp = server.test("/v2/models/ext/infer",
+                               body={"inputs": [1, 2, 3]})
+            assert resp["outputs"] == [6]
+        finally:
+            httpd.shutdown()
