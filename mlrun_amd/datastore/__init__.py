@@ -480,7 +480,7 @@ def parse_store_uri(url: str, default_project: str = ""):
     iteration = 0
     if "@" in key:
         key, tree = key.rsplit("@", 1)
-    elif ":" in key:
+    if ":" in key:
         key, tag = key.rsplit(":", 1)
     if "#" in key:
         key, it = key.rsplit("#", 1)

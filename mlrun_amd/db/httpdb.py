@@ -111,7 +111,10 @@ class HTTPRunDB(RunDBInterface):
         if labels:
             params["label"] = labels if isinstance(labels, list) else [
                 f"{k}={v}" for k, v in labels.items()]
-        return self.api_call("GET", "runs", params=params).get("runs", [])
+        from ..lists import RunList
+
+        return RunList(self.api_call("GET", "runs",
+                                     params=params).get("runs", []))
 
     def del_run(self, uid, project="", iter=0):
         self.api_call("DELETE", f"run/{project or 'default'}/{uid}",
@@ -160,8 +163,12 @@ class HTTPRunDB(RunDBInterface):
         if labels:
             params["label"] = labels if isinstance(labels, list) else [
                 f"{k}={v}" for k, v in labels.items()]
-        return self.api_call("GET", "artifacts", params=params).get(
-            "artifacts", [])
+        from ..lists import ArtifactList
+
+        result = ArtifactList(self.api_call(
+            "GET", "artifacts", params=params).get("artifacts", []))
+        result.tag = tag
+        return result
 
     def del_artifact(self, key, tag="", project="", uid=None, tree=None):
         self.api_call("DELETE", f"artifact/{project or 'default'}/{key}",

@@ -392,7 +392,9 @@ class SQLRunDB(RunDBExtras, RunDBInterface):
                 bucket.append(body)
                 capped.append(body)
             out = capped
-        return out
+        from ..lists import RunList
+
+        return RunList(out)
 
     def del_run(self, uid, project="", iter=0):
         project = project or "default"
@@ -631,7 +633,11 @@ class SQLRunDB(RunDBExtras, RunDBInterface):
             body = json.loads(row["body"])
             if _match_labels(body, labels):
                 out.append(body)
-        return out
+        from ..lists import ArtifactList
+
+        result = ArtifactList(out)
+        result.tag = tag
+        return result
 
     def del_artifact(self, key, tag="", project="", uid=None, tree=None):
         project = project or "default"

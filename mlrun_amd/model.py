@@ -514,19 +514,21 @@ def _get_db():
 
 
 def get_artifact_target(item: dict, project: str = None) -> str:
-    """Build a store:// uri for an artifact dict (or return its target path)."""
+    """Build a store:// uri for an artifact dict (or return its target
+    path).  Reference helpers.py:109: the store uri is used only when
+    the artifact was registered in the DB (``spec.db_key`` set);
+    otherwise the raw target path is the address."""
     metadata = item.get("metadata", item)
     spec = item.get("spec", item)
     kind = item.get("kind", "artifact")
-    key = metadata.get("key")
+    db_key = spec.get("db_key")
     project = metadata.get("project") or project or "default"
     tree = metadata.get("tree")
     tag = metadata.get("tag")
-    if kind in ["dataset", "model", "artifact"] and key:
-        uri = f"store://artifacts/{project}/{key}"
-        if tag:
-            uri += f":{tag}"
-        elif tree:
+    if kind in ["dataset", "model", "artifact"] and db_key:
+        uri = f"store://artifacts/{project}/{db_key}"
+        uri += f":{tag}" if tag else ":latest"
+        if tree:
             uri += f"@{tree}"
         return uri
     return spec.get("target_path", "")

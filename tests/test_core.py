@@ -58,7 +58,7 @@ class TestModelObjects:
         run.status.artifacts = [
             {"kind": "model", "metadata": {"key": "m", "project": "p",
                                            "tree": "abc"},
-             "spec": {"target_path": "/tmp/m"}}]
+             "spec": {"target_path": "/tmp/m", "db_key": "m"}}]
         assert run.output("acc") == 0.9
         assert run.output("m").startswith("store://")
         outputs = run.outputs
