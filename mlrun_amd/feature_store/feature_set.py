@@ -44,13 +44,35 @@ class Entity(ModelObj):
 
 
 class Feature(ModelObj):
+    _dict_fields = ["name", "value_type", "description", "aggregate",
+                    "labels", "validator"]
+
     def __init__(self, value_type=None, description=None, name=None,
-                 aggregate=None, labels=None):
+                 aggregate=None, labels=None, validator=None):
         self.name = name
         self.value_type = value_type or "float"
         self.description = description
         self.aggregate = aggregate
         self.labels = labels or {}
+        self._validator = None
+        if validator is not None:
+            self.validator = validator
+
+    @property
+    def validator(self):
+        return self._validator
+
+    @validator.setter
+    def validator(self, validator):
+        """Accepts a Validator instance or its dict form (reference
+        features.py:125)."""
+        if isinstance(validator, dict):
+            from ..features import validator_from_dict
+
+            validator = validator_from_dict(validator)
+        if validator is not None:
+            validator.set_feature(self)
+        self._validator = validator
 
 
 class FeatureAggregation(ModelObj):

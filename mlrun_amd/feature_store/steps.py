@@ -30,7 +30,10 @@ class MLRunStep:
 
 
 class FeaturesetValidator(MLRunStep):
-    """Validate values against feature value_type; drop/flag bad rows."""
+    """Validate values against per-feature validators
+    (``Feature.validator`` — MinMax/MinMaxLen/Regex, reference
+    steps.py FeaturesetValidator) and feature value_type; violations
+    are logged (severity-tagged) and optionally dropped."""
 
     def __init__(self, context=None, name=None, featureset=None,
                  columns=None, drop_invalid=True, **kwargs):
@@ -38,20 +41,47 @@ class FeaturesetValidator(MLRunStep):
         self.featureset = featureset
         self.columns = columns
         self.drop_invalid = drop_invalid
+        self.violations: list = []
+
+    def _validators(self, df):
+        out = {}
+        if self.featureset is None:
+            return out
+        for feature in self.featureset.spec.features:
+            validator = getattr(feature, "validator", None)
+            if validator is not None and feature.name in df.columns and \
+                    (not self.columns or feature.name in self.columns):
+                validator.set_feature(feature)
+                out[feature.name] = validator
+        return out
 
     def do(self, df):
         import pandas as pd
 
+        validators = self._validators(df)
         columns = self.columns or []
         if self.featureset is not None and not columns:
             columns = [f.name for f in self.featureset.spec.features
                        if not f.aggregate and f.name in df.columns]
         bad_mask = None
+        for col, validator in validators.items():
+            checked = df[col].map(lambda v: validator.check(v)[0])
+            col_bad = ~checked
+            for idx in df.index[col_bad]:
+                ok, info = validator.check(df.at[idx, col])
+                violation = {"feature": col,
+                             "severity": validator.severity, **info}
+                self.violations.append(violation)
+                from ..utils import logger
+
+                logger.warning("feature validation failed", **{
+                    ("detail" if k == "message" else k): str(v)
+                    for k, v in violation.items()})
+            bad_mask = col_bad if bad_mask is None else (bad_mask | col_bad)
         for col in columns:
-            if col not in df.columns:
+            if col in validators or col not in df.columns:
                 continue
             numeric = pd.to_numeric(df[col], errors="coerce")
-            col_bad = numeric.isna() & df[col].notna() == False  # noqa
             col_bad = numeric.isna()
             bad_mask = col_bad if bad_mask is None else (bad_mask | col_bad)
         if bad_mask is not None and self.drop_invalid:

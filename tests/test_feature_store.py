@@ -748,3 +748,73 @@ class TestOfflineQuerySurface:
                            match="run_config"):
             fstore.get_online_feature_service(vector,
                                               run_config=object())
+
+
+class TestFeatureValidators:
+    """Per-value validators (reference mlrun/features.py)."""
+
+    def test_minmax(self):
+        from mlrun_amd.features import MinMaxValidator
+
+        v = MinMaxValidator(min=0, max=10, severity="info")
+        assert v.check(5)[0]
+        ok, info = v.check(-1)
+        assert not ok and "min" in info
+        ok, info = v.check(11)
+        assert not ok and "max" in info
+
+    def test_minmaxlen_and_regex(self):
+        from mlrun_amd.features import (MinMaxLenValidator,
+                                        RegexValidator)
+
+        lv = MinMaxLenValidator(min=2, max=4)
+        assert lv.check("abc")[0]
+        assert not lv.check("a")[0]
+        assert not lv.check("abcde")[0]
+        rv = RegexValidator(regex=r"[A-Z]\d{3}")
+        assert rv.check("A123")[0]
+        assert not rv.check("abc")[0]
+
+    def test_type_check(self):
+        from mlrun_amd.feature_store.feature_set import Feature
+        from mlrun_amd.features import Validator
+
+        v = Validator(check_type=True)
+        v.set_feature(Feature(name="x", value_type="int8"))
+        assert v.check(5)[0]
+        assert not v.check(300)[0]  # out of int8 range
+
+    def test_feature_validator_dict_roundtrip(self):
+        from mlrun_amd.feature_store.feature_set import Feature
+        from mlrun_amd.features import MinMaxValidator
+
+        f = Feature(name="bid", value_type="float")
+        f.validator = {"kind": "minmax", "min": 52, "severity": "info"}
+        assert isinstance(f.validator, MinMaxValidator)
+        d = f.to_dict()
+        assert d["validator"]["kind"] == "minmax"
+        f2 = Feature.from_dict(d)
+        assert isinstance(f2.validator, MinMaxValidator)
+        assert f2.validator.min == 52
+
+    def test_validator_step_drops_and_logs(self):
+        import pandas as pd
+
+        import mlrun_amd.feature_store as fstore
+        from mlrun_amd.feature_store.feature_set import Feature
+        from mlrun_amd.feature_store.steps import FeaturesetValidator
+        from mlrun_amd.features import MinMaxValidator
+
+        fset = fstore.FeatureSet("quotes", entities=["t"])
+        fset.spec.features = [
+            Feature(name="bid", value_type="float",
+                    validator=MinMaxValidator(min=0, severity="info"))]
+        step = FeaturesetValidator(featureset=fset)
+        df = pd.DataFrame({"t": ["a", "b"], "bid": [5.0, -3.0]})
+        out = step.do(df)
+        assert len(out) == 1
+        assert step.violations[0]["feature"] == "bid"
+        assert step.violations[0]["severity"] == "info"
+
+    def test_mlrun_features_alias(self):
+        from mlrun.features import MinMaxValidator  # noqa: F401
