@@ -372,8 +372,11 @@ def validate_alert_config(struct: dict) -> dict:
     validated = AlertConfigSchema(**{k: v for k, v in struct.items()
                                      if k != "entities"})
     out = validated.model_dump(exclude_none=True)
-    if "entities" in struct:
-        out["entities"] = struct["entities"]
+    # client AlertConfig carries extra fields (entities, description,
+    # state, id, count, created) — preserve them verbatim
+    for key, value in struct.items():
+        if key not in out and value is not None:
+            out[key] = value
     for spec in out.get("notifications", []):
         inner = spec.get("notification", spec) if isinstance(
             spec, dict) else spec

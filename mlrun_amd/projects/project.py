@@ -328,6 +328,67 @@ class MlrunProject(ModelObj):
     def list_model_monitoring_functions(self):
         return []
 
+    # ----------------------------------------------------------- alerts
+    def store_alert_config(self, alert_data, alert_name: str = None):
+        """Create/modify an alert (reference project.py:4205); accepts
+        an ``mlrun.alerts.AlertConfig`` or a dict."""
+        from ..errors import MLRunInvalidArgumentError
+        from ..utils import logger
+
+        if not alert_data:
+            raise MLRunInvalidArgumentError("Alert data must be provided")
+        name = alert_name or (alert_data.get("name") if isinstance(
+            alert_data, dict) else alert_data.name)
+        alert_project = (alert_data.get("project") if isinstance(
+            alert_data, dict) else alert_data.project)
+        if alert_project is not None and alert_project != self.name:
+            logger.warning("Project in alert does not match project in "
+                           "operation", project=alert_project)
+        if isinstance(alert_data, dict):
+            alert_data["project"] = self.name
+        else:
+            alert_data.project = self.name
+        return self._get_db().store_alert_config(self.name, name,
+                                                 alert_data)
+
+    def get_alert_config(self, alert_name: str):
+        from ..alerts import AlertConfig
+
+        struct = self._get_db().get_alert_config(self.name, alert_name)
+        return AlertConfig.from_dict(struct) if isinstance(
+            struct, dict) else struct
+
+    def list_alerts_configs(self) -> list:
+        return self._get_db().list_alert_configs(self.name)
+
+    def delete_alert_config(self, alert_data=None, alert_name: str = None):
+        alert_name = self._resolve_alert_name(alert_data, alert_name)
+        self._get_db().delete_alert_config(self.name, alert_name)
+
+    def reset_alert_config(self, alert_data=None, alert_name: str = None):
+        alert_name = self._resolve_alert_name(alert_data, alert_name)
+        self._get_db().reset_alert_config(self.name, alert_name)
+
+    @staticmethod
+    def _resolve_alert_name(alert_data, alert_name):
+        if alert_data is None and alert_name is None:
+            raise ValueError(
+                "At least one of alert_data or alert_name must be "
+                "provided")
+        data_name = None if alert_data is None else (
+            alert_data.get("name") if isinstance(alert_data, dict)
+            else alert_data.name)
+        if data_name and alert_name and data_name != alert_name:
+            raise ValueError(
+                "Alert_data name does not match the provided alert_name")
+        return alert_name or data_name
+
+    def get_alert_template(self, template_name: str):
+        return self._get_db().get_alert_template(template_name)
+
+    def list_alert_templates(self) -> list:
+        return self._get_db().list_alert_templates()
+
     # --------------------------------------------------------- workflows
     def set_workflow(self, name, workflow_path: str, embed=False,
                      engine=None, args_schema=None, handler=None, **args):
