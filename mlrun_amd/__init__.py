@@ -74,7 +74,12 @@ from .projects import (  # noqa: F401,E402
     new_project,
     run_function,
 )
-from .package import handler, ArtifactType  # noqa: F401,E402
+from .package import (  # noqa: F401,E402
+    ArtifactType,
+    DefaultPackager,
+    Packager,
+    handler,
+)
 from .secrets import get_secret_or_env  # noqa: F401,E402
 from .platforms import (  # noqa: F401,E402
     VolumeMount,
@@ -112,17 +117,48 @@ def set_environment(api_path: str = None, artifact_path: str = "",
 
 
 def _load_env_file(path: str):
+    set_env_from_file(path)
+
+
+def set_env_from_file(env_file: str, return_dict: bool = False):
+    """Read KEY=VALUE lines from a .env file into the process env and
+    reload mlrun config (reference mlrun/__init__.py:187)."""
     import os
 
-    from .utils import list_to_dict
+    from .errors import MLRunInvalidArgumentError, MLRunNotFoundError
 
-    with open(os.path.expanduser(path)) as fp:
-        for key, value in list_to_dict(fp.readlines()).items():
-            if key and not key.startswith("#"):
-                os.environ[key] = value
+    env_file = os.path.expanduser(env_file)
+    if not os.path.isfile(env_file):
+        raise MLRunNotFoundError(f"env file {env_file} does not exist")
+    env_vars = {}
+    with open(env_file) as fp:
+        for line in fp:
+            line = line.strip()
+            if not line or line.startswith("#"):
+                continue
+            if "=" not in line:
+                raise MLRunInvalidArgumentError(
+                    "env file lines must be in the form key=value")
+            key, _, value = line.partition("=")
+            env_vars[key.strip()] = value.strip()
+    for key, value in env_vars.items():
+        os.environ[key] = value
     from .config import _populate
 
     _populate()
+    return env_vars if return_dict else None
+
+
+def get_sample_path(subpath: str = "") -> str:
+    """Url of a sample dataset or model (reference
+    mlrun/__init__.py:175)."""
+    import os
+
+    samples_path = os.environ.get("SAMPLE_DATA_SOURCE_URL_PREFIX",
+                                  mlconf.default_samples_path)
+    if subpath:
+        samples_path = os.path.join(samples_path, subpath.lstrip("/"))
+    return samples_path
 
 
 def get_current_project(silent: bool = False):

@@ -29,6 +29,8 @@ default_config = {
     "dbpath": "",
     "base_dir": os.path.expanduser("~/.mlrun_amd"),
     "default_project": "default",
+    # url prefix for get_sample_path (reference config.py:181)
+    "default_samples_path": "https://s3.wasabisys.com/iguazio/",
     "artifact_path": "",  # default artifact path template
     "log_level": "INFO",
     "log_format": "human",  # human | json

@@ -553,3 +553,46 @@ class TestSignatureParity:
             if missing:
                 failures.append((name, sorted(missing)))
         assert not failures, failures
+
+
+class TestTopLevelHelpers:
+    def test_set_env_from_file(self, tmp_path):
+        import os
+
+        import mlrun
+
+        env_file = tmp_path / "test.env"
+        env_file.write_text("# comment\nMY_TEST_VAR=abc\nOTHER=1\n")
+        result = mlrun.set_env_from_file(str(env_file), return_dict=True)
+        assert result == {"MY_TEST_VAR": "abc", "OTHER": "1"}
+        assert os.environ["MY_TEST_VAR"] == "abc"
+        os.environ.pop("MY_TEST_VAR", None)
+        os.environ.pop("OTHER", None)
+
+    def test_set_env_from_file_errors(self, tmp_path):
+        import pytest as _pytest
+
+        import mlrun
+        from mlrun_amd.errors import (MLRunInvalidArgumentError,
+                                      MLRunNotFoundError)
+
+        with _pytest.raises(MLRunNotFoundError):
+            mlrun.set_env_from_file(str(tmp_path / "nope.env"))
+        bad = tmp_path / "bad.env"
+        bad.write_text("NOT A PAIR\n")
+        with _pytest.raises(MLRunInvalidArgumentError):
+            mlrun.set_env_from_file(str(bad))
+
+    def test_get_sample_path(self):
+        import mlrun
+
+        assert mlrun.get_sample_path("data/x.csv").endswith("data/x.csv")
+        base = mlrun.get_sample_path()
+        assert base.startswith("http")
+
+    def test_packager_exports(self):
+        import mlrun
+
+        assert hasattr(mlrun, "Packager")
+        assert hasattr(mlrun, "DefaultPackager")
+        assert hasattr(mlrun, "ArtifactType")
