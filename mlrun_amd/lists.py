@@ -121,31 +121,14 @@ class RunList(list):
     def compare(self, hide_identical: bool = True, exclude: list = None,
                 show: bool = None, extend_iterations: bool = True,
                 filename: str = None, colorscale: str = None):
-        """Tabular comparison of the runs' parameters and results
-        (reference returns a parallel-coordinates plot; headless
-        equivalent returns/saves the comparison table HTML)."""
-        df = self.to_df(flat=True, extend_iterations=extend_iterations,
-                        cache=False)
-        drop = [c for c in df.columns
-                if c in ("labels", "inputs", "artifacts",
-                         "artifact_uris", "error", "kind", "project")]
-        df = df.drop(columns=drop, errors="ignore")
-        if exclude:
-            df = df.drop(columns=list(exclude), errors="ignore")
-        if hide_identical and len(df) > 1:
-            keep = [c for c in df.columns
-                    if c in ("uid", "name", "state", "start")
-                    or df[c].astype(str).nunique() > 1]
-            df = df[keep]
-        html = df.to_html(index=False)
-        if filename:
-            with open(filename, "w") as f:
-                f.write(html)
-        if show:
-            from .render import _display
+        """Parallel-coordinates plot + table comparing the runs
+        (reference lists.py:136 → frameworks.parallel_coordinates)."""
+        from .frameworks.parallel_coordinates import compare_run_objects
 
-            _display(html)
-        return html
+        return compare_run_objects(
+            self, hide_identical=hide_identical, exclude=exclude,
+            show=show, extend_iterations=extend_iterations,
+            filename=filename, colorscale=colorscale)
 
 
 class ArtifactList(list):

@@ -115,3 +115,39 @@ class TestArtifactList:
         items = db.list_artifacts(project="p").dataitems()
         assert len(items) == 1
         assert "a,b" in items[0].get().decode()
+
+
+class TestParallelCoordinates:
+    def test_compare_run_objects_svg(self, db, tmp_path):
+        from mlrun_amd.frameworks.parallel_coordinates import (
+            compare_run_objects)
+
+        _store_runs(db)
+        runs = db.list_runs(project="p").to_objects()
+        out = tmp_path / "cmp.html"
+        html = compare_run_objects(runs, filename=str(out),
+                                   extend_iterations=False)
+        assert "<svg" in html and "polyline" in html
+        assert "param.lr" in html
+        assert out.exists()
+
+    def test_compare_db_runs(self, db, monkeypatch):
+        import mlrun_amd.db as db_mod
+        from mlrun_amd.frameworks.parallel_coordinates import (
+            compare_db_runs)
+
+        _store_runs(db)
+        monkeypatch.setattr(db_mod, "_run_db", db)
+        html = compare_db_runs(project_name="p", run_name="~train")
+        assert "<svg" in html
+
+    def test_categorical_axis(self):
+        import pandas as pd
+
+        from mlrun_amd.frameworks.parallel_coordinates import (
+            gen_pcp_plot)
+
+        df = pd.DataFrame({"param.optim": ["sgd", "adam", "sgd"],
+                           "output.acc": [0.8, 0.9, 0.85]})
+        html = gen_pcp_plot(df, "iter")
+        assert "adam" in html and "polyline" in html
