@@ -131,14 +131,19 @@ class TestParallelCoordinates:
         assert "param.lr" in html
         assert out.exists()
 
-    def test_compare_db_runs(self, db, monkeypatch):
+    def test_compare_db_runs(self, db):
         import mlrun_amd.db as db_mod
         from mlrun_amd.frameworks.parallel_coordinates import (
             compare_db_runs)
 
         _store_runs(db)
-        monkeypatch.setattr(db_mod, "_run_db", db)
-        html = compare_db_runs(project_name="p", run_name="~train")
+        prev = db_mod._run_db
+        db_mod.set_run_db(db)
+        try:
+            html = compare_db_runs(project_name="p",
+                                   run_name="~train")
+        finally:
+            db_mod._run_db = prev
         assert "<svg" in html
 
     def test_categorical_axis(self):
