@@ -22,6 +22,7 @@ from .sqldb import SQLRunDB
 _lock = threading.Lock()
 _run_db = None
 _run_db_url = None
+_run_db_pinned = False
 
 
 def create_run_db(url: str = "", secrets=None) -> RunDBInterface:
@@ -37,18 +38,25 @@ def get_run_db(url: str = None, secrets=None, force_reconnect=False) -> RunDBInt
     """Return the process-wide run DB (created on first use)."""
     global _run_db, _run_db_url
 
+    global _run_db_pinned
+    explicit_url = url is not None
     url = url if url is not None else str(config.dbpath or "")
     with _lock:
+        if _run_db_pinned and not explicit_url and not force_reconnect:
+            return _run_db
         if _run_db is None or force_reconnect or url != _run_db_url:
             _run_db = create_run_db(url, secrets)
             _run_db_url = url
+            _run_db_pinned = False
         return _run_db
 
 
 def set_run_db(db: RunDBInterface):
-    """Install a DB instance (used by tests to inject mocks)."""
-    global _run_db, _run_db_url
+    """Install (pin) a DB instance; get_run_db() returns it until a
+    reconnect or an explicit url is requested (test/mocking hook)."""
+    global _run_db, _run_db_url, _run_db_pinned
 
     with _lock:
         _run_db = db
         _run_db_url = getattr(db, "dsn", getattr(db, "base_url", "injected"))
+        _run_db_pinned = True

@@ -144,6 +144,7 @@ class TestParallelCoordinates:
                                    run_name="~train")
         finally:
             db_mod._run_db = prev
+            db_mod._run_db_pinned = False
         assert "<svg" in html
 
     def test_categorical_axis(self):
