@@ -39,8 +39,13 @@ class HuggingFaceModelServer(V2ModelServer):
         elif model_dir:
             import transformers
 
-            self.tokenizer = transformers.AutoTokenizer.from_pretrained(
-                model_dir)
+            try:
+                self.tokenizer = transformers.AutoTokenizer \
+                    .from_pretrained(model_dir)
+            except (OSError, ValueError):
+                # no tokenizer files in the model dir: serve raw
+                # input_ids (callers pre-tokenize)
+                self.tokenizer = None
             if task == "text-generation":
                 self.model = transformers.AutoModelForCausalLM \
                     .from_pretrained(model_dir, torch_dtype=torch.bfloat16)

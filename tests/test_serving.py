@@ -799,3 +799,61 @@ class TestRemoteModelUrl:
             assert resp["outputs"] == [6]
         finally:
             httpd.shutdown()
+
+
+def _tiny_hf_model_dir(tmp_path):
+    import transformers
+
+    cfg = transformers.LlamaConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=2, max_position_embeddings=128)
+    model = transformers.LlamaForCausalLM(cfg)
+    model.save_pretrained(str(tmp_path / "hf"))
+    return str(tmp_path / "hf")
+
+
+class TestHuggingFaceServer:
+    """HuggingFaceModelServer serves a real transformers model
+    (reference frameworks/huggingface/model_server.py:24)."""
+
+    def test_text_generation_cpu(self, tmp_path):
+        import mlrun_amd
+
+        model_dir = _tiny_hf_model_dir(tmp_path)
+        fn = mlrun_amd.new_function("hf", kind="serving")
+        fn.set_topology("router")
+        fn.add_model(
+            "tiny",
+            class_name="mlrun_amd.frameworks.huggingface."
+                       "HuggingFaceModelServer",
+            model_path=model_dir, task="text-generation",
+            device="cpu")
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/tiny/infer",
+                           body={"inputs": [[1, 2, 3, 4]],
+                                 "max_tokens": 4})
+        out = resp["outputs"]
+        assert len(out[0]) == 8  # 4 prompt + 4 generated ids
+
+    @pytest.mark.gpu
+    def test_text_generation_gpu(self, tmp_path):
+        import torch
+
+        import mlrun_amd
+
+        assert torch.cuda.is_available()
+        model_dir = _tiny_hf_model_dir(tmp_path)
+        fn = mlrun_amd.new_function("hf", kind="serving")
+        fn.set_topology("router")
+        fn.add_model(
+            "tiny",
+            class_name="mlrun_amd.frameworks.huggingface."
+                       "HuggingFaceModelServer",
+            model_path=model_dir, task="text-generation",
+            device="cuda:0")
+        server = fn.to_mock_server()
+        resp = server.test("/v2/models/tiny/infer",
+                           body={"inputs": [[1, 2, 3, 4]],
+                                 "max_tokens": 4})
+        assert len(resp["outputs"][0]) == 8
