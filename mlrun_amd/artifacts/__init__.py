@@ -288,13 +288,193 @@ class PlotArtifact(Artifact):
         self.spec.viewer = "chart"
 
 
+class DirArtifact(Artifact):
+    """Directory artifact (reference artifacts/base.py DirArtifact):
+    target_path is a directory tree; upload copies it recursively."""
+
+    kind = "dir"
+
+    def write_body(self, target_path: str):
+        import shutil
+
+        src = self.spec.src_path
+        if not src or not os.path.isdir(src):
+            raise MLRunInvalidArgumentError(
+                f"dir artifact needs a src_path directory (got {src!r})")
+        if os.path.abspath(src) != os.path.abspath(target_path):
+            shutil.copytree(src, target_path, dirs_exist_ok=True)
+
+
+class TableArtifact(Artifact):
+    """Tabular artifact from a DataFrame or csv body
+    (reference artifacts/plots.py TableArtifact)."""
+
+    kind = "table"
+
+    def __init__(self, key=None, body=None, df=None, viewer=None,
+                 visible=False, format=None, header=None, **kwargs):
+        if df is not None:
+            format = format or "csv"
+            body = df
+        super().__init__(key=key, body=body, format=format, **kwargs)
+        self._df = df
+        self.header = header
+        self.spec.viewer = viewer or "table"
+        self.visible = visible
+
+    def get_body(self):
+        if self._df is not None:
+            return self._df.to_csv(index=False)
+        body = self.spec.get_body() if hasattr(self.spec, "get_body") \
+            else self._body
+        return body
+
+
 artifact_types = {
     "artifact": Artifact,
     "": Artifact,
     "dataset": DatasetArtifact,
     "model": ModelArtifact,
     "plot": PlotArtifact,
+    "dir": DirArtifact,
+    "table": TableArtifact,
 }
+
+
+def dict_to_artifact(struct: dict) -> Artifact:
+    """Rebuild a typed artifact object from its dict form (reference
+    artifacts/__init__.py dict_to_artifact)."""
+    return Artifact.from_dict(struct)
+
+
+def get_artifact_meta(artifact):
+    """Resolve an artifact path / DataItem to (artifact object,
+    extra_data dict) — reference artifacts/base.py:801."""
+    import yaml as _yaml
+
+    from ..datastore import is_store_uri, store_manager
+
+    if hasattr(artifact, "artifact_url"):
+        artifact = artifact.artifact_url or artifact.url
+    if isinstance(artifact, Artifact):
+        spec = artifact
+    elif is_store_uri(artifact):
+        struct, _ = store_manager.get_store_artifact(artifact)
+        spec = dict_to_artifact(struct)
+    elif str(artifact).lower().endswith((".yaml", ".yml")):
+        data = store_manager.object(url=artifact).get()
+        spec = dict_to_artifact(_yaml.safe_load(data))
+    else:
+        raise MLRunInvalidArgumentError(
+            f"cant resolve artifact file for {artifact}")
+    extra_dataitems = {}
+    from ..run import get_dataitem
+
+    for key, item in (getattr(spec.spec, "extra_data", None) or {}).items():
+        extra_dataitems[key] = get_dataitem(item) if isinstance(
+            item, str) else item
+    return spec, extra_dataitems
+
+
+def _resolve_stored_artifact(artifact, expected_kind: str):
+    from ..datastore import is_store_uri, store_manager
+
+    if hasattr(artifact, "artifact_url"):
+        artifact = artifact.artifact_url or artifact.url
+    if isinstance(artifact, Artifact):
+        spec = artifact
+    elif isinstance(artifact, str) and is_store_uri(artifact):
+        struct, _ = store_manager.get_store_artifact(artifact)
+        spec = dict_to_artifact(struct)
+    else:
+        raise MLRunInvalidArgumentError(
+            f"{expected_kind} path must be a {expected_kind} store "
+            "object/URL/DataItem")
+    if spec.kind != expected_kind:
+        raise MLRunInvalidArgumentError(
+            f"store artifact ({artifact}) is not {expected_kind} kind")
+    return spec
+
+
+def _store_artifact_object(spec):
+    from ..db import get_run_db
+
+    meta = spec.metadata
+    get_run_db().store_artifact(
+        meta.key, spec.to_dict(), tag=meta.tag or "latest",
+        project=meta.project, tree=meta.tree)
+
+
+def update_dataset_meta(artifact, from_df=None, schema: dict = None,
+                        header: list = None, preview: list = None,
+                        stats: dict = None, extra_data: dict = None,
+                        column_metadata: dict = None, labels: dict = None,
+                        ignore_preview_limits: bool = False):
+    """Edit/add metadata on a stored dataset artifact (reference
+    artifacts/dataset.py:396)."""
+    spec = _resolve_stored_artifact(artifact, "dataset")
+    if from_df is not None:
+        spec._df = from_df
+        spec.before_log()
+    if header:
+        spec.status.header = header
+    if preview:
+        spec.status.preview = preview
+    if stats:
+        spec.status.stats = stats
+    if schema:
+        spec.spec.schema = schema
+    if column_metadata:
+        spec.spec.column_metadata = column_metadata
+    if extra_data:
+        existing = dict(getattr(spec.spec, "extra_data", None) or {})
+        existing.update(extra_data)
+        spec.spec.extra_data = existing
+    if labels:
+        spec.metadata.labels.update(labels)
+    _store_artifact_object(spec)
+    return spec
+
+
+def update_model(model_artifact, parameters: dict = None,
+                 metrics: dict = None, extra_data: dict = None,
+                 inputs: list = None, outputs: list = None,
+                 feature_vector: str = None, feature_weights: list = None,
+                 key_prefix: str = "", labels: dict = None,
+                 write_spec_copy: bool = True, store_object: bool = True):
+    """Edit/add attributes on a stored model artifact (reference
+    artifacts/model.py:515)."""
+    spec = _resolve_stored_artifact(model_artifact, "model")
+    for key, val in (parameters or {}).items():
+        spec.parameters[key] = val
+    for key, val in (metrics or {}).items():
+        spec.metrics[key_prefix + key] = val
+    for key, val in (labels or {}).items():
+        spec.metadata.labels[key] = val
+    if inputs:
+        spec.inputs = inputs
+    if outputs:
+        spec.outputs = outputs
+    if feature_vector:
+        spec.feature_vector = feature_vector
+    if feature_weights:
+        spec.feature_weights = feature_weights
+    if extra_data:
+        existing = dict(getattr(spec.spec, "extra_data", None) or {})
+        for key, item in extra_data.items():
+            if hasattr(item, "target_path"):
+                item = item.target_path
+            existing[key_prefix + key] = item
+        spec.spec.extra_data = existing
+    target_dir = os.path.dirname(spec.spec.target_path or "")
+    if write_spec_copy and target_dir and os.path.isdir(target_dir):
+        import yaml as _yaml
+
+        with open(os.path.join(target_dir, "model_spec.yaml"), "w") as f:
+            _yaml.safe_dump(spec.to_dict(), f, default_flow_style=False)
+    if store_object:
+        _store_artifact_object(spec)
+    return spec
 
 
 class ArtifactProducer:

@@ -501,3 +501,73 @@ def get_object(url: str, secrets=None, size=None, offset=0) -> bytes:
 
 def get_dataitem(url: str, secrets=None) -> DataItem:
     return store_manager.object(url)
+
+
+def is_store_uri(url) -> bool:
+    """True when the url is a store:// artifact reference
+    (reference store_resources.py:28)."""
+    return isinstance(url, str) and url.startswith("store://")
+
+
+def get_store_uri(kind: str, uri: str) -> str:
+    """Build a store://<kind>/<uri> reference
+    (reference store_resources.py:48)."""
+    return f"store://{kind}/{uri}"
+
+
+def parse_kafka_url(url: str, brokers=None):
+    """Split kafka://[broker[,broker]]/topic into (topic, brokers)
+    (reference datastore/utils.py:28)."""
+    from urllib.parse import urlparse
+
+    brokers = brokers or []
+    if isinstance(brokers, str):
+        brokers = brokers.split(",")
+    parsed = urlparse(url)
+    topic = parsed.path.strip("/")
+    if parsed.netloc:
+        brokers = brokers or parsed.netloc.split(",")
+    return topic, brokers
+
+
+def get_stream_pusher(stream_path: str, **kwargs):
+    """Stream pusher from a path/url (reference datastore
+    __init__.py:98): kafka:// -> KafkaOutputStream, http(s):// ->
+    HTTPOutputStream, dummy:// -> in-memory, else file-backed
+    OutputStream."""
+    from ..platforms import (HTTPOutputStream, KafkaOutputStream,
+                             OutputStream)
+
+    kafka_brokers = kwargs.get("kafka_brokers")
+    if stream_path.startswith("kafka://") or kafka_brokers:
+        topic, brokers = parse_kafka_url(stream_path, kafka_brokers)
+        return KafkaOutputStream(
+            topic, brokers, kwargs.get("kafka_producer_options"))
+    if stream_path.startswith(("http://", "https://")):
+        return HTTPOutputStream(stream_path)
+    if stream_path.startswith("dummy://"):
+        return OutputStream(stream_path)
+    return OutputStream(stream_path, **{
+        k: v for k, v in kwargs.items() if k in ("shards", "create")})
+
+
+from .sources import (  # noqa: F401,E402
+    BaseSource,
+    CSVSource,
+    DataFrameSource,
+    KafkaSource,
+    ParquetSource,
+    SQLSource,
+    StreamSource,
+)
+from .targets import (  # noqa: F401,E402
+    BaseStoreTarget,
+    CSVTarget,
+    KafkaTarget,
+    NoSqlTarget,
+    ParquetTarget,
+    RedisNoSqlTarget,
+    SQLTarget,
+    StreamTarget,
+    TSDBTarget,
+)

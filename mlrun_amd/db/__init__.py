@@ -60,3 +60,18 @@ def set_run_db(db: RunDBInterface):
         _run_db = db
         _run_db_url = getattr(db, "dsn", getattr(db, "base_url", "injected"))
         _run_db_pinned = True
+
+
+class RunDBError(Exception):
+    """Generic run-DB failure (reference db/base.py:29)."""
+
+
+def get_or_set_dburl(default: str = "") -> str:
+    """Return config.dbpath, setting it (and MLRUN_DBPATH) to default
+    first when unset (reference db/__init__.py:20)."""
+    import os
+
+    if not config.dbpath and default:
+        config.dbpath = default
+        os.environ["MLRUN_DBPATH"] = default
+    return config.dbpath

@@ -497,3 +497,66 @@ def get_online_feature_service(feature_vector, impute_policy: dict = None,
         tables[fs_name] = get_online_table(fset)
     return OnlineVectorService(vector, tables,
                                impute_policy=impute_policy or {})
+
+
+def get_feature_set(uri, project: str = None) -> FeatureSet:
+    """Get a feature set object from the db by uri
+    ({project}/{name}[:tag]) — reference api.py:1223."""
+    if project and "/" not in str(uri) and \
+            not str(uri).startswith("store://"):
+        uri = f"{project}/{uri}"
+    return _resolve_feature_set(uri)
+
+
+def get_feature_vector(uri, project: str = None):
+    """Get a feature vector object from the db by uri
+    ({project}/{name}[:tag]) — reference api.py:1232."""
+    from ..db import get_run_db
+    from .vector import FeatureVector
+
+    name = str(uri)
+    proj = project or "default"
+    if name.startswith("store://feature-vectors/"):
+        body = name[len("store://feature-vectors/"):]
+        proj, _, name = body.partition("/")
+    elif "/" in name:
+        proj, _, name = name.partition("/")
+    name, _, tag = name.partition(":")
+    struct = get_run_db().get_feature_vector(name, proj, tag=tag or None)
+    return FeatureVector.from_dict(struct)
+
+
+def delete_feature_set(name, project: str = "", tag: str = None,
+                       uid: str = None, force: bool = False):
+    """Delete a FeatureSet from the DB (reference api.py:1241); with
+    force=False refuses when materialized targets still exist."""
+    from ..db import get_run_db
+    from ..errors import MLRunInvalidArgumentError
+
+    if tag and uid:
+        raise MLRunInvalidArgumentError(
+            "both tag and uid must not be specified")
+    db = get_run_db()
+    if not force:
+        try:
+            struct = db.get_feature_set(name, project, tag=tag)
+        except Exception:
+            struct = None
+        targets = ((struct or {}).get("status") or {}).get("targets")
+        if targets:
+            raise MLRunInvalidArgumentError(
+                f"feature set {name} has materialized targets; delete "
+                "them first or pass force=True")
+    db.delete_feature_set(name, project, tag=tag)
+
+
+def delete_feature_vector(name, project: str = "", tag: str = None,
+                          uid: str = None):
+    """Delete a FeatureVector from the DB (reference api.py:1264)."""
+    from ..db import get_run_db
+    from ..errors import MLRunInvalidArgumentError
+
+    if tag and uid:
+        raise MLRunInvalidArgumentError(
+            "both tag and uid must not be specified")
+    get_run_db().delete_feature_vector(name, project, tag=tag)

@@ -11,6 +11,7 @@ aggregation serves the whole request batch) instead of per-row
 asyncio emits.
 """
 
+import enum
 import typing
 
 from ..errors import MLRunInvalidArgumentError
@@ -295,3 +296,12 @@ class OnlineVectorService:
 
     def __exit__(self, *exc):
         self.close()
+
+
+class FixedWindowType(enum.Enum):
+    """Which fixed window the online service reads (reference
+    feature_vector.py:449): the still-open current window or the last
+    closed one."""
+
+    CurrentOpenWindow = 1
+    LastClosedWindow = 2

@@ -29,3 +29,11 @@ from .apps import (  # noqa: F401
     ResultStatusApp,
 )
 from .writer import ModelMonitoringWriter  # noqa: F401
+from .model_endpoint import (  # noqa: F401
+    ModelEndpoint,
+    ModelEndpointMetadata,
+    ModelEndpointSpec,
+    ModelEndpointStatus,
+    TrackingPolicy,
+    get_stream_path,
+)

@@ -61,7 +61,39 @@ class OutputStream:
 
 class KafkaOutputStream(OutputStream):
     """Alias kind — the node-local build has no Kafka; kept for spec
-    compatibility (push/drain semantics are identical)."""
+    compatibility (push/drain semantics are identical).  Accepts the
+    reference (topic, brokers, producer_options) signature."""
+
+    def __init__(self, topic: str, brokers=None,
+                 producer_options: dict = None, **kwargs):
+        self.brokers = brokers if isinstance(brokers, list) else (
+            (brokers or "").split(",") if brokers else [])
+        self.producer_options = producer_options or {}
+        super().__init__(topic, **kwargs)
+
+
+class HTTPOutputStream(OutputStream):
+    """Push records by POSTing them to an HTTP endpoint (reference
+    platforms/iguazio.py HTTPOutputStream)."""
+
+    def __init__(self, stream_path: str, **kwargs):
+        self.path = stream_path
+        self._queue = None
+
+    def push(self, data):
+        import requests
+
+        records = data if isinstance(data, list) else [data]
+        for record in records:
+            if isinstance(record, (dict, list)):
+                resp = requests.post(self.path, json=record, timeout=10)
+            else:
+                resp = requests.post(self.path, data=record, timeout=10)
+            resp.raise_for_status()
+
+    def drain(self, max_batch: int = 4096) -> list:
+        raise NotImplementedError(
+            "HTTPOutputStream is write-only (consume at the endpoint)")
 
 
 def mount_v3io(*args, **kwargs):

@@ -16,3 +16,4 @@ from .operations import (  # noqa: F401
     deploy_function,
     run_function,
 )
+from .pipelines import load_and_run  # noqa: F401,E402

@@ -186,3 +186,40 @@ def run_workflow(project, path=None, handler=None, arguments=None,
     if state == "error":
         raise MLRunRuntimeError(f"workflow failed: {error}")
     return status
+
+
+def load_and_run(context, url: str = None, project_name: str = "",
+                 init_git: bool = None, subpath: str = None,
+                 clone: bool = False, save: bool = True,
+                 workflow_name: str = None, workflow_path: str = None,
+                 workflow_arguments: dict = None,
+                 artifact_path: str = None, workflow_handler=None,
+                 namespace: str = None, sync: bool = False,
+                 dirty: bool = False, engine: str = None,
+                 local: bool = None, schedule=None,
+                 cleanup_ttl: int = None, load_only: bool = False,
+                 wait_for_completion: bool = False,
+                 project_context: str = None):
+    """Load a project from a source and run one of its workflows —
+    the handler the remote/scheduled workflow runner executes
+    (reference pipelines.py:987)."""
+    from .project import load_project
+
+    project = load_project(
+        context=project_context or f"./{project_name}", url=url,
+        name=project_name, init_git=init_git, subpath=subpath,
+        clone=clone, save=save)
+    if context is not None and hasattr(context, "logger"):
+        context.logger.info(f"Loaded project {project.name} successfully")
+    if load_only:
+        return
+    status = project.run(
+        name=workflow_name or "", workflow_path=workflow_path or "",
+        arguments=workflow_arguments, artifact_path=artifact_path or "",
+        workflow_handler=workflow_handler, namespace=namespace,
+        sync=sync)
+    if context is not None and hasattr(context, "log_result"):
+        context.log_result("workflow_id", getattr(status, "run_id", ""))
+        context.log_result("workflow_state",
+                           getattr(status, "state", "completed"))
+    return status

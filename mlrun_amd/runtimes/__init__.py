@@ -84,3 +84,30 @@ def get_runtime_class(kind: str):
             f"kind='job' with hyper_param_options.parallel_runs for task "
             f"fan-out or kind='mpijob' for distributed (RCCL/xGMI) runs")
     raise MLRunInvalidArgumentError(f"unsupported runtime kind {kind!r}")
+
+
+class RunError(Exception):
+    """A run failed inside a runtime (reference runtimes/utils.py
+    RunError)."""
+
+
+def is_local_runtime(kind: str) -> bool:
+    return kind in RuntimeKinds.local_runtimes() or not kind
+
+
+from .mpijob import MpiRuntime  # noqa: F401,E402
+from .remote import ApplicationRuntime, RemoteRuntime  # noqa: F401,E402
+from .serving import ServingRuntime  # noqa: F401,E402
+from ..serving.v1_serving import (  # noqa: F401,E402
+    MLModelServer,
+    new_v1_model_server,
+)
+
+
+def new_model_server(*args, **kwargs):
+    """Create a serving function pre-loaded with model routes
+    (delegates to run.new_model_server — import deferred to avoid a
+    cycle)."""
+    from ..run import new_model_server as _factory
+
+    return _factory(*args, **kwargs)

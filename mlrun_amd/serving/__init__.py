@@ -6,8 +6,10 @@
 
 from .states import (  # noqa: F401
     BaseStep,
+    ErrorStep,
     FlowStep,
     GraphError,
+    MonitoringApplicationStep,
     QueueStep,
     RootFlowStep,
     RouterStep,
@@ -21,6 +23,7 @@ from .server import (  # noqa: F401
     MockEvent,
     create_graph_server,
 )
+from .v1_serving import MLModelServer, new_v1_model_server  # noqa: F401
 from .v2_serving import V2ModelServer  # noqa: F401
 from .routers import (  # noqa: F401
     BaseModelRouter,

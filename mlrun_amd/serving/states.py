@@ -92,11 +92,20 @@ class _StepBase:
     def error_handler(self, name: str = None, class_name=None, handler=None,
                       before=None, function=None, full_event=None,
                       **class_args):
-        """Route step errors to a named (or new) step."""
+        """Route step errors to a named (or new ErrorStep) step
+        (reference states.py:155); ``before`` names the step(s) the
+        flow continues with after the handler runs."""
         if class_name or handler:
-            step = self.parent.add_step(class_name, name=name,
-                                        handler=handler, full_event=full_event,
-                                        **class_args)
+            step = ErrorStep(class_name, class_args=class_args or None,
+                             handler=handler, name=name,
+                             full_event=full_event, function=function)
+            step.base_step = getattr(self, "name", None)
+            parent = self.parent if getattr(self, "parent", None) \
+                is not None else self
+            parent._attach(step)
+            if before:
+                step.before = [before] if isinstance(before, str) \
+                    else list(before)
             self.on_error = step.name
         else:
             self.on_error = name
@@ -559,6 +568,16 @@ class FlowStep(BaseStep):
                     self.steps[upstream]._next.append(step.name)
             elif step.name not in error_targets:
                 self._start_steps.append(step)
+        # ErrorStep.before: the flow continues with those steps after
+        # the handler runs (reference error_handler `before` param)
+        for step in self.steps.values():
+            for next_name in getattr(step, "before", None) or []:
+                if next_name not in self.steps:
+                    raise GraphError(
+                        f"error step {step.name} is before unknown "
+                        f"step {next_name}")
+                if next_name not in step._next:
+                    step._next.append(next_name)
         if self.final_step and self.final_step in self.steps:
             pass
 
@@ -650,6 +669,12 @@ class FlowStep(BaseStep):
             event.error = error_text
             event.origin_state = step.fullname
             result = self._run_from(self.steps[handler_name], event)
+            if result is not None and getattr(result, "error", None) == \
+                    error_text:
+                # the handler consumed the error: its output is a normal
+                # response (async recovery-step semantics; origin_state
+                # stays readable)
+                result.error = None
             return result
         event.error = error_text
         event.origin_state = step.fullname
@@ -693,11 +718,47 @@ class RootFlowStep(FlowStep):
     kind = "root"
 
 
+class ErrorStep(TaskStep):
+    """Error-handler execution step (reference states.py:635): a task
+    step attached via ``step.error_handler(...)``; ``before`` names
+    steps whose errors it handles, ``base_step`` the one it resumes."""
+
+    kind = "error_step"
+
+    def __init__(self, class_name=None, class_args=None, handler=None,
+                 name=None, after=None, full_event=None, function=None,
+                 responder=False, input_path=None, result_path=None):
+        super().__init__(class_name=class_name, class_args=class_args,
+                         handler=handler, name=name, after=after,
+                         full_event=full_event, function=function,
+                         responder=responder, input_path=input_path,
+                         result_path=result_path)
+        self.before = None
+        self.base_step = None
+
+    def to_dict(self):
+        struct = super().to_dict()
+        if self.before:
+            struct["before"] = self.before
+        if self.base_step:
+            struct["base_step"] = self.base_step
+        return struct
+
+
+class MonitoringApplicationStep(TaskStep):
+    """Model-monitoring application step (reference states.py:602):
+    a task step running a monitoring app class."""
+
+    kind = "monitoring_application"
+
+
 classes_map = {
     "task": TaskStep,
     "router": RouterStep,
     "flow": FlowStep,
     "queue": QueueStep,
+    "error_step": ErrorStep,
+    "monitoring_application": MonitoringApplicationStep,
 }
 
 
@@ -743,7 +804,8 @@ def step_from_dict(struct: dict) -> BaseStep:
         step = QueueStep(name=struct.get("name"), path=struct.get("path"))
         step.after = struct.get("after", [])
         return step
-    step = TaskStep(class_name=struct.get("class_name"),
+    step_cls = classes_map.get(kind, TaskStep)
+    step = step_cls(class_name=struct.get("class_name"),
                     class_args=struct.get("class_args"),
                     handler=struct.get("handler"),
                     name=struct.get("name"),
@@ -751,4 +813,7 @@ def step_from_dict(struct: dict) -> BaseStep:
     step.after = struct.get("after", [])
     step.on_error = struct.get("on_error")
     step.responder = struct.get("responder", False)
+    if isinstance(step, ErrorStep):
+        step.before = struct.get("before")
+        step.base_step = struct.get("base_step")
     return step

@@ -65,3 +65,19 @@ class MLModelServer:
         logger.debug("v1 predict", model=self.name,
                      ms=(time.perf_counter() - start) * 1000.0)
         return event
+
+
+def new_v1_model_server(name, model_class: str, models: dict = None,
+                        filename="", protocol="", image="", endpoint="",
+                        workers=8, canary=None):
+    """Create a (legacy) v1 model-server function
+    (reference v1_serving.py:33): a serving function whose routes use
+    the flask-style MLModelServer protocol."""
+    from ..run import new_function
+
+    fn = new_function(name=name, kind="serving", command=filename,
+                      image=image)
+    fn.spec.parameters["protocol"] = protocol or "v1"
+    for key, model_path in (models or {}).items():
+        fn.add_model(key, model_path=model_path, class_name=model_class)
+    return fn
