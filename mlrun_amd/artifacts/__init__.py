@@ -651,11 +651,26 @@ class ArtifactManager:
                     body = body.encode()
                 with open(os.path.join(target_dir, model.model_file), "wb") as fp:
                     fp.write(body)
-            elif model.spec.src_path and os.path.isdir(model.spec.src_path):
+            elif model_dir and os.path.isdir(model_dir):
                 import shutil
 
-                shutil.copytree(model.spec.src_path, target_dir,
+                src_abs = os.path.abspath(model_dir)
+                dst_abs = os.path.abspath(target_dir)
+                if dst_abs.startswith(src_abs + os.sep):
+                    raise MLRunInvalidArgumentError(
+                        f"model target dir {target_dir} is inside the "
+                        f"source dir {model_dir}; use a different "
+                        "artifact_path")
+                shutil.copytree(model_dir, target_dir,
                                 dirs_exist_ok=True)
+            elif model_file and os.path.isfile(model_file):
+                # only the named model file travels (its directory may
+                # contain unrelated files — or the target itself)
+                import shutil
+
+                dst = os.path.join(target_dir, model.model_file)
+                if os.path.abspath(model_file) != os.path.abspath(dst):
+                    shutil.copyfile(model_file, dst)
             for name, value in (extra_data or {}).items():
                 if isinstance(value, (str, bytes)):
                     data = value.encode() if isinstance(value, str) else value

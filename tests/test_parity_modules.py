@@ -596,3 +596,166 @@ class TestTopLevelHelpers:
         assert hasattr(mlrun, "Packager")
         assert hasattr(mlrun, "DefaultPackager")
         assert hasattr(mlrun, "ArtifactType")
+
+
+class TestExportSurfaceBatch:
+    """Round-2 export-surface additions (reference subpackage
+    __init__ exports)."""
+
+    def test_serving_exports(self):
+        from mlrun_amd.serving import (ErrorStep, MLModelServer,
+                                       MonitoringApplicationStep,
+                                       new_v1_model_server)
+
+        step = ErrorStep(handler="h", name="err")
+        step.before = ["echo"]
+        d = step.to_dict()
+        assert d["kind"] == "error_step" and d["before"] == ["echo"]
+        from mlrun_amd.serving.states import step_from_dict
+
+        back = step_from_dict(d)
+        assert isinstance(back, ErrorStep) and back.before == ["echo"]
+        assert MonitoringApplicationStep(
+            handler="h").to_dict()["kind"] == "monitoring_application"
+        assert MLModelServer is not None
+        fn = new_v1_model_server("v1srv", model_class="MyModel")
+        assert fn.kind == "serving"
+
+    def test_error_handler_before_both_engines(self):
+        import mlrun_amd
+
+        def raising_step(x):
+            raise ValueError("boom")
+
+        def handle_error(event):
+            event.body = {"handled": str(event.error)}
+            return event
+
+        def echo(x):
+            return {"echo": x}
+
+        namespace = {"raising_step": raising_step,
+                     "handle_error": handle_error, "echo": echo}
+        for engine in ("sync", "async"):
+            fn = mlrun_amd.new_function("g", kind="serving")
+            graph = fn.set_topology("flow", engine=engine)
+            graph.to(name="raise", handler="raising_step") \
+                .error_handler(name="error_catcher",
+                               handler="handle_error",
+                               full_event=True, before="echo")
+            graph.add_step(name="echo", handler="echo",
+                           after="raise").respond()
+            server = fn.to_mock_server(namespace=namespace)
+            resp = server.test("/", body={"a": 1})
+            assert resp["echo"]["handled"] == "ValueError: boom", \
+                (engine, resp)
+            server.graph.shutdown()
+
+    def test_artifacts_helpers(self, tmp_path):
+        import pandas as pd
+
+        from mlrun_amd.artifacts import (DirArtifact, TableArtifact,
+                                         dict_to_artifact)
+
+        table = TableArtifact("t1", df=pd.DataFrame({"a": [1, 2]}))
+        assert "a" in table.get_body()
+        art = dict_to_artifact({"kind": "dir",
+                                "metadata": {"key": "d1"}})
+        assert isinstance(art, DirArtifact)
+
+    def test_update_model_and_dataset_meta(self, tmp_path):
+        import mlrun_amd
+        import mlrun_amd.db as db_mod
+        from mlrun_amd.artifacts import update_model
+        from mlrun_amd.db.sqldb import SQLRunDB
+
+        db = SQLRunDB(str(tmp_path / "m.db"))
+        prev = db_mod._run_db
+        db_mod.set_run_db(db)
+        try:
+            from mlrun_amd.config import config as _cfg
+
+            prev_ap = _cfg.artifact_path
+            _cfg.artifact_path = str(tmp_path / "arts")
+            ctx = mlrun_amd.get_or_create_ctx("upd", project="p")
+            model_file = tmp_path / "model.bin"
+            model_file.write_bytes(b"weights")
+            model = ctx.log_model("mymodel",
+                                  model_file=str(model_file))
+            update_model(model.uri, metrics={"accuracy": 0.99},
+                         labels={"stage": "prod"})
+            stored = db.read_artifact("mymodel", project="p")
+            assert stored["spec"]["metrics"]["accuracy"] == 0.99
+            assert stored["metadata"]["labels"]["stage"] == "prod"
+        finally:
+            _cfg.artifact_path = prev_ap
+            db_mod._run_db = prev
+            db_mod._run_db_pinned = False
+
+    def test_datastore_helpers(self):
+        from mlrun_amd.datastore import (CSVSource, ParquetTarget,
+                                         get_store_uri,
+                                         get_stream_pusher,
+                                         is_store_uri, parse_kafka_url)
+
+        assert is_store_uri("store://artifacts/p/k")
+        assert not is_store_uri("/tmp/x")
+        assert get_store_uri("models", "p/m") == "store://models/p/m"
+        topic, brokers = parse_kafka_url("kafka://b1:9092/topic")
+        assert topic == "topic" and brokers == ["b1:9092"]
+        pusher = get_stream_pusher("teststream")
+        pusher.push({"x": 1})
+        assert len(pusher.drain()) == 1
+        assert CSVSource is not None and ParquetTarget is not None
+
+    def test_feature_store_exports(self, tmp_path):
+        import mlrun_amd.db as db_mod
+        import mlrun_amd.feature_store as fstore
+        from mlrun_amd.db.sqldb import SQLRunDB
+
+        assert fstore.FixedWindowType.LastClosedWindow.value == 2
+        config = fstore.RunConfig(local=True, parameters={"x": 1})
+        config.with_secret("inline", {"k": "v"})
+        assert config.secret_sources[0]["kind"] == "inline"
+        db = SQLRunDB(str(tmp_path / "fs.db"))
+        prev = db_mod._run_db
+        db_mod.set_run_db(db)
+        try:
+            fset = fstore.FeatureSet("fs1", entities=["uid"])
+            db.store_feature_set(fset.to_dict(), name="fs1",
+                                 project="p")
+            got = fstore.get_feature_set("p/fs1")
+            assert got.metadata.name == "fs1"
+            fstore.delete_feature_set("fs1", project="p", force=True)
+        finally:
+            db_mod._run_db = prev
+            db_mod._run_db_pinned = False
+
+    def test_model_monitoring_objects(self):
+        from mlrun_amd.model_monitoring import (ModelEndpoint,
+                                                TrackingPolicy,
+                                                get_stream_path)
+
+        ep = ModelEndpoint()
+        ep.spec.function_uri = "p/f:latest"
+        ep.spec.model = "m:v1"
+        uid = ep.create_endpoint_id()
+        assert len(uid) == 40
+        back = ModelEndpoint.from_dict(ep.to_dict())
+        assert back.spec.model == "m:v1"
+        assert TrackingPolicy().base_period == 10
+        assert get_stream_path("p").endswith("p/stream")
+
+    def test_db_and_runtime_helpers(self):
+        from mlrun_amd.db import RunDBError, get_or_set_dburl
+        from mlrun_amd.runtimes import (RunError, ServingRuntime,
+                                        is_local_runtime,
+                                        new_model_server)
+
+        assert issubclass(RunDBError, Exception)
+        assert issubclass(RunError, Exception)
+        assert is_local_runtime("local") and not is_local_runtime("job")
+        assert get_or_set_dburl() is not None
+        fn = new_model_server("msrv", model_class="MyServer",
+                              models={"m": "/tmp/m"})
+        assert fn.kind == "serving" and ServingRuntime is not None
