@@ -730,3 +730,32 @@ def get_model(model_dir_or_uri: str, suffix: str = "") -> typing.Tuple[
                 key=files[0], model_file=files[0]), {}
     raise MLRunInvalidArgumentError(
         f"cannot resolve model from {model_dir_or_uri}")
+
+
+class PlotlyArtifact(Artifact):
+    """Plotly-figure artifact stored as html (reference
+    artifacts/plots.py:77).  Requires plotly at construction time;
+    the stored form is plain html so readers don't need plotly."""
+
+    kind = "plotly"
+
+    def __init__(self, figure=None, key: str = None,
+                 target_path: str = None, **kwargs):
+        super().__init__(key=key, target_path=target_path,
+                         format="html", **kwargs)
+        self.spec.viewer = "plotly"
+        self._figure = figure
+        if figure is not None:
+            module = type(figure).__module__
+            if not module.startswith("plotly"):
+                raise MLRunInvalidArgumentError(
+                    "PlotlyArtifact requires a plotly Figure object "
+                    f"(got {module}.{type(figure).__name__})")
+
+    def get_body(self):
+        if self._figure is not None:
+            return self._figure.to_html()
+        return super().get_body()
+
+
+artifact_types["plotly"] = PlotlyArtifact

@@ -571,3 +571,105 @@ from .targets import (  # noqa: F401,E402
     StreamTarget,
     TSDBTarget,
 )
+
+
+def parse_path(url: str):
+    """Split a scheme url into (endpoint, path) (reference
+    datastore/__init__.py parse_path)."""
+    parsed = urlparse(url)
+    endpoint = parsed.netloc
+    return endpoint, parsed.path.lstrip("/")
+
+
+def get_kafka_brokers_from_dict(options: dict, pop: bool = False):
+    """Extract kafka brokers from an options dict (reference
+    datastore/utils.py)."""
+    if not isinstance(options, dict):
+        return None
+    key = "kafka_brokers"
+    value = options.pop(key, None) if pop else options.get(key)
+    return value
+
+
+def uri_to_ipython(link: str) -> str:
+    """Render a data uri as a notebook link target (reference
+    datastore/__init__.py uri_to_ipython)."""
+    if is_store_uri(link):
+        return ""
+    return link
+
+
+def get_store_resource(uri: str, db=None, secrets=None, project: str = "",
+                       data_store_secrets=None):
+    """Resolve a store:// uri to its object (artifact / feature set /
+    feature vector) — reference store_resources.py:144."""
+    kind, project_, name, tag, tree, iteration = parse_store_uri(
+        uri, project)
+    if db is None:
+        from ..db import get_run_db
+
+        db = get_run_db(secrets=secrets) if secrets else get_run_db()
+    if kind in ("feature-sets", "feature_sets"):
+        from ..feature_store.feature_set import FeatureSet
+
+        return FeatureSet.from_dict(
+            db.get_feature_set(name, project_, tag=tag))
+    if kind in ("feature-vectors", "feature_vectors"):
+        from ..feature_store.vector import FeatureVector
+
+        return FeatureVector.from_dict(
+            db.get_feature_vector(name, project_, tag=tag))
+    if kind in ("artifacts", "models", "datasets", "artifact", "model",
+                "dataset"):
+        from ..artifacts import dict_to_artifact
+
+        return dict_to_artifact(
+            db.read_artifact(name, tag=tag, project=project_, tree=tree,
+                             iter=iteration))
+    raise MLRunInvalidArgumentError(
+        f"unsupported store uri kind {kind!r} in {uri}")
+
+
+class HttpSource:
+    """Read a remote http(s) object as a source (reference
+    sources.py HttpSource) — node-local build fetches via requests."""
+
+    kind = "http"
+
+    def __init__(self, name: str = "", path: str = None, **kwargs):
+        self.name = name
+        self.path = path
+
+    def to_dataframe(self, **kwargs):
+        import io as _io
+
+        import pandas as pd
+        import requests
+
+        resp = requests.get(self.path, timeout=30)
+        resp.raise_for_status()
+        if self.path.endswith(".parquet") or self.path.endswith(".pq"):
+            return pd.read_parquet(_io.BytesIO(resp.content))
+        return pd.read_csv(_io.BytesIO(resp.content))
+
+
+in_memory_store = MemoryStore(None, "memory", "memory")
+
+
+def set_in_memory_item(key: str, value):
+    """Store an object under memory://<key> (reference
+    datastore/__init__.py set_in_memory_item)."""
+    in_memory_store.put_object(key.lstrip("/"), value)
+    return DataItem(key, in_memory_store, key.lstrip("/"),
+                    f"memory://{key.lstrip('/')}")
+
+
+def get_in_memory_items() -> dict:
+    return dict(MemoryStore._items)
+
+
+from ..platforms import (  # noqa: F401,E402
+    HTTPOutputStream,
+    KafkaOutputStream,
+    OutputStream,
+)

@@ -26,6 +26,7 @@ from .api import (  # noqa: F401
     delete_feature_set,
     delete_feature_vector,
     deploy_ingestion_service,
+    deploy_ingestion_service_v2,
     get_feature_set,
     get_feature_vector,
     get_offline_features,

@@ -560,3 +560,13 @@ def delete_feature_vector(name, project: str = "", tag: str = None,
         raise MLRunInvalidArgumentError(
             "both tag and uid must not be specified")
     get_run_db().delete_feature_vector(name, project, tag=tag)
+
+
+def deploy_ingestion_service_v2(featureset, source=None, targets=None,
+                                name: str = None, run_config=None,
+                                verbose=False):
+    """V2 alias of deploy_ingestion_service (reference api.py) —
+    returns (endpoint/url, function object)."""
+    service = deploy_ingestion_service(featureset, source=source,
+                                       targets=targets)
+    return getattr(service, "endpoint", ""), service

@@ -37,3 +37,19 @@ from .model_endpoint import (  # noqa: F401
     TrackingPolicy,
     get_stream_path,
 )
+
+
+def get_store_object(project: str = "", **kwargs):
+    """Endpoint-record store for a project (reference
+    model_monitoring/db get_store_object): the node-local build keeps
+    endpoint records in the run DB."""
+    from ..db import get_run_db
+
+    return get_run_db()
+
+
+def get_tsdb_connector(project: str = "", **kwargs):
+    """Time-series store connector (reference get_tsdb_connector):
+    the node-local build records window stats in the stream
+    processor's in-memory TSDB."""
+    return get_stream_processor(project=project)

@@ -17,3 +17,4 @@ from .operations import (  # noqa: F401
     run_function,
 )
 from .pipelines import load_and_run  # noqa: F401,E402
+from .project import ProjectStatus  # noqa: F401,E402

@@ -599,3 +599,4 @@ class _PipelineContext:
 
 
 pipeline_context = _PipelineContext()
+

@@ -111,3 +111,39 @@ def new_model_server(*args, **kwargs):
     from ..run import new_model_server as _factory
 
     return _factory(*args, **kwargs)
+
+
+# reference runtimes/__init__ aliases & constants
+MpiRuntimeV1 = MpiRuntime
+serving_subkind = "v2"
+
+
+class MPIJobCRDVersions:
+    v1 = "v1"
+    v1alpha1 = "v1alpha1"
+
+    @staticmethod
+    def all():
+        return [MPIJobCRDVersions.v1, MPIJobCRDVersions.v1alpha1]
+
+    @staticmethod
+    def default():
+        return MPIJobCRDVersions.v1
+
+
+class RuntimeClassMode:
+    """How a runtime class is used (reference runtimes/__init__
+    RuntimeClassMode): building a run object vs monitoring one."""
+
+    run = "run"
+    build = "build"
+
+
+def new_v2_model_server(name: str, model_class: str, models: dict = None,
+                        filename: str = "", protocol: str = "",
+                        image: str = "", workers: int = 8, **kwargs):
+    """Create a V2 (KFServing-v2 protocol) model-server function
+    (reference runtimes/__init__.py new_v2_model_server)."""
+    return new_model_server(name, model_class=model_class, models=models,
+                            filename=filename, protocol=protocol or "v2",
+                            image=image, **kwargs)
