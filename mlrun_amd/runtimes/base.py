@@ -134,6 +134,13 @@ class BaseRuntime(ModelObj):
     def status(self, value):
         self._status = self._verify_dict(value, "status", FunctionStatus)
 
+    def apply(self, modifier):
+        """Apply a function modifier (reference KubeResource.apply —
+        e.g. ``fn.apply(mlrun.platforms.auto_mount())``); modifiers
+        are callables taking the runtime object."""
+        modifier(self)
+        return self
+
     def is_deployed(self) -> bool:
         return True
 

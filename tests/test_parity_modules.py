@@ -759,3 +759,94 @@ class TestExportSurfaceBatch:
         fn = new_model_server("msrv", model_class="MyServer",
                               models={"m": "/tmp/m"})
         assert fn.kind == "serving" and ServingRuntime is not None
+
+
+class TestExportSurfaceBatch2:
+    def test_get_store_resource_feature_set(self, tmp_path):
+        import mlrun_amd.db as db_mod
+        import mlrun_amd.feature_store as fstore
+        from mlrun_amd.datastore import get_store_resource
+        from mlrun_amd.db.sqldb import SQLRunDB
+
+        db = SQLRunDB(str(tmp_path / "sr.db"))
+        prev = db_mod._run_db
+        db_mod.set_run_db(db)
+        try:
+            fset = fstore.FeatureSet("fs1", entities=["uid"])
+            db.store_feature_set(fset.to_dict(), name="fs1",
+                                 project="p")
+            got = get_store_resource("store://feature-sets/p/fs1")
+            assert got.metadata.name == "fs1"
+        finally:
+            db_mod._run_db = prev
+            db_mod._run_db_pinned = False
+
+    def test_in_memory_items(self):
+        from mlrun_amd.datastore import (get_in_memory_items,
+                                         set_in_memory_item)
+
+        item = set_in_memory_item("box/data", b"abc123")
+        assert item.get() == b"abc123"
+        assert any(k.startswith("box/")
+                   for k in get_in_memory_items())
+
+    def test_plotly_artifact_rejects_non_figures(self):
+        import pytest as _pytest
+
+        from mlrun_amd.artifacts import PlotlyArtifact
+        from mlrun_amd.errors import MLRunInvalidArgumentError
+
+        with _pytest.raises(MLRunInvalidArgumentError):
+            PlotlyArtifact(figure=object(), key="p1")
+
+    def test_run_config_to_function(self):
+        import mlrun_amd
+        from mlrun_amd.feature_store import RunConfig
+
+        fn = mlrun_amd.new_function("ing", kind="job")
+        config = RunConfig(fn, parameters={"x": 1})
+        materialized = config.to_function()
+        assert materialized.kind == "job"
+        config2 = RunConfig(kind="job", image="mlrun/mlrun")
+        fn2 = config2.to_function()
+        assert fn2.spec.image == "mlrun/mlrun"
+
+    def test_runtime_aliases(self):
+        from mlrun_amd.runtimes import (MPIJobCRDVersions, MpiRuntime,
+                                        MpiRuntimeV1,
+                                        new_v2_model_server)
+
+        assert MpiRuntimeV1 is MpiRuntime
+        assert MPIJobCRDVersions.default() == "v1"
+        fn = new_v2_model_server("v2srv", model_class="S")
+        assert fn.kind == "serving"
+
+    def test_http_output_stream_and_source(self):
+        import json
+        import threading
+        from http.server import BaseHTTPRequestHandler, HTTPServer
+
+        from mlrun_amd.datastore import get_stream_pusher
+
+        received = []
+
+        class Handler(BaseHTTPRequestHandler):
+            def do_POST(self):
+                n = int(self.headers.get("content-length", 0))
+                received.append(json.loads(self.rfile.read(n)))
+                self.send_response(200)
+                self.end_headers()
+
+            def log_message(self, *a):
+                pass
+
+        httpd = HTTPServer(("127.0.0.1", 0), Handler)
+        port = httpd.server_address[1]
+        threading.Thread(target=httpd.serve_forever,
+                         daemon=True).start()
+        try:
+            pusher = get_stream_pusher(f"http://127.0.0.1:{port}/s")
+            pusher.push({"k": 1})
+            assert received == [{"k": 1}]
+        finally:
+            httpd.shutdown()
