@@ -34,3 +34,8 @@ from .onnx import ONNXModelServer  # noqa: F401
 from .tf_keras import TFKerasModelServer  # noqa: F401
 from .torch_nn import PyTorchModelServer  # noqa: F401
 from .auto import apply_mlrun, detect_framework, get_model_server_class  # noqa: F401
+from .huggingface import HuggingFaceModelServer  # noqa: F401
+from .parallel_coordinates import (  # noqa: F401
+    compare_db_runs,
+    compare_run_objects,
+)
