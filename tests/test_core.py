@@ -299,3 +299,83 @@ class TestSQLRunDBConcurrency:
         assert not errors, errors[:2]
         assert len(rundb.list_runs(project="default",
                                    last=1000)) >= 240
+
+
+class TestUtilsHelpers:
+    """Reference utils/helpers.py surface (curated public set)."""
+
+    def test_serialization_helpers(self):
+        from mlrun_amd.utils import (dict_to_json, dict_to_str,
+                                     dict_to_yaml)
+
+        assert "a: 1" in dict_to_yaml({"a": 1})
+        assert dict_to_json({"a": 1}) == '{"a": 1}'
+        assert dict_to_str({"a": 1, "b": 2}) == "a=1,b=2"
+
+    def test_uri_helpers(self):
+        from mlrun_amd.utils import (generate_artifact_uri,
+                                     generate_object_uri,
+                                     parse_artifact_uri)
+
+        assert generate_artifact_uri("p", "k", tag="t", iter=1) == \
+            "p/k#1:t"
+        assert generate_object_uri("p", "f", hash_key="abc") == \
+            "p/f@abc"
+        assert parse_artifact_uri("p/key#2:tag@tree") == \
+            ("p", "key", 2, "tag", "tree")
+        assert parse_artifact_uri("key", "dflt")[0] == "dflt"
+
+    def test_object_hash_stable(self):
+        from mlrun_amd.utils import fill_object_hash
+
+        a = {"metadata": {"name": "x", "updated": "t1"},
+             "spec": {"v": 1}, "status": {"state": "ready"}}
+        b = {"metadata": {"name": "x", "updated": "t2"},
+             "spec": {"v": 1}, "status": {"state": "error"}}
+        assert fill_object_hash(a) == fill_object_hash(b)
+        c = {"metadata": {"name": "x"}, "spec": {"v": 2}}
+        assert fill_object_hash(c) != fill_object_hash(a)
+
+    def test_dynamic_loading(self):
+        from mlrun_amd.utils import get_class, get_function
+
+        assert get_class("mlrun_amd.artifacts.Artifact").kind == \
+            "artifact"
+        fn = get_function("mlrun_amd.utils.dict_to_json")
+        assert fn({"x": 1}) == '{"x": 1}'
+
+    def test_time_and_chunk_helpers(self):
+        import pandas as pd
+
+        from mlrun_amd.utils import (iterate_list_by_chunks,
+                                     str_to_timestamp)
+
+        assert str_to_timestamp("now - 1h") < pd.Timestamp.now()
+        assert str_to_timestamp("2026-01-01").year == 2026
+        assert list(iterate_list_by_chunks([1, 2, 3, 4, 5], 2)) == \
+            [[1, 2], [3, 4], [5]]
+
+    def test_path_safety_and_templates(self):
+        from mlrun_amd.utils import (StorePrefix, is_safe_path,
+                                     template_artifact_path)
+
+        assert is_safe_path("/tmp/base", "/tmp/base/sub/x")
+        assert not is_safe_path("/tmp/base", "/tmp/base/../etc")
+        assert StorePrefix.kind_to_prefix("model") == "models"
+        assert template_artifact_path(
+            "/data/{{project}}/{{run.uid}}", "p", "u1") == "/data/p/u1"
+
+    def test_retry_until_successful(self):
+        from mlrun_amd.utils import retry_until_successful
+
+        calls = {"n": 0}
+
+        def flaky():
+            calls["n"] += 1
+            if calls["n"] < 3:
+                raise RuntimeError("not yet")
+            return "done"
+
+        assert retry_until_successful(0.01, 5, None, False,
+                                      flaky) == "done"
+        assert calls["n"] == 3
