@@ -392,7 +392,130 @@ class MLClientCtx:
         else:
             self.commit_db()
 
+    # context-manager: `with ctx.get_child_context(..) as child` —
+    # exit commits the child (errors mark it failed)
+    def __enter__(self):
+        return self
+
+    def __exit__(self, exc_type, exc_value, exc_tb):
+        if exc_type is not None:
+            self.set_state(RunStates.error, str(exc_value))
+        elif getattr(self, "_parent", None) is not None:
+            self.commit(completed=True)
+        return False
+
+    def get_cached_artifact(self, key: str):
+        """Logged-artifact object by key (reference execution.py
+        get_cached_artifact)."""
+        if self._artifacts_manager is None:
+            return None
+        return self._artifacts_manager.artifacts.get(key)
+
+    def get_notifications(self, unmask_secret_params: bool = False):
+        return getattr(self, "_notifications", []) or []
+
+    def get_project_object(self):
+        """Load this run's project object from the DB (reference
+        execution.py get_project_object)."""
+        from .projects import get_or_create_project
+
+        return get_or_create_project(self._project or "default")
+
+    def get_project_param(self, key: str, default=None):
+        project = self.get_project_object()
+        if project is None:
+            return default
+        return project.spec.params.get(key, default)
+
+    @property
+    def log_level(self) -> str:
+        import logging as _logging
+
+        return _logging.getLevelName(
+            getattr(self._logger, "level", _logging.INFO)).lower()
+
+    def set_logger_stream(self, stream):
+        handler = getattr(self._logger, "_handler", None)
+        if handler is not None:
+            handler.stream = stream
+
+    def get_child_context(self, with_parent_params: bool = False,
+                          **params) -> "MLClientCtx":
+        """Child context for a sub-experiment iteration (reference
+        execution.py:223): log_xx on the child updates that iteration
+        only; ``update_child_iterations`` folds them into the parent;
+        ``child.mark_as_best()`` marks its iteration.
+
+        Example::
+
+            for param in param_list:
+                with context.get_child_context(p=param) as child:
+                    accuracy = child_handler(child, **child.parameters)
+                    child.log_result("accuracy", accuracy)
+        """
+        from .errors import MLRunInvalidArgumentError
+
+        if self._iteration != 0:
+            raise MLRunInvalidArgumentError(
+                "cannot create child from a child iteration!")
+        struct = self.to_dict()
+        spec = struct.setdefault("spec", {})
+        spec["parameters"] = dict(self._parameters) \
+            if with_parent_params else {}
+        spec["parameters"].update(params)
+        struct["status"] = {}
+        struct.setdefault("metadata", {})["iteration"] = \
+            len(self._children) + 1
+        child = MLClientCtx.from_dict(
+            struct, rundb=self._db, autocommit=self._autocommit,
+            log_stream=self._logger, store_run=False)
+        child._artifacts_manager = self._artifacts_manager
+        child._parent = self
+        self._children.append(child)
+        return child
+
+    def update_child_iterations(self, best_run: int = 0,
+                                commit_children: bool = False,
+                                completed: bool = True):
+        """Fold child-iteration results into this (parent) run
+        (reference execution.py:271): builds the iteration table,
+        records the best child's results."""
+        if not self._children:
+            return
+        if commit_children:
+            for child in self._children:
+                child.commit(completed=completed)
+        # iteration table: header + one row per child
+        result_keys: list = []
+        param_keys: list = []
+        for child in self._children:
+            for key in child._parameters:
+                if key not in param_keys:
+                    param_keys.append(key)
+            for key in child._results:
+                if key not in result_keys:
+                    result_keys.append(key)
+        header = ["state", "iter"] + [f"param.{k}" for k in param_keys] \
+            + [f"output.{k}" for k in result_keys]
+        rows = [header]
+        for child in self._children:
+            rows.append([child._state, child._iteration]
+                        + [child._parameters.get(k) for k in param_keys]
+                        + [child._results.get(k) for k in result_keys])
+        task = self._children[best_run - 1].to_dict() if best_run else None
+        self.log_iteration_results(best_run, rows, task)
+
+    @property
+    def _children(self) -> list:
+        return self._child_iterations
+
     def mark_as_best(self):
+        """Mark a child run as the best iteration (reference
+        execution.py:291)."""
+        parent = getattr(self, "_parent", None)
+        if parent is not None and self._iteration:
+            parent.log_iteration_results(self._iteration, None,
+                                         self.to_dict())
         self.set_label("best_iteration", self._iteration)
 
     # ------------------------------------------------------------ dicts

@@ -308,6 +308,22 @@ class RunTemplate(ModelObj):
         self.spec.inputs[name] = path
         return self
 
+    def with_param_file(self, param_file, selector=None, strategy=None,
+                        **options):
+        """Hyper-param values from a csv/json file url (reference
+        model.py:1379)."""
+        opts = self.spec.hyper_param_options or HyperParamOptions()
+        for key, value in options.items():
+            setattr(opts, key, value)
+        opts.param_file = param_file
+        if selector:
+            opts.selector = selector
+        if strategy:
+            opts.strategy = strategy
+        self.spec.hyper_param_options = opts
+        self.spec.selector = selector or self.spec.selector
+        return self
+
     def with_hyper_params(self, hyperparams, selector=None, strategy=None,
                           **options):
         self.spec.hyperparams = hyperparams
@@ -416,6 +432,15 @@ class RunObject(RunTemplate):
         """Refresh from DB if possible and return current state."""
         self.refresh()
         return self.status.state or RunStates.unknown
+
+    def abort(self):
+        """Abort this run (reference model.py RunObject.abort)."""
+        from .db import get_run_db
+
+        get_run_db().abort_run(self.metadata.uid,
+                               project=self.metadata.project,
+                               iter=self.metadata.iteration or 0)
+        return self.refresh()
 
     def refresh(self):
         from .db import get_run_db

@@ -59,6 +59,13 @@ class SecretsStore:
     def to_serial(self) -> list:
         return [{"kind": "inline", "source": dict(self._secrets)}]
 
+    def to_dict(self, struct: dict = None) -> dict:
+        """Serialize into a run-spec dict (reference secrets.py
+        to_dict — populates spec.secret_sources)."""
+        struct = struct if struct is not None else {}
+        struct["secret_sources"] = self.to_serial()
+        return struct
+
     def items(self):
         return dict(self._secrets)
 

@@ -379,3 +379,100 @@ class TestUtilsHelpers:
         assert retry_until_successful(0.01, 5, None, False,
                                       flaky) == "done"
         assert calls["n"] == 3
+
+
+class TestChildContext:
+    """get_child_context / update_child_iterations / mark_as_best
+    (reference execution.py:223-291)."""
+
+    def _parent(self, tmp_path):
+        import mlrun_amd.db as db_mod
+        from mlrun_amd.db.sqldb import SQLRunDB
+        from mlrun_amd.execution import MLClientCtx
+
+        db = SQLRunDB(str(tmp_path / "cc.db"))
+        db_mod.set_run_db(db)
+        ctx = MLClientCtx.from_dict(
+            {"metadata": {"name": "parent", "project": "p"},
+             "spec": {"parameters": {"base": 1}}}, rundb=db)
+        return ctx, db
+
+    def test_children_and_best(self, tmp_path):
+        import mlrun_amd.db as db_mod
+
+        ctx, db = self._parent(tmp_path)
+        try:
+            best_acc = 0
+            for i, lr in enumerate([0.1, 0.2, 0.3]):
+                with ctx.get_child_context(lr=lr) as child:
+                    acc = 1 - abs(lr - 0.2)  # best at 0.2
+                    child.log_result("accuracy", acc)
+                    if acc > best_acc:
+                        child.mark_as_best()
+                        best_acc = acc
+                    assert child.get_param("lr") == lr
+                    assert child._iteration == i + 1
+            ctx.update_child_iterations(best_run=2)
+            assert ctx._results["best_iteration"] == 2
+            assert ctx._results["accuracy"] == 1.0
+            rows = ctx._iteration_results
+            assert rows[0][:2] == ["state", "iter"]
+            assert "param.lr" in rows[0]
+            assert len(rows) == 4
+        finally:
+            db_mod._run_db = None
+            db_mod._run_db_pinned = False
+
+    def test_with_parent_params_and_nesting_guard(self, tmp_path):
+        import pytest as _pytest
+
+        import mlrun_amd.db as db_mod
+        from mlrun_amd.errors import MLRunInvalidArgumentError
+
+        ctx, db = self._parent(tmp_path)
+        try:
+            child = ctx.get_child_context(with_parent_params=True,
+                                          extra=2)
+            assert child.get_param("base") == 1
+            assert child.get_param("extra") == 2
+            with _pytest.raises(MLRunInvalidArgumentError):
+                child.get_child_context()
+        finally:
+            db_mod._run_db = None
+            db_mod._run_db_pinned = False
+
+    def test_child_error_marks_failed(self, tmp_path):
+        import mlrun_amd.db as db_mod
+
+        ctx, db = self._parent(tmp_path)
+        try:
+            with pytest.raises(ValueError):
+                with ctx.get_child_context(lr=1) as child:
+                    raise ValueError("boom")
+            assert child._state == "error"
+        finally:
+            db_mod._run_db = None
+            db_mod._run_db_pinned = False
+
+
+class TestRunTemplateParamFile:
+    def test_with_param_file(self):
+        from mlrun_amd.model import new_task
+
+        task = new_task("grid").with_param_file(
+            "/tmp/params.json", selector="max.accuracy",
+            strategy="grid")
+        opts = task.spec.hyper_param_options
+        assert opts.param_file == "/tmp/params.json"
+        assert opts.selector == "max.accuracy"
+        assert opts.strategy == "grid"
+
+
+class TestSecretsToDict:
+    def test_to_dict(self):
+        from mlrun_amd.secrets import SecretsStore
+
+        store = SecretsStore()
+        store.add_source("inline", {"K": "V"})
+        struct = store.to_dict()
+        assert struct["secret_sources"][0]["source"]["K"] == "V"
