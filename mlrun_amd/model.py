@@ -12,6 +12,7 @@ HyperParamOptions :856), re-designed without k8s-specific fields.
 """
 
 import inspect
+import os
 import time
 import typing
 import uuid
@@ -589,3 +590,229 @@ def new_task(name=None, project=None, handler=None, params=None, hyper_params=No
 
 def generate_uid() -> str:
     return uuid.uuid4().hex
+
+
+class ObjectDict:
+    """Dict of typed child objects keyed by name, kind-dispatched
+    (reference model.py:272) — used by router routes & graph steps."""
+
+    kind = "object_dict"
+
+    def __init__(self, classes_map: dict, default_kind: str = ""):
+        self._children: dict = {}
+        self._default_kind = default_kind
+        self._classes_map = classes_map
+
+    def values(self):
+        return self._children.values()
+
+    def keys(self):
+        return self._children.keys()
+
+    def items(self):
+        return self._children.items()
+
+    def __len__(self):
+        return len(self._children)
+
+    def __iter__(self):
+        yield from self._children.keys()
+
+    def __getitem__(self, name):
+        return self._children[name]
+
+    def __setitem__(self, name, item):
+        self._children[name] = self._resolve(name, item)
+
+    def __delitem__(self, name):
+        del self._children[name]
+
+    def update(self, key, item):
+        self._children[key] = self._resolve(key, item)
+        return self._children[key]
+
+    def to_dict(self) -> dict:
+        return {name: child.to_dict()
+                for name, child in self._children.items()}
+
+    @classmethod
+    def from_dict(cls, classes_map: dict, children: dict,
+                  default_kind: str = ""):
+        new_obj = cls(classes_map, default_kind)
+        for name, child in (children or {}).items():
+            new_obj[name] = child
+        return new_obj
+
+    def _resolve(self, name, item):
+        if hasattr(item, "to_dict") and not isinstance(item, dict):
+            return item
+        if isinstance(item, dict):
+            kind = item.get("kind", self._default_kind)
+            cls = self._classes_map.get(kind)
+            if cls is None:
+                raise MLRunInvalidArgumentError(
+                    f"illegal object kind {kind!r} for child {name}")
+            if hasattr(cls, "from_dict"):
+                child = cls.from_dict(item)
+            else:
+                from .serving.states import step_from_dict
+
+                child = step_from_dict(item)
+            if hasattr(child, "name"):
+                child.name = name
+            return child
+        raise MLRunInvalidArgumentError(
+            f"child {name} must be an object or dict")
+
+
+class ObjectList:
+    """Ordered list of typed child objects addressable by name
+    (reference model.py:361) — e.g. feature/entity lists."""
+
+    def __init__(self, child_class):
+        self._children: dict = {}
+        self._child_class = child_class
+
+    def values(self):
+        return self._children.values()
+
+    def keys(self):
+        return self._children.keys()
+
+    def items(self):
+        return self._children.items()
+
+    def __len__(self):
+        return len(self._children)
+
+    def __iter__(self):
+        yield from self._children.values()
+
+    def __getitem__(self, name):
+        if isinstance(name, int):
+            return list(self._children.values())[name]
+        return self._children[name]
+
+    def __setitem__(self, key, item):
+        self.update(item, key)
+
+    def __delitem__(self, key):
+        del self._children[key]
+
+    def __contains__(self, key):
+        return key in self._children
+
+    def update(self, item, key=None):
+        if isinstance(item, dict):
+            item = self._child_class.from_dict(item)
+        key = key or getattr(item, "name", None)
+        if key is None:
+            raise MLRunInvalidArgumentError("child item has no name")
+        self._children[key] = item
+        return item
+
+    def to_dict(self) -> list:
+        return [child.to_dict() for child in self._children.values()]
+
+    @classmethod
+    def from_list(cls, child_class, children: list = None):
+        new_obj = cls(child_class)
+        for child in children or []:
+            new_obj.update(child)
+        return new_obj
+
+
+class Credentials(ModelObj):
+    """Function/run credentials spec (reference model.py:427)."""
+
+    generate_access_key = "$generate"
+    secret_reference_prefix = "$ref:"
+
+    def __init__(self, access_key: str = None):
+        self.access_key = access_key
+
+
+class ImageBuilder(ModelObj):
+    """Function build spec (reference model.py:485): source, base
+    image, commands, requirements."""
+
+    def __init__(self, functionSourceCode=None, source=None, image=None,  # noqa: N803
+                 base_image=None, commands=None, extra=None, secret=None,
+                 code_origin=None, registry=None, load_source_on_run=None,
+                 origin_filename=None, with_mlrun=None, auto_build=None,
+                 build_pod=None, requirements: list = None,
+                 extra_args=None, source_code_target_dir=None):
+        self.functionSourceCode = functionSourceCode  # noqa: N803
+        self.source = source
+        self.image = image
+        self.base_image = base_image
+        self.commands = commands or []
+        self.extra = extra
+        self.secret = secret
+        self.code_origin = code_origin
+        self.registry = registry
+        self.load_source_on_run = load_source_on_run
+        self.origin_filename = origin_filename
+        self.with_mlrun = with_mlrun
+        self.auto_build = auto_build
+        self.requirements = requirements or []
+        self.extra_args = extra_args
+        self.source_code_target_dir = source_code_target_dir
+
+
+class EntrypointParam(ModelObj):
+    """One handler parameter's doc/type/default (reference
+    model.py:1865)."""
+
+    def __init__(self, name="", type=None, default=None, doc="",
+                 required=None, choices: list = None):
+        self.name = name
+        self.type = type
+        self.default = default
+        self.doc = doc
+        self.required = required
+        self.choices = choices
+
+
+class FunctionEntrypoint(ModelObj):
+    """One handler's signature summary (reference model.py:1883)."""
+
+    def __init__(self, name="", doc="", parameters=None, outputs=None,
+                 lineno=-1, has_varargs=None, has_kwargs=None):
+        self.name = name
+        self.doc = doc
+        self.parameters = parameters or []
+        self.outputs = outputs or []
+        self.lineno = lineno
+        self.has_varargs = has_varargs
+        self.has_kwargs = has_kwargs
+
+
+class TargetPathObject:
+    """Target path with optional {run_id} templating (reference
+    model.py:1983)."""
+
+    def __init__(self, base_path=None, run_id=None,
+                 is_single_file=False):
+        self.run_id = run_id
+        self.full_path_template = base_path
+        self.is_single_file = is_single_file
+        if run_id is not None and "{run_id}" not in (base_path or ""):
+            if not is_single_file:
+                self.full_path_template = os.path.join(
+                    base_path or "", "{run_id}") + "/"
+            else:
+                directory, fname = os.path.split(base_path or "")
+                self.full_path_template = os.path.join(
+                    directory, "{run_id}", fname)
+
+    def get_templated_path(self) -> str:
+        return self.full_path_template
+
+    def get_absolute_path(self, project_name: str = None) -> str:
+        path = self.full_path_template or ""
+        if self.run_id is not None:
+            path = path.replace("{run_id}", str(self.run_id))
+        if project_name:
+            path = path.replace("{project}", project_name)
+        return path

@@ -476,3 +476,56 @@ class TestSecretsToDict:
         store.add_source("inline", {"K": "V"})
         struct = store.to_dict()
         assert struct["secret_sources"][0]["source"]["K"] == "V"
+
+
+class TestModelAuxClasses:
+    """Reference model.py aux objects (ObjectDict/ObjectList,
+    Credentials, ImageBuilder, entrypoints, TargetPathObject)."""
+
+    def test_object_list(self):
+        from mlrun_amd.feature_store.feature_set import Feature
+        from mlrun_amd.model import ObjectList
+
+        features = ObjectList.from_list(
+            Feature, [{"name": "a", "value_type": "float"},
+                      {"name": "b", "value_type": "str"}])
+        assert len(features) == 2
+        assert features["a"].value_type == "float"
+        assert features[1].name == "b"
+        assert [f.name for f in features] == ["a", "b"]
+        assert features.to_dict()[0]["name"] == "a"
+
+    def test_object_dict_kind_dispatch(self):
+        from mlrun_amd.model import ObjectDict
+        from mlrun_amd.serving.states import TaskStep, classes_map
+
+        steps = ObjectDict(classes_map, default_kind="task")
+        steps["s1"] = {"kind": "task", "handler": "h"}
+        assert isinstance(steps["s1"], TaskStep)
+        assert steps.to_dict()["s1"]["handler"] == "h"
+
+    def test_target_path_object(self):
+        from mlrun_amd.model import TargetPathObject
+
+        tp = TargetPathObject("/data/out", run_id="r1")
+        assert "{run_id}" in tp.get_templated_path()
+        assert tp.get_absolute_path() == "/data/out/r1/"
+        tpf = TargetPathObject("/data/out/f.pq", run_id="r2",
+                               is_single_file=True)
+        assert tpf.get_absolute_path() == "/data/out/r2/f.pq"
+
+    def test_builder_and_entrypoints(self):
+        from mlrun_amd.model import (Credentials, EntrypointParam,
+                                     FunctionEntrypoint, ImageBuilder)
+
+        builder = ImageBuilder(base_image="rocm/pytorch",
+                               requirements=["einops"])
+        back = ImageBuilder.from_dict(builder.to_dict())
+        assert back.base_image == "rocm/pytorch"
+        assert back.requirements == ["einops"]
+        entry = FunctionEntrypoint(
+            name="train", doc="trains",
+            parameters=[EntrypointParam("lr", type="float",
+                                        default=0.1).to_dict()])
+        assert entry.to_dict()["parameters"][0]["name"] == "lr"
+        assert Credentials.generate_access_key == "$generate"
