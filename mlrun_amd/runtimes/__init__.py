@@ -147,3 +147,8 @@ def new_v2_model_server(name: str, model_class: str, models: dict = None,
     return new_model_server(name, model_class=model_class, models=models,
                             filename=filename, protocol=protocol or "v2",
                             image=image, **kwargs)
+from .api_gateway import (  # noqa: F401,E402
+    APIGateway,
+    APIGatewayMetadata,
+    APIGatewaySpec,
+)
