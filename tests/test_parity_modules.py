@@ -909,7 +909,7 @@ class TestAPIGateway:
                 APIGatewaySpec(functions=["f1"], project="p",
                                host=f"http://127.0.0.1:{port}"))
             gw.save()
-            assert db.get_api_gateway("gw1", "p")["spec"][
+            assert db.get_api_gateway("p", "gw1")["spec"][
                 "functions"] == ["f1"]
             resp = gw.invoke(body={"x": 1})
             assert resp.json()["echo"] == {"x": 1}
