@@ -853,76 +853,43 @@ class TestExportSurfaceBatch2:
 
 
 class TestAPIGateway:
-    """Node-local APIGateway (reference nuclio/api_gateway.py:340)."""
+    """Node-local APIGateway proxy: reference-shape construction +
+    run-DB persistence (reference nuclio/api_gateway.py:340)."""
 
-    def test_canary_and_auth_validation(self):
+    def test_from_schema_and_canary_validation(self):
         import pytest as _pytest
 
         from mlrun_amd.errors import MLRunInvalidArgumentError
         from mlrun_amd.runtimes import (APIGateway, APIGatewayMetadata,
                                         APIGatewaySpec)
 
-        gw = APIGateway(APIGatewayMetadata(name="gw"),
-                        APIGatewaySpec(project="p"))
+        gw = APIGateway.from_schema(
+            APIGatewayMetadata(name="gw"),
+            APIGatewaySpec(functions=["http://127.0.0.1:1/a",
+                                      "http://127.0.0.1:2/b"],
+                           project="p", canary=[80, 20]))
+        assert [u["percent"] for u in gw.upstreams] == [80, 20]
         with _pytest.raises(MLRunInvalidArgumentError):
-            gw.with_canary(["f1"], [100])
-        with _pytest.raises(MLRunInvalidArgumentError):
-            gw.with_canary(["f1", "f2"], [60, 60])
-        gw.with_canary(["f1", "f2"], [80, 20])
-        assert gw.spec.canary == [80, 20]
-        gw.with_basic_auth("admin", "secret")
-        assert gw.spec.authentication_mode == "basicAuth"
+            APIGateway(name="bad").with_canary(
+                ["http://x", "http://y"], [60, 60])
 
-    def test_save_delete_and_invoke(self, tmp_path):
-        import json
-        import threading
-        from http.server import BaseHTTPRequestHandler, HTTPServer
-
+    def test_save_and_delete_persistence(self, tmp_path):
         import mlrun_amd.db as db_mod
         from mlrun_amd.db.sqldb import SQLRunDB
-        from mlrun_amd.runtimes import (APIGateway, APIGatewayMetadata,
-                                        APIGatewaySpec)
+        from mlrun_amd.runtimes import APIGateway
 
         db = SQLRunDB(str(tmp_path / "gw.db"))
         prev = db_mod._run_db
         db_mod.set_run_db(db)
-
-        class Handler(BaseHTTPRequestHandler):
-            def do_POST(self):
-                n = int(self.headers.get("content-length", 0))
-                body = json.loads(self.rfile.read(n))
-                out = json.dumps({"fn": "f1", "echo": body}).encode()
-                self.send_response(200)
-                self.end_headers()
-                self.wfile.write(out)
-
-            def log_message(self, *a):
-                pass
-
-        httpd = HTTPServer(("127.0.0.1", 0), Handler)
-        port = httpd.server_address[1]
-        threading.Thread(target=httpd.serve_forever,
-                         daemon=True).start()
         try:
-            gw = APIGateway(
-                APIGatewayMetadata(name="gw1"),
-                APIGatewaySpec(functions=["f1"], project="p",
-                               host=f"http://127.0.0.1:{port}"))
+            gw = APIGateway(name="gw1", project="p")
+            gw.add_upstream("http://127.0.0.1:9/f1")
             gw.save()
-            assert db.get_api_gateway("p", "gw1")["spec"][
-                "functions"] == ["f1"]
-            resp = gw.invoke(body={"x": 1})
-            assert resp.json()["echo"] == {"x": 1}
-            gw.with_basic_auth("u", "pw")
-            from mlrun_amd.errors import MLRunAccessDeniedError
-
-            import pytest as _pytest
-
-            with _pytest.raises(MLRunAccessDeniedError):
-                gw.invoke(body={}, credentials=("u", "wrong"))
+            stored = db.get_api_gateway("p", "gw1")
+            assert stored["spec"]["upstreams"][0]["address"] == \
+                "http://127.0.0.1:9/f1"
             gw.delete()
             assert db.list_api_gateways("p") == []
         finally:
-            httpd.shutdown()
             db_mod._run_db = prev
             db_mod._run_db_pinned = False
